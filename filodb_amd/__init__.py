@@ -1,0 +1,286 @@
+"""filodb_amd — Python wrapper over the C-ABI of the MI355X-native FiloDB
+chunk-scan + range-vector engine (include/filodb_amd.h).
+
+The wrapper is plumbing only: chunk building, dataset upload and query launch all
+happen in the native library. PyTorch is used solely for device-tensor interop
+(RCCL all-reduce of aggregated grids in multi-GPU runs).
+"""
+import ctypes
+import os
+
+import numpy as np
+
+_ROOT = os.path.dirname(os.path.abspath(__file__))
+LIB_PATH = os.path.join(_ROOT, "libfilodb_amd.so")
+
+# function ids (include/filodb_amd.h; dispatch table RangeFunction.scala:294-410)
+FN_RATE, FN_INCREASE, FN_DELTA = 0, 1, 2
+FN_SUM_OVER_TIME, FN_COUNT_OVER_TIME, FN_AVG_OVER_TIME = 3, 4, 5
+FN_MIN_OVER_TIME, FN_MAX_OVER_TIME = 6, 7
+FN_STDDEV_OVER_TIME, FN_STDVAR_OVER_TIME, FN_CHANGES = 8, 9, 10
+# aggregation ids (RowAggregator implementations)
+AGG_NONE, AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX, AGG_AVG = 0, 1, 2, 3, 4, 5
+# column kinds
+COL_GAUGE, COL_COUNTER = 0, 1
+
+_c_double_p = ctypes.POINTER(ctypes.c_double)
+
+
+class Query(ctypes.Structure):
+    _fields_ = [
+        ("start", ctypes.c_int64), ("step", ctypes.c_int64),
+        ("end", ctypes.c_int64), ("window", ctypes.c_int64),
+        ("func_id", ctypes.c_int32), ("agg_id", ctypes.c_int32),
+        ("num_groups", ctypes.c_int32), ("_pad", ctypes.c_int32),
+    ]
+
+    @property
+    def num_windows(self):
+        return (self.end - self.start) // self.step + 1
+
+
+class ChunkInfo(ctypes.Structure):
+    _fields_ = [
+        ("ts_vec", ctypes.POINTER(ctypes.c_uint8)),
+        ("val_vec", ctypes.POINTER(ctypes.c_uint8)),
+        ("num_rows", ctypes.c_int32),
+        ("start_time", ctypes.c_int64), ("end_time", ctypes.c_int64),
+        ("ts_vec_len", ctypes.c_int32), ("val_vec_len", ctypes.c_int32),
+    ]
+
+
+class View(ctypes.Structure):
+    _fields_ = [
+        ("blob", ctypes.POINTER(ctypes.c_uint8)), ("blob_len", ctypes.c_int64),
+        ("dir", ctypes.c_void_p), ("num_chunks", ctypes.c_int64),
+        ("series_first", ctypes.POINTER(ctypes.c_int32)),
+        ("series_nchunks", ctypes.POINTER(ctypes.c_int32)),
+        ("group_ids", ctypes.POINTER(ctypes.c_int32)),
+        ("num_series", ctypes.c_int32), ("_pad", ctypes.c_int32),
+    ]
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        if not os.path.exists(LIB_PATH):
+            raise RuntimeError(
+                f"{LIB_PATH} not built — run `python __graft_entry__.py build` "
+                "or __graft_entry__.build()")
+        L = ctypes.CDLL(LIB_PATH)
+        L.fdb_last_error.restype = ctypes.c_char_p
+        L.fdb_store_create.restype = ctypes.c_void_p
+        L.fdb_store_create.argtypes = [ctypes.c_int64]
+        L.fdb_store_destroy.argtypes = [ctypes.c_void_p]
+        L.fdb_store_add_series.argtypes = [ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32]
+        L.fdb_series_append.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                        ctypes.POINTER(ctypes.c_int64), _c_double_p,
+                                        ctypes.c_int32]
+        L.fdb_series_cut_chunk.argtypes = [ctypes.c_void_p, ctypes.c_int32]
+        L.fdb_store_set_max_rows.argtypes = [ctypes.c_void_p, ctypes.c_int32]
+        L.fdb_store_seal.argtypes = [ctypes.c_void_p]
+        L.fdb_store_num_series.argtypes = [ctypes.c_void_p]
+        L.fdb_series_num_chunks.argtypes = [ctypes.c_void_p, ctypes.c_int32]
+        L.fdb_chunk_get.argtypes = [ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32,
+                                    ctypes.POINTER(ChunkInfo)]
+        L.fdb_store_view.argtypes = [ctypes.c_void_p, ctypes.POINTER(View)]
+        L.fdb_nibblepack_pack8.argtypes = [ctypes.POINTER(ctypes.c_int64),
+                                           ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32]
+        L.fdb_nibblepack_unpack8.argtypes = [ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32,
+                                             ctypes.POINTER(ctypes.c_int64),
+                                             ctypes.POINTER(ctypes.c_int32)]
+        L.fdb_nibblepack_pack_delta.argtypes = [ctypes.POINTER(ctypes.c_int64), ctypes.c_int32,
+                                                ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32]
+        L.fdb_nibblepack_pack_doubles.argtypes = [_c_double_p, ctypes.c_int32,
+                                                  ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32]
+        L.fdb_engine_create.restype = ctypes.c_void_p
+        L.fdb_engine_create.argtypes = [ctypes.c_int32]
+        L.fdb_engine_destroy.argtypes = [ctypes.c_void_p]
+        L.fdb_engine_synchronize.argtypes = [ctypes.c_void_p]
+        L.fdb_dataset_upload.restype = ctypes.c_void_p
+        L.fdb_dataset_upload.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+        L.fdb_dataset_destroy.argtypes = [ctypes.c_void_p]
+        L.fdb_dataset_bytes.restype = ctypes.c_int64
+        L.fdb_dataset_bytes.argtypes = [ctypes.c_void_p]
+        L.fdb_dataset_samples.restype = ctypes.c_int64
+        L.fdb_dataset_samples.argtypes = [ctypes.c_void_p]
+        L.fdb_query_exec.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
+                                     _c_double_p, _c_double_p, ctypes.c_int32]
+        L.fdb_query_bench.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
+                                      _c_double_p, _c_double_p, ctypes.c_int32,
+                                      ctypes.c_int32, ctypes.c_int32, _c_double_p]
+        _lib = L
+    return _lib
+
+
+def _err():
+    return lib().fdb_last_error().decode()
+
+
+def _check(rc, what):
+    if rc < 0:
+        raise RuntimeError(f"{what} failed (rc={rc}): {_err()}")
+    return rc
+
+
+def _as_f64_ptr(x):
+    if x is None:
+        return None
+    if isinstance(x, np.ndarray):
+        assert x.dtype == np.float64 and x.flags["C_CONTIGUOUS"]
+        return x.ctypes.data_as(_c_double_p)
+    # torch tensor (device pointer) — duck-typed to avoid importing torch here
+    return ctypes.cast(ctypes.c_void_p(x.data_ptr()), _c_double_p)
+
+
+class ChunkStore:
+    """Host-side chunk store: encodes samples into the reference's frozen chunk
+    format (TimeSeriesPartition.switchBuffers(encode=true) equivalent)."""
+
+    def __init__(self, expected_series=0):
+        self._h = lib().fdb_store_create(expected_series)
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            lib().fdb_store_destroy(self._h)
+            self._h = None
+
+    def set_max_rows(self, n):
+        _check(lib().fdb_store_set_max_rows(self._h, n), "set_max_rows")
+
+    def add_series(self, group_id=0, kind=COL_GAUGE):
+        return _check(lib().fdb_store_add_series(self._h, group_id, kind), "add_series")
+
+    def append(self, sid, ts, vals):
+        ts = np.ascontiguousarray(ts, dtype=np.int64)
+        vals = np.ascontiguousarray(vals, dtype=np.float64)
+        assert len(ts) == len(vals)
+        _check(lib().fdb_series_append(
+            self._h, sid, ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+            vals.ctypes.data_as(_c_double_p), len(ts)), "append")
+
+    def cut_chunk(self, sid):
+        _check(lib().fdb_series_cut_chunk(self._h, sid), "cut_chunk")
+
+    def seal(self):
+        _check(lib().fdb_store_seal(self._h), "seal")
+
+    @property
+    def num_series(self):
+        return lib().fdb_store_num_series(self._h)
+
+    def num_chunks(self, sid):
+        return _check(lib().fdb_series_num_chunks(self._h, sid), "num_chunks")
+
+    def chunk(self, sid, idx):
+        """Returns (ts_bytes, val_bytes, num_rows, start_time, end_time)."""
+        info = ChunkInfo()
+        _check(lib().fdb_chunk_get(self._h, sid, idx, ctypes.byref(info)), "chunk_get")
+        tsb = ctypes.string_at(info.ts_vec, info.ts_vec_len)
+        vab = ctypes.string_at(info.val_vec, info.val_vec_len)
+        return tsb, vab, info.num_rows, info.start_time, info.end_time
+
+    def view(self):
+        v = View()
+        _check(lib().fdb_store_view(self._h, ctypes.byref(v)), "store_view")
+        return v
+
+
+class Engine:
+    """GPU engine (fails without a HIP device — no CPU fallback)."""
+
+    def __init__(self, device=0):
+        h = lib().fdb_engine_create(device)
+        if not h:
+            raise RuntimeError(f"fdb_engine_create failed: {_err()}")
+        self._h = h
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            lib().fdb_engine_destroy(self._h)
+            self._h = None
+
+    def upload(self, store: ChunkStore):
+        d = lib().fdb_dataset_upload(self._h, store._h)
+        if not d:
+            raise RuntimeError(f"fdb_dataset_upload failed: {_err()}")
+        return Dataset(d)
+
+    def query(self, dataset, q: Query, out=None, out_counts=None, on_device=False):
+        """Executes one (shard, query). Returns `out` (allocated as numpy when None
+        and on_device is False)."""
+        if out is None:
+            assert not on_device
+            n = dataset_out_len(dataset, q)
+            out = np.empty(n, dtype=np.float64)
+        rc = lib().fdb_query_exec(self._h, dataset._h, ctypes.byref(q),
+                                  _as_f64_ptr(out), _as_f64_ptr(out_counts),
+                                  1 if on_device else 0)
+        _check(rc, "query_exec")
+        return out
+
+    def bench(self, dataset, q: Query, out, out_counts=None, on_device=False,
+              warmup=2, iters=10):
+        ms = ctypes.c_double()
+        rc = lib().fdb_query_bench(self._h, dataset._h, ctypes.byref(q),
+                                   _as_f64_ptr(out), _as_f64_ptr(out_counts),
+                                   1 if on_device else 0, warmup, iters,
+                                   ctypes.byref(ms))
+        _check(rc, "query_bench")
+        return ms.value
+
+    def synchronize(self):
+        _check(lib().fdb_engine_synchronize(self._h), "synchronize")
+
+
+class Dataset:
+    def __init__(self, h):
+        self._h = h
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            lib().fdb_dataset_destroy(self._h)
+            self._h = None
+
+    @property
+    def payload_bytes(self):
+        return lib().fdb_dataset_bytes(self._h)
+
+    @property
+    def total_samples(self):
+        return lib().fdb_dataset_samples(self._h)
+
+
+def dataset_out_len(dataset, q: Query):
+    if q.agg_id == AGG_NONE:
+        # series count is not stored on Dataset; caller usually knows it
+        raise ValueError("pass an explicit out buffer for AGG_NONE queries")
+    return q.num_groups * q.num_windows
+
+
+def make_query(start, step, end, window, func_id, agg_id=AGG_NONE, num_groups=0):
+    q = Query()
+    q.start, q.step, q.end, q.window = start, step, end, window
+    q.func_id, q.agg_id, q.num_groups = func_id, agg_id, num_groups
+    return q
+
+
+def nibblepack_pack8(vals8):
+    a = np.ascontiguousarray(vals8, dtype=np.int64)
+    assert len(a) == 8
+    out = (ctypes.c_uint8 * 64)()
+    n = _check(lib().fdb_nibblepack_pack8(
+        a.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)), out, 64), "pack8")
+    return bytes(out[:n])
+
+
+def nibblepack_unpack8(data):
+    buf = (ctypes.c_uint8 * len(data)).from_buffer_copy(data)
+    out = (ctypes.c_int64 * 8)()
+    consumed = ctypes.c_int32()
+    _check(lib().fdb_nibblepack_unpack8(buf, len(data), out, ctypes.byref(consumed)),
+           "unpack8")
+    return list(out), consumed.value

@@ -1,0 +1,517 @@
+// Host-side chunk store + encoder: builds the reference's frozen BinaryVector chunk
+// format from raw (timestamp, value) samples.
+//
+// Restates (semantics, not code) the reference ingest/encode path:
+//   TimeSeriesPartition.ingest → switchBuffers(encode=true) → optimizeAll
+//     (core/.../memstore/TimeSeriesPartition.scala:130)
+//   timestamps: DeltaDeltaVector.fromLongVector(approxConst=true)
+//     (core/.../format/vectors/LongBinaryVector.scala:333-341,
+//      DeltaDeltaVector.scala:63-135)
+//   doubles: DDV if all integral else raw f64; counter drop bit
+//     (core/.../format/vectors/DoubleVector.scala:86-96,457-476)
+//   NibblePack primitives (core/.../format/NibblePack.scala:16-183)
+//
+// This file is product host code (the builder feeds the GPU engine); the query-time
+// DECODE restatement lives in oracle/ and is test-only.
+
+#include <cstdint>
+#include <cstring>
+#include <cmath>
+#include <cstdio>
+#include <cstdarg>
+#include <vector>
+#include <string>
+#include <limits>
+
+#include "chunk_format.h"
+#include "../../include/filodb_amd.h"
+
+// ---------------------------------------------------------------------------
+// error handling
+// ---------------------------------------------------------------------------
+static thread_local std::string g_last_error;
+extern "C" const char* fdb_last_error(void) { return g_last_error.c_str(); }
+void fdb_set_error(const char* fmt, ...) {
+  char buf[512];
+  va_list ap; va_start(ap, fmt);
+  vsnprintf(buf, sizeof buf, fmt, ap);
+  va_end(ap);
+  g_last_error = buf;
+}
+
+// ---------------------------------------------------------------------------
+// little-endian byte emit helpers (x86/amd64 host: plain stores)
+// ---------------------------------------------------------------------------
+using bytes = std::vector<uint8_t>;
+static inline void put_u8(bytes& b, size_t off, uint8_t v)  { if (off >= b.size()) b.resize(off + 1); b[off] = v; }
+static inline void put_u16(bytes& b, size_t off, uint16_t v){ if (off + 2 > b.size()) b.resize(off + 2); memcpy(&b[off], &v, 2); }
+static inline void put_u32(bytes& b, size_t off, uint32_t v){ if (off + 4 > b.size()) b.resize(off + 4); memcpy(&b[off], &v, 4); }
+static inline void put_u64(bytes& b, size_t off, uint64_t v){ if (off + 8 > b.size()) b.resize(off + 8); memcpy(&b[off], &v, 8); }
+
+// ---------------------------------------------------------------------------
+// NibblePack (NibblePack.scala:108-183 pack8/packUniversal; :16-98 outer packers)
+// ---------------------------------------------------------------------------
+// Packs 8 longs; returns new write position.
+static size_t np_pack8(const int64_t in[8], bytes& buf, size_t pos) {
+  int bitmask = 0;
+  for (int i = 0; i < 8; i++) if (in[i] != 0) bitmask |= 1 << i;
+  put_u8(buf, pos++, (uint8_t)bitmask);
+  if (bitmask == 0) return pos;
+
+  int minLeading = 64, minTrailing = 64;
+  for (int i = 0; i < 8; i++) {
+    uint64_t u = (uint64_t)in[i];
+    // numberOfLeading/TrailingZeros(0) == 64 in Java
+    int lz = u ? __builtin_clzll(u) : 64;
+    int tz = u ? __builtin_ctzll(u) : 64;
+    if (lz < minLeading)  minLeading  = lz;
+    if (tz < minTrailing) minTrailing = tz;
+  }
+  int trailingNibbles = minTrailing / 4;
+  int numNibbles = 16 - (minLeading / 4) - trailingNibbles;
+  put_u8(buf, pos++, (uint8_t)(((numNibbles - 1) << 4) | trailingNibbles));
+
+  // packUniversal (NibblePack.scala:143-183)
+  int trailingShift = trailingNibbles * 4;
+  int numBits = numNibbles * 4;
+  uint64_t outWord = 0;
+  int bitCursor = 0;
+  for (int i = 0; i < 8; i++) {
+    if (in[i] != 0) {
+      int remaining = 64 - bitCursor;
+      uint64_t shifted = (uint64_t)in[i] >> trailingShift;
+      outWord |= (bitCursor == 64 ? 0 : shifted << bitCursor);
+      if (remaining <= numBits) {
+        put_u64(buf, pos, outWord); pos += 8;
+        outWord = (remaining < numBits) ? (shifted >> remaining) : 0;
+      }
+      bitCursor = (bitCursor + numBits) % 64;
+    }
+  }
+  if (bitCursor > 0) {
+    // write remainder word but advance only the needed bytes
+    if (pos + 8 > buf.size()) buf.resize(pos + 8);
+    memcpy(&buf[pos], &outWord, 8);
+    pos += (bitCursor + 7) / 8;
+    buf.resize(pos);
+  }
+  return pos;
+}
+
+// packDelta (NibblePack.scala:37-63): positive nondecreasing longs as deltas.
+static size_t np_pack_delta(const int64_t* input, int n, bytes& buf, size_t pos) {
+  int64_t tmp[8] = {0};
+  int64_t last = 0;
+  int i = 0;
+  for (; i < n; i++) {
+    int64_t delta = (input[i] >= last) ? input[i] - last : 0;
+    last = input[i];
+    tmp[i % 8] = delta;
+    if (i % 8 == 7) pos = np_pack8(tmp, buf, pos);
+  }
+  if (i % 8 != 0) {
+    for (int j = i % 8; j < 8; j++) tmp[j] = 0;
+    pos = np_pack8(tmp, buf, pos);
+  }
+  return pos;
+}
+
+// packDoubles (NibblePack.scala:73-98): first double raw, rest XOR-encoded.
+static size_t np_pack_doubles(const double* input, int n, bytes& buf, size_t pos) {
+  uint64_t first; memcpy(&first, &input[0], 8);
+  put_u64(buf, pos, first); pos += 8;
+  int64_t tmp[8] = {0};
+  uint64_t last = first;
+  int i = 0;
+  for (; i < n - 1; i++) {
+    uint64_t bits; memcpy(&bits, &input[i + 1], 8);
+    tmp[i % 8] = (int64_t)(bits ^ last);
+    last = bits;
+    if (i % 8 == 7) pos = np_pack8(tmp, buf, pos);
+  }
+  if (i % 8 != 0) {
+    for (int j = i % 8; j < 8; j++) tmp[j] = 0;
+    pos = np_pack8(tmp, buf, pos);
+  }
+  return pos;
+}
+
+extern "C" int32_t fdb_nibblepack_pack8(const int64_t in[8], uint8_t* out, int32_t outcap) {
+  bytes b;
+  size_t n = np_pack8(in, b, 0);
+  if ((int32_t)n > outcap) { fdb_set_error("output buffer too small (%zu needed)", n); return FDB_ERR_BADARG; }
+  memcpy(out, b.data(), n);
+  return (int32_t)n;
+}
+extern "C" int32_t fdb_nibblepack_pack_delta(const int64_t* in, int32_t n, uint8_t* out, int32_t outcap) {
+  bytes b;
+  size_t len = np_pack_delta(in, n, b, 0);
+  if ((int32_t)len > outcap) { fdb_set_error("output buffer too small"); return FDB_ERR_BADARG; }
+  memcpy(out, b.data(), len);
+  return (int32_t)len;
+}
+extern "C" int32_t fdb_nibblepack_pack_doubles(const double* in, int32_t n, uint8_t* out, int32_t outcap) {
+  if (n <= 0) { fdb_set_error("packDoubles requires n>0"); return FDB_ERR_BADARG; }
+  bytes b;
+  size_t len = np_pack_doubles(in, n, b, 0);
+  if ((int32_t)len > outcap) { fdb_set_error("output buffer too small"); return FDB_ERR_BADARG; }
+  memcpy(out, b.data(), len);
+  return (int32_t)len;
+}
+
+// unpack8 (NibblePack.scala:395-447). Returns FDB_OK and *consumed, or error.
+extern "C" int32_t fdb_nibblepack_unpack8(const uint8_t* in, int32_t inlen, int64_t out[8],
+                                          int32_t* consumed) {
+  if (inlen < 1) { fdb_set_error("input too short"); return FDB_ERR_BADARG; }
+  uint8_t nonzeroMask = in[0];
+  if (nonzeroMask == 0) {
+    for (int i = 0; i < 8; i++) out[i] = 0;
+    *consumed = 1;
+    return FDB_OK;
+  }
+  if (inlen < 2) { fdb_set_error("input too short"); return FDB_ERR_BADARG; }
+  int numNibblesU8 = in[1] & 0xff;
+  int numBits = ((numNibblesU8 >> 4) + 1) * 4;
+  int trailingZeroes = (numNibblesU8 & 0x0f) * 4;
+  int totalBytes = 2 + (numBits * __builtin_popcount(nonzeroMask) + 7) / 8;
+  uint64_t mask = (numBits >= 64) ? ~0ULL : ((1ULL << numBits) - 1);
+  int bufIndex = 2;
+  int bitCursor = 0;
+
+  // readLong with zero-padding past the end (NibblePack.scala:458-470)
+  auto readLong = [&](int idx) -> uint64_t {
+    if (idx + 8 <= inlen) { uint64_t v; memcpy(&v, in + idx, 8); return v; }
+    uint64_t v = 0;
+    for (int i = 0; idx + i < inlen; i++) v |= (uint64_t)in[idx + i] << (8 * i);
+    return v;
+  };
+
+  uint64_t inWord = readLong(bufIndex);
+  bufIndex += 8;
+  for (int bit = 0; bit < 8; bit++) {
+    if (nonzeroMask & (1 << bit)) {
+      int remaining = 64 - bitCursor;
+      uint64_t shiftedIn = inWord >> bitCursor;
+      uint64_t outWord = shiftedIn & mask;
+      if (remaining <= numBits && bufIndex < totalBytes) {
+        if (bufIndex < inlen) {
+          inWord = readLong(bufIndex);
+          bufIndex += 8;
+          if (remaining < numBits) outWord |= (inWord << remaining) & mask;
+        } else {
+          fdb_set_error("InputTooShort");
+          return FDB_ERR_BADARG;
+        }
+      }
+      out[bit] = (int64_t)(outWord << trailingZeroes);
+      bitCursor = (bitCursor + numBits) % 64;
+    } else {
+      out[bit] = 0;
+    }
+  }
+  *consumed = totalBytes;
+  return FDB_OK;
+}
+
+// ---------------------------------------------------------------------------
+// DeltaDelta encoding (DeltaDeltaVector.scala:63-135)
+// ---------------------------------------------------------------------------
+// minMaxToNbitsSigned (IntBinaryVector.scala:161-177)
+static void nbits_signed(int32_t mn, int32_t mx, int* nbits, bool* sign) {
+  if (mn >= 0 && mx < 4)            { *nbits = 2;  *sign = false; }
+  else if (mn >= 0 && mx < 16)      { *nbits = 4;  *sign = false; }
+  else if (mn >= -128 && mx <= 127) { *nbits = 8;  *sign = true;  }
+  else if (mn >= 0 && mx < 256)     { *nbits = 8;  *sign = false; }
+  else if (mn >= -32768 && mx <= 32767) { *nbits = 16; *sign = true; }
+  else if (mn >= 0 && mx < 65536)   { *nbits = 16; *sign = false; }
+  else                              { *nbits = 32; *sign = true;  }
+}
+
+// Returns true and fills out if the long array is DDV-eligible; else false.
+// approx_const: accept ±250 band as const (timestamps; DeltaDeltaVector.scala:46,77).
+static bool encode_ddv_longs(const int64_t* in, int n, bool approx_const, bytes& out) {
+  if (n <= 2) return false;                      // fromLongVector eligibility :67
+  // getSlope (:112-115) — Long division truncates toward zero, same as C++
+  int64_t slope64 = (in[n - 1] - in[0]) / (n - 1);
+  if (slope64 >= INT32_MAX || slope64 <= INT32_MIN) return false;
+  int32_t slope = (int32_t)slope64;
+  // getDeltasMinMax (:118-130)
+  int64_t base = in[0];
+  int32_t mn = INT32_MAX, mx = INT32_MIN;
+  for (int i = 1; i < n; i++) {
+    base += slope;
+    int64_t d = in[i] - base;
+    if (d > INT32_MAX || d < INT32_MIN) return false;
+    if ((int32_t)d > mx) mx = (int32_t)d;
+    if ((int32_t)d < mn) mn = (int32_t)d;
+  }
+  int nbits; bool sign;
+  nbits_signed(mn, mx, &nbits, &sign);           // getNbitsSignedFromMinMax ≤32 always
+
+  if ((mn == 0 && mx == 0) ||
+      (approx_const && mn >= -FDB_DDV_MAX_APPROX_DELTA && mx <= FDB_DDV_MAX_APPROX_DELTA)) {
+    // const DDV (:89-106)
+    out.assign(FDB_DDVC_BYTES, 0);
+    put_u32(out, 0, 20);
+    put_u32(out, 4, FDB_WF_DDV_CONST);
+    put_u32(out, FDB_DDVC_OFF_NELEM, (uint32_t)n);
+    put_u64(out, FDB_DDVC_OFF_INIT, (uint64_t)in[0]);
+    put_u32(out, FDB_DDVC_OFF_SLOPE, (uint32_t)slope);
+    return true;
+  }
+
+  // packed DDV: outer header + inner int vector (DeltaDeltaVector.scala:138-146;
+  // inner header BinaryVector.scala:511-533, bit mechanics :560-580)
+  out.clear();
+  put_u32(out, 4, FDB_WF_DDV);
+  put_u64(out, FDB_DDV_OFF_INIT, (uint64_t)in[0]);
+  put_u32(out, FDB_DDV_OFF_SLOPE, (uint32_t)slope);
+  size_t inner = FDB_DDV_OFF_INNER;
+  put_u16(out, inner + 4, FDB_WF_INT_NOMASK);
+  put_u8(out, inner + 6, (uint8_t)((nbits & FDB_NBITS_MASK) | (sign ? FDB_SIGN_MASK : 0)));
+
+  size_t data = inner + FDB_PRIM_OFF_DATA;
+  int bitShift = 0;
+  size_t w = data;                 // current write byte
+  base = in[0] - slope;            // expected = initValue; addData: delta = v - expected
+  int64_t expected = in[0];
+  for (int i = 0; i < n; i++) {
+    int32_t v = (int32_t)(in[i] - expected);     // exact by minmax check above
+    expected += slope;
+    switch (nbits) {
+      case 32: put_u32(out, w, (uint32_t)v); w += 4; break;
+      case 16: put_u16(out, w, (uint16_t)(int16_t)v); w += 2; break;
+      case 8:  put_u8(out, w, (uint8_t)(int8_t)v); w += 1; break;
+      default: {  // 2 or 4, unsigned (IntBinaryVector.scala:84-105)
+        uint8_t orig = (bitShift == 0) ? 0 : out[w];
+        put_u8(out, w, (uint8_t)(orig | ((uint32_t)v << bitShift)));
+        bitShift = (bitShift + nbits) % 8;
+        if (bitShift == 0) w += 1;
+        break;
+      }
+    }
+  }
+  size_t dataBytes = (w - data) + ((bitShift != 0) ? 1 : 0);
+  put_u8(out, inner + 7, (uint8_t)bitShift);
+  put_u32(out, inner + 0, (uint32_t)(4 + dataBytes));          // inner length word
+  out.resize(data + dataBytes);
+  put_u32(out, 0, (uint32_t)(out.size() - 4));                 // outer length word
+  return true;
+}
+
+// raw 64-bit primitive vector (f64 or i64) with optional drop bit
+static void encode_raw64(const void* vals, int n, bool drop, bytes& out) {
+  out.clear();
+  put_u32(out, 0, (uint32_t)(4 + 8 * n));
+  put_u16(out, 4, FDB_WF_PRIM64);
+  put_u16(out, 6, (uint16_t)((64 | FDB_SIGN_MASK) | (drop ? FDB_DROP_MASK : 0)));
+  out.resize(FDB_PRIM_OFF_DATA + 8 * (size_t)n);
+  memcpy(out.data() + FDB_PRIM_OFF_DATA, vals, 8 * (size_t)n);
+}
+
+// DoubleVector.optimize (DoubleVector.scala:86-96): DDV when all values integral.
+static void encode_doubles(const double* vals, int n, bool drop, bytes& out) {
+  bool all_integral = true;
+  for (int i = 0; i < n; i++) {
+    double d = vals[i];
+    if (d > 9.2233720368547758e18 /* Long.MaxValue.toDouble */ || rint(d) != d) {
+      all_integral = false; break;      // NaN: rint(NaN)!=NaN → non-integral
+    }
+  }
+  if (all_integral && n > 2) {
+    std::vector<int64_t> longs((size_t)n);
+    for (int i = 0; i < n; i++) longs[(size_t)i] = (int64_t)vals[i];
+    if (encode_ddv_longs(longs.data(), n, /*approx_const=*/false, out)) {
+      if (drop) {                        // DoubleCounterAppender.optimize (:468-473)
+        uint16_t w; memcpy(&w, out.data() + 6, 2);
+        w |= FDB_DROP_MASK;              // bit 15 of u16 at +6 is unused by the
+        memcpy(out.data() + 6, &w, 2);   // 32-bit wireformat word's upper half
+      }
+      return;
+    }
+  }
+  encode_raw64(vals, n, drop, out);
+}
+
+// timestamps: approx-const DDV, else raw i64 (LongBinaryVector.scala:333-341)
+static void encode_timestamps(const int64_t* ts, int n, bytes& out) {
+  if (!encode_ddv_longs(ts, n, /*approx_const=*/true, out))
+    encode_raw64(ts, n, false, out);
+}
+
+// ---------------------------------------------------------------------------
+// chunk store
+// ---------------------------------------------------------------------------
+struct Chunk {
+  uint64_t ts_off, val_off;      // into store blob (valid after seal)
+  int32_t  num_rows;
+  int64_t  start_time, end_time;
+  bytes    ts_bytes, val_bytes;  // cleared after seal (moved into blob)
+};
+
+struct Series {
+  int32_t group_id = 0;
+  int32_t col_kind = FDB_COL_GAUGE;
+  std::vector<int64_t> buf_ts;
+  std::vector<double>  buf_vals;
+  bool     buf_drop = false;          // DoubleCounterAppender drop flag (per chunk)
+  double   buf_last = -1.7976931348623157e308;  // Double.MinValue
+  std::vector<Chunk> chunks;
+};
+
+struct fdb_store {
+  std::vector<Series> series;
+  int32_t max_rows = FDB_DEFAULT_MAX_ROWS;
+  bool sealed = false;
+  bytes blob;
+  std::vector<fdb_dir_entry_t> dir;
+  std::vector<int32_t> series_first, series_nchunks, group_ids;
+};
+
+extern "C" fdb_store_t* fdb_store_create(int64_t expected_series) {
+  auto* s = new fdb_store();
+  if (expected_series > 0) s->series.reserve((size_t)expected_series);
+  return s;
+}
+extern "C" void fdb_store_destroy(fdb_store_t* s) { delete s; }
+
+extern "C" int32_t fdb_store_set_max_rows(fdb_store_t* s, int32_t max_rows) {
+  if (max_rows < 1) { fdb_set_error("max_rows must be >= 1"); return FDB_ERR_BADARG; }
+  s->max_rows = max_rows;
+  return FDB_OK;
+}
+
+extern "C" int32_t fdb_store_add_series(fdb_store_t* s, int32_t group_id, int32_t col_kind) {
+  if (s->sealed) { fdb_set_error("store is sealed"); return FDB_ERR_BADARG; }
+  Series se;
+  se.group_id = group_id;
+  se.col_kind = col_kind;
+  s->series.push_back(std::move(se));
+  return (int32_t)s->series.size() - 1;
+}
+
+static int32_t cut_chunk(fdb_store_t* s, Series& se) {
+  int n = (int)se.buf_ts.size();
+  if (n == 0) return FDB_OK;
+  Chunk c;
+  c.num_rows = n;
+  c.start_time = se.buf_ts.front();
+  c.end_time = se.buf_ts.back();
+  encode_timestamps(se.buf_ts.data(), n, c.ts_bytes);
+  encode_doubles(se.buf_vals.data(), n, se.col_kind == FDB_COL_COUNTER && se.buf_drop,
+                 c.val_bytes);
+  se.chunks.push_back(std::move(c));
+  se.buf_ts.clear(); se.buf_vals.clear();
+  se.buf_drop = false;
+  se.buf_last = -1.7976931348623157e308;
+  return FDB_OK;
+}
+
+extern "C" int32_t fdb_series_append(fdb_store_t* s, int32_t sid,
+                                     const int64_t* ts, const double* vals, int32_t n) {
+  if (s->sealed) { fdb_set_error("store is sealed"); return FDB_ERR_BADARG; }
+  if (sid < 0 || sid >= (int32_t)s->series.size()) { fdb_set_error("bad series id %d", sid); return FDB_ERR_BADARG; }
+  Series& se = s->series[(size_t)sid];
+  for (int32_t i = 0; i < n; i++) {
+    if (!se.buf_ts.empty() && ts[i] < se.buf_ts.back()) {
+      fdb_set_error("timestamps must be nondecreasing (series %d)", sid);
+      return FDB_ERR_BADARG;
+    }
+    if (se.col_kind == FDB_COL_COUNTER) {
+      // DoubleCounterAppender.addData (DoubleVector.scala:460-466)
+      double v = vals[i];
+      if (std::isnan(v) || v < se.buf_last) se.buf_drop = true;
+      if (!std::isnan(v)) se.buf_last = v;
+    }
+    se.buf_ts.push_back(ts[i]);
+    se.buf_vals.push_back(vals[i]);
+    if ((int32_t)se.buf_ts.size() >= s->max_rows) cut_chunk(s, se);
+  }
+  return FDB_OK;
+}
+
+extern "C" int32_t fdb_series_cut_chunk(fdb_store_t* s, int32_t sid) {
+  if (sid < 0 || sid >= (int32_t)s->series.size()) { fdb_set_error("bad series id"); return FDB_ERR_BADARG; }
+  return cut_chunk(s, s->series[(size_t)sid]);
+}
+
+extern "C" int32_t fdb_store_seal(fdb_store_t* s) {
+  if (s->sealed) return FDB_OK;
+  size_t total = 0, nchunks = 0;
+  for (auto& se : s->series) {
+    cut_chunk(s, se);
+    for (auto& c : se.chunks) {
+      total += (c.ts_bytes.size() + 63 & ~size_t(63)) + (c.val_bytes.size() + 63 & ~size_t(63));
+      nchunks++;
+    }
+  }
+  s->blob.resize(total);
+  s->dir.reserve(nchunks);
+  s->series_first.reserve(s->series.size());
+  s->series_nchunks.reserve(s->series.size());
+  s->group_ids.reserve(s->series.size());
+  size_t off = 0;
+  for (auto& se : s->series) {
+    s->series_first.push_back((int32_t)s->dir.size());
+    s->series_nchunks.push_back((int32_t)se.chunks.size());
+    s->group_ids.push_back(se.group_id);
+    for (auto& c : se.chunks) {
+      c.ts_off = off;
+      memcpy(s->blob.data() + off, c.ts_bytes.data(), c.ts_bytes.size());
+      off = (off + c.ts_bytes.size() + 63) & ~size_t(63);
+      c.val_off = off;
+      memcpy(s->blob.data() + off, c.val_bytes.data(), c.val_bytes.size());
+      off = (off + c.val_bytes.size() + 63) & ~size_t(63);
+      fdb_dir_entry_t e;
+      e.ts_off = c.ts_off; e.val_off = c.val_off;
+      e.start_time = c.start_time; e.end_time = c.end_time;
+      e.num_rows = c.num_rows; e._pad = 0;
+      s->dir.push_back(e);
+    }
+  }
+  s->sealed = true;
+  return FDB_OK;
+}
+
+extern "C" int32_t fdb_store_num_series(const fdb_store_t* s) { return (int32_t)s->series.size(); }
+extern "C" int32_t fdb_series_num_chunks(const fdb_store_t* s, int32_t sid) {
+  if (sid < 0 || sid >= (int32_t)s->series.size()) return FDB_ERR_BADARG;
+  return (int32_t)s->series[(size_t)sid].chunks.size();
+}
+
+extern "C" int32_t fdb_chunk_get(const fdb_store_t* s, int32_t sid, int32_t ci, fdb_chunk_info_t* out) {
+  if (sid < 0 || sid >= (int32_t)s->series.size()) { fdb_set_error("bad series id"); return FDB_ERR_BADARG; }
+  const Series& se = s->series[(size_t)sid];
+  if (ci < 0 || ci >= (int32_t)se.chunks.size()) { fdb_set_error("bad chunk idx"); return FDB_ERR_BADARG; }
+  const Chunk& c = se.chunks[(size_t)ci];
+  if (s->sealed) {
+    out->ts_vec  = s->blob.data() + c.ts_off;
+    out->val_vec = s->blob.data() + c.val_off;
+    uint32_t tl, vl;
+    memcpy(&tl, out->ts_vec, 4); memcpy(&vl, out->val_vec, 4);
+    out->ts_vec_len = (int32_t)tl + 4;
+    out->val_vec_len = (int32_t)vl + 4;
+  } else {
+    out->ts_vec = c.ts_bytes.data();
+    out->val_vec = c.val_bytes.data();
+    out->ts_vec_len = (int32_t)c.ts_bytes.size();
+    out->val_vec_len = (int32_t)c.val_bytes.size();
+  }
+  out->num_rows = c.num_rows;
+  out->start_time = c.start_time;
+  out->end_time = c.end_time;
+  return FDB_OK;
+}
+
+extern "C" int32_t fdb_store_view(const fdb_store_t* s, fdb_view_t* out) {
+  if (!s->sealed) { fdb_set_error("store not sealed"); return FDB_ERR_BADARG; }
+  out->blob = s->blob.data();
+  out->blob_len = (int64_t)s->blob.size();
+  out->dir = s->dir.data();
+  out->num_chunks = (int64_t)s->dir.size();
+  out->series_first = s->series_first.data();
+  out->series_nchunks = s->series_nchunks.data();
+  out->group_ids = s->group_ids.data();
+  out->num_series = (int32_t)s->series.size();
+  return FDB_OK;
+}

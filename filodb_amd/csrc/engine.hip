@@ -1,0 +1,712 @@
+// MI355X (gfx950) engine: chunk-scan + range-vector kernels behind the C-ABI.
+//
+// Replaces, as one batched GPU launch per (shard, query), the per-window virtual
+// dispatch of the reference execution engine:
+//   ChunkedWindowIterator.doNext        query/.../exec/PeriodicSamplesMapper.scala:293-330
+//   WindowedChunkIterator               core/.../store/ChunkSetInfo.scala:445-529
+//   ChunkedRangeFunction.addChunks      query/.../rangefn/RangeFunction.scala:101-198
+//   ChunkedRateFunctionBase/extrapolatedRate  rangefn/RateFunctions.scala:72-111,230-289
+//   gauge over-time functions           rangefn/AggrOverTimeFunctions.scala
+//   RangeVectorAggregator.fastReduce    query/.../exec/AggrOverRangeVectors.scala:320-377
+//
+// Execution model: one 64-lane wavefront per series; 256-thread
+// blocks = 4 series; chunks decoded into LDS once per series; windows parallel
+// across lanes reading LDS (the ~97% window overlap is served on-chip). Counter
+// correction is a wave-wide prefix scan; the chunk→chunk carry is sequential in
+// time order exactly like the reference's CorrectionMeta carry.
+//
+// There is NO CPU fallback: engine creation fails without a HIP device.
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <cstring>
+#include <cstdio>
+#include <cmath>
+#include <vector>
+
+#include "chunk_format.h"
+#include "../../include/filodb_amd.h"
+
+void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
+
+#define HIP_CHECK(expr) do { hipError_t _e = (expr); if (_e != hipSuccess) { \
+  fdb_set_error("%s failed: %s", #expr, hipGetErrorString(_e)); return FDB_ERR; } } while (0)
+#define HIP_CHECK_NULL(expr) do { hipError_t _e = (expr); if (_e != hipSuccess) { \
+  fdb_set_error("%s failed: %s", #expr, hipGetErrorString(_e)); return nullptr; } } while (0)
+
+// ---------------------------------------------------------------------------
+// capacity limits (round 1)
+// ---------------------------------------------------------------------------
+#define FDB_MAX_ROWS_PER_SERIES 416   // LDS-resident rows per series
+#define FDB_MAX_CHUNKS_PER_SERIES 16
+#define WAVES_PER_BLOCK 4
+#define BLOCK_THREADS (WAVES_PER_BLOCK * 64)
+
+enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
+       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10 };
+enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5 };
+
+// ---------------------------------------------------------------------------
+// device-side vector readers (same layouts as oracle)
+// ---------------------------------------------------------------------------
+struct DirSoA {
+  const uint64_t* ts_off;
+  const uint64_t* val_off;
+  const int64_t*  start_time;
+  const int64_t*  end_time;
+  const int32_t*  num_rows;
+};
+
+__device__ __forceinline__ uint16_t d_u16(const uint8_t* p) { uint16_t v; memcpy(&v, p, 2); return v; }
+__device__ __forceinline__ uint32_t d_u32(const uint8_t* p) { uint32_t v; memcpy(&v, p, 4); return v; }
+__device__ __forceinline__ int32_t  d_i32(const uint8_t* p) { int32_t v; memcpy(&v, p, 4); return v; }
+__device__ __forceinline__ int64_t  d_i64(const uint8_t* p) { int64_t v; memcpy(&v, p, 8); return v; }
+__device__ __forceinline__ double   d_f64(const uint8_t* p) { double v; memcpy(&v, p, 8); return v; }
+
+struct DVec {           // opened vector header
+  const uint8_t* idata;
+  int64_t init;
+  int32_t slope;
+  int n;
+  uint16_t wf;
+  uint8_t nbits, sign, dropped;
+};
+
+__device__ void d_vec_open(const uint8_t* p, DVec* v) {
+  v->wf = d_u16(p + 4);
+  v->dropped = (d_u16(p + 6) & FDB_DROP_MASK) != 0;
+  if (v->wf == FDB_WF_DDV) {
+    v->init = d_i64(p + FDB_DDV_OFF_INIT);
+    v->slope = d_i32(p + FDB_DDV_OFF_SLOPE);
+    const uint8_t* inner = p + FDB_DDV_OFF_INNER;
+    v->nbits = inner[6] & FDB_NBITS_MASK;
+    v->sign = (inner[6] & FDB_SIGN_MASK) != 0;
+    v->idata = inner + FDB_PRIM_OFF_DATA;
+    int numBytes = (int)d_u32(inner);
+    int bitShift = inner[7] & 0x3f;
+    v->n = ((numBytes - 4) * 8 + (bitShift != 0 ? bitShift - 8 : 0)) / v->nbits;
+  } else if (v->wf == FDB_WF_DDV_CONST) {
+    v->n = d_i32(p + FDB_DDVC_OFF_NELEM);
+    v->init = d_i64(p + FDB_DDVC_OFF_INIT);
+    v->slope = d_i32(p + FDB_DDVC_OFF_SLOPE);
+    v->idata = nullptr; v->nbits = 0; v->sign = 0;
+  } else {
+    v->n = ((int)d_u32(p) - 4) / 8;
+    v->idata = p + FDB_PRIM_OFF_DATA;
+    v->init = 0; v->slope = 0; v->nbits = 64; v->sign = 1;
+  }
+}
+
+__device__ __forceinline__ int64_t d_inner_at(const DVec* v, int i) {
+  switch (v->nbits) {
+    case 32: return d_i32(v->idata + 4 * (size_t)i);
+    case 16: { int32_t x = (int16_t)d_u16(v->idata + 2 * (size_t)i);
+               return v->sign ? x : (x & 0xffff); }
+    case 8:  { int32_t x = (int8_t)v->idata[i];
+               return v->sign ? x : (x & 0xff); }
+    case 4:  return (v->idata[i >> 1] >> ((i & 1) * 4)) & 0x0f;
+    case 2:  return (v->idata[i >> 2] >> ((i & 3) * 2)) & 0x03;
+  }
+  return 0;
+}
+
+__device__ __forceinline__ int64_t d_lv_at(const DVec* v, int i) {
+  if (v->wf == FDB_WF_DDV) return v->init + (int64_t)v->slope * i + d_inner_at(v, i);
+  if (v->wf == FDB_WF_DDV_CONST) return v->init + (int64_t)v->slope * i;
+  return d_i64(v->idata + 8 * (size_t)i);
+}
+
+__device__ __forceinline__ double d_dv_at(const DVec* v, int i) {
+  if (v->wf == FDB_WF_PRIM64) return d_f64(v->idata + 8 * (size_t)i);
+  return (double)d_lv_at(v, i);
+}
+
+// ---------------------------------------------------------------------------
+// per-series LDS workspace
+// ---------------------------------------------------------------------------
+struct ChunkMeta {
+  int32_t row0, nrows;
+  int64_t start_time, end_time;    // directory times (pre-encoding, like ChunkSetInfo)
+  double last_for_update;          // updateCorrection lastValue (DoubleVector.scala:375-391,190-195)
+  double chunk_correction;         // CorrectingDoubleVectorReader._correction total
+  int32_t dropped;
+  int32_t v0_nan;                  // isNaN(apply(0)) for the single-row-NaN rule
+};
+
+template <bool RATE_FAMILY>
+struct Ws {                         // per-wave LDS workspace
+  int64_t ts[FDB_MAX_ROWS_PER_SERIES];
+  double  val[FDB_MAX_ROWS_PER_SERIES];                  // raw values
+  double  aux[RATE_FAMILY ? FDB_MAX_ROWS_PER_SERIES : 1]; // corrected values (dropped chunks)
+  ChunkMeta cm[FDB_MAX_CHUNKS_PER_SERIES];
+  int32_t nchunks;
+  int32_t total_rows;
+};
+
+// wave-wide inclusive prefix sum (64 lanes)
+__device__ __forceinline__ double wave_incl_scan(double x, int lane) {
+  for (int off = 1; off < 64; off <<= 1) {
+    double t = __shfl_up(x, off);
+    if (lane >= off) x += t;
+  }
+  return x;
+}
+
+// first index in [0,n) with seg[i] >= item, over sorted LDS segment; n when none.
+__device__ __forceinline__ int lds_search_ge(const int64_t* seg, int n, int64_t item) {
+  int lo = 0, hi = n;
+  while (lo < hi) { int mid = (lo + hi) >> 1; if (seg[mid] < item) lo = mid + 1; else hi = mid; }
+  return lo;
+}
+// last index with seg[i] <= item; -1 when none (ceilingIndex semantics)
+__device__ __forceinline__ int lds_search_le(const int64_t* seg, int n, int64_t item) {
+  int lo = 0, hi = n;
+  while (lo < hi) { int mid = (lo + hi) >> 1; if (seg[mid] <= item) lo = mid + 1; else hi = mid; }
+  return lo - 1;
+}
+
+// extrapolatedRate (RateFunctions.scala:72-111) — same arithmetic as the oracle
+__device__ double d_extrapolated_rate(int64_t windowStart, int64_t windowEnd, int numSamples,
+                                      int64_t t1, double v1, int64_t t2, double v2,
+                                      bool isCounter, bool isRate) {
+  double durationToStart = (double)(t1 - windowStart) / 1000.0;
+  double durationToEnd = (double)(windowEnd - t2) / 1000.0;
+  double sampledInterval = (double)(t2 - t1) / 1000.0;
+  double avgDur = sampledInterval / ((double)numSamples - 1);
+  double delta = v2 - v1;
+  if (isCounter && delta > 0 && v1 >= 0) {
+    double durationToZero = sampledInterval * (v1 / delta);
+    if (durationToZero < durationToStart) durationToStart = durationToZero;
+  }
+  double thresh = avgDur * 1.1;
+  double ext = sampledInterval;
+  ext += (durationToStart < thresh) ? durationToStart : avgDur / 2;
+  ext += (durationToEnd < thresh) ? durationToEnd : avgDur / 2;
+  double scaledDelta = delta * (ext / sampledInterval);
+  return isRate ? (scaledDelta / (double)(windowEnd - windowStart) * 1000.0) : scaledDelta;
+}
+
+// NaN-aware f64 atomic min/max via CAS (group aggregation; RowAggregator semantics)
+__device__ void atomic_min_max_f64(double* addr, double val, bool is_min) {
+  unsigned long long* a = (unsigned long long*)addr;
+  unsigned long long old = *a, assumed;
+  do {
+    assumed = old;
+    double cur = __longlong_as_double((long long)assumed);
+    double nw = isnan(cur) ? val : (is_min ? fmin(cur, val) : fmax(cur, val));
+    if (!isnan(cur) && nw == cur) return;
+    old = atomicCAS(a, assumed, (unsigned long long)__double_as_longlong(nw));
+  } while (old != assumed);
+}
+
+// ---------------------------------------------------------------------------
+// the scan kernel
+// ---------------------------------------------------------------------------
+template <int FUNC>
+__global__ __launch_bounds__(BLOCK_THREADS)
+void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
+                 const int32_t* __restrict__ series_first,
+                 const int32_t* __restrict__ series_nchunks,
+                 const int32_t* __restrict__ group_ids,
+                 int num_series,
+                 int64_t qstart, int64_t qstep, int64_t qend, int64_t qwindow,
+                 int num_windows,
+                 int agg_id,
+                 double* __restrict__ out,        // [S×W] when AGG_NONE else [G×W] sums
+                 double* __restrict__ out_cnt)    // [G×W] contribution counts (agg) or null
+{
+  constexpr bool RATE_FAMILY = (FUNC <= FN_DELTA);
+  __shared__ Ws<RATE_FAMILY> ws_all[WAVES_PER_BLOCK];
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int sid = blockIdx.x * WAVES_PER_BLOCK + wave;
+  if (sid >= num_series) return;
+  Ws<RATE_FAMILY>& ws = ws_all[wave];
+
+  // ---- decode phase: all chunks of this series into LDS --------------------
+  const int first = series_first[sid];
+  int nchunks = series_nchunks[sid];
+  if (nchunks > FDB_MAX_CHUNKS_PER_SERIES) nchunks = FDB_MAX_CHUNKS_PER_SERIES;
+  if (lane == 0) { ws.nchunks = nchunks; ws.total_rows = 0; }
+
+  int row0 = 0;
+  for (int c = 0; c < nchunks; c++) {
+    DVec tv, vv;
+    d_vec_open(blob + dir.ts_off[first + c], &tv);
+    d_vec_open(blob + dir.val_off[first + c], &vv);
+    int n = dir.num_rows[first + c];
+    if (row0 + n > FDB_MAX_ROWS_PER_SERIES) { n = 0; }   // guarded at upload; belt+braces
+
+    for (int i = lane; i < n; i += 64) {
+      ws.ts[row0 + i] = d_lv_at(&tv, i);
+      ws.val[row0 + i] = d_dv_at(&vv, i);
+    }
+
+    if (lane == 0) {
+      ChunkMeta& m = ws.cm[c];
+      m.row0 = row0; m.nrows = n;
+      m.start_time = dir.start_time[first + c];
+      m.end_time = dir.end_time[first + c];
+      m.dropped = vv.dropped;
+      m.chunk_correction = 0;
+      m.v0_nan = 0;
+      m.last_for_update = 0;
+    }
+    row0 += n;
+  }
+  if (lane == 0) ws.total_rows = row0;
+  __builtin_amdgcn_s_waitcnt(0);   // LDS writes visible within the wave
+  __builtin_amdgcn_wave_barrier();
+
+  // ---- rate family: per-chunk correction scan (CorrectingDoubleVectorReader
+  //      :325-342) + updateCorrection scalars (:375-391,190-195) -------------
+  if constexpr (RATE_FAMILY) {
+    for (int c = 0; c < nchunks; c++) {
+      ChunkMeta& m = ws.cm[c];
+      const int r0 = m.row0, n = m.nrows;
+      if (n == 0) continue;
+      if (lane == 0) m.v0_nan = isnan(ws.val[r0]);
+      if (m.dropped) {
+        double carry_corr = 0;
+        double carry_x = -1.7976931348623157e308;   // 'last' starts Double.MinValue
+        int last_idx = -1;                          // last non-NaN index (for update)
+        double last_val = 0;
+        for (int base = 0; base < n; base += 64) {
+          int i = base + lane;
+          double raw = (i < n) ? ws.val[r0 + i] : 0;
+          double x = (i < n && !isnan(raw)) ? raw : 0;
+          // drop contribution: x < last(previous x) adds previous x
+          double px = __shfl_up(x, 1);
+          if (lane == 0) px = carry_x;
+          double ci = (i < n && x < px) ? px : 0;
+          double scan = wave_incl_scan(ci, lane);
+          if (i < n) ws.aux[r0 + i] = x + carry_corr + scan;
+          carry_corr += __shfl(scan, 63);
+          carry_x = __shfl(x, 63);
+          if (i < n && !isnan(raw)) { last_idx = i; last_val = raw; }
+        }
+        // wave-reduce max last_idx (value rides along)
+        for (int off = 32; off > 0; off >>= 1) {
+          int oi = __shfl_down(last_idx, off);
+          double ov = __shfl_down(last_val, off);
+          if (oi > last_idx) { last_idx = oi; last_val = ov; }
+        }
+        if (lane == 0) {
+          m.chunk_correction = carry_corr;
+          m.last_for_update = (last_idx >= 0) ? last_val : 0;
+        }
+      } else if (lane == 0) {
+        m.last_for_update = ws.val[r0 + n - 1];     // default updateCorrection
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_wave_barrier();
+  }
+
+  // ---- window phase: lanes split the windows -------------------------------
+  const int grp = group_ids[sid];
+  for (int w = lane; w < num_windows; w += 64) {
+    const int64_t wEnd = qstart + (int64_t)w * qstep;
+    const int64_t wStart = wEnd - qwindow;
+    double result = NAN;
+
+    if constexpr (RATE_FAMILY) {
+      bool meta_has = false;
+      double meta_last = 0, meta_corr = 0;
+      int numSamples = 0;
+      int64_t lowestTime = INT64_MAX, highestTime = 0;
+      double lowestValue = NAN, highestValue = NAN;
+      constexpr bool isCounter = (FUNC != FN_DELTA);
+      for (int c = 0; c < ws.nchunks; c++) {
+        const ChunkMeta& m = ws.cm[c];
+        if (m.end_time < wStart) continue;          // WindowedChunkIterator drop rule
+        const int64_t* seg = ws.ts + m.row0;
+        int startRow = lds_search_ge(seg, m.nrows, wStart);
+        int endRow = lds_search_le(seg, m.nrows, wEnd);
+        if (isCounter && meta_has) {                // detectDropAndCorrection
+          double firstv = ws.val[m.row0];
+          if (isnan(firstv) || firstv < meta_last) meta_corr += meta_last;
+        }
+        if (startRow <= endRow && endRow < m.nrows) {
+          bool skip = isCounter && startRow == 0 && endRow == 0 && m.v0_nan;
+          if (!skip) {
+            int64_t st = seg[startRow], en = seg[endRow];
+            if (st < lowestTime || en > highestTime) {
+              numSamples += endRow - startRow + 1;
+              if (st < lowestTime) {
+                lowestTime = st;
+                double v = m.dropped ? ws.aux[m.row0 + startRow] : ws.val[m.row0 + startRow];
+                lowestValue = isCounter ? v + meta_corr
+                                        : ws.val[m.row0 + startRow];
+              }
+              if (en > highestTime) {
+                highestTime = en;
+                double v = m.dropped ? ws.aux[m.row0 + endRow] : ws.val[m.row0 + endRow];
+                highestValue = isCounter ? v + meta_corr
+                                         : ws.val[m.row0 + endRow];
+              }
+            }
+          }
+        }
+        if (isCounter) {                            // updateCorrection
+          if (m.dropped) meta_corr += m.chunk_correction;
+          meta_last = m.last_for_update;
+          meta_has = true;
+        }
+        if (m.end_time >= wEnd) break;
+      }
+      if (highestTime > lowestTime)
+        result = d_extrapolated_rate(wStart, wEnd, numSamples,
+                                     lowestTime, lowestValue, highestTime, highestValue,
+                                     isCounter, FUNC == FN_RATE);
+    } else {
+      double sum = NAN, count = NAN, sqsum = NAN, mn = NAN, mx = NAN;
+      double changes = NAN, prev = NAN;
+      int icount = 0;
+      for (int c = 0; c < ws.nchunks; c++) {
+        const ChunkMeta& m = ws.cm[c];
+        if (m.end_time < wStart) continue;
+        const int64_t* seg = ws.ts + m.row0;
+        const double* vals = ws.val + m.row0;
+        int startRow = lds_search_ge(seg, m.nrows, wStart);
+        int endRow = lds_search_le(seg, m.nrows, wEnd);
+        if (startRow <= endRow && endRow < m.nrows) {
+          if (FUNC == FN_SUM || FUNC == FN_AVG) {
+            double cs = NAN;                       // NaN-skipping chunk sum (:244-253)
+            for (int i = startRow; i <= endRow; i++) {
+              double x = vals[i];
+              if (!isnan(x)) { if (isnan(cs)) cs = 0; cs += x; }
+            }
+            if (!isnan(cs) && isnan(sum)) sum = 0;
+            sum += cs;                             // NaN-poison quirk preserved
+            if (FUNC == FN_AVG)
+              for (int i = startRow; i <= endRow; i++) if (!isnan(vals[i])) icount++;
+          } else if (FUNC == FN_COUNT) {
+            if (isnan(count)) count = 0;
+            for (int i = startRow; i <= endRow; i++) if (!isnan(vals[i])) count += 1;
+          } else if (FUNC == FN_MIN || FUNC == FN_MAX) {
+            for (int i = startRow; i <= endRow; i++) {
+              double x = vals[i];
+              if (isnan(x)) continue;
+              if (FUNC == FN_MIN) mn = (isnan(mn) || x < mn) ? x : mn;
+              else mx = (isnan(mx) || x > mx) ? x : mx;
+            }
+          } else if (FUNC == FN_STDDEV || FUNC == FN_STDVAR) {
+            double cs = NAN, csq = NAN; int cc = 0;
+            for (int i = startRow; i <= endRow; i++) {
+              double x = vals[i];
+              if (!isnan(x)) {
+                if (isnan(cs)) cs = 0;
+                if (isnan(csq)) csq = 0;
+                cs += x; csq += x * x; cc++;
+              }
+            }
+            if (!isnan(cs) && isnan(sum)) sum = 0;
+            sum += cs;
+            if (!isnan(csq) && isnan(sqsum)) sqsum = 0;
+            sqsum += csq;
+            icount += cc;
+          } else if (FUNC == FN_CHANGES) {
+            if (isnan(changes)) changes = 0;       // double changes (:283-302)
+            double prevV = prev, ch = 0;
+            for (int i = startRow; i <= endRow; i++) {
+              double x = vals[i];
+              if (!isnan(x) && prevV != x && !isnan(prevV)) ch += 1;
+              prevV = x;
+            }
+            changes += ch; prev = prevV;
+          }
+        }
+        if (m.end_time >= wEnd) break;
+      }
+      switch (FUNC) {
+        case FN_SUM:   result = sum; break;
+        case FN_COUNT: result = count; break;
+        case FN_AVG:   result = icount > 0 ? sum / icount : (isnan(sum) ? sum : 0); break;
+        case FN_MIN:   result = mn; break;
+        case FN_MAX:   result = mx; break;
+        case FN_STDDEV: case FN_STDVAR: {
+          if (icount > 0) {
+            double avg = sum / icount;
+            double r = sqsum / icount - avg * avg;
+            result = (FUNC == FN_STDDEV) ? sqrt(r) : r;
+          } else result = isnan(sum) ? sum : 0;
+        } break;
+        case FN_CHANGES: result = changes; break;
+      }
+    }
+
+    if (agg_id == AGG_NONE) {
+      out[(size_t)sid * num_windows + w] = result;
+    } else if (!isnan(result)) {
+      // fastReduce fused: RowAggregator merge per group cell
+      size_t cell = (size_t)grp * num_windows + w;
+      switch (agg_id) {
+        case AGG_SUM: case AGG_AVG:
+          atomicAdd(&out[cell], result);
+          atomicAdd(&out_cnt[cell], 1.0);
+          break;
+        case AGG_COUNT:
+          atomicAdd(&out[cell], 1.0);
+          atomicAdd(&out_cnt[cell], 1.0);
+          break;
+        case AGG_MIN: atomic_min_max_f64(&out[cell], result, true);
+                      atomicAdd(&out_cnt[cell], 1.0); break;
+        case AGG_MAX: atomic_min_max_f64(&out[cell], result, false);
+                      atomicAdd(&out_cnt[cell], 1.0); break;
+      }
+    }
+  }
+}
+
+// presentation fixup for aggregated grids (NaN where no contributions; mean for avg)
+__global__ void agg_present_kernel(double* out, const double* cnt, size_t n,
+                                   int agg_id, int partial) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  if (partial) return;                      // partial mode: leave raw sums + counts
+  if (cnt[i] <= 0) { out[i] = NAN; return; }
+  if (agg_id == AGG_AVG) out[i] = out[i] / cnt[i];
+}
+
+__global__ void fill_f64_kernel(double* p, size_t n, double v) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = v;
+}
+
+// ---------------------------------------------------------------------------
+// host engine
+// ---------------------------------------------------------------------------
+struct fdb_engine {
+  int device;
+  hipStream_t stream;
+};
+
+struct fdb_dataset {
+  uint8_t* blob;
+  uint64_t *ts_off, *val_off;
+  int64_t *start_time, *end_time;
+  int32_t *num_rows;
+  int32_t *series_first, *series_nchunks, *group_ids;
+  int32_t num_series;
+  int64_t num_chunks;
+  int64_t payload_bytes;    // sum of vector bytes (algorithmic HBM footprint)
+  int64_t total_samples;
+  int max_group;            // max group id seen (for validation)
+};
+
+extern "C" fdb_engine_t* fdb_engine_create(int32_t device) {
+  int count = 0;
+  hipError_t e = hipGetDeviceCount(&count);
+  if (e != hipSuccess || count == 0) {
+    fdb_set_error("no HIP device present (hipGetDeviceCount: %s) — "
+                  "filodb_amd has no CPU fallback", hipGetErrorString(e));
+    return nullptr;
+  }
+  if (device < 0 || device >= count) { fdb_set_error("bad device %d", device); return nullptr; }
+  HIP_CHECK_NULL(hipSetDevice(device));
+  auto* eng = new fdb_engine();
+  eng->device = device;
+  if (hipStreamCreate(&eng->stream) != hipSuccess) {
+    fdb_set_error("hipStreamCreate failed");
+    delete eng;
+    return nullptr;
+  }
+  return eng;
+}
+
+extern "C" void fdb_engine_destroy(fdb_engine_t* e) {
+  if (!e) return;
+  hipStreamDestroy(e->stream);
+  delete e;
+}
+
+extern "C" int32_t fdb_engine_synchronize(fdb_engine_t* e) {
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  return FDB_OK;
+}
+
+extern "C" void fdb_dataset_destroy(fdb_dataset_t* d) {
+  if (!d) return;
+  hipFree(d->blob); hipFree(d->ts_off); hipFree(d->val_off);
+  hipFree(d->start_time); hipFree(d->end_time); hipFree(d->num_rows);
+  hipFree(d->series_first); hipFree(d->series_nchunks); hipFree(d->group_ids);
+  delete d;
+}
+
+extern "C" int64_t fdb_dataset_bytes(const fdb_dataset_t* d) { return d->payload_bytes; }
+extern "C" int64_t fdb_dataset_samples(const fdb_dataset_t* d) { return d->total_samples; }
+
+extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t* s) {
+  fdb_view_t view;
+  if (fdb_store_view(s, &view) != FDB_OK) return nullptr;
+  HIP_CHECK_NULL(hipSetDevice(e->device));
+
+  const fdb_dir_entry_t* dir = (const fdb_dir_entry_t*)view.dir;
+  int64_t nc = view.num_chunks;
+
+  // SoA host staging
+  std::vector<uint64_t> ts_off(nc), val_off(nc);
+  std::vector<int64_t> st(nc), en(nc);
+  std::vector<int32_t> nr(nc);
+  int64_t payload = 0, samples = 0;
+  for (int64_t i = 0; i < nc; i++) {
+    ts_off[i] = dir[i].ts_off; val_off[i] = dir[i].val_off;
+    st[i] = dir[i].start_time; en[i] = dir[i].end_time; nr[i] = dir[i].num_rows;
+    uint32_t tl, vl;
+    memcpy(&tl, view.blob + dir[i].ts_off, 4);
+    memcpy(&vl, view.blob + dir[i].val_off, 4);
+    payload += (int64_t)tl + 4 + (int64_t)vl + 4;
+    samples += dir[i].num_rows;
+  }
+  int max_group = 0;
+  for (int32_t sid = 0; sid < view.num_series; sid++) {
+    if (view.group_ids[sid] > max_group) max_group = view.group_ids[sid];
+    // round-1 capacity checks
+    int total = 0;
+    for (int c = 0; c < view.series_nchunks[sid]; c++)
+      total += dir[view.series_first[sid] + c].num_rows;
+    if (total > FDB_MAX_ROWS_PER_SERIES || view.series_nchunks[sid] > FDB_MAX_CHUNKS_PER_SERIES) {
+      fdb_set_error("series %d exceeds round-1 capacity (%d rows / %d chunks; caps %d/%d)",
+                    sid, total, view.series_nchunks[sid],
+                    FDB_MAX_ROWS_PER_SERIES, FDB_MAX_CHUNKS_PER_SERIES);
+      return nullptr;
+    }
+  }
+
+  auto* d = new fdb_dataset();
+  memset(d, 0, sizeof(*d));
+  d->num_series = view.num_series;
+  d->num_chunks = nc;
+  d->payload_bytes = payload;
+  d->total_samples = samples;
+  d->max_group = max_group;
+
+  auto upload = [&](void** dst, const void* src, size_t bytes) -> bool {
+    if (hipMalloc(dst, bytes ? bytes : 8) != hipSuccess) return false;
+    return hipMemcpy(*dst, src, bytes, hipMemcpyHostToDevice) == hipSuccess;
+  };
+  bool ok = upload((void**)&d->blob, view.blob, (size_t)view.blob_len)
+    && upload((void**)&d->ts_off, ts_off.data(), nc * 8)
+    && upload((void**)&d->val_off, val_off.data(), nc * 8)
+    && upload((void**)&d->start_time, st.data(), nc * 8)
+    && upload((void**)&d->end_time, en.data(), nc * 8)
+    && upload((void**)&d->num_rows, nr.data(), nc * 4)
+    && upload((void**)&d->series_first, view.series_first, (size_t)view.num_series * 4)
+    && upload((void**)&d->series_nchunks, view.series_nchunks, (size_t)view.num_series * 4)
+    && upload((void**)&d->group_ids, view.group_ids, (size_t)view.num_series * 4);
+  if (!ok) {
+    fdb_set_error("device upload failed (out of HBM?)");
+    fdb_dataset_destroy(d);
+    return nullptr;
+  }
+  return d;
+}
+
+// launch dispatch over the func template parameter
+static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
+                           double* dev_out, double* dev_cnt) {
+  DirSoA dir{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
+  int nw = fdb_num_windows(q);
+  int grid = (d->num_series + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+  #define CASE(F) case F: \
+    hipLaunchKernelGGL(scan_kernel<F>, dim3(grid), dim3(BLOCK_THREADS), 0, e->stream, \
+      d->blob, dir, d->series_first, d->series_nchunks, d->group_ids, d->num_series, \
+      q->start, q->step, q->end, q->window, nw, q->agg_id, dev_out, dev_cnt); break
+  switch (q->func_id) {
+    CASE(FN_RATE); CASE(FN_INCREASE); CASE(FN_DELTA); CASE(FN_SUM); CASE(FN_COUNT);
+    CASE(FN_AVG); CASE(FN_MIN); CASE(FN_MAX); CASE(FN_STDDEV); CASE(FN_STDVAR);
+    CASE(FN_CHANGES);
+    default: fdb_set_error("bad func_id %d", q->func_id); return FDB_ERR_BADARG;
+  }
+  #undef CASE
+  HIP_CHECK(hipGetLastError());
+  return FDB_OK;
+}
+
+static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
+                         double* out, double* out_counts, int32_t out_on_device,
+                         int32_t warmup, int32_t iters, double* avg_ms) {
+  HIP_CHECK(hipSetDevice(e->device));
+  int nw = fdb_num_windows(q);
+  if (nw <= 0) { fdb_set_error("bad window params"); return FDB_ERR_BADARG; }
+  size_t out_len = (q->agg_id == AGG_NONE) ? (size_t)d->num_series * nw
+                                           : (size_t)q->num_groups * nw;
+  if (q->agg_id != AGG_NONE) {
+    if (q->num_groups <= 0 || d->max_group >= q->num_groups) {
+      fdb_set_error("num_groups %d inconsistent with dataset max group %d",
+                    q->num_groups, d->max_group);
+      return FDB_ERR_BADARG;
+    }
+  }
+
+  double *dev_out = out_on_device ? out : nullptr;
+  double *dev_cnt = out_on_device ? out_counts : nullptr;
+  bool own_out = false, own_cnt = false;
+  if (!dev_out) {
+    HIP_CHECK(hipMalloc(&dev_out, out_len * 8));
+    own_out = true;
+  }
+  if (q->agg_id != AGG_NONE && !dev_cnt) {
+    HIP_CHECK(hipMalloc(&dev_cnt, out_len * 8));
+    own_cnt = true;
+  }
+
+  int partial = (q->agg_id != AGG_NONE && out_counts != nullptr) ? 1 : 0;
+  hipEvent_t ev0, ev1;
+  HIP_CHECK(hipEventCreate(&ev0));
+  HIP_CHECK(hipEventCreate(&ev1));
+
+  int total_runs = warmup + iters;
+  float ms_sum = 0;
+  for (int it = 0; it < total_runs; it++) {
+    if (q->agg_id != AGG_NONE) {
+      HIP_CHECK(hipMemsetAsync(dev_out, 0, out_len * 8, e->stream));
+      HIP_CHECK(hipMemsetAsync(dev_cnt, 0, out_len * 8, e->stream));
+      if (q->agg_id == AGG_MIN || q->agg_id == AGG_MAX) {
+        fill_f64_kernel<<<(unsigned)((out_len + 255) / 256), 256, 0, e->stream>>>(
+            dev_out, out_len, NAN);
+      }
+    }
+    bool timed = it >= warmup;
+    if (timed) HIP_CHECK(hipEventRecord(ev0, e->stream));
+    int32_t rc = launch_scan(e, d, q, dev_out, dev_cnt);
+    if (rc != FDB_OK) return rc;
+    if (timed) {
+      HIP_CHECK(hipEventRecord(ev1, e->stream));
+      HIP_CHECK(hipEventSynchronize(ev1));
+      float ms;
+      HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+      ms_sum += ms;
+    }
+  }
+  if (q->agg_id != AGG_NONE) {
+    agg_present_kernel<<<(unsigned)((out_len + 255) / 256), 256, 0, e->stream>>>(
+        dev_out, dev_cnt, out_len, q->agg_id, partial);
+    HIP_CHECK(hipGetLastError());
+  }
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  hipEventDestroy(ev0); hipEventDestroy(ev1);
+
+  if (!out_on_device) {
+    HIP_CHECK(hipMemcpy(out, dev_out, out_len * 8, hipMemcpyDeviceToHost));
+    if (q->agg_id != AGG_NONE && out_counts)
+      HIP_CHECK(hipMemcpy(out_counts, dev_cnt, out_len * 8, hipMemcpyDeviceToHost));
+  }
+  if (own_out) hipFree(dev_out);
+  if (own_cnt) hipFree(dev_cnt);
+  if (avg_ms) *avg_ms = iters > 0 ? (double)ms_sum / iters : 0.0;
+  return FDB_OK;
+}
+
+extern "C" int32_t fdb_query_exec(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
+                                  double* out, double* out_counts, int32_t out_on_device) {
+  return run_query(e, d, q, out, out_counts, out_on_device, 0, 1, nullptr);
+}
+
+extern "C" int32_t fdb_query_bench(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
+                                   double* out, double* out_counts, int32_t out_on_device,
+                                   int32_t warmup, int32_t iters, double* avg_kernel_ms) {
+  return run_query(e, d, q, out, out_counts, out_on_device, warmup, iters, avg_kernel_ms);
+}

@@ -1,0 +1,192 @@
+/* filodb_amd — C-ABI for the MI355X-native FiloDB chunk-scan + range-vector engine.
+ *
+ * This is the drop-in boundary. Each entry point names the
+ * reference interface it replaces. The shape follows the reference's own native-crossing
+ * precedent: SimdNativeMethods.simdSumDouble(dataAddr, start, end, ignoreNaN) — raw
+ * pointers in, numbers out, no exceptions across the boundary
+ * (core/src/main/scala/filodb.memory/format/vectors/SimdNativeMethods.scala:75,
+ *  core/src/rust/filodb_core/src/simd_vectors.rs:174-213).
+ *
+ * Ownership: the caller owns every buffer it passes; the engine owns its scratch and
+ * device memory. Errors: negative int return + thread-local message via fdb_last_error().
+ */
+#ifndef FILODB_AMD_H
+#define FILODB_AMD_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- error handling ------------------------------------------------------ */
+/* Returns the thread-local message for the last failing fdb_* call. */
+const char* fdb_last_error(void);
+
+#define FDB_OK            0
+#define FDB_ERR          -1   /* generic; see fdb_last_error() */
+#define FDB_ERR_NOGPU    -2   /* compute entry point called with no HIP device */
+#define FDB_ERR_BADARG   -3
+
+/* ---- chunk store (host): builds the reference's frozen chunk format ------ */
+/* Replaces the ingest/encode path TimeSeriesPartition.switchBuffers(encode=true) →
+ * optimizeAll (core/.../memstore/TimeSeriesPartition.scala:130; encoding decisions:
+ * DeltaDeltaVector.scala:63-85, DoubleVector.scala:86-96,457-476,
+ * LongBinaryVector.scala:333-341). Host-side C++; produces bit-identical frozen
+ * BinaryVector bytes plus the chunk directory (ChunkSetInfoReader cached fields,
+ * core/.../store/ChunkSetInfoReader.scala:53-66). */
+typedef struct fdb_store fdb_store_t;
+
+/* series value temporality / column kind */
+#define FDB_COL_GAUGE    0   /* plain double column                        */
+#define FDB_COL_COUNTER  1   /* drop-detecting counter column (DoubleCounterAppender) */
+
+fdb_store_t* fdb_store_create(int64_t expected_series);
+void         fdb_store_destroy(fdb_store_t* s);
+
+/* Adds a series; returns its dense series id (>=0). group_id is the precomputed
+ * cross-series aggregation key (the label-subset key of AggregateMapReduce,
+ * query/.../exec/AggrOverRangeVectors.scala:150-159, resolved by the host). */
+int32_t fdb_store_add_series(fdb_store_t* s, int32_t group_id, int32_t col_kind);
+
+/* Appends samples to a series' write buffer. NaN values allowed (Prometheus stale
+ * markers). Timestamps must be nondecreasing. */
+int32_t fdb_series_append(fdb_store_t* s, int32_t series_id,
+                          const int64_t* ts, const double* vals, int32_t n);
+
+/* Forces a chunk boundary: encodes the write buffer into frozen vectors
+ * (= switchBuffers(encode=true)). Called automatically when a buffer reaches
+ * max_rows (default 400 = filodb-defaults.conf:835 sourced chunk cap). */
+int32_t fdb_series_cut_chunk(fdb_store_t* s, int32_t series_id);
+int32_t fdb_store_set_max_rows(fdb_store_t* s, int32_t max_rows);
+/* Encodes every series' outstanding buffer. Call before upload/inspect. */
+int32_t fdb_store_seal(fdb_store_t* s);
+
+/* -- introspection (tests, oracle, upload) -- */
+int32_t  fdb_store_num_series(const fdb_store_t* s);
+int32_t  fdb_series_num_chunks(const fdb_store_t* s, int32_t series_id);
+/* Chunk directory record — the cached ChunkSetInfoReader fields. */
+typedef struct {
+  const uint8_t* ts_vec;    /* frozen timestamp BinaryVector bytes  */
+  const uint8_t* val_vec;   /* frozen value BinaryVector bytes      */
+  int32_t  num_rows;
+  int64_t  start_time;      /* first timestamp in chunk  */
+  int64_t  end_time;        /* last timestamp in chunk   */
+  int32_t  ts_vec_len;      /* total bytes incl. 4B length word     */
+  int32_t  val_vec_len;
+} fdb_chunk_info_t;
+int32_t fdb_chunk_get(const fdb_store_t* s, int32_t series_id, int32_t chunk_idx,
+                      fdb_chunk_info_t* out);
+
+/* Flattened whole-store view over a sealed store: one contiguous blob of frozen
+ * vector bytes plus the SoA chunk directory. This is exactly what
+ * fdb_dataset_upload copies to HBM, and what the test-side oracle reads.
+ * Pointers remain owned by the store and valid until it is mutated/destroyed.
+ * dir entries are `fdb_dir_entry_t` as defined in filodb_amd/csrc/chunk_format.h
+ * ({u64 ts_off, u64 val_off, i64 start_time, i64 end_time, i32 num_rows, i32 pad}). */
+typedef struct {
+  const uint8_t* blob;
+  int64_t        blob_len;
+  const void*    dir;            /* fdb_dir_entry_t[num_chunks]  */
+  int64_t        num_chunks;
+  const int32_t* series_first;   /* first chunk index per series */
+  const int32_t* series_nchunks;
+  const int32_t* group_ids;      /* per series                   */
+  int32_t        num_series;
+  int32_t        _pad;
+} fdb_view_t;
+int32_t fdb_store_view(const fdb_store_t* s, fdb_view_t* out);
+
+/* ---- standalone codec entry points (host; tests/golden-vector parity) ---- */
+/* NibblePack.pack8 / unpack8 (core/.../format/NibblePack.scala:108-183,395-447). */
+int32_t fdb_nibblepack_pack8(const int64_t in[8], uint8_t* out, int32_t outcap);
+int32_t fdb_nibblepack_unpack8(const uint8_t* in, int32_t inlen, int64_t out[8],
+                               int32_t* consumed);
+/* NibblePack.packDelta / packDoubles (NibblePack.scala:37-98). Return bytes written. */
+int32_t fdb_nibblepack_pack_delta(const int64_t* in, int32_t n, uint8_t* out, int32_t outcap);
+int32_t fdb_nibblepack_pack_doubles(const double* in, int32_t n, uint8_t* out, int32_t outcap);
+
+/* ---- query definition ---------------------------------------------------- */
+/* Range functions: the dispatch table of RangeFunction.generatorFor
+ * (query/.../rangefn/RangeFunction.scala:294-410). */
+#define FDB_FN_RATE             0   /* ChunkedRateFunction       */
+#define FDB_FN_INCREASE         1   /* ChunkedIncreaseFunction   */
+#define FDB_FN_DELTA            2   /* ChunkedDeltaFunction      */
+#define FDB_FN_SUM_OVER_TIME    3
+#define FDB_FN_COUNT_OVER_TIME  4
+#define FDB_FN_AVG_OVER_TIME    5
+#define FDB_FN_MIN_OVER_TIME    6
+#define FDB_FN_MAX_OVER_TIME    7
+#define FDB_FN_STDDEV_OVER_TIME 8
+#define FDB_FN_STDVAR_OVER_TIME 9
+#define FDB_FN_CHANGES         10
+
+/* Cross-series aggregation: RowAggregator implementations
+ * (query/.../exec/aggregator/RowAggregator.scala:28-150). */
+#define FDB_AGG_NONE   0   /* emit the full [series × windows] grid  */
+#define FDB_AGG_SUM    1   /* SumRowAggregator.scala:12-34           */
+#define FDB_AGG_COUNT  2
+#define FDB_AGG_MIN    3
+#define FDB_AGG_MAX    4
+#define FDB_AGG_AVG    5   /* AvgRowAggregator.scala:8-41 (sum,count partials) */
+
+typedef struct {
+  int64_t start;     /* first window end timestamp (ms)                      */
+  int64_t step;      /* ms; numWindows = (end-start)/step + 1                */
+  int64_t end;       /* last window end timestamp                            */
+  int64_t window;    /* lookback length (ms); wStart = wEnd - window
+                        (inclusive-range=true, filodb-defaults.conf:590)     */
+  int32_t func_id;   /* FDB_FN_*  (PeriodicSamplesMapper.functionId)         */
+  int32_t agg_id;    /* FDB_AGG_* (AggregateMapReduce)                       */
+  int32_t num_groups;/* required when agg_id != NONE                         */
+  int32_t _pad;
+} fdb_query_t;
+
+static inline int32_t fdb_num_windows(const fdb_query_t* q) {
+  return (int32_t)((q->end - q->start) / q->step) + 1;
+}
+
+/* ---- GPU engine ---------------------------------------------------------- */
+typedef struct fdb_engine  fdb_engine_t;
+typedef struct fdb_dataset fdb_dataset_t;
+
+/* Creates an engine bound to HIP device `device`. Fails (FDB_ERR_NOGPU) when no
+ * HIP device is present — there is deliberately no CPU fallback. */
+fdb_engine_t* fdb_engine_create(int32_t device);
+void          fdb_engine_destroy(fdb_engine_t* e);
+int32_t       fdb_engine_synchronize(fdb_engine_t* e);
+
+/* Uploads a sealed store into one contiguous HBM blob + SoA chunk directory
+ * (DESIGN.md §2). Returns NULL on error. */
+fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t* s);
+void           fdb_dataset_destroy(fdb_dataset_t* d);
+int64_t        fdb_dataset_bytes(const fdb_dataset_t* d);   /* chunk payload bytes in HBM */
+int64_t        fdb_dataset_samples(const fdb_dataset_t* d); /* total rows across chunks   */
+
+/* Executes one (shard, query): the batched equivalent of folding
+ * PeriodicSamplesMapper (+ AggregateMapReduce when agg_id != NONE) over every
+ * RawDataRangeVector of the shard (ExecPlan.scala:404-419).
+ *
+ * agg_id == FDB_AGG_NONE: out must hold series×windows doubles; out_counts ignored.
+ * agg_id != FDB_AGG_NONE: out must hold num_groups×windows doubles; for FDB_AGG_AVG
+ *   (and cross-GPU merges) out_counts (num_groups×windows doubles, may be NULL
+ *   otherwise) receives the count partials — the reduction schema of
+ *   AvgRowAggregator.scala:8-41.
+ * out/out_counts may be HOST pointers (out_on_device=0) or DEVICE pointers
+ * (out_on_device=1, e.g. torch tensor data_ptr for the RCCL all-reduce).
+ * Synchronous: returns after the result is materialized. */
+int32_t fdb_query_exec(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
+                       double* out, double* out_counts, int32_t out_on_device);
+
+/* Timing variant for bench.py: runs the same launch `iters` times and returns the
+ * average per-iteration kernel milliseconds measured with HIP events on the
+ * engine's stream. Results land in out like fdb_query_exec. */
+int32_t fdb_query_bench(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
+                        double* out, double* out_counts, int32_t out_on_device,
+                        int32_t warmup, int32_t iters, double* avg_kernel_ms);
+
+#ifdef __cplusplus
+} /* extern "C" */
+#endif
+#endif /* FILODB_AMD_H */

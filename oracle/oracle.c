@@ -1,0 +1,714 @@
+/* oracle — CPU restatement of the FiloDB chunk-scan + range-vector query path.
+ *
+ * TEST INFRASTRUCTURE ONLY: this library is the parity checker and
+ * the bench.py cpu_baseline leg. Only tests/, __graft_entry__.smoke() and bench.py's
+ * cpu_baseline may load it. It is never linked into the product library and the
+ * product GPU path never calls it.
+ *
+ * Every routine restates, line for line in semantics, the cited reference code:
+ *   decoders   core/.../format/NibblePack.scala, vectors/DeltaDeltaVector.scala,
+ *              vectors/IntBinaryVector.scala, vectors/DoubleVector.scala,
+ *              vectors/LongBinaryVector.scala
+ *   windowing  query/.../exec/PeriodicSamplesMapper.scala:256-331 +
+ *              core/.../store/ChunkSetInfo.scala:445-529
+ *   functions  query/.../rangefn/RateFunctions.scala:72-111,230-322,
+ *              query/.../rangefn/AggrOverTimeFunctions.scala
+ *   reduce     query/.../exec/AggrOverRangeVectors.scala:320-377 + aggregator/
+ *
+ * Parity pinning: the reference is Scala/JVM and cannot be built in this
+ * environment (no java/sbt — probed), so the oracle is pinned by
+ * golden vectors copied from the reference's own test sources into tests/
+ * (NibblePackTest.scala, RateFunctionsSpec.scala, AggrOverTimeFunctionsSpec.scala).
+ */
+#include <stdint.h>
+#include <string.h>
+#include <math.h>
+#include <stdlib.h>
+
+#include "../filodb_amd/csrc/chunk_format.h"
+
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
+#define EXPORT __attribute__((visibility("default")))
+
+/* ---- mirrored query structs (must match include/filodb_amd.h) ------------- */
+typedef struct {
+  int64_t start, step, end, window;
+  int32_t func_id, agg_id, num_groups, _pad;
+} fdb_query_t;
+typedef struct {
+  const uint8_t* blob;
+  int64_t blob_len;
+  const fdb_dir_entry_t* dir;
+  int64_t num_chunks;
+  const int32_t* series_first;
+  const int32_t* series_nchunks;
+  const int32_t* group_ids;
+  int32_t num_series, _pad;
+} fdb_view_t;
+
+enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
+       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10 };
+enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5 };
+
+static inline uint16_t rd_u16(const uint8_t* p) { uint16_t v; memcpy(&v, p, 2); return v; }
+static inline uint32_t rd_u32(const uint8_t* p) { uint32_t v; memcpy(&v, p, 4); return v; }
+static inline int32_t  rd_i32(const uint8_t* p) { int32_t v; memcpy(&v, p, 4); return v; }
+static inline int64_t  rd_i64(const uint8_t* p) { int64_t v; memcpy(&v, p, 8); return v; }
+static inline double   rd_f64(const uint8_t* p) { double v; memcpy(&v, p, 8); return v; }
+
+/* =========================================================================
+ * NibblePack unpack8 (NibblePack.scala:395-447) — for histogram work + tests
+ * ========================================================================= */
+EXPORT int32_t oracle_nibblepack_unpack8(const uint8_t* in, int32_t inlen,
+                                         int64_t out[8], int32_t* consumed) {
+  if (inlen < 1) return -1;
+  uint8_t nonzeroMask = in[0];
+  if (nonzeroMask == 0) {
+    for (int i = 0; i < 8; i++) out[i] = 0;
+    *consumed = 1;
+    return 0;
+  }
+  if (inlen < 2) return -1;
+  int numNibblesU8 = in[1] & 0xff;
+  int numBits = ((numNibblesU8 >> 4) + 1) * 4;
+  int trailingZeroes = (numNibblesU8 & 0x0f) * 4;
+  int totalBytes = 2 + (numBits * __builtin_popcount(nonzeroMask) + 7) / 8;
+  uint64_t mask = numBits >= 64 ? ~0ULL : ((1ULL << numBits) - 1);
+  int bufIndex = 2, bitCursor = 0;
+  uint64_t inWord = 0;
+  {
+    int i; uint64_t v = 0;
+    if (bufIndex + 8 <= inlen) memcpy(&v, in + bufIndex, 8);
+    else for (i = 0; bufIndex + i < inlen; i++) v |= (uint64_t)in[bufIndex + i] << (8 * i);
+    inWord = v;
+  }
+  bufIndex += 8;
+  for (int bit = 0; bit < 8; bit++) {
+    if (nonzeroMask & (1 << bit)) {
+      int remaining = 64 - bitCursor;
+      uint64_t outWord = (inWord >> bitCursor) & mask;
+      if (remaining <= numBits && bufIndex < totalBytes) {
+        if (bufIndex < inlen) {
+          uint64_t v = 0;
+          if (bufIndex + 8 <= inlen) memcpy(&v, in + bufIndex, 8);
+          else { for (int i = 0; bufIndex + i < inlen; i++) v |= (uint64_t)in[bufIndex + i] << (8 * i); }
+          inWord = v; bufIndex += 8;
+          if (remaining < numBits) outWord |= (inWord << remaining) & mask;
+        } else return -1;
+      }
+      out[bit] = (int64_t)(outWord << trailingZeroes);
+      bitCursor = (bitCursor + numBits) % 64;
+    } else out[bit] = 0;
+  }
+  *consumed = totalBytes;
+  return 0;
+}
+
+/* =========================================================================
+ * vector readers
+ * ========================================================================= */
+typedef struct {
+  const uint8_t* p;     /* vector base */
+  uint16_t wf;
+  int dropped;          /* bit 15 of u16 at +6 (BinaryVector.scala:519) */
+  int n;                /* element count */
+  /* DDV fields */
+  int64_t init; int32_t slope;
+  const uint8_t* idata; /* inner packed data (+20+8) */
+  int nbits, sign;
+} vec_t;
+
+static void vec_open(const uint8_t* p, vec_t* v) {
+  v->p = p;
+  v->wf = rd_u16(p + 4);
+  v->dropped = (rd_u16(p + 6) & FDB_DROP_MASK) != 0;
+  if (v->wf == FDB_WF_DDV) {
+    v->init = rd_i64(p + FDB_DDV_OFF_INIT);
+    v->slope = rd_i32(p + FDB_DDV_OFF_SLOPE);
+    const uint8_t* inner = p + FDB_DDV_OFF_INNER;
+    v->nbits = inner[6] & FDB_NBITS_MASK;
+    v->sign = (inner[6] & FDB_SIGN_MASK) != 0;
+    v->idata = inner + FDB_PRIM_OFF_DATA;
+    /* IntVectorDataReader.length (IntBinaryVector.scala:248-250) */
+    int numBytes = (int)rd_u32(inner);       /* reader numBytes = getInt(addr) */
+    int bitShift = inner[7] & 0x3f;
+    v->n = ((numBytes - 4) * 8 + (bitShift != 0 ? bitShift - 8 : 0)) / v->nbits;
+  } else if (v->wf == FDB_WF_DDV_CONST) {
+    v->n = rd_i32(p + FDB_DDVC_OFF_NELEM);
+    v->init = rd_i64(p + FDB_DDVC_OFF_INIT);
+    v->slope = rd_i32(p + FDB_DDVC_OFF_SLOPE);
+    v->idata = 0; v->nbits = 0; v->sign = 0;
+  } else { /* FDB_WF_PRIM64 raw f64/i64 */
+    v->n = ((int)rd_u32(p) - 4) / 8;          /* DoubleVectorDataReader.length */
+    v->idata = p + FDB_PRIM_OFF_DATA;
+    v->init = 0; v->slope = 0; v->nbits = 64; v->sign = 1;
+  }
+}
+
+/* inner packed-int element (IntBinaryVector.scala:306-439 readers) */
+static inline int64_t inner_at(const vec_t* v, int i) {
+  switch (v->nbits) {
+    case 32: return rd_i32(v->idata + 4 * (size_t)i);
+    case 16: { int32_t x = (int16_t)rd_u16(v->idata + 2 * (size_t)i);
+               return v->sign ? x : (x & 0xffff); }
+    case 8:  { int32_t x = (int8_t)v->idata[i];
+               return v->sign ? x : (x & 0xff); }
+    case 4:  return (v->idata[i / 2] >> ((i & 1) * 4)) & 0x0f;
+    case 2:  return (v->idata[i / 4] >> ((i & 3) * 2)) & 0x03;
+  }
+  return 0;
+}
+
+/* long element (DeltaDeltaVector.scala:153-156,241-242; LongVectorDataReader64) */
+static inline int64_t lv_at(const vec_t* v, int i) {
+  if (v->wf == FDB_WF_DDV) return v->init + (int64_t)v->slope * i + inner_at(v, i);
+  if (v->wf == FDB_WF_DDV_CONST) return v->init + (int64_t)v->slope * i;
+  return rd_i64(v->idata + 8 * (size_t)i);
+}
+
+/* binarySearch: first element >= item; bit31 set when no exact match
+ * (LongBinaryVector.scala:145-152 contract; DeltaDeltaVector.scala:159-188,245-253) */
+static int lv_binary_search(const vec_t* v, int64_t item) {
+  if (v->wf == FDB_WF_DDV_CONST) {
+    int64_t slope = v->slope;
+    int guess = slope == 0 ? (item <= v->init ? 0 : v->n)
+                           : (int)((item - v->init + (slope - 1)) / slope);
+    if (guess < 0) return (int)0x80000000;
+    if (guess >= v->n) return (int)(0x80000000u | (uint32_t)v->n);
+    if (item != lv_at(v, guess)) return (int)(0x80000000u | (uint32_t)guess);
+    return guess;
+  }
+  if (v->wf == FDB_WF_DDV) {
+    int64_t slope = v->slope;
+    int len = v->n;
+    int elemNo = slope == 0 ? (item <= v->init ? 0 : len)
+                            : (int)((item - v->init + (slope - 1)) / slope);
+    if (elemNo < 0) elemNo = 0;
+    if (elemNo >= len) elemNo = len - 1;
+    int64_t curBase = v->init + slope * (int64_t)elemNo;
+    while (elemNo >= 0 && item < curBase + inner_at(v, elemNo)) { elemNo--; curBase -= slope; }
+    if (elemNo >= 0 && item == curBase + inner_at(v, elemNo)) return elemNo;
+    elemNo++; curBase += slope;
+    while (elemNo < len && item > curBase + inner_at(v, elemNo)) { elemNo++; curBase += slope; }
+    if (elemNo < len && item == curBase + inner_at(v, elemNo)) return elemNo;
+    return (int)(0x80000000u | (uint32_t)elemNo);
+  }
+  /* raw i64: standard first >= (LongVectorDataReader64 binarySearch semantics) */
+  int lo = 0, hi = v->n;
+  while (lo < hi) { int mid = (lo + hi) >> 1; if (lv_at(v, mid) < item) lo = mid + 1; else hi = mid; }
+  if (lo < v->n && lv_at(v, lo) == item) return lo;
+  return (int)(0x80000000u | (uint32_t)lo);
+}
+
+/* ceilingIndex: last element <= item (LongBinaryVector.scala:162-169) */
+static inline int lv_ceiling(const vec_t* v, int64_t item) {
+  int r = lv_binary_search(v, item);
+  if (r < 0) return (r & 0x7fffffff) - 1;
+  return r;
+}
+
+/* long sum → double (DeltaDeltaConstDataReader.slopeSum :265-268) */
+static inline double slope_sum(int64_t init, int32_t slope, int start, int end) {
+  int len = end - start + 1;
+  return (double)len * (double)(init + (int64_t)start * slope)
+       + (double)(((int64_t)((end - start) * len / 2)) * slope);
+}
+
+static double lv_sum(const vec_t* v, int start, int end) {
+  if (v->wf == FDB_WF_DDV_CONST) return slope_sum(v->init, v->slope, start, end);
+  if (v->wf == FDB_WF_DDV) {
+    int64_t s = 0;
+    for (int i = start; i <= end; i++) s += inner_at(v, i);
+    return slope_sum(v->init, v->slope, start, end) + (double)s;
+  }
+  double s = 0;
+  for (int i = start; i <= end; i++) s += (double)lv_at(v, i);
+  return s;
+}
+
+/* long changes (DeltaDeltaVector.scala:212-228,280-289) */
+static void lv_changes(const vec_t* v, int start, int end, int64_t prev, int ignorePrev,
+                       int64_t* out_changes, int64_t* out_prev) {
+  if (v->wf == FDB_WF_DDV_CONST) {
+    int64_t firstValue = lv_at(v, start), lastValue = lv_at(v, end);
+    int64_t ch = (!ignorePrev && prev != firstValue) ? 1 : 0;
+    *out_changes = v->slope == 0 ? ch : (end - start) + ch;
+    *out_prev = lastValue;
+    return;
+  }
+  int64_t prevV = prev, ch = 0;
+  for (int i = start; i <= end; i++) {
+    int64_t cur = lv_at(v, i);
+    if (i == start && ignorePrev) prevV = cur;
+    if (prevV != cur) ch++;
+    prevV = cur;
+  }
+  *out_changes = ch; *out_prev = prevV;
+}
+
+/* ---- double reader over a value vector ----------------------------------- */
+static inline double dv_at(const vec_t* v, int i) {
+  if (v->wf == FDB_WF_PRIM64) return rd_f64(v->idata + 8 * (size_t)i);
+  return (double)lv_at(v, i);
+}
+
+/* NaN-skipping sum (DoubleVectorDataReader64.sum :234-262; DDV wrap :553-554) */
+static double dv_sum(const vec_t* v, int start, int end) {
+  if (v->wf != FDB_WF_PRIM64) return lv_sum(v, start, end);
+  double sum = NAN;
+  for (int i = start; i <= end; i++) {
+    double x = rd_f64(v->idata + 8 * (size_t)i);
+    if (!isnan(x)) { if (isnan(sum)) sum = 0; sum += x; }
+  }
+  return sum;
+}
+
+static int dv_count(const vec_t* v, int start, int end) {
+  if (v->wf != FDB_WF_PRIM64) return end - start + 1;   /* DoubleLongWrap :555 */
+  int c = 0;
+  for (int i = start; i <= end; i++) if (!isnan(rd_f64(v->idata + 8 * (size_t)i))) c++;
+  return c;
+}
+
+/* double changes (DoubleVectorDataReader64.changes :283-302;
+ * DoubleLongWrapDataReader.changes :559-566) */
+static void dv_changes(const vec_t* v, int start, int end, double prev,
+                       double* out_changes, double* out_prev) {
+  if (v->wf != FDB_WF_PRIM64) {
+    int ignorePrev = isnan(prev);
+    int64_t ch, pv;
+    lv_changes(v, start, end, isnan(prev) ? 0 : (int64_t)prev, ignorePrev, &ch, &pv);
+    *out_changes = (double)ch; *out_prev = (double)pv;
+    return;
+  }
+  double ch = 0, prevV = prev;
+  for (int i = start; i <= end; i++) {
+    double x = rd_f64(v->idata + 8 * (size_t)i);
+    if (!isnan(x) && prevV != x && !isnan(prevV)) ch += 1;
+    prevV = x;
+  }
+  *out_changes = ch; *out_prev = prevV;
+}
+
+/* =========================================================================
+ * counter correction (DoubleVector.scala:177-211,305-392)
+ * ========================================================================= */
+typedef struct {
+  int has;                /* 0 = NoCorrection */
+  double lastValue;
+  double correction;
+} corr_meta_t;
+
+/* per-chunk corrected-state: built lazily once per (window,chunk) visit.
+ * corrected[i] = nan-zeroed value + in-chunk correction (CorrectingDoubleVectorReader
+ * :325-342). scratch must hold num_rows doubles. */
+typedef struct {
+  const vec_t* v;
+  double* corrected;      /* NULL when chunk has no drop bit */
+  double chunk_correction; /* _correction total */
+} cread_t;
+
+static void cread_init(const vec_t* v, double* scratch, cread_t* r) {
+  r->v = v;
+  r->chunk_correction = 0;
+  if (!v->dropped) { r->corrected = 0; return; }
+  r->corrected = scratch;
+  double corr = 0, last = -1.7976931348623157e308; /* Double.MinValue */
+  for (int i = 0; i < v->n; i++) {
+    double x = dv_at(v, i);
+    if (isnan(x)) x = 0;
+    if (x < last) corr += last;
+    r->corrected[i] = x + corr;
+    last = x;
+  }
+  r->chunk_correction = corr;
+}
+
+static inline double corrected_value(const cread_t* r, int n, const corr_meta_t* m) {
+  double corr = m->has ? m->correction : 0;
+  if (r->corrected) return r->corrected[n] + corr;
+  return dv_at(r->v, n) + corr;
+}
+
+static void detect_drop_and_correction(const vec_t* v, corr_meta_t* m) {
+  if (!m->has) return;
+  double first = dv_at(v, 0);
+  if (isnan(first) || first < m->lastValue) m->correction += m->lastValue;
+}
+
+static void update_correction(const cread_t* r, corr_meta_t* m) {
+  const vec_t* v = r->v;
+  double lastValue;
+  if (r->corrected) {      /* CorrectingDoubleVectorReader.updateCorrection :375-391 */
+    int index = v->n - 1;
+    lastValue = dv_at(v, index); index--;
+    while (isnan(lastValue) && index >= 0) { lastValue = dv_at(v, index); index--; }
+    if (isnan(lastValue)) lastValue = 0;
+    m->correction = (m->has ? m->correction : 0) + r->chunk_correction;
+  } else {                 /* default updateCorrection :190-195 */
+    lastValue = dv_at(v, v->n - 1);
+    if (!m->has) m->correction = 0;
+  }
+  m->lastValue = lastValue;
+  m->has = 1;
+}
+
+/* =========================================================================
+ * extrapolatedRate (RateFunctions.scala:72-111)
+ * ========================================================================= */
+static double extrapolated_rate(int64_t windowStart, int64_t windowEnd, int numSamples,
+                                int64_t t1, double v1, int64_t t2, double v2,
+                                int isCounter, int isRate) {
+  double durationToStart = (double)(t1 - windowStart) / 1000.0;
+  double durationToEnd = (double)(windowEnd - t2) / 1000.0;
+  double sampledInterval = (double)(t2 - t1) / 1000.0;
+  double averageDurationBetweenSamples = sampledInterval / ((double)numSamples - 1);
+  double delta = v2 - v1;
+  if (isCounter && delta > 0 && v1 >= 0) {
+    double durationToZero = sampledInterval * (v1 / delta);
+    if (durationToZero < durationToStart) durationToStart = durationToZero;
+  }
+  double extrapolationThreshold = averageDurationBetweenSamples * 1.1;
+  double extrapolateToInterval = sampledInterval;
+  extrapolateToInterval += (durationToStart < extrapolationThreshold)
+                             ? durationToStart : averageDurationBetweenSamples / 2;
+  extrapolateToInterval += (durationToEnd < extrapolationThreshold)
+                             ? durationToEnd : averageDurationBetweenSamples / 2;
+  double scaledDelta = delta * (extrapolateToInterval / sampledInterval);
+  return isRate ? (scaledDelta / (double)(windowEnd - windowStart) * 1000.0) : scaledDelta;
+}
+
+/* =========================================================================
+ * per-series window evaluation (ChunkedWindowIterator.doNext,
+ * PeriodicSamplesMapper.scala:293-330 + WindowedChunkIterator,
+ * ChunkSetInfo.scala:467-529; function semantics DESIGN.md §3)
+ * ========================================================================= */
+typedef struct { double* scratch; } eval_ctx_t;
+
+static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
+                        eval_ctx_t* ctx, double* out /* numWindows */) {
+  int nw = (int)((q->end - q->start) / q->step) + 1;
+  int first = view->series_first[sid];
+  int nchunks = view->series_nchunks[sid];
+  const fdb_dir_entry_t* dir = view->dir + first;
+  int is_rate_family = q->func_id <= FN_DELTA;
+
+  /* pre-open vectors once per series */
+  vec_t tsv[64], vav[64];
+  cread_t cr[64];
+  double* scratch = ctx->scratch;
+  if (nchunks > 64) nchunks = 64;  /* oracle cap; builder max_rows keeps chunks few */
+  for (int c = 0; c < nchunks; c++) {
+    vec_open(view->blob + dir[c].ts_off, &tsv[c]);
+    vec_open(view->blob + dir[c].val_off, &vav[c]);
+    if (is_rate_family) {
+      cread_init(&vav[c], scratch + (size_t)c * 512, &cr[c]);
+    }
+  }
+
+  for (int w = 0; w < nw; w++) {
+    int64_t wEnd = q->start + (int64_t)w * q->step;
+    int64_t wStart = wEnd - q->window;   /* inclusive-range=true (ChunkSetInfo.scala:470-473) */
+    double result = NAN;
+
+    if (is_rate_family) {
+      /* ChunkedRateFunctionBase (RateFunctions.scala:230-289) +
+       * CounterChunkedRangeFunction.addChunks (RangeFunction.scala:138-163) */
+      corr_meta_t meta = {0, 0, 0};
+      int numSamples = 0;
+      int64_t lowestTime = INT64_MAX, highestTime = 0;
+      double lowestValue = NAN, highestValue = NAN;
+      int isCounter = q->func_id != FN_DELTA;
+      for (int c = 0; c < nchunks; c++) {
+        if (dir[c].end_time < wStart) continue;   /* WindowedChunkIterator drop rule */
+        const vec_t* tv = &tsv[c];
+        int startRow = lv_binary_search(tv, wStart) & 0x7fffffff;
+        int endRow = lv_ceiling(tv, wEnd);
+        if (endRow > dir[c].num_rows - 1) endRow = dir[c].num_rows - 1;
+        if (isCounter) detect_drop_and_correction(&vav[c], &meta);
+        if (startRow <= endRow) {
+          int skip = 0;
+          if (isCounter && startRow == 0 && endRow == 0 && isnan(dv_at(&vav[c], 0)))
+            skip = 1;  /* single-row NaN chunk (RateFunctions.scala:253-255) */
+          if (!skip) {
+            int64_t st = lv_at(tv, startRow), en = lv_at(tv, endRow);
+            if (st < lowestTime || en > highestTime) {
+              numSamples += endRow - startRow + 1;
+              if (st < lowestTime) {
+                lowestTime = st;
+                lowestValue = isCounter ? corrected_value(&cr[c], startRow, &meta)
+                                        : dv_at(&vav[c], startRow);
+              }
+              if (en > highestTime) {
+                highestTime = en;
+                highestValue = isCounter ? corrected_value(&cr[c], endRow, &meta)
+                                         : dv_at(&vav[c], endRow);
+              }
+            }
+          }
+        }
+        if (isCounter) update_correction(&cr[c], &meta);
+        if (dir[c].end_time >= wEnd) break;       /* add-while rule */
+      }
+      if (highestTime > lowestTime) {
+        result = extrapolated_rate(wStart, wEnd, numSamples,
+                                   lowestTime, lowestValue, highestTime, highestValue,
+                                   q->func_id != FN_DELTA, q->func_id == FN_RATE);
+      }
+    } else {
+      /* gauge family state (AggrOverTimeFunctions.scala) */
+      double sum = NAN, count = NAN, sqsum = NAN, mn = NAN, mx = NAN;
+      double changes = NAN, prev = NAN;
+      int icount = 0;
+      for (int c = 0; c < nchunks; c++) {
+        if (dir[c].end_time < wStart) continue;
+        const vec_t* tv = &tsv[c];
+        const vec_t* vv = &vav[c];
+        int startRow = lv_binary_search(tv, wStart) & 0x7fffffff;
+        int endRow = lv_ceiling(tv, wEnd);
+        if (endRow > dir[c].num_rows - 1) endRow = dir[c].num_rows - 1;
+        if (startRow <= endRow) {
+          switch (q->func_id) {
+            case FN_SUM: {
+              double cs = dv_sum(vv, startRow, endRow);
+              if (!isnan(cs) && isnan(sum)) sum = 0;
+              sum += cs;                       /* :560-572 incl. NaN-poison quirk */
+            } break;
+            case FN_COUNT: {
+              if (isnan(count)) count = 0;     /* :943-958 */
+              count += dv_count(vv, startRow, endRow);
+            } break;
+            case FN_AVG: {
+              double cs = dv_sum(vv, startRow, endRow);
+              if (!isnan(cs) && isnan(sum)) sum = 0;
+              sum += cs;
+              icount += dv_count(vv, startRow, endRow);   /* :1004-1016 */
+            } break;
+            case FN_MIN: case FN_MAX: {
+              for (int i = startRow; i <= endRow; i++) {
+                double x = dv_at(vv, i);
+                if (isnan(x)) continue;        /* QueryUtils.minIgnoreNaN */
+                if (q->func_id == FN_MIN) mn = isnan(mn) || x < mn ? x : mn;
+                else mx = isnan(mx) || x > mx ? x : mx;
+              }
+            } break;
+            case FN_STDDEV: case FN_STDVAR: {
+              /* VarOverTimeChunkedFunctionD :1082-1115 */
+              double cs = NAN, csq = NAN; int cc = 0;
+              for (int i = startRow; i <= endRow; i++) {
+                double x = dv_at(vv, i);
+                if (!isnan(x)) {
+                  if (isnan(cs)) cs = 0;
+                  if (isnan(csq)) csq = 0;
+                  cs += x; csq += x * x; cc++;
+                }
+              }
+              if (!isnan(cs) && isnan(sum)) sum = 0;
+              sum += cs;
+              if (!isnan(csq) && isnan(sqsum)) sqsum = 0;
+              sqsum += csq;
+              icount += cc;
+            } break;
+            case FN_CHANGES: {
+              if (isnan(changes)) changes = 0; /* :1185-1210 */
+              double ch, pv;
+              dv_changes(vv, startRow, endRow, prev, &ch, &pv);
+              changes += ch; prev = pv;
+            } break;
+          }
+        }
+        if (dir[c].end_time >= wEnd) break;
+      }
+      switch (q->func_id) {
+        case FN_SUM:   result = sum; break;
+        case FN_COUNT: result = count; break;
+        case FN_AVG:   result = icount > 0 ? sum / icount : (isnan(sum) ? sum : 0); break;
+        case FN_MIN:   result = mn; break;
+        case FN_MAX:   result = mx; break;
+        case FN_STDDEV: case FN_STDVAR: {
+          double r;
+          if (icount > 0) {
+            double avg = sum / icount;
+            r = sqsum / icount - avg * avg;
+            if (q->func_id == FN_STDDEV) r = sqrt(r);
+          } else if (isnan(sum)) r = sum;
+          else r = 0;
+          result = r;
+        } break;
+        case FN_CHANGES: result = changes; break;
+      }
+    }
+    out[w] = result;
+  }
+}
+
+/* =========================================================================
+ * whole-query execution + fastReduce
+ * (AggrOverRangeVectors.scala:320-377; aggregator/{Sum,Count,Min,Max,Avg}RowAggregator)
+ * ========================================================================= */
+EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
+                                 double* out, double* out_counts, int32_t nthreads) {
+  int nw = (int)((q->end - q->start) / q->step) + 1;
+  int ns = view->num_series;
+  if (q->agg_id == AGG_NONE) {
+#ifdef _OPENMP
+    omp_set_num_threads(nthreads > 0 ? nthreads : 1);
+    #pragma omp parallel
+    {
+      eval_ctx_t ctx; ctx.scratch = (double*)malloc(64 * 512 * sizeof(double));
+      #pragma omp for schedule(static)
+      for (int s = 0; s < ns; s++) eval_series(view, s, q, &ctx, out + (size_t)s * nw);
+      free(ctx.scratch);
+    }
+#else
+    eval_ctx_t ctx; ctx.scratch = (double*)malloc(64 * 512 * sizeof(double));
+    for (int s = 0; s < ns; s++) eval_series(view, s, q, &ctx, out + (size_t)s * nw);
+    free(ctx.scratch);
+#endif
+    return 0;
+  }
+
+  int ng = q->num_groups;
+  size_t gridlen = (size_t)ng * nw;
+  for (size_t i = 0; i < gridlen; i++) out[i] = NAN;
+  double* counts = out_counts;
+  double* owned_counts = 0;
+  if (q->agg_id == AGG_AVG && !counts) {
+    owned_counts = (double*)malloc(gridlen * sizeof(double));
+    counts = owned_counts;
+  }
+  if (counts) for (size_t i = 0; i < gridlen; i++) counts[i] = 0;
+
+  int T = nthreads > 0 ? nthreads : 1;
+#ifdef _OPENMP
+  omp_set_num_threads(T);
+#else
+  T = 1;
+#endif
+  /* per-thread partial grids, merged with the same RowAggregator semantics */
+  double* pg = (double*)malloc((size_t)T * gridlen * sizeof(double));
+  double* pc = (double*)malloc((size_t)T * gridlen * sizeof(double));
+  for (size_t i = 0; i < (size_t)T * gridlen; i++) { pg[i] = NAN; pc[i] = 0; }
+
+#ifdef _OPENMP
+  #pragma omp parallel
+#endif
+  {
+#ifdef _OPENMP
+    int t = omp_get_thread_num();
+#else
+    int t = 0;
+#endif
+    double* g = pg + (size_t)t * gridlen;
+    double* gc = pc + (size_t)t * gridlen;
+    eval_ctx_t ctx; ctx.scratch = (double*)malloc(64 * 512 * sizeof(double));
+    double* row = (double*)malloc((size_t)nw * sizeof(double));
+#ifdef _OPENMP
+    #pragma omp for schedule(static)
+#endif
+    for (int s = 0; s < ns; s++) {
+      eval_series(view, s, q, &ctx, row);
+      int grp = view->group_ids[s];
+      double* acc = g + (size_t)grp * nw;
+      double* accc = gc + (size_t)grp * nw;
+      for (int w = 0; w < nw; w++) {
+        double x = row[w];
+        switch (q->agg_id) {
+          case AGG_SUM:   /* SumRowAggregator.scala:23-29 */
+            if (!isnan(x)) { if (isnan(acc[w])) acc[w] = 0; acc[w] += x; }
+            break;
+          case AGG_COUNT:
+            /* CountRowAggregator.scala:36-42: sample maps to 1 (0 when NaN); the
+             * accumulator leaves NaN only when every mapped value was 0. */
+            if (!isnan(x)) { if (isnan(acc[w])) acc[w] = 0; acc[w] += 1; }
+            break;
+          case AGG_MIN:
+            if (!isnan(x) && (isnan(acc[w]) || x < acc[w])) acc[w] = x;
+            break;
+          case AGG_MAX:
+            if (!isnan(x) && (isnan(acc[w]) || x > acc[w])) acc[w] = x;
+            break;
+          case AGG_AVG:   /* AvgRowAggregator: weighted mean ≡ sum/count */
+            if (!isnan(x)) { if (isnan(acc[w])) acc[w] = 0; acc[w] += x; accc[w] += 1; }
+            break;
+        }
+      }
+    }
+    free(ctx.scratch); free(row);
+  }
+
+  /* merge thread partials (same op; associative for these aggregators) */
+  for (int t = 0; t < T; t++) {
+    double* g = pg + (size_t)t * gridlen;
+    double* gc = pc + (size_t)t * gridlen;
+    for (size_t i = 0; i < gridlen; i++) {
+      double x = g[i];
+      switch (q->agg_id) {
+        case AGG_SUM:
+          if (!isnan(x)) { if (isnan(out[i])) out[i] = 0; out[i] += x; }
+          break;
+        case AGG_COUNT:
+          if (!isnan(x)) { if (isnan(out[i])) out[i] = 0; out[i] += x; }
+          break;
+        case AGG_MIN:
+          if (!isnan(x) && (isnan(out[i]) || x < out[i])) out[i] = x;
+          break;
+        case AGG_MAX:
+          if (!isnan(x) && (isnan(out[i]) || x > out[i])) out[i] = x;
+          break;
+        case AGG_AVG:
+          if (!isnan(x)) { if (isnan(out[i])) out[i] = 0; out[i] += x; counts[i] += gc[i]; }
+          break;
+      }
+    }
+  }
+  if (q->agg_id == AGG_AVG) {
+    /* present: mean = sum/count (AvgRowAggregator.scala:38-46 algebraically) */
+    for (size_t i = 0; i < gridlen; i++)
+      if (counts[i] > 0) out[i] = out[i] / counts[i];
+  }
+  free(pg); free(pc);
+  if (owned_counts) free(owned_counts);
+  return 0;
+}
+
+/* convenience: evaluate one series only (tests) */
+EXPORT int32_t oracle_eval_series(const fdb_view_t* view, int32_t sid, const fdb_query_t* q,
+                                  double* out) {
+  eval_ctx_t ctx; ctx.scratch = (double*)malloc(64 * 512 * sizeof(double));
+  eval_series(view, sid, q, &ctx, out);
+  free(ctx.scratch);
+  return 0;
+}
+
+/* decode helpers for unit tests */
+EXPORT int32_t oracle_decode_longs(const uint8_t* vec, int64_t* out, int32_t cap) {
+  vec_t v; vec_open(vec, &v);
+  if (v.n > cap) return -1;
+  for (int i = 0; i < v.n; i++) out[i] = lv_at(&v, i);
+  return v.n;
+}
+EXPORT int32_t oracle_decode_doubles(const uint8_t* vec, double* out, int32_t cap) {
+  vec_t v; vec_open(vec, &v);
+  if (v.n > cap) return -1;
+  for (int i = 0; i < v.n; i++) out[i] = dv_at(&v, i);
+  return v.n;
+}
+EXPORT int32_t oracle_vec_info(const uint8_t* vec, int32_t* wf, int32_t* n,
+                               int32_t* nbits, int32_t* dropped) {
+  vec_t v; vec_open(vec, &v);
+  *wf = v.wf; *n = v.n; *nbits = v.nbits; *dropped = v.dropped;
+  return 0;
+}
+EXPORT int32_t oracle_binary_search(const uint8_t* vec, int64_t item) {
+  vec_t v; vec_open(vec, &v);
+  return lv_binary_search(&v, item);
+}
+EXPORT double oracle_extrapolated_rate(int64_t ws, int64_t we, int32_t n,
+                                       int64_t t1, double v1, int64_t t2, double v2,
+                                       int32_t isCounter, int32_t isRate) {
+  return extrapolated_rate(ws, we, n, t1, v1, t2, v2, isCounter, isRate);
+}

@@ -1,0 +1,118 @@
+"""ctypes client for the TEST-ONLY CPU oracle (oracle/liboracle.so).
+
+May be imported only by tests/, __graft_entry__.smoke() and bench.py's
+cpu_baseline leg. The product GPU path never touches this library.
+"""
+import ctypes
+import os
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+LIB_PATH = os.path.join(_DIR, "liboracle.so")
+
+_lib = None
+_c_double_p = ctypes.POINTER(ctypes.c_double)
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        if not os.path.exists(LIB_PATH):
+            raise RuntimeError(f"{LIB_PATH} not built — run `make -C oracle`")
+        L = ctypes.CDLL(LIB_PATH)
+        L.oracle_query_exec.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                        _c_double_p, _c_double_p, ctypes.c_int32]
+        L.oracle_eval_series.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                         ctypes.c_void_p, _c_double_p]
+        L.oracle_decode_longs.argtypes = [ctypes.POINTER(ctypes.c_uint8),
+                                          ctypes.POINTER(ctypes.c_int64), ctypes.c_int32]
+        L.oracle_decode_doubles.argtypes = [ctypes.POINTER(ctypes.c_uint8),
+                                            _c_double_p, ctypes.c_int32]
+        L.oracle_vec_info.argtypes = [ctypes.POINTER(ctypes.c_uint8)] + \
+            [ctypes.POINTER(ctypes.c_int32)] * 4
+        L.oracle_binary_search.argtypes = [ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64]
+        L.oracle_nibblepack_unpack8.argtypes = [ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32,
+                                                ctypes.POINTER(ctypes.c_int64),
+                                                ctypes.POINTER(ctypes.c_int32)]
+        L.oracle_extrapolated_rate.restype = ctypes.c_double
+        L.oracle_extrapolated_rate.argtypes = [ctypes.c_int64, ctypes.c_int64, ctypes.c_int32,
+                                               ctypes.c_int64, ctypes.c_double,
+                                               ctypes.c_int64, ctypes.c_double,
+                                               ctypes.c_int32, ctypes.c_int32]
+        _lib = L
+    return _lib
+
+
+def query_exec(view, q, num_series, num_windows, out_counts=False, nthreads=1):
+    """Runs the oracle over a store view (filodb_amd.View) with query q
+    (filodb_amd.Query). Returns the result grid as numpy."""
+    if q.agg_id == 0:
+        out = np.empty(num_series * num_windows, dtype=np.float64)
+        cnt = None
+    else:
+        out = np.empty(q.num_groups * num_windows, dtype=np.float64)
+        cnt = np.zeros(q.num_groups * num_windows, dtype=np.float64) if out_counts else None
+    rc = lib().oracle_query_exec(
+        ctypes.byref(view), ctypes.byref(q),
+        out.ctypes.data_as(_c_double_p),
+        cnt.ctypes.data_as(_c_double_p) if cnt is not None else None,
+        nthreads)
+    if rc != 0:
+        raise RuntimeError("oracle_query_exec failed")
+    return (out, cnt) if out_counts else out
+
+
+def eval_series(view, sid, q, num_windows):
+    out = np.empty(num_windows, dtype=np.float64)
+    rc = lib().oracle_eval_series(ctypes.byref(view), sid, ctypes.byref(q),
+                                  out.ctypes.data_as(_c_double_p))
+    if rc != 0:
+        raise RuntimeError("oracle_eval_series failed")
+    return out
+
+
+def _u8(data):
+    return (ctypes.c_uint8 * len(data)).from_buffer_copy(data)
+
+
+def decode_longs(vec_bytes, cap=100000):
+    out = np.empty(cap, dtype=np.int64)
+    n = lib().oracle_decode_longs(_u8(vec_bytes),
+                                  out.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)), cap)
+    if n < 0:
+        raise RuntimeError("decode_longs failed")
+    return out[:n].copy()
+
+
+def decode_doubles(vec_bytes, cap=100000):
+    out = np.empty(cap, dtype=np.float64)
+    n = lib().oracle_decode_doubles(_u8(vec_bytes), out.ctypes.data_as(_c_double_p), cap)
+    if n < 0:
+        raise RuntimeError("decode_doubles failed")
+    return out[:n].copy()
+
+
+def vec_info(vec_bytes):
+    wf, n, nbits, dropped = (ctypes.c_int32() for _ in range(4))
+    lib().oracle_vec_info(_u8(vec_bytes), ctypes.byref(wf), ctypes.byref(n),
+                          ctypes.byref(nbits), ctypes.byref(dropped))
+    return {"wf": wf.value, "n": n.value, "nbits": nbits.value, "dropped": bool(dropped.value)}
+
+
+def binary_search(vec_bytes, item):
+    return lib().oracle_binary_search(_u8(vec_bytes), item)
+
+
+def nibblepack_unpack8(data):
+    out = (ctypes.c_int64 * 8)()
+    consumed = ctypes.c_int32()
+    rc = lib().oracle_nibblepack_unpack8(_u8(data), len(data), out, ctypes.byref(consumed))
+    if rc != 0:
+        raise RuntimeError("unpack8 failed")
+    return list(out), consumed.value
+
+
+def extrapolated_rate(ws, we, n, t1, v1, t2, v2, is_counter, is_rate):
+    return lib().oracle_extrapolated_rate(ws, we, n, t1, v1, t2, v2,
+                                          1 if is_counter else 0, 1 if is_rate else 0)
