@@ -1,0 +1,228 @@
+#!/usr/bin/env python3
+"""bench.py — measures BASELINE.json's metric on its named config.
+
+Metric: "samples/sec scanned+aggregated, 1M series rate()[5m]" — whole-job
+samples scanned+aggregated per second. Default (N=1) workload is BASELINE
+configs[1]: 1M Prometheus counter series, 1h@15s, rate()[5m] step=15s, on one
+MI355X. A "step" = one execution of the hot path (one fdb_query_exec launch)
+over the resident dataset; inputs are built host-side and uploaded to HBM
+BEFORE the timed region (DESIGN.md §5 — no PCIe in the timed region).
+
+N>1 (launched by the driver via torch.distributed.run): weak scaling — each
+rank owns its own 1M-series shard on its GPU (the reference's shard model;
+series are independent until the cross-series reduce, which configs[1] does not
+include — see SURVEY.md §8e). value aggregates all ranks' samples.
+
+Roofline: HBM-bound. achieved = algorithmic bytes per launch (encoded chunk
+payload + chunk directory reads + output grid writes) ÷ avg kernel ms from HIP
+events on the engine's stream. traffic: rocprofv3 PMC-derived bytes per launch
+via env FDB_TRAFFIC_BYTES_PER_LAUNCH (collected separately; profiles/), else null.
+
+cpu_baseline: the CPU oracle (reference-algorithm restatement, kind "port")
+timed on the host cores over a bounded sample of the same workload (N=1 rank 0
+only).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+import filodb_amd as fdb  # noqa: E402
+
+T0 = 100000  # epoch-ms origin of the synthetic hour
+
+WORKLOADS = {
+    # BASELINE configs[1]: the headline config
+    "rate_1m": dict(kind="counter", n_series=1_000_000, n_samples=240,
+                    step_ms=15000, func=fdb.FN_RATE, agg=fdb.AGG_NONE,
+                    window=300_000, qstep=15000, n_groups=1000,
+                    label="1M counter series 1h@15s, rate()[5m] step=15s"),
+    # BASELINE configs[2]: gauge windows (LDS sliding)
+    "gauge_1m": dict(kind="gauge", n_series=1_000_000, n_samples=240,
+                     step_ms=15000, func=fdb.FN_AVG_OVER_TIME, agg=fdb.AGG_NONE,
+                     window=600_000, qstep=15000, n_groups=1000,
+                     label="1M gauge series 1h@15s, avg_over_time[10m]"),
+    # BASELINE configs[4] shape (per-GPU shard): sum by(job)(rate[5m])
+    "sumrate_1m": dict(kind="counter", n_series=1_000_000, n_samples=240,
+                       step_ms=15000, func=fdb.FN_RATE, agg=fdb.AGG_SUM,
+                       window=300_000, qstep=15000, n_groups=1000,
+                       label="1M counter series, sum by(job)(rate()[5m]), 1000 groups"),
+    "smoke": dict(kind="counter", n_series=20_000, n_samples=240,
+                  step_ms=15000, func=fdb.FN_RATE, agg=fdb.AGG_NONE,
+                  window=300_000, qstep=15000, n_groups=100,
+                  label="20k counter series (reduced)"),
+}
+
+
+def build_store(w, rank):
+    st = fdb.ChunkStore()
+    kind = fdb.COL_COUNTER if w["kind"] == "counter" else fdb.COL_GAUGE
+    st.synth_generate(kind, w["n_series"], w["n_samples"], start_ts=T0,
+                      step_ms=w["step_ms"], jitter_ms=250, lam=10.0,
+                      reset_p=0.001, n_groups=w["n_groups"], seed=42 + rank)
+    st.seal()
+    return st
+
+
+def make_query(w):
+    span = w["n_samples"] * w["step_ms"]
+    return fdb.make_query(T0, w["qstep"], T0 + span, w["window"], w["func"],
+                          w["agg"], w["n_groups"] if w["agg"] != fdb.AGG_NONE else 0)
+
+
+def cpu_baseline(w, budget_s=12.0):
+    """Oracle (kind 'port') on host cores over a bounded sample of the workload."""
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import pyclient as oracle
+    cores = os.cpu_count() or 1
+    n = 2000
+    q = make_query(w)
+    elapsed, samples = 0.0, 0
+    while True:
+        sub = dict(w)
+        sub["n_series"] = n
+        st = build_store(sub, rank=0)
+        nw = q.num_windows
+        t0 = time.perf_counter()
+        if w["agg"] == fdb.AGG_NONE:
+            oracle.query_exec(st.view(), q, st.num_series, nw, nthreads=cores)
+        else:
+            oracle.query_exec(st.view(), q, st.num_series, nw, nthreads=cores)
+        elapsed = time.perf_counter() - t0
+        samples = n * w["n_samples"]
+        if elapsed > 3.0 or n >= w["n_series"]:
+            break
+        n = min(w["n_series"], int(n * max(2, 6.0 / max(elapsed, 0.05))))
+    return {
+        "value": samples / elapsed,
+        "unit": "samples/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": f"{n} series x {w['n_samples']} samples of the same workload, "
+                  f"OpenMP {cores} threads, {elapsed:.1f}s",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--workload", default="rate_1m", choices=list(WORKLOADS))
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    n_gpus = max(world, args.gpus)
+
+    dist = None
+    if world > 1:
+        import torch
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+
+    import torch
+    w = WORKLOADS[args.workload]
+    st = build_store(w, rank)
+    q = make_query(w)
+    nw = q.num_windows
+
+    eng = fdb.Engine(local_rank)
+    ds = eng.upload(st)
+    samples_per_step = ds.total_samples
+
+    # output buffers resident on device (torch tensors so an RCCL reduce could
+    # consume them directly); allocated before the timed region
+    if q.agg_id == fdb.AGG_NONE:
+        out = torch.empty(st.num_series * nw, dtype=torch.float64,
+                          device=f"cuda:{local_rank}")
+        cnt = None
+    else:
+        out = torch.empty(q.num_groups * nw, dtype=torch.float64,
+                          device=f"cuda:{local_rank}")
+        cnt = torch.empty_like(out)
+
+    # warmup pass (also yields avg HIP-event kernel ms for the roofline), then
+    # wall-time EXACTLY `steps` launches between barriers+synchronize.
+    kernel_ms = eng.bench(ds, q, out, out_counts=cnt, on_device=True,
+                          warmup=args.warmup, iters=args.steps)
+    if dist:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    _ = eng.bench(ds, q, out, out_counts=cnt, on_device=True,
+                  warmup=0, iters=args.steps)
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    ms_per_step = elapsed / args.steps * 1000
+    if dist:
+        t = torch.tensor([ms_per_step], device=f"cuda:{local_rank}")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        ms_per_step = float(t.item())
+
+    total_samples_per_step = samples_per_step * n_gpus
+    value = total_samples_per_step / (ms_per_step / 1000.0)
+
+    if rank == 0:
+        # algorithmic bytes per launch: encoded chunk payload read + output
+        # grid written (directory reads are <1% and omitted)
+        out_bytes = out.numel() * 8 + (cnt.numel() * 8 if cnt is not None else 0)
+        algo_bytes = ds.payload_bytes + out_bytes
+        ach = algo_bytes / (kernel_ms / 1000.0) if kernel_ms > 0 else None
+        traffic = os.environ.get("FDB_TRAFFIC_BYTES_PER_LAUNCH")
+        roofline = {
+            "bound": "hbm",
+            "achieved": round(ach / 1e9, 2) if ach else None,
+            "peak": 8000.0,
+            "unit": "GB/s",
+            "frac": round(ach / 8e12, 4) if ach else None,
+            "traffic": float(traffic) if traffic else None,
+        }
+        cb = None
+        if not args.no_cpu_baseline and n_gpus == 1:
+            cb = cpu_baseline(w)
+        line = {
+            "metric": "samples/sec scanned+aggregated, 1M series rate()[5m]",
+            "value": value,
+            "unit": "samples/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "kernel_ms_per_step": kernel_ms,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,   # no published reference numbers
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {
+                "workload": w["label"],
+                "series_per_gpu": w["n_series"],
+                "samples_per_series": w["n_samples"],
+                "windows": nw,
+                "payload_bytes_per_gpu": ds.payload_bytes,
+            },
+            "roofline": roofline,
+            "cpu_baseline": cb,
+        }
+        print(json.dumps(line), flush=True)
+
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

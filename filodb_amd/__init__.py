@@ -87,6 +87,10 @@ def lib():
         L.fdb_chunk_get.argtypes = [ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32,
                                     ctypes.POINTER(ChunkInfo)]
         L.fdb_store_view.argtypes = [ctypes.c_void_p, ctypes.POINTER(View)]
+        L.fdb_synth_generate.argtypes = [ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32,
+                                         ctypes.c_int32, ctypes.c_int64, ctypes.c_int32,
+                                         ctypes.c_int32, ctypes.c_double, ctypes.c_double,
+                                         ctypes.c_int32, ctypes.c_uint64]
         L.fdb_nibblepack_pack8.argtypes = [ctypes.POINTER(ctypes.c_int64),
                                            ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32]
         L.fdb_nibblepack_unpack8.argtypes = [ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32,
@@ -161,6 +165,13 @@ class ChunkStore:
         _check(lib().fdb_series_append(
             self._h, sid, ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
             vals.ctypes.data_as(_c_double_p), len(ts)), "append")
+
+    def synth_generate(self, kind, n_series, n_samples, start_ts=100000,
+                       step_ms=15000, jitter_ms=250, lam=10.0, reset_p=0.001,
+                       n_groups=1, seed=42):
+        _check(lib().fdb_synth_generate(self._h, kind, n_series, n_samples,
+                                        start_ts, step_ms, jitter_ms, lam, reset_p,
+                                        n_groups, seed), "synth_generate")
 
     def cut_chunk(self, sid):
         _check(lib().fdb_series_cut_chunk(self._h, sid), "cut_chunk")

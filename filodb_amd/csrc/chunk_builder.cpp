@@ -503,6 +503,72 @@ extern "C" int32_t fdb_chunk_get(const fdb_store_t* s, int32_t sid, int32_t ci, 
   return FDB_OK;
 }
 
+// ---------------------------------------------------------------------------
+// synthetic workload generator (TestTimeseriesProducer shapes; BASELINE configs)
+// ---------------------------------------------------------------------------
+static inline uint64_t splitmix64(uint64_t& x) {
+  x += 0x9E3779B97F4A7C15ULL;
+  uint64_t z = x;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+  return z ^ (z >> 31);
+}
+static inline double u01(uint64_t& st) { return (double)(splitmix64(st) >> 11) * 0x1.0p-53; }
+static int poisson_knuth(uint64_t& st, double lam) {
+  double L = exp(-lam), p = 1.0;
+  int k = 0;
+  do { k++; p *= u01(st); } while (p > L);
+  return k - 1;
+}
+
+extern "C" int32_t fdb_synth_generate(fdb_store_t* s, int32_t kind, int32_t n_series,
+                                      int32_t n_samples, int64_t start_ts, int32_t step_ms,
+                                      int32_t jitter_ms, double lam, double reset_p,
+                                      int32_t n_groups, uint64_t seed) {
+  if (s->sealed) { fdb_set_error("store is sealed"); return FDB_ERR_BADARG; }
+  size_t base = s->series.size();
+  s->series.resize(base + (size_t)n_series);
+  for (int32_t i = 0; i < n_series; i++) {
+    Series& se = s->series[base + i];
+    se.group_id = n_groups > 0 ? (int32_t)((base + i) % n_groups) : 0;
+    se.col_kind = kind;
+  }
+  #pragma omp parallel for schedule(static)
+  for (int32_t i = 0; i < n_series; i++) {
+    Series& se = s->series[base + i];
+    uint64_t st = seed * 0x9E3779B97F4A7C15ULL + (uint64_t)(base + i) * 1000003ULL + 12345ULL;
+    se.buf_ts.reserve(s->max_rows);
+    se.buf_vals.reserve(s->max_rows);
+    int64_t last_ts = 0;
+    double counter = 0, walk = 0;
+    for (int32_t k = 0; k < n_samples; k++) {
+      int64_t jitter = jitter_ms > 0
+        ? (int64_t)(splitmix64(st) % (uint64_t)(2 * jitter_ms + 1)) - jitter_ms : 0;
+      int64_t ts = start_ts + (int64_t)k * step_ms + jitter;
+      if (ts < last_ts) ts = last_ts;
+      last_ts = ts;
+      double v;
+      if (kind == FDB_COL_COUNTER) {
+        if (u01(st) < reset_p) counter = 0;          // counter reset
+        counter += (double)poisson_knuth(st, lam);
+        v = counter;
+      } else {
+        walk += (u01(st) - 0.5) * 2.0;
+        v = walk + u01(st);                          // non-integral → raw f64 path
+      }
+      if (se.col_kind == FDB_COL_COUNTER) {
+        if (std::isnan(v) || v < se.buf_last) se.buf_drop = true;
+        if (!std::isnan(v)) se.buf_last = v;
+      }
+      se.buf_ts.push_back(ts);
+      se.buf_vals.push_back(v);
+      if ((int32_t)se.buf_ts.size() >= s->max_rows) cut_chunk(s, se);
+    }
+    cut_chunk(s, se);
+  }
+  return FDB_OK;
+}
+
 extern "C" int32_t fdb_store_view(const fdb_store_t* s, fdb_view_t* out) {
   if (!s->sealed) { fdb_set_error("store not sealed"); return FDB_ERR_BADARG; }
   out->blob = s->blob.data();

@@ -98,6 +98,18 @@ typedef struct {
 } fdb_view_t;
 int32_t fdb_store_view(const fdb_store_t* s, fdb_view_t* out);
 
+/* Synthetic workload generator (host, parallel): reproduces the data shapes of
+ * the reference's TestTimeseriesProducer gauge/counter series
+ * (gateway/src/main/scala/filodb/timeseries/TestTimeseriesProducer.scala:75-199)
+ * at BASELINE.json config scale. kind: FDB_COL_COUNTER → cumulative Poisson(lam)
+ * counts with reset probability reset_p per sample; FDB_COL_GAUGE → non-integral
+ * random walk. Timestamps jittered ±jitter_ms around the step grid. Series s
+ * gets group id s % n_groups. Deterministic in seed. */
+int32_t fdb_synth_generate(fdb_store_t* s, int32_t kind, int32_t n_series,
+                           int32_t n_samples, int64_t start_ts, int32_t step_ms,
+                           int32_t jitter_ms, double lam, double reset_p,
+                           int32_t n_groups, uint64_t seed);
+
 /* ---- standalone codec entry points (host; tests/golden-vector parity) ---- */
 /* NibblePack.pack8 / unpack8 (core/.../format/NibblePack.scala:108-183,395-447). */
 int32_t fdb_nibblepack_pack8(const int64_t in[8], uint8_t* out, int32_t outcap);

@@ -616,23 +616,28 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
       double* accc = gc + (size_t)grp * nw;
       for (int w = 0; w < nw; w++) {
         double x = row[w];
+        if (isnan(x)) continue;   /* NaN rows never contribute (all RowAggregators) */
+        accc[w] += 1;
         switch (q->agg_id) {
           case AGG_SUM:   /* SumRowAggregator.scala:23-29 */
-            if (!isnan(x)) { if (isnan(acc[w])) acc[w] = 0; acc[w] += x; }
+            if (isnan(acc[w])) acc[w] = 0;
+            acc[w] += x;
             break;
           case AGG_COUNT:
             /* CountRowAggregator.scala:36-42: sample maps to 1 (0 when NaN); the
              * accumulator leaves NaN only when every mapped value was 0. */
-            if (!isnan(x)) { if (isnan(acc[w])) acc[w] = 0; acc[w] += 1; }
+            if (isnan(acc[w])) acc[w] = 0;
+            acc[w] += 1;
             break;
           case AGG_MIN:
-            if (!isnan(x) && (isnan(acc[w]) || x < acc[w])) acc[w] = x;
+            if (isnan(acc[w]) || x < acc[w]) acc[w] = x;
             break;
           case AGG_MAX:
-            if (!isnan(x) && (isnan(acc[w]) || x > acc[w])) acc[w] = x;
+            if (isnan(acc[w]) || x > acc[w]) acc[w] = x;
             break;
           case AGG_AVG:   /* AvgRowAggregator: weighted mean ≡ sum/count */
-            if (!isnan(x)) { if (isnan(acc[w])) acc[w] = 0; acc[w] += x; accc[w] += 1; }
+            if (isnan(acc[w])) acc[w] = 0;
+            acc[w] += x;
             break;
         }
       }
@@ -641,36 +646,44 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
   }
 
   /* merge thread partials (same op; associative for these aggregators) */
+  double* cnt_total = counts ? counts : (double*)calloc(gridlen, sizeof(double));
   for (int t = 0; t < T; t++) {
     double* g = pg + (size_t)t * gridlen;
     double* gc = pc + (size_t)t * gridlen;
     for (size_t i = 0; i < gridlen; i++) {
       double x = g[i];
+      cnt_total[i] += gc[i];
+      if (isnan(x)) continue;
       switch (q->agg_id) {
-        case AGG_SUM:
-          if (!isnan(x)) { if (isnan(out[i])) out[i] = 0; out[i] += x; }
-          break;
-        case AGG_COUNT:
-          if (!isnan(x)) { if (isnan(out[i])) out[i] = 0; out[i] += x; }
+        case AGG_SUM: case AGG_COUNT: case AGG_AVG:
+          if (isnan(out[i])) out[i] = 0;
+          out[i] += x;
           break;
         case AGG_MIN:
-          if (!isnan(x) && (isnan(out[i]) || x < out[i])) out[i] = x;
+          if (isnan(out[i]) || x < out[i]) out[i] = x;
           break;
         case AGG_MAX:
-          if (!isnan(x) && (isnan(out[i]) || x > out[i])) out[i] = x;
-          break;
-        case AGG_AVG:
-          if (!isnan(x)) { if (isnan(out[i])) out[i] = 0; out[i] += x; counts[i] += gc[i]; }
+          if (isnan(out[i]) || x > out[i]) out[i] = x;
           break;
       }
     }
   }
-  if (q->agg_id == AGG_AVG) {
+  if (out_counts) {
+    /* PARTIAL mode (cross-shard merge inputs, ReduceAggregateExec contract):
+     * raw sums with 0 where empty + contribution counts; caller merges then
+     * presents (NaN where total count 0; /count for avg). */
+    if (q->agg_id == AGG_SUM || q->agg_id == AGG_COUNT || q->agg_id == AGG_AVG)
+      for (size_t i = 0; i < gridlen; i++)
+        if (isnan(out[i]) && cnt_total[i] == 0) out[i] = 0;
+    /* MIN/MAX partials keep NaN for empty cells; the merging caller maps them
+     * to ±inf before the collective. */
+  } else if (q->agg_id == AGG_AVG) {
     /* present: mean = sum/count (AvgRowAggregator.scala:38-46 algebraically) */
     for (size_t i = 0; i < gridlen; i++)
-      if (counts[i] > 0) out[i] = out[i] / counts[i];
+      if (cnt_total[i] > 0) out[i] = out[i] / cnt_total[i];
   }
   free(pg); free(pc);
+  if (!counts) free(cnt_total);
   if (owned_counts) free(owned_counts);
   return 0;
 }

@@ -1,0 +1,89 @@
+"""Multi-shard reduce correctness on CPU (gloo, world_size=2).
+
+Covers the distributed merge the 8-GPU path uses: each rank computes partial
+(sum, count) grids for its shard of series, all-reduces them (RCCL on GPU, gloo
+here), and presents. This is the reference's shard model — per-shard leaf plans
+reduced by ReduceAggregateExec (AggrOverRangeVectors.scala:18-102) — with the
+Akka merge replaced by one collective.
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from conftest import REPO, build_store, synth_gauge_series
+
+WORLD = 2
+
+
+def _make_shard(fdb, rank, n_groups):
+    rng = np.random.default_rng(1000 + rank)
+    series, groups = [], []
+    for s in range(20):
+        ts, vs = synth_gauge_series(rng, 50, step=10000, jitter=400, nan_p=0.1)
+        series.append([[(int(t), float(v)) for t, v in zip(ts, vs)]])
+        groups.append((rank * 20 + s) % n_groups)
+    return build_store(fdb, series, groups=groups), groups
+
+
+def _worker(rank, result_queue):
+    import sys
+    sys.path.insert(0, REPO)
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import filodb_amd as fdb
+    import pyclient as oracle
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29517"
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+
+    n_groups = 4
+    st, _ = _make_shard(fdb, rank, n_groups)
+    start = 100000 + 20 * 10000
+    q = fdb.make_query(start, 30000, start + 200000, 100000,
+                       fdb.FN_AVG_OVER_TIME, fdb.AGG_AVG, n_groups)
+    nw = q.num_windows
+    # rank-local partials (the oracle stands in for the GPU engine on CPU;
+    # the GPU parity suite covers engine==oracle)
+    sums, counts = oracle.query_exec(st.view(), q, st.num_series, nw, out_counts=True)
+    t_sum = torch.from_numpy(sums)
+    t_cnt = torch.from_numpy(counts)
+    dist.all_reduce(t_sum)
+    dist.all_reduce(t_cnt)
+    merged = torch.where(t_cnt > 0, t_sum / t_cnt, torch.full_like(t_sum, float("nan")))
+
+    if rank == 0:
+        result_queue.put(merged.numpy())
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_two_shard_avg_merge_equals_global(fdb, oracle):
+    ctx = mp.get_context("spawn")
+    queue = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, queue)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    merged = queue.get(timeout=120)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+
+    # global recomputation: both shards in one store
+    n_groups = 4
+    series, groups = [], []
+    for rank in range(WORLD):
+        rng = np.random.default_rng(1000 + rank)
+        for s in range(20):
+            ts, vs = synth_gauge_series(rng, 50, step=10000, jitter=400, nan_p=0.1)
+            series.append([[(int(t), float(v)) for t, v in zip(ts, vs)]])
+            groups.append((rank * 20 + s) % n_groups)
+    st = build_store(fdb, series, groups=groups)
+    start = 100000 + 20 * 10000
+    q = fdb.make_query(start, 30000, start + 200000, 100000,
+                       fdb.FN_AVG_OVER_TIME, fdb.AGG_AVG, n_groups)
+    expected = oracle.query_exec(st.view(), q, st.num_series, q.num_windows)
+    np.testing.assert_allclose(merged, expected, rtol=1e-9, equal_nan=True)

@@ -1,0 +1,142 @@
+"""Oracle gauge window functions vs an independent Python/numpy recomputation.
+
+Mirrors the reference's own strategy (AggrOverTimeFunctionsSpec.scala:289-400):
+the chunked result must equal a naive recomputation over the same samples —
+exactly for counts/min/max, within 1e-9 relative for FP sums (the spec uses
++-1e-7 abs; our bar is tighter per north_star).
+"""
+import numpy as np
+import pytest
+
+from conftest import build_store, synth_gauge_series
+
+REL = 1e-9
+
+
+def naive_window(ts, vs, chunk_bounds, w_start, w_end, func):
+    """Independent restatement of the chunked per-window semantics over decoded
+    samples. chunk_bounds: list of (lo, hi) row index ranges per chunk."""
+    sum_, count, sqsum, icount = np.nan, np.nan, np.nan, 0
+    mn = mx = np.nan
+    changes, prev = np.nan, np.nan
+    for lo, hi in chunk_bounds:
+        cts, cvs = ts[lo:hi], vs[lo:hi]
+        m = (cts >= w_start) & (cts <= w_end)
+        if not m.any():
+            continue
+        x = cvs[m]
+        nn = x[~np.isnan(x)]
+        if func == "sum" or func == "avg":
+            cs = np.nan if len(nn) == 0 else nn.sum()
+            if not np.isnan(cs) and np.isnan(sum_):
+                sum_ = 0.0
+            sum_ = sum_ + cs
+            icount += len(nn)
+        elif func == "count":
+            if np.isnan(count):
+                count = 0.0
+            count += len(nn)
+        elif func in ("min", "max"):
+            if len(nn):
+                mn = np.nanmin([mn, nn.min()]) if not np.isnan(mn) else nn.min()
+                mx = np.nanmax([mx, nn.max()]) if not np.isnan(mx) else nn.max()
+        elif func in ("stddev", "stdvar"):
+            cs = np.nan if len(nn) == 0 else nn.sum()
+            csq = np.nan if len(nn) == 0 else (nn * nn).sum()
+            if not np.isnan(cs) and np.isnan(sum_):
+                sum_ = 0.0
+            sum_ = sum_ + cs
+            if not np.isnan(csq) and np.isnan(sqsum):
+                sqsum = 0.0
+            sqsum = sqsum + csq
+            icount += len(nn)
+        elif func == "changes":
+            if np.isnan(changes):
+                changes = 0.0
+            for v in x:
+                if not np.isnan(v) and prev != v and not np.isnan(prev):
+                    changes += 1
+                prev = v
+            else:
+                pass
+            prev = x[-1]
+    if func == "sum":
+        return sum_
+    if func == "count":
+        return count
+    if func == "avg":
+        return sum_ / icount if icount > 0 else (sum_ if np.isnan(sum_) else 0.0)
+    if func == "min":
+        return mn
+    if func == "max":
+        return mx
+    if func in ("stddev", "stdvar"):
+        if icount > 0:
+            avg = sum_ / icount
+            r = sqsum / icount - avg * avg
+            return np.sqrt(r) if func == "stddev" else r
+        return sum_ if np.isnan(sum_) else 0.0
+    if func == "changes":
+        return changes
+    raise ValueError(func)
+
+
+FUNC_IDS = {"sum": 3, "count": 4, "avg": 5, "min": 6, "max": 7,
+            "stddev": 8, "stdvar": 9, "changes": 10}
+
+
+@pytest.mark.parametrize("func", list(FUNC_IDS))
+@pytest.mark.parametrize("nan_p,nchunks", [(0.0, 1), (0.0, 3), (0.1, 1), (0.1, 3)])
+def test_gauge_chunked_vs_naive(fdb, oracle, func, nan_p, nchunks):
+    rng = np.random.default_rng(hash((func, nan_p, nchunks)) % 2**31)
+    n = 120
+    ts, vs = synth_gauge_series(rng, n, step=10000, jitter=400, nan_p=nan_p)
+    per = n // nchunks
+    chunks, bounds = [], []
+    for c in range(nchunks):
+        lo = c * per
+        hi = n if c == nchunks - 1 else (c + 1) * per
+        chunks.append([(int(ts[i]), float(vs[i])) for i in range(lo, hi)])
+        bounds.append((lo, hi))
+    st = build_store(fdb, [chunks])
+    # windowSize=20 samples ≈ 200s, step 3 samples ≈ 30s (spec's sliding recipe)
+    start = int(ts[25])
+    end = int(ts[-1])
+    step, window = 30000, 200000
+    q = fdb.make_query(start, step, end, window, FUNC_IDS[func])
+    out = oracle.eval_series(st.view(), 0, q, q.num_windows)
+    for w in range(q.num_windows):
+        w_end = start + w * step
+        expected = naive_window(ts, vs, bounds, w_end - window, w_end, func)
+        got = out[w]
+        if np.isnan(expected):
+            assert np.isnan(got), (func, w)
+        else:
+            assert got == pytest.approx(expected, rel=REL, abs=1e-12), (func, w)
+
+
+def test_sum_matches_simple_sliding(fdb, oracle):
+    # AggrOverTimeFunctionsSpec:289-310 — chunked sum == data.sliding(...).sum
+    rng = np.random.default_rng(11)
+    n = 100
+    ts = (100000 + np.arange(n) * 10000).astype(np.int64)
+    vs = rng.random(n) * 100
+    st = build_store(fdb, [[[(int(t), float(v)) for t, v in zip(ts, vs)]]], max_rows=50)
+    # window = exactly 20 samples: [wEnd-190000, wEnd]
+    step, window = 10000 * 5, 190000
+    start, end = int(ts[19]), int(ts[-1])
+    q = fdb.make_query(start, step, end, window, 3)
+    out = oracle.eval_series(st.view(), 0, q, q.num_windows)
+    for w in range(q.num_windows):
+        i_end = 19 + 5 * w
+        expected = vs[i_end - 19:i_end + 1].sum()
+        assert out[w] == pytest.approx(expected, rel=REL)
+
+
+def test_empty_window_emits_nan(fdb, oracle):
+    ts = (100000 + np.arange(10) * 1000).astype(np.int64)
+    vs = np.arange(10, dtype=np.float64) + 0.5
+    st = build_store(fdb, [[[(int(t), float(v)) for t, v in zip(ts, vs)]]])
+    q = fdb.make_query(200000, 1000, 205000, 500, 3)  # windows beyond the data
+    out = oracle.eval_series(st.view(), 0, q, q.num_windows)
+    assert np.isnan(out).all()
