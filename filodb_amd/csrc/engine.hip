@@ -213,7 +213,8 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                  int num_windows,
                  int agg_id,
                  double* __restrict__ out,        // [S×W] when AGG_NONE else [G×W] sums
-                 double* __restrict__ out_cnt)    // [G×W] contribution counts (agg) or null
+                 double* __restrict__ out_cnt,    // [G×W] contribution counts (agg) or null
+                 int phase_mask)                  // debug ablation: 1=decode 2=windows
 {
   constexpr bool RATE_FAMILY = (FUNC <= FN_DELTA);
   __shared__ Ws<RATE_FAMILY> ws_all[WAVES_PER_BLOCK];
@@ -233,6 +234,23 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   int row0 = 0;
   for (int c = 0; c < nchunks; c++) {
     DVec tv, vv;
+    if (!(phase_mask & 1)) {   // window-only ablation: synthesize rows in LDS
+      int n = dir.num_rows[first + c];
+      if (row0 + n > FDB_MAX_ROWS_PER_SERIES) n = 0;
+      for (int i = lane; i < n; i += 64) {
+        ws.ts[row0 + i] = dir.start_time[first + c] + (int64_t)i * 15000;
+        ws.val[row0 + i] = (double)(i * 10);
+      }
+      if (lane == 0) {
+        ChunkMeta& m = ws.cm[c];
+        m.row0 = row0; m.nrows = n;
+        m.start_time = dir.start_time[first + c];
+        m.end_time = dir.end_time[first + c];
+        m.dropped = 0; m.chunk_correction = 0; m.v0_nan = 0; m.last_for_update = 0;
+      }
+      row0 += n;
+      continue;
+    }
     d_vec_open(blob + dir.ts_off[first + c], &tv);
     d_vec_open(blob + dir.val_off[first + c], &vv);
     int n = dir.num_rows[first + c];
@@ -305,6 +323,11 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   }
 
   // ---- window phase: lanes split the windows -------------------------------
+  if (!(phase_mask & 2)) {            // decode-only ablation: publish a checksum
+    if (lane == 0 && ws.total_rows > 0)
+      out[(size_t)sid * num_windows] = ws.val[0] + (double)ws.ts[0];
+    return;
+  }
   const int grp = group_ids[sid];
   for (int w = lane; w < num_windows; w += 64) {
     const int64_t wEnd = qstart + (int64_t)w * qstep;
@@ -613,7 +636,8 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
   #define CASE(F) case F: \
     hipLaunchKernelGGL(scan_kernel<F>, dim3(grid), dim3(BLOCK_THREADS), 0, e->stream, \
       d->blob, dir, d->series_first, d->series_nchunks, d->group_ids, d->num_series, \
-      q->start, q->step, q->end, q->window, nw, q->agg_id, dev_out, dev_cnt); break
+      q->start, q->step, q->end, q->window, nw, q->agg_id, dev_out, dev_cnt, \
+      q->_pad == 0 ? 3 : q->_pad); break
   switch (q->func_id) {
     CASE(FN_RATE); CASE(FN_INCREASE); CASE(FN_DELTA); CASE(FN_SUM); CASE(FN_COUNT);
     CASE(FN_AVG); CASE(FN_MIN); CASE(FN_MAX); CASE(FN_STDDEV); CASE(FN_STDVAR);

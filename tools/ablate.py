@@ -1,0 +1,40 @@
+#!/usr/bin/env python3
+"""Phase ablation of the scan kernel on the 1M-series rate workload.
+Phases: 3=full, 1=decode-only, 2=window-only (synthesized LDS rows)."""
+import os
+import sys
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+import filodb_amd as fdb  # noqa: E402
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+T0 = 100000
+
+
+def main():
+    workload = sys.argv[1] if len(sys.argv) > 1 else "rate"
+    n_series = int(sys.argv[2]) if len(sys.argv) > 2 else 1_000_000
+    st = fdb.ChunkStore()
+    kind = fdb.COL_COUNTER if workload == "rate" else fdb.COL_GAUGE
+    st.synth_generate(kind, n_series, 240, start_ts=T0, step_ms=15000,
+                      jitter_ms=250, lam=10.0, reset_p=0.001, n_groups=1000, seed=42)
+    st.seal()
+    eng = fdb.Engine(0)
+    ds = eng.upload(st)
+    func = fdb.FN_RATE if workload == "rate" else fdb.FN_AVG_OVER_TIME
+    window = 300_000 if workload == "rate" else 600_000
+    import torch
+    for phase, name in [(3, "full"), (1, "decode-only"), (2, "window-only")]:
+        q = fdb.make_query(T0, 15000, T0 + 240 * 15000, window, func)
+        q._pad = phase
+        out = torch.empty(n_series * q.num_windows, dtype=torch.float64, device="cuda")
+        ms = eng.bench(ds, q, out, on_device=True, warmup=2, iters=10)
+        print(f"{workload} {name:12s} kernel_ms={ms:8.3f}", flush=True)
+
+
+if __name__ == "__main__":
+    main()
