@@ -127,17 +127,27 @@ __device__ __forceinline__ double d_dv_at(const DVec* v, int i) {
 struct ChunkMeta {
   int32_t row0, nrows;
   int64_t start_time, end_time;    // directory times (pre-encoding, like ChunkSetInfo)
+  int64_t ts0;                     // decoded first timestamp (search guess base)
+  float   inv_slope;               // (nrows-1)/(ts_last-ts0); 0 when degenerate
+  int32_t dropped;
   double last_for_update;          // updateCorrection lastValue (DoubleVector.scala:375-391,190-195)
   double chunk_correction;         // CorrectingDoubleVectorReader._correction total
-  int32_t dropped;
-  int32_t v0_nan;                  // isNaN(apply(0)) for the single-row-NaN rule
+  int16_t dstart, dcount;          // slice of the wave drop table
+  int16_t dense_corr;              // >MAX_DROPS drops: recompute corrections serially
+  int16_t v0_nan;                  // isNaN(apply(0)) for the single-row-NaN rule
 };
+
+#define FDB_MAX_DROPS 32            // wave-wide counter-reset table capacity
 
 template <bool RATE_FAMILY>
 struct Ws {                         // per-wave LDS workspace
   int64_t ts[FDB_MAX_ROWS_PER_SERIES];
   double  val[FDB_MAX_ROWS_PER_SERIES];                  // raw values
-  double  aux[RATE_FAMILY ? FDB_MAX_ROWS_PER_SERIES : 1]; // corrected values (dropped chunks)
+  // counter resets are rare: store (position, cumulative in-chunk correction)
+  // pairs instead of a full corrected[] copy (CorrectingDoubleVectorReader
+  // :325-342 materializes corrected[]; the correction is a step function)
+  int16_t dpos[RATE_FAMILY ? FDB_MAX_DROPS : 1];
+  double  dcum[RATE_FAMILY ? FDB_MAX_DROPS : 1];
   ChunkMeta cm[FDB_MAX_CHUNKS_PER_SERIES];
   int32_t nchunks;
   int32_t total_rows;
@@ -152,17 +162,26 @@ __device__ __forceinline__ double wave_incl_scan(double x, int lane) {
   return x;
 }
 
-// first index in [0,n) with seg[i] >= item, over sorted LDS segment; n when none.
-__device__ __forceinline__ int lds_search_ge(const int64_t* seg, int n, int64_t item) {
-  int lo = 0, hi = n;
-  while (lo < hi) { int mid = (lo + hi) >> 1; if (seg[mid] < item) lo = mid + 1; else hi = mid; }
-  return lo;
+// first index in [0,n) with seg[i] >= item; n when none. Timestamps sit near a
+// slope line (the DDV premise), so an interpolation guess + short walk replaces
+// the O(log n) dependent-LDS binary-search chain (usually 1-2 LDS reads) —
+// same trick as DeltaDeltaDataReader.binarySearch (DeltaDeltaVector.scala:159-188).
+__device__ __forceinline__ int lds_search_ge(const int64_t* seg, int n, int64_t item,
+                                             int64_t ts0, float inv_slope) {
+  if (n <= 0) return 0;
+  int g = (int)((float)(item - ts0) * inv_slope);
+  if (g < 0) g = 0;
+  if (g > n - 1) g = n - 1;
+  while (g > 0 && seg[g - 1] >= item) g--;
+  if (seg[g] >= item && (g == 0 || seg[g - 1] < item)) return g;
+  while (g < n && seg[g] < item) g++;
+  return g;
 }
-// last index with seg[i] <= item; -1 when none (ceilingIndex semantics)
-__device__ __forceinline__ int lds_search_le(const int64_t* seg, int n, int64_t item) {
-  int lo = 0, hi = n;
-  while (lo < hi) { int mid = (lo + hi) >> 1; if (seg[mid] <= item) lo = mid + 1; else hi = mid; }
-  return lo - 1;
+// last index with seg[i] <= item; -1 when none (ceilingIndex semantics).
+// Integer timestamps: last <= item == (first >= item+1) - 1.
+__device__ __forceinline__ int lds_search_le(const int64_t* seg, int n, int64_t item,
+                                             int64_t ts0, float inv_slope) {
+  return lds_search_ge(seg, n, item + 1, ts0, inv_slope) - 1;
 }
 
 // extrapolatedRate (RateFunctions.scala:72-111) — same arithmetic as the oracle
@@ -184,6 +203,34 @@ __device__ double d_extrapolated_rate(int64_t windowStart, int64_t windowEnd, in
   ext += (durationToEnd < thresh) ? durationToEnd : avgDur / 2;
   double scaledDelta = delta * (ext / sampledInterval);
   return isRate ? (scaledDelta / (double)(windowEnd - windowStart) * 1000.0) : scaledDelta;
+}
+
+// in-chunk correction lookup from the sparse drop table (the step function
+// CorrectingDoubleVectorReader :325-342 materializes as corrected[])
+template <bool RF>
+__device__ __forceinline__ double d_corr_at(const Ws<RF>& ws, const ChunkMeta& m, int i) {
+  if (m.dense_corr) {
+    // serial recompute — only when one chunk held >FDB_MAX_DROPS resets
+    double corr = 0, last = -1.7976931348623157e308;
+    for (int j = 0; j <= i; j++) {
+      double x = ws.val[m.row0 + j];
+      if (isnan(x)) x = 0;
+      if (x < last) corr += last;
+      last = x;
+    }
+    return corr;
+  }
+  double c = 0;
+  for (int j = m.dstart; j < m.dstart + m.dcount; j++) {
+    if (ws.dpos[j] <= i) c = ws.dcum[j]; else break;
+  }
+  return c;
+}
+template <bool RF>
+__device__ __forceinline__ double d_corrected(const Ws<RF>& ws, const ChunkMeta& m, int i) {
+  double x = ws.val[m.row0 + i];
+  if (m.dropped) { if (isnan(x)) x = 0; x += d_corr_at(ws, m, i); }
+  return x;
 }
 
 // NaN-aware f64 atomic min/max via CAS (group aggregation; RowAggregator semantics)
@@ -247,6 +294,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         m.start_time = dir.start_time[first + c];
         m.end_time = dir.end_time[first + c];
         m.dropped = 0; m.chunk_correction = 0; m.v0_nan = 0; m.last_for_update = 0;
+        m.dcount = 0; m.dense_corr = 0;
       }
       row0 += n;
       continue;
@@ -277,19 +325,28 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   __builtin_amdgcn_s_waitcnt(0);   // LDS writes visible within the wave
   __builtin_amdgcn_wave_barrier();
 
-  // ---- rate family: per-chunk correction scan (CorrectingDoubleVectorReader
-  //      :325-342) + updateCorrection scalars (:375-391,190-195) -------------
-  if constexpr (RATE_FAMILY) {
-    for (int c = 0; c < nchunks; c++) {
-      ChunkMeta& m = ws.cm[c];
-      const int r0 = m.row0, n = m.nrows;
-      if (n == 0) continue;
+  // ---- meta phase: search-guess slopes for every func; rate family adds the
+  // counter-correction scan (CorrectingDoubleVectorReader :325-342) recorded
+  // as a sparse drop table + updateCorrection scalars (:375-391,190-195) -----
+  int table_used = 0;
+  for (int c = 0; c < nchunks; c++) {
+    ChunkMeta& m = ws.cm[c];
+    const int r0 = m.row0, n = m.nrows;
+    if (n == 0) continue;
+    if (lane == 0) {
+      m.ts0 = ws.ts[r0];
+      int64_t tl = ws.ts[r0 + n - 1];
+      m.inv_slope = (tl > m.ts0) ? (float)(n - 1) / (float)(tl - m.ts0) : 0.0f;
+    }
+    if constexpr (RATE_FAMILY) {
       if (lane == 0) m.v0_nan = isnan(ws.val[r0]);
       if (m.dropped) {
         double carry_corr = 0;
         double carry_x = -1.7976931348623157e308;   // 'last' starts Double.MinValue
         int last_idx = -1;                          // last non-NaN index (for update)
         double last_val = 0;
+        int dstart = table_used, dcnt = 0;
+        bool overflow = false;
         for (int base = 0; base < n; base += 64) {
           int i = base + lane;
           double raw = (i < n) ? ws.val[r0 + i] : 0;
@@ -299,7 +356,19 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
           if (lane == 0) px = carry_x;
           double ci = (i < n && x < px) ? px : 0;
           double scan = wave_incl_scan(ci, lane);
-          if (i < n) ws.aux[r0 + i] = x + carry_corr + scan;
+          // record (pos, cumulative correction) for each drop in this iteration
+          uint64_t mask = __ballot(ci != 0);
+          int here = __popcll(mask);
+          if (here) {
+            if (dstart + dcnt + here > FDB_MAX_DROPS) {
+              overflow = true;
+            } else if (ci != 0) {
+              int slot = dstart + dcnt + __popcll(mask & ((1ULL << lane) - 1));
+              ws.dpos[slot] = (int16_t)i;
+              ws.dcum[slot] = carry_corr + scan;
+            }
+            dcnt += here;
+          }
           carry_corr += __shfl(scan, 63);
           carry_x = __shfl(x, 63);
           if (i < n && !isnan(raw)) { last_idx = i; last_val = raw; }
@@ -310,17 +379,22 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
           double ov = __shfl_down(last_val, off);
           if (oi > last_idx) { last_idx = oi; last_val = ov; }
         }
+        if (!overflow) table_used = dstart + dcnt;
         if (lane == 0) {
           m.chunk_correction = carry_corr;
           m.last_for_update = (last_idx >= 0) ? last_val : 0;
+          m.dstart = (int16_t)dstart;
+          m.dcount = overflow ? 0 : (int16_t)dcnt;
+          m.dense_corr = overflow ? 1 : 0;
         }
       } else if (lane == 0) {
         m.last_for_update = ws.val[r0 + n - 1];     // default updateCorrection
+        m.dcount = 0; m.dense_corr = 0;
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);
-    __builtin_amdgcn_wave_barrier();
   }
+  __builtin_amdgcn_s_waitcnt(0);
+  __builtin_amdgcn_wave_barrier();
 
   // ---- window phase: lanes split the windows -------------------------------
   if (!(phase_mask & 2)) {            // decode-only ablation: publish a checksum
@@ -345,8 +419,8 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         const ChunkMeta& m = ws.cm[c];
         if (m.end_time < wStart) continue;          // WindowedChunkIterator drop rule
         const int64_t* seg = ws.ts + m.row0;
-        int startRow = lds_search_ge(seg, m.nrows, wStart);
-        int endRow = lds_search_le(seg, m.nrows, wEnd);
+        int startRow = lds_search_ge(seg, m.nrows, wStart, m.ts0, m.inv_slope);
+        int endRow = lds_search_le(seg, m.nrows, wEnd, m.ts0, m.inv_slope);
         if (isCounter && meta_has) {                // detectDropAndCorrection
           double firstv = ws.val[m.row0];
           if (isnan(firstv) || firstv < meta_last) meta_corr += meta_last;
@@ -359,14 +433,12 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
               numSamples += endRow - startRow + 1;
               if (st < lowestTime) {
                 lowestTime = st;
-                double v = m.dropped ? ws.aux[m.row0 + startRow] : ws.val[m.row0 + startRow];
-                lowestValue = isCounter ? v + meta_corr
+                lowestValue = isCounter ? d_corrected(ws, m, startRow) + meta_corr
                                         : ws.val[m.row0 + startRow];
               }
               if (en > highestTime) {
                 highestTime = en;
-                double v = m.dropped ? ws.aux[m.row0 + endRow] : ws.val[m.row0 + endRow];
-                highestValue = isCounter ? v + meta_corr
+                highestValue = isCounter ? d_corrected(ws, m, endRow) + meta_corr
                                          : ws.val[m.row0 + endRow];
               }
             }
@@ -392,8 +464,8 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         if (m.end_time < wStart) continue;
         const int64_t* seg = ws.ts + m.row0;
         const double* vals = ws.val + m.row0;
-        int startRow = lds_search_ge(seg, m.nrows, wStart);
-        int endRow = lds_search_le(seg, m.nrows, wEnd);
+        int startRow = lds_search_ge(seg, m.nrows, wStart, m.ts0, m.inv_slope);
+        int endRow = lds_search_le(seg, m.nrows, wEnd, m.ts0, m.inv_slope);
         if (startRow <= endRow && endRow < m.nrows) {
           if (FUNC == FN_SUM || FUNC == FN_AVG) {
             double cs = NAN;                       // NaN-skipping chunk sum (:244-253)

@@ -5,6 +5,7 @@ import os
 import sys
 
 import numpy as np
+import torch  # must precede HIP init by the engine
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
@@ -27,7 +28,6 @@ def main():
     ds = eng.upload(st)
     func = fdb.FN_RATE if workload == "rate" else fdb.FN_AVG_OVER_TIME
     window = 300_000 if workload == "rate" else 600_000
-    import torch
     for phase, name in [(3, "full"), (1, "decode-only"), (2, "window-only")]:
         q = fdb.make_query(T0, 15000, T0 + 240 * 15000, window, func)
         q._pad = phase
