@@ -139,15 +139,35 @@ struct ChunkMeta {
 
 #define FDB_MAX_DROPS 32            // wave-wide counter-reset table capacity
 
-template <bool RATE_FAMILY>
+// which auxiliary LDS arrays a function family needs
+#define K_RATE    0   // ts, val, drop table
+#define K_PFX     1   // ts, prefix-sum (in place of val) + count prefix
+#define K_PFX_SQ  2   // + squared prefix (stddev/stdvar)
+#define K_MINMAX  3   // ts, val, 8-element group min/max
+#define K_CHANGES 4   // ts, val, change-indicator prefix
+
+template <int FUNC> struct KKind { static constexpr int v =
+    (FUNC <= FN_DELTA) ? K_RATE :
+    (FUNC == FN_SUM || FUNC == FN_AVG || FUNC == FN_COUNT) ? K_PFX :
+    (FUNC == FN_STDDEV || FUNC == FN_STDVAR) ? K_PFX_SQ :
+    (FUNC == FN_MIN || FUNC == FN_MAX) ? K_MINMAX : K_CHANGES; };
+
+#define FDB_NGROUPS ((FDB_MAX_ROWS_PER_SERIES + 7) / 8)
+
+template <int KIND>
 struct Ws {                         // per-wave LDS workspace
   int64_t ts[FDB_MAX_ROWS_PER_SERIES];
-  double  val[FDB_MAX_ROWS_PER_SERIES];                  // raw values
-  // counter resets are rare: store (position, cumulative in-chunk correction)
-  // pairs instead of a full corrected[] copy (CorrectingDoubleVectorReader
-  // :325-342 materializes corrected[]; the correction is a step function)
-  int16_t dpos[RATE_FAMILY ? FDB_MAX_DROPS : 1];
-  double  dcum[RATE_FAMILY ? FDB_MAX_DROPS : 1];
+  double  val[FDB_MAX_ROWS_PER_SERIES];   // raw values, or inclusive NaN-zeroed
+                                          // prefix sums for the PFX kinds
+  // counter resets are rare: (position, cumulative in-chunk correction) pairs
+  // instead of a full corrected[] copy (CorrectingDoubleVectorReader :325-342)
+  int16_t dpos[KIND == K_RATE ? FDB_MAX_DROPS : 1];
+  double  dcum[KIND == K_RATE ? FDB_MAX_DROPS : 1];
+  // non-NaN count prefix (PFX kinds) / change-indicator prefix (CHANGES)
+  uint16_t cnt[(KIND == K_PFX || KIND == K_PFX_SQ || KIND == K_CHANGES)
+               ? FDB_MAX_ROWS_PER_SERIES : 1];
+  double  sq[KIND == K_PFX_SQ ? FDB_MAX_ROWS_PER_SERIES : 1];   // squared prefix
+  double  grp[KIND == K_MINMAX ? (FDB_NGROUPS + FDB_MAX_CHUNKS_PER_SERIES) : 1]; // 8-elem group min/max (chunk-relative, padded per chunk)
   ChunkMeta cm[FDB_MAX_CHUNKS_PER_SERIES];
   int32_t nchunks;
   int32_t total_rows;
@@ -161,11 +181,96 @@ __device__ __forceinline__ double wave_incl_scan(double x, int lane) {
   }
   return x;
 }
+__device__ __forceinline__ int wave_incl_scan_i(int x, int lane) {
+  for (int off = 1; off < 64; off <<= 1) {
+    int t = __shfl_up(x, off);
+    if (lane >= off) x += t;
+  }
+  return x;
+}
+
+// ---------------------------------------------------------------------------
+// vectorized decode: whole chunk into LDS, wide loads, high ILP.
+// (the frozen formats of DESIGN.md §2; element semantics identical to
+//  d_lv_at/d_dv_at, which remain the reference implementations / fallback)
+// ---------------------------------------------------------------------------
+template <bool AS_DOUBLE>
+__device__ void d_decode_chunk(const DVec& v, int n, int64_t* tout, double* dout,
+                               int lane) {
+  if (v.wf == FDB_WF_DDV_CONST) {
+    for (int i = lane; i < n; i += 64) {
+      int64_t x = v.init + (int64_t)v.slope * i;
+      if (AS_DOUBLE) dout[i] = (double)x; else tout[i] = x;
+    }
+    return;
+  }
+  if (v.wf == FDB_WF_PRIM64) {
+    // raw 64-bit payload, 8-byte aligned (+8 from a 64B-aligned base)
+    for (int i = lane; i < n; i += 64) {
+      if (AS_DOUBLE) dout[i] = d_f64(v.idata + 8 * (size_t)i);
+      else           tout[i] = d_i64(v.idata + 8 * (size_t)i);
+    }
+    return;
+  }
+  // packed DDV inner data starts at +28 (4-byte aligned only)
+  if (v.nbits == 16) {
+    for (int i0 = 4 * lane; i0 < n; i0 += 256) {
+      uint32_t lo = d_u32(v.idata + 2 * (size_t)i0);
+      uint32_t hi = d_u32(v.idata + 2 * (size_t)i0 + 4);
+      uint64_t w = ((uint64_t)hi << 32) | lo;
+      #pragma unroll
+      for (int k = 0; k < 4; k++) {
+        if (i0 + k < n) {
+          int32_t d = (int32_t)(int16_t)(uint16_t)(w >> (16 * k));
+          if (!v.sign) d &= 0xffff;
+          int64_t x = v.init + (int64_t)v.slope * (i0 + k) + d;
+          if (AS_DOUBLE) dout[i0 + k] = (double)x; else tout[i0 + k] = x;
+        }
+      }
+    }
+    return;
+  }
+  if (v.nbits == 8) {
+    for (int i0 = 8 * lane; i0 < n; i0 += 512) {
+      uint32_t lo = d_u32(v.idata + (size_t)i0);
+      uint32_t hi = d_u32(v.idata + (size_t)i0 + 4);
+      uint64_t w = ((uint64_t)hi << 32) | lo;
+      #pragma unroll
+      for (int k = 0; k < 8; k++) {
+        if (i0 + k < n) {
+          int32_t d = (int32_t)(int8_t)(uint8_t)(w >> (8 * k));
+          if (!v.sign) d &= 0xff;
+          int64_t x = v.init + (int64_t)v.slope * (i0 + k) + d;
+          if (AS_DOUBLE) dout[i0 + k] = (double)x; else tout[i0 + k] = x;
+        }
+      }
+    }
+    return;
+  }
+  if (v.nbits == 32) {
+    for (int i0 = 2 * lane; i0 < n; i0 += 128) {
+      #pragma unroll
+      for (int k = 0; k < 2; k++) {
+        if (i0 + k < n) {
+          int64_t x = v.init + (int64_t)v.slope * (i0 + k)
+                    + d_i32(v.idata + 4 * (size_t)(i0 + k));
+          if (AS_DOUBLE) dout[i0 + k] = (double)x; else tout[i0 + k] = x;
+        }
+      }
+    }
+    return;
+  }
+  // nbits 2/4 fallback (rare)
+  for (int i = lane; i < n; i += 64) {
+    int64_t x = v.init + (int64_t)v.slope * i + d_inner_at(&v, i);
+    if (AS_DOUBLE) dout[i] = (double)x; else tout[i] = x;
+  }
+}
 
 // first index in [0,n) with seg[i] >= item; n when none. Timestamps sit near a
 // slope line (the DDV premise), so an interpolation guess + short walk replaces
-// the O(log n) dependent-LDS binary-search chain (usually 1-2 LDS reads) —
-// same trick as DeltaDeltaDataReader.binarySearch (DeltaDeltaVector.scala:159-188).
+// a dependent binary-search chain (usually 1-2 LDS reads) — the same idea as
+// DeltaDeltaDataReader.binarySearch (DeltaDeltaVector.scala:159-188).
 __device__ __forceinline__ int lds_search_ge(const int64_t* seg, int n, int64_t item,
                                              int64_t ts0, float inv_slope) {
   if (n <= 0) return 0;
@@ -173,7 +278,6 @@ __device__ __forceinline__ int lds_search_ge(const int64_t* seg, int n, int64_t 
   if (g < 0) g = 0;
   if (g > n - 1) g = n - 1;
   while (g > 0 && seg[g - 1] >= item) g--;
-  if (seg[g] >= item && (g == 0 || seg[g - 1] < item)) return g;
   while (g < n && seg[g] < item) g++;
   return g;
 }
@@ -207,8 +311,8 @@ __device__ double d_extrapolated_rate(int64_t windowStart, int64_t windowEnd, in
 
 // in-chunk correction lookup from the sparse drop table (the step function
 // CorrectingDoubleVectorReader :325-342 materializes as corrected[])
-template <bool RF>
-__device__ __forceinline__ double d_corr_at(const Ws<RF>& ws, const ChunkMeta& m, int i) {
+template <int KIND>
+__device__ __forceinline__ double d_corr_at(const Ws<KIND>& ws, const ChunkMeta& m, int i) {
   if (m.dense_corr) {
     // serial recompute — only when one chunk held >FDB_MAX_DROPS resets
     double corr = 0, last = -1.7976931348623157e308;
@@ -226,8 +330,8 @@ __device__ __forceinline__ double d_corr_at(const Ws<RF>& ws, const ChunkMeta& m
   }
   return c;
 }
-template <bool RF>
-__device__ __forceinline__ double d_corrected(const Ws<RF>& ws, const ChunkMeta& m, int i) {
+template <int KIND>
+__device__ __forceinline__ double d_corrected(const Ws<KIND>& ws, const ChunkMeta& m, int i) {
   double x = ws.val[m.row0 + i];
   if (m.dropped) { if (isnan(x)) x = 0; x += d_corr_at(ws, m, i); }
   return x;
@@ -247,7 +351,9 @@ __device__ void atomic_min_max_f64(double* addr, double val, bool is_min) {
 }
 
 // ---------------------------------------------------------------------------
-// the scan kernel
+// the scan kernel: one wavefront per series
+//   decode → per-chunk meta (search slopes, correction table, prefixes) →
+//   per-window evaluation with O(1)-ish row location and O(1) prefix lookups
 // ---------------------------------------------------------------------------
 template <int FUNC>
 __global__ __launch_bounds__(BLOCK_THREADS)
@@ -263,14 +369,15 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                  double* __restrict__ out_cnt,    // [G×W] contribution counts (agg) or null
                  int phase_mask)                  // debug ablation: 1=decode 2=windows
 {
-  constexpr bool RATE_FAMILY = (FUNC <= FN_DELTA);
-  __shared__ Ws<RATE_FAMILY> ws_all[WAVES_PER_BLOCK];
+  constexpr int KIND = KKind<FUNC>::v;
+  constexpr bool RATE_FAMILY = (KIND == K_RATE);
+  __shared__ Ws<KIND> ws_all[WAVES_PER_BLOCK];
 
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int sid = blockIdx.x * WAVES_PER_BLOCK + wave;
   if (sid >= num_series) return;
-  Ws<RATE_FAMILY>& ws = ws_all[wave];
+  Ws<KIND>& ws = ws_all[wave];
 
   // ---- decode phase: all chunks of this series into LDS --------------------
   const int first = series_first[sid];
@@ -280,54 +387,45 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 
   int row0 = 0;
   for (int c = 0; c < nchunks; c++) {
-    DVec tv, vv;
-    if (!(phase_mask & 1)) {   // window-only ablation: synthesize rows in LDS
-      int n = dir.num_rows[first + c];
-      if (row0 + n > FDB_MAX_ROWS_PER_SERIES) n = 0;
+    int n = dir.num_rows[first + c];
+    if (row0 + n > FDB_MAX_ROWS_PER_SERIES) n = 0;   // guarded at upload
+    DVec vv;
+    if (phase_mask & 1) {
+      DVec tv;
+      d_vec_open(blob + dir.ts_off[first + c], &tv);
+      d_vec_open(blob + dir.val_off[first + c], &vv);
+      d_decode_chunk<false>(tv, n, ws.ts + row0, nullptr, lane);
+      d_decode_chunk<true>(vv, n, nullptr, ws.val + row0, lane);
+    } else {                       // window-only ablation: synthesize rows
+      vv.dropped = 0;
       for (int i = lane; i < n; i += 64) {
         ws.ts[row0 + i] = dir.start_time[first + c] + (int64_t)i * 15000;
         ws.val[row0 + i] = (double)(i * 10);
       }
-      if (lane == 0) {
-        ChunkMeta& m = ws.cm[c];
-        m.row0 = row0; m.nrows = n;
-        m.start_time = dir.start_time[first + c];
-        m.end_time = dir.end_time[first + c];
-        m.dropped = 0; m.chunk_correction = 0; m.v0_nan = 0; m.last_for_update = 0;
-        m.dcount = 0; m.dense_corr = 0;
-      }
-      row0 += n;
-      continue;
     }
-    d_vec_open(blob + dir.ts_off[first + c], &tv);
-    d_vec_open(blob + dir.val_off[first + c], &vv);
-    int n = dir.num_rows[first + c];
-    if (row0 + n > FDB_MAX_ROWS_PER_SERIES) { n = 0; }   // guarded at upload; belt+braces
-
-    for (int i = lane; i < n; i += 64) {
-      ws.ts[row0 + i] = d_lv_at(&tv, i);
-      ws.val[row0 + i] = d_dv_at(&vv, i);
-    }
-
     if (lane == 0) {
       ChunkMeta& m = ws.cm[c];
       m.row0 = row0; m.nrows = n;
       m.start_time = dir.start_time[first + c];
       m.end_time = dir.end_time[first + c];
       m.dropped = vv.dropped;
-      m.chunk_correction = 0;
-      m.v0_nan = 0;
-      m.last_for_update = 0;
+      m.chunk_correction = 0; m.last_for_update = 0;
+      m.dcount = 0; m.dense_corr = 0; m.v0_nan = 0;
     }
     row0 += n;
   }
   if (lane == 0) ws.total_rows = row0;
-  __builtin_amdgcn_s_waitcnt(0);   // LDS writes visible within the wave
+  __builtin_amdgcn_s_waitcnt(0);   // wave-local LDS visibility
   __builtin_amdgcn_wave_barrier();
 
-  // ---- meta phase: search-guess slopes for every func; rate family adds the
-  // counter-correction scan (CorrectingDoubleVectorReader :325-342) recorded
-  // as a sparse drop table + updateCorrection scalars (:375-391,190-195) -----
+  // ---- meta phase ----------------------------------------------------------
+  // every kind: search-guess slope per chunk
+  // K_RATE: counter-correction scan (CorrectingDoubleVectorReader :325-342)
+  //         recorded as a sparse drop table + updateCorrection scalars
+  // K_PFX(_SQ): NaN-zeroed value (and square) prefix + non-NaN count prefix —
+  //         the ~97% window overlap then costs O(1) per window, not O(rows)
+  // K_MINMAX: 8-element group min/max
+  // K_CHANGES: change-indicator prefix (DoubleVectorDataReader64.changes :283-302)
   int table_used = 0;
   for (int c = 0; c < nchunks; c++) {
     ChunkMeta& m = ws.cm[c];
@@ -338,12 +436,13 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       int64_t tl = ws.ts[r0 + n - 1];
       m.inv_slope = (tl > m.ts0) ? (float)(n - 1) / (float)(tl - m.ts0) : 0.0f;
     }
+
     if constexpr (RATE_FAMILY) {
       if (lane == 0) m.v0_nan = isnan(ws.val[r0]);
       if (m.dropped) {
         double carry_corr = 0;
         double carry_x = -1.7976931348623157e308;   // 'last' starts Double.MinValue
-        int last_idx = -1;                          // last non-NaN index (for update)
+        int last_idx = -1;
         double last_val = 0;
         int dstart = table_used, dcnt = 0;
         bool overflow = false;
@@ -351,12 +450,10 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
           int i = base + lane;
           double raw = (i < n) ? ws.val[r0 + i] : 0;
           double x = (i < n && !isnan(raw)) ? raw : 0;
-          // drop contribution: x < last(previous x) adds previous x
           double px = __shfl_up(x, 1);
           if (lane == 0) px = carry_x;
           double ci = (i < n && x < px) ? px : 0;
           double scan = wave_incl_scan(ci, lane);
-          // record (pos, cumulative correction) for each drop in this iteration
           uint64_t mask = __ballot(ci != 0);
           int here = __popcll(mask);
           if (here) {
@@ -373,7 +470,6 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
           carry_x = __shfl(x, 63);
           if (i < n && !isnan(raw)) { last_idx = i; last_val = raw; }
         }
-        // wave-reduce max last_idx (value rides along)
         for (int off = 32; off > 0; off >>= 1) {
           int oi = __shfl_down(last_idx, off);
           double ov = __shfl_down(last_val, off);
@@ -389,7 +485,64 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         }
       } else if (lane == 0) {
         m.last_for_update = ws.val[r0 + n - 1];     // default updateCorrection
-        m.dcount = 0; m.dense_corr = 0;
+      }
+    }
+
+    if constexpr (KIND == K_PFX || KIND == K_PFX_SQ) {
+      double carry = 0, carry_sq = 0;
+      int ccarry = 0;
+      for (int base = 0; base < n; base += 64) {
+        int i = base + lane;
+        double raw = (i < n) ? ws.val[r0 + i] : NAN;
+        bool ok = (i < n) && !isnan(raw);
+        double x = ok ? raw : 0;
+        double s = wave_incl_scan(x, lane);
+        int cs = wave_incl_scan_i(ok ? 1 : 0, lane);
+        double sqs = 0;
+        if constexpr (KIND == K_PFX_SQ) sqs = wave_incl_scan(x * x, lane);
+        __builtin_amdgcn_wave_barrier();            // all raw reads precede writes
+        if (i < n) {
+          ws.val[r0 + i] = carry + s;               // val[] becomes the prefix
+          ws.cnt[r0 + i] = (uint16_t)(ccarry + cs);
+          if constexpr (KIND == K_PFX_SQ) ws.sq[r0 + i] = carry_sq + sqs;
+        }
+        carry += __shfl(s, 63);
+        ccarry += __shfl(cs, 63);
+        if constexpr (KIND == K_PFX_SQ) carry_sq += __shfl(sqs, 63);
+      }
+    }
+
+    if constexpr (KIND == K_MINMAX) {
+      constexpr bool IS_MIN = (FUNC == FN_MIN);
+      for (int base = 0; base < n; base += 64) {
+        int i = base + lane;
+        double x = (i < n) ? ws.val[r0 + i] : NAN;
+        // 8-lane group reduce (elements 8g..8g+7 are lanes 8k..8k+7)
+        #pragma unroll
+        for (int off = 1; off < 8; off <<= 1) {
+          double o = __shfl_xor(x, off);
+          if (!isnan(o) && (isnan(x) || (IS_MIN ? o < x : o > x))) x = o;
+        }
+        if ((lane & 7) == 0 && i < n)
+          ws.grp[(r0 >> 3) + c + (i >> 3)] = x;   // chunk-relative groups;
+                                                  // +c pads across chunk seams
+      }
+    }
+
+    if constexpr (KIND == K_CHANGES) {
+      // ind[i] = in-chunk change at i (i>0); prefix stored in cnt[]
+      int ccarry = 0;
+      double carry_x = NAN;
+      for (int base = 0; base < n; base += 64) {
+        int i = base + lane;
+        double x = (i < n) ? ws.val[r0 + i] : NAN;
+        double px = __shfl_up(x, 1);
+        if (lane == 0) px = carry_x;
+        int ind = (i > 0 && i < n && !isnan(x) && !isnan(px) && x != px) ? 1 : 0;
+        int s = wave_incl_scan_i(ind, lane);
+        if (i < n) ws.cnt[r0 + i] = (uint16_t)(ccarry + s);
+        ccarry += __shfl(s, 63);
+        carry_x = __shfl(x, 63);
       }
     }
   }
@@ -402,7 +555,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       out[(size_t)sid * num_windows] = ws.val[0] + (double)ws.ts[0];
     return;
   }
-  const int grp = group_ids[sid];
+  const int grp_id = group_ids[sid];
   for (int w = lane; w < num_windows; w += 64) {
     const int64_t wEnd = qstart + (int64_t)w * qstep;
     const int64_t wStart = wEnd - qwindow;
@@ -456,71 +609,66 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                                      lowestTime, lowestValue, highestTime, highestValue,
                                      isCounter, FUNC == FN_RATE);
     } else {
-      double sum = NAN, count = NAN, sqsum = NAN, mn = NAN, mx = NAN;
+      double sum = NAN, sqsum = NAN, mm = NAN;
       double changes = NAN, prev = NAN;
       int icount = 0;
+      bool started = false;                        // CountOverTime: saw a nonempty range
       for (int c = 0; c < ws.nchunks; c++) {
         const ChunkMeta& m = ws.cm[c];
         if (m.end_time < wStart) continue;
         const int64_t* seg = ws.ts + m.row0;
-        const double* vals = ws.val + m.row0;
         int startRow = lds_search_ge(seg, m.nrows, wStart, m.ts0, m.inv_slope);
         int endRow = lds_search_le(seg, m.nrows, wEnd, m.ts0, m.inv_slope);
         if (startRow <= endRow && endRow < m.nrows) {
-          if (FUNC == FN_SUM || FUNC == FN_AVG) {
-            double cs = NAN;                       // NaN-skipping chunk sum (:244-253)
-            for (int i = startRow; i <= endRow; i++) {
-              double x = vals[i];
-              if (!isnan(x)) { if (isnan(cs)) cs = 0; cs += x; }
-            }
+          const int a = m.row0 + startRow, b = m.row0 + endRow;
+          if constexpr (KIND == K_PFX || KIND == K_PFX_SQ) {
+            // chunk-range aggregates from the prefixes (O(1) per window)
+            double ps = ws.val[b] - (startRow ? ws.val[a - 1] : 0.0);
+            int pc = (int)ws.cnt[b] - (startRow ? (int)ws.cnt[a - 1] : 0);
+            double cs = pc > 0 ? ps : NAN;          // all-NaN range sums to NaN
             if (!isnan(cs) && isnan(sum)) sum = 0;
-            sum += cs;                             // NaN-poison quirk preserved
-            if (FUNC == FN_AVG)
-              for (int i = startRow; i <= endRow; i++) if (!isnan(vals[i])) icount++;
-          } else if (FUNC == FN_COUNT) {
-            if (isnan(count)) count = 0;
-            for (int i = startRow; i <= endRow; i++) if (!isnan(vals[i])) count += 1;
-          } else if (FUNC == FN_MIN || FUNC == FN_MAX) {
-            for (int i = startRow; i <= endRow; i++) {
-              double x = vals[i];
-              if (isnan(x)) continue;
-              if (FUNC == FN_MIN) mn = (isnan(mn) || x < mn) ? x : mn;
-              else mx = (isnan(mx) || x > mx) ? x : mx;
+            sum += cs;                              // NaN-poison quirk preserved
+            icount += pc;
+            started = true;
+            if constexpr (KIND == K_PFX_SQ) {
+              double qs = ws.sq[b] - (startRow ? ws.sq[a - 1] : 0.0);
+              double csq = pc > 0 ? qs : NAN;
+              if (!isnan(csq) && isnan(sqsum)) sqsum = 0;
+              sqsum += csq;
             }
-          } else if (FUNC == FN_STDDEV || FUNC == FN_STDVAR) {
-            double cs = NAN, csq = NAN; int cc = 0;
-            for (int i = startRow; i <= endRow; i++) {
-              double x = vals[i];
-              if (!isnan(x)) {
-                if (isnan(cs)) cs = 0;
-                if (isnan(csq)) csq = 0;
-                cs += x; csq += x * x; cc++;
-              }
+          } else if constexpr (KIND == K_MINMAX) {
+            constexpr bool IS_MIN = (FUNC == FN_MIN);
+            auto acc = [&](double x) {
+              if (!isnan(x) && (isnan(mm) || (IS_MIN ? x < mm : x > mm))) mm = x;
+            };
+            // chunk-relative full groups in [ga, gb), slot base sb
+            int ga = (startRow + 7) >> 3, gb = (endRow + 1) >> 3;
+            int sb = (m.row0 >> 3) + c;
+            if (ga < gb) {
+              for (int i = startRow; i < ga * 8; i++) acc(ws.val[m.row0 + i]);
+              for (int g = ga; g < gb; g++) acc(ws.grp[sb + g]);
+              for (int i = gb * 8; i <= endRow; i++) acc(ws.val[m.row0 + i]);
+            } else {
+              for (int i = a; i <= b; i++) acc(ws.val[i]);
             }
-            if (!isnan(cs) && isnan(sum)) sum = 0;
-            sum += cs;
-            if (!isnan(csq) && isnan(sqsum)) sqsum = 0;
-            sqsum += csq;
-            icount += cc;
-          } else if (FUNC == FN_CHANGES) {
-            if (isnan(changes)) changes = 0;       // double changes (:283-302)
-            double prevV = prev, ch = 0;
-            for (int i = startRow; i <= endRow; i++) {
-              double x = vals[i];
-              if (!isnan(x) && prevV != x && !isnan(prevV)) ch += 1;
-              prevV = x;
-            }
-            changes += ch; prev = prevV;
+          } else {  // K_CHANGES
+            if (isnan(changes)) changes = 0;
+            // in-chunk changes from the indicator prefix over (startRow, endRow]
+            double ch = (double)((int)ws.cnt[b] - (int)ws.cnt[a]);
+            // boundary: first sample vs carry from the previous chunk range
+            double vs = ws.val[a];
+            if (!isnan(vs) && !isnan(prev) && vs != prev) ch += 1;
+            changes += ch;
+            prev = ws.val[b];                       // carried raw (NaN possible)
           }
         }
         if (m.end_time >= wEnd) break;
       }
       switch (FUNC) {
         case FN_SUM:   result = sum; break;
-        case FN_COUNT: result = count; break;
+        case FN_COUNT: result = started ? (double)icount : NAN; break;
         case FN_AVG:   result = icount > 0 ? sum / icount : (isnan(sum) ? sum : 0); break;
-        case FN_MIN:   result = mn; break;
-        case FN_MAX:   result = mx; break;
+        case FN_MIN: case FN_MAX: result = mm; break;
         case FN_STDDEV: case FN_STDVAR: {
           if (icount > 0) {
             double avg = sum / icount;
@@ -536,7 +684,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       out[(size_t)sid * num_windows + w] = result;
     } else if (!isnan(result)) {
       // fastReduce fused: RowAggregator merge per group cell
-      size_t cell = (size_t)grp * num_windows + w;
+      size_t cell = (size_t)grp_id * num_windows + w;
       switch (agg_id) {
         case AGG_SUM: case AGG_AVG:
           atomicAdd(&out[cell], result);
