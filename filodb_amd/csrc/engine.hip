@@ -375,10 +375,14 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int sid = blockIdx.x * WAVES_PER_BLOCK + wave;
-  if (sid >= num_series) return;
   Ws<KIND>& ws = ws_all[wave];
 
+  // grid-stride over series: a few resident blocks loop over the shard instead
+  // of one block per 4 series — 250k tiny-workgroup dispatches cost multiple ms
+  // of pure launch churn (measured ~20-26ns each), so the loop, not the grid,
+  // walks the series
+  for (int sid = blockIdx.x * WAVES_PER_BLOCK + wave; sid < num_series;
+       sid += gridDim.x * WAVES_PER_BLOCK) {
   // ---- decode phase: all chunks of this series into LDS --------------------
   const int first = series_first[sid];
   int nchunks = series_nchunks[sid];
@@ -553,7 +557,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   if (!(phase_mask & 2)) {            // decode-only ablation: publish a checksum
     if (lane == 0 && ws.total_rows > 0)
       out[(size_t)sid * num_windows] = ws.val[0] + (double)ws.ts[0];
-    return;
+    continue;
   }
   const int grp_id = group_ids[sid];
   for (int w = lane; w < num_windows; w += 64) {
@@ -701,6 +705,10 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       }
     }
   }
+  // next series reuses this wave's LDS slot: wave-local ordering is enough
+  __builtin_amdgcn_s_waitcnt(0);
+  __builtin_amdgcn_wave_barrier();
+  }  // series grid-stride loop
 }
 
 // presentation fixup for aggregated grids (NaN where no contributions; mean for avg)
@@ -853,6 +861,9 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
   DirSoA dir{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
   int nw = fdb_num_windows(q);
   int grid = (d->num_series + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+  // enough blocks to fill every CU at worst-case occupancy, few enough that
+  // workgroup dispatch is off the critical path
+  if (grid > 2048) grid = 2048;
   #define CASE(F) case F: \
     hipLaunchKernelGGL(scan_kernel<F>, dim3(grid), dim3(BLOCK_THREADS), 0, e->stream, \
       d->blob, dir, d->series_first, d->series_nchunks, d->group_ids, d->num_series, \
