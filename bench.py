@@ -183,6 +183,15 @@ def main():
         algo_bytes = ds.payload_bytes + out_bytes
         ach = algo_bytes / (kernel_ms / 1000.0) if kernel_ms > 0 else None
         traffic = os.environ.get("FDB_TRAFFIC_BYTES_PER_LAUNCH")
+        if not traffic:
+            # PMC-derived HBM bytes per launch, measured by rocprofv3 --pmc
+            # FETCH_SIZE/WRITE_SIZE runs (profiles/; FETCH doubled per the
+            # gfx950 read-side calibration, MI355X_MICROARCH.md §HBM)
+            cal = os.path.join(REPO, "profiles", "traffic_calibration.json")
+            if os.path.exists(cal):
+                with open(cal) as f:
+                    c = json.load(f)
+                traffic = c.get(args.workload)
         roofline = {
             "bound": "hbm",
             "achieved": round(ach / 1e9, 2) if ach else None,

@@ -560,6 +560,89 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     continue;
   }
   const int grp_id = group_ids[sid];
+  // single-chunk fast path (the common shape: one 400-row-capped chunk per
+  // query span): chunk meta lives in registers, no per-window chunk loop
+  if (ws.nchunks == 1 && ws.cm[0].nrows > 0) {
+    const ChunkMeta mr = ws.cm[0];       // one LDS read burst, then registers
+    const int64_t* seg = ws.ts;
+    #pragma unroll 2
+    for (int w = lane; w < num_windows; w += 64) {
+      const int64_t wEnd = qstart + (int64_t)w * qstep;
+      const int64_t wStart = wEnd - qwindow;
+      double result = NAN;
+      if (mr.end_time >= wStart) {
+        int startRow = lds_search_ge(seg, mr.nrows, wStart, mr.ts0, mr.inv_slope);
+        int endRow = lds_search_le(seg, mr.nrows, wEnd, mr.ts0, mr.inv_slope);
+        if (startRow <= endRow && endRow < mr.nrows) {
+          if constexpr (RATE_FAMILY) {
+            constexpr bool isCounter = (FUNC != FN_DELTA);
+            if (!(isCounter && startRow == 0 && endRow == 0 && mr.v0_nan)) {
+              int64_t st = seg[startRow], en = seg[endRow];
+              if (en > st) {
+                int numSamples = endRow - startRow + 1;
+                double lo = isCounter ? d_corrected(ws, mr, startRow)
+                                      : ws.val[startRow];
+                double hi = isCounter ? d_corrected(ws, mr, endRow)
+                                      : ws.val[endRow];
+                result = d_extrapolated_rate(wStart, wEnd, numSamples, st, lo,
+                                             en, hi, isCounter, FUNC == FN_RATE);
+              }
+            }
+          } else if constexpr (KIND == K_PFX || KIND == K_PFX_SQ) {
+            double ps = ws.val[endRow] - (startRow ? ws.val[startRow - 1] : 0.0);
+            int pc = (int)ws.cnt[endRow] - (startRow ? (int)ws.cnt[startRow - 1] : 0);
+            if (FUNC == FN_SUM) result = pc > 0 ? ps : NAN;
+            else if (FUNC == FN_COUNT) result = (double)pc;
+            else if (FUNC == FN_AVG) result = pc > 0 ? ps / pc : NAN;
+            else {  // stddev / stdvar over one chunk range
+              if (pc > 0) {
+                double qs = ws.sq[endRow] - (startRow ? ws.sq[startRow - 1] : 0.0);
+                double avg = ps / pc;
+                double r = qs / pc - avg * avg;
+                result = (FUNC == FN_STDDEV) ? sqrt(r) : r;
+              } else result = NAN;
+            }
+          } else if constexpr (KIND == K_MINMAX) {
+            constexpr bool IS_MIN = (FUNC == FN_MIN);
+            double mm = NAN;
+            auto acc = [&](double x) {
+              if (!isnan(x) && (isnan(mm) || (IS_MIN ? x < mm : x > mm))) mm = x;
+            };
+            int ga = (startRow + 7) >> 3, gb = (endRow + 1) >> 3;
+            if (ga < gb) {
+              for (int i = startRow; i < ga * 8; i++) acc(ws.val[i]);
+              for (int g = ga; g < gb; g++) acc(ws.grp[g]);
+              for (int i = gb * 8; i <= endRow; i++) acc(ws.val[i]);
+            } else {
+              for (int i = startRow; i <= endRow; i++) acc(ws.val[i]);
+            }
+            result = mm;
+          } else {  // K_CHANGES, single chunk: prefix diff, prev starts NaN
+            result = (double)((int)ws.cnt[endRow] - (int)ws.cnt[startRow]);
+          }
+        }
+      }
+      if (agg_id == AGG_NONE) {
+        out[(size_t)sid * num_windows + w] = result;
+      } else if (!isnan(result)) {
+        size_t cell = (size_t)grp_id * num_windows + w;
+        switch (agg_id) {
+          case AGG_SUM: case AGG_AVG:
+            atomicAdd(&out[cell], result); atomicAdd(&out_cnt[cell], 1.0); break;
+          case AGG_COUNT:
+            atomicAdd(&out[cell], 1.0); atomicAdd(&out_cnt[cell], 1.0); break;
+          case AGG_MIN: atomic_min_max_f64(&out[cell], result, true);
+                        atomicAdd(&out_cnt[cell], 1.0); break;
+          case AGG_MAX: atomic_min_max_f64(&out[cell], result, false);
+                        atomicAdd(&out_cnt[cell], 1.0); break;
+        }
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_wave_barrier();
+    continue;
+  }
+
   for (int w = lane; w < num_windows; w += 64) {
     const int64_t wEnd = qstart + (int64_t)w * qstep;
     const int64_t wStart = wEnd - qwindow;
