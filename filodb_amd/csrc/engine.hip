@@ -23,6 +23,7 @@
 #include <cstdio>
 #include <cmath>
 #include <vector>
+#include <cstdlib>
 
 #include "chunk_format.h"
 #include "../../include/filodb_amd.h"
@@ -946,7 +947,9 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
   int grid = (d->num_series + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
   // enough blocks to fill every CU at worst-case occupancy, few enough that
   // workgroup dispatch is off the critical path
-  if (grid > 2048) grid = 2048;
+  int cap = 2048;
+  if (const char* g = getenv("FDB_GRID")) cap = atoi(g);   // perf experiments
+  if (cap > 0 && grid > cap) grid = cap;
   #define CASE(F) case F: \
     hipLaunchKernelGGL(scan_kernel<F>, dim3(grid), dim3(BLOCK_THREADS), 0, e->stream, \
       d->blob, dir, d->series_first, d->series_nchunks, d->group_ids, d->num_series, \
