@@ -357,7 +357,7 @@ __device__ void atomic_min_max_f64(double* addr, double val, bool is_min) {
 //   per-window evaluation with O(1)-ish row location and O(1) prefix lookups
 // ---------------------------------------------------------------------------
 template <int FUNC>
-__global__ __launch_bounds__(BLOCK_THREADS)
+__global__ __launch_bounds__(BLOCK_THREADS, 5)
 void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                  const int32_t* __restrict__ series_first,
                  const int32_t* __restrict__ series_nchunks,
@@ -947,7 +947,7 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
   int grid = (d->num_series + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
   // enough blocks to fill every CU at worst-case occupancy, few enough that
   // workgroup dispatch is off the critical path
-  int cap = 2048;
+  int cap = 8192;
   if (const char* g = getenv("FDB_GRID")) cap = atoi(g);   // perf experiments
   if (cap > 0 && grid > cap) grid = cap;
   #define CASE(F) case F: \
