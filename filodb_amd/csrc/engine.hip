@@ -38,8 +38,8 @@ void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
 // ---------------------------------------------------------------------------
 // capacity limits (round 1)
 // ---------------------------------------------------------------------------
-#define FDB_MAX_ROWS_PER_SERIES 416   // LDS-resident rows per series
-#define FDB_MAX_CHUNKS_PER_SERIES 16
+#define FDB_MAX_ROWS_PER_SERIES 400   // LDS-resident rows per series (= chunk row cap)
+#define FDB_MAX_CHUNKS_PER_SERIES 12
 #define WAVES_PER_BLOCK 4
 #define BLOCK_THREADS (WAVES_PER_BLOCK * 64)
 
@@ -357,7 +357,9 @@ __device__ void atomic_min_max_f64(double* addr, double val, bool is_min) {
 //   per-window evaluation with O(1)-ish row location and O(1) prefix lookups
 // ---------------------------------------------------------------------------
 template <int FUNC>
-__global__ __launch_bounds__(BLOCK_THREADS, 5)
+// rate family carries more live state (correction meta): forcing 5 waves/SIMD
+// makes it spill; the gauge kinds fit 5 waves cleanly
+__global__ __launch_bounds__(BLOCK_THREADS, (FUNC <= FN_DELTA) ? 4 : 5)
 void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                  const int32_t* __restrict__ series_first,
                  const int32_t* __restrict__ series_nchunks,
