@@ -53,6 +53,13 @@ WORKLOADS = {
                        step_ms=15000, func=fdb.FN_RATE, agg=fdb.AGG_SUM,
                        window=300_000, qstep=15000, n_groups=1000,
                        label="1M counter series, sum by(job)(rate()[5m]), 1000 groups"),
+    # BASELINE configs[3]: 100k hist series x 64 buckets,
+    # histogram_quantile(0.99, sum(rate([5m])) by (le-group))
+    "hist_100k": dict(kind="hist", n_series=100_000, n_samples=240,
+                      step_ms=15000, func=11, agg=1, nb=64,
+                      window=300_000, qstep=15000, n_groups=10,
+                      label="100k hist series x 64 buckets, "
+                            "histogram_quantile(0.99, sum(rate()[5m])) by group"),
     "smoke": dict(kind="counter", n_series=20_000, n_samples=240,
                   step_ms=15000, func=fdb.FN_RATE, agg=fdb.AGG_NONE,
                   window=300_000, qstep=15000, n_groups=100,
@@ -62,7 +69,8 @@ WORKLOADS = {
 
 def build_store(w, rank):
     st = fdb.ChunkStore()
-    kind = fdb.COL_COUNTER if w["kind"] == "counter" else fdb.COL_GAUGE
+    kind = {"counter": fdb.COL_COUNTER, "gauge": fdb.COL_GAUGE,
+            "hist": fdb.COL_HIST}[w["kind"]]
     st.synth_generate(kind, w["n_series"], w["n_samples"], start_ts=T0,
                       step_ms=w["step_ms"], jitter_ms=250, lam=10.0,
                       reset_p=0.001, n_groups=w["n_groups"], seed=42 + rank)
@@ -73,7 +81,8 @@ def build_store(w, rank):
 def make_query(w):
     span = w["n_samples"] * w["step_ms"]
     return fdb.make_query(T0, w["qstep"], T0 + span, w["window"], w["func"],
-                          w["agg"], w["n_groups"] if w["agg"] != fdb.AGG_NONE else 0)
+                          w["agg"], w["n_groups"] if w["agg"] != fdb.AGG_NONE else 0,
+                          param=0.99 if w["kind"] == "hist" else 0.0)
 
 
 def cpu_baseline(w, budget_s=12.0):
@@ -90,12 +99,12 @@ def cpu_baseline(w, budget_s=12.0):
         st = build_store(sub, rank=0)
         nw = q.num_windows
         t0 = time.perf_counter()
-        if w["agg"] == fdb.AGG_NONE:
-            oracle.query_exec(st.view(), q, st.num_series, nw, nthreads=cores)
+        if w["kind"] == "hist":
+            oracle.query_exec_hist(st.view(), q, w["nb"])
         else:
             oracle.query_exec(st.view(), q, st.num_series, nw, nthreads=cores)
         elapsed = time.perf_counter() - t0
-        samples = n * w["n_samples"]
+        samples = n * w["n_samples"] * (w.get("nb", 1) if w["kind"] == "hist" else 1)
         if elapsed > 3.0 or n >= w["n_series"]:
             break
         n = min(w["n_series"], int(n * max(2, 6.0 / max(elapsed, 0.05))))
@@ -143,7 +152,10 @@ def main():
 
     # output buffers resident on device (torch tensors so an RCCL reduce could
     # consume them directly); allocated before the timed region
-    if q.agg_id == fdb.AGG_NONE:
+    if w["kind"] == "hist":
+        out = torch.zeros(1, dtype=torch.float64, device=f"cuda:{local_rank}")
+        cnt = None
+    elif q.agg_id == fdb.AGG_NONE:
         out = torch.empty(st.num_series * nw, dtype=torch.float64,
                           device=f"cuda:{local_rank}")
         cnt = None
@@ -152,16 +164,43 @@ def main():
                           device=f"cuda:{local_rank}")
         cnt = torch.empty_like(out)
 
-    # warmup pass (also yields avg HIP-event kernel ms for the roofline), then
-    # wall-time EXACTLY `steps` launches between barriers+synchronize.
-    kernel_ms = eng.bench(ds, q, out, out_counts=cnt, on_device=True,
-                          warmup=args.warmup, iters=args.steps)
+    is_hist = w["kind"] == "hist"
+    nb = w.get("nb", 0)
+    if is_hist:
+        hist_sums = torch.zeros(q.num_groups * nw * nb, dtype=torch.float64,
+                                device=f"cuda:{local_rank}")
+        hist_cnt = torch.zeros(q.num_groups * nw, dtype=torch.float64,
+                               device=f"cuda:{local_rank}")
+        hist_quant = torch.zeros(q.num_groups * nw, dtype=torch.float64,
+                                 device=f"cuda:{local_rank}")
+
+        def hist_step():
+            eng.query_hist(ds, q, nb, out_bucket_sums=hist_sums,
+                           out_counts=hist_cnt, out_quantile=hist_quant,
+                           on_device=True)
+        for _ in range(args.warmup):
+            hist_step()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            hist_step()
+        torch.cuda.synchronize()
+        kernel_ms = (time.perf_counter() - t0) / args.steps * 1000  # sync per call
+    else:
+        # warmup pass (also yields avg HIP-event kernel ms for the roofline), then
+        # wall-time EXACTLY `steps` launches between barriers+synchronize.
+        kernel_ms = eng.bench(ds, q, out, out_counts=cnt, on_device=True,
+                              warmup=args.warmup, iters=args.steps)
     if dist:
         dist.barrier()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
-    _ = eng.bench(ds, q, out, out_counts=cnt, on_device=True,
-                  warmup=0, iters=args.steps)
+    if is_hist:
+        for _ in range(args.steps):
+            hist_step()
+    else:
+        _ = eng.bench(ds, q, out, out_counts=cnt, on_device=True,
+                      warmup=0, iters=args.steps)
     torch.cuda.synchronize()
     if dist:
         dist.barrier()
@@ -173,6 +212,8 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         ms_per_step = float(t.item())
 
+    if w["kind"] == "hist":
+        samples_per_step *= w.get("nb", 1)   # bucket-samples (SURVEY §8d numerator)
     total_samples_per_step = samples_per_step * n_gpus
     value = total_samples_per_step / (ms_per_step / 1000.0)
 

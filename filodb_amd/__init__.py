@@ -18,10 +18,11 @@ FN_RATE, FN_INCREASE, FN_DELTA = 0, 1, 2
 FN_SUM_OVER_TIME, FN_COUNT_OVER_TIME, FN_AVG_OVER_TIME = 3, 4, 5
 FN_MIN_OVER_TIME, FN_MAX_OVER_TIME = 6, 7
 FN_STDDEV_OVER_TIME, FN_STDVAR_OVER_TIME, FN_CHANGES = 8, 9, 10
+FN_HIST_RATE = 11
 # aggregation ids (RowAggregator implementations)
 AGG_NONE, AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX, AGG_AVG = 0, 1, 2, 3, 4, 5
 # column kinds
-COL_GAUGE, COL_COUNTER = 0, 1
+COL_GAUGE, COL_COUNTER, COL_HIST = 0, 1, 2
 
 _c_double_p = ctypes.POINTER(ctypes.c_double)
 
@@ -32,6 +33,7 @@ class Query(ctypes.Structure):
         ("end", ctypes.c_int64), ("window", ctypes.c_int64),
         ("func_id", ctypes.c_int32), ("agg_id", ctypes.c_int32),
         ("num_groups", ctypes.c_int32), ("_pad", ctypes.c_int32),
+        ("param", ctypes.c_double),
     ]
 
     @property
@@ -87,6 +89,15 @@ def lib():
         L.fdb_chunk_get.argtypes = [ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32,
                                     ctypes.POINTER(ChunkInfo)]
         L.fdb_store_view.argtypes = [ctypes.c_void_p, ctypes.POINTER(View)]
+        L.fdb_series_append_hist.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                             ctypes.POINTER(ctypes.c_int64),
+                                             ctypes.POINTER(ctypes.c_uint64),
+                                             ctypes.c_int32, ctypes.c_int32,
+                                             ctypes.c_double, ctypes.c_double]
+        L.fdb_query_exec_hist.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                          ctypes.POINTER(Query), ctypes.c_int32,
+                                          _c_double_p, _c_double_p, _c_double_p,
+                                          ctypes.c_int32]
         L.fdb_synth_generate.argtypes = [ctypes.c_void_p, ctypes.c_int32, ctypes.c_int32,
                                          ctypes.c_int32, ctypes.c_int64, ctypes.c_int32,
                                          ctypes.c_int32, ctypes.c_double, ctypes.c_double,
@@ -173,6 +184,16 @@ class ChunkStore:
                                         start_ts, step_ms, jitter_ms, lam, reset_p,
                                         n_groups, seed), "synth_generate")
 
+    def append_hist(self, sid, ts, bucket_values, first=2.0, mult=2.0):
+        ts = np.ascontiguousarray(ts, dtype=np.int64)
+        bv = np.ascontiguousarray(bucket_values, dtype=np.uint64)
+        n, nb = bv.shape
+        assert len(ts) == n
+        _check(lib().fdb_series_append_hist(
+            self._h, sid, ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+            bv.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)), n, nb,
+            first, mult), "append_hist")
+
     def cut_chunk(self, sid):
         _check(lib().fdb_series_cut_chunk(self._h, sid), "cut_chunk")
 
@@ -243,6 +264,15 @@ class Engine:
         _check(rc, "query_bench")
         return ms.value
 
+    def query_hist(self, dataset, q: Query, num_buckets,
+                   out_bucket_sums=None, out_counts=None, out_quantile=None,
+                   on_device=False):
+        rc = lib().fdb_query_exec_hist(
+            self._h, dataset._h, ctypes.byref(q), num_buckets,
+            _as_f64_ptr(out_bucket_sums), _as_f64_ptr(out_counts),
+            _as_f64_ptr(out_quantile), 1 if on_device else 0)
+        _check(rc, "query_exec_hist")
+
     def synchronize(self):
         _check(lib().fdb_engine_synchronize(self._h), "synchronize")
 
@@ -272,10 +302,12 @@ def dataset_out_len(dataset, q: Query):
     return q.num_groups * q.num_windows
 
 
-def make_query(start, step, end, window, func_id, agg_id=AGG_NONE, num_groups=0):
+def make_query(start, step, end, window, func_id, agg_id=AGG_NONE, num_groups=0,
+               param=0.0):
     q = Query()
     q.start, q.step, q.end, q.window = start, step, end, window
     q.func_id, q.agg_id, q.num_groups = func_id, agg_id, num_groups
+    q.param = param
     return q
 
 

@@ -356,6 +356,10 @@ struct Series {
   std::vector<double>  buf_vals;
   bool     buf_drop = false;          // DoubleCounterAppender drop flag (per chunk)
   double   buf_last = -1.7976931348623157e308;  // Double.MinValue
+  // histogram column state (FDB_COL_HIST)
+  std::vector<uint64_t> buf_hist;     // row-major [rows × num_buckets]
+  int32_t  num_buckets = 0;
+  double   bucket_first = 0, bucket_mult = 0;
   std::vector<Chunk> chunks;
 };
 
@@ -390,6 +394,96 @@ extern "C" int32_t fdb_store_add_series(fdb_store_t* s, int32_t group_id, int32_
   return (int32_t)s->series.size() - 1;
 }
 
+// packDelta of one histogram's cumulative bucket values (the wire format of
+// BinaryHistogram.writeDelta, HistogramVector.scala:196-209)
+static void hist_pack_raw(const uint64_t* vals, int nb, bytes& out) {
+  std::vector<int64_t> v(nb);
+  for (int b = 0; b < nb; b++) v[(size_t)b] = (int64_t)vals[b];
+  out.clear();
+  np_pack_delta(v.data(), nb, out, 0);
+}
+
+// pack8 stream of (bucket-consecutive deltas − section-base deltas)
+// (NibblePack.DeltaSectDiffPackSink.process, NibblePack.scala:318-338)
+static void hist_pack_diff(const int64_t* deltas, const int64_t* orig, int nb, bytes& out) {
+  out.clear();
+  size_t pos = 0;
+  int64_t tmp[8];
+  for (int i = 0; i < nb; i += 8) {
+    int m = nb - i < 8 ? nb - i : 8;
+    for (int k = 0; k < m; k++) tmp[k] = deltas[i + k] - orig[i + k];
+    for (int k = m; k < 8; k++) tmp[k] = 0;
+    pos = np_pack8(tmp, out, pos);
+  }
+}
+
+// Encodes the buffered histogram rows as a sect-delta histogram vector
+static void encode_hist_chunk(Series& se, int n, bytes& out) {
+  const int nb = se.num_buckets;
+  out.clear();
+  put_u16(out, 4, FDB_WF_HIST_SECTDELTA);
+  put_u16(out, FDB_HIST_OFF_NUMHIST, (uint16_t)n);
+  put_u8(out, FDB_HIST_OFF_FMT, FDB_HIST_FMT_GEOMETRIC_DELTA);
+  put_u16(out, FDB_HIST_OFF_DEFSIZE, 18);
+  put_u16(out, FDB_HIST_OFF_DEF, (uint16_t)nb);
+  uint64_t fb, mu;
+  memcpy(&fb, &se.bucket_first, 8); memcpy(&mu, &se.bucket_mult, 8);
+  put_u64(out, FDB_HIST_OFF_DEF + 2, fb);
+  put_u64(out, FDB_HIST_OFF_DEF + 10, mu);
+
+  std::vector<int64_t> lastDeltas(nb, 0), origDeltas(nb, 0), deltas(nb);
+  bytes blob;
+  size_t sect = 0;                        // current section header offset; 0 = none
+  auto sect_bytes = [&]() { uint16_t v; memcpy(&v, &out[sect], 2); return (int)v; };
+  auto sect_elems = [&]() { return (int)out[sect + 2]; };
+  auto new_section = [&](int type) {
+    sect = out.size();
+    put_u16(out, sect, 0);
+    put_u8(out, sect + 2, 0);
+    put_u8(out, sect + 3, (uint8_t)type);
+  };
+  for (int e = 0; e < n; e++) {
+    const uint64_t* row = se.buf_hist.data() + (size_t)e * nb;
+    bool dropped = false;
+    for (int b = 0; b < nb; b++) {
+      int64_t d = (int64_t)row[b] - (b ? (int64_t)row[b - 1] : 0);
+      if (d < lastDeltas[(size_t)b]) dropped = true;   // sink drop rule :321
+      deltas[(size_t)b] = d;
+    }
+    bytes raw;
+    hist_pack_raw(row, nb, raw);
+    bool fresh;
+    if (dropped) {                                     // TypeDrop section, raw base
+      new_section(1);
+      blob = raw;
+      fresh = true;
+    } else if (e == 0 || sect_elems() >= FDB_HIST_MAX_PER_SECTION ||
+               sect_bytes() + (int)raw.size() >= 65536) {
+      new_section(0);
+      blob = raw;
+      fresh = true;
+    } else {
+      hist_pack_diff(deltas.data(), origDeltas.data(), nb, blob);
+      fresh = false;
+      if (sect_elems() >= FDB_HIST_MAX_PER_SECTION ||
+          sect_bytes() + (int)blob.size() >= 65536) {  // appendBlob's own check
+        new_section(0);
+        blob = raw;                                    // cannot happen after above; safety
+        fresh = true;
+      }
+    }
+    size_t at = out.size();
+    put_u16(out, at, (uint16_t)blob.size());
+    out.resize(at + 2 + blob.size());
+    memcpy(out.data() + at + 2, blob.data(), blob.size());
+    put_u16(out, sect, (uint16_t)(sect_bytes() + 2 + (int)blob.size()));
+    put_u8(out, sect + 2, (uint8_t)(sect_elems() + 1));
+    lastDeltas = deltas;
+    if (fresh) origDeltas = deltas;
+  }
+  put_u32(out, 0, (uint32_t)(out.size() - 4));
+}
+
 static int32_t cut_chunk(fdb_store_t* s, Series& se) {
   int n = (int)se.buf_ts.size();
   if (n == 0) return FDB_OK;
@@ -398,12 +492,45 @@ static int32_t cut_chunk(fdb_store_t* s, Series& se) {
   c.start_time = se.buf_ts.front();
   c.end_time = se.buf_ts.back();
   encode_timestamps(se.buf_ts.data(), n, c.ts_bytes);
-  encode_doubles(se.buf_vals.data(), n, se.col_kind == FDB_COL_COUNTER && se.buf_drop,
-                 c.val_bytes);
+  if (se.col_kind == FDB_COL_HIST) {
+    encode_hist_chunk(se, n, c.val_bytes);
+    se.buf_hist.clear();
+  } else {
+    encode_doubles(se.buf_vals.data(), n, se.col_kind == FDB_COL_COUNTER && se.buf_drop,
+                   c.val_bytes);
+  }
   se.chunks.push_back(std::move(c));
   se.buf_ts.clear(); se.buf_vals.clear();
   se.buf_drop = false;
   se.buf_last = -1.7976931348623157e308;
+  return FDB_OK;
+}
+
+extern "C" int32_t fdb_series_append_hist(fdb_store_t* s, int32_t sid,
+                                          const int64_t* ts, const uint64_t* bucket_values,
+                                          int32_t n, int32_t num_buckets,
+                                          double bucket_first, double bucket_mult) {
+  if (s->sealed) { fdb_set_error("store is sealed"); return FDB_ERR_BADARG; }
+  if (sid < 0 || sid >= (int32_t)s->series.size()) { fdb_set_error("bad series id"); return FDB_ERR_BADARG; }
+  Series& se = s->series[(size_t)sid];
+  if (se.col_kind != FDB_COL_HIST) { fdb_set_error("series %d is not FDB_COL_HIST", sid); return FDB_ERR_BADARG; }
+  if (num_buckets < 1 || num_buckets > 64) { fdb_set_error("num_buckets must be 1..64"); return FDB_ERR_BADARG; }
+  if (se.num_buckets == 0) {
+    se.num_buckets = num_buckets;
+    se.bucket_first = bucket_first;
+    se.bucket_mult = bucket_mult;
+  } else if (se.num_buckets != num_buckets) {
+    fdb_set_error("bucket scheme mismatch"); return FDB_ERR_BADARG;
+  }
+  for (int32_t i = 0; i < n; i++) {
+    if (!se.buf_ts.empty() && ts[i] < se.buf_ts.back()) {
+      fdb_set_error("timestamps must be nondecreasing"); return FDB_ERR_BADARG;
+    }
+    se.buf_ts.push_back(ts[i]);
+    se.buf_hist.insert(se.buf_hist.end(), bucket_values + (size_t)i * num_buckets,
+                       bucket_values + (size_t)(i + 1) * num_buckets);
+    if ((int32_t)se.buf_ts.size() >= s->max_rows) cut_chunk(s, se);
+  }
   return FDB_OK;
 }
 
@@ -547,6 +674,28 @@ extern "C" int32_t fdb_synth_generate(fdb_store_t* s, int32_t kind, int32_t n_se
       int64_t ts = start_ts + (int64_t)k * step_ms + jitter;
       if (ts < last_ts) ts = last_ts;
       last_ts = ts;
+      if (kind == FDB_COL_HIST) {
+        // config #4 shape: 64 geometric buckets, per-interval cumulative counts
+        const int nb = 64;
+        if (se.num_buckets == 0) {
+          se.num_buckets = nb; se.bucket_first = 2.0; se.bucket_mult = 2.0;
+          se.buf_hist.reserve((size_t)s->max_rows * nb);
+        }
+        static thread_local std::vector<uint64_t> cum_int;
+        if (k == 0) { cum_int.assign(nb, 0); }
+        if (u01(st) < reset_p) cum_int.assign(nb, 0);
+        uint64_t row = 0;
+        se.buf_ts.push_back(ts);
+        size_t at = se.buf_hist.size();
+        se.buf_hist.resize(at + nb);
+        for (int b = 0; b < nb; b++) {
+          cum_int[(size_t)b] += (uint64_t)poisson_knuth(st, lam / 8.0);
+          row += cum_int[(size_t)b];
+          se.buf_hist[at + (size_t)b] = row;
+        }
+        if ((int32_t)se.buf_ts.size() >= s->max_rows) cut_chunk(s, se);
+        continue;
+      }
       double v;
       if (kind == FDB_COL_COUNTER) {
         if (u01(st) < reset_p) counter = 0;          // counter reset

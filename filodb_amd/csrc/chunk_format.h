@@ -27,6 +27,24 @@
 #define FDB_SUBTYPE_INT_NOMASK       0x08
 #define FDB_SUBTYPE_H_SECTDELTA      0x12
 
+/* Sect-delta histogram vector (HistogramVector.scala:237-254,491-545):
+ *  +0  i32 length word (bytes after it)
+ *  +4  u16 wireformat 0x1209
+ *  +6  u16 numHistograms
+ *  +8  u8  format code (0x03 = geometric + NibblePacked delta longs)
+ *  +9  u16 bucketDefNumBytes
+ *  +11 bucket def: u16 numBuckets, f64 firstBucket, f64 multiplier (geometric)
+ *  +11+def: sections; each section (Section.scala:17-24):
+ *     +0 u16 bytes after 4-byte header; +2 u8 numElements; +3 u8 type(0|1=drop)
+ *     elements: u16 len + NibblePack stream */
+#define FDB_WF_HIST_SECTDELTA 0x1209
+#define FDB_HIST_FMT_GEOMETRIC_DELTA 0x03
+#define FDB_HIST_OFF_NUMHIST   6
+#define FDB_HIST_OFF_FMT       8
+#define FDB_HIST_OFF_DEFSIZE   9
+#define FDB_HIST_OFF_DEF       11
+#define FDB_HIST_MAX_PER_SECTION 16   /* AppendableSectDeltaHistVector :501 */
+
 #define FDB_WF(major, sub) ((uint16_t)((((sub) & 0xff) << 8) | ((major) & 0xff)))
 
 /* Composite wireformat words actually produced on this path */

@@ -797,6 +797,208 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   }  // series grid-stride loop
 }
 
+
+// ---------------------------------------------------------------------------
+// histogram scan kernel (BASELINE config #4)
+//   histogram_quantile(q, sum(rate(hist[w])) by group)
+//   wave = one series, lane = one bucket (<=64); elements walked in time order
+//   decoding the sect-delta NibblePack streams in lock-step, with a ring of
+//   active windows (window/step+2 <= FDB_HIST_RING slots).
+//   Semantics: SectDeltaHistogramReader (HistogramVector.scala:628-737),
+//   HistogramRateFunctionBase (RateFunctions.scala:330-400),
+//   HistSumRowAggregator.scala:20-29.
+// ---------------------------------------------------------------------------
+#define FDB_HIST_RING 24
+#define HIST_WAVES 2
+
+struct HistWs {
+  int64_t ts[FDB_MAX_ROWS_PER_SERIES];
+  double  ring_lo[FDB_HIST_RING][64];   // lowestValue per bucket per active window
+  int32_t ring_w[FDB_HIST_RING];        // window id the slot holds (-1 none)
+  int32_t ring_e[FDB_HIST_RING];        // start element index
+};
+
+__device__ __forceinline__ int64_t d_fdiv(int64_t a, int64_t b) {
+  return a >= 0 ? a / b : -((-a + b - 1) / b);
+}
+
+// wave-wide inclusive prefix sum over i64 (bucket-cumulative reconstruction)
+__device__ __forceinline__ int64_t wave_incl_scan_i64(int64_t x, int lane) {
+  for (int off = 1; off < 64; off <<= 1) {
+    int64_t t = __shfl_up(x, off);
+    if (lane >= off) x += t;
+  }
+  return x;
+}
+
+__global__ __launch_bounds__(HIST_WAVES * 64, 4)
+void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
+                      const int32_t* __restrict__ series_first,
+                      const int32_t* __restrict__ series_nchunks,
+                      const int32_t* __restrict__ group_ids,
+                      int num_series,
+                      int64_t qstart, int64_t qstep, int64_t qend, int64_t qwindow,
+                      int num_windows, int nb,
+                      double* __restrict__ out_sums,   // [G × W × nb]
+                      double* __restrict__ out_cnt)    // [G × W]
+{
+  __shared__ HistWs ws_all[HIST_WAVES];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  HistWs& ws = ws_all[wave];
+
+  for (int sid = blockIdx.x * HIST_WAVES + wave; sid < num_series;
+       sid += gridDim.x * HIST_WAVES) {
+    if (series_nchunks[sid] != 1) continue;            // guarded at exec
+    const int ci = series_first[sid];
+    const int n = dir.num_rows[ci];
+    if (n <= 0 || n > FDB_MAX_ROWS_PER_SERIES) continue;
+
+    DVec tv;
+    d_vec_open(blob + dir.ts_off[ci], &tv);
+    d_decode_chunk<false>(tv, n, ws.ts, nullptr, lane);
+    for (int r = lane; r < FDB_HIST_RING; r += 64) ws.ring_w[r] = -1;
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_wave_barrier();
+
+    const uint8_t* hv = blob + dir.val_off[ci];
+    const int defsz = d_u16(hv + FDB_HIST_OFF_DEFSIZE);
+    const uint8_t* sp = hv + FDB_HIST_OFF_DEF + defsz;  // first section
+
+    const int b = lane;                                 // bucket owned by lane
+    const bool live = b < nb;
+    double raw = 0, C = 0, base = 0;
+    const int grp = group_ids[sid];
+
+    int sect_left = 0;        // elements left in current section
+    int sect_first = 0;       // next element is the section's base element
+    const uint8_t* ep = sp;   // element cursor
+
+    for (int e = 0; e < n; e++) {
+      if (sect_left == 0) {   // enter next section
+        int stype = sp[3];
+        sect_left = sp[2];
+        ep = sp + 4;
+        sp += 4 + d_u16(sp);  // advance to the section after this one
+        sect_first = 1;
+        if (stype == 1 && e > 0) C += raw;  // TypeDrop: corr += apply(e-1)
+      }
+      const int elen = d_u16(ep);
+      const uint8_t* s = ep + 2;
+      // parse the 8-value group headers serially (uniform across lanes)
+      int my_group = b >> 3;
+      int off = 0, gOff = 0, gBits = 0, gTrail = 0;
+      uint32_t gMask = 0;
+      for (int g = 0; g * 8 < nb; g++) {
+        uint32_t mask = s[off];
+        int numBits = 0, trail = 0, glen;
+        if (mask == 0) {
+          glen = 1;
+        } else {
+          int widths = s[off + 1];
+          numBits = ((widths >> 4) + 1) * 4;
+          trail = (widths & 0x0f) * 4;
+          glen = 2 + (numBits * __popc(mask) + 7) / 8;
+        }
+        if (g == my_group) { gOff = off; gBits = numBits; gTrail = trail; gMask = mask; }
+        off += glen;
+      }
+      // extract this lane's value from its group
+      int64_t delta = 0;
+      const int bit = b & 7;
+      if (live && (gMask & (1u << bit))) {
+        int slot = __popc(gMask & ((1u << bit) - 1));
+        int bitpos = slot * gBits;
+        const uint8_t* p = s + gOff + 2 + (bitpos >> 3);
+        uint64_t w64;
+        memcpy(&w64, p, 8);                  // unaligned-safe via memcpy
+        int sh = bitpos & 7;
+        uint64_t v = w64 >> sh;
+        if (gBits > 64 - sh) v |= (uint64_t)p[8] << (64 - sh);
+        uint64_t m = gBits >= 64 ? ~0ULL : ((1ULL << gBits) - 1);
+        delta = (int64_t)((v & m) << gTrail);
+      }
+      // reconstruct this element's cumulative bucket value
+      int64_t scan = wave_incl_scan_i64(live ? delta : 0, lane);
+      if (sect_first) { raw = (double)scan; base = raw; }
+      else            raw = base + (double)scan;
+      sect_first = 0;
+      sect_left--;
+      ep += 2 + elen;
+      const double corrected = raw + C;
+
+      // window triggers (inversion of the row-range search)
+      const int64_t ts_e = ws.ts[e];
+      const int64_t ts_prev = e > 0 ? ws.ts[e - 1] : INT64_MIN / 4;
+      const int64_t ts_next = e + 1 < n ? ws.ts[e + 1] : INT64_MAX / 4;
+      // windows starting at e: wStart in (ts_prev, ts_e]
+      int ws_lo = (int)(d_fdiv(ts_prev - qstart + qwindow, qstep) + 1);
+      int ws_hi = (int)d_fdiv(ts_e - qstart + qwindow, qstep);
+      if (ws_lo < 0) ws_lo = 0;
+      if (ws_hi > num_windows - 1) ws_hi = num_windows - 1;
+      for (int w = ws_lo; w <= ws_hi; w++) {
+        int slot = w % FDB_HIST_RING;
+        if (live) ws.ring_lo[slot][b] = corrected;
+        if (lane == 0) { ws.ring_w[slot] = w; ws.ring_e[slot] = e; }
+      }
+      // windows ending at e: wEnd in [ts_e, ts_next)
+      int we_lo = (int)d_fdiv(ts_e - qstart + qstep - 1, qstep);
+      int we_hi = e + 1 < n ? (int)(d_fdiv(ts_next - qstart + qstep - 1, qstep) - 1)
+                            : num_windows - 1;
+      if (we_lo < 0) we_lo = 0;
+      if (we_hi > num_windows - 1) we_hi = num_windows - 1;
+      if (ws_hi >= we_lo)   // a slot written this element may be read below
+        { __builtin_amdgcn_s_waitcnt(0); __builtin_amdgcn_wave_barrier(); }
+      for (int w = we_lo; w <= we_hi; w++) {
+        int slot = w % FDB_HIST_RING;
+        if (ws.ring_w[slot] != w) continue;     // window never started (empty)
+        int e0 = ws.ring_e[slot];
+        int64_t t1 = ws.ts[e0];
+        if (!(ts_e > t1)) continue;             // highestTime > lowestTime rule
+        int64_t wEnd = qstart + (int64_t)w * qstep;
+        int64_t wStart = wEnd - qwindow;
+        int numSamples = e - e0 + 1;
+        if (live) {
+          double r = d_extrapolated_rate(wStart, wEnd, numSamples,
+                                         t1, ws.ring_lo[slot][b],
+                                         ts_e, corrected, true, true);
+          atomicAdd(&out_sums[((size_t)grp * num_windows + w) * nb + b], r);
+        }
+        if (lane == 0)
+          atomicAdd(&out_cnt[(size_t)grp * num_windows + w], 1.0);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_wave_barrier();
+  }
+}
+
+// quantile present step (Histogram.quantile, Histogram.scala:63-108; geometric)
+__global__ void hist_quantile_kernel(const double* __restrict__ sums,
+                                     const double* __restrict__ cnt,
+                                     double* __restrict__ out,
+                                     size_t cells, int nb, double q,
+                                     double first, double mult) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= cells) return;
+  if (!(cnt[i] > 0)) { out[i] = NAN; return; }
+  const double* v = sums + i * nb;
+  double top = v[nb - 1];
+  if (q < 0) { out[i] = -INFINITY; return; }
+  if (q > 1) { out[i] = INFINITY; return; }
+  if (nb < 2 || !(top > 0)) { out[i] = NAN; return; }
+  double rank = q * top;
+  int bucket = 0;
+  while (v[bucket] < rank) bucket++;
+  double bucketStart = bucket == 0 ? 0 : first * pow(mult, bucket - 1);
+  double bucketEnd = first * pow(mult, bucket);
+  if (bucket == nb - 1 && isinf(bucketEnd)) { out[i] = first * pow(mult, nb - 2); return; }
+  if (bucket == 0 && first <= 0) { out[i] = first; return; }
+  double count = bucket == 0 ? v[0] : v[bucket] - v[bucket - 1];
+  rank -= bucket == 0 ? 0 : v[bucket - 1];
+  out[i] = bucketStart + (bucketEnd - bucketStart) * (rank / count);
+}
+
 // presentation fixup for aggregated grids (NaN where no contributions; mean for avg)
 __global__ void agg_present_kernel(double* out, const double* cnt, size_t n,
                                    int agg_id, int partial) {
@@ -1040,6 +1242,71 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   if (own_out) hipFree(dev_out);
   if (own_cnt) hipFree(dev_cnt);
   if (avg_ms) *avg_ms = iters > 0 ? (double)ms_sum / iters : 0.0;
+  return FDB_OK;
+}
+
+
+extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
+                                       const fdb_query_t* q, int32_t nb,
+                                       double* out_bucket_sums, double* out_counts,
+                                       double* out_quantile, int32_t out_on_device) {
+  HIP_CHECK(hipSetDevice(e->device));
+  int nw = fdb_num_windows(q);
+  if (nw <= 0 || q->num_groups <= 0) { fdb_set_error("bad hist query params"); return FDB_ERR_BADARG; }
+  if (nb < 1 || nb > 64) { fdb_set_error("num_buckets must be 1..64"); return FDB_ERR_BADARG; }
+  if (q->window / q->step + 2 > FDB_HIST_RING) {
+    fdb_set_error("window/step ratio %lld exceeds round-1 ring capacity %d",
+                  (long long)(q->window / q->step), FDB_HIST_RING);
+    return FDB_ERR_BADARG;
+  }
+  if (d->max_group >= q->num_groups) { fdb_set_error("num_groups too small"); return FDB_ERR_BADARG; }
+  size_t cells = (size_t)q->num_groups * nw;
+
+  double *dev_sums = nullptr, *dev_cnt = nullptr, *dev_quant = nullptr;
+  bool own_sums = true, own_cnt = true, own_quant = false;
+  if (out_on_device) {
+    dev_sums = out_bucket_sums; own_sums = dev_sums == nullptr;
+    dev_cnt = out_counts; own_cnt = dev_cnt == nullptr;
+    dev_quant = out_quantile;
+  }
+  if (!dev_sums) HIP_CHECK(hipMalloc(&dev_sums, cells * nb * 8));
+  if (!dev_cnt) HIP_CHECK(hipMalloc(&dev_cnt, cells * 8));
+  if (out_quantile && !dev_quant) { HIP_CHECK(hipMalloc(&dev_quant, cells * 8)); own_quant = true; }
+  HIP_CHECK(hipMemsetAsync(dev_sums, 0, cells * nb * 8, e->stream));
+  HIP_CHECK(hipMemsetAsync(dev_cnt, 0, cells * 8, e->stream));
+
+  DirSoA dir{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
+  int grid = (d->num_series + HIST_WAVES - 1) / HIST_WAVES;
+  if (grid > 8192) grid = 8192;
+  hipLaunchKernelGGL(hist_scan_kernel, dim3(grid), dim3(HIST_WAVES * 64), 0, e->stream,
+                     d->blob, dir, d->series_first, d->series_nchunks, d->group_ids,
+                     d->num_series, q->start, q->step, q->end, q->window, nw, nb,
+                     dev_sums, dev_cnt);
+  HIP_CHECK(hipGetLastError());
+
+  if (out_quantile) {
+    // bucket scheme (first, mult) from the first hist vector in the blob
+    uint64_t voff0;
+    HIP_CHECK(hipMemcpy(&voff0, d->val_off, 8, hipMemcpyDeviceToHost));
+    double fm[2];
+    HIP_CHECK(hipMemcpy(fm, d->blob + voff0 + FDB_HIST_OFF_DEF + 2, 16,
+                        hipMemcpyDeviceToHost));
+    hist_quantile_kernel<<<(unsigned)((cells + 255) / 256), 256, 0, e->stream>>>(
+        dev_sums, dev_cnt, dev_quant, cells, nb, q->param, fm[0], fm[1]);
+    HIP_CHECK(hipGetLastError());
+  }
+  HIP_CHECK(hipStreamSynchronize(e->stream));
+  if (!out_on_device) {
+    if (out_bucket_sums)
+      HIP_CHECK(hipMemcpy(out_bucket_sums, dev_sums, cells * nb * 8, hipMemcpyDeviceToHost));
+    if (out_counts)
+      HIP_CHECK(hipMemcpy(out_counts, dev_cnt, cells * 8, hipMemcpyDeviceToHost));
+    if (out_quantile)
+      HIP_CHECK(hipMemcpy(out_quantile, dev_quant, cells * 8, hipMemcpyDeviceToHost));
+  }
+  if (own_sums) hipFree(dev_sums);
+  if (own_cnt) hipFree(dev_cnt);
+  if (own_quant) hipFree(dev_quant);
   return FDB_OK;
 }
 

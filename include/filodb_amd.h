@@ -41,6 +41,8 @@ typedef struct fdb_store fdb_store_t;
 /* series value temporality / column kind */
 #define FDB_COL_GAUGE    0   /* plain double column                        */
 #define FDB_COL_COUNTER  1   /* drop-detecting counter column (DoubleCounterAppender) */
+#define FDB_COL_HIST     2   /* sect-delta histogram column (AppendableSectDeltaHistVector,
+                                core/.../vectors/HistogramVector.scala:491-545) */
 
 fdb_store_t* fdb_store_create(int64_t expected_series);
 void         fdb_store_destroy(fdb_store_t* s);
@@ -54,6 +56,20 @@ int32_t fdb_store_add_series(fdb_store_t* s, int32_t group_id, int32_t col_kind)
  * markers). Timestamps must be nondecreasing. */
 int32_t fdb_series_append(fdb_store_t* s, int32_t series_id,
                           const int64_t* ts, const double* vals, int32_t n);
+
+/* Appends histogram samples to a FDB_COL_HIST series. bucket_values is
+ * row-major [n × num_buckets] cumulative-LE counts (each bucket an increasing
+ * counter). The bucket scheme is geometric: top(i) = first * mult^i
+ * (GeometricBuckets, core/.../vectors/Histogram.scala:609-626); it must be
+ * identical across a series. Encoded per the reference's sect-delta format:
+ * sections of <=16 histograms, first element NibblePack-delta packed raw,
+ * later elements packed as bucket-delta diffs vs the section base, counter
+ * drops forcing a TypeDrop section (HistogramVector.scala:491-545,
+ * NibblePack.scala:304-350, Section.scala). */
+int32_t fdb_series_append_hist(fdb_store_t* s, int32_t series_id,
+                               const int64_t* ts, const uint64_t* bucket_values,
+                               int32_t n, int32_t num_buckets,
+                               double bucket_first, double bucket_mult);
 
 /* Forces a chunk boundary: encodes the write buffer into frozen vectors
  * (= switchBuffers(encode=true)). Called automatically when a buffer reaches
@@ -133,6 +149,8 @@ int32_t fdb_nibblepack_pack_doubles(const double* in, int32_t n, uint8_t* out, i
 #define FDB_FN_STDDEV_OVER_TIME 8
 #define FDB_FN_STDVAR_OVER_TIME 9
 #define FDB_FN_CHANGES         10
+#define FDB_FN_HIST_RATE       11   /* HistRateFunction (RateFunctions.scala:330-400):
+                                       per-bucket counter-corrected extrapolated rate */
 
 /* Cross-series aggregation: RowAggregator implementations
  * (query/.../exec/aggregator/RowAggregator.scala:28-150). */
@@ -153,6 +171,8 @@ typedef struct {
   int32_t agg_id;    /* FDB_AGG_* (AggregateMapReduce)                       */
   int32_t num_groups;/* required when agg_id != NONE                         */
   int32_t _pad;
+  double  param;     /* function parameter (quantile q for the histogram
+                        present step); 0 when unused                          */
 } fdb_query_t;
 
 static inline int32_t fdb_num_windows(const fdb_query_t* q) {
@@ -190,6 +210,23 @@ int64_t        fdb_dataset_samples(const fdb_dataset_t* d); /* total rows across
  * Synchronous: returns after the result is materialized. */
 int32_t fdb_query_exec(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
                        double* out, double* out_counts, int32_t out_on_device);
+
+/* Histogram pipeline for BASELINE config #4:
+ * histogram_quantile(param, sum(rate(hist[window])) by group)
+ * — HistRateFunction per series (per-bucket extrapolated rate with counter
+ * correction), HistSumRowAggregator across series per group
+ * (aggregator/HistSumRowAggregator.scala:20-29), Histogram.quantile present
+ * step (core/.../vectors/Histogram.scala:63-108).
+ * out_bucket_sums: [num_groups × windows × num_buckets] per-bucket rate sums
+ *                  (the cross-shard/RCCL merge payload); may be NULL.
+ * out_counts:      [num_groups × windows] contributing-series counts; may be NULL.
+ * out_quantile:    [num_groups × windows] quantile(param) of the summed rate
+ *                  histogram (NaN where no contributions); may be NULL.
+ * Round-1 engine limit: one chunk per series per query span. */
+int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
+                            const fdb_query_t* q, int32_t num_buckets,
+                            double* out_bucket_sums, double* out_counts,
+                            double* out_quantile, int32_t out_on_device);
 
 /* Timing variant for bench.py: runs the same launch `iters` times and returns the
  * average per-iteration kernel milliseconds measured with HIP events on the

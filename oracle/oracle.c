@@ -37,6 +37,7 @@
 typedef struct {
   int64_t start, step, end, window;
   int32_t func_id, agg_id, num_groups, _pad;
+  double param;
 } fdb_query_t;
 typedef struct {
   const uint8_t* blob;
@@ -724,4 +725,241 @@ EXPORT double oracle_extrapolated_rate(int64_t ws, int64_t we, int32_t n,
                                        int64_t t1, double v1, int64_t t2, double v2,
                                        int32_t isCounter, int32_t isRate) {
   return extrapolated_rate(ws, we, n, t1, v1, t2, v2, isCounter, isRate);
+}
+
+/* =========================================================================
+ * sect-delta histogram path (config #4)
+ *   vector format  HistogramVector.scala:237-254,491-545; Section.scala:17-24
+ *   reader         SectDeltaHistogramReader (HistogramVector.scala:628-737)
+ *   rate           HistogramRateFunctionBase (RateFunctions.scala:330-400)
+ *   cross-series   HistSumRowAggregator.scala:20-29
+ *   quantile       Histogram.quantile (Histogram.scala:63-108)
+ * ========================================================================= */
+typedef struct {
+  int n, nb;
+  double first, mult;
+  int64_t* cum;      /* [n × nb] raw cumulative bucket values               */
+  int64_t* corr;     /* [n × nb] cumulative in-chunk corrections at element */
+  int64_t* chunk_corr; /* [nb] total corrections (updateCorrection)         */
+} hist_chunk_t;
+
+/* NibblePack delta-decode nb longs from a stream (DeltaSink semantics) */
+static int np_unpack_delta(const uint8_t* in, int inlen, int64_t* out, int nb) {
+  int pos = 0, i = 0;
+  int64_t current = 0;
+  while (i < nb) {
+    int64_t grp[8]; int consumed;
+    if (oracle_nibblepack_unpack8(in + pos, inlen - pos, grp, &consumed) != 0) return -1;
+    pos += consumed;
+    for (int k = 0; k < 8 && i < nb; k++, i++) { current += grp[k]; out[i] = current; }
+  }
+  return 0;
+}
+
+/* decodes a sect-delta hist vector; returns 0 and fills hc (malloc'd) */
+static int hist_open(const uint8_t* p, hist_chunk_t* hc) {
+  if (rd_u16(p + 4) != FDB_WF_HIST_SECTDELTA) return -1;
+  int n = rd_u16(p + FDB_HIST_OFF_NUMHIST);
+  int nb = rd_u16(p + FDB_HIST_OFF_DEF);
+  hc->n = n; hc->nb = nb;
+  hc->first = rd_f64(p + FDB_HIST_OFF_DEF + 2);
+  hc->mult = rd_f64(p + FDB_HIST_OFF_DEF + 10);
+  hc->cum = (int64_t*)malloc((size_t)n * nb * 8);
+  hc->corr = (int64_t*)calloc((size_t)n * nb, 8);
+  hc->chunk_corr = (int64_t*)calloc((size_t)nb, 8);
+  const uint8_t* sp = p + FDB_HIST_OFF_DEF + rd_u16(p + FDB_HIST_OFF_DEFSIZE);
+  int e = 0;
+  int64_t* base = (int64_t*)malloc((size_t)nb * 8);
+  int64_t* runc = (int64_t*)calloc((size_t)nb, 8);
+  while (e < n) {
+    int sbytes = rd_u16(sp);
+    int selems = sp[2];
+    int stype = sp[3];
+    const uint8_t* ep = sp + 4;
+    /* TypeDrop section (not first element): correction += value before drop
+     * (SectDeltaHistogramReader.corrections, HistogramVector.scala:663-676) */
+    if (stype == 1 && e > 0)
+      for (int b = 0; b < nb; b++) runc[b] += hc->cum[(size_t)(e - 1) * nb + b];
+    for (int se = 0; se < selems && e < n; se++, e++) {
+      int elen = rd_u16(ep);
+      int64_t* row = hc->cum + (size_t)e * nb;
+      if (se == 0) {               /* section base: raw packDelta of cum values */
+        if (np_unpack_delta(ep + 2, elen, row, nb) != 0) return -1;
+        memcpy(base, row, (size_t)nb * 8);
+      } else {                     /* delta element: base + delta-decoded diffs */
+        if (np_unpack_delta(ep + 2, elen, row, nb) != 0) return -1;
+        for (int b = 0; b < nb; b++) row[b] += base[b];
+      }
+      memcpy(hc->corr + (size_t)e * nb, runc, (size_t)nb * 8);
+      ep += 2 + elen;
+    }
+    sp += 4 + sbytes;
+  }
+  memcpy(hc->chunk_corr, runc, (size_t)nb * 8);
+  free(base); free(runc);
+  return 0;
+}
+static void hist_close(hist_chunk_t* hc) { free(hc->cum); free(hc->corr); free(hc->chunk_corr); }
+
+/* Histogram.compare (Histogram.scala:204-214), equal schemes: top-down values */
+static int hist_less(const int64_t* a, const int64_t* b, int nb) {
+  for (int i = nb - 1; i >= 0; i--) {
+    if (a[i] != b[i]) return a[i] < b[i];
+  }
+  return 0;
+}
+
+/* Histogram.quantile (Histogram.scala:63-108), geometric buckets */
+EXPORT double oracle_hist_quantile(double q, const double* values, int nb,
+                                   double first, double mult) {
+  if (q < 0) return -INFINITY;
+  if (q > 1) return INFINITY;
+  double top = nb <= 0 ? NAN : values[nb - 1];
+  if (nb < 2 || !(top > 0)) return NAN;
+  double rank = q * top;
+  int bucket = 0;
+  while (values[bucket] < rank) bucket++;
+  double bucketStart = bucket == 0 ? 0 : first * pow(mult, bucket - 1);
+  double bucketEnd = first * pow(mult, bucket);
+  if (bucket == nb - 1 && isinf(bucketEnd)) return first * pow(mult, nb - 2);
+  if (bucket == 0 && first <= 0) return first;
+  double count = bucket == 0 ? values[0] : values[bucket] - values[bucket - 1];
+  rank -= bucket == 0 ? 0 : values[bucket - 1];
+  double fraction = rank / count;
+  return bucketStart + (bucketEnd - bucketStart) * fraction;
+}
+
+/* test helper: decode all cumulative bucket values of a hist vector */
+EXPORT int32_t oracle_hist_decode(const uint8_t* vec, int64_t* out, int32_t cap,
+                                  int32_t* out_n, int32_t* out_nb) {
+  hist_chunk_t hc;
+  if (hist_open(vec, &hc) != 0) return -1;
+  if (hc.n * hc.nb > cap) { hist_close(&hc); return -1; }
+  memcpy(out, hc.cum, (size_t)hc.n * hc.nb * 8);
+  *out_n = hc.n; *out_nb = hc.nb;
+  hist_close(&hc);
+  return 0;
+}
+EXPORT int32_t oracle_hist_corrections(const uint8_t* vec, int64_t* out, int32_t cap) {
+  hist_chunk_t hc;
+  if (hist_open(vec, &hc) != 0) return -1;
+  if (hc.n * hc.nb > cap) { hist_close(&hc); return -1; }
+  memcpy(out, hc.corr, (size_t)hc.n * hc.nb * 8);
+  hist_close(&hc);
+  return 0;
+}
+
+/* histogram_quantile(param, sum by(group)(rate(hist[window]))) — the full
+ * config-#4 pipeline. out_bucket_sums [G×W×nb] / out_counts [G×W] /
+ * out_quantile [G×W]; any may be NULL. */
+EXPORT int32_t oracle_query_exec_hist(const fdb_view_t* view, const fdb_query_t* q,
+                                      int32_t nb,
+                                      double* out_bucket_sums, double* out_counts,
+                                      double* out_quantile, int32_t nthreads) {
+  int nw = (int)((q->end - q->start) / q->step) + 1;
+  int ng = q->num_groups;
+  size_t cells = (size_t)ng * nw;
+  double* sums = out_bucket_sums ? out_bucket_sums
+                                 : (double*)malloc(cells * nb * sizeof(double));
+  double* cnts = out_counts ? out_counts : (double*)malloc(cells * sizeof(double));
+  for (size_t i = 0; i < cells * nb; i++) sums[i] = 0;
+  for (size_t i = 0; i < cells; i++) cnts[i] = 0;
+
+  (void)nthreads;  /* serial: hist oracle runs at test sizes */
+  int64_t* lastv = (int64_t*)malloc((size_t)nb * 8);
+  int64_t* corr  = (int64_t*)malloc((size_t)nb * 8);
+  double* lo = (double*)malloc((size_t)nb * 8);
+  double* hi = (double*)malloc((size_t)nb * 8);
+  for (int sid = 0; sid < view->num_series; sid++) {
+    int first = view->series_first[sid];
+    int nchunks = view->series_nchunks[sid];
+    if (nchunks > 64) nchunks = 64;
+    const fdb_dir_entry_t* dir = view->dir + first;
+    vec_t tsv[64];
+    hist_chunk_t hv[64];
+    for (int c = 0; c < nchunks; c++) {
+      vec_open(view->blob + dir[c].ts_off, &tsv[c]);
+      if (hist_open(view->blob + dir[c].val_off, &hv[c]) != 0) return -1;
+    }
+    int grp = view->group_ids[sid];
+    for (int w = 0; w < nw; w++) {
+      int64_t wEnd = q->start + (int64_t)w * q->step;
+      int64_t wStart = wEnd - q->window;
+      int meta_has = 0;
+      memset(corr, 0, (size_t)nb * 8);
+      int numSamples = 0;
+      int64_t lowestTime = INT64_MAX, highestTime = 0;
+      int have_lo = 0, have_hi = 0;
+      for (int c = 0; c < nchunks; c++) {
+        if (dir[c].end_time < wStart) continue;
+        const vec_t* tv = &tsv[c];
+        const hist_chunk_t* h = &hv[c];
+        int startRow = lv_binary_search(tv, wStart) & 0x7fffffff;
+        int endRow = lv_ceiling(tv, wEnd);
+        if (endRow > dir[c].num_rows - 1) endRow = dir[c].num_rows - 1;
+        if (meta_has) {              /* detectDropAndCorrection :654-663 */
+          if (hist_less(h->cum, lastv, nb))
+            for (int b = 0; b < nb; b++) corr[b] += lastv[b];
+        }
+        if (startRow <= endRow) {
+          int64_t st = lv_at(tv, startRow), en = lv_at(tv, endRow);
+          if (st < lowestTime || en > highestTime) {
+            numSamples += endRow - startRow + 1;
+            if (st < lowestTime) {
+              lowestTime = st;
+              for (int b = 0; b < nb; b++)
+                lo[b] = (double)(h->cum[(size_t)startRow * nb + b]
+                                 + h->corr[(size_t)startRow * nb + b] + corr[b]);
+              have_lo = 1;
+            }
+            if (en > highestTime) {
+              highestTime = en;
+              for (int b = 0; b < nb; b++)
+                hi[b] = (double)(h->cum[(size_t)endRow * nb + b]
+                                 + h->corr[(size_t)endRow * nb + b] + corr[b]);
+              have_hi = 1;
+            }
+          }
+        }
+        /* updateCorrection :699-711: lastValue = apply(len-1), the RAW value */
+        for (int b = 0; b < nb; b++) {
+          corr[b] += h->chunk_corr[b];
+          lastv[b] = h->cum[(size_t)(h->n - 1) * nb + b];
+        }
+        meta_has = 1;
+        if (dir[c].end_time >= wEnd) break;
+      }
+      if (highestTime > lowestTime && have_lo && have_hi) {
+        size_t cell = (size_t)grp * nw + w;
+        for (int b = 0; b < nb; b++) {
+          double r = extrapolated_rate(wStart, wEnd, numSamples,
+                                       lowestTime, lo[b], highestTime, hi[b], 1, 1);
+          sums[cell * nb + b] += r;   /* HistSumRowAggregator bucket-wise add */
+        }
+        cnts[cell] += 1;
+      }
+    }
+    for (int c = 0; c < nchunks; c++) hist_close(&hv[c]);
+  }
+  free(lastv); free(corr); free(lo); free(hi);
+
+  if (out_quantile) {
+    double first = 0, mult = 0;
+    /* bucket scheme from the first hist chunk */
+    for (int sid = 0; sid < view->num_series && first == 0; sid++) {
+      const fdb_dir_entry_t* dir = view->dir + view->series_first[sid];
+      hist_chunk_t hc;
+      if (hist_open(view->blob + dir[0].val_off, &hc) == 0) {
+        first = hc.first; mult = hc.mult;
+        hist_close(&hc);
+      }
+    }
+    for (size_t i = 0; i < cells; i++) {
+      out_quantile[i] = cnts[i] > 0
+        ? oracle_hist_quantile(q->param, sums + i * nb, nb, first, mult) : NAN;
+    }
+  }
+  if (!out_bucket_sums) free(sums);
+  if (!out_counts) free(cnts);
+  return 0;
 }

@@ -44,6 +44,52 @@ def lib():
     return _lib
 
 
+def query_exec_hist(view, q, num_buckets, with_quantile=True):
+    import ctypes as ct
+    nw = q.num_windows
+    ng = q.num_groups
+    sums = np.zeros(ng * nw * num_buckets, dtype=np.float64)
+    cnts = np.zeros(ng * nw, dtype=np.float64)
+    quant = np.zeros(ng * nw, dtype=np.float64) if with_quantile else None
+    L = lib()
+    L.oracle_query_exec_hist.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_int32,
+                                         _c_double_p, _c_double_p, _c_double_p,
+                                         ct.c_int32]
+    rc = L.oracle_query_exec_hist(
+        ct.byref(view), ct.byref(q), num_buckets,
+        sums.ctypes.data_as(_c_double_p), cnts.ctypes.data_as(_c_double_p),
+        quant.ctypes.data_as(_c_double_p) if quant is not None else None, 1)
+    if rc != 0:
+        raise RuntimeError("oracle_query_exec_hist failed")
+    return sums, cnts, quant
+
+
+def hist_decode(vec_bytes, cap=200000):
+    import ctypes as ct
+    out = np.empty(cap, dtype=np.int64)
+    n = ct.c_int32(); nb = ct.c_int32()
+    L = lib()
+    L.oracle_hist_decode.argtypes = [ct.POINTER(ct.c_uint8), ct.POINTER(ct.c_int64),
+                                     ct.c_int32, ct.POINTER(ct.c_int32),
+                                     ct.POINTER(ct.c_int32)]
+    rc = L.oracle_hist_decode(_u8(vec_bytes),
+                              out.ctypes.data_as(ct.POINTER(ct.c_int64)), cap,
+                              ct.byref(n), ct.byref(nb))
+    if rc != 0:
+        raise RuntimeError("hist_decode failed")
+    return out[:n.value * nb.value].reshape(n.value, nb.value).copy()
+
+
+def hist_quantile(q, values, first, mult):
+    import ctypes as ct
+    L = lib()
+    L.oracle_hist_quantile.restype = ct.c_double
+    L.oracle_hist_quantile.argtypes = [ct.c_double, _c_double_p, ct.c_int32,
+                                       ct.c_double, ct.c_double]
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    return L.oracle_hist_quantile(q, v.ctypes.data_as(_c_double_p), len(v), first, mult)
+
+
 def query_exec(view, q, num_series, num_windows, out_counts=False, nthreads=1):
     """Runs the oracle over a store view (filodb_amd.View) with query q
     (filodb_amd.Query). Returns the result grid as numpy."""

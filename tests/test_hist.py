@@ -1,0 +1,176 @@
+"""Histogram path (BASELINE config #4): builder ↔ oracle parity.
+
+Golden values from the reference's own tests:
+core/src/test/scala/filodb.memory/format/vectors/HistogramTest.scala:5-45,109-118
+(GeometricBuckets(1,2,8) fixtures and quantile(0.50) known answers).
+Sect-delta encode semantics: HistogramVector.scala:491-545; NibblePack.scala:304-350.
+"""
+import numpy as np
+import pytest
+
+# HistogramTest.scala:8-13 — raw bucket arrays; :44 — quantile(0.5) answers
+RAW_HIST_BUCKETS = [
+    [10, 15, 17, 20, 25, 34, 76, 82],
+    [6, 16, 26, 26, 36, 38, 56, 59],
+    [11, 16, 26, 27, 33, 42, 46, 55],
+    [4, 4, 5, 33, 35, 67, 91, 121],
+]
+QUANTILE50 = [37.333333333333336, 10.8, 8.666666666666666, 28.75]
+
+
+def test_quantile_golden(oracle):
+    # GeometricBuckets(1.0, 2.0, 8)
+    for values, expected in zip(RAW_HIST_BUCKETS, QUANTILE50):
+        got = oracle.hist_quantile(0.5, np.array(values, dtype=np.float64), 1.0, 2.0)
+        assert got == pytest.approx(expected, rel=1e-12)
+
+
+def make_hist_store(fdb, series_list, nb=8, max_rows=400, groups=None):
+    st = fdb.ChunkStore()
+    st.set_max_rows(max_rows)
+    for i, (ts, bv) in enumerate(series_list):
+        gid = groups[i] if groups else 0
+        sid = st.add_series(gid, fdb.COL_HIST)
+        st.append_hist(sid, ts, bv, first=2.0, mult=2.0)
+    st.seal()
+    return st
+
+
+def synth_hist(rng, n, nb=8, reset_p=0.0, start_ts=100000, step=15000):
+    """Valid cumulative-LE histograms: per-interval counts are increasing
+    counters; resets zero everything."""
+    ts = start_ts + np.arange(n) * step + rng.integers(-250, 251, n)
+    ts = np.maximum.accumulate(ts).astype(np.int64)
+    intervals = np.zeros((n, nb), dtype=np.uint64)
+    cur = np.zeros(nb, dtype=np.uint64)
+    for i in range(n):
+        if reset_p and rng.random() < reset_p:
+            cur = np.zeros(nb, dtype=np.uint64)
+        cur = cur + rng.poisson(2.0, nb).astype(np.uint64)
+        intervals[i] = cur
+    cum = np.cumsum(intervals, axis=1)  # cumulative-LE across buckets
+    return ts, cum
+
+
+def test_hist_encode_decode_roundtrip(fdb, oracle):
+    rng = np.random.default_rng(5)
+    for reset_p in (0.0, 0.05):
+        ts, cum = synth_hist(rng, 50, reset_p=reset_p)
+        st = make_hist_store(fdb, [(ts, cum)])
+        _, vab, n, _, _ = st.chunk(0, 0)
+        dec = oracle.hist_decode(vab)
+        assert dec.shape == (50, 8)
+        np.testing.assert_array_equal(dec, cum.astype(np.int64))
+
+
+def test_hist_sections_and_length(fdb, oracle):
+    # 40 elements, no drops → sections of 16: raw elements at 0, 16, 32
+    rng = np.random.default_rng(9)
+    ts, cum = synth_hist(rng, 40)
+    st = make_hist_store(fdb, [(ts, cum)])
+    _, vab, _, _, _ = st.chunk(0, 0)
+    import struct
+    wf = struct.unpack("<H", vab[4:6])[0]
+    assert wf == 0x1209
+    assert struct.unpack("<H", vab[6:8])[0] == 40  # numHistograms
+    assert vab[8] == 0x03
+    nb = struct.unpack("<H", vab[11:13])[0]
+    assert nb == 8
+    first, mult = struct.unpack("<dd", vab[13:29])
+    assert (first, mult) == (2.0, 2.0)
+
+
+def naive_hist_rate(oracle, ts, cum, wstart, wend, nsamples=None):
+    m = (ts >= wstart) & (ts <= wend)
+    if m.sum() < 2:
+        return None
+    idx = np.nonzero(m)[0]
+    t1, t2 = int(ts[idx[0]]), int(ts[idx[-1]])
+    return [oracle.extrapolated_rate(int(wstart), int(wend), int(m.sum()),
+                                     t1, float(cum[idx[0], b]),
+                                     t2, float(cum[idx[-1], b]), True, True)
+            for b in range(cum.shape[1])]
+
+
+def test_hist_rate_vs_naive(fdb, oracle):
+    """sum(rate(hist[5m])) with a single clean series == naive per-bucket
+    extrapolated rate (no resets → corrections zero)."""
+    rng = np.random.default_rng(21)
+    ts, cum = synth_hist(rng, 100)
+    st = make_hist_store(fdb, [(ts, cum)])
+    q_start = int(ts[30])
+    q = __import__("filodb_amd").make_query(q_start, 15000, q_start + 50 * 15000,
+                                            300000, 11, 1, 1, param=0.99)
+    # naive needs the ENCODED timestamps (approx-const may apply)
+    tsb, _, _, _, _ = st.chunk(0, 0)
+    enc_ts = oracle.decode_longs(tsb)
+    sums, cnts, quant = oracle.query_exec_hist(st.view(), q, 8)
+    nw = q.num_windows
+    for w in range(nw):
+        wend = q_start + w * 15000
+        expected = naive_hist_rate(oracle, enc_ts, cum, wend - 300000, wend)
+        cell = sums[w * 8:(w + 1) * 8]
+        if expected is None:
+            assert cnts[w] == 0
+            assert np.isnan(quant[w])
+        else:
+            np.testing.assert_allclose(cell, expected, rtol=1e-9)
+            assert quant[w] == pytest.approx(
+                oracle.hist_quantile(0.99, np.array(expected), 2.0, 2.0), rel=1e-12)
+
+
+def test_hist_rate_with_reset(fdb, oracle):
+    """A counter reset mid-chunk starts a TypeDrop section; the rate must use
+    the corrected (reset-compensated) values (SectDeltaHistogramReader
+    corrections, HistogramVector.scala:663-737)."""
+    nb = 4
+    ts = (100000 + np.arange(10) * 10000).astype(np.int64)
+    inc = np.full((10, nb), 5, dtype=np.uint64)
+    cum_int = np.cumsum(inc, axis=0)         # per-interval cumulative over time
+    cum_int[6:] = cum_int[6:] - cum_int[5]   # reset after element 5
+    cum = np.cumsum(cum_int, axis=1)         # cumulative-LE across buckets
+    st = make_hist_store(fdb, [(ts, cum.astype(np.uint64))], nb=nb)
+    _, vab, _, _, _ = st.chunk(0, 0)
+    dec = oracle.hist_decode(vab)
+    np.testing.assert_array_equal(dec, cum)
+    # one window over everything: corrected last = last + value-before-reset
+    end = int(ts[-1]) + 50
+    q = __import__("filodb_amd").make_query(end, 10000, end, end - int(ts[0]) + 50,
+                                            11, 1, 1, param=0.5)
+    sums, cnts, _ = oracle.query_exec_hist(st.view(), q, nb)
+    tsb, _, _, _, _ = st.chunk(0, 0)
+    enc_ts = oracle.decode_longs(tsb)
+    corrected_last = cum[-1] + cum[5]        # correction adds apply(drop-1)
+    expected = [oracle.extrapolated_rate(end - (end - int(ts[0]) + 50), end, 10,
+                                         int(enc_ts[0]), float(cum[0][b]),
+                                         int(enc_ts[-1]), float(corrected_last[b]),
+                                         True, True) for b in range(nb)]
+    np.testing.assert_allclose(sums[:nb], expected, rtol=1e-9)
+
+
+def test_hist_group_sum(fdb, oracle):
+    """Multiple series sum bucket-wise per group (HistSumRowAggregator)."""
+    rng = np.random.default_rng(33)
+    series = [synth_hist(rng, 60) for _ in range(6)]
+    st = make_hist_store(fdb, series, groups=[0, 1, 0, 1, 0, 1])
+    q_start = int(series[0][0][20])
+    import filodb_amd as f
+    q = f.make_query(q_start, 15000, q_start + 30 * 15000, 300000, 11, 1, 2,
+                     param=0.9)
+    sums, cnts, quant = oracle.query_exec_hist(st.view(), q, 8)
+    nw = q.num_windows
+    # group sums == sum of single-series runs
+    singles = []
+    for i, (ts, cum) in enumerate(series):
+        st1 = make_hist_store(fdb, [(ts, cum)])
+        s1, c1, _ = oracle.query_exec_hist(st1.view(), q, 8)
+        singles.append((i % 2, s1, c1))
+    for g in range(2):
+        exp = np.zeros(nw * 8)
+        expc = np.zeros(nw)
+        for gg, s1, c1 in singles:
+            if gg == g:
+                exp += s1[:nw * 8]
+                expc += c1[:nw]
+        np.testing.assert_allclose(sums[g * nw * 8:(g + 1) * nw * 8], exp, rtol=1e-9)
+        np.testing.assert_array_equal(cnts[g * nw:(g + 1) * nw], expc)
