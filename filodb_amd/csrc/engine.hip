@@ -932,21 +932,24 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       const int64_t ts_prev = e > 0 ? ws.ts[e - 1] : INT64_MIN / 4;
       const int64_t ts_next = e + 1 < n ? ws.ts[e + 1] : INT64_MAX / 4;
       // windows starting at e: wStart in (ts_prev, ts_e]
-      int ws_lo = (int)(d_fdiv(ts_prev - qstart + qwindow, qstep) + 1);
-      int ws_hi = (int)d_fdiv(ts_e - qstart + qwindow, qstep);
-      if (ws_lo < 0) ws_lo = 0;
-      if (ws_hi > num_windows - 1) ws_hi = num_windows - 1;
+      // (clamp in i64 BEFORE narrowing: the e=0 sentinel is a huge negative)
+      int64_t ws_lo64 = d_fdiv(ts_prev - qstart + qwindow, qstep) + 1;
+      int64_t ws_hi64 = d_fdiv(ts_e - qstart + qwindow, qstep);
+      if (ws_lo64 < 0) ws_lo64 = 0;
+      if (ws_hi64 > num_windows - 1) ws_hi64 = num_windows - 1;
+      int ws_lo = (int)ws_lo64, ws_hi = (int)ws_hi64;
       for (int w = ws_lo; w <= ws_hi; w++) {
         int slot = w % FDB_HIST_RING;
         if (live) ws.ring_lo[slot][b] = corrected;
         if (lane == 0) { ws.ring_w[slot] = w; ws.ring_e[slot] = e; }
       }
       // windows ending at e: wEnd in [ts_e, ts_next)
-      int we_lo = (int)d_fdiv(ts_e - qstart + qstep - 1, qstep);
-      int we_hi = e + 1 < n ? (int)(d_fdiv(ts_next - qstart + qstep - 1, qstep) - 1)
-                            : num_windows - 1;
-      if (we_lo < 0) we_lo = 0;
-      if (we_hi > num_windows - 1) we_hi = num_windows - 1;
+      int64_t we_lo64 = d_fdiv(ts_e - qstart + qstep - 1, qstep);
+      int64_t we_hi64 = e + 1 < n ? d_fdiv(ts_next - qstart + qstep - 1, qstep) - 1
+                                  : num_windows - 1;
+      if (we_lo64 < 0) we_lo64 = 0;
+      if (we_hi64 > num_windows - 1) we_hi64 = num_windows - 1;
+      int we_lo = (int)we_lo64, we_hi = (int)we_hi64;
       if (ws_hi >= we_lo)   // a slot written this element may be read below
         { __builtin_amdgcn_s_waitcnt(0); __builtin_amdgcn_wave_barrier(); }
       for (int w = we_lo; w <= we_hi; w++) {
