@@ -169,10 +169,17 @@ struct Ws {                         // per-wave LDS workspace
                ? FDB_MAX_ROWS_PER_SERIES : 1];
   double  sq[KIND == K_PFX_SQ ? FDB_MAX_ROWS_PER_SERIES : 1];   // squared prefix
   double  grp[KIND == K_MINMAX ? (FDB_NGROUPS + FDB_MAX_CHUNKS_PER_SERIES) : 1]; // 8-elem group min/max (chunk-relative, padded per chunk)
+  // single-chunk fast path: winSE[w] = startRow | (endRow+1)<<16, built by
+  // inversion during the meta phase (no per-window searches)
+  uint32_t winSE[256];
   ChunkMeta cm[FDB_MAX_CHUNKS_PER_SERIES];
   int32_t nchunks;
   int32_t total_rows;
 };
+
+__device__ __forceinline__ int64_t d_fdiv_s(int64_t a, int64_t b) {
+  return a >= 0 ? a / b : -((-a + b - 1) / b);
+}
 
 // wave-wide inclusive prefix sum (64 lanes)
 __device__ __forceinline__ double wave_incl_scan(double x, int lane) {
@@ -553,6 +560,36 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       }
     }
   }
+  // single-chunk fast path: invert rows -> per-window row ranges, so the
+  // window phase does one LDS read instead of two interpolation searches.
+  // Trigger math mirrors the searches exactly: startRow = first ts >= wStart,
+  // endRow = last ts <= wEnd (RangeFunction.scala:142-143 semantics).
+  const bool use_table = (nchunks == 1) && (num_windows <= 256) && ws.cm[0].nrows > 0;
+  if (use_table) {
+    const int n = ws.cm[0].nrows;
+    for (int w = lane; w < num_windows; w += 64)
+      ws.winSE[w] = (uint32_t)n;               // empty: startRow=n, endRow+1=0
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_wave_barrier();
+    for (int i = lane; i < n; i += 64) {
+      int64_t ts_i = ws.ts[i];
+      int64_t ts_prev = i > 0 ? ws.ts[i - 1] : INT64_MIN / 4;
+      int64_t ts_next = i + 1 < n ? ws.ts[i + 1] : INT64_MAX / 4;
+      int64_t lo = d_fdiv_s(ts_prev - qstart + qwindow, qstep) + 1;
+      int64_t hi = d_fdiv_s(ts_i - qstart + qwindow, qstep);
+      if (lo < 0) lo = 0;
+      if (hi > num_windows - 1) hi = num_windows - 1;
+      for (int64_t w = lo; w <= hi; w++)       // rows starting window w
+        ws.winSE[w] = (ws.winSE[w] & 0xffff0000u) | (uint32_t)i;
+      lo = d_fdiv_s(ts_i - qstart + qstep - 1, qstep);
+      hi = i + 1 < n ? d_fdiv_s(ts_next - qstart + qstep - 1, qstep) - 1
+                     : num_windows - 1;
+      if (lo < 0) lo = 0;
+      if (hi > num_windows - 1) hi = num_windows - 1;
+      for (int64_t w = lo; w <= hi; w++)       // rows ending window w
+        ws.winSE[w] = (ws.winSE[w] & 0x0000ffffu) | ((uint32_t)(i + 1) << 16);
+    }
+  }
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
 
@@ -574,8 +611,15 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       const int64_t wStart = wEnd - qwindow;
       double result = NAN;
       if (mr.end_time >= wStart) {
-        int startRow = lds_search_ge(seg, mr.nrows, wStart, mr.ts0, mr.inv_slope);
-        int endRow = lds_search_le(seg, mr.nrows, wEnd, mr.ts0, mr.inv_slope);
+        int startRow, endRow;
+        if (use_table) {
+          uint32_t se = ws.winSE[w];
+          startRow = (int)(se & 0xffff);
+          endRow = (int)(se >> 16) - 1;
+        } else {
+          startRow = lds_search_ge(seg, mr.nrows, wStart, mr.ts0, mr.inv_slope);
+          endRow = lds_search_le(seg, mr.nrows, wEnd, mr.ts0, mr.inv_slope);
+        }
         if (startRow <= endRow && endRow < mr.nrows) {
           if constexpr (RATE_FAMILY) {
             constexpr bool isCounter = (FUNC != FN_DELTA);
