@@ -169,26 +169,10 @@ struct Ws {                         // per-wave LDS workspace
                ? FDB_MAX_ROWS_PER_SERIES : 1];
   double  sq[KIND == K_PFX_SQ ? FDB_MAX_ROWS_PER_SERIES : 1];   // squared prefix
   double  grp[KIND == K_MINMAX ? (FDB_NGROUPS + FDB_MAX_CHUNKS_PER_SERIES) : 1]; // 8-elem group min/max (chunk-relative, padded per chunk)
-  // single-chunk fast path: winSE[w] = startRow | (endRow+1)<<16, built by
-  // inversion during the meta phase (no per-window searches)
-  uint32_t winSE[256];
   ChunkMeta cm[FDB_MAX_CHUNKS_PER_SERIES];
   int32_t nchunks;
   int32_t total_rows;
 };
-
-__device__ __forceinline__ int64_t d_fdiv_s(int64_t a, int64_t b) {
-  return a >= 0 ? a / b : -((-a + b - 1) / b);
-}
-
-// floor(a / b) via double reciprocal + exact integer fixup — i64 division is
-// software-emulated (hundreds of cycles); this is 2 multiplies + a short walk
-__device__ __forceinline__ int64_t d_fdiv_fast(int64_t a, int64_t b, double inv_b) {
-  int64_t w = (int64_t)floor((double)a * inv_b);
-  while ((w + 1) * b <= a) w++;
-  while (w * b > a) w--;
-  return w;
-}
 
 // wave-wide inclusive prefix sum (64 lanes)
 __device__ __forceinline__ double wave_incl_scan(double x, int lane) {
@@ -569,38 +553,6 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       }
     }
   }
-  // single-chunk fast path: invert rows -> per-window row ranges, so the
-  // window phase does one LDS read instead of two interpolation searches.
-  // Trigger math mirrors the searches exactly: startRow = first ts >= wStart,
-  // endRow = last ts <= wEnd (RangeFunction.scala:142-143 semantics).
-  const bool use_table = (nchunks == 1) && (num_windows <= 256) && ws.cm[0].nrows > 0;
-  if (use_table) {
-    const int n = ws.cm[0].nrows;
-    const double inv_step = 1.0 / (double)qstep;
-    for (int w = lane; w < num_windows; w += 64)
-      ws.winSE[w] = (uint32_t)n;               // empty: startRow=n, endRow+1=0
-    __builtin_amdgcn_s_waitcnt(0);
-    __builtin_amdgcn_wave_barrier();
-    for (int i = lane; i < n; i += 64) {
-      int64_t ts_i = ws.ts[i];
-      // sentinels kept small so the double conversion stays exact-enough
-      int64_t ts_prev = i > 0 ? ws.ts[i - 1] : ts_i - ((int64_t)1 << 40);
-      int64_t ts_next = i + 1 < n ? ws.ts[i + 1] : ts_i + ((int64_t)1 << 40);
-      int64_t lo = d_fdiv_fast(ts_prev - qstart + qwindow, qstep, inv_step) + 1;
-      int64_t hi = d_fdiv_fast(ts_i - qstart + qwindow, qstep, inv_step);
-      if (lo < 0) lo = 0;
-      if (hi > num_windows - 1) hi = num_windows - 1;
-      for (int64_t w = lo; w <= hi; w++)       // rows starting window w
-        ws.winSE[w] = (ws.winSE[w] & 0xffff0000u) | (uint32_t)i;
-      lo = d_fdiv_fast(ts_i - qstart + qstep - 1, qstep, inv_step);
-      hi = i + 1 < n ? d_fdiv_fast(ts_next - qstart + qstep - 1, qstep, inv_step) - 1
-                     : num_windows - 1;
-      if (lo < 0) lo = 0;
-      if (hi > num_windows - 1) hi = num_windows - 1;
-      for (int64_t w = lo; w <= hi; w++)       // rows ending window w
-        ws.winSE[w] = (ws.winSE[w] & 0x0000ffffu) | ((uint32_t)(i + 1) << 16);
-    }
-  }
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
 
@@ -622,15 +574,8 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       const int64_t wStart = wEnd - qwindow;
       double result = NAN;
       if (mr.end_time >= wStart) {
-        int startRow, endRow;
-        if (use_table) {
-          uint32_t se = ws.winSE[w];
-          startRow = (int)(se & 0xffff);
-          endRow = (int)(se >> 16) - 1;
-        } else {
-          startRow = lds_search_ge(seg, mr.nrows, wStart, mr.ts0, mr.inv_slope);
-          endRow = lds_search_le(seg, mr.nrows, wEnd, mr.ts0, mr.inv_slope);
-        }
+        int startRow = lds_search_ge(seg, mr.nrows, wStart, mr.ts0, mr.inv_slope);
+        int endRow = lds_search_le(seg, mr.nrows, wEnd, mr.ts0, mr.inv_slope);
         if (startRow <= endRow && endRow < mr.nrows) {
           if constexpr (RATE_FAMILY) {
             constexpr bool isCounter = (FUNC != FN_DELTA);
@@ -928,7 +873,6 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     int sect_left = 0;        // elements left in current section
     int sect_first = 0;       // next element is the section's base element
     const uint8_t* ep = sp;   // element cursor
-    const double inv_step = 1.0 / (double)qstep;
 
     for (int e = 0; e < n; e++) {
       if (sect_left == 0) {   // enter next section
@@ -985,12 +929,12 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 
       // window triggers (inversion of the row-range search)
       const int64_t ts_e = ws.ts[e];
-      const int64_t ts_prev = e > 0 ? ws.ts[e - 1] : ts_e - ((int64_t)1 << 40);
-      const int64_t ts_next = e + 1 < n ? ws.ts[e + 1] : ts_e + ((int64_t)1 << 40);
+      const int64_t ts_prev = e > 0 ? ws.ts[e - 1] : INT64_MIN / 4;
+      const int64_t ts_next = e + 1 < n ? ws.ts[e + 1] : INT64_MAX / 4;
       // windows starting at e: wStart in (ts_prev, ts_e]
       // (clamp in i64 BEFORE narrowing: the e=0 sentinel is a huge negative)
-      int64_t ws_lo64 = d_fdiv_fast(ts_prev - qstart + qwindow, qstep, inv_step) + 1;
-      int64_t ws_hi64 = d_fdiv_fast(ts_e - qstart + qwindow, qstep, inv_step);
+      int64_t ws_lo64 = d_fdiv(ts_prev - qstart + qwindow, qstep) + 1;
+      int64_t ws_hi64 = d_fdiv(ts_e - qstart + qwindow, qstep);
       if (ws_lo64 < 0) ws_lo64 = 0;
       if (ws_hi64 > num_windows - 1) ws_hi64 = num_windows - 1;
       int ws_lo = (int)ws_lo64, ws_hi = (int)ws_hi64;
@@ -1000,10 +944,9 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         if (lane == 0) { ws.ring_w[slot] = w; ws.ring_e[slot] = e; }
       }
       // windows ending at e: wEnd in [ts_e, ts_next)
-      int64_t we_lo64 = d_fdiv_fast(ts_e - qstart + qstep - 1, qstep, inv_step);
-      int64_t we_hi64 = e + 1 < n
-          ? d_fdiv_fast(ts_next - qstart + qstep - 1, qstep, inv_step) - 1
-          : num_windows - 1;
+      int64_t we_lo64 = d_fdiv(ts_e - qstart + qstep - 1, qstep);
+      int64_t we_hi64 = e + 1 < n ? d_fdiv(ts_next - qstart + qstep - 1, qstep) - 1
+                                  : num_windows - 1;
       if (we_lo64 < 0) we_lo64 = 0;
       if (we_hi64 > num_windows - 1) we_hi64 = num_windows - 1;
       int we_lo = (int)we_lo64, we_hi = (int)we_hi64;
