@@ -360,7 +360,7 @@ template <int FUNC>
 // rate family carries more live state (correction meta): forcing 5 waves/SIMD
 // makes it spill; the gauge kinds fit 5 waves cleanly
 __global__ __launch_bounds__(BLOCK_THREADS, (FUNC <= FN_DELTA) ? 4 : 5)
-void scan_kernel(const uint8_t* __restrict__ blob, int64_t blob_len, DirSoA dir,
+void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                  const int32_t* __restrict__ series_first,
                  const int32_t* __restrict__ series_nchunks,
                  const int32_t* __restrict__ group_ids,
@@ -556,24 +556,6 @@ void scan_kernel(const uint8_t* __restrict__ blob, int64_t blob_len, DirSoA dir,
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
 
-  // ---- cross-series prefetch: touch the NEXT series' chunk bytes now so its
-  // decode hits L2 (~200cyc) instead of cold HBM (~900cyc); the consuming use
-  // sits after the window phase, so the fills overlap the window math ---------
-  int64_t pf_acc = 0;
-  {
-    int nsid = sid + (int)gridDim.x * WAVES_PER_BLOCK;
-    if (nsid < num_series) {
-      int nf = series_first[nsid];
-      int64_t to = (int64_t)dir.ts_off[nf], vo = (int64_t)dir.val_off[nf];
-      if (lane < 8) {
-        int64_t ta = to + (int64_t)lane * 128;
-        int64_t va = vo + (int64_t)lane * 128;
-        if (ta + 8 <= blob_len) pf_acc += d_i64(blob + ta);
-        if (va + 8 <= blob_len) pf_acc += d_i64(blob + va);
-      }
-    }
-  }
-
   // ---- window phase: lanes split the windows -------------------------------
   if (!(phase_mask & 2)) {            // decode-only ablation: publish a checksum
     if (lane == 0 && ws.total_rows > 0)
@@ -659,7 +641,6 @@ void scan_kernel(const uint8_t* __restrict__ blob, int64_t blob_len, DirSoA dir,
         }
       }
     }
-    if (pf_acc == INT64_MIN + 1) out[0] = 0;   // keep prefetch loads alive
     __builtin_amdgcn_s_waitcnt(0);
     __builtin_amdgcn_wave_barrier();
     continue;
@@ -810,8 +791,6 @@ void scan_kernel(const uint8_t* __restrict__ blob, int64_t blob_len, DirSoA dir,
       }
     }
   }
-  // consume the prefetch accumulator (never true; keeps the loads alive)
-  if (pf_acc == INT64_MIN + 1) out[0] = 0;
   // next series reuses this wave's LDS slot: wave-local ordering is enough
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
@@ -1048,7 +1027,6 @@ struct fdb_engine {
 
 struct fdb_dataset {
   uint8_t* blob;
-  int64_t blob_len;
   uint64_t *ts_off, *val_off;
   int64_t *start_time, *end_time;
   int32_t *num_rows;
@@ -1142,7 +1120,6 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
   auto* d = new fdb_dataset();
   memset(d, 0, sizeof(*d));
   d->num_series = view.num_series;
-  d->blob_len = view.blob_len;
   d->num_chunks = nc;
   d->payload_bytes = payload;
   d->total_samples = samples;
@@ -1182,7 +1159,7 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
   if (cap > 0 && grid > cap) grid = cap;
   #define CASE(F) case F: \
     hipLaunchKernelGGL(scan_kernel<F>, dim3(grid), dim3(BLOCK_THREADS), 0, e->stream, \
-      d->blob, d->blob_len, dir, d->series_first, d->series_nchunks, d->group_ids, d->num_series, \
+      d->blob, dir, d->series_first, d->series_nchunks, d->group_ids, d->num_series, \
       q->start, q->step, q->end, q->window, nw, q->agg_id, dev_out, dev_cnt, \
       q->_pad == 0 ? 3 : q->_pad); break
   switch (q->func_id) {
