@@ -44,7 +44,7 @@ void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
 #define BLOCK_THREADS (WAVES_PER_BLOCK * 64)
 
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
-       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10 };
+       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12 };
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5 };
 
 // ---------------------------------------------------------------------------
@@ -146,12 +146,14 @@ struct ChunkMeta {
 #define K_PFX_SQ  2   // + squared prefix (stddev/stdvar)
 #define K_MINMAX  3   // ts, val, 8-element group min/max
 #define K_CHANGES 4   // ts, val, change-indicator prefix
+#define K_LAST    5   // ts, val only (LastSampleChunkedFunctionD)
 
 template <int FUNC> struct KKind { static constexpr int v =
     (FUNC <= FN_DELTA) ? K_RATE :
     (FUNC == FN_SUM || FUNC == FN_AVG || FUNC == FN_COUNT) ? K_PFX :
     (FUNC == FN_STDDEV || FUNC == FN_STDVAR) ? K_PFX_SQ :
-    (FUNC == FN_MIN || FUNC == FN_MAX) ? K_MINMAX : K_CHANGES; };
+    (FUNC == FN_MIN || FUNC == FN_MAX) ? K_MINMAX :
+    (FUNC == FN_LAST) ? K_LAST : K_CHANGES; };
 
 #define FDB_NGROUPS ((FDB_MAX_ROWS_PER_SERIES + 7) / 8)
 
@@ -620,6 +622,9 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
               for (int i = startRow; i <= endRow; i++) acc(ws.val[i]);
             }
             result = mm;
+          } else if constexpr (KIND == K_LAST) {
+            // last sample <= wEnd within window; raw value (NaN propagates)
+            result = seg[endRow] >= wStart ? ws.val[endRow] : NAN;
           } else {  // K_CHANGES, single chunk: prefix diff, prev starts NaN
             result = (double)((int)ws.cnt[endRow] - (int)ws.cnt[startRow]);
           }
@@ -701,6 +706,8 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     } else {
       double sum = NAN, sqsum = NAN, mm = NAN;
       double changes = NAN, prev = NAN;
+      double last_val = NAN;
+      int64_t last_ts = -1;
       int icount = 0;
       bool started = false;                        // CountOverTime: saw a nonempty range
       for (int c = 0; c < ws.nchunks; c++) {
@@ -709,6 +716,16 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         const int64_t* seg = ws.ts + m.row0;
         int startRow = lds_search_ge(seg, m.nrows, wStart, m.ts0, m.inv_slope);
         int endRow = lds_search_le(seg, m.nrows, wEnd, m.ts0, m.inv_slope);
+        if constexpr (KIND == K_LAST) {
+          // LastSampleChunkedFunction.addChunks (RangeFunction.scala:599-614):
+          // no startRow search; last ts <= wEnd wins if within the window
+          if (endRow >= 0 && endRow < m.nrows) {
+            int64_t t = seg[endRow];
+            if (t >= wStart && t > last_ts) { last_ts = t; last_val = ws.val[m.row0 + endRow]; }
+          }
+          if (m.end_time >= wEnd) break;
+          continue;
+        }
         if (startRow <= endRow && endRow < m.nrows) {
           const int a = m.row0 + startRow, b = m.row0 + endRow;
           if constexpr (KIND == K_PFX || KIND == K_PFX_SQ) {
@@ -767,6 +784,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
           } else result = isnan(sum) ? sum : 0;
         } break;
         case FN_CHANGES: result = changes; break;
+        case FN_LAST:    result = last_val; break;
       }
     }
 
@@ -1165,7 +1183,7 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
   switch (q->func_id) {
     CASE(FN_RATE); CASE(FN_INCREASE); CASE(FN_DELTA); CASE(FN_SUM); CASE(FN_COUNT);
     CASE(FN_AVG); CASE(FN_MIN); CASE(FN_MAX); CASE(FN_STDDEV); CASE(FN_STDVAR);
-    CASE(FN_CHANGES);
+    CASE(FN_CHANGES); CASE(FN_LAST);
     default: fdb_set_error("bad func_id %d", q->func_id); return FDB_ERR_BADARG;
   }
   #undef CASE

@@ -151,6 +151,11 @@ int32_t fdb_nibblepack_pack_doubles(const double* in, int32_t n, uint8_t* out, i
 #define FDB_FN_CHANGES         10
 #define FDB_FN_HIST_RATE       11   /* HistRateFunction (RateFunctions.scala:330-400):
                                        per-bucket counter-corrected extrapolated rate */
+#define FDB_FN_LAST            12   /* LastSampleChunkedFunctionD (RangeFunction.scala:595-694):
+                                       raw value of the last sample <= wEnd within the window
+                                       (NaN stale markers propagate); the reference uses
+                                       window = stale-sample-after + 1 = 300001ms for raw
+                                       queries (PeriodicSamplesMapper.scala:79-81) */
 
 /* Cross-series aggregation: RowAggregator implementations
  * (query/.../exec/aggregator/RowAggregator.scala:28-150). */

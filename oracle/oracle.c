@@ -51,7 +51,7 @@ typedef struct {
 } fdb_view_t;
 
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
-       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10 };
+       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12 };
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5 };
 
 static inline uint16_t rd_u16(const uint8_t* p) { uint16_t v; memcpy(&v, p, 2); return v; }
@@ -463,6 +463,8 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
       /* gauge family state (AggrOverTimeFunctions.scala) */
       double sum = NAN, count = NAN, sqsum = NAN, mn = NAN, mx = NAN;
       double changes = NAN, prev = NAN;
+      double last_val = NAN;
+      int64_t last_ts = -1;
       int icount = 0;
       for (int c = 0; c < nchunks; c++) {
         if (dir[c].end_time < wStart) continue;
@@ -471,6 +473,16 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
         int startRow = lv_binary_search(tv, wStart) & 0x7fffffff;
         int endRow = lv_ceiling(tv, wEnd);
         if (endRow > dir[c].num_rows - 1) endRow = dir[c].num_rows - 1;
+        if (q->func_id == FN_LAST) {
+          /* LastSampleChunkedFunction.addChunks (RangeFunction.scala:599-614):
+           * no startRow search; last ts <= wEnd wins if within the window */
+          if (endRow >= 0) {
+            int64_t t = lv_at(tv, endRow);
+            if (t >= wStart && t > last_ts) { last_ts = t; last_val = dv_at(vv, endRow); }
+          }
+          if (dir[c].end_time >= wEnd) break;
+          continue;
+        }
         if (startRow <= endRow) {
           switch (q->func_id) {
             case FN_SUM: {
@@ -540,6 +552,7 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
           result = r;
         } break;
         case FN_CHANGES: result = changes; break;
+        case FN_LAST:    result = last_val; break;
       }
     }
     out[w] = result;

@@ -78,11 +78,16 @@ def naive_window(ts, vs, chunk_bounds, w_start, w_end, func):
         return sum_ if np.isnan(sum_) else 0.0
     if func == "changes":
         return changes
+    if func == "last":
+        # LastSampleChunkedFunctionD: raw value of last sample <= wEnd within
+        # the window; NaN stale markers propagate (RangeFunction.scala:595-694)
+        m = (ts >= w_start) & (ts <= w_end)
+        return vs[m][-1] if m.any() else np.nan
     raise ValueError(func)
 
 
 FUNC_IDS = {"sum": 3, "count": 4, "avg": 5, "min": 6, "max": 7,
-            "stddev": 8, "stdvar": 9, "changes": 10}
+            "stddev": 8, "stdvar": 9, "changes": 10, "last": 12}
 
 
 @pytest.mark.parametrize("func", list(FUNC_IDS))
