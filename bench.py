@@ -133,14 +133,16 @@ def main():
     n_gpus = max(world, args.gpus)
 
     dist = None
+    import torch
     if world > 1:
-        import torch
         import torch.distributed as dist_mod
         dist = dist_mod
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
-
-    import torch
+        # device: one GPU per rank (the driver's 8-GPU launch); modulo lets the
+        # distributed path be smoke-tested on fewer GPUs with the gloo backend
+        device = local_rank % max(torch.cuda.device_count(), 1)
+        torch.cuda.set_device(device)
+        dist.init_process_group(os.environ.get("FDB_DIST_BACKEND", "nccl"))
+        local_rank = device
     w = WORKLOADS[args.workload]
     st = build_store(w, rank)
     q = make_query(w)
@@ -208,7 +210,8 @@ def main():
 
     ms_per_step = elapsed / args.steps * 1000
     if dist:
-        t = torch.tensor([ms_per_step], device=f"cuda:{local_rank}")
+        dev = f"cuda:{local_rank}" if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([ms_per_step], device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         ms_per_step = float(t.item())
 
