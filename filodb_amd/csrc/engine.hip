@@ -1078,7 +1078,7 @@ extern "C" fdb_engine_t* fdb_engine_create(int32_t device) {
 
 extern "C" void fdb_engine_destroy(fdb_engine_t* e) {
   if (!e) return;
-  hipStreamDestroy(e->stream);
+  (void)hipStreamDestroy(e->stream);
   delete e;
 }
 
@@ -1089,9 +1089,9 @@ extern "C" int32_t fdb_engine_synchronize(fdb_engine_t* e) {
 
 extern "C" void fdb_dataset_destroy(fdb_dataset_t* d) {
   if (!d) return;
-  hipFree(d->blob); hipFree(d->ts_off); hipFree(d->val_off);
-  hipFree(d->start_time); hipFree(d->end_time); hipFree(d->num_rows);
-  hipFree(d->series_first); hipFree(d->series_nchunks); hipFree(d->group_ids);
+  (void)hipFree(d->blob); (void)hipFree(d->ts_off); (void)hipFree(d->val_off);
+  (void)hipFree(d->start_time); (void)hipFree(d->end_time); (void)hipFree(d->num_rows);
+  (void)hipFree(d->series_first); (void)hipFree(d->series_nchunks); (void)hipFree(d->group_ids);
   delete d;
 }
 
@@ -1253,15 +1253,15 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
     HIP_CHECK(hipGetLastError());
   }
   HIP_CHECK(hipStreamSynchronize(e->stream));
-  hipEventDestroy(ev0); hipEventDestroy(ev1);
+  (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
 
   if (!out_on_device) {
     HIP_CHECK(hipMemcpy(out, dev_out, out_len * 8, hipMemcpyDeviceToHost));
     if (q->agg_id != AGG_NONE && out_counts)
       HIP_CHECK(hipMemcpy(out_counts, dev_cnt, out_len * 8, hipMemcpyDeviceToHost));
   }
-  if (own_out) hipFree(dev_out);
-  if (own_cnt) hipFree(dev_cnt);
+  if (own_out) (void)hipFree(dev_out);
+  if (own_cnt) (void)hipFree(dev_cnt);
   if (avg_ms) *avg_ms = iters > 0 ? (double)ms_sum / iters : 0.0;
   return FDB_OK;
 }
@@ -1325,9 +1325,9 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
     if (out_quantile)
       HIP_CHECK(hipMemcpy(out_quantile, dev_quant, cells * 8, hipMemcpyDeviceToHost));
   }
-  if (own_sums) hipFree(dev_sums);
-  if (own_cnt) hipFree(dev_cnt);
-  if (own_quant) hipFree(dev_quant);
+  if (own_sums) (void)hipFree(dev_sums);
+  if (own_cnt) (void)hipFree(dev_cnt);
+  if (own_quant) (void)hipFree(dev_quant);
   return FDB_OK;
 }
 
