@@ -382,6 +382,11 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   const int lane = threadIdx.x & 63;
   Ws<KIND>& ws = ws_all[wave];
 
+  // phase timing (ablation: phase_mask & 8) — wall cycles per wave per phase
+  const bool timing = (phase_mask & 8) != 0;
+  uint64_t t_decode = 0, t_meta = 0, t_win = 0, tt0 = 0;
+  if (timing) tt0 = __builtin_amdgcn_s_memtime();
+
   // grid-stride over series: a few resident blocks loop over the shard instead
   // of one block per 4 series — 250k tiny-workgroup dispatches cost multiple ms
   // of pure launch churn (measured ~20-26ns each), so the loop, not the grid,
@@ -426,6 +431,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   if (lane == 0) ws.total_rows = row0;
   __builtin_amdgcn_s_waitcnt(0);   // wave-local LDS visibility
   __builtin_amdgcn_wave_barrier();
+  if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_decode += t - tt0; tt0 = t; }
 
   // ---- meta phase ----------------------------------------------------------
   // every kind: search-guess slope per chunk
@@ -557,6 +563,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   }
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
+  if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_meta += t - tt0; tt0 = t; }
 
   // ---- window phase: lanes split the windows -------------------------------
   if (!(phase_mask & 2)) {            // decode-only ablation: publish a checksum
@@ -648,6 +655,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     }
     __builtin_amdgcn_s_waitcnt(0);
     __builtin_amdgcn_wave_barrier();
+    if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_win += t - tt0; tt0 = t; }
     continue;
   }
 
@@ -812,7 +820,15 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   // next series reuses this wave's LDS slot: wave-local ordering is enough
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
+  if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_win += t - tt0; tt0 = t; }
   }  // series grid-stride loop
+  if (timing && lane == 0) {
+    size_t gw = (size_t)blockIdx.x * WAVES_PER_BLOCK + wave;
+    out[gw * 4 + 0] = (double)t_decode;
+    out[gw * 4 + 1] = (double)t_meta;
+    out[gw * 4 + 2] = (double)t_win;
+    out[gw * 4 + 3] = 0;
+  }
 }
 
 

@@ -28,6 +28,22 @@ def main():
     ds = eng.upload(st)
     func = fdb.FN_RATE if workload == "rate" else fdb.FN_AVG_OVER_TIME
     window = 300_000 if workload == "rate" else 600_000
+    if len(sys.argv) > 3 and sys.argv[3] == "time":
+        # phase-timing mode: kernel reports wall cycles per wave per phase
+        q = fdb.make_query(T0, 15000, T0 + 240 * 15000, window, func)
+        q._pad = 3 | 8
+        out = torch.zeros(n_series * q.num_windows, dtype=torch.float64, device="cuda")
+        eng.query(ds, q, out=out, on_device=True)
+        eng.synchronize()
+        waves = 8192 * 4
+        t = out[:waves * 4].reshape(waves, 4).cpu().numpy()
+        t = t[t.sum(axis=1) > 0]
+        tot = t.sum(axis=0)
+        per = tot / tot.sum()
+        print(f"phase cycles (avg/wave): decode={t[:,0].mean():,.0f} "
+              f"meta={t[:,1].mean():,.0f} windows={t[:,2].mean():,.0f}")
+        print(f"phase share: decode={per[0]:.2%} meta={per[1]:.2%} windows={per[2]:.2%}")
+        return
     for phase, name in [(3, "full"), (1, "decode-only"), (2, "window-only")]:
         q = fdb.make_query(T0, 15000, T0 + 240 * 15000, window, func)
         q._pad = phase
