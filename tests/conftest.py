@@ -4,6 +4,11 @@ import sys
 import numpy as np
 import pytest
 
+# torch must initialize before the native engine touches the HIP runtime, or
+# torch.cuda sees no devices (observed on ROCm 7.2); importing here guarantees
+# the order for any single-file pytest invocation.
+import torch  # noqa: F401
+
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
 
