@@ -188,6 +188,21 @@ def main():
             hist_step()
         torch.cuda.synchronize()
         kernel_ms = (time.perf_counter() - t0) / args.steps * 1000  # sync per call
+    elif dist is not None and q.agg_id != fdb.AGG_NONE:
+        # config #5: the per-shard [G×W] partials merge with ONE collective —
+        # RCCL all-reduce over xGMI (ReduceAggregateExec equivalent).
+        # The collective is part of the timed step.
+        def agg_step():
+            eng.query(ds, q, out=out, out_counts=cnt, on_device=True)
+            dist.all_reduce(out)
+            dist.all_reduce(cnt)
+        for _ in range(args.warmup):
+            agg_step()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        agg_step()
+        torch.cuda.synchronize()
+        kernel_ms = (time.perf_counter() - t0) * 1000
     else:
         # warmup pass (also yields avg HIP-event kernel ms for the roofline), then
         # wall-time EXACTLY `steps` launches between barriers+synchronize.
@@ -200,6 +215,9 @@ def main():
     if is_hist:
         for _ in range(args.steps):
             hist_step()
+    elif dist is not None and q.agg_id != fdb.AGG_NONE:
+        for _ in range(args.steps):
+            agg_step()
     else:
         _ = eng.bench(ds, q, out, out_counts=cnt, on_device=True,
                       warmup=0, iters=args.steps)
