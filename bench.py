@@ -130,6 +130,17 @@ def main():
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if args.gpus > 1 and world == 1:
+        # not under torchrun: re-exec ourselves through it (one rank per GPU)
+        import subprocess
+        cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+               f"--nproc-per-node={args.gpus}", "--master-addr", "127.0.0.1",
+               "--master-port", "29507", os.path.abspath(__file__),
+               f"--gpus={args.gpus}", f"--steps={args.steps}",
+               f"--warmup={args.warmup}", f"--workload={args.workload}"]
+        if args.no_cpu_baseline:
+            cmd.append("--no-cpu-baseline")
+        sys.exit(subprocess.run(cmd).returncode)
     n_gpus = max(world, args.gpus)
 
     dist = None
