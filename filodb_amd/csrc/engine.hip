@@ -45,7 +45,8 @@ void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
 
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12 };
-enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5 };
+enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
+       AGG_TOPK=6, AGG_BOTTOMK=7 };
 
 // ---------------------------------------------------------------------------
 // device-side vector readers (same layouts as oracle)
@@ -1097,6 +1098,38 @@ __global__ void agg_present_kernel(double* out, const double* cnt, size_t n,
   if (agg_id == AGG_AVG) out[i] = out[i] / cnt[i];
 }
 
+// top/bottom-k presenter (TopBottomKRowAggregator.scala:29-100): one thread
+// per (group, window) walks its group's series in ascending id order over the
+// [S×W] grid, maintaining a sorted k-list — identical insertion order to the
+// oracle's fold, so results (including ties) match exactly.
+__global__ void topk_kernel(const double* __restrict__ grid,
+                            const int32_t* __restrict__ sbg,
+                            const int32_t* __restrict__ goff,
+                            int ng, int nw, int k, int top,
+                            double* __restrict__ out, double* __restrict__ ids) {
+  size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (size_t)ng * nw) return;
+  int g = (int)(idx / nw), w = (int)(idx % nw);
+  double best[16], bid[16];
+  for (int j = 0; j < 16; j++) { best[j] = NAN; bid[j] = -1; }
+  for (int i = goff[g]; i < goff[g + 1]; i++) {
+    int s = sbg[i];
+    double x = grid[(size_t)s * nw + w];
+    if (isnan(x)) continue;
+    int pos = -1;
+    for (int j = 0; j < k; j++)
+      if (isnan(best[j]) || (top ? x > best[j] : x < best[j])) { pos = j; break; }
+    if (pos >= 0) {
+      for (int j = k - 1; j > pos; j--) { best[j] = best[j - 1]; bid[j] = bid[j - 1]; }
+      best[pos] = x; bid[pos] = (double)s;
+    }
+  }
+  for (int j = 0; j < k; j++) {
+    out[idx * k + j] = best[j];
+    if (ids) ids[idx * k + j] = bid[j];
+  }
+}
+
 __global__ void fill_f64_kernel(double* p, size_t n, double v) {
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) p[i] = v;
@@ -1116,6 +1149,7 @@ struct fdb_dataset {
   int64_t *start_time, *end_time;
   int32_t *num_rows;
   int32_t *series_first, *series_nchunks, *group_ids;
+  int32_t *series_by_group, *group_offsets;   // group-sorted series index (topk)
   int32_t num_series;
   int64_t num_chunks;
   int64_t payload_bytes;    // sum of vector bytes (algorithmic HBM footprint)
@@ -1214,6 +1248,18 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
     if (hipMalloc(dst, bytes ? bytes : 8) != hipSuccess) return false;
     return hipMemcpy(*dst, src, bytes, hipMemcpyHostToDevice) == hipSuccess;
   };
+  // group-sorted series index for the top/bottom-k presenter: stable ascending
+  // series order per group (ties then resolve identically to the oracle's fold)
+  std::vector<int32_t> goff((size_t)max_group + 2, 0);
+  for (int32_t s2 = 0; s2 < view.num_series; s2++) goff[(size_t)view.group_ids[s2] + 1]++;
+  for (size_t g = 1; g < goff.size(); g++) goff[g] += goff[g - 1];
+  std::vector<int32_t> sbg((size_t)view.num_series);
+  {
+    std::vector<int32_t> cur(goff.begin(), goff.end() - 1);
+    for (int32_t s2 = 0; s2 < view.num_series; s2++)
+      sbg[(size_t)cur[(size_t)view.group_ids[s2]]++] = s2;
+  }
+
   bool ok = upload((void**)&d->blob, view.blob, (size_t)view.blob_len)
     && upload((void**)&d->ts_off, ts_off.data(), nc * 8)
     && upload((void**)&d->val_off, val_off.data(), nc * 8)
@@ -1222,7 +1268,9 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
     && upload((void**)&d->num_rows, nr.data(), nc * 4)
     && upload((void**)&d->series_first, view.series_first, (size_t)view.num_series * 4)
     && upload((void**)&d->series_nchunks, view.series_nchunks, (size_t)view.num_series * 4)
-    && upload((void**)&d->group_ids, view.group_ids, (size_t)view.num_series * 4);
+    && upload((void**)&d->group_ids, view.group_ids, (size_t)view.num_series * 4)
+    && upload((void**)&d->series_by_group, sbg.data(), sbg.size() * 4)
+    && upload((void**)&d->group_offsets, goff.data(), goff.size() * 4);
   if (!ok) {
     fdb_set_error("device upload failed (out of HBM?)");
     fdb_dataset_destroy(d);
@@ -1264,8 +1312,15 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   HIP_CHECK(hipSetDevice(e->device));
   int nw = fdb_num_windows(q);
   if (nw <= 0) { fdb_set_error("bad window params"); return FDB_ERR_BADARG; }
+  const bool is_topk = q->agg_id == AGG_TOPK || q->agg_id == AGG_BOTTOMK;
+  const int kk = is_topk ? (int)q->param : 0;
+  if (is_topk && (kk < 1 || kk > 16)) {
+    fdb_set_error("topk k=%d out of range 1..16 (q.param)", kk);
+    return FDB_ERR_BADARG;
+  }
   size_t out_len = (q->agg_id == AGG_NONE) ? (size_t)d->num_series * nw
-                                           : (size_t)q->num_groups * nw;
+                 : is_topk ? (size_t)q->num_groups * nw * kk
+                           : (size_t)q->num_groups * nw;
   if (q->agg_id != AGG_NONE) {
     if (q->num_groups <= 0 || d->max_group >= q->num_groups) {
       fdb_set_error("num_groups %d inconsistent with dataset max group %d",
@@ -1291,10 +1346,18 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   HIP_CHECK(hipEventCreate(&ev0));
   HIP_CHECK(hipEventCreate(&ev1));
 
+  // topk: the scan fills an internal [S×W] grid, the presenter reduces it
+  double* topk_grid = nullptr;
+  fdb_query_t qscan = *q;
+  if (is_topk) {
+    HIP_CHECK(hipMalloc(&topk_grid, (size_t)d->num_series * nw * 8));
+    qscan.agg_id = AGG_NONE;
+  }
+
   int total_runs = warmup + iters;
   float ms_sum = 0;
   for (int it = 0; it < total_runs; it++) {
-    if (q->agg_id != AGG_NONE) {
+    if (q->agg_id != AGG_NONE && !is_topk) {
       HIP_CHECK(hipMemsetAsync(dev_out, 0, out_len * 8, e->stream));
       HIP_CHECK(hipMemsetAsync(dev_cnt, 0, out_len * 8, e->stream));
       if (q->agg_id == AGG_MIN || q->agg_id == AGG_MAX) {
@@ -1304,8 +1367,17 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
     }
     bool timed = it >= warmup;
     if (timed) HIP_CHECK(hipEventRecord(ev0, e->stream));
-    int32_t rc = launch_scan(e, d, q, dev_out, dev_cnt);
+    int32_t rc = launch_scan(e, d, &qscan, is_topk ? topk_grid : dev_out,
+                             is_topk ? nullptr : dev_cnt);
     if (rc != FDB_OK) return rc;
+    if (is_topk) {
+      size_t cells = (size_t)q->num_groups * nw;
+      topk_kernel<<<(unsigned)((cells + 255) / 256), 256, 0, e->stream>>>(
+          topk_grid, d->series_by_group, d->group_offsets,
+          q->num_groups, nw, kk, q->agg_id == AGG_TOPK ? 1 : 0,
+          dev_out, dev_cnt);
+      HIP_CHECK(hipGetLastError());
+    }
     if (timed) {
       HIP_CHECK(hipEventRecord(ev1, e->stream));
       HIP_CHECK(hipEventSynchronize(ev1));
@@ -1314,11 +1386,12 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
       ms_sum += ms;
     }
   }
-  if (q->agg_id != AGG_NONE) {
+  if (q->agg_id != AGG_NONE && !is_topk) {
     agg_present_kernel<<<(unsigned)((out_len + 255) / 256), 256, 0, e->stream>>>(
         dev_out, dev_cnt, out_len, q->agg_id, partial);
     HIP_CHECK(hipGetLastError());
   }
+  if (topk_grid) (void)hipFree(topk_grid);
   HIP_CHECK(hipStreamSynchronize(e->stream));
   (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
 

@@ -165,6 +165,13 @@ int32_t fdb_nibblepack_pack_doubles(const double* in, int32_t n, uint8_t* out, i
 #define FDB_AGG_MIN    3
 #define FDB_AGG_MAX    4
 #define FDB_AGG_AVG    5   /* AvgRowAggregator.scala:8-41 (sum,count partials) */
+#define FDB_AGG_TOPK   6   /* TopBottomKRowAggregator.scala:29-100: k largest
+                              non-NaN series values per (group, window).
+                              q.param = k (<=16). out: [G × W × k] values sorted
+                              descending (NaN-padded); out_counts reinterpreted
+                              as [G × W × k] doubles holding the series ids
+                              (-1 padding). */
+#define FDB_AGG_BOTTOMK 7  /* same, k smallest; values sorted ascending */
 
 typedef struct {
   int64_t start;     /* first window end timestamp (ms)                      */

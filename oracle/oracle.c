@@ -52,7 +52,8 @@ typedef struct {
 
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12 };
-enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5 };
+enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
+       AGG_TOPK=6, AGG_BOTTOMK=7 };
 
 static inline uint16_t rd_u16(const uint8_t* p) { uint16_t v; memcpy(&v, p, 2); return v; }
 static inline uint32_t rd_u32(const uint8_t* p) { uint32_t v; memcpy(&v, p, 4); return v; }
@@ -587,6 +588,46 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
 
   int ng = q->num_groups;
   size_t gridlen = (size_t)ng * nw;
+
+  if (q->agg_id == AGG_TOPK || q->agg_id == AGG_BOTTOMK) {
+    /* TopBottomKRowAggregator.scala:29-100: per (group, window) keep the k
+     * largest (topk) / smallest (bottomk) non-NaN series values with their
+     * keys; NaN padding. Output sorted descending (topk) / ascending. */
+    int k = (int)q->param;
+    if (k < 1 || k > 16) return -1;
+    int top = q->agg_id == AGG_TOPK;
+    size_t cells = gridlen * (size_t)k;
+    double* ids = out_counts;         /* [G × W × k] series ids as doubles */
+    for (size_t i = 0; i < cells; i++) { out[i] = NAN; if (ids) ids[i] = -1; }
+    double* row = (double*)malloc((size_t)nw * sizeof(double));
+    eval_ctx_t ctx; ctx.scratch = (double*)malloc(64 * 512 * sizeof(double));
+    for (int s = 0; s < ns; s++) {
+      eval_series(view, s, q, &ctx, row);
+      int grp = view->group_ids[s];
+      for (int w = 0; w < nw; w++) {
+        double x = row[w];
+        if (isnan(x)) continue;
+        double* cell = out + ((size_t)grp * nw + w) * k;
+        double* cid = ids ? ids + ((size_t)grp * nw + w) * k : 0;
+        /* insertion into the sorted-k list (k <= 16) */
+        int pos = -1;
+        for (int j = 0; j < k; j++) {
+          if (isnan(cell[j]) || (top ? x > cell[j] : x < cell[j])) { pos = j; break; }
+        }
+        if (pos >= 0) {
+          for (int j = k - 1; j > pos; j--) {
+            cell[j] = cell[j - 1];
+            if (cid) cid[j] = cid[j - 1];
+          }
+          cell[pos] = x;
+          if (cid) cid[pos] = (double)s;
+        }
+      }
+    }
+    free(row); free(ctx.scratch);
+    return 0;
+  }
+
   for (size_t i = 0; i < gridlen; i++) out[i] = NAN;
   double* counts = out_counts;
   double* owned_counts = 0;

@@ -97,8 +97,11 @@ def query_exec(view, q, num_series, num_windows, out_counts=False, nthreads=1):
         out = np.empty(num_series * num_windows, dtype=np.float64)
         cnt = None
     else:
-        out = np.empty(q.num_groups * num_windows, dtype=np.float64)
-        cnt = np.zeros(q.num_groups * num_windows, dtype=np.float64) if out_counts else None
+        cells = q.num_groups * num_windows
+        if q.agg_id in (6, 7):        # top/bottom-k: [G × W × k] values + ids
+            cells *= int(q.param)
+        out = np.empty(cells, dtype=np.float64)
+        cnt = np.zeros(cells, dtype=np.float64) if out_counts else None
     rc = lib().oracle_query_exec(
         ctypes.byref(view), ctypes.byref(q),
         out.ctypes.data_as(_c_double_p),
