@@ -271,68 +271,6 @@ __device__ void d_decode_chunk(const DVec& v, int n, int64_t* tout, double* dout
   }
 }
 
-// gather 4 contiguous elements [i0, i0+4) into registers — the fused
-// decode+scan phase gives each lane a contiguous run so one wave scan per
-// 256-element block replaces four strided scans, and prefix arrays are
-// written once with no raw-value LDS round trip (DESIGN.md §9 round-2 lever)
-template <bool AS_DOUBLE>
-__device__ __forceinline__ void d_gather4(const DVec& v, int i0, int n,
-                                          int64_t* tq, double* dq) {
-  if (i0 >= n) {
-    #pragma unroll
-    for (int k = 0; k < 4; k++) { if (AS_DOUBLE) dq[k] = 0; else tq[k] = 0; }
-    return;
-  }
-  if (v.wf == FDB_WF_DDV_CONST) {
-    #pragma unroll
-    for (int k = 0; k < 4; k++) {
-      int64_t x = v.init + (int64_t)v.slope * (i0 + k);
-      if (AS_DOUBLE) dq[k] = (double)x; else tq[k] = x;
-    }
-    return;
-  }
-  if (v.wf == FDB_WF_PRIM64) {
-    #pragma unroll
-    for (int k = 0; k < 4; k++) {
-      if (AS_DOUBLE) dq[k] = d_f64(v.idata + 8 * (size_t)(i0 + k));
-      else           tq[k] = d_i64(v.idata + 8 * (size_t)(i0 + k));
-    }
-    return;
-  }
-  int64_t d4[4] = {0, 0, 0, 0};
-  if (v.nbits == 16) {          // 8 bytes at a 4-aligned offset (idata = +28)
-    uint32_t lo = d_u32(v.idata + 2 * (size_t)i0);
-    uint32_t hi = d_u32(v.idata + 2 * (size_t)i0 + 4);
-    uint64_t w = ((uint64_t)hi << 32) | lo;
-    #pragma unroll
-    for (int k = 0; k < 4; k++) {
-      int32_t d = (int32_t)(int16_t)(uint16_t)(w >> (16 * k));
-      if (!v.sign) d &= 0xffff;
-      d4[k] = d;
-    }
-  } else if (v.nbits == 8) {
-    uint32_t w = d_u32(v.idata + (size_t)i0);
-    #pragma unroll
-    for (int k = 0; k < 4; k++) {
-      int32_t d = (int32_t)(int8_t)(uint8_t)(w >> (8 * k));
-      if (!v.sign) d &= 0xff;
-      d4[k] = d;
-    }
-  } else if (v.nbits == 32) {
-    #pragma unroll
-    for (int k = 0; k < 4; k++) d4[k] = d_i32(v.idata + 4 * (size_t)(i0 + k));
-  } else {                      // nbits 2/4 (rare)
-    #pragma unroll
-    for (int k = 0; k < 4; k++)
-      if (i0 + k < n) d4[k] = d_inner_at(&v, i0 + k);
-  }
-  #pragma unroll
-  for (int k = 0; k < 4; k++) {
-    int64_t x = v.init + (int64_t)v.slope * (i0 + k) + d4[k];
-    if (AS_DOUBLE) dq[k] = (double)x; else tq[k] = x;
-  }
-}
-
 // first index in [0,n) with seg[i] >= item; n when none. Timestamps sit near a
 // slope line (the DDV premise), so an interpolation guess + short walk replaces
 // a dependent binary-search chain (usually 1-2 LDS reads) — the same idea as
@@ -424,7 +362,7 @@ __device__ void atomic_min_max_f64(double* addr, double val, bool is_min) {
 template <int FUNC>
 // rate family carries more live state (correction meta): forcing 5 waves/SIMD
 // makes it spill; the gauge kinds fit 5 waves cleanly
-__global__ __launch_bounds__(BLOCK_THREADS, 4)
+__global__ __launch_bounds__(BLOCK_THREADS, (FUNC <= FN_DELTA) ? 4 : 5)
 void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                  const int32_t* __restrict__ series_first,
                  const int32_t* __restrict__ series_nchunks,
@@ -463,195 +401,23 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   if (lane == 0) { ws.nchunks = nchunks; ws.total_rows = 0; }
 
   int row0 = 0;
-  int table_used = 0;
   for (int c = 0; c < nchunks; c++) {
     int n = dir.num_rows[first + c];
     if (row0 + n > FDB_MAX_ROWS_PER_SERIES) n = 0;   // guarded at upload
-    DVec tv, vv;
-    vv.dropped = 0;
+    DVec vv;
     if (phase_mask & 1) {
+      DVec tv;
       d_vec_open(blob + dir.ts_off[first + c], &tv);
       d_vec_open(blob + dir.val_off[first + c], &vv);
-    }
-
-    // ---- fused decode + per-chunk scans: lane owns elements [4L, 4L+4) of
-    // each 256-element block; one wave scan per block -----
-    // carries across 256-element blocks
-    double carryP = 0, carrySq = 0;        // PFX value/square prefixes
-    int carryC = 0;                        // PFX count / CHANGES indicator prefix
-    double carry_corr = 0;                 // RATE correction running total
-    double carry_x = -1.7976931348623157e308;
-    double carry_prev = NAN;               // CHANGES cross-block previous value
-    int dstart = table_used, dcnt = 0;     // RATE drop-table slice
-    bool overflow = false;
-    int best_idx = -1; double best_val = 0;   // RATE last non-NaN (lane-local)
-    bool v0_nan = false;
-
-    for (int base = 0; base < n; base += 256) {
-      const int i0 = base + 4 * lane;
-      int64_t t4[4];
-      double v4[4];
-      if (phase_mask & 1) {
-        d_gather4<false>(tv, i0, n, t4, nullptr);
-        d_gather4<true>(vv, i0, n, nullptr, v4);
-      } else {                       // window-only ablation: synthesize rows
-        #pragma unroll
-        for (int k = 0; k < 4; k++) {
-          t4[k] = dir.start_time[first + c] + (int64_t)(i0 + k) * 15000;
-          v4[k] = (double)((i0 + k) * 10);
-        }
-      }
-      const int live = i0 < n ? (n - i0 < 4 ? n - i0 : 4) : 0;
-      #pragma unroll
-      for (int k = 0; k < 4; k++)
-        if (k < live) ws.ts[row0 + i0 + k] = t4[k];
-      if constexpr (KIND != K_PFX && KIND != K_PFX_SQ) {
-        #pragma unroll
-        for (int k = 0; k < 4; k++)
-          if (k < live) ws.val[row0 + i0 + k] = v4[k];
-      }
-
-      if constexpr (KIND == K_PFX || KIND == K_PFX_SQ) {
-        // NaN-zeroed value (and square) prefix + non-NaN count prefix
-        double lp[4], lq[4];
-        int lc[4];
-        double run = 0, runq = 0;
-        int runc = 0;
-        #pragma unroll
-        for (int k = 0; k < 4; k++) {
-          bool ok = k < live && !isnan(v4[k]);
-          double x = ok ? v4[k] : 0;
-          run += x; runc += ok ? 1 : 0;
-          lp[k] = run; lc[k] = runc;
-          if constexpr (KIND == K_PFX_SQ) { runq += x * x; lq[k] = runq; }
-        }
-        double s = wave_incl_scan(run, lane);
-        int cs = wave_incl_scan_i(runc, lane);
-        double excl = s - run;
-        int excl_c = cs - runc;
-        double sq_s = 0, excl_q = 0;
-        if constexpr (KIND == K_PFX_SQ) {
-          sq_s = wave_incl_scan(runq, lane);
-          excl_q = sq_s - runq;
-        }
-        #pragma unroll
-        for (int k = 0; k < 4; k++) {
-          if (k < live) {
-            ws.val[row0 + i0 + k] = carryP + excl + lp[k];
-            ws.cnt[row0 + i0 + k] = (uint16_t)(carryC + excl_c + lc[k]);
-            if constexpr (KIND == K_PFX_SQ)
-              ws.sq[row0 + i0 + k] = carrySq + excl_q + lq[k];
-          }
-        }
-        carryP += __shfl(s, 63);
-        carryC += __shfl(cs, 63);
-        if constexpr (KIND == K_PFX_SQ) carrySq += __shfl(sq_s, 63);
-      }
-
-      if constexpr (RATE_FAMILY) {
-        if (base == 0 && lane == 0) v0_nan = isnan(v4[0]);
-        if (vv.dropped) {
-          // counter-correction scan (CorrectingDoubleVectorReader :325-342):
-          // NaN→0, a decrease adds the previous value; recorded sparsely
-          double x[4];
-          #pragma unroll
-          for (int k = 0; k < 4; k++) x[k] = (k < live && !isnan(v4[k])) ? v4[k] : 0;
-          double p = __shfl_up(x[3], 1);
-          if (lane == 0) p = carry_x;
-          double cdrop[4], lcum[4];
-          double run = 0;
-          int nj = 0;
-          #pragma unroll
-          for (int k = 0; k < 4; k++) {
-            double prev = k == 0 ? p : x[k - 1];
-            cdrop[k] = (k < live && x[k] < prev) ? prev : 0;
-            run += cdrop[k];
-            lcum[k] = run;
-            nj += cdrop[k] != 0 ? 1 : 0;
-          }
-          double s = wave_incl_scan(run, lane);
-          double excl = s - run;
-          int sb = wave_incl_scan_i(nj, lane);
-          int slot0 = dstart + dcnt + (sb - nj);
-          int total = __shfl(sb, 63);
-          if (dcnt + dstart + total > FDB_MAX_DROPS) {
-            overflow = true;
-          } else {
-            int li = 0;
-            #pragma unroll
-            for (int k = 0; k < 4; k++) {
-              if (cdrop[k] != 0) {
-                ws.dpos[slot0 + li] = (int16_t)(i0 + k);
-                ws.dcum[slot0 + li] = carry_corr + excl + lcum[k];
-                li++;
-              }
-            }
-          }
-          dcnt += total;
-          carry_corr += __shfl(s, 63);
-          carry_x = __shfl(x[3], 63);
-          #pragma unroll
-          for (int k = 0; k < 4; k++)
-            if (k < live && !isnan(v4[k])) { best_idx = i0 + k; best_val = v4[k]; }
-        }
-      }
-
-      if constexpr (KIND == K_MINMAX) {
-        constexpr bool IS_MIN = (FUNC == FN_MIN);
-        double g = NAN;
-        #pragma unroll
-        for (int k = 0; k < 4; k++) {
-          double xx = k < live ? v4[k] : NAN;
-          if (!isnan(xx) && (isnan(g) || (IS_MIN ? xx < g : xx > g))) g = xx;
-        }
-        double o = __shfl_xor(g, 1);      // lanes 2g,2g+1 hold elements 8g..8g+7
-        if (!isnan(o) && (isnan(g) || (IS_MIN ? o < g : o > g))) g = o;
-        if ((lane & 1) == 0 && i0 < n)
-          ws.grp[(row0 >> 3) + c + (i0 >> 3)] = g;
-      }
-
-      if constexpr (KIND == K_CHANGES) {
-        // change-indicator prefix (DoubleVectorDataReader64.changes :283-302)
-        double p = __shfl_up(v4[3], 1);
-        if (lane == 0) p = carry_prev;
-        int li[4];
-        int run = 0;
-        #pragma unroll
-        for (int k = 0; k < 4; k++) {
-          double prev = k == 0 ? p : v4[k - 1];
-          bool ind = (i0 + k) > 0 && k < live && !isnan(v4[k]) && !isnan(prev)
-                     && v4[k] != prev;
-          run += ind ? 1 : 0;
-          li[k] = run;
-        }
-        int s = wave_incl_scan_i(run, lane);
-        int excl = s - run;
-        #pragma unroll
-        for (int k = 0; k < 4; k++)
-          if (k < live) ws.cnt[row0 + i0 + k] = (uint16_t)(carryC + excl + li[k]);
-        carryC += __shfl(s, 63);
-        carry_prev = __shfl(v4[3], 63);
+      d_decode_chunk<false>(tv, n, ws.ts + row0, nullptr, lane);
+      d_decode_chunk<true>(vv, n, nullptr, ws.val + row0, lane);
+    } else {                       // window-only ablation: synthesize rows
+      vv.dropped = 0;
+      for (int i = lane; i < n; i += 64) {
+        ws.ts[row0 + i] = dir.start_time[first + c] + (int64_t)i * 15000;
+        ws.val[row0 + i] = (double)(i * 10);
       }
     }
-
-    // RATE: reduce the lane-local last-non-NaN candidates once per chunk
-    double red_val = 0;
-    int red_idx = -1;
-    if constexpr (RATE_FAMILY) {
-      if (vv.dropped) {
-        red_idx = best_idx; red_val = best_val;
-        for (int off = 32; off > 0; off >>= 1) {
-          int oi = __shfl_down(red_idx, off);
-          double ov = __shfl_down(red_val, off);
-          if (oi > red_idx) { red_idx = oi; red_val = ov; }
-        }
-      }
-    }
-
-    __builtin_amdgcn_s_waitcnt(0);       // chunk's LDS writes visible
-    __builtin_amdgcn_wave_barrier();
-
-    if (!overflow) table_used = dstart + dcnt;
     if (lane == 0) {
       ChunkMeta& m = ws.cm[c];
       m.row0 = row0; m.nrows = n;
@@ -660,30 +426,144 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       m.dropped = vv.dropped;
       m.chunk_correction = 0; m.last_for_update = 0;
       m.dcount = 0; m.dense_corr = 0; m.v0_nan = 0;
-      if (n > 0) {
-        m.ts0 = ws.ts[row0];
-        int64_t tl = ws.ts[row0 + n - 1];
-        m.inv_slope = (tl > m.ts0) ? (float)(n - 1) / (float)(tl - m.ts0) : 0.0f;
-        if constexpr (RATE_FAMILY) {
-          m.v0_nan = v0_nan;
-          if (vv.dropped) {
-            m.chunk_correction = carry_corr;
-            m.last_for_update = (red_idx >= 0) ? red_val : 0;
-            m.dstart = (int16_t)dstart;
-            m.dcount = overflow ? 0 : (int16_t)dcnt;
-            m.dense_corr = overflow ? 1 : 0;
-          } else {
-            m.last_for_update = ws.val[row0 + n - 1];  // default updateCorrection
-          }
-        }
-      }
     }
     row0 += n;
   }
   if (lane == 0) ws.total_rows = row0;
-  __builtin_amdgcn_s_waitcnt(0);         // meta writes visible
+  __builtin_amdgcn_s_waitcnt(0);   // wave-local LDS visibility
   __builtin_amdgcn_wave_barrier();
   if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_decode += t - tt0; tt0 = t; }
+
+  // ---- meta phase ----------------------------------------------------------
+  // every kind: search-guess slope per chunk
+  // K_RATE: counter-correction scan (CorrectingDoubleVectorReader :325-342)
+  //         recorded as a sparse drop table + updateCorrection scalars
+  // K_PFX(_SQ): NaN-zeroed value (and square) prefix + non-NaN count prefix —
+  //         the ~97% window overlap then costs O(1) per window, not O(rows)
+  // K_MINMAX: 8-element group min/max
+  // K_CHANGES: change-indicator prefix (DoubleVectorDataReader64.changes :283-302)
+  int table_used = 0;
+  for (int c = 0; c < nchunks; c++) {
+    ChunkMeta& m = ws.cm[c];
+    const int r0 = m.row0, n = m.nrows;
+    if (n == 0) continue;
+    if (lane == 0) {
+      m.ts0 = ws.ts[r0];
+      int64_t tl = ws.ts[r0 + n - 1];
+      m.inv_slope = (tl > m.ts0) ? (float)(n - 1) / (float)(tl - m.ts0) : 0.0f;
+    }
+
+    if constexpr (RATE_FAMILY) {
+      if (lane == 0) m.v0_nan = isnan(ws.val[r0]);
+      if (m.dropped) {
+        double carry_corr = 0;
+        double carry_x = -1.7976931348623157e308;   // 'last' starts Double.MinValue
+        int last_idx = -1;
+        double last_val = 0;
+        int dstart = table_used, dcnt = 0;
+        bool overflow = false;
+        for (int base = 0; base < n; base += 64) {
+          int i = base + lane;
+          double raw = (i < n) ? ws.val[r0 + i] : 0;
+          double x = (i < n && !isnan(raw)) ? raw : 0;
+          double px = __shfl_up(x, 1);
+          if (lane == 0) px = carry_x;
+          double ci = (i < n && x < px) ? px : 0;
+          double scan = wave_incl_scan(ci, lane);
+          uint64_t mask = __ballot(ci != 0);
+          int here = __popcll(mask);
+          if (here) {
+            if (dstart + dcnt + here > FDB_MAX_DROPS) {
+              overflow = true;
+            } else if (ci != 0) {
+              int slot = dstart + dcnt + __popcll(mask & ((1ULL << lane) - 1));
+              ws.dpos[slot] = (int16_t)i;
+              ws.dcum[slot] = carry_corr + scan;
+            }
+            dcnt += here;
+          }
+          carry_corr += __shfl(scan, 63);
+          carry_x = __shfl(x, 63);
+          if (i < n && !isnan(raw)) { last_idx = i; last_val = raw; }
+        }
+        for (int off = 32; off > 0; off >>= 1) {
+          int oi = __shfl_down(last_idx, off);
+          double ov = __shfl_down(last_val, off);
+          if (oi > last_idx) { last_idx = oi; last_val = ov; }
+        }
+        if (!overflow) table_used = dstart + dcnt;
+        if (lane == 0) {
+          m.chunk_correction = carry_corr;
+          m.last_for_update = (last_idx >= 0) ? last_val : 0;
+          m.dstart = (int16_t)dstart;
+          m.dcount = overflow ? 0 : (int16_t)dcnt;
+          m.dense_corr = overflow ? 1 : 0;
+        }
+      } else if (lane == 0) {
+        m.last_for_update = ws.val[r0 + n - 1];     // default updateCorrection
+      }
+    }
+
+    if constexpr (KIND == K_PFX || KIND == K_PFX_SQ) {
+      double carry = 0, carry_sq = 0;
+      int ccarry = 0;
+      for (int base = 0; base < n; base += 64) {
+        int i = base + lane;
+        double raw = (i < n) ? ws.val[r0 + i] : NAN;
+        bool ok = (i < n) && !isnan(raw);
+        double x = ok ? raw : 0;
+        double s = wave_incl_scan(x, lane);
+        int cs = wave_incl_scan_i(ok ? 1 : 0, lane);
+        double sqs = 0;
+        if constexpr (KIND == K_PFX_SQ) sqs = wave_incl_scan(x * x, lane);
+        __builtin_amdgcn_wave_barrier();            // all raw reads precede writes
+        if (i < n) {
+          ws.val[r0 + i] = carry + s;               // val[] becomes the prefix
+          ws.cnt[r0 + i] = (uint16_t)(ccarry + cs);
+          if constexpr (KIND == K_PFX_SQ) ws.sq[r0 + i] = carry_sq + sqs;
+        }
+        carry += __shfl(s, 63);
+        ccarry += __shfl(cs, 63);
+        if constexpr (KIND == K_PFX_SQ) carry_sq += __shfl(sqs, 63);
+      }
+    }
+
+    if constexpr (KIND == K_MINMAX) {
+      constexpr bool IS_MIN = (FUNC == FN_MIN);
+      for (int base = 0; base < n; base += 64) {
+        int i = base + lane;
+        double x = (i < n) ? ws.val[r0 + i] : NAN;
+        // 8-lane group reduce (elements 8g..8g+7 are lanes 8k..8k+7)
+        #pragma unroll
+        for (int off = 1; off < 8; off <<= 1) {
+          double o = __shfl_xor(x, off);
+          if (!isnan(o) && (isnan(x) || (IS_MIN ? o < x : o > x))) x = o;
+        }
+        if ((lane & 7) == 0 && i < n)
+          ws.grp[(r0 >> 3) + c + (i >> 3)] = x;   // chunk-relative groups;
+                                                  // +c pads across chunk seams
+      }
+    }
+
+    if constexpr (KIND == K_CHANGES) {
+      // ind[i] = in-chunk change at i (i>0); prefix stored in cnt[]
+      int ccarry = 0;
+      double carry_x = NAN;
+      for (int base = 0; base < n; base += 64) {
+        int i = base + lane;
+        double x = (i < n) ? ws.val[r0 + i] : NAN;
+        double px = __shfl_up(x, 1);
+        if (lane == 0) px = carry_x;
+        int ind = (i > 0 && i < n && !isnan(x) && !isnan(px) && x != px) ? 1 : 0;
+        int s = wave_incl_scan_i(ind, lane);
+        if (i < n) ws.cnt[r0 + i] = (uint16_t)(ccarry + s);
+        ccarry += __shfl(s, 63);
+        carry_x = __shfl(x, 63);
+      }
+    }
+  }
+  __builtin_amdgcn_s_waitcnt(0);
+  __builtin_amdgcn_wave_barrier();
   if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_meta += t - tt0; tt0 = t; }
 
   // ---- window phase: lanes split the windows -------------------------------
