@@ -46,7 +46,7 @@ void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12 };
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
-       AGG_TOPK=6, AGG_BOTTOMK=7 };
+       AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9 };
 
 // ---------------------------------------------------------------------------
 // device-side vector readers (same layouts as oracle)
@@ -373,6 +373,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                  int agg_id,
                  double* __restrict__ out,        // [S×W] when AGG_NONE else [G×W] sums
                  double* __restrict__ out_cnt,    // [G×W] contribution counts (agg) or null
+                 double* __restrict__ out_sq,     // [G×W] sum-of-squares (stddev aggs) or null
                  int phase_mask)                  // debug ablation: 1=decode 2=windows
 {
   constexpr int KIND = KKind<FUNC>::v;
@@ -695,6 +696,9 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         switch (agg_id) {
           case AGG_SUM: case AGG_AVG:
             atomicAdd(&out[cell], result); atomicAdd(&out_cnt[cell], 1.0); break;
+          case AGG_STDDEV: case AGG_STDVAR:
+            atomicAdd(&out[cell], result); atomicAdd(&out_sq[cell], result * result);
+            atomicAdd(&out_cnt[cell], 1.0); break;
           case AGG_COUNT:
             atomicAdd(&out[cell], 1.0); atomicAdd(&out_cnt[cell], 1.0); break;
           case AGG_MIN: atomic_min_max_f64(&out[cell], result, true);
@@ -856,6 +860,11 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       switch (agg_id) {
         case AGG_SUM: case AGG_AVG:
           atomicAdd(&out[cell], result);
+          atomicAdd(&out_cnt[cell], 1.0);
+          break;
+        case AGG_STDDEV: case AGG_STDVAR:
+          atomicAdd(&out[cell], result);
+          atomicAdd(&out_sq[cell], result * result);
           atomicAdd(&out_cnt[cell], 1.0);
           break;
         case AGG_COUNT:
@@ -1089,13 +1098,18 @@ __global__ void hist_quantile_kernel(const double* __restrict__ sums,
 }
 
 // presentation fixup for aggregated grids (NaN where no contributions; mean for avg)
-__global__ void agg_present_kernel(double* out, const double* cnt, size_t n,
-                                   int agg_id, int partial) {
+__global__ void agg_present_kernel(double* out, const double* cnt, const double* sq,
+                                   size_t n, int agg_id, int partial) {
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   if (partial) return;                      // partial mode: leave raw sums + counts
   if (cnt[i] <= 0) { out[i] = NAN; return; }
   if (agg_id == AGG_AVG) out[i] = out[i] / cnt[i];
+  else if (agg_id == AGG_STDDEV || agg_id == AGG_STDVAR) {
+    double mean = out[i] / cnt[i];
+    double var = sq[i] / cnt[i] - mean * mean;   // StddevRowAggregator.scala:49-52
+    out[i] = agg_id == AGG_STDDEV ? sqrt(var) : var;
+  }
 }
 
 // top/bottom-k presenter (TopBottomKRowAggregator.scala:29-100): one thread
@@ -1281,7 +1295,7 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
 
 // launch dispatch over the func template parameter
 static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
-                           double* dev_out, double* dev_cnt) {
+                           double* dev_out, double* dev_cnt, double* dev_sq) {
   DirSoA dir{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
   int nw = fdb_num_windows(q);
   int grid = (d->num_series + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
@@ -1293,7 +1307,7 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
   #define CASE(F) case F: \
     hipLaunchKernelGGL(scan_kernel<F>, dim3(grid), dim3(BLOCK_THREADS), 0, e->stream, \
       d->blob, dir, d->series_first, d->series_nchunks, d->group_ids, d->num_series, \
-      q->start, q->step, q->end, q->window, nw, q->agg_id, dev_out, dev_cnt, \
+      q->start, q->step, q->end, q->window, nw, q->agg_id, dev_out, dev_cnt, dev_sq, \
       q->_pad == 0 ? 3 : q->_pad); break
   switch (q->func_id) {
     CASE(FN_RATE); CASE(FN_INCREASE); CASE(FN_DELTA); CASE(FN_SUM); CASE(FN_COUNT);
@@ -1341,7 +1355,14 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
     own_cnt = true;
   }
 
-  int partial = (q->agg_id != AGG_NONE && out_counts != nullptr) ? 1 : 0;
+  const bool needs_sq = q->agg_id == AGG_STDDEV || q->agg_id == AGG_STDVAR;
+  int partial = (q->agg_id != AGG_NONE && !is_topk && out_counts != nullptr) ? 1 : 0;
+  if (needs_sq && partial) {
+    fdb_set_error("stddev/stdvar cross-shard partial mode is not supported yet");
+    return FDB_ERR_BADARG;
+  }
+  double* dev_sq = nullptr;
+  if (needs_sq) HIP_CHECK(hipMalloc(&dev_sq, out_len * 8));
   hipEvent_t ev0, ev1;
   HIP_CHECK(hipEventCreate(&ev0));
   HIP_CHECK(hipEventCreate(&ev1));
@@ -1360,6 +1381,7 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
     if (q->agg_id != AGG_NONE && !is_topk) {
       HIP_CHECK(hipMemsetAsync(dev_out, 0, out_len * 8, e->stream));
       HIP_CHECK(hipMemsetAsync(dev_cnt, 0, out_len * 8, e->stream));
+      if (dev_sq) HIP_CHECK(hipMemsetAsync(dev_sq, 0, out_len * 8, e->stream));
       if (q->agg_id == AGG_MIN || q->agg_id == AGG_MAX) {
         fill_f64_kernel<<<(unsigned)((out_len + 255) / 256), 256, 0, e->stream>>>(
             dev_out, out_len, NAN);
@@ -1368,7 +1390,7 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
     bool timed = it >= warmup;
     if (timed) HIP_CHECK(hipEventRecord(ev0, e->stream));
     int32_t rc = launch_scan(e, d, &qscan, is_topk ? topk_grid : dev_out,
-                             is_topk ? nullptr : dev_cnt);
+                             is_topk ? nullptr : dev_cnt, dev_sq);
     if (rc != FDB_OK) return rc;
     if (is_topk) {
       size_t cells = (size_t)q->num_groups * nw;
@@ -1388,10 +1410,11 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   }
   if (q->agg_id != AGG_NONE && !is_topk) {
     agg_present_kernel<<<(unsigned)((out_len + 255) / 256), 256, 0, e->stream>>>(
-        dev_out, dev_cnt, out_len, q->agg_id, partial);
+        dev_out, dev_cnt, dev_sq, out_len, q->agg_id, partial);
     HIP_CHECK(hipGetLastError());
   }
   if (topk_grid) (void)hipFree(topk_grid);
+  if (dev_sq) (void)hipFree(dev_sq);
   HIP_CHECK(hipStreamSynchronize(e->stream));
   (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
 

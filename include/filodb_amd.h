@@ -172,6 +172,10 @@ int32_t fdb_nibblepack_pack_doubles(const double* in, int32_t n, uint8_t* out, i
                               as [G × W × k] doubles holding the series ids
                               (-1 padding). */
 #define FDB_AGG_BOTTOMK 7  /* same, k smallest; values sorted ascending */
+#define FDB_AGG_STDDEV  8  /* StddevRowAggregator.scala:39-58 (merge is
+                              algebraically sum/sumsq/count; presented only —
+                              cross-shard partials are round-2) */
+#define FDB_AGG_STDVAR  9  /* StdvarRowAggregator (same, without the sqrt) */
 
 typedef struct {
   int64_t start;     /* first window end timestamp (ms)                      */

@@ -53,7 +53,7 @@ typedef struct {
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12 };
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
-       AGG_TOPK=6, AGG_BOTTOMK=7 };
+       AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9 };
 
 static inline uint16_t rd_u16(const uint8_t* p) { uint16_t v; memcpy(&v, p, 2); return v; }
 static inline uint32_t rd_u32(const uint8_t* p) { uint32_t v; memcpy(&v, p, 4); return v; }
@@ -646,7 +646,10 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
   /* per-thread partial grids, merged with the same RowAggregator semantics */
   double* pg = (double*)malloc((size_t)T * gridlen * sizeof(double));
   double* pc = (double*)malloc((size_t)T * gridlen * sizeof(double));
-  for (size_t i = 0; i < (size_t)T * gridlen; i++) { pg[i] = NAN; pc[i] = 0; }
+  int needs_sq = q->agg_id == AGG_STDDEV || q->agg_id == AGG_STDVAR;
+  double* pq = needs_sq ? (double*)malloc((size_t)T * gridlen * sizeof(double)) : 0;
+  for (size_t i = 0; i < (size_t)T * gridlen; i++) { pg[i] = NAN; pc[i] = 0; if (pq) pq[i] = 0; }
+  double* sq_total = needs_sq ? (double*)calloc(gridlen, sizeof(double)) : 0;
 
 #ifdef _OPENMP
   #pragma omp parallel
@@ -659,6 +662,7 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
 #endif
     double* g = pg + (size_t)t * gridlen;
     double* gc = pc + (size_t)t * gridlen;
+    double* gq = pq ? pq + (size_t)t * gridlen : 0;
     eval_ctx_t ctx; ctx.scratch = (double*)malloc(64 * 512 * sizeof(double));
     double* row = (double*)malloc((size_t)nw * sizeof(double));
 #ifdef _OPENMP
@@ -694,6 +698,13 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
             if (isnan(acc[w])) acc[w] = 0;
             acc[w] += x;
             break;
+          case AGG_STDDEV: case AGG_STDVAR:
+            /* StddevRowAggregator.scala:39-58: the running merge is
+             * algebraically (sum, sumsq, count) */
+            if (isnan(acc[w])) acc[w] = 0;
+            acc[w] += x;
+            gq[(size_t)grp * nw + w] += x * x;
+            break;
         }
       }
     }
@@ -705,14 +716,17 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
   for (int t = 0; t < T; t++) {
     double* g = pg + (size_t)t * gridlen;
     double* gc = pc + (size_t)t * gridlen;
+    double* gq = pq ? pq + (size_t)t * gridlen : 0;
     for (size_t i = 0; i < gridlen; i++) {
       double x = g[i];
       cnt_total[i] += gc[i];
       if (isnan(x)) continue;
       switch (q->agg_id) {
         case AGG_SUM: case AGG_COUNT: case AGG_AVG:
+        case AGG_STDDEV: case AGG_STDVAR:
           if (isnan(out[i])) out[i] = 0;
           out[i] += x;
+          if (sq_total) sq_total[i] += gq[i];
           break;
         case AGG_MIN:
           if (isnan(out[i]) || x < out[i]) out[i] = x;
@@ -736,8 +750,19 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
     /* present: mean = sum/count (AvgRowAggregator.scala:38-46 algebraically) */
     for (size_t i = 0; i < gridlen; i++)
       if (cnt_total[i] > 0) out[i] = out[i] / cnt_total[i];
+  } else if (needs_sq) {
+    /* present: sqrt(sumsq/n - mean^2) (StddevRowAggregator.scala:49-52) */
+    for (size_t i = 0; i < gridlen; i++) {
+      if (cnt_total[i] > 0) {
+        double mean = out[i] / cnt_total[i];
+        double var = sq_total[i] / cnt_total[i] - mean * mean;
+        out[i] = q->agg_id == AGG_STDDEV ? sqrt(var) : var;
+      }
+    }
   }
   free(pg); free(pc);
+  if (pq) free(pq);
+  if (sq_total) free(sq_total);
   if (!counts) free(cnt_total);
   if (owned_counts) free(owned_counts);
   return 0;

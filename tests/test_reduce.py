@@ -9,7 +9,8 @@ import pytest
 
 from conftest import build_store, synth_gauge_series
 
-AGGS = {"sum": 1, "count": 2, "min": 3, "max": 4, "avg": 5}
+AGGS = {"sum": 1, "count": 2, "min": 3, "max": 4, "avg": 5,
+        "stddev": 8, "stdvar": 9}
 
 
 def make_multi(fdb, rng, n_series=40, n_groups=5, n=60, nan_p=0.15):
@@ -53,8 +54,13 @@ def test_group_reduce_vs_numpy(fdb, oracle, agg):
                 exp = np.nan if len(nn) == 0 else nn.min()
             elif agg == "max":
                 exp = np.nan if len(nn) == 0 else nn.max()
-            else:  # avg
+            elif agg == "avg":
                 exp = np.nan if len(nn) == 0 else nn.mean()
+            elif agg == "stddev":
+                exp = np.nan if len(nn) == 0 else np.sqrt(
+                    (nn * nn).mean() - nn.mean() ** 2)
+            else:  # stdvar
+                exp = np.nan if len(nn) == 0 else (nn * nn).mean() - nn.mean() ** 2
             got = grid[g * nw + w]
             if np.isnan(exp):
                 assert np.isnan(got)
