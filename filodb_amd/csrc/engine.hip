@@ -158,21 +158,26 @@ template <int FUNC> struct KKind { static constexpr int v =
 
 #define FDB_NGROUPS ((FDB_MAX_ROWS_PER_SERIES + 7) / 8)
 
-template <int KIND>
+// two capacity tiers: the common one-chunk-per-span shape (4 waves/block) and
+// a long-lookback tier for series spanning several 400-row chunks
+// (2 waves/block; SURVEY §5: long windows crossing many chunks)
+#define FDB_ROWS_LARGE 1600
+#define FDB_CHUNKS_LARGE 16
+
+template <int KIND, int CAP, int MAXC>
 struct Ws {                         // per-wave LDS workspace
-  int64_t ts[FDB_MAX_ROWS_PER_SERIES];
-  double  val[FDB_MAX_ROWS_PER_SERIES];   // raw values, or inclusive NaN-zeroed
-                                          // prefix sums for the PFX kinds
+  int64_t ts[CAP];
+  double  val[CAP];                 // raw values, or inclusive NaN-zeroed
+                                    // prefix sums for the PFX kinds
   // counter resets are rare: (position, cumulative in-chunk correction) pairs
   // instead of a full corrected[] copy (CorrectingDoubleVectorReader :325-342)
   int16_t dpos[KIND == K_RATE ? FDB_MAX_DROPS : 1];
   double  dcum[KIND == K_RATE ? FDB_MAX_DROPS : 1];
   // non-NaN count prefix (PFX kinds) / change-indicator prefix (CHANGES)
-  uint16_t cnt[(KIND == K_PFX || KIND == K_PFX_SQ || KIND == K_CHANGES)
-               ? FDB_MAX_ROWS_PER_SERIES : 1];
-  double  sq[KIND == K_PFX_SQ ? FDB_MAX_ROWS_PER_SERIES : 1];   // squared prefix
-  double  grp[KIND == K_MINMAX ? (FDB_NGROUPS + FDB_MAX_CHUNKS_PER_SERIES) : 1]; // 8-elem group min/max (chunk-relative, padded per chunk)
-  ChunkMeta cm[FDB_MAX_CHUNKS_PER_SERIES];
+  uint16_t cnt[(KIND == K_PFX || KIND == K_PFX_SQ || KIND == K_CHANGES) ? CAP : 1];
+  double  sq[KIND == K_PFX_SQ ? CAP : 1];   // squared prefix
+  double  grp[KIND == K_MINMAX ? ((CAP + 7) / 8 + MAXC) : 1]; // 8-elem group min/max (chunk-relative, padded per chunk)
+  ChunkMeta cm[MAXC];
   int32_t nchunks;
   int32_t total_rows;
 };
@@ -315,8 +320,8 @@ __device__ double d_extrapolated_rate(int64_t windowStart, int64_t windowEnd, in
 
 // in-chunk correction lookup from the sparse drop table (the step function
 // CorrectingDoubleVectorReader :325-342 materializes as corrected[])
-template <int KIND>
-__device__ __forceinline__ double d_corr_at(const Ws<KIND>& ws, const ChunkMeta& m, int i) {
+template <typename WS>
+__device__ __forceinline__ double d_corr_at(const WS& ws, const ChunkMeta& m, int i) {
   if (m.dense_corr) {
     // serial recompute — only when one chunk held >FDB_MAX_DROPS resets
     double corr = 0, last = -1.7976931348623157e308;
@@ -334,8 +339,8 @@ __device__ __forceinline__ double d_corr_at(const Ws<KIND>& ws, const ChunkMeta&
   }
   return c;
 }
-template <int KIND>
-__device__ __forceinline__ double d_corrected(const Ws<KIND>& ws, const ChunkMeta& m, int i) {
+template <typename WS>
+__device__ __forceinline__ double d_corrected(const WS& ws, const ChunkMeta& m, int i) {
   double x = ws.val[m.row0 + i];
   if (m.dropped) { if (isnan(x)) x = 0; x += d_corr_at(ws, m, i); }
   return x;
@@ -359,10 +364,11 @@ __device__ void atomic_min_max_f64(double* addr, double val, bool is_min) {
 //   decode → per-chunk meta (search slopes, correction table, prefixes) →
 //   per-window evaluation with O(1)-ish row location and O(1) prefix lookups
 // ---------------------------------------------------------------------------
-template <int FUNC>
 // rate family carries more live state (correction meta): forcing 5 waves/SIMD
 // makes it spill; the gauge kinds fit 5 waves cleanly
-__global__ __launch_bounds__(BLOCK_THREADS, (FUNC <= FN_DELTA) ? 4 : 5)
+template <int FUNC, int CAP = FDB_MAX_ROWS_PER_SERIES,
+          int MAXC = FDB_MAX_CHUNKS_PER_SERIES, int WAVES = WAVES_PER_BLOCK>
+__global__ __launch_bounds__(WAVES * 64, (FUNC <= FN_DELTA) ? 4 : 5)
 void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                  const int32_t* __restrict__ series_first,
                  const int32_t* __restrict__ series_nchunks,
@@ -378,11 +384,11 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 {
   constexpr int KIND = KKind<FUNC>::v;
   constexpr bool RATE_FAMILY = (KIND == K_RATE);
-  __shared__ Ws<KIND> ws_all[WAVES_PER_BLOCK];
+  __shared__ Ws<KIND, CAP, MAXC> ws_all[WAVES];
 
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  Ws<KIND>& ws = ws_all[wave];
+  Ws<KIND, CAP, MAXC>& ws = ws_all[wave];
 
   // phase timing (ablation: phase_mask & 8) — wall cycles per wave per phase
   const bool timing = (phase_mask & 8) != 0;
@@ -393,18 +399,18 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   // of one block per 4 series — 250k tiny-workgroup dispatches cost multiple ms
   // of pure launch churn (measured ~20-26ns each), so the loop, not the grid,
   // walks the series
-  for (int sid = blockIdx.x * WAVES_PER_BLOCK + wave; sid < num_series;
-       sid += gridDim.x * WAVES_PER_BLOCK) {
+  for (int sid = blockIdx.x * WAVES + wave; sid < num_series;
+       sid += gridDim.x * WAVES) {
   // ---- decode phase: all chunks of this series into LDS --------------------
   const int first = series_first[sid];
   int nchunks = series_nchunks[sid];
-  if (nchunks > FDB_MAX_CHUNKS_PER_SERIES) nchunks = FDB_MAX_CHUNKS_PER_SERIES;
+  if (nchunks > MAXC) nchunks = MAXC;
   if (lane == 0) { ws.nchunks = nchunks; ws.total_rows = 0; }
 
   int row0 = 0;
   for (int c = 0; c < nchunks; c++) {
     int n = dir.num_rows[first + c];
-    if (row0 + n > FDB_MAX_ROWS_PER_SERIES) n = 0;   // guarded at upload
+    if (row0 + n > CAP) n = 0;   // guarded at upload
     DVec vv;
     if (phase_mask & 1) {
       DVec tv;
@@ -1169,6 +1175,8 @@ struct fdb_dataset {
   int64_t payload_bytes;    // sum of vector bytes (algorithmic HBM footprint)
   int64_t total_samples;
   int max_group;            // max group id seen (for validation)
+  int max_rows;             // max rows in one series (capacity tier selection)
+  int max_chunks;           // max chunks in one series
 };
 
 extern "C" fdb_engine_t* fdb_engine_create(int32_t device) {
@@ -1235,17 +1243,18 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
     payload += (int64_t)tl + 4 + (int64_t)vl + 4;
     samples += dir[i].num_rows;
   }
-  int max_group = 0;
+  int max_group = 0, max_rows = 0, max_chunks = 0;
   for (int32_t sid = 0; sid < view.num_series; sid++) {
     if (view.group_ids[sid] > max_group) max_group = view.group_ids[sid];
-    // round-1 capacity checks
     int total = 0;
     for (int c = 0; c < view.series_nchunks[sid]; c++)
       total += dir[view.series_first[sid] + c].num_rows;
-    if (total > FDB_MAX_ROWS_PER_SERIES || view.series_nchunks[sid] > FDB_MAX_CHUNKS_PER_SERIES) {
-      fdb_set_error("series %d exceeds round-1 capacity (%d rows / %d chunks; caps %d/%d)",
-                    sid, total, view.series_nchunks[sid],
-                    FDB_MAX_ROWS_PER_SERIES, FDB_MAX_CHUNKS_PER_SERIES);
+    if (total > max_rows) max_rows = total;
+    if (view.series_nchunks[sid] > max_chunks) max_chunks = view.series_nchunks[sid];
+    if (total > 1600 || view.series_nchunks[sid] > 16) {
+      fdb_set_error("series %d exceeds capacity (%d rows / %d chunks; caps 1600/16 "
+                    "— longer lookbacks are round-2 streaming work)",
+                    sid, total, view.series_nchunks[sid]);
       return nullptr;
     }
   }
@@ -1257,6 +1266,8 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
   d->payload_bytes = payload;
   d->total_samples = samples;
   d->max_group = max_group;
+  d->max_rows = max_rows;
+  d->max_chunks = max_chunks;
 
   auto upload = [&](void** dst, const void* src, size_t bytes) -> bool {
     if (hipMalloc(dst, bytes ? bytes : 8) != hipSuccess) return false;
@@ -1298,17 +1309,26 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
                            double* dev_out, double* dev_cnt, double* dev_sq) {
   DirSoA dir{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
   int nw = fdb_num_windows(q);
-  int grid = (d->num_series + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+  // capacity tier: the common one-chunk shape runs 4 waves/block; series
+  // spanning several chunks (long lookbacks) use the 1600-row/2-wave variant
+  const bool large = d->max_rows > FDB_MAX_ROWS_PER_SERIES ||
+                     d->max_chunks > FDB_MAX_CHUNKS_PER_SERIES;
+  const int waves = large ? 2 : WAVES_PER_BLOCK;
+  int grid = (d->num_series + waves - 1) / waves;
   // enough blocks to fill every CU at worst-case occupancy, few enough that
   // workgroup dispatch is off the critical path
   int cap = 8192;
   if (const char* g = getenv("FDB_GRID")) cap = atoi(g);   // perf experiments
   if (cap > 0 && grid > cap) grid = cap;
+  #define ARGS d->blob, dir, d->series_first, d->series_nchunks, d->group_ids, \
+      d->num_series, q->start, q->step, q->end, q->window, nw, q->agg_id, \
+      dev_out, dev_cnt, dev_sq, q->_pad == 0 ? 3 : q->_pad
   #define CASE(F) case F: \
-    hipLaunchKernelGGL(scan_kernel<F>, dim3(grid), dim3(BLOCK_THREADS), 0, e->stream, \
-      d->blob, dir, d->series_first, d->series_nchunks, d->group_ids, d->num_series, \
-      q->start, q->step, q->end, q->window, nw, q->agg_id, dev_out, dev_cnt, dev_sq, \
-      q->_pad == 0 ? 3 : q->_pad); break
+    if (large) hipLaunchKernelGGL((scan_kernel<F, FDB_ROWS_LARGE, FDB_CHUNKS_LARGE, 2>), \
+                                  dim3(grid), dim3(128), 0, e->stream, ARGS); \
+    else hipLaunchKernelGGL((scan_kernel<F>), dim3(grid), dim3(BLOCK_THREADS), 0, \
+                            e->stream, ARGS); \
+    break
   switch (q->func_id) {
     CASE(FN_RATE); CASE(FN_INCREASE); CASE(FN_DELTA); CASE(FN_SUM); CASE(FN_COUNT);
     CASE(FN_AVG); CASE(FN_MIN); CASE(FN_MAX); CASE(FN_STDDEV); CASE(FN_STDVAR);
@@ -1316,6 +1336,7 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
     default: fdb_set_error("bad func_id %d", q->func_id); return FDB_ERR_BADARG;
   }
   #undef CASE
+  #undef ARGS
   HIP_CHECK(hipGetLastError());
   return FDB_OK;
 }

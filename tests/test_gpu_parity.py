@@ -211,3 +211,18 @@ def test_device_output_pointer(fdb, oracle, engine):
     want_s, want_c = oracle.query_exec(st.view(), q, st.num_series, nw, out_counts=True)
     np.testing.assert_allclose(t_sum.cpu().numpy(), want_s, rtol=1e-9, equal_nan=True)
     np.testing.assert_array_equal(t_cnt.cpu().numpy(), want_c)
+
+
+def test_long_lookback_large_capacity(fdb, oracle, engine):
+    """Series spanning several 400-row chunks route to the 1600-row kernel
+    tier (long lookbacks); parity must hold there too."""
+    st = counter_store(fdb, n_series=64, n=800, seed=31, reset_p=0.01,
+                       chunking=(400, 400))
+    q = mkq(fdb, FUNCS["rate"], end=Q["start"] + 700 * 15000)
+    got, want = run_both(fdb, oracle, engine, st, q)
+    check(got, want)
+    st2 = gauge_store(fdb, n_series=64, n=800, seed=32, nan_p=0.1,
+                      chunking=(300, 300, 200))
+    q2 = mkq(fdb, FUNCS["avg"], end=Q["start"] + 700 * 15000)
+    got2, want2 = run_both(fdb, oracle, engine, st2, q2)
+    check(got2, want2)
