@@ -1177,6 +1177,7 @@ struct fdb_dataset {
   int max_group;            // max group id seen (for validation)
   int max_rows;             // max rows in one series (capacity tier selection)
   int max_chunks;           // max chunks in one series
+  int has_hist;             // dataset holds sect-delta histogram vectors
 };
 
 extern "C" fdb_engine_t* fdb_engine_create(int32_t device) {
@@ -1244,6 +1245,17 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
     samples += dir[i].num_rows;
   }
   int max_group = 0, max_rows = 0, max_chunks = 0;
+  int has_hist = 0, has_scalar = 0;
+  for (int64_t i2 = 0; i2 < nc; i2++) {
+    uint16_t wf;
+    memcpy(&wf, view.blob + dir[i2].val_off + 4, 2);
+    if (wf == FDB_WF_HIST_SECTDELTA) has_hist = 1; else has_scalar = 1;
+  }
+  if (has_hist && has_scalar) {
+    fdb_set_error("mixed histogram and scalar series in one store are not "
+                  "supported — upload them as separate datasets");
+    return nullptr;
+  }
   for (int32_t sid = 0; sid < view.num_series; sid++) {
     if (view.group_ids[sid] > max_group) max_group = view.group_ids[sid];
     int total = 0;
@@ -1268,6 +1280,7 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
   d->max_group = max_group;
   d->max_rows = max_rows;
   d->max_chunks = max_chunks;
+  d->has_hist = has_hist;
 
   auto upload = [&](void** dst, const void* src, size_t bytes) -> bool {
     if (hipMalloc(dst, bytes ? bytes : 8) != hipSuccess) return false;
@@ -1345,6 +1358,10 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
                          double* out, double* out_counts, int32_t out_on_device,
                          int32_t warmup, int32_t iters, double* avg_ms) {
   HIP_CHECK(hipSetDevice(e->device));
+  if (d->has_hist) {
+    fdb_set_error("histogram dataset: use fdb_query_exec_hist");
+    return FDB_ERR_BADARG;
+  }
   int nw = fdb_num_windows(q);
   if (nw <= 0) { fdb_set_error("bad window params"); return FDB_ERR_BADARG; }
   const bool is_topk = q->agg_id == AGG_TOPK || q->agg_id == AGG_BOTTOMK;
@@ -1459,6 +1476,15 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
   int nw = fdb_num_windows(q);
   if (nw <= 0 || q->num_groups <= 0) { fdb_set_error("bad hist query params"); return FDB_ERR_BADARG; }
   if (nb < 1 || nb > 64) { fdb_set_error("num_buckets must be 1..64"); return FDB_ERR_BADARG; }
+  if (!d->has_hist) {
+    fdb_set_error("not a histogram dataset");
+    return FDB_ERR_BADARG;
+  }
+  if (d->max_chunks > 1) {
+    fdb_set_error("histogram series spanning multiple chunks are round-2 work "
+                  "(streaming element walk)");
+    return FDB_ERR_BADARG;
+  }
   if (q->window / q->step + 2 > FDB_HIST_RING) {
     fdb_set_error("window/step ratio %lld exceeds round-1 ring capacity %d",
                   (long long)(q->window / q->step), FDB_HIST_RING);
