@@ -163,7 +163,10 @@ class ChunkStore:
 
     def __del__(self):
         if getattr(self, "_h", None):
-            lib().fdb_store_destroy(self._h)
+            try:
+                lib().fdb_store_destroy(self._h)
+            except (TypeError, AttributeError):
+                pass  # interpreter shutdown: ctypes globals already torn down
             self._h = None
 
     def set_max_rows(self, n):
@@ -235,7 +238,10 @@ class Engine:
 
     def __del__(self):
         if getattr(self, "_h", None):
-            lib().fdb_engine_destroy(self._h)
+            try:
+                lib().fdb_engine_destroy(self._h)
+            except (TypeError, AttributeError):
+                pass
             self._h = None
 
     def upload(self, store: ChunkStore):
@@ -286,7 +292,10 @@ class Dataset:
 
     def __del__(self):
         if getattr(self, "_h", None):
-            lib().fdb_dataset_destroy(self._h)
+            try:
+                lib().fdb_dataset_destroy(self._h)
+            except (TypeError, AttributeError):
+                pass
             self._h = None
 
     @property

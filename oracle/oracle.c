@@ -311,12 +311,18 @@ typedef struct {
   const vec_t* v;
   double* corrected;      /* NULL when chunk has no drop bit */
   double chunk_correction; /* _correction total */
+  int owned;              /* corrected was malloc'd (chunks > 512 rows) */
 } cread_t;
 
 static void cread_init(const vec_t* v, double* scratch, cread_t* r) {
   r->v = v;
   r->chunk_correction = 0;
+  r->owned = 0;
   if (!v->dropped) { r->corrected = 0; return; }
+  if (v->n > 512) {            /* eval_series scratch is 512 rows per chunk */
+    scratch = (double*)malloc((size_t)v->n * sizeof(double));
+    r->owned = 1;
+  }
   r->corrected = scratch;
   double corr = 0, last = -1.7976931348623157e308; /* Double.MinValue */
   for (int i = 0; i < v->n; i++) {
@@ -558,6 +564,9 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
     }
     out[w] = result;
   }
+  if (is_rate_family)
+    for (int c = 0; c < nchunks; c++)
+      if (cr[c].owned) free(cr[c].corrected);
 }
 
 /* =========================================================================
