@@ -60,6 +60,11 @@ WORKLOADS = {
                       window=300_000, qstep=15000, n_groups=10,
                       label="100k hist series x 64 buckets, "
                             "histogram_quantile(0.99, sum(rate()[5m])) by group"),
+    # long-lookback tier exercise: 3h@15s spans two 400-row chunks/series
+    "rate_long": dict(kind="counter", n_series=400_000, n_samples=720,
+                      step_ms=15000, func=fdb.FN_RATE, agg=fdb.AGG_NONE,
+                      window=300_000, qstep=15000, n_groups=1000,
+                      label="400k counter series 3h@15s (2 chunks/series), rate()[5m]"),
     "smoke": dict(kind="counter", n_series=20_000, n_samples=240,
                   step_ms=15000, func=fdb.FN_RATE, agg=fdb.AGG_NONE,
                   window=300_000, qstep=15000, n_groups=100,

@@ -1,0 +1,115 @@
+"""Property-based round-trip and parity fuzzing (hypothesis).
+
+Mirrors the reference's ScalaCheck layer (EncodingPropertiesTest.scala): any
+valid sample stream must encode → decode bit-exactly, and the chunked window
+engine must match the naive recomputation for arbitrary windows.
+"""
+import numpy as np
+from hypothesis import given, settings, strategies as st
+
+from conftest import build_store
+
+import filodb_amd as fdb_mod
+
+
+def _oracle():
+    import sys
+    import os
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "oracle"))
+    import pyclient
+    return pyclient
+
+
+ts_strategy = st.lists(st.integers(min_value=0, max_value=10_000), min_size=3,
+                       max_size=120)
+val_strategy = st.lists(
+    st.one_of(st.floats(min_value=-1e12, max_value=1e12, allow_nan=False),
+              st.integers(min_value=-10**12, max_value=10**12).map(float),
+              st.just(float("nan"))),
+    min_size=3, max_size=120)
+
+
+@settings(max_examples=60, deadline=None)
+@given(deltas=ts_strategy, vals=val_strategy)
+def test_encode_decode_roundtrip(deltas, vals):
+    """Arbitrary nondecreasing timestamps + arbitrary doubles (NaNs included)
+    survive the encoder bit-exactly — whatever encoding gets chosen, EXCEPT
+    the approx-const timestamp loss the reference accepts by design
+    (DeltaDeltaVector.scala:46-47): when that triggers, decoded timestamps lie
+    on the slope line within ±250."""
+    oracle = _oracle()
+    n = min(len(deltas), len(vals))
+    ts = np.cumsum(np.array(deltas[:n], dtype=np.int64)) + 100_000
+    vs = np.array(vals[:n], dtype=np.float64)
+    st_ = build_store(fdb_mod, [[[(int(t), float(v)) for t, v in zip(ts, vs)]]],
+                      kind=fdb_mod.COL_GAUGE, max_rows=400)
+    tsb, vab, nrows, t0, t1 = st_.chunk(0, 0)
+    assert nrows == n and t0 == ts[0] and t1 == ts[-1]
+    dec_ts = oracle.decode_longs(tsb)
+    info = oracle.vec_info(tsb)
+    if info["wf"] == 0x0608 and not np.array_equal(dec_ts, ts):
+        assert np.abs(dec_ts - ts).max() <= 250   # approx-const band
+        assert dec_ts[0] == ts[0]
+    else:
+        np.testing.assert_array_equal(dec_ts, ts)
+    dec_vs = oracle.decode_doubles(vab)
+    np.testing.assert_array_equal(
+        dec_vs.view(np.uint64), vs.view(np.uint64))  # bit-exact incl. NaNs
+
+
+@settings(max_examples=40, deadline=None)
+@given(deltas=ts_strategy,
+       incs=st.lists(st.integers(min_value=0, max_value=1000), min_size=3,
+                     max_size=120),
+       resets=st.lists(st.booleans(), min_size=3, max_size=120),
+       window=st.integers(min_value=1, max_value=20_000),
+       step=st.integers(min_value=1, max_value=5_000))
+def test_rate_window_parity_fuzz(deltas, incs, resets, window, step):
+    """Counter series with arbitrary resets: the oracle's windowed rate equals
+    a naive recomputation (corrected first/last + extrapolatedRate) for every
+    window — the oracle is the GPU's reference, the naive model is its check."""
+    oracle = _oracle()
+    n = min(len(deltas), len(incs), len(resets))
+    if n < 3:
+        return
+    ts = np.cumsum(np.array(deltas[:n], dtype=np.int64) + 1) + 100_000
+    vals = np.zeros(n)
+    cur = 0.0
+    for i in range(n):
+        if resets[i]:
+            cur = 0.0
+        cur += incs[i]
+        vals[i] = cur
+    st_ = build_store(fdb_mod, [[[(int(t), float(v)) for t, v in zip(ts, vals)]]],
+                      kind=fdb_mod.COL_COUNTER)
+    tsb, _, _, _, _ = st_.chunk(0, 0)
+    enc_ts = oracle.decode_longs(tsb)
+    start = int(ts[0])
+    end = int(ts[-1]) + step
+    q = fdb_mod.make_query(start, step, end, window, fdb_mod.FN_RATE)
+    out = oracle.eval_series(st_.view(), 0, q, q.num_windows)
+
+    # naive corrected series (CorrectingDoubleVectorReader semantics)
+    corrected = vals.copy()
+    corr = 0.0
+    last = -np.inf
+    for i in range(n):
+        x = vals[i]
+        if x < last:
+            corr += last
+        corrected[i] = x + corr
+        last = x
+    for w in range(q.num_windows):
+        w_end = start + w * step
+        w_start = w_end - window
+        m = (enc_ts >= w_start) & (enc_ts <= w_end)
+        if m.sum() < 2 or enc_ts[m][-1] <= enc_ts[m][0]:
+            assert np.isnan(out[w]), w
+            continue
+        idx = np.nonzero(m)[0]
+        expect = oracle.extrapolated_rate(
+            int(w_start), int(w_end), int(m.sum()),
+            int(enc_ts[idx[0]]), float(corrected[idx[0]]),
+            int(enc_ts[idx[-1]]), float(corrected[idx[-1]]), True, True)
+        assert out[w] == expect or abs(out[w] - expect) <= 1e-9 * abs(expect), w
