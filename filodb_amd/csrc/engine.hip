@@ -394,6 +394,13 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   const bool timing = (phase_mask & 8) != 0;
   uint64_t t_decode = 0, t_meta = 0, t_win = 0, tt0 = 0;
   if (timing) tt0 = __builtin_amdgcn_s_memtime();
+  // residency census (ablation: phase_mask & 16): concurrent-block high-water
+  // in out_cnt[0..1] (u64 counters)
+  if ((phase_mask & 16) && threadIdx.x == 0) {
+    unsigned long long cur =
+        atomicAdd((unsigned long long*)&out_cnt[0], 1ULL) + 1;
+    atomicMax((unsigned long long*)&out_cnt[1], cur);
+  }
 
   // grid-stride over series: a few resident blocks loop over the shard instead
   // of one block per 4 series — 250k tiny-workgroup dispatches cost multiple ms
@@ -889,6 +896,8 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   __builtin_amdgcn_wave_barrier();
   if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_win += t - tt0; tt0 = t; }
   }  // series grid-stride loop
+  if ((phase_mask & 16) && threadIdx.x == 0)
+    atomicAdd((unsigned long long*)&out_cnt[0], (unsigned long long)-1ll);
   if (timing && lane == 0) {
     size_t gw = (size_t)blockIdx.x * WAVES_PER_BLOCK + wave;
     out[gw * 4 + 0] = (double)t_decode;

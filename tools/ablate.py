@@ -28,6 +28,24 @@ def main():
     ds = eng.upload(st)
     func = fdb.FN_RATE if workload == "rate" else fdb.FN_AVG_OVER_TIME
     window = 300_000 if workload == "rate" else 600_000
+    if len(sys.argv) > 3 and sys.argv[3] == "census":
+        # concurrent-block high-water during a full launch
+        q = fdb.make_query(T0, 15000, T0 + 240 * 15000, window, func)
+        q._pad = 3 | 16
+        out = torch.zeros(n_series * q.num_windows, dtype=torch.float64, device="cuda")
+        cnt = torch.zeros(1024, dtype=torch.float64, device="cuda")
+        nw = q.num_windows
+        import ctypes
+        rc = fdb.lib().fdb_query_exec(eng._h, ds._h, ctypes.byref(q),
+                                      fdb._as_f64_ptr(out), fdb._as_f64_ptr(cnt), 1)
+        assert rc == 0, fdb.lib().fdb_last_error()
+        eng.synchronize()
+        vals = cnt[:2].cpu().numpy().view('uint64') if hasattr(cnt[:2].cpu().numpy(), 'view') else None
+        import numpy as np
+        raw = cnt[:2].cpu().numpy()
+        u = raw.view(np.uint64)
+        print(f"census: final_active={u[0]} peak_blocks={u[1]}")
+        return
     if len(sys.argv) > 3 and sys.argv[3] == "time":
         # phase-timing mode: kernel reports wall cycles per wave per phase
         q = fdb.make_query(T0, 15000, T0 + 240 * 15000, window, func)
