@@ -39,7 +39,7 @@ void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
 // capacity limits (round 1)
 // ---------------------------------------------------------------------------
 #define FDB_MAX_ROWS_PER_SERIES 400   // LDS-resident rows per series (= chunk row cap)
-#define FDB_MAX_CHUNKS_PER_SERIES 12
+#define FDB_MAX_CHUNKS_PER_SERIES 8   /* small tier; longer series use the large tier */
 #define WAVES_PER_BLOCK 4
 #define BLOCK_THREADS (WAVES_PER_BLOCK * 64)
 
