@@ -932,6 +932,15 @@ __device__ __forceinline__ int64_t d_fdiv(int64_t a, int64_t b) {
   return a >= 0 ? a / b : -((-a + b - 1) / b);
 }
 
+// floor(a / b) via double reciprocal + exact integer fixup — i64 division is
+// software-emulated (hundreds of cycles); this is 2 multiplies + a short walk
+__device__ __forceinline__ int64_t d_fdiv_fast(int64_t a, int64_t b, double inv_b) {
+  int64_t w = (int64_t)floor((double)a * inv_b);
+  while ((w + 1) * b <= a) w++;
+  while (w * b > a) w--;
+  return w;
+}
+
 // wave-wide inclusive prefix sum over i64 (bucket-cumulative reconstruction)
 __device__ __forceinline__ int64_t wave_incl_scan_i64(int64_t x, int lane) {
   for (int off = 1; off < 64; off <<= 1) {
@@ -983,6 +992,7 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     int sect_left = 0;        // elements left in current section
     int sect_first = 0;       // next element is the section's base element
     const uint8_t* ep = sp;   // element cursor
+    const double inv_step = 1.0 / (double)qstep;
 
     for (int e = 0; e < n; e++) {
       if (sect_left == 0) {   // enter next section
@@ -1039,12 +1049,13 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 
       // window triggers (inversion of the row-range search)
       const int64_t ts_e = ws.ts[e];
-      const int64_t ts_prev = e > 0 ? ws.ts[e - 1] : INT64_MIN / 4;
-      const int64_t ts_next = e + 1 < n ? ws.ts[e + 1] : INT64_MAX / 4;
+      // sentinels kept near the data so the double conversion stays exact
+      const int64_t ts_prev = e > 0 ? ws.ts[e - 1] : ts_e - ((int64_t)1 << 40);
+      const int64_t ts_next = e + 1 < n ? ws.ts[e + 1] : ts_e + ((int64_t)1 << 40);
       // windows starting at e: wStart in (ts_prev, ts_e]
-      // (clamp in i64 BEFORE narrowing: the e=0 sentinel is a huge negative)
-      int64_t ws_lo64 = d_fdiv(ts_prev - qstart + qwindow, qstep) + 1;
-      int64_t ws_hi64 = d_fdiv(ts_e - qstart + qwindow, qstep);
+      // (clamp in i64 BEFORE narrowing)
+      int64_t ws_lo64 = d_fdiv_fast(ts_prev - qstart + qwindow, qstep, inv_step) + 1;
+      int64_t ws_hi64 = d_fdiv_fast(ts_e - qstart + qwindow, qstep, inv_step);
       if (ws_lo64 < 0) ws_lo64 = 0;
       if (ws_hi64 > num_windows - 1) ws_hi64 = num_windows - 1;
       int ws_lo = (int)ws_lo64, ws_hi = (int)ws_hi64;
@@ -1054,9 +1065,10 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         if (lane == 0) { ws.ring_w[slot] = w; ws.ring_e[slot] = e; }
       }
       // windows ending at e: wEnd in [ts_e, ts_next)
-      int64_t we_lo64 = d_fdiv(ts_e - qstart + qstep - 1, qstep);
-      int64_t we_hi64 = e + 1 < n ? d_fdiv(ts_next - qstart + qstep - 1, qstep) - 1
-                                  : num_windows - 1;
+      int64_t we_lo64 = d_fdiv_fast(ts_e - qstart + qstep - 1, qstep, inv_step);
+      int64_t we_hi64 = e + 1 < n
+          ? d_fdiv_fast(ts_next - qstart + qstep - 1, qstep, inv_step) - 1
+          : num_windows - 1;
       if (we_lo64 < 0) we_lo64 = 0;
       if (we_hi64 > num_windows - 1) we_hi64 = num_windows - 1;
       int we_lo = (int)we_lo64, we_hi = (int)we_hi64;
