@@ -58,6 +58,53 @@ def test_encode_decode_roundtrip(deltas, vals):
         dec_vs.view(np.uint64), vs.view(np.uint64))  # bit-exact incl. NaNs
 
 
+@settings(max_examples=50, deadline=None)
+@given(deltas=ts_strategy, vals=val_strategy,
+       nchunks=st.integers(min_value=1, max_value=4),
+       window=st.integers(min_value=1, max_value=20_000),
+       step=st.integers(min_value=1, max_value=5_000),
+       func=st.sampled_from(["sum", "count", "avg", "min", "max",
+                             "stddev", "stdvar", "changes", "last"]))
+def test_gauge_window_parity_fuzz(deltas, vals, nchunks, window, step, func):
+    """Gauge family under arbitrary samples/NaNs/chunking/windows: the oracle's
+    chunked evaluation equals the naive per-window model (the same one
+    test_oracle_windows pins against AggrOverTimeFunctionsSpec.scala:289-400),
+    including the NaN-poison sum, started-count and chunk-boundary semantics."""
+    from test_oracle_windows import naive_window, FUNC_IDS
+    oracle = _oracle()
+    n = min(len(deltas), len(vals))
+    if n < 3:
+        return
+    ts = np.cumsum(np.array(deltas[:n], dtype=np.int64) + 1) + 100_000
+    vs = np.array(vals[:n], dtype=np.float64)
+    per = max(1, n // nchunks)
+    chunks, bounds = [], []
+    for c in range(nchunks):
+        lo = c * per
+        hi = n if c == nchunks - 1 else min(n, (c + 1) * per)
+        if lo >= hi:
+            break
+        chunks.append([(int(ts[i]), float(vs[i])) for i in range(lo, hi)])
+        bounds.append((lo, hi))
+    st_ = build_store(fdb_mod, [chunks], kind=fdb_mod.COL_GAUGE, max_rows=400)
+    # use STORED timestamps (approx-const encoding may shift them within ±250)
+    dec = np.concatenate([oracle.decode_longs(st_.chunk(0, c)[0])
+                          for c in range(len(chunks))])
+    start = int(dec[0])
+    end = int(dec[-1]) + step
+    q = fdb_mod.make_query(start, step, end, window, FUNC_IDS[func])
+    out = oracle.eval_series(st_.view(), 0, q, q.num_windows)
+    for w in range(q.num_windows):
+        w_end = start + w * step
+        expected = naive_window(dec, vs, bounds, w_end - window, w_end, func)
+        got = out[w]
+        if np.isnan(expected):
+            assert np.isnan(got), (func, w)
+        else:
+            assert got == expected or abs(got - expected) <= 1e-9 * abs(expected), \
+                (func, w, got, expected)
+
+
 @settings(max_examples=40, deadline=None)
 @given(deltas=ts_strategy,
        incs=st.lists(st.integers(min_value=0, max_value=1000), min_size=3,
