@@ -1171,6 +1171,87 @@ __global__ void topk_kernel(const double* __restrict__ grid,
   }
 }
 
+// cross-series group reduction from a per-series [S×W] grid. Replaces the
+// contended-atomic accumulation path: the scan writes plain per-series window
+// results (identical to AGG_NONE) and this kernel folds each group's
+// contiguous members from the group-sorted index — one coalesced read of the
+// grid instead of 2-3 f64 RMWs per (series, window) on a hot [G×W] grid.
+// Semantics: RangeVectorAggregator.fastReduce (AggrOverRangeVectors.scala:
+// 320-377) with the RowAggregator map/reduce rules (NaN rows skipped; MIN/MAX
+// stay NaN until a value arrives). Raw sums+counts out — agg_present_kernel
+// applies the presentation step, as the atomic path did.
+template <int VARIANT>
+__global__ void group_reduce_kernel(const double* __restrict__ grid,
+                                    const int32_t* __restrict__ sbg,
+                                    const int32_t* __restrict__ goff,
+                                    int ng, int nw, int agg_id,
+                                    double* __restrict__ out,
+                                    double* __restrict__ cnt,
+                                    double* __restrict__ sq) {
+  size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (size_t)ng * nw) return;
+  int g = (int)(idx / nw), w = (int)(idx % nw);
+  // consecutive threads share g and walk consecutive w, so each member's row
+  // grid[s*nw + w..] is read coalesced across the wave
+  const int i0 = goff[g], i1 = goff[g + 1];
+  if (agg_id == AGG_MIN || agg_id == AGG_MAX) {
+    const bool is_min = agg_id == AGG_MIN;
+    double acc = NAN, c = 0.0;
+    for (int i = i0; i < i1; i++) {
+      double x = grid[(size_t)sbg[i] * nw + w];
+      if (isnan(x)) continue;
+      c += 1.0;
+      acc = isnan(acc) || (is_min ? x < acc : x > acc) ? x : acc;
+    }
+    out[idx] = acc;
+    cnt[idx] = c;
+    if (sq) sq[idx] = 0.0;
+    return;
+  }
+  if (VARIANT == 0) {
+    double acc = 0.0, c = 0.0, s2 = 0.0;
+    for (int i2 = i0; i2 < i1; i2++) {
+      double x = grid[(size_t)sbg[i2] * nw + w];
+      if (isnan(x)) continue;
+      c += 1.0;
+      acc += x;
+      s2 += x * x;
+    }
+    out[idx] = agg_id == AGG_COUNT ? c : acc;
+    cnt[idx] = c;
+    if (sq) sq[idx] = s2;
+    return;
+  }
+  // sum-shaped reductions: 4 independent accumulator chains keep 4 loads in
+  // flight (a single chain leaves the loop latency-bound far below the HBM
+  // rate). Reordering the member sum is fine: the atomic path this replaced
+  // already accumulated in nondeterministic order.
+  double a0 = 0, a1 = 0, a2 = 0, a3 = 0;
+  double c0 = 0, c1 = 0, q0 = 0, q1 = 0;
+  int i = i0;
+  for (; i + 3 < i1; i += 4) {
+    double x0 = grid[(size_t)sbg[i] * nw + w];
+    double x1 = grid[(size_t)sbg[i + 1] * nw + w];
+    double x2 = grid[(size_t)sbg[i + 2] * nw + w];
+    double x3 = grid[(size_t)sbg[i + 3] * nw + w];
+    bool n0 = !isnan(x0), n1 = !isnan(x1), n2 = !isnan(x2), n3 = !isnan(x3);
+    a0 += n0 ? x0 : 0.0; a1 += n1 ? x1 : 0.0;
+    a2 += n2 ? x2 : 0.0; a3 += n3 ? x3 : 0.0;
+    c0 += (double)n0 + (double)n2; c1 += (double)n1 + (double)n3;
+    q0 += n0 ? x0 * x0 : 0.0; q1 += n1 ? x1 * x1 : 0.0;
+    q0 += n2 ? x2 * x2 : 0.0; q1 += n3 ? x3 * x3 : 0.0;
+  }
+  for (; i < i1; i++) {
+    double x = grid[(size_t)sbg[i] * nw + w];
+    if (isnan(x)) continue;
+    a0 += x; c0 += 1.0; q0 += x * x;
+  }
+  double c = c0 + c1;
+  out[idx] = agg_id == AGG_COUNT ? c : (a0 + a1) + (a2 + a3);
+  cnt[idx] = c;
+  if (sq) sq[idx] = q0 + q1;
+}
+
 __global__ void fill_f64_kernel(double* p, size_t n, double v) {
   size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) p[i] = v;
@@ -1426,37 +1507,48 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   HIP_CHECK(hipEventCreate(&ev0));
   HIP_CHECK(hipEventCreate(&ev1));
 
-  // topk: the scan fills an internal [S×W] grid, the presenter reduces it
-  double* topk_grid = nullptr;
+  // every agg runs two-phase: the scan fills an internal per-series [S×W]
+  // grid with plain stores, then a presenter (topk_kernel / group_reduce_
+  // kernel) folds it along the group-sorted index — no data-path atomics
+  double* per_grid = nullptr;
   fdb_query_t qscan = *q;
-  if (is_topk) {
-    HIP_CHECK(hipMalloc(&topk_grid, (size_t)d->num_series * nw * 8));
+  if (q->agg_id != AGG_NONE) {
+    HIP_CHECK(hipMalloc(&per_grid, (size_t)d->num_series * nw * 8));
     qscan.agg_id = AGG_NONE;
   }
 
   int total_runs = warmup + iters;
   float ms_sum = 0;
   for (int it = 0; it < total_runs; it++) {
-    if (q->agg_id != AGG_NONE && !is_topk) {
-      HIP_CHECK(hipMemsetAsync(dev_out, 0, out_len * 8, e->stream));
-      HIP_CHECK(hipMemsetAsync(dev_cnt, 0, out_len * 8, e->stream));
-      if (dev_sq) HIP_CHECK(hipMemsetAsync(dev_sq, 0, out_len * 8, e->stream));
-      if (q->agg_id == AGG_MIN || q->agg_id == AGG_MAX) {
-        fill_f64_kernel<<<(unsigned)((out_len + 255) / 256), 256, 0, e->stream>>>(
-            dev_out, out_len, NAN);
-      }
-    }
     bool timed = it >= warmup;
     if (timed) HIP_CHECK(hipEventRecord(ev0, e->stream));
-    int32_t rc = launch_scan(e, d, &qscan, is_topk ? topk_grid : dev_out,
-                             is_topk ? nullptr : dev_cnt, dev_sq);
+    int32_t rc = launch_scan(e, d, &qscan, per_grid ? per_grid : dev_out,
+                             per_grid ? nullptr : dev_cnt, nullptr);
     if (rc != FDB_OK) return rc;
     if (is_topk) {
       size_t cells = (size_t)q->num_groups * nw;
       topk_kernel<<<(unsigned)((cells + 255) / 256), 256, 0, e->stream>>>(
-          topk_grid, d->series_by_group, d->group_offsets,
+          per_grid, d->series_by_group, d->group_offsets,
           q->num_groups, nw, kk, q->agg_id == AGG_TOPK ? 1 : 0,
           dev_out, dev_cnt);
+      HIP_CHECK(hipGetLastError());
+    } else if (q->agg_id != AGG_NONE) {
+      size_t cells = (size_t)q->num_groups * nw;
+      static int variant = -1;
+      if (variant < 0) {
+        const char* v = getenv("FDB_REDUCE_VARIANT");   // perf experiments
+        variant = v ? atoi(v) : 1;
+      }
+      if (variant == 1)
+        group_reduce_kernel<1><<<(unsigned)((cells + 255) / 256), 256, 0,
+                                 e->stream>>>(
+            per_grid, d->series_by_group, d->group_offsets,
+            q->num_groups, nw, q->agg_id, dev_out, dev_cnt, dev_sq);
+      else
+        group_reduce_kernel<0><<<(unsigned)((cells + 255) / 256), 256, 0,
+                                 e->stream>>>(
+            per_grid, d->series_by_group, d->group_offsets,
+            q->num_groups, nw, q->agg_id, dev_out, dev_cnt, dev_sq);
       HIP_CHECK(hipGetLastError());
     }
     if (timed) {
@@ -1472,7 +1564,7 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
         dev_out, dev_cnt, dev_sq, out_len, q->agg_id, partial);
     HIP_CHECK(hipGetLastError());
   }
-  if (topk_grid) (void)hipFree(topk_grid);
+  if (per_grid) (void)hipFree(per_grid);
   if (dev_sq) (void)hipFree(dev_sq);
   HIP_CHECK(hipStreamSynchronize(e->stream));
   (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
