@@ -919,13 +919,16 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 //   HistSumRowAggregator.scala:20-29.
 // ---------------------------------------------------------------------------
 #define FDB_HIST_RING 24
+#define FDB_HIST_MAX_CHUNKS 4
 #define HIST_WAVES 2
 
 struct HistWs {
-  int64_t ts[FDB_MAX_ROWS_PER_SERIES];
+  int64_t ts[FDB_MAX_ROWS_PER_SERIES];  // current chunk's timestamps
   double  ring_lo[FDB_HIST_RING][64];   // lowestValue per bucket per active window
+  int64_t ring_t1[FDB_HIST_RING];       // lowestTime (start element's timestamp)
   int32_t ring_w[FDB_HIST_RING];        // window id the slot holds (-1 none)
-  int32_t ring_e[FDB_HIST_RING];        // start element index
+  int32_t ring_e[FDB_HIST_RING];        // start element index (series-global)
+  int32_t ring_c0[FDB_HIST_RING];       // chunk index the window started in
 };
 
 __device__ __forceinline__ int64_t d_fdiv(int64_t a, int64_t b) {
@@ -950,6 +953,14 @@ __device__ __forceinline__ int64_t wave_incl_scan_i64(int64_t x, int lane) {
   return x;
 }
 
+// MAXC = max chunks per series this instantiation handles. The walk streams
+// chunks in time order carrying a running per-bucket correction C; per-window
+// values subtract C0[c0] (C at entry of the window's FIRST chunk) so results
+// equal the reference's per-window CorrectionMeta that starts NoCorrection at
+// that chunk (RangeFunction.scala:138-165). Chunk-boundary drop detection is
+// Histogram.compare's top-bucket-down lexicographic order (Histogram.scala:
+// 204-214) via ballot + highest-differing-lane broadcast.
+template <int MAXC>
 __global__ __launch_bounds__(HIST_WAVES * 64, 4)
 void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                       const int32_t* __restrict__ series_first,
@@ -968,31 +979,48 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 
   for (int sid = blockIdx.x * HIST_WAVES + wave; sid < num_series;
        sid += gridDim.x * HIST_WAVES) {
-    if (series_nchunks[sid] != 1) continue;            // guarded at exec
-    const int ci = series_first[sid];
+    const int nchunks = series_nchunks[sid];
+    if (nchunks < 1 || nchunks > MAXC) continue;       // guarded at exec
+    const int first = series_first[sid];
+
+    for (int r = lane; r < FDB_HIST_RING; r += 64) ws.ring_w[r] = -1;
+
+    const int b = lane;                                 // bucket owned by lane
+    const bool live = b < nb;
+    double raw = 0, C = 0, base = 0;
+    double C0[MAXC];            // C at entry of each chunk (boundary drop incl.)
+    double prevLast = 0;        // raw apply(len-1) of the previous chunk
+    int64_t prev_ts = 0;        // last timestamp of the previous chunk
+    int EG0 = 0;                // series-global index of current chunk's row 0
+    const int grp = group_ids[sid];
+    const double inv_step = 1.0 / (double)qstep;
+
+    for (int c = 0; c < nchunks; c++) {
+    const int ci = first + c;
     const int n = dir.num_rows[ci];
-    if (n <= 0 || n > FDB_MAX_ROWS_PER_SERIES) continue;
+    if (n <= 0 || n > FDB_MAX_ROWS_PER_SERIES) break;  // guarded at exec
 
     DVec tv;
     d_vec_open(blob + dir.ts_off[ci], &tv);
     d_decode_chunk<false>(tv, n, ws.ts, nullptr, lane);
-    for (int r = lane; r < FDB_HIST_RING; r += 64) ws.ring_w[r] = -1;
     __builtin_amdgcn_s_waitcnt(0);
     __builtin_amdgcn_wave_barrier();
+
+    // first timestamp of the next chunk: closes windows ending in the gap
+    int64_t next_first_ts = 0;
+    if (c + 1 < nchunks) {
+      DVec tnext;
+      d_vec_open(blob + dir.ts_off[ci + 1], &tnext);
+      next_first_ts = d_lv_at(&tnext, 0);
+    }
 
     const uint8_t* hv = blob + dir.val_off[ci];
     const int defsz = d_u16(hv + FDB_HIST_OFF_DEFSIZE);
     const uint8_t* sp = hv + FDB_HIST_OFF_DEF + defsz;  // first section
 
-    const int b = lane;                                 // bucket owned by lane
-    const bool live = b < nb;
-    double raw = 0, C = 0, base = 0;
-    const int grp = group_ids[sid];
-
     int sect_left = 0;        // elements left in current section
     int sect_first = 0;       // next element is the section's base element
     const uint8_t* ep = sp;   // element cursor
-    const double inv_step = 1.0 / (double)qstep;
 
     for (int e = 0; e < n; e++) {
       if (sect_left == 0) {   // enter next section
@@ -1045,13 +1073,32 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       sect_first = 0;
       sect_left--;
       ep += 2 + elen;
+      if (e == 0) {
+        if (c > 0) {
+          // detectDropAndCorrection (HistogramVector.scala:670-681):
+          // firstValue < lastValue in Histogram.compare's top-bucket-down
+          // order => correction += lastValue (per bucket)
+          unsigned long long diff = __ballot(live && raw != prevLast);
+          if (diff) {
+            int L = 63 - __clzll(diff);          // highest differing bucket
+            int lt = __shfl((int)(raw < prevLast), L);
+            if (lt) C += prevLast;
+          }
+        }
+        C0[c] = C;   // window-local values subtract this (C==0 when c==0)
+      }
       const double corrected = raw + C;
 
       // window triggers (inversion of the row-range search)
       const int64_t ts_e = ws.ts[e];
-      // sentinels kept near the data so the double conversion stays exact
-      const int64_t ts_prev = e > 0 ? ws.ts[e - 1] : ts_e - ((int64_t)1 << 40);
-      const int64_t ts_next = e + 1 < n ? ws.ts[e + 1] : ts_e + ((int64_t)1 << 40);
+      // sentinels kept near the data so the double conversion stays exact;
+      // at chunk boundaries the neighbor timestamp comes from the adjacent
+      // chunk so windows starting/ending in the inter-chunk gap fire here
+      const int64_t ts_prev = e > 0 ? ws.ts[e - 1]
+                            : (c > 0 ? prev_ts : ts_e - ((int64_t)1 << 40));
+      const int64_t ts_next = e + 1 < n ? ws.ts[e + 1]
+                            : (c + 1 < nchunks ? next_first_ts
+                                               : ts_e + ((int64_t)1 << 40));
       // windows starting at e: wStart in (ts_prev, ts_e]
       // (clamp in i64 BEFORE narrowing)
       int64_t ws_lo64 = d_fdiv_fast(ts_prev - qstart + qwindow, qstep, inv_step) + 1;
@@ -1062,11 +1109,14 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       for (int w = ws_lo; w <= ws_hi; w++) {
         int slot = w % FDB_HIST_RING;
         if (live) ws.ring_lo[slot][b] = corrected;
-        if (lane == 0) { ws.ring_w[slot] = w; ws.ring_e[slot] = e; }
+        if (lane == 0) {
+          ws.ring_w[slot] = w; ws.ring_e[slot] = EG0 + e;
+          ws.ring_t1[slot] = ts_e; ws.ring_c0[slot] = c;
+        }
       }
       // windows ending at e: wEnd in [ts_e, ts_next)
       int64_t we_lo64 = d_fdiv_fast(ts_e - qstart + qstep - 1, qstep, inv_step);
-      int64_t we_hi64 = e + 1 < n
+      int64_t we_hi64 = (e + 1 < n || c + 1 < nchunks)
           ? d_fdiv_fast(ts_next - qstart + qstep - 1, qstep, inv_step) - 1
           : num_windows - 1;
       if (we_lo64 < 0) we_lo64 = 0;
@@ -1078,23 +1128,33 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         int slot = w % FDB_HIST_RING;
         if (ws.ring_w[slot] != w) continue;     // window never started (empty)
         int e0 = ws.ring_e[slot];
-        int64_t t1 = ws.ts[e0];
+        int64_t t1 = ws.ring_t1[slot];
         if (!(ts_e > t1)) continue;             // highestTime > lowestTime rule
         int64_t wEnd = qstart + (int64_t)w * qstep;
         int64_t wStart = wEnd - qwindow;
-        int numSamples = e - e0 + 1;
+        int numSamples = (EG0 + e) - e0 + 1;
         if (live) {
+          // correction total at entry of the window's first chunk: the
+          // reference's per-window meta starts NoCorrection there
+          double c0v = C0[0];
+          #pragma unroll
+          for (int k = 1; k < MAXC; k++)
+            if (ws.ring_c0[slot] == k) c0v = C0[k];
           double r = d_extrapolated_rate(wStart, wEnd, numSamples,
-                                         t1, ws.ring_lo[slot][b],
-                                         ts_e, corrected, true, true);
+                                         t1, ws.ring_lo[slot][b] - c0v,
+                                         ts_e, corrected - c0v, true, true);
           atomicAdd(&out_sums[((size_t)grp * num_windows + w) * nb + b], r);
         }
         if (lane == 0)
           atomicAdd(&out_cnt[(size_t)grp * num_windows + w], 1.0);
       }
-    }
+    }                                           // elements
+    prevLast = raw;                             // updateCorrection: RAW apply(n-1)
+    prev_ts = ws.ts[n - 1];
+    EG0 += n;
     __builtin_amdgcn_s_waitcnt(0);
     __builtin_amdgcn_wave_barrier();
+    }                                           // chunks
   }
 }
 
@@ -1279,6 +1339,7 @@ struct fdb_dataset {
   int max_group;            // max group id seen (for validation)
   int max_rows;             // max rows in one series (capacity tier selection)
   int max_chunks;           // max chunks in one series
+  int max_chunk_rows;       // max rows in one chunk (hist per-chunk LDS cap)
   int has_hist;             // dataset holds sect-delta histogram vectors
 };
 
@@ -1346,12 +1407,13 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
     payload += (int64_t)tl + 4 + (int64_t)vl + 4;
     samples += dir[i].num_rows;
   }
-  int max_group = 0, max_rows = 0, max_chunks = 0;
+  int max_group = 0, max_rows = 0, max_chunks = 0, max_chunk_rows = 0;
   int has_hist = 0, has_scalar = 0;
   for (int64_t i2 = 0; i2 < nc; i2++) {
     uint16_t wf;
     memcpy(&wf, view.blob + dir[i2].val_off + 4, 2);
     if (wf == FDB_WF_HIST_SECTDELTA) has_hist = 1; else has_scalar = 1;
+    if (dir[i2].num_rows > max_chunk_rows) max_chunk_rows = dir[i2].num_rows;
   }
   if (has_hist && has_scalar) {
     fdb_set_error("mixed histogram and scalar series in one store are not "
@@ -1382,6 +1444,7 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
   d->max_group = max_group;
   d->max_rows = max_rows;
   d->max_chunks = max_chunks;
+  d->max_chunk_rows = max_chunk_rows;
   d->has_hist = has_hist;
 
   auto upload = [&](void** dst, const void* src, size_t bytes) -> bool {
@@ -1593,9 +1656,14 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
     fdb_set_error("not a histogram dataset");
     return FDB_ERR_BADARG;
   }
-  if (d->max_chunks > 1) {
-    fdb_set_error("histogram series spanning multiple chunks are round-2 work "
-                  "(streaming element walk)");
+  if (d->max_chunks > FDB_HIST_MAX_CHUNKS) {
+    fdb_set_error("histogram series span %d chunks; the streaming walk handles "
+                  "up to %d", d->max_chunks, FDB_HIST_MAX_CHUNKS);
+    return FDB_ERR_BADARG;
+  }
+  if (d->max_chunk_rows > FDB_MAX_ROWS_PER_SERIES) {
+    fdb_set_error("histogram chunk has %d rows; per-chunk cap is %d",
+                  d->max_chunk_rows, FDB_MAX_ROWS_PER_SERIES);
     return FDB_ERR_BADARG;
   }
   if (q->window / q->step + 2 > FDB_HIST_RING) {
@@ -1622,10 +1690,18 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
   DirSoA dir{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
   int grid = (d->num_series + HIST_WAVES - 1) / HIST_WAVES;
   if (grid > 8192) grid = 8192;
-  hipLaunchKernelGGL(hist_scan_kernel, dim3(grid), dim3(HIST_WAVES * 64), 0, e->stream,
-                     d->blob, dir, d->series_first, d->series_nchunks, d->group_ids,
-                     d->num_series, q->start, q->step, q->end, q->window, nw, nb,
-                     dev_sums, dev_cnt);
+  if (d->max_chunks > 1)
+    hipLaunchKernelGGL(hist_scan_kernel<FDB_HIST_MAX_CHUNKS>, dim3(grid),
+                       dim3(HIST_WAVES * 64), 0, e->stream,
+                       d->blob, dir, d->series_first, d->series_nchunks, d->group_ids,
+                       d->num_series, q->start, q->step, q->end, q->window, nw, nb,
+                       dev_sums, dev_cnt);
+  else
+    hipLaunchKernelGGL(hist_scan_kernel<1>, dim3(grid), dim3(HIST_WAVES * 64), 0,
+                       e->stream,
+                       d->blob, dir, d->series_first, d->series_nchunks, d->group_ids,
+                       d->num_series, q->start, q->step, q->end, q->window, nw, nb,
+                       dev_sums, dev_cnt);
   HIP_CHECK(hipGetLastError());
 
   if (out_quantile) {

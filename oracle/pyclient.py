@@ -80,6 +80,22 @@ def hist_decode(vec_bytes, cap=200000):
     return out[:n.value * nb.value].reshape(n.value, nb.value).copy()
 
 
+def hist_corrections(vec_bytes, cap=200000):
+    """Cumulative in-chunk TypeDrop corrections at each element [n × nb]."""
+    import ctypes as ct
+    dec = hist_decode(vec_bytes, cap)     # for the shape
+    out = np.empty(dec.size, dtype=np.int64)
+    L = lib()
+    L.oracle_hist_corrections.argtypes = [ct.POINTER(ct.c_uint8),
+                                          ct.POINTER(ct.c_int64), ct.c_int32]
+    rc = L.oracle_hist_corrections(_u8(vec_bytes),
+                                   out.ctypes.data_as(ct.POINTER(ct.c_int64)),
+                                   dec.size)
+    if rc != 0:
+        raise RuntimeError("hist_corrections failed")
+    return out.reshape(dec.shape).copy()
+
+
 def hist_quantile(q, values, first, mult):
     import ctypes as ct
     L = lib()

@@ -54,6 +54,12 @@ def test_encode_decode_roundtrip(deltas, vals):
     else:
         np.testing.assert_array_equal(dec_ts, ts)
     dec_vs = oracle.decode_doubles(vab)
+    vinfo = oracle.vec_info(vab)
+    if vinfo["wf"] != 0x0506:
+        # integral doubles ride the DeltaDelta-long path, where the reference
+        # itself canonicalizes -0.0 to +0.0 (LongDoubleWrapper.toLong,
+        # DoubleVector.scala:517-538)
+        vs = np.where(vs == 0.0, 0.0, vs)
     np.testing.assert_array_equal(
         dec_vs.view(np.uint64), vs.view(np.uint64))  # bit-exact incl. NaNs
 

@@ -55,3 +55,51 @@ def test_hist_empty_windows(fdb, oracle, engine):
     np.testing.assert_array_equal(gc, wc)
     np.testing.assert_allclose(gs, ws, rtol=1e-9, atol=1e-12)
     np.testing.assert_allclose(gq, wq, rtol=1e-9, atol=1e-12, equal_nan=True)
+
+
+def test_hist_multichunk(fdb, oracle, engine):
+    """Multi-chunk histogram series: streaming walk with cross-chunk
+    correction carry must match the oracle's per-window CorrectionMeta
+    (boundary drop detection, in-chunk TypeDrop carry, gap windows)."""
+    from test_hist import _multichunk_cases
+    rng = np.random.default_rng(47)
+    nb = 8
+    cases = _multichunk_cases(rng, nb)
+    # one store with all cases as separate series in separate groups,
+    # plus a single-chunk series to exercise the mixed dispatch
+    ts0, c0 = synth_hist(rng, 40, nb=nb)
+    series = cases + [(ts0, c0)]
+    groups = list(range(len(series)))
+    import filodb_amd as f
+    start = int(ts0[10])
+    end = start + 200 * 15000
+    q = f.make_query(start, 15000, end, 300000, f.FN_HIST_RATE,
+                     f.AGG_SUM, len(series), param=0.9)
+    (gs, gc, gq), (ws, wc, wq) = run_pair(fdb, oracle, engine, series, nb,
+                                          groups, q)
+    np.testing.assert_array_equal(gc, wc)
+    np.testing.assert_allclose(gs, ws, rtol=1e-9, atol=1e-12)
+    np.testing.assert_allclose(gq, wq, rtol=1e-9, atol=1e-12, equal_nan=True)
+
+
+def test_hist_too_many_chunks_rejected(fdb, engine):
+    """> FDB_HIST_MAX_CHUNKS chunks must error loudly, not silently skip."""
+    rng = np.random.default_rng(3)
+    nb = 8
+    chunks = []
+    t = 100000
+    for _ in range(5):
+        ts, cum = synth_hist(rng, 20, nb=nb, start_ts=t)
+        chunks.append((ts, cum))
+        t = int(ts[-1]) + 15000
+    st = make_hist_store(fdb, [chunks])
+    ds = engine.upload(st)
+    import filodb_amd as f
+    q = f.make_query(200000, 15000, 500000, 300000, f.FN_HIST_RATE,
+                     f.AGG_SUM, 1, param=0.5)
+    nw = q.num_windows
+    with pytest.raises(RuntimeError):
+        engine.query_hist(ds, q, nb,
+                          out_bucket_sums=np.zeros(nw * nb),
+                          out_counts=np.zeros(nw),
+                          out_quantile=np.zeros(nw))

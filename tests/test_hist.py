@@ -26,12 +26,18 @@ def test_quantile_golden(oracle):
 
 
 def make_hist_store(fdb, series_list, nb=8, max_rows=400, groups=None):
+    """series_list entries: (ts, cum) for one chunk, or a list of such pairs
+    for a multi-chunk series (explicit cut between chunks)."""
     st = fdb.ChunkStore()
     st.set_max_rows(max_rows)
-    for i, (ts, bv) in enumerate(series_list):
+    for i, entry in enumerate(series_list):
         gid = groups[i] if groups else 0
         sid = st.add_series(gid, fdb.COL_HIST)
-        st.append_hist(sid, ts, bv, first=2.0, mult=2.0)
+        chunks = entry if isinstance(entry, list) else [entry]
+        for ci, (ts, bv) in enumerate(chunks):
+            if ci > 0:
+                st.cut_chunk(sid)
+            st.append_hist(sid, ts, bv, first=2.0, mult=2.0)
     st.seal()
     return st
 
@@ -174,3 +180,117 @@ def test_hist_group_sum(fdb, oracle):
                 expc += c1[:nw]
         np.testing.assert_allclose(sums[g * nw * 8:(g + 1) * nw * 8], exp, rtol=1e-9)
         np.testing.assert_array_equal(cnts[g * nw:(g + 1) * nw], expc)
+
+
+# ---------------------------------------------------------------------------
+# multi-chunk histogram series: per-window CorrectionMeta semantics
+# (CounterChunkedRangeFunction, RangeFunction.scala:131-165;
+#  SectDeltaHistogramReader.detectDropAndCorrection/updateCorrection,
+#  HistogramVector.scala:670-719; Histogram.compare, Histogram.scala:204-214)
+# ---------------------------------------------------------------------------
+
+def _lex_less(a, b):
+    """Histogram.compare with equal schemes: top bucket down, first difference."""
+    for i in range(len(a) - 1, -1, -1):
+        if a[i] != b[i]:
+            return a[i] < b[i]
+    return False
+
+
+def naive_hist_rate_window(oracle, chunks, w_start, w_end, nb):
+    """chunks: list of (ts, raw [n×nb], corr [n×nb], start_t, end_t)."""
+    carry = np.zeros(nb)
+    lastv = None
+    samples = 0
+    low_t, hi_t = None, None
+    lo = hi = None
+    for ts, raw, corr, st_t, en_t in chunks:
+        if en_t < w_start:
+            continue
+        s = int(np.searchsorted(ts, w_start, side="left"))
+        e = int(np.searchsorted(ts, w_end, side="right")) - 1
+        e = min(e, len(ts) - 1)
+        if lastv is not None and _lex_less(raw[0], lastv):
+            carry = carry + lastv
+        if s <= e:
+            if low_t is None or ts[s] < low_t or ts[e] > hi_t:
+                samples += e - s + 1
+                if low_t is None or ts[s] < low_t:
+                    low_t = int(ts[s])
+                    lo = raw[s] + corr[s] + carry
+                if hi_t is None or ts[e] > hi_t:
+                    hi_t = int(ts[e])
+                    hi = raw[e] + corr[e] + carry
+        carry = carry + corr[-1]
+        lastv = raw[-1]
+        if en_t >= w_end:
+            break
+    if low_t is None or hi_t is None or not hi_t > low_t:
+        return None
+    return np.array([oracle.extrapolated_rate(w_start, w_end, samples,
+                                              low_t, float(lo[b]),
+                                              hi_t, float(hi[b]), True, True)
+                     for b in range(nb)])
+
+
+def _multichunk_cases(rng, nb):
+    """Chunked hist series exercising: in-chunk resets, a hard counter reset
+    at a chunk boundary, a long inter-chunk gap, and equal boundary values."""
+    cases = []
+    # (a) plain 3 chunks, contiguous time, no resets
+    ts1, c1 = synth_hist(rng, 40, nb=nb)
+    ts2, c2 = synth_hist(rng, 40, nb=nb, start_ts=int(ts1[-1]) + 15000)
+    c2 = c2 + c1[-1]          # continue the counters
+    ts3, c3 = synth_hist(rng, 40, nb=nb, start_ts=int(ts2[-1]) + 15000)
+    c3 = c3 + c2[-1]
+    cases.append([(ts1, c1), (ts2, c2), (ts3, c3)])
+    # (b) counter reset exactly at the chunk boundary (drop detection)
+    ts4, c4 = synth_hist(rng, 50, nb=nb)
+    ts5, c5 = synth_hist(rng, 50, nb=nb, start_ts=int(ts4[-1]) + 15000)
+    cases.append([(ts4, c4), (ts5, c5)])  # c5 restarts near zero => drop
+    # (c) in-chunk resets + boundary drop + a 30-minute gap between chunks
+    ts6, c6 = synth_hist(rng, 60, nb=nb, reset_p=0.05)
+    ts7, c7 = synth_hist(rng, 60, nb=nb, reset_p=0.05,
+                         start_ts=int(ts6[-1]) + 1_800_000)
+    cases.append([(ts6, c6), (ts7, c7)])
+    # (d) identical value at the boundary (compare == 0: NOT a drop)
+    ts8, c8 = synth_hist(rng, 30, nb=nb)
+    ts9 = (ts8[-1] + 15000 + np.arange(30) * 15000).astype(np.int64)
+    inc = rng.poisson(2.0, (30, nb)).astype(np.uint64)
+    inc[0] = 0                # first element equals the previous last exactly
+    c9 = c8[-1] + np.cumsum(np.cumsum(inc, axis=1), axis=0)
+    cases.append([(ts8, c8), (ts9, c9)])
+    return cases
+
+
+def test_hist_multichunk_oracle_vs_naive(fdb, oracle):
+    rng = np.random.default_rng(31)
+    nb = 8
+    for case in _multichunk_cases(rng, nb):
+        st = make_hist_store(fdb, [case])
+        view = st.view()
+        chunks = []
+        for ci in range(len(case)):
+            tsb, vab, n, st_t, en_t = st.chunk(0, ci)
+            raw = oracle.hist_decode(vab).astype(np.float64)
+            corr = oracle.hist_corrections(vab).astype(np.float64)
+            # stored timestamps (approx-const encoding may shift within ±250)
+            dec_ts = oracle.decode_longs(tsb)
+            chunks.append((dec_ts, raw, corr, st_t, en_t))
+        start = int(chunks[0][0][10])
+        end = int(chunks[-1][0][-1]) + 30000
+        q = fdb.make_query(start, 30000, end, 300000, fdb.FN_HIST_RATE,
+                           fdb.AGG_SUM, 1, param=0.5)
+        nw = q.num_windows
+        sums, cnts, _ = oracle.query_exec_hist(view, q, nb)
+        for w in range(nw):
+            w_end = start + w * 30000
+            expect = naive_hist_rate_window(oracle, chunks, w_end - 300000,
+                                            w_end, nb)
+            got = sums[w * nb:(w + 1) * nb]
+            if expect is None:
+                assert cnts[w] == 0, w
+            else:
+                assert cnts[w] == 1, w
+                np.testing.assert_allclose(got, expect, rtol=1e-9, atol=1e-12,
+                                           err_msg=f"window {w}")
