@@ -11,7 +11,7 @@ import os
 import numpy as np
 
 _ROOT = os.path.dirname(os.path.abspath(__file__))
-LIB_PATH = os.path.join(_ROOT, "libfilodb_amd.so")
+LIB_PATH = os.environ.get("FDB_LIB", os.path.join(_ROOT, "libfilodb_amd.so"))
 
 # function ids (include/filodb_amd.h; dispatch table RangeFunction.scala:294-410)
 FN_RATE, FN_INCREASE, FN_DELTA = 0, 1, 2

@@ -676,7 +676,13 @@ extern "C" int32_t fdb_synth_generate(fdb_store_t* s, int32_t kind, int32_t n_se
       last_ts = ts;
       if (kind == FDB_COL_HIST) {
         // config #4 shape: 64 geometric buckets, per-interval cumulative counts
-        const int nb = 64;
+        // (FDB_SYNTH_HIST_NB overrides for perf experiments)
+        static const int nb_env = [] {
+          const char* v = getenv("FDB_SYNTH_HIST_NB");
+          int x = v ? atoi(v) : 64;
+          return (x >= 1 && x <= 64) ? x : 64;
+        }();
+        const int nb = nb_env;
         if (se.num_buckets == 0) {
           se.num_buckets = nb; se.bucket_first = 2.0; se.bucket_mult = 2.0;
           se.buf_hist.reserve((size_t)s->max_rows * nb);

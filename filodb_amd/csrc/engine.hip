@@ -1689,7 +1689,9 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
 
   DirSoA dir{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
   int grid = (d->num_series + HIST_WAVES - 1) / HIST_WAVES;
-  if (grid > 8192) grid = 8192;
+  int hcap = 8192;
+  if (const char* g = getenv("FDB_HIST_GRID")) hcap = atoi(g);  // perf experiments
+  if (hcap > 0 && grid > hcap) grid = hcap;
   if (d->max_chunks > 1)
     hipLaunchKernelGGL(hist_scan_kernel<FDB_HIST_MAX_CHUNKS>, dim3(grid),
                        dim3(HIST_WAVES * 64), 0, e->stream,
