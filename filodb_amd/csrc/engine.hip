@@ -112,6 +112,14 @@ __device__ __forceinline__ int64_t d_inner_at(const DVec* v, int i) {
   return 0;
 }
 
+// wave-local LDS ordering only: s_waitcnt lgkmcnt(0) — unlike s_waitcnt(0)
+// this does NOT drain vmcnt, so outstanding global loads/stores/atomics keep
+// flying across the fence (a full flush on the per-element hot path costs a
+// full memory round trip per element)
+__device__ __forceinline__ void d_wait_lds() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+}
+
 __device__ __forceinline__ int64_t d_lv_at(const DVec* v, int i) {
   if (v->wf == FDB_WF_DDV) return v->init + (int64_t)v->slope * i + d_inner_at(v, i);
   if (v->wf == FDB_WF_DDV_CONST) return v->init + (int64_t)v->slope * i;
@@ -444,7 +452,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     row0 += n;
   }
   if (lane == 0) ws.total_rows = row0;
-  __builtin_amdgcn_s_waitcnt(0);   // wave-local LDS visibility
+  d_wait_lds();                    // wave-local LDS visibility
   __builtin_amdgcn_wave_barrier();
   if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_decode += t - tt0; tt0 = t; }
 
@@ -576,7 +584,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       }
     }
   }
-  __builtin_amdgcn_s_waitcnt(0);
+  d_wait_lds();
   __builtin_amdgcn_wave_barrier();
   if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_meta += t - tt0; tt0 = t; }
 
@@ -722,7 +730,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       }
       }  // stage C per-window
     }  // window batches
-    __builtin_amdgcn_s_waitcnt(0);
+    d_wait_lds();
     __builtin_amdgcn_wave_barrier();
     if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_win += t - tt0; tt0 = t; }
     continue;
@@ -892,7 +900,7 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     }
   }
   // next series reuses this wave's LDS slot: wave-local ordering is enough
-  __builtin_amdgcn_s_waitcnt(0);
+  d_wait_lds();
   __builtin_amdgcn_wave_barrier();
   if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_win += t - tt0; tt0 = t; }
   }  // series grid-stride loop
@@ -1003,7 +1011,7 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     DVec tv;
     d_vec_open(blob + dir.ts_off[ci], &tv);
     d_decode_chunk<false>(tv, n, ws.ts, nullptr, lane);
-    __builtin_amdgcn_s_waitcnt(0);
+    d_wait_lds();
     __builtin_amdgcn_wave_barrier();
 
     // first timestamp of the next chunk: closes windows ending in the gap
@@ -1123,7 +1131,7 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       if (we_hi64 > num_windows - 1) we_hi64 = num_windows - 1;
       int we_lo = (int)we_lo64, we_hi = (int)we_hi64;
       if (ws_hi >= we_lo)   // a slot written this element may be read below
-        { __builtin_amdgcn_s_waitcnt(0); __builtin_amdgcn_wave_barrier(); }
+        { d_wait_lds(); __builtin_amdgcn_wave_barrier(); }
       for (int w = we_lo; w <= we_hi; w++) {
         int slot = w % FDB_HIST_RING;
         if (ws.ring_w[slot] != w) continue;     // window never started (empty)
@@ -1152,7 +1160,7 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     prevLast = raw;                             // updateCorrection: RAW apply(n-1)
     prev_ts = ws.ts[n - 1];
     EG0 += n;
-    __builtin_amdgcn_s_waitcnt(0);
+    d_wait_lds();
     __builtin_amdgcn_wave_barrier();
     }                                           // chunks
   }
