@@ -961,6 +961,60 @@ __device__ __forceinline__ int64_t wave_incl_scan_i64(int64_t x, int lane) {
   return x;
 }
 
+// --- wave-staged element stream -------------------------------------------
+// One coalesced dword-per-lane load stages 256 B of an element's NibblePack
+// stream into the wave's registers; all subsequent byte/u16/u64 reads are
+// register shuffles (ds_bpermute) instead of dependent global loads. The
+// stage for element e+1 is issued before element e is parsed, so the load
+// latency overlaps a full element's parse work. Elements longer than the
+// staged range (elen+shift > 252, i.e. huge sections) fall back to direct
+// global reads; the blob is tail-padded 256 B at upload so staging never
+// faults.
+__device__ __forceinline__ uint32_t estream_stage(const uint8_t* p, int lane) {
+  const uint8_t* base = (const uint8_t*)((uintptr_t)p & ~(uintptr_t)3);
+  uint32_t d;
+  memcpy(&d, base + lane * 4, 4);
+  return d;
+}
+// `staged` must be WAVE-UNIFORM and every call site must have the full wave
+// active: __shfl is ds_bpermute, and a source lane that is inactive (or a
+// divergent caller) yields undefined data.
+__device__ __forceinline__ uint32_t estream_byte(bool staged, uint32_t buf,
+                                                 int shift, const uint8_t* g,
+                                                 int k) {
+  if (staged) {
+    int kk = k + shift;
+    uint32_t d = __shfl(buf, kk >> 2);
+    return (d >> ((kk & 3) * 8)) & 0xff;
+  }
+  return g[k];
+}
+__device__ __forceinline__ uint32_t estream_u16(bool staged, uint32_t buf,
+                                                int shift, const uint8_t* g,
+                                                int k) {
+  return estream_byte(staged, buf, shift, g, k) |
+         (estream_byte(staged, buf, shift, g, k + 1) << 8);
+}
+// little-endian 8-byte window at offset k (per-lane k; bpermute shuffles)
+__device__ __forceinline__ uint64_t estream_w64(bool staged, uint32_t buf,
+                                                int shift, const uint8_t* g,
+                                                int k) {
+  if (staged) {
+    int kk = k + shift;
+    int dw = kk >> 2;
+    uint64_t a = __shfl(buf, dw);
+    uint64_t b = __shfl(buf, dw + 1);
+    uint64_t c = __shfl(buf, dw + 2);
+    int sh = (kk & 3) * 8;
+    uint64_t w = (a | (b << 32)) >> sh;
+    if (sh) w |= c << (64 - sh);
+    return w;
+  }
+  uint64_t w;
+  memcpy(&w, g + k, 8);
+  return w;
+}
+
 // MAXC = max chunks per series this instantiation handles. The walk streams
 // chunks in time order carrying a running per-bucket correction C; per-window
 // values subtract C0[c0] (C at entry of the window's FIRST chunk) so results
@@ -978,12 +1032,14 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                       int64_t qstart, int64_t qstep, int64_t qend, int64_t qwindow,
                       int num_windows, int nb,
                       double* __restrict__ out_sums,   // [G × W × nb]
-                      double* __restrict__ out_cnt)    // [G × W]
+                      double* __restrict__ out_cnt,    // [G × W]
+                      int dbg)   // FDB_HIST_TIME: phase cycles into out_cnt[0..3]
 {
   __shared__ HistWs ws_all[HIST_WAVES];
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   HistWs& ws = ws_all[wave];
+  uint64_t t_parse = 0, t_scan = 0, t_win = 0, tt = 0, n_elem = 0;
 
   for (int sid = blockIdx.x * HIST_WAVES + wave; sid < num_series;
        sid += gridDim.x * HIST_WAVES) {
@@ -1030,7 +1086,12 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     int sect_first = 0;       // next element is the section's base element
     const uint8_t* ep = sp;   // element cursor
 
+    // pre-stage element 0 (first element of the first section)
+    uint32_t ebuf = estream_stage(sp + 4, lane);
+    int eshift = (int)((uintptr_t)(sp + 4) & 3);
+
     for (int e = 0; e < n; e++) {
+      if (dbg) tt = __builtin_amdgcn_s_memtime();
       if (sect_left == 0) {   // enter next section
         int stype = sp[3];
         sect_left = sp[2];
@@ -1039,19 +1100,27 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         sect_first = 1;
         if (stype == 1 && e > 0) C += raw;  // TypeDrop: corr += apply(e-1)
       }
-      const int elen = d_u16(ep);
-      const uint8_t* s = ep + 2;
+      // elen's two bytes sit at staged offsets 0-1: always in range
+      const int elen = (int)estream_u16(true, ebuf, eshift, ep, 0);
+      // whole-element staging test, uniform across the wave (the max read
+      // offset is koff+8 <= elen+10; +shift must stay inside the 256 B stage)
+      const bool est = elen + 14 + eshift <= 256;
+      // issue the stage load for the NEXT element before parsing this one —
+      // its latency overlaps the whole parse below
+      const uint8_t* nxt = sect_left > 1 ? ep + 2 + elen : sp + 4;
+      uint32_t nbuf = estream_stage(nxt, lane);
+      int nshift = (int)((uintptr_t)nxt & 3);
       // parse the 8-value group headers serially (uniform across lanes)
       int my_group = b >> 3;
       int off = 0, gOff = 0, gBits = 0, gTrail = 0;
       uint32_t gMask = 0;
       for (int g = 0; g * 8 < nb; g++) {
-        uint32_t mask = s[off];
+        uint32_t mask = estream_byte(est, ebuf, eshift, ep, 2 + off);
         int numBits = 0, trail = 0, glen;
         if (mask == 0) {
           glen = 1;
         } else {
-          int widths = s[off + 1];
+          int widths = (int)estream_byte(est, ebuf, eshift, ep, 2 + off + 1);
           numBits = ((widths >> 4) + 1) * 4;
           trail = (widths & 0x0f) * 4;
           glen = 2 + (numBits * __popc(mask) + 7) / 8;
@@ -1059,21 +1128,26 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         if (g == my_group) { gOff = off; gBits = numBits; gTrail = trail; gMask = mask; }
         off += glen;
       }
-      // extract this lane's value from its group
-      int64_t delta = 0;
+      // extract this lane's value from its group. The stream reads run with
+      // the FULL wave active (shuffle sources must be live); only the final
+      // delta is predicated.
       const int bit = b & 7;
-      if (live && (gMask & (1u << bit))) {
-        int slot = __popc(gMask & ((1u << bit) - 1));
-        int bitpos = slot * gBits;
-        const uint8_t* p = s + gOff + 2 + (bitpos >> 3);
-        uint64_t w64;
-        memcpy(&w64, p, 8);                  // unaligned-safe via memcpy
+      const uint32_t in_mask = gMask & (1u << bit);
+      int slot = __popc(gMask & ((1u << bit) - 1));
+      int bitpos = slot * gBits;
+      int koff = 2 + gOff + 2 + (bitpos >> 3);     // byte offset from ep
+      uint64_t w64 = estream_w64(est, ebuf, eshift, ep, koff);
+      uint32_t b8 = estream_byte(est, ebuf, eshift, ep, koff + 8);
+      int64_t delta = 0;
+      if (live && in_mask) {
         int sh = bitpos & 7;
         uint64_t v = w64 >> sh;
-        if (gBits > 64 - sh) v |= (uint64_t)p[8] << (64 - sh);
+        if (gBits > 64 - sh) v |= (uint64_t)b8 << (64 - sh);
         uint64_t m = gBits >= 64 ? ~0ULL : ((1ULL << gBits) - 1);
         delta = (int64_t)((v & m) << gTrail);
       }
+      ebuf = nbuf; eshift = nshift;
+      if (dbg) { uint64_t t = __builtin_amdgcn_s_memtime(); t_parse += t - tt; tt = t; }
       // reconstruct this element's cumulative bucket value
       int64_t scan = wave_incl_scan_i64(live ? delta : 0, lane);
       if (sect_first) { raw = (double)scan; base = raw; }
@@ -1096,6 +1170,7 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         C0[c] = C;   // window-local values subtract this (C==0 when c==0)
       }
       const double corrected = raw + C;
+      if (dbg) { uint64_t t = __builtin_amdgcn_s_memtime(); t_scan += t - tt; tt = t; }
 
       // window triggers (inversion of the row-range search)
       const int64_t ts_e = ws.ts[e];
@@ -1156,6 +1231,7 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         if (lane == 0)
           atomicAdd(&out_cnt[(size_t)grp * num_windows + w], 1.0);
       }
+      if (dbg) { t_win += __builtin_amdgcn_s_memtime() - tt; n_elem++; }
     }                                           // elements
     prevLast = raw;                             // updateCorrection: RAW apply(n-1)
     prev_ts = ws.ts[n - 1];
@@ -1163,6 +1239,12 @@ void hist_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     d_wait_lds();
     __builtin_amdgcn_wave_barrier();
     }                                           // chunks
+  }
+  if (dbg && lane == 0) {
+    atomicAdd(&out_cnt[0], (double)t_parse);
+    atomicAdd(&out_cnt[1], (double)t_scan);
+    atomicAdd(&out_cnt[2], (double)t_win);
+    atomicAdd(&out_cnt[3], (double)n_elem);
   }
 }
 
@@ -1471,7 +1553,14 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
       sbg[(size_t)cur[(size_t)view.group_ids[s2]]++] = s2;
   }
 
-  bool ok = upload((void**)&d->blob, view.blob, (size_t)view.blob_len)
+  // +256 B tail pad: the hist walk's wave-staged element reads may extend up
+  // to 256 B past the last element's start (estream staging)
+  auto upload_pad = [&](void** dst, const void* src, size_t bytes) -> bool {
+    if (hipMalloc(dst, bytes + 256) != hipSuccess) return false;
+    if (hipMemset((char*)*dst + bytes, 0, 256) != hipSuccess) return false;
+    return hipMemcpy(*dst, src, bytes, hipMemcpyHostToDevice) == hipSuccess;
+  };
+  bool ok = upload_pad((void**)&d->blob, view.blob, (size_t)view.blob_len)
     && upload((void**)&d->ts_off, ts_off.data(), nc * 8)
     && upload((void**)&d->val_off, val_off.data(), nc * 8)
     && upload((void**)&d->start_time, st.data(), nc * 8)
@@ -1700,18 +1789,19 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
   int hcap = 8192;
   if (const char* g = getenv("FDB_HIST_GRID")) hcap = atoi(g);  // perf experiments
   if (hcap > 0 && grid > hcap) grid = hcap;
+  int dbg = getenv("FDB_HIST_TIME") ? 1 : 0;
   if (d->max_chunks > 1)
     hipLaunchKernelGGL(hist_scan_kernel<FDB_HIST_MAX_CHUNKS>, dim3(grid),
                        dim3(HIST_WAVES * 64), 0, e->stream,
                        d->blob, dir, d->series_first, d->series_nchunks, d->group_ids,
                        d->num_series, q->start, q->step, q->end, q->window, nw, nb,
-                       dev_sums, dev_cnt);
+                       dev_sums, dev_cnt, dbg);
   else
     hipLaunchKernelGGL(hist_scan_kernel<1>, dim3(grid), dim3(HIST_WAVES * 64), 0,
                        e->stream,
                        d->blob, dir, d->series_first, d->series_nchunks, d->group_ids,
                        d->num_series, q->start, q->step, q->end, q->window, nw, nb,
-                       dev_sums, dev_cnt);
+                       dev_sums, dev_cnt, dbg);
   HIP_CHECK(hipGetLastError());
 
   if (out_quantile) {
