@@ -1643,11 +1643,18 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
     }
   }
 
+  const bool needs_sq = q->agg_id == AGG_STDDEV || q->agg_id == AGG_STDVAR;
+  int partial = (q->agg_id != AGG_NONE && !is_topk && out_counts != nullptr) ? 1 : 0;
+  // stddev/stdvar partials ship (raw sums, raw sumsq) stacked: the caller's
+  // `out` is [2 × G × W]; shards merge both halves + counts by addition —
+  // algebraically StddevRowAggregator.scala:36-52's reduction
+  size_t buf_len = (needs_sq && partial) ? out_len * 2 : out_len;
+
   double *dev_out = out_on_device ? out : nullptr;
   double *dev_cnt = out_on_device ? out_counts : nullptr;
   bool own_out = false, own_cnt = false;
   if (!dev_out) {
-    HIP_CHECK(hipMalloc(&dev_out, out_len * 8));
+    HIP_CHECK(hipMalloc(&dev_out, buf_len * 8));
     own_out = true;
   }
   if (q->agg_id != AGG_NONE && !dev_cnt) {
@@ -1655,14 +1662,11 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
     own_cnt = true;
   }
 
-  const bool needs_sq = q->agg_id == AGG_STDDEV || q->agg_id == AGG_STDVAR;
-  int partial = (q->agg_id != AGG_NONE && !is_topk && out_counts != nullptr) ? 1 : 0;
-  if (needs_sq && partial) {
-    fdb_set_error("stddev/stdvar cross-shard partial mode is not supported yet");
-    return FDB_ERR_BADARG;
-  }
   double* dev_sq = nullptr;
-  if (needs_sq) HIP_CHECK(hipMalloc(&dev_sq, out_len * 8));
+  if (needs_sq) {
+    if (partial) dev_sq = dev_out + out_len;   // second half of the caller grid
+    else HIP_CHECK(hipMalloc(&dev_sq, out_len * 8));
+  }
   hipEvent_t ev0, ev1;
   HIP_CHECK(hipEventCreate(&ev0));
   HIP_CHECK(hipEventCreate(&ev1));
@@ -1725,12 +1729,12 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
     HIP_CHECK(hipGetLastError());
   }
   if (per_grid) (void)hipFree(per_grid);
-  if (dev_sq) (void)hipFree(dev_sq);
+  if (dev_sq && !(needs_sq && partial)) (void)hipFree(dev_sq);
   HIP_CHECK(hipStreamSynchronize(e->stream));
   (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
 
   if (!out_on_device) {
-    HIP_CHECK(hipMemcpy(out, dev_out, out_len * 8, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(out, dev_out, buf_len * 8, hipMemcpyDeviceToHost));
     if (q->agg_id != AGG_NONE && out_counts)
       HIP_CHECK(hipMemcpy(out_counts, dev_cnt, out_len * 8, hipMemcpyDeviceToHost));
   }

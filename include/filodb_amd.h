@@ -221,6 +221,11 @@ int64_t        fdb_dataset_samples(const fdb_dataset_t* d); /* total rows across
  *   (and cross-GPU merges) out_counts (num_groups×windows doubles, may be NULL
  *   otherwise) receives the count partials — the reduction schema of
  *   AvgRowAggregator.scala:8-41.
+ * Passing out_counts selects PARTIAL mode: out holds raw sums (0 where empty;
+ *   MIN/MAX keep NaN) for the caller to merge across shards and present.
+ *   For FDB_AGG_STDDEV/STDVAR partials, out must hold 2×num_groups×windows
+ *   doubles — raw sums then raw sum-of-squares — merged by addition
+ *   (algebraically StddevRowAggregator.scala:36-52's reduction schema).
  * out/out_counts may be HOST pointers (out_on_device=0) or DEVICE pointers
  * (out_on_device=1, e.g. torch tensor data_ptr for the RCCL all-reduce).
  * Synchronous: returns after the result is materialized. */

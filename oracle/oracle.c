@@ -749,10 +749,15 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
   if (out_counts) {
     /* PARTIAL mode (cross-shard merge inputs, ReduceAggregateExec contract):
      * raw sums with 0 where empty + contribution counts; caller merges then
-     * presents (NaN where total count 0; /count for avg). */
-    if (q->agg_id == AGG_SUM || q->agg_id == AGG_COUNT || q->agg_id == AGG_AVG)
+     * presents (NaN where total count 0; /count for avg). For stddev/stdvar
+     * `out` is [2 x G x W]: raw sums then raw sumsq — merging by addition is
+     * algebraically StddevRowAggregator.scala:36-52's reduction. */
+    if (q->agg_id == AGG_SUM || q->agg_id == AGG_COUNT || q->agg_id == AGG_AVG ||
+        needs_sq)
       for (size_t i = 0; i < gridlen; i++)
         if (isnan(out[i]) && cnt_total[i] == 0) out[i] = 0;
+    if (needs_sq)
+      for (size_t i = 0; i < gridlen; i++) out[gridlen + i] = sq_total[i];
     /* MIN/MAX partials keep NaN for empty cells; the merging caller maps them
      * to ±inf before the collective. */
   } else if (q->agg_id == AGG_AVG) {

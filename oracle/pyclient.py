@@ -116,7 +116,10 @@ def query_exec(view, q, num_series, num_windows, out_counts=False, nthreads=1):
         cells = q.num_groups * num_windows
         if q.agg_id in (6, 7):        # top/bottom-k: [G × W × k] values + ids
             cells *= int(q.param)
-        out = np.empty(cells, dtype=np.float64)
+        buf = cells
+        if out_counts and q.agg_id in (8, 9):
+            buf = cells * 2           # stddev partials: (sums, sumsq) stacked
+        out = np.empty(buf, dtype=np.float64)
         cnt = np.zeros(cells, dtype=np.float64) if out_counts else None
     rc = lib().oracle_query_exec(
         ctypes.byref(view), ctypes.byref(q),

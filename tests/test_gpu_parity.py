@@ -226,3 +226,29 @@ def test_long_lookback_large_capacity(fdb, oracle, engine):
     q2 = mkq(fdb, FUNCS["avg"], end=Q["start"] + 700 * 15000)
     got2, want2 = run_both(fdb, oracle, engine, st2, q2)
     check(got2, want2)
+
+
+def test_stddev_partial_mode_gpu(fdb, oracle, engine):
+    """Stddev partials (raw sums + sumsq stacked, counts) via the engine match
+    the oracle's — the cross-shard merge contract for AGG_STDDEV/STDVAR."""
+    rng = np.random.default_rng(77)
+    n_groups = 3
+    series, groups = [], []
+    for s in range(30):
+        ts, vs = synth_gauge_series(rng, 80, nan_p=0.15)
+        series.append([[(int(t), float(v)) for t, v in zip(ts, vs)]])
+        groups.append(s % n_groups)
+    st = build_store(fdb, series, groups=groups)
+    ds = engine.upload(st)
+    for agg in ("stddev", "stdvar"):
+        q = mkq(fdb, FUNCS["sum"], AGGS[agg], n_groups,
+                end=Q["start"] + 40 * 15000)
+        nw = q.num_windows
+        cells = n_groups * nw
+        want_s, want_c = oracle.query_exec(st.view(), q, st.num_series, nw,
+                                           out_counts=True)
+        got_s = np.empty(2 * cells)
+        got_c = np.empty(cells)
+        engine.query(ds, q, out=got_s, out_counts=got_c)
+        check(got_s, want_s)
+        np.testing.assert_array_equal(got_c, want_c)

@@ -87,3 +87,42 @@ def test_partial_mode_merge_equals_presented(fdb, oracle):
                           (sums / counts) if agg == "avg" else sums,
                           np.nan)
         np.testing.assert_allclose(merged, presented, rtol=1e-12, equal_nan=True)
+
+
+def test_stddev_partial_two_shard_merge(fdb, oracle):
+    """Stddev/stdvar partials = (raw sums, raw sumsq, counts); merging two
+    shards by addition and presenting sqrt(sq/n - mean²) must equal the
+    single-shard presented result (StddevRowAggregator.scala:36-52)."""
+    rng = np.random.default_rng(41)
+    n_groups = 3
+    series, groups = [], []
+    for s in range(24):
+        ts, vs = synth_gauge_series(rng, 50, step=10000, jitter=300, nan_p=0.1)
+        series.append([[(int(t), float(v)) for t, v in zip(ts, vs)]])
+        groups.append(s % n_groups)
+    start = 100000 + 15 * 10000
+    for agg in ("stddev", "stdvar"):
+        q = fdb.make_query(start, 30000, start + 200000, 120000,
+                           fdb.FN_SUM_OVER_TIME, AGGS[agg], n_groups)
+        nw = q.num_windows
+        cells = n_groups * nw
+        # whole dataset, presented single-pass
+        st_all = build_store(fdb, series, groups=groups)
+        presented = oracle.query_exec(st_all.view(), q, st_all.num_series, nw)
+        # two shards: first half / second half of the series
+        tot_s = np.zeros(2 * cells)
+        tot_c = np.zeros(cells)
+        for half in (slice(0, 12), slice(12, 24)):
+            st = build_store(fdb, series[half], groups=groups[half])
+            sums, counts = oracle.query_exec(st.view(), q, st.num_series, nw,
+                                             out_counts=True)
+            assert sums.shape == (2 * cells,)
+            tot_s += sums
+            tot_c += counts
+        merged = np.full(cells, np.nan)
+        m = tot_c > 0
+        mean = tot_s[:cells][m] / tot_c[m]
+        var = tot_s[cells:][m] / tot_c[m] - mean * mean
+        merged[m] = np.sqrt(var) if agg == "stddev" else var
+        np.testing.assert_allclose(merged, presented, rtol=1e-9, atol=1e-12,
+                                   equal_nan=True)
