@@ -166,3 +166,37 @@ def test_rate_window_parity_fuzz(deltas, incs, resets, window, step):
             int(enc_ts[idx[0]]), float(corrected[idx[0]]),
             int(enc_ts[idx[-1]]), float(corrected[idx[-1]]), True, True)
         assert out[w] == expect or abs(out[w] - expect) <= 1e-9 * abs(expect), w
+
+
+@settings(max_examples=40, deadline=None)
+@given(rows=st.lists(
+           st.lists(st.integers(min_value=0, max_value=5000), min_size=4,
+                    max_size=4),
+           min_size=2, max_size=80),
+       reset_at=st.lists(st.integers(min_value=1, max_value=79), max_size=4))
+def test_hist_encode_decode_fuzz(rows, reset_at):
+    """Arbitrary cumulative-LE histogram streams with arbitrary resets must
+    survive the sect-delta encoder bit-exactly, with TypeDrop corrections
+    equal to the raw value before each drop (SectDeltaHistogramReader
+    semantics; drop rule of DeltaSectDiffPackSink, HistogramVector.scala:
+    491-545)."""
+    oracle = _oracle()
+    from test_hist import make_hist_store
+    n = len(rows)
+    cum = np.cumsum(np.array(rows, dtype=np.uint64), axis=1)  # cumulative-LE
+    running = cum.copy()
+    for i in range(1, n):
+        if i not in reset_at:
+            running[i] = running[i - 1] + cum[i]   # keep counters increasing
+    ts = (100000 + np.arange(n) * 15000).astype(np.int64)
+    st_ = make_hist_store(fdb_mod, [(ts, running)], nb=4)
+    _, vab, _, _, _ = st_.chunk(0, 0)
+    dec = oracle.hist_decode(vab)
+    np.testing.assert_array_equal(dec, running.astype(np.int64))
+    # corrections: walk the decoded values with the reference's rule — a new
+    # section starts TypeDrop iff some bucket-consecutive delta decreased
+    corr = oracle.hist_corrections(vab)
+    assert corr.shape == dec.shape
+    # invariant: corrected values are nondecreasing over time per bucket
+    corrected = dec + corr
+    assert (np.diff(corrected, axis=0) >= 0).all()

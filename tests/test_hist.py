@@ -294,3 +294,46 @@ def test_hist_multichunk_oracle_vs_naive(fdb, oracle):
                 assert cnts[w] == 1, w
                 np.testing.assert_allclose(got, expect, rtol=1e-9, atol=1e-12,
                                            err_msg=f"window {w}")
+
+
+# HistogramVectorTest.scala:276-307 — the 14-element bucketData fixture whose
+# SectDelta encoding must place TypeDrop sections at elements 6 and 10
+# (`reader.dropPositions shouldEqual debox.Buffer(6, 10)`).
+BUCKET_DATA_14 = [
+    [0, 0, 1], [0, 2, 3], [2, 5, 6], [2, 5, 9], [2, 5, 10], [2, 8, 14],
+    [0, 0, 2], [1, 7, 9], [1, 15, 19], [2, 16, 21],
+    [0, 1, 1], [0, 15, 15], [1, 16, 19], [4, 20, 25],
+]
+
+
+def test_drop_positions_golden(fdb, oracle):
+    """Our encoder + oracle must reproduce the reference's drop positions and
+    correction amounts on its own fixture: drops at elements 6 and 10, each
+    correction = the raw value just before the drop
+    (SectDeltaHistogramReader.corrections, HistogramVector.scala:683-699)."""
+    data = np.array(BUCKET_DATA_14, dtype=np.uint64)
+    ts = (100000 + np.arange(len(data)) * 10000).astype(np.int64)
+    st = make_hist_store(fdb, [(ts, data)], nb=3)
+    _, vab, _, _, _ = st.chunk(0, 0)
+    dec = oracle.hist_decode(vab)
+    np.testing.assert_array_equal(dec, data.astype(np.int64))
+    corr = oracle.hist_corrections(vab)
+    expect = np.zeros_like(dec)
+    expect[6:] += dec[5]          # dropPosition 6: += apply(5)
+    expect[10:] += dec[9]         # dropPosition 10: += apply(9)
+    np.testing.assert_array_equal(corr, expect)
+
+
+def test_update_correction_golden(fdb, oracle):
+    """HistogramVectorTest.scala:333-363: incrHistBuckets appended twice gives
+    one normal + one TypeDrop section; updateCorrection(NoCorrection) yields
+    correction == lastIncrHist (the full cumulative last row)."""
+    incr = np.cumsum(np.array(RAW_HIST_BUCKETS, dtype=np.uint64), axis=0)
+    data = np.vstack([incr, incr])          # second copy restarts => drop at 4
+    ts = (100000 + np.arange(8) * 10000).astype(np.int64)
+    st = make_hist_store(fdb, [(ts, data)], nb=8)
+    _, vab, _, _, _ = st.chunk(0, 0)
+    corr = oracle.hist_corrections(vab)
+    last_incr = incr[-1].astype(np.int64)
+    np.testing.assert_array_equal(corr[:4], np.zeros((4, 8), dtype=np.int64))
+    np.testing.assert_array_equal(corr[4:], np.tile(last_incr, (4, 1)))
