@@ -1402,11 +1402,6 @@ __global__ void group_reduce_kernel(const double* __restrict__ grid,
   if (sq) sq[idx] = q0 + q1;
 }
 
-__global__ void fill_f64_kernel(double* p, size_t n, double v) {
-  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) p[i] = v;
-}
-
 // ---------------------------------------------------------------------------
 // host engine
 // ---------------------------------------------------------------------------
