@@ -1,0 +1,154 @@
+"""Window-machinery goldens mined from the reference's own WindowIteratorSpec
+(query/src/test/scala/filodb/query/exec/WindowIteratorSpec.scala): literal
+sample streams with literal expected per-window outputs through the chunked
+iterator — the exact results ChunkedWindowIteratorD produces, filtered of NaN
+windows the way the spec filters them."""
+import numpy as np
+import pytest
+
+from conftest import build_store
+
+SAMPLES_GAUGE = [
+    (100000, 1.0), (153000, 2.0), (250000, 3.0), (270000, 4.0), (280000, 5.0),
+    (360000, 6.0), (430000, 7.0), (690000, 8.0), (700000, 9.0),
+    (710000, float("nan")),   # Prom end-of-time-series marker
+]
+
+
+def _eval(fdb, oracle, samples, start, step, end, window, func,
+          kind=None):
+    kind = kind if kind is not None else fdb.COL_GAUGE
+    st = build_store(fdb, [[[(t, v) for t, v in samples]]], kind=kind)
+    q = fdb.make_query(start, step, end, window, func)
+    out = oracle.eval_series(st.view(), 0, q, q.num_windows)
+    return {start + w * step: out[w] for w in range(q.num_windows)}
+
+
+@pytest.mark.parametrize("func,expected", [
+    # WindowIteratorSpec.scala:180-218 (SumOverTime incl. the NaN marker)
+    ("sum", {150000: 1.0, 250000: 5.0, 350000: 12.0, 450000: 13.0,
+             750000: 17.0}),
+    # :466-502 (AvgOverTime)
+    ("avg", {150000: 1.0, 250000: 2.5, 350000: 4.0, 450000: 6.5}),
+    # :503-539 (CountOverTime)
+    ("count", {150000: 1.0, 250000: 2.0, 350000: 3.0, 450000: 2.0}),
+    # :593-629 (MinOverTime)
+    ("min", {150000: 1.0, 250000: 2.0, 350000: 3.0, 450000: 6.0}),
+    # :630-667 (MaxOverTime)
+    ("max", {150000: 1.0, 250000: 3.0, 350000: 5.0, 450000: 7.0}),
+])
+def test_window_iterator_goldens(fdb, oracle, func, expected):
+    ids = {"sum": fdb.FN_SUM_OVER_TIME, "avg": fdb.FN_AVG_OVER_TIME,
+           "count": fdb.FN_COUNT_OVER_TIME, "min": fdb.FN_MIN_OVER_TIME,
+           "max": fdb.FN_MAX_OVER_TIME}
+    end = 1100000 if func == "sum" else 700000
+    got = _eval(fdb, oracle, SAMPLES_GAUGE, 50000, 100000, end, 100000,
+                ids[func])
+    finite = {t: v for t, v in got.items() if not np.isnan(v)}
+    assert finite == pytest.approx(expected)
+
+
+def test_prometheus_rate_golden(fdb, oracle):
+    """:219-256 — rate() matching the Prometheus rate function on its sample
+    stream, to 1e-10 as the spec asserts."""
+    samples = [(1548191486000 + i * 10000, v) for i, v in enumerate(
+        [84.0, 152.0, 195.0, 222.0, 245.0, 251.0, 329.0, 374.0, 431.0])]
+    expected = {
+        1548191496000: 0.34,
+        1548191511000: 0.555,
+        1548191526000: 0.60375,
+        1548191541000: 0.668,
+        1548191556000: 1.0357142857142858,
+    }
+    got = _eval(fdb, oracle, samples, 1548191496000, 15000, 1548191796000,
+                300000, fdb.FN_RATE, kind=fdb.COL_COUNTER)
+    for t, v in expected.items():
+        assert got[t] == pytest.approx(v, abs=1e-10), t
+
+
+def test_rate_nan_markers_golden(fdb, oracle):
+    """:257-285 — NaN end-of-series markers interleaved with counter resets;
+    the instant-query rate must equal the spec's 0.5870753512132821."""
+    samples = [
+        (1614821996000, float("nan")), (1614821996100, 489.0),
+        (1614821997000, float("nan")), (1614822566000, 19.0),
+        (1614822596000, 26.0), (1614822626000, 26.0), (1614822656000, 26.0),
+        (1614822686000, 26.0), (1614822716000, 26.0),
+        (1614822717000, float("nan")), (1614822866000, 5.0),
+    ]
+    got = _eval(fdb, oracle, samples, 1614822880000, 15000, 1614822880000,
+                900000, fdb.FN_RATE, kind=fdb.COL_COUNTER)
+    assert got[1614822880000] == pytest.approx(0.5870753512132821, abs=1e-12)
+
+
+def test_last_sample_stale_golden(fdb, oracle):
+    """:436-465 — default instant selector: value present at
+    time - staleSampleAfterMs (300000, filodb-defaults.conf:604) is returned;
+    older samples go stale (NaN)."""
+    samples = [(100000, 100.0), (153000, 160.0), (200000, 200.0)]
+    got = _eval(fdb, oracle, samples, 100000, 100000, 600000, 300001,
+                fdb.FN_LAST)
+    finite = {t: v for t, v in got.items() if not np.isnan(v)}
+    assert finite == pytest.approx({100000: 100.0, 200000: 200.0,
+                                    300000: 200.0, 400000: 200.0,
+                                    500000: 200.0})
+
+
+@pytest.mark.gpu
+def test_window_iterator_goldens_on_gpu(fdb, oracle):
+    """The same WindowIteratorSpec literal expectations through the GPU
+    engine (AGG_NONE [S×W] grid) — reference numbers, not oracle numbers."""
+    engine = fdb.Engine(0)
+    cases = [
+        (fdb.FN_SUM_OVER_TIME, 1100000,
+         {150000: 1.0, 250000: 5.0, 350000: 12.0, 450000: 13.0, 750000: 17.0}),
+        (fdb.FN_AVG_OVER_TIME, 700000,
+         {150000: 1.0, 250000: 2.5, 350000: 4.0, 450000: 6.5}),
+        (fdb.FN_COUNT_OVER_TIME, 700000,
+         {150000: 1.0, 250000: 2.0, 350000: 3.0, 450000: 2.0}),
+        (fdb.FN_MIN_OVER_TIME, 700000,
+         {150000: 1.0, 250000: 2.0, 350000: 3.0, 450000: 6.0}),
+        (fdb.FN_MAX_OVER_TIME, 700000,
+         {150000: 1.0, 250000: 3.0, 350000: 5.0, 450000: 7.0}),
+    ]
+    st = build_store(fdb, [[[(t, v) for t, v in SAMPLES_GAUGE]]])
+    ds = engine.upload(st)
+    for func, end, expected in cases:
+        q = fdb.make_query(50000, 100000, end, 100000, func)
+        nw = q.num_windows
+        out = np.empty(nw)
+        engine.query(ds, q, out=out)
+        finite = {50000 + w * 100000: out[w] for w in range(nw)
+                  if not np.isnan(out[w])}
+        assert finite == pytest.approx(expected), func
+
+    # rate on the Prometheus fixture
+    samples = [(1548191486000 + i * 10000, v) for i, v in enumerate(
+        [84.0, 152.0, 195.0, 222.0, 245.0, 251.0, 329.0, 374.0, 431.0])]
+    st2 = build_store(fdb, [[[(t, v) for t, v in samples]]],
+                      kind=fdb.COL_COUNTER)
+    q = fdb.make_query(1548191496000, 15000, 1548191796000, 300000,
+                       fdb.FN_RATE)
+    out = np.empty(q.num_windows)
+    engine.query(engine.upload(st2), q, out=out)
+    got = {1548191496000 + w * 15000: out[w] for w in range(q.num_windows)}
+    for t, v in {1548191496000: 0.34, 1548191511000: 0.555,
+                 1548191526000: 0.60375, 1548191541000: 0.668,
+                 1548191556000: 1.0357142857142858}.items():
+        assert got[t] == pytest.approx(v, abs=1e-10), t
+
+    # NaN-marker counter instant rate
+    samples3 = [
+        (1614821996000, float("nan")), (1614821996100, 489.0),
+        (1614821997000, float("nan")), (1614822566000, 19.0),
+        (1614822596000, 26.0), (1614822626000, 26.0), (1614822656000, 26.0),
+        (1614822686000, 26.0), (1614822716000, 26.0),
+        (1614822717000, float("nan")), (1614822866000, 5.0),
+    ]
+    st3 = build_store(fdb, [[[(t, v) for t, v in samples3]]],
+                      kind=fdb.COL_COUNTER)
+    q = fdb.make_query(1614822880000, 15000, 1614822880000, 900000,
+                       fdb.FN_RATE)
+    out = np.empty(1)
+    engine.query(engine.upload(st3), q, out=out)
+    assert out[0] == pytest.approx(0.5870753512132821, abs=1e-12)
