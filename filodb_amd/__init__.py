@@ -23,7 +23,7 @@ FN_LAST = 12
 # aggregation ids (RowAggregator implementations)
 AGG_NONE, AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX, AGG_AVG = 0, 1, 2, 3, 4, 5
 AGG_TOPK, AGG_BOTTOMK = 6, 7
-AGG_STDDEV, AGG_STDVAR = 8, 9
+AGG_STDDEV, AGG_STDVAR, AGG_GROUP = 8, 9, 10
 # column kinds
 COL_GAUGE, COL_COUNTER, COL_HIST = 0, 1, 2
 

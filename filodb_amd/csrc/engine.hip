@@ -46,7 +46,7 @@ void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12 };
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
-       AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9 };
+       AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9, AGG_GROUP=10 };
 
 // ---------------------------------------------------------------------------
 // device-side vector readers (same layouts as oracle)
@@ -1281,7 +1281,8 @@ __global__ void agg_present_kernel(double* out, const double* cnt, const double*
   if (i >= n) return;
   if (partial) return;                      // partial mode: leave raw sums + counts
   if (cnt[i] <= 0) { out[i] = NAN; return; }
-  if (agg_id == AGG_AVG) out[i] = out[i] / cnt[i];
+  if (agg_id == AGG_GROUP) out[i] = 1.0;   // GroupRowAggregator.scala:12-31
+  else if (agg_id == AGG_AVG) out[i] = out[i] / cnt[i];
   else if (agg_id == AGG_STDDEV || agg_id == AGG_STDVAR) {
     double mean = out[i] / cnt[i];
     double var = sq[i] / cnt[i] - mean * mean;   // StddevRowAggregator.scala:49-52
@@ -1367,7 +1368,7 @@ __global__ void group_reduce_kernel(const double* __restrict__ grid,
       acc += x;
       s2 += x * x;
     }
-    out[idx] = agg_id == AGG_COUNT ? c : acc;
+    out[idx] = (agg_id == AGG_COUNT || agg_id == AGG_GROUP) ? c : acc;
     cnt[idx] = c;
     if (sq) sq[idx] = s2;
     return;
@@ -1397,7 +1398,8 @@ __global__ void group_reduce_kernel(const double* __restrict__ grid,
     a0 += x; c0 += 1.0; q0 += x * x;
   }
   double c = c0 + c1;
-  out[idx] = agg_id == AGG_COUNT ? c : (a0 + a1) + (a2 + a3);
+  out[idx] = (agg_id == AGG_COUNT || agg_id == AGG_GROUP)
+                 ? c : (a0 + a1) + (a2 + a3);
   cnt[idx] = c;
   if (sq) sq[idx] = q0 + q1;
 }

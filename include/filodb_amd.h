@@ -173,9 +173,11 @@ int32_t fdb_nibblepack_pack_doubles(const double* in, int32_t n, uint8_t* out, i
                               (-1 padding). */
 #define FDB_AGG_BOTTOMK 7  /* same, k smallest; values sorted ascending */
 #define FDB_AGG_STDDEV  8  /* StddevRowAggregator.scala:39-58 (merge is
-                              algebraically sum/sumsq/count; presented only —
-                              cross-shard partials are round-2) */
+                              algebraically sum/sumsq/count; partials stack
+                              raw sums+sumsq in a 2x out grid, see below) */
 #define FDB_AGG_STDVAR  9  /* StdvarRowAggregator (same, without the sqrt) */
+#define FDB_AGG_GROUP  10  /* GroupRowAggregator: 1 where any non-NaN row
+                              contributed, NaN otherwise */
 
 typedef struct {
   int64_t start;     /* first window end timestamp (ms)                      */

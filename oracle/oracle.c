@@ -53,7 +53,7 @@ typedef struct {
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12 };
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
-       AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9 };
+       AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9, AGG_GROUP=10 };
 
 static inline uint16_t rd_u16(const uint8_t* p) { uint16_t v; memcpy(&v, p, 2); return v; }
 static inline uint32_t rd_u32(const uint8_t* p) { uint32_t v; memcpy(&v, p, 4); return v; }
@@ -691,9 +691,11 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
             if (isnan(acc[w])) acc[w] = 0;
             acc[w] += x;
             break;
-          case AGG_COUNT:
+          case AGG_COUNT: case AGG_GROUP:
             /* CountRowAggregator.scala:36-42: sample maps to 1 (0 when NaN); the
-             * accumulator leaves NaN only when every mapped value was 0. */
+             * accumulator leaves NaN only when every mapped value was 0.
+             * GROUP folds identically; its present step emits 1
+             * (GroupRowAggregator.scala:12-31). */
             if (isnan(acc[w])) acc[w] = 0;
             acc[w] += 1;
             break;
@@ -732,7 +734,7 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
       if (isnan(x)) continue;
       switch (q->agg_id) {
         case AGG_SUM: case AGG_COUNT: case AGG_AVG:
-        case AGG_STDDEV: case AGG_STDVAR:
+        case AGG_STDDEV: case AGG_STDVAR: case AGG_GROUP:
           if (isnan(out[i])) out[i] = 0;
           out[i] += x;
           if (sq_total) sq_total[i] += gq[i];
@@ -753,13 +755,17 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
      * `out` is [2 x G x W]: raw sums then raw sumsq — merging by addition is
      * algebraically StddevRowAggregator.scala:36-52's reduction. */
     if (q->agg_id == AGG_SUM || q->agg_id == AGG_COUNT || q->agg_id == AGG_AVG ||
-        needs_sq)
+        q->agg_id == AGG_GROUP || needs_sq)
       for (size_t i = 0; i < gridlen; i++)
         if (isnan(out[i]) && cnt_total[i] == 0) out[i] = 0;
     if (needs_sq)
       for (size_t i = 0; i < gridlen; i++) out[gridlen + i] = sq_total[i];
     /* MIN/MAX partials keep NaN for empty cells; the merging caller maps them
      * to ±inf before the collective. */
+  } else if (q->agg_id == AGG_GROUP) {
+    /* present: 1 wherever any row contributed (GroupRowAggregator.scala:12-31) */
+    for (size_t i = 0; i < gridlen; i++)
+      if (!isnan(out[i])) out[i] = 1;
   } else if (q->agg_id == AGG_AVG) {
     /* present: mean = sum/count (AvgRowAggregator.scala:38-46 algebraically) */
     for (size_t i = 0; i < gridlen; i++)

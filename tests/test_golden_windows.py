@@ -152,3 +152,53 @@ def test_window_iterator_goldens_on_gpu(fdb, oracle):
     out = np.empty(1)
     engine.query(engine.upload(st3), q, out=out)
     assert out[0] == pytest.approx(0.5870753512132821, abs=1e-12)
+
+
+# ---------------------------------------------------------------------------
+# Cross-series aggregator goldens from AggrOverRangeVectorsSpec.scala
+# ---------------------------------------------------------------------------
+
+def _agg_grid(fdb, oracle, series_samples, agg_id, start, step, end):
+    """Aligned per-timestamp cross-series aggregation: window=0 makes each
+    window exactly one sample, mirroring mapReduce over TransientRows."""
+    st = build_store(fdb, [[[(t, v) for t, v in s]] for s in series_samples],
+                     groups=[0] * len(series_samples))
+    q = fdb.make_query(start, step, end, 0, fdb.FN_SUM_OVER_TIME, agg_id, 1)
+    return oracle.query_exec(st.view(), q, st.num_series, q.num_windows)
+
+
+def test_avg_nan_golden(fdb, oracle):
+    """:367-386 — avg of one all-1.0 series and one mostly-NaN series must be
+    1.0 everywhere (NaN rows never contribute)."""
+    nan = float("nan")
+    s1 = [(1541190600 + i * 60, nan) for i in range(5)] + \
+         [(1541190900, 1.0), (1541190960, 1.0)]
+    s2 = [(1541190600 + i * 60, 1.0) for i in range(7)]
+    grid = _agg_grid(fdb, oracle, [s1, s2], fdb.AGG_AVG,
+                     1541190600, 60, 1541190960)
+    np.testing.assert_allclose(grid, np.ones(7), rtol=1e-12)
+
+
+def test_stddev_stdvar_nan_golden(fdb, oracle):
+    """:387-419 — 11 series (8 all-NaN) at two timestamps; the spec's literal
+    stdvar/stddev values."""
+    nan = float("nan")
+    vals = [(3247.0, 3297.0)] + [(nan, nan)] * 6 + [(5173.0, 5173.0),
+            (nan, nan), (11583.0, 11583.0), (nan, nan)]
+    series = [[(1, a), (2, b)] for a, b in vals]
+    got_var = _agg_grid(fdb, oracle, series, fdb.AGG_STDVAR, 1, 1, 2)
+    np.testing.assert_allclose(
+        got_var, [12698496.88888889, 12585030.222222222], rtol=1e-12)
+    got_dev = _agg_grid(fdb, oracle, series, fdb.AGG_STDDEV, 1, 1, 2)
+    np.testing.assert_allclose(
+        got_dev, [3563.4950384263, 3547.5386146203], rtol=1e-10)
+
+
+def test_group_aggregator_golden(fdb, oracle):
+    """GroupRowAggregator (aggregator/GroupRowAggregator.scala:12-31): 1 where
+    any series contributed a non-NaN row, NaN where all were NaN."""
+    nan = float("nan")
+    s1 = [(1, 5.0), (2, nan), (3, nan)]
+    s2 = [(1, 7.0), (2, 3.0), (3, nan)]
+    grid = _agg_grid(fdb, oracle, [s1, s2], fdb.AGG_GROUP, 1, 1, 3)
+    assert grid[0] == 1.0 and grid[1] == 1.0 and np.isnan(grid[2])

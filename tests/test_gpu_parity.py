@@ -14,7 +14,7 @@ pytestmark = pytest.mark.gpu
 FUNCS = {"rate": 0, "increase": 1, "delta": 2, "sum": 3, "count": 4, "avg": 5,
          "min": 6, "max": 7, "stddev": 8, "stdvar": 9, "changes": 10, "last": 12}
 AGGS = {"sum": 1, "count": 2, "min": 3, "max": 4, "avg": 5,
-        "stddev": 8, "stdvar": 9}
+        "stddev": 8, "stdvar": 9, "group": 10}
 
 
 @pytest.fixture(scope="module")

@@ -10,7 +10,7 @@ import pytest
 from conftest import build_store, synth_gauge_series
 
 AGGS = {"sum": 1, "count": 2, "min": 3, "max": 4, "avg": 5,
-        "stddev": 8, "stdvar": 9}
+        "stddev": 8, "stdvar": 9, "group": 10}
 
 
 def make_multi(fdb, rng, n_series=40, n_groups=5, n=60, nan_p=0.15):
@@ -59,6 +59,8 @@ def test_group_reduce_vs_numpy(fdb, oracle, agg):
             elif agg == "stddev":
                 exp = np.nan if len(nn) == 0 else np.sqrt(
                     (nn * nn).mean() - nn.mean() ** 2)
+            elif agg == "group":
+                exp = np.nan if len(nn) == 0 else 1.0
             else:  # stdvar
                 exp = np.nan if len(nn) == 0 else (nn * nn).mean() - nn.mean() ** 2
             got = grid[g * nw + w]
