@@ -51,7 +51,8 @@ typedef struct {
 } fdb_view_t;
 
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
-       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12 };
+       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12,
+       FN_PRESENT=13, FN_TIMESTAMP=14, FN_ZSCORE=15 };
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
        AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9, AGG_GROUP=10 };
 
@@ -471,6 +472,7 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
       double sum = NAN, count = NAN, sqsum = NAN, mn = NAN, mx = NAN;
       double changes = NAN, prev = NAN;
       double last_val = NAN;
+      double last_sample = NAN;    /* zscore: endRow value when non-NaN */
       int64_t last_ts = -1;
       int icount = 0;
       for (int c = 0; c < nchunks; c++) {
@@ -480,12 +482,32 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
         int startRow = lv_binary_search(tv, wStart) & 0x7fffffff;
         int endRow = lv_ceiling(tv, wEnd);
         if (endRow > dir[c].num_rows - 1) endRow = dir[c].num_rows - 1;
-        if (q->func_id == FN_LAST) {
+        if (q->func_id == FN_LAST || q->func_id == FN_PRESENT) {
           /* LastSampleChunkedFunction.addChunks (RangeFunction.scala:599-614):
-           * no startRow search; last ts <= wEnd wins if within the window */
+           * no startRow search; last ts <= wEnd wins if within the window.
+           * PresentOverTimeChunkedFunctionD (:725-745): non-NaN -> 1; a NaN
+           * stale marker steps back one row (marker-before-marker -> NaN). */
           if (endRow >= 0) {
             int64_t t = lv_at(tv, endRow);
-            if (t >= wStart && t > last_ts) { last_ts = t; last_val = dv_at(vv, endRow); }
+            if (t >= wStart && t > last_ts) {
+              double v = dv_at(vv, endRow);
+              if (q->func_id == FN_LAST) { last_ts = t; last_val = v; }
+              else if (!isnan(v)) { last_ts = t; last_val = 1; }
+              else if (endRow > 0) {
+                last_ts = t;
+                last_val = isnan(dv_at(vv, endRow - 1)) ? NAN : 1;
+              }
+            }
+          }
+          if (dir[c].end_time >= wEnd) break;
+          continue;
+        }
+        if (q->func_id == FN_TIMESTAMP) {
+          /* TimestampChunkedFunction (RangeFunction.scala:705-723): last
+           * ts <= wEnd in the chunk list, no window-start bound; seconds */
+          if (endRow >= 0) {
+            int64_t t = lv_at(tv, endRow);
+            if (t > last_ts) { last_ts = t; last_val = (double)t / 1000.0; }
           }
           if (dir[c].end_time >= wEnd) break;
           continue;
@@ -515,14 +537,16 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
                 else mx = isnan(mx) || x > mx ? x : mx;
               }
             } break;
-            case FN_STDDEV: case FN_STDVAR: {
-              /* VarOverTimeChunkedFunctionD :1082-1115 */
+            case FN_STDDEV: case FN_STDVAR: case FN_ZSCORE: {
+              /* VarOverTimeChunkedFunctionD :1082-1115; lastSample is set
+               * only when the chunk range's endRow value is non-NaN (:1103) */
               double cs = NAN, csq = NAN; int cc = 0;
               for (int i = startRow; i <= endRow; i++) {
                 double x = dv_at(vv, i);
                 if (!isnan(x)) {
                   if (isnan(cs)) cs = 0;
                   if (isnan(csq)) csq = 0;
+                  if (i == endRow) last_sample = x;
                   cs += x; csq += x * x; cc++;
                 }
               }
@@ -559,7 +583,17 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
           result = r;
         } break;
         case FN_CHANGES: result = changes; break;
-        case FN_LAST:    result = last_val; break;
+        case FN_LAST: case FN_PRESENT: case FN_TIMESTAMP:
+          result = last_val; break;
+        case FN_ZSCORE: {
+          /* ZScoreChunkedFunctionD (AggrOverTimeFunctions.scala:1592-1603) */
+          if (icount > 0) {
+            double avg = sum / icount;
+            double sd = sqrt(sqsum / icount - avg * avg);
+            result = (last_sample - avg) / sd;
+          } else if (isnan(sum)) result = sum;
+          else result = 0;
+        } break;
       }
     }
     out[w] = result;

@@ -70,7 +70,8 @@ def test_encode_decode_roundtrip(deltas, vals):
        window=st.integers(min_value=1, max_value=20_000),
        step=st.integers(min_value=1, max_value=5_000),
        func=st.sampled_from(["sum", "count", "avg", "min", "max",
-                             "stddev", "stdvar", "changes", "last"]))
+                             "stddev", "stdvar", "changes", "last",
+                             "present", "timestamp", "zscore"]))
 def test_gauge_window_parity_fuzz(deltas, vals, nchunks, window, step, func):
     """Gauge family under arbitrary samples/NaNs/chunking/windows: the oracle's
     chunked evaluation equals the naive per-window model (the same one

@@ -40,7 +40,7 @@ def naive_window(ts, vs, chunk_bounds, w_start, w_end, func):
             if len(nn):
                 mn = np.nanmin([mn, nn.min()]) if not np.isnan(mn) else nn.min()
                 mx = np.nanmax([mx, nn.max()]) if not np.isnan(mx) else nn.max()
-        elif func in ("stddev", "stdvar"):
+        elif func in ("stddev", "stdvar", "zscore"):
             cs = np.nan if len(nn) == 0 else nn.sum()
             csq = np.nan if len(nn) == 0 else (nn * nn).sum()
             if not np.isnan(cs) and np.isnan(sum_):
@@ -83,11 +83,61 @@ def naive_window(ts, vs, chunk_bounds, w_start, w_end, func):
         # the window; NaN stale markers propagate (RangeFunction.scala:595-694)
         m = (ts >= w_start) & (ts <= w_end)
         return vs[m][-1] if m.any() else np.nan
+    if func == "present":
+        # PresentOverTimeChunkedFunctionD (RangeFunction.scala:725-745): 1 for
+        # a non-NaN last sample; a NaN stale marker steps back ONE chunk row
+        best_t, val = -1, np.nan
+        for lo, hi in chunk_bounds:
+            cts, cvs = ts[lo:hi], vs[lo:hi]
+            if cts[-1] < w_start:
+                continue
+            idx = np.nonzero(cts <= w_end)[0]
+            if len(idx) == 0:
+                continue
+            e = int(idx[-1])
+            t = int(cts[e])
+            if t >= w_start and t > best_t:
+                if not np.isnan(cvs[e]):
+                    best_t, val = t, 1.0
+                elif e > 0:
+                    best_t, val = t, (np.nan if np.isnan(cvs[e - 1]) else 1.0)
+        return val
+    if func == "timestamp":
+        # TimestampChunkedFunction (RangeFunction.scala:705-723): last ts <=
+        # wEnd over the window's chunk list (no window-start bound), seconds
+        best_t = -1
+        for lo, hi in chunk_bounds:
+            cts = ts[lo:hi]
+            if cts[-1] < w_start:
+                continue
+            idx = np.nonzero(cts <= w_end)[0]
+            if len(idx) and int(cts[idx[-1]]) > best_t:
+                best_t = int(cts[idx[-1]])
+        return best_t / 1000.0 if best_t >= 0 else np.nan
+    if func == "zscore":
+        # ZScoreChunkedFunctionD (AggrOverTimeFunctions.scala:1592-1603) over
+        # VarOverTime accumulation; lastSample only from a non-NaN range end
+        last = np.nan
+        for lo, hi in chunk_bounds:
+            cts, cvs = ts[lo:hi], vs[lo:hi]
+            if cts[-1] < w_start:
+                continue
+            m = (cts >= w_start) & (cts <= w_end)
+            if m.any():
+                e = int(np.nonzero(m)[0][-1])
+                if not np.isnan(cvs[e]):
+                    last = cvs[e]
+        if icount > 0:
+            avg = sum_ / icount
+            sd = np.sqrt(sqsum / icount - avg * avg)
+            return (last - avg) / sd
+        return sum_ if np.isnan(sum_) else 0.0
     raise ValueError(func)
 
 
 FUNC_IDS = {"sum": 3, "count": 4, "avg": 5, "min": 6, "max": 7,
-            "stddev": 8, "stdvar": 9, "changes": 10, "last": 12}
+            "stddev": 8, "stdvar": 9, "changes": 10, "last": 12,
+            "present": 13, "timestamp": 14, "zscore": 15}
 
 
 @pytest.mark.parametrize("func", list(FUNC_IDS))
