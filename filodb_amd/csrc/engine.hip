@@ -44,7 +44,8 @@ void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
 #define BLOCK_THREADS (WAVES_PER_BLOCK * 64)
 
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
-       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12 };
+       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12,
+       FN_PRESENT=13, FN_TIMESTAMP=14, FN_ZSCORE=15 };
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
        AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9, AGG_GROUP=10 };
 
@@ -162,7 +163,8 @@ template <int FUNC> struct KKind { static constexpr int v =
     (FUNC == FN_SUM || FUNC == FN_AVG || FUNC == FN_COUNT) ? K_PFX :
     (FUNC == FN_STDDEV || FUNC == FN_STDVAR) ? K_PFX_SQ :
     (FUNC == FN_MIN || FUNC == FN_MAX) ? K_MINMAX :
-    (FUNC == FN_LAST) ? K_LAST : K_CHANGES; };
+    (FUNC == FN_LAST || FUNC == FN_PRESENT || FUNC == FN_TIMESTAMP)
+        ? K_LAST : K_CHANGES; };   // FN_ZSCORE uses K_CHANGES raw values
 
 #define FDB_NGROUPS ((FDB_MAX_ROWS_PER_SERIES + 7) / 8)
 
@@ -655,7 +657,13 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       const int64_t wEnd = qstart + (int64_t)w * qstep;
       const int64_t wStart = wEnd - qwindow;
       double result = NAN;
-      {
+      if constexpr (FUNC == FN_TIMESTAMP) {
+        // TimestampChunkedFunction (RangeFunction.scala:705-723): last ts <=
+        // wEnd in the window's chunk list, no window-start bound; seconds
+        int endRow = sRk[k] == 0x7fffffff ? -1 : eRk[k];
+        if (endRow >= 0 && endRow < mr.nrows)
+          result = (double)seg[endRow] / 1000.0;
+      } else {
         int startRow = sRk[k] == 0x7fffffff ? 1 : sRk[k];
         int endRow = sRk[k] == 0x7fffffff ? 0 : eRk[k];
         if (startRow <= endRow && endRow < mr.nrows) {
@@ -702,9 +710,32 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
               for (int i = startRow; i <= endRow; i++) acc(ws.val[i]);
             }
             result = mm;
+          } else if constexpr (FUNC == FN_PRESENT) {
+            // PresentOverTimeChunkedFunctionD (RangeFunction.scala:725-745):
+            // non-NaN -> 1; a NaN stale marker steps back one row
+            double v = ws.val[endRow];
+            if (!isnan(v)) result = 1.0;
+            else if (endRow > 0)
+              result = isnan(ws.val[endRow - 1]) ? NAN : 1.0;
           } else if constexpr (KIND == K_LAST) {
             // last sample <= wEnd within window; raw value (NaN propagates)
             result = seg[endRow] >= wStart ? ws.val[endRow] : NAN;
+          } else if constexpr (FUNC == FN_ZSCORE) {
+            // ZScoreChunkedFunctionD (AggrOverTimeFunctions.scala:1592-1603)
+            double s = NAN, sq = NAN, lastv = NAN;
+            int pc = 0;
+            for (int i = startRow; i <= endRow; i++) {
+              double x = ws.val[i];
+              if (isnan(x)) continue;
+              if (isnan(s)) { s = 0; sq = 0; }
+              if (i == endRow) lastv = x;
+              s += x; sq += x * x; pc++;
+            }
+            if (pc > 0) {
+              double avg = s / pc;
+              double sd = sqrt(sq / pc - avg * avg);
+              result = (lastv - avg) / sd;
+            } else result = isnan(s) ? s : 0;
           } else {  // K_CHANGES, single chunk: prefix diff, prev starts NaN
             result = (double)((int)ws.cnt[endRow] - (int)ws.cnt[startRow]);
           }
@@ -803,10 +834,26 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         int endRow = lds_search_le(seg, m.nrows, wEnd, m.ts0, m.inv_slope);
         if constexpr (KIND == K_LAST) {
           // LastSampleChunkedFunction.addChunks (RangeFunction.scala:599-614):
-          // no startRow search; last ts <= wEnd wins if within the window
+          // no startRow search; last ts <= wEnd wins if within the window.
+          // FN_PRESENT maps to 1 with a one-row step-back on NaN markers
+          // (:725-745); FN_TIMESTAMP takes the bare ts without the
+          // window-start bound (:705-723).
           if (endRow >= 0 && endRow < m.nrows) {
             int64_t t = seg[endRow];
-            if (t >= wStart && t > last_ts) { last_ts = t; last_val = ws.val[m.row0 + endRow]; }
+            if constexpr (FUNC == FN_TIMESTAMP) {
+              if (t > last_ts) { last_ts = t; last_val = (double)t / 1000.0; }
+            } else if constexpr (FUNC == FN_PRESENT) {
+              if (t >= wStart && t > last_ts) {
+                double v = ws.val[m.row0 + endRow];
+                if (!isnan(v)) { last_ts = t; last_val = 1.0; }
+                else if (endRow > 0) {
+                  last_ts = t;
+                  last_val = isnan(ws.val[m.row0 + endRow - 1]) ? NAN : 1.0;
+                }
+              }
+            } else {
+              if (t >= wStart && t > last_ts) { last_ts = t; last_val = ws.val[m.row0 + endRow]; }
+            }
           }
           if (m.end_time >= wEnd) break;
           continue;
@@ -843,6 +890,24 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
             } else {
               for (int i = a; i <= b; i++) acc(ws.val[i]);
             }
+          } else if constexpr (FUNC == FN_ZSCORE) {
+            // VarOverTimeChunkedFunctionD accumulation over raw values
+            // (AggrOverTimeFunctions.scala:1082-1115): lastSample set only
+            // when the range's endRow value is non-NaN (:1103)
+            double cs = NAN, csq = NAN;
+            int cc = 0;
+            for (int i = a; i <= b; i++) {
+              double x = ws.val[i];
+              if (isnan(x)) continue;
+              if (isnan(cs)) { cs = 0; csq = 0; }
+              if (i == b) last_val = x;       // lastSample carry across chunks
+              cs += x; csq += x * x; cc++;
+            }
+            if (!isnan(cs) && isnan(sum)) sum = 0;
+            sum += cs;
+            if (!isnan(csq) && isnan(sqsum)) sqsum = 0;
+            sqsum += csq;
+            icount += cc;
           } else {  // K_CHANGES
             if (isnan(changes)) changes = 0;
             // in-chunk changes from the indicator prefix over (startRow, endRow]
@@ -869,7 +934,16 @@ void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
           } else result = isnan(sum) ? sum : 0;
         } break;
         case FN_CHANGES: result = changes; break;
-        case FN_LAST:    result = last_val; break;
+        case FN_LAST: case FN_PRESENT: case FN_TIMESTAMP:
+          result = last_val; break;
+        case FN_ZSCORE: {
+          // ZScoreChunkedFunctionD (AggrOverTimeFunctions.scala:1592-1603)
+          if (icount > 0) {
+            double avg = sum / icount;
+            double sd = sqrt(sqsum / icount - avg * avg);
+            result = (last_val - avg) / sd;
+          } else result = isnan(sum) ? sum : 0;
+        } break;
       }
     }
 
@@ -1605,6 +1679,7 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
     CASE(FN_RATE); CASE(FN_INCREASE); CASE(FN_DELTA); CASE(FN_SUM); CASE(FN_COUNT);
     CASE(FN_AVG); CASE(FN_MIN); CASE(FN_MAX); CASE(FN_STDDEV); CASE(FN_STDVAR);
     CASE(FN_CHANGES); CASE(FN_LAST);
+    CASE(FN_PRESENT); CASE(FN_TIMESTAMP); CASE(FN_ZSCORE);
     default: fdb_set_error("bad func_id %d", q->func_id); return FDB_ERR_BADARG;
   }
   #undef CASE

@@ -156,6 +156,14 @@ int32_t fdb_nibblepack_pack_doubles(const double* in, int32_t n, uint8_t* out, i
                                        (NaN stale markers propagate); the reference uses
                                        window = stale-sample-after + 1 = 300001ms for raw
                                        queries (PeriodicSamplesMapper.scala:79-81) */
+#define FDB_FN_PRESENT         13   /* PresentOverTimeChunkedFunctionD
+                                       (RangeFunction.scala:725-745): 1 for a non-NaN last
+                                       sample; NaN stale markers step back one row */
+#define FDB_FN_TIMESTAMP       14   /* TimestampChunkedFunction (RangeFunction.scala:705-723):
+                                       last sample's timestamp <= wEnd, in seconds */
+#define FDB_FN_ZSCORE          15   /* ZScoreChunkedFunctionD
+                                       (AggrOverTimeFunctions.scala:1592-1603):
+                                       (lastSample - mean) / stddev over the window */
 
 /* Cross-series aggregation: RowAggregator implementations
  * (query/.../exec/aggregator/RowAggregator.scala:28-150). */

@@ -12,7 +12,8 @@ from conftest import build_store, synth_counter_series, synth_gauge_series
 pytestmark = pytest.mark.gpu
 
 FUNCS = {"rate": 0, "increase": 1, "delta": 2, "sum": 3, "count": 4, "avg": 5,
-         "min": 6, "max": 7, "stddev": 8, "stdvar": 9, "changes": 10, "last": 12}
+         "min": 6, "max": 7, "stddev": 8, "stdvar": 9, "changes": 10, "last": 12,
+         "present": 13, "timestamp": 14, "zscore": 15}
 AGGS = {"sum": 1, "count": 2, "min": 3, "max": 4, "avg": 5,
         "stddev": 8, "stdvar": 9, "group": 10}
 
@@ -104,7 +105,8 @@ def test_counter_funcs(fdb, oracle, engine, func, case):
 
 
 @pytest.mark.parametrize("func", ["sum", "count", "avg", "min", "max",
-                                  "stddev", "stdvar", "changes", "last"])
+                                  "stddev", "stdvar", "changes", "last",
+                                  "present", "timestamp", "zscore"])
 @pytest.mark.parametrize("case", ["raw", "nans", "multichunk", "integral"])
 def test_gauge_funcs(fdb, oracle, engine, func, case):
     seed = hash((func, case)) % 2**31
