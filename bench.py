@@ -30,6 +30,13 @@ import time
 
 import numpy as np
 
+# torchrun exports OMP_NUM_THREADS=1, which would serialize each rank's
+# host-side synthetic chunk build (OpenMP in libfilodb_amd). Give every rank a
+# fair share of the host cores instead; must happen BEFORE the library loads.
+_world = int(os.environ.get("WORLD_SIZE", "1"))
+if os.environ.get("OMP_NUM_THREADS") in (None, "1"):
+    os.environ["OMP_NUM_THREADS"] = str(max(1, (os.cpu_count() or 1) // _world))
+
 REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
