@@ -130,7 +130,8 @@ def naive_window(ts, vs, chunk_bounds, w_start, w_end, func):
         if icount > 0:
             avg = sum_ / icount
             sd = np.sqrt(sqsum / icount - avg * avg)
-            return (last - avg) / sd
+            with np.errstate(invalid="ignore", divide="ignore"):
+                return (last - avg) / sd    # 0/0 -> NaN, x/0 -> inf, as in C
         return sum_ if np.isnan(sum_) else 0.0
     raise ValueError(func)
 
