@@ -1012,7 +1012,11 @@ EXPORT int32_t oracle_query_exec_hist(const fdb_view_t* view, const fdb_query_t*
     hist_chunk_t hv[64];
     for (int c = 0; c < nchunks; c++) {
       vec_open(view->blob + dir[c].ts_off, &tsv[c]);
-      if (hist_open(view->blob + dir[c].val_off, &hv[c]) != 0) return -1;
+      if (hist_open(view->blob + dir[c].val_off, &hv[c]) != 0) {
+        for (int c2 = 0; c2 < c; c2++) hist_close(&hv[c2]);
+        free(lastv); free(corr); free(lo); free(hi);
+        return -1;
+      }
     }
     int grp = view->group_ids[sid];
     for (int w = 0; w < nw; w++) {
