@@ -202,3 +202,35 @@ def test_group_aggregator_golden(fdb, oracle):
     s2 = [(1, 7.0), (2, 3.0), (3, nan)]
     grid = _agg_grid(fdb, oracle, [s1, s2], fdb.AGG_GROUP, 1, 1, 3)
     assert grid[0] == 1.0 and grid[1] == 1.0 and np.isnan(grid[2])
+
+
+def test_last_sample_floor_property(fdb, oracle):
+    """LastSampleFunctionSpec's validation property: for sweeps of start
+    offsets and steps over a sparse random series, every window's value is the
+    floor sample (last ts <= wEnd) unless it is older than 5 minutes
+    (LastSampleFunctionSpec.scala:138-158; chunked window w = 300000)."""
+    rng = np.random.default_rng(99)
+    now = 1_600_000_000_000
+    # generateRandomRawCounterSeries-like: ~20-25s intervals, 200 samples
+    gaps = rng.integers(20_000, 25_000, 200)
+    ts = now - 4_000_000 + np.cumsum(gaps).astype(np.int64)
+    vs = ts.astype(np.float64)
+    # three chunks to exercise the chunk-list rule
+    n = len(ts)
+    chunks = [[(int(t), float(v)) for t, v in zip(ts[i:i + 70], vs[i:i + 70])]
+              for i in range(0, n, 70)]
+    st = build_store(fdb, [chunks])
+    W = 300_000
+    for diff in range(-40_000, 40_001, 12_500):
+        for step in (2_000, 7_000, 33_000):
+            start = int(ts[60]) + diff
+            end = start + 100_000
+            q = fdb.make_query(start, step, end, W, fdb.FN_LAST)
+            out = oracle.eval_series(st.view(), 0, q, q.num_windows)
+            for w in range(q.num_windows):
+                cur = start + w * step
+                idx = np.searchsorted(ts, cur, side="right") - 1
+                if idx < 0 or cur - int(ts[idx]) > W:
+                    assert np.isnan(out[w]), (diff, step, w)
+                else:
+                    assert out[w] == vs[idx], (diff, step, w)
