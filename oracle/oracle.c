@@ -336,6 +336,25 @@ static void cread_init(const vec_t* v, double* scratch, cread_t* r) {
   r->chunk_correction = corr;
 }
 
+/* test helper: the corrected series of one counter chunk — the values
+ * BufferableCounterCorrectionIteratorSpec pins ([3,5,7,13,2,34] ->
+ * [3,5,7,13,15,47]) and CorrectingDoubleVectorReader materializes. */
+EXPORT int32_t oracle_corrected_doubles(const uint8_t* vec, double* out,
+                                        int32_t cap) {
+  vec_t v;
+  vec_open(vec, &v);
+  if (v.n > cap) return -1;
+  double corr = 0, last = -1.7976931348623157e308;
+  for (int i = 0; i < v.n; i++) {
+    double x = dv_at(&v, i);
+    if (isnan(x)) x = 0;
+    if (x < last) corr += last;
+    out[i] = x + corr;
+    last = x;
+  }
+  return v.n;
+}
+
 static inline double corrected_value(const cread_t* r, int n, const corr_meta_t* m) {
   double corr = m->has ? m->correction : 0;
   if (r->corrected) return r->corrected[n] + corr;

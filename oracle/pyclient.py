@@ -184,3 +184,17 @@ def nibblepack_unpack8(data):
 def extrapolated_rate(ws, we, n, t1, v1, t2, v2, is_counter, is_rate):
     return lib().oracle_extrapolated_rate(ws, we, n, t1, v1, t2, v2,
                                           1 if is_counter else 0, 1 if is_rate else 0)
+
+
+def corrected_doubles(vec_bytes, cap=100000):
+    """Corrected counter series of one chunk (CorrectingDoubleVectorReader)."""
+    import ctypes as ct
+    out = np.empty(cap, dtype=np.float64)
+    L = lib()
+    L.oracle_corrected_doubles.argtypes = [ct.POINTER(ct.c_uint8), _c_double_p,
+                                           ct.c_int32]
+    n = L.oracle_corrected_doubles(_u8(vec_bytes),
+                                   out.ctypes.data_as(_c_double_p), cap)
+    if n < 0:
+        raise RuntimeError("corrected_doubles failed")
+    return out[:n].copy()

@@ -144,3 +144,20 @@ def test_rate_many_windows_vs_naive(fdb, oracle):
                                             int(t1), float(vs[m][0]),
                                             int(t2), float(vs[m][-1]), True, True)
         assert out[w] == pytest.approx(expected, abs=ERR), w
+
+
+def test_counter_correction_goldens(fdb, oracle):
+    """BufferableCounterCorrectionIteratorSpec.scala:9-26 literal fixtures:
+    the corrected counter series for dips / multiple dips / no dips."""
+    cases = [
+        ([3, 5, 7, 13, 2, 34], [3, 5, 7, 13, 15, 47]),
+        ([3, 5, 7, 13, 2, 34, 4, 6], [3, 5, 7, 13, 15, 47, 51, 53]),
+        ([3, 5, 7, 13, 22, 34], [3, 5, 7, 13, 22, 34]),
+    ]
+    for raw, expected in cases:
+        ts = (100000 + np.arange(len(raw)) * 10000).astype(np.int64)
+        st = build_store(fdb, [[[(int(t), float(v)) for t, v in zip(ts, raw)]]],
+                         kind=fdb.COL_COUNTER)
+        _, vab, _, _, _ = st.chunk(0, 0)
+        got = oracle.corrected_doubles(vab)
+        np.testing.assert_array_equal(got, np.array(expected, dtype=np.float64))
