@@ -52,7 +52,24 @@ typedef struct {
 
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12,
-       FN_PRESENT=13, FN_TIMESTAMP=14, FN_ZSCORE=15 };
+       FN_PRESENT=13, FN_TIMESTAMP=14, FN_ZSCORE=15,
+       FN_QUANTILE=16, FN_MAD=17 };
+
+/* QuantileOverTimeFunction.calculateRank (AggrOverTimeFunctions.scala:400-406)
+ * + the sorted linear interpolation both quantile_over_time and
+ * median_absolute_deviation_over_time share (:1227-1267). vals is sorted. */
+static int cmp_dbl(const void* a, const void* b) {
+  double x = *(const double*)a, y = *(const double*)b;
+  return x < y ? -1 : x > y ? 1 : 0;
+}
+static double interp_quantile(double q, const double* vals, int n) {
+  double rank = q * (n - 1);
+  int lower = (int)floor(rank);
+  if (lower < 0) lower = 0;
+  int upper = lower + 1 < n - 1 ? lower + 1 : n - 1;
+  double weight = rank - floor(rank);
+  return vals[lower] * (1 - weight) + vals[upper] * weight;
+}
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
        AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9, AGG_GROUP=10 };
 
@@ -429,13 +446,20 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
   cread_t cr[64];
   double* scratch = ctx->scratch;
   if (nchunks > 64) nchunks = 64;  /* oracle cap; builder max_rows keeps chunks few */
+  int total_rows = 0;
   for (int c = 0; c < nchunks; c++) {
     vec_open(view->blob + dir[c].ts_off, &tsv[c]);
     vec_open(view->blob + dir[c].val_off, &vav[c]);
+    total_rows += dir[c].num_rows;
     if (is_rate_family) {
       cread_init(&vav[c], scratch + (size_t)c * 512, &cr[c]);
     }
   }
+  /* quantile/MAD need the window's raw samples materialized */
+  double* qbuf = (q->func_id == FN_QUANTILE || q->func_id == FN_MAD)
+                     ? (double*)malloc((size_t)(total_rows > 0 ? total_rows : 1)
+                                       * sizeof(double))
+                     : NULL;
 
   for (int w = 0; w < nw; w++) {
     int64_t wEnd = q->start + (int64_t)w * q->step;
@@ -494,6 +518,7 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
       double last_sample = NAN;    /* zscore: endRow value when non-NaN */
       int64_t last_ts = -1;
       int icount = 0;
+      int qn = 0, touched = 0;     /* quantile/MAD window sample buffer */
       for (int c = 0; c < nchunks; c++) {
         if (dir[c].end_time < wStart) continue;
         const vec_t* tv = &tsv[c];
@@ -581,6 +606,17 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
               dv_changes(vv, startRow, endRow, prev, &ch, &pv);
               changes += ch; prev = pv;
             } break;
+            case FN_QUANTILE: case FN_MAD: {
+              /* QuantileOverTimeChunkedFunctionD (:1272-1299) /
+               * MedianAbsoluteDeviationOverTimeChunkedFunctionD (:1302-1330):
+               * collect the window's non-NaN samples */
+              touched = 1;
+              if (!(q->func_id == FN_QUANTILE && (q->param < 0 || q->param > 1)))
+                for (int i = startRow; i <= endRow; i++) {
+                  double x = dv_at(vv, i);
+                  if (!isnan(x)) qbuf[qn++] = x;
+                }
+            } break;
           }
         }
         if (dir[c].end_time >= wEnd) break;
@@ -613,6 +649,23 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
           } else if (isnan(sum)) result = sum;
           else result = 0;
         } break;
+        case FN_QUANTILE: {
+          if (touched && q->param < 0) result = -INFINITY;
+          else if (touched && q->param > 1) result = INFINITY;
+          else if (qn > 0) {
+            qsort(qbuf, (size_t)qn, sizeof(double), cmp_dbl);
+            result = interp_quantile(q->param, qbuf, qn);
+          }
+        } break;
+        case FN_MAD: {
+          if (qn > 0) {
+            qsort(qbuf, (size_t)qn, sizeof(double), cmp_dbl);
+            double median = interp_quantile(0.5, qbuf, qn);
+            for (int i = 0; i < qn; i++) qbuf[i] = fabs(median - qbuf[i]);
+            qsort(qbuf, (size_t)qn, sizeof(double), cmp_dbl);
+            result = interp_quantile(0.5, qbuf, qn);
+          }
+        } break;
       }
     }
     out[w] = result;
@@ -620,6 +673,7 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
   if (is_rate_family)
     for (int c = 0; c < nchunks; c++)
       if (cr[c].owned) free(cr[c].corrected);
+  free(qbuf);
 }
 
 /* =========================================================================

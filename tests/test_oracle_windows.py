@@ -196,3 +196,84 @@ def test_empty_window_emits_nan(fdb, oracle):
     q = fdb.make_query(200000, 1000, 205000, 500, 3)  # windows beyond the data
     out = oracle.eval_series(st.view(), 0, q, q.num_windows)
     assert np.isnan(out).all()
+
+
+def _np_rank_interp(q, vals):
+    """calculateRank + sorted interpolation (AggrOverTimeFunctions.scala:400-406)."""
+    v = np.sort(vals)
+    n = len(v)
+    rank = q * (n - 1)
+    lower = max(0, int(np.floor(rank)))
+    upper = min(n - 1, lower + 1)
+    weight = rank - np.floor(rank)
+    return v[lower] * (1 - weight) + v[upper] * weight
+
+
+@pytest.mark.parametrize("nchunks", [1, 3])
+def test_quantile_over_time_vs_numpy(fdb, oracle, nchunks):
+    """quantile_over_time (QuantileOverTimeChunkedFunctionD,
+    AggrOverTimeFunctions.scala:1227-1299): sorted interpolation over the
+    window's non-NaN samples; q<0 -> -Inf, q>1 -> +Inf, empty -> NaN."""
+    rng = np.random.default_rng(61)
+    n = 120
+    ts, vs = synth_gauge_series(rng, n, step=10000, jitter=300, nan_p=0.12)
+    per = n // nchunks
+    chunks, bounds = [], []
+    for c in range(nchunks):
+        lo, hi = c * per, (n if c == nchunks - 1 else (c + 1) * per)
+        chunks.append([(int(ts[i]), float(vs[i])) for i in range(lo, hi)])
+    st = build_store(fdb, [chunks])
+    # use STORED timestamps (approx-const encoding may shift within ±250)
+    tsd = np.concatenate([oracle.decode_longs(st.chunk(0, c)[0])
+                          for c in range(nchunks)])
+    start, step, window = int(tsd[20]), 30000, 200000
+    end = int(tsd[-1]) + step
+    for q_param in (-0.5, 0.0, 0.25, 0.5, 0.9, 1.0, 1.5):
+        q = fdb.make_query(start, step, end, window, fdb.FN_QUANTILE_OVER_TIME,
+                           param=q_param)
+        out = oracle.eval_series(st.view(), 0, q, q.num_windows)
+        for w in range(q.num_windows):
+            w_end = start + w * step
+            m = (tsd >= w_end - window) & (tsd <= w_end)
+            nn = vs[m][~np.isnan(vs[m])]
+            if not m.any():
+                assert np.isnan(out[w]), (q_param, w)
+            elif q_param < 0:
+                assert out[w] == -np.inf, (q_param, w)
+            elif q_param > 1:
+                assert out[w] == np.inf, (q_param, w)
+            elif len(nn) == 0:
+                assert np.isnan(out[w]), (q_param, w)
+            else:
+                assert out[w] == pytest.approx(_np_rank_interp(q_param, nn),
+                                               rel=1e-12), (q_param, w)
+
+
+@pytest.mark.parametrize("nchunks", [1, 3])
+def test_mad_over_time_vs_numpy(fdb, oracle, nchunks):
+    """median_absolute_deviation_over_time
+    (AggrOverTimeFunctions.scala:1248-1330): median of |median - v|."""
+    rng = np.random.default_rng(67)
+    n = 100
+    ts, vs = synth_gauge_series(rng, n, step=10000, jitter=300, nan_p=0.1)
+    per = n // nchunks
+    chunks = [[(int(ts[i]), float(vs[i]))
+               for i in range(c * per, n if c == nchunks - 1 else (c + 1) * per)]
+              for c in range(nchunks)]
+    st = build_store(fdb, [chunks])
+    tsd = np.concatenate([oracle.decode_longs(st.chunk(0, c)[0])
+                          for c in range(nchunks)])
+    start, step, window = int(tsd[15]), 30000, 250000
+    end = int(tsd[-1]) + step
+    q = fdb.make_query(start, step, end, window, fdb.FN_MAD_OVER_TIME)
+    out = oracle.eval_series(st.view(), 0, q, q.num_windows)
+    for w in range(q.num_windows):
+        w_end = start + w * step
+        m = (tsd >= w_end - window) & (tsd <= w_end)
+        nn = vs[m][~np.isnan(vs[m])]
+        if len(nn) == 0:
+            assert np.isnan(out[w]), w
+        else:
+            med = _np_rank_interp(0.5, nn)
+            expect = _np_rank_interp(0.5, np.abs(med - nn))
+            assert out[w] == pytest.approx(expect, rel=1e-12), w
