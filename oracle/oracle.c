@@ -53,7 +53,7 @@ typedef struct {
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12,
        FN_PRESENT=13, FN_TIMESTAMP=14, FN_ZSCORE=15,
-       FN_QUANTILE=16, FN_MAD=17 };
+       FN_QUANTILE=16, FN_MAD=17, FN_PREDICT_LINEAR=18 };
 
 /* QuantileOverTimeFunction.calculateRank (AggrOverTimeFunctions.scala:400-406)
  * + the sorted linear interpolation both quantile_over_time and
@@ -519,6 +519,8 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
       int64_t last_ts = -1;
       int icount = 0;
       int qn = 0, touched = 0;     /* quantile/MAD window sample buffer */
+      double plX = NAN, plY = NAN, plXY = NAN, plX2 = NAN;  /* predict_linear */
+      int plN = 0;
       for (int c = 0; c < nchunks; c++) {
         if (dir[c].end_time < wStart) continue;
         const vec_t* tv = &tsv[c];
@@ -606,6 +608,18 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
               dv_changes(vv, startRow, endRow, prev, &ch, &pv);
               changes += ch; prev = pv;
             } break;
+            case FN_PREDICT_LINEAR: {
+              /* PredictLinearChunkedFunctionD (AggrOverTimeFunctions.scala:
+               * 1520-1554): regression sums over x=(ts-wEnd)/1000, y=value */
+              for (int i = startRow; i <= endRow; i++) {
+                double y = dv_at(vv, i);
+                if (isnan(y)) continue;
+                double x = (double)(lv_at(tv, i) - wEnd) / 1000.0;
+                if (isnan(plY)) { plY = y; plX = x; plXY = x * y; plX2 = x * x; }
+                else { plY += y; plX += x; plXY += x * y; plX2 += x * x; }
+                plN++;
+              }
+            } break;
             case FN_QUANTILE: case FN_MAD: {
               /* QuantileOverTimeChunkedFunctionD (:1272-1299) /
                * MedianAbsoluteDeviationOverTimeChunkedFunctionD (:1302-1330):
@@ -655,6 +669,16 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
           else if (qn > 0) {
             qsort(qbuf, (size_t)qn, sizeof(double), cmp_dbl);
             result = interp_quantile(q->param, qbuf, qn);
+          }
+        } break;
+        case FN_PREDICT_LINEAR: {
+          /* emit (AggrOverTimeFunctions.scala:1507-1517): counter >= 2 */
+          if (plN >= 2) {
+            double covXY = plXY - plX * plY / plN;
+            double varX = plX2 - plX * plX / plN;
+            double slope = covXY / varX;
+            double intercept = plY / plN - slope * plX / plN;
+            result = slope * q->param + intercept;
           }
         } break;
         case FN_MAD: {

@@ -277,3 +277,42 @@ def test_mad_over_time_vs_numpy(fdb, oracle, nchunks):
             med = _np_rank_interp(0.5, nn)
             expect = _np_rank_interp(0.5, np.abs(med - nn))
             assert out[w] == pytest.approx(expect, rel=1e-12), w
+
+
+@pytest.mark.parametrize("nchunks", [1, 3])
+def test_predict_linear_vs_numpy(fdb, oracle, nchunks):
+    """predict_linear (PredictLinearChunkedFunctionD,
+    AggrOverTimeFunctions.scala:1496-1554): least-squares extrapolation over
+    x=(ts-wEnd)/1000; NaN below 2 samples."""
+    rng = np.random.default_rng(71)
+    n = 90
+    ts, vs = synth_gauge_series(rng, n, step=10000, jitter=300, nan_p=0.1)
+    per = n // nchunks
+    chunks = [[(int(ts[i]), float(vs[i]))
+               for i in range(c * per, n if c == nchunks - 1 else (c + 1) * per)]
+              for c in range(nchunks)]
+    st = build_store(fdb, [chunks])
+    tsd = np.concatenate([oracle.decode_longs(st.chunk(0, c)[0])
+                          for c in range(nchunks)])
+    start, step, window = int(tsd[12]), 30000, 200000
+    end = int(tsd[-1]) + step
+    for duration in (60.0, 600.0):
+        q = fdb.make_query(start, step, end, window, fdb.FN_PREDICT_LINEAR,
+                           param=duration)
+        out = oracle.eval_series(st.view(), 0, q, q.num_windows)
+        for w in range(q.num_windows):
+            w_end = start + w * step
+            m = (tsd >= w_end - window) & (tsd <= w_end)
+            sel = ~np.isnan(vs[m])
+            x = (tsd[m][sel] - w_end) / 1000.0
+            y = vs[m][sel]
+            if len(y) < 2:
+                assert np.isnan(out[w]), (duration, w)
+                continue
+            cn = len(y)
+            cov = (x * y).sum() - x.sum() * y.sum() / cn
+            var = (x * x).sum() - x.sum() ** 2 / cn
+            slope = cov / var
+            intercept = y.sum() / cn - slope * x.sum() / cn
+            assert out[w] == pytest.approx(slope * duration + intercept,
+                                           rel=1e-9), (duration, w)
