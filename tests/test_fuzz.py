@@ -11,6 +11,9 @@ from conftest import build_store
 
 import filodb_amd as fdb_mod
 
+# FDB_FUZZ_SCALE multiplies example counts for one-off deep runs (default 1)
+_SCALE = int(__import__("os").environ.get("FDB_FUZZ_SCALE", "1"))
+
 
 def _oracle():
     import sys
@@ -30,7 +33,7 @@ val_strategy = st.lists(
     min_size=3, max_size=120)
 
 
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=60 * _SCALE, deadline=None)
 @given(deltas=ts_strategy, vals=val_strategy)
 def test_encode_decode_roundtrip(deltas, vals):
     """Arbitrary nondecreasing timestamps + arbitrary doubles (NaNs included)
@@ -64,7 +67,7 @@ def test_encode_decode_roundtrip(deltas, vals):
         dec_vs.view(np.uint64), vs.view(np.uint64))  # bit-exact incl. NaNs
 
 
-@settings(max_examples=50, deadline=None)
+@settings(max_examples=50 * _SCALE, deadline=None)
 @given(deltas=ts_strategy, vals=val_strategy,
        nchunks=st.integers(min_value=1, max_value=4),
        window=st.integers(min_value=1, max_value=20_000),
@@ -112,7 +115,7 @@ def test_gauge_window_parity_fuzz(deltas, vals, nchunks, window, step, func):
                 (func, w, got, expected)
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40 * _SCALE, deadline=None)
 @given(deltas=ts_strategy,
        incs=st.lists(st.integers(min_value=0, max_value=1000), min_size=3,
                      max_size=120),
@@ -169,7 +172,7 @@ def test_rate_window_parity_fuzz(deltas, incs, resets, window, step):
         assert out[w] == expect or abs(out[w] - expect) <= 1e-9 * abs(expect), w
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40 * _SCALE, deadline=None)
 @given(rows=st.lists(
            st.lists(st.integers(min_value=0, max_value=5000), min_size=4,
                     max_size=4),
