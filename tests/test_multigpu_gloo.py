@@ -87,3 +87,72 @@ def test_two_shard_avg_merge_equals_global(fdb, oracle):
                        fdb.FN_AVG_OVER_TIME, fdb.AGG_AVG, n_groups)
     expected = oracle.query_exec(st.view(), q, st.num_series, q.num_windows)
     np.testing.assert_allclose(merged, expected, rtol=1e-9, equal_nan=True)
+
+
+def _worker_stddev(rank, result_queue):
+    import sys
+    sys.path.insert(0, REPO)
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import filodb_amd as fdb
+    import pyclient as oracle
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29519"
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+
+    n_groups = 3
+    st, _ = _make_shard(fdb, rank, n_groups)
+    start = 100000 + 20 * 10000
+    q = fdb.make_query(start, 30000, start + 200000, 100000,
+                       fdb.FN_SUM_OVER_TIME, fdb.AGG_STDDEV, n_groups)
+    nw = q.num_windows
+    cells = n_groups * nw
+    # stacked (raw sums, raw sumsq) partials + counts — the header contract
+    stacked, counts = oracle.query_exec(st.view(), q, st.num_series, nw,
+                                        out_counts=True)
+    assert stacked.shape == (2 * cells,)
+    t_sc = torch.from_numpy(stacked)
+    t_cnt = torch.from_numpy(counts)
+    dist.all_reduce(t_sc)          # ONE collective merges sums and sumsq
+    dist.all_reduce(t_cnt)
+    s, sq, c = t_sc[:cells], t_sc[cells:], t_cnt
+    mean = s / c
+    var = sq / c - mean * mean
+    merged = torch.where(c > 0, torch.sqrt(var),
+                         torch.full_like(s, float("nan")))
+
+    if rank == 0:
+        result_queue.put(merged.numpy())
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_two_shard_stddev_merge_equals_global(fdb, oracle):
+    """Cross-shard stddev via the stacked (sums, sumsq) partial contract
+    (StddevRowAggregator.scala:36-52 algebra over one all-reduce)."""
+    ctx = mp.get_context("spawn")
+    queue = ctx.Queue()
+    procs = [ctx.Process(target=_worker_stddev, args=(r, queue))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    merged = queue.get(timeout=120)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+
+    n_groups = 3
+    series, groups = [], []
+    for rank in range(WORLD):
+        rng = np.random.default_rng(1000 + rank)
+        for s in range(20):
+            ts, vs = synth_gauge_series(rng, 50, step=10000, jitter=400, nan_p=0.1)
+            series.append([[(int(t), float(v)) for t, v in zip(ts, vs)]])
+            groups.append((rank * 20 + s) % n_groups)
+    st = build_store(fdb, series, groups=groups)
+    start = 100000 + 20 * 10000
+    q = fdb.make_query(start, 30000, start + 200000, 100000,
+                       fdb.FN_SUM_OVER_TIME, fdb.AGG_STDDEV, n_groups)
+    expected = oracle.query_exec(st.view(), q, st.num_series, q.num_windows)
+    np.testing.assert_allclose(merged, expected, rtol=1e-9, atol=1e-12,
+                               equal_nan=True)
