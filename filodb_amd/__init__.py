@@ -23,6 +23,7 @@ FN_LAST = 12
 FN_PRESENT, FN_TIMESTAMP, FN_ZSCORE = 13, 14, 15
 FN_QUANTILE_OVER_TIME, FN_MAD_OVER_TIME = 16, 17   # oracle-only (kernel: round 2)
 FN_PREDICT_LINEAR = 18                             # oracle-only (kernel: round 2)
+FN_RATE_OVER_DELTA = 19                            # oracle-only (kernel: round 2)
 # aggregation ids (RowAggregator implementations)
 AGG_NONE, AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX, AGG_AVG = 0, 1, 2, 3, 4, 5
 AGG_TOPK, AGG_BOTTOMK = 6, 7

@@ -53,7 +53,8 @@ typedef struct {
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12,
        FN_PRESENT=13, FN_TIMESTAMP=14, FN_ZSCORE=15,
-       FN_QUANTILE=16, FN_MAD=17, FN_PREDICT_LINEAR=18 };
+       FN_QUANTILE=16, FN_MAD=17, FN_PREDICT_LINEAR=18,
+       FN_RATE_OVER_DELTA=19 };
 
 /* QuantileOverTimeFunction.calculateRank (AggrOverTimeFunctions.scala:400-406)
  * + the sorted linear interpolation both quantile_over_time and
@@ -560,7 +561,7 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
         }
         if (startRow <= endRow) {
           switch (q->func_id) {
-            case FN_SUM: {
+            case FN_SUM: case FN_RATE_OVER_DELTA: {
               double cs = dv_sum(vv, startRow, endRow);
               if (!isnan(cs) && isnan(sum)) sum = 0;
               sum += cs;                       /* :560-572 incl. NaN-poison quirk */
@@ -637,6 +638,11 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
       }
       switch (q->func_id) {
         case FN_SUM:   result = sum; break;
+        case FN_RATE_OVER_DELTA:
+          /* delta-temporality rate: RateOverDeltaChunkedFunctionD
+           * (RateFunctions.scala:424-445) = sum_over_time / window seconds */
+          result = sum / (double)(wEnd - wStart) * 1000;
+          break;
         case FN_COUNT: result = count; break;
         case FN_AVG:   result = icount > 0 ? sum / icount : (isnan(sum) ? sum : 0); break;
         case FN_MIN:   result = mn; break;

@@ -316,3 +316,28 @@ def test_predict_linear_vs_numpy(fdb, oracle, nchunks):
             intercept = y.sum() / cn - slope * x.sum() / cn
             assert out[w] == pytest.approx(slope * duration + intercept,
                                            rel=1e-9), (duration, w)
+
+
+def test_rate_over_delta_vs_numpy(fdb, oracle):
+    """Delta-temporality rate (RateOverDeltaChunkedFunctionD,
+    RateFunctions.scala:424-445): sum of per-sample deltas / window seconds.
+    The increase-over-delta arm is plain sum_over_time."""
+    rng = np.random.default_rng(83)
+    n = 100
+    ts, vs = synth_gauge_series(rng, n, step=10000, jitter=300, nan_p=0.1)
+    vs = np.abs(vs)                # delta counters carry per-interval counts
+    st = build_store(fdb, [[[(int(t), float(v)) for t, v in zip(ts, vs)]]])
+    tsd = oracle.decode_longs(st.chunk(0, 0)[0])
+    start, step, window = int(tsd[15]), 30000, 200000
+    end = int(tsd[-1]) + step
+    q = fdb.make_query(start, step, end, window, fdb.FN_RATE_OVER_DELTA)
+    out = oracle.eval_series(st.view(), 0, q, q.num_windows)
+    for w in range(q.num_windows):
+        w_end = start + w * step
+        m = (tsd >= w_end - window) & (tsd <= w_end)
+        nn = vs[m][~np.isnan(vs[m])]
+        if len(nn) == 0:
+            assert np.isnan(out[w]), w
+        else:
+            expect = nn.sum() / window * 1000
+            assert out[w] == pytest.approx(expect, rel=1e-12), w
