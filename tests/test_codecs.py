@@ -156,3 +156,28 @@ def test_max_rows_auto_cut(fdb):
     st.append(sid, ts, np.arange(250, dtype=np.float64) + 0.5)
     st.seal()
     assert st.num_chunks(sid) == 3  # 100+100+50 (maxChunksSize cut)
+
+
+def test_huge_timestamp_jump_falls_back_to_raw(fdb, oracle):
+    """Timestamp deltas beyond the 32-bit inner-vector range make
+    DeltaDeltaVector.fromLongVector bail (DeltaDeltaVector.scala:63-85) —
+    the encoder must fall back to raw i64 and stay bit-exact, and the window
+    engine must still evaluate across the jump."""
+    ts = np.array([100000, 115000, 130000,
+                   100000 + (1 << 33), 100000 + (1 << 33) + 15000],
+                  dtype=np.int64)
+    vs = np.array([1.5, 2.5, 3.5, 4.5, 5.5])
+    st = build_store(fdb, [[[(int(t), float(v)) for t, v in zip(ts, vs)]]])
+    tsb, _, n, t0, t1 = st.chunk(0, 0)
+    assert (n, t0, t1) == (5, int(ts[0]), int(ts[-1]))
+    info = oracle.vec_info(tsb)
+    assert info["wf"] == WF_PRIM64          # raw i64 fallback
+    np.testing.assert_array_equal(oracle.decode_longs(tsb), ts)
+    # windows on both sides of the jump
+    q = fdb.make_query(int(ts[-1]), 15000, int(ts[-1]), 30000,
+                       fdb.FN_SUM_OVER_TIME)
+    out = oracle.eval_series(st.view(), 0, q, 1)
+    assert out[0] == 4.5 + 5.5
+    q2 = fdb.make_query(130000, 15000, 130000, 30000, fdb.FN_SUM_OVER_TIME)
+    out2 = oracle.eval_series(st.view(), 0, q2, 1)
+    assert out2[0] == 1.5 + 2.5 + 3.5
