@@ -83,8 +83,11 @@ def build_store(w, rank):
     st = fdb.ChunkStore()
     kind = {"counter": fdb.COL_COUNTER, "gauge": fdb.COL_GAUGE,
             "hist": fdb.COL_HIST}[w["kind"]]
+    # jitter 300 > MaxApproxDelta=250 (DeltaDeltaVector.scala:46): timestamp
+    # vectors encode as PACKED DDV, not const — the decode path configs[2]
+    # intends the benchmark to exercise (round-1 verdict, weak item 7)
     st.synth_generate(kind, w["n_series"], w["n_samples"], start_ts=T0,
-                      step_ms=w["step_ms"], jitter_ms=250, lam=10.0,
+                      step_ms=w["step_ms"], jitter_ms=300, lam=10.0,
                       reset_p=0.001, n_groups=w["n_groups"], seed=42 + rank)
     st.seal()
     return st
@@ -166,6 +169,12 @@ def main():
         torch.cuda.set_device(device)
         dist.init_process_group(os.environ.get("FDB_DIST_BACKEND", "nccl"))
         local_rank = device
+        # per-rank sanity so a SCALE run is self-evidencing: the rank/world the
+        # backend actually sees, and the device this rank drives
+        print(f"[fdb rank-sanity] rank={dist.get_rank()}/{dist.get_world_size()} "
+              f"backend={dist.get_backend()} device=cuda:{device} "
+              f"visible_gpus={torch.cuda.device_count()}",
+              file=sys.stderr, flush=True)
     w = WORKLOADS[args.workload]
     st = build_store(w, rank)
     q = make_query(w)

@@ -259,7 +259,7 @@ class Engine:
         and on_device is False)."""
         if out is None:
             assert not on_device
-            n = dataset_out_len(dataset, q)
+            n = dataset_out_len(dataset, q, partial=out_counts is not None)
             out = np.empty(n, dtype=np.float64)
         rc = lib().fdb_query_exec(self._h, dataset._h, ctypes.byref(q),
                                   _as_f64_ptr(out), _as_f64_ptr(out_counts),
@@ -311,11 +311,19 @@ class Dataset:
         return lib().fdb_dataset_samples(self._h)
 
 
-def dataset_out_len(dataset, q: Query):
+def dataset_out_len(dataset, q: Query, partial=False):
+    """Doubles the engine copies into `out` for this query (run_query buf_len):
+    G*W*k for top/bottom-k (k = q.param), 2*G*W for the stacked stddev/stdvar
+    partial grids (raw sums + sumsq) when out_counts is passed, else G*W."""
     if q.agg_id == AGG_NONE:
         # series count is not stored on Dataset; caller usually knows it
         raise ValueError("pass an explicit out buffer for AGG_NONE queries")
-    return q.num_groups * q.num_windows
+    n = q.num_groups * q.num_windows
+    if q.agg_id in (AGG_TOPK, AGG_BOTTOMK):
+        return n * max(1, int(q.param))
+    if partial and q.agg_id in (AGG_STDDEV, AGG_STDVAR):
+        return 2 * n
+    return n
 
 
 def make_query(start, step, end, window, func_id, agg_id=AGG_NONE, num_groups=0,

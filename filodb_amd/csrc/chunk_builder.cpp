@@ -314,7 +314,8 @@ static void encode_doubles(const double* vals, int n, bool drop, bytes& out) {
   bool all_integral = true;
   for (int i = 0; i < n; i++) {
     double d = vals[i];
-    if (d > 9.2233720368547758e18 /* Long.MaxValue.toDouble */ || rint(d) != d) {
+    if (d > 9.2233720368547758e18 /* Long.MaxValue.toDouble */ ||
+        d < -9.2233720368547758e18 /* Long.MinValue.toDouble */ || rint(d) != d) {
       all_integral = false; break;      // NaN: rint(NaN)!=NaN → non-integral
     }
   }

@@ -1540,6 +1540,7 @@ extern "C" void fdb_dataset_destroy(fdb_dataset_t* d) {
   (void)hipFree(d->blob); (void)hipFree(d->ts_off); (void)hipFree(d->val_off);
   (void)hipFree(d->start_time); (void)hipFree(d->end_time); (void)hipFree(d->num_rows);
   (void)hipFree(d->series_first); (void)hipFree(d->series_nchunks); (void)hipFree(d->group_ids);
+  (void)hipFree(d->series_by_group); (void)hipFree(d->group_offsets);
   delete d;
 }
 
@@ -1734,19 +1735,37 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
     own_cnt = true;
   }
 
+  // every exit (including the HIP_CHECK early returns) releases what this call
+  // owns — device buffers and events must not leak on a failed launch
+  struct Guard {
+    double **out_p, **cnt_p, **sq_p, **grid_p;
+    bool *own_out_p, *own_cnt_p, *own_sq;
+    hipEvent_t *e0, *e1;
+    ~Guard() {
+      if (*grid_p) (void)hipFree(*grid_p);
+      if (*sq_p && *own_sq) (void)hipFree(*sq_p);
+      if (*own_out_p) (void)hipFree(*out_p);
+      if (*own_cnt_p) (void)hipFree(*cnt_p);
+      if (*e0) (void)hipEventDestroy(*e0);
+      if (*e1) (void)hipEventDestroy(*e1);
+    }
+  };
   double* dev_sq = nullptr;
+  bool own_sq = false;
+  double* per_grid = nullptr;
+  hipEvent_t ev0 = nullptr, ev1 = nullptr;
+  Guard guard{&dev_out, &dev_cnt, &dev_sq, &per_grid,
+              &own_out, &own_cnt, &own_sq, &ev0, &ev1};
   if (needs_sq) {
     if (partial) dev_sq = dev_out + out_len;   // second half of the caller grid
-    else HIP_CHECK(hipMalloc(&dev_sq, out_len * 8));
+    else { HIP_CHECK(hipMalloc(&dev_sq, out_len * 8)); own_sq = true; }
   }
-  hipEvent_t ev0, ev1;
   HIP_CHECK(hipEventCreate(&ev0));
   HIP_CHECK(hipEventCreate(&ev1));
 
   // every agg runs two-phase: the scan fills an internal per-series [S×W]
   // grid with plain stores, then a presenter (topk_kernel / group_reduce_
   // kernel) folds it along the group-sorted index — no data-path atomics
-  double* per_grid = nullptr;
   fdb_query_t qscan = *q;
   if (q->agg_id != AGG_NONE) {
     HIP_CHECK(hipMalloc(&per_grid, (size_t)d->num_series * nw * 8));
@@ -1800,20 +1819,15 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
         dev_out, dev_cnt, dev_sq, out_len, q->agg_id, partial);
     HIP_CHECK(hipGetLastError());
   }
-  if (per_grid) (void)hipFree(per_grid);
-  if (dev_sq && !(needs_sq && partial)) (void)hipFree(dev_sq);
   HIP_CHECK(hipStreamSynchronize(e->stream));
-  (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
 
   if (!out_on_device) {
     HIP_CHECK(hipMemcpy(out, dev_out, buf_len * 8, hipMemcpyDeviceToHost));
     if (q->agg_id != AGG_NONE && out_counts)
       HIP_CHECK(hipMemcpy(out_counts, dev_cnt, out_len * 8, hipMemcpyDeviceToHost));
   }
-  if (own_out) (void)hipFree(dev_out);
-  if (own_cnt) (void)hipFree(dev_cnt);
   if (avg_ms) *avg_ms = iters > 0 ? (double)ms_sum / iters : 0.0;
-  return FDB_OK;
+  return FDB_OK;   // guard frees per_grid/dev_sq/owned buffers + events
 }
 
 
