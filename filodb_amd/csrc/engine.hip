@@ -26,9 +26,21 @@
 #include <cstdlib>
 
 #include "chunk_format.h"
+#include "scan_common.h"
 #include "../../include/filodb_amd.h"
 
 void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
+
+// scan_fast.hip: single-chunk-per-series fast path
+int32_t fdb_launch_fast_scan(hipStream_t stream, const uint8_t* blob, DirSoA dir,
+                             const int32_t* series_first, const int32_t* series_nchunks,
+                             const int32_t* group_ids, const int32_t* series_by_group,
+                             int num_series,
+                             int64_t qstart, int64_t qstep, int64_t qwindow,
+                             int num_windows, int func_id, int agg_id, int emit_group,
+                             double* out, double* out_cnt, double* out_sq,
+                             int phase_mask);
+bool fdb_fast_scan_supported(int func_id);
 
 #define HIP_CHECK(expr) do { hipError_t _e = (expr); if (_e != hipSuccess) { \
   fdb_set_error("%s failed: %s", #expr, hipGetErrorString(_e)); return FDB_ERR; } } while (0)
@@ -43,94 +55,8 @@ void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
 #define WAVES_PER_BLOCK 4
 #define BLOCK_THREADS (WAVES_PER_BLOCK * 64)
 
-enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
-       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12,
-       FN_PRESENT=13, FN_TIMESTAMP=14, FN_ZSCORE=15 };
-enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
-       AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9, AGG_GROUP=10 };
-
-// ---------------------------------------------------------------------------
-// device-side vector readers (same layouts as oracle)
-// ---------------------------------------------------------------------------
-struct DirSoA {
-  const uint64_t* ts_off;
-  const uint64_t* val_off;
-  const int64_t*  start_time;
-  const int64_t*  end_time;
-  const int32_t*  num_rows;
-};
-
-__device__ __forceinline__ uint16_t d_u16(const uint8_t* p) { uint16_t v; memcpy(&v, p, 2); return v; }
-__device__ __forceinline__ uint32_t d_u32(const uint8_t* p) { uint32_t v; memcpy(&v, p, 4); return v; }
-__device__ __forceinline__ int32_t  d_i32(const uint8_t* p) { int32_t v; memcpy(&v, p, 4); return v; }
-__device__ __forceinline__ int64_t  d_i64(const uint8_t* p) { int64_t v; memcpy(&v, p, 8); return v; }
-__device__ __forceinline__ double   d_f64(const uint8_t* p) { double v; memcpy(&v, p, 8); return v; }
-
-struct DVec {           // opened vector header
-  const uint8_t* idata;
-  int64_t init;
-  int32_t slope;
-  int n;
-  uint16_t wf;
-  uint8_t nbits, sign, dropped;
-};
-
-__device__ void d_vec_open(const uint8_t* p, DVec* v) {
-  v->wf = d_u16(p + 4);
-  v->dropped = (d_u16(p + 6) & FDB_DROP_MASK) != 0;
-  if (v->wf == FDB_WF_DDV) {
-    v->init = d_i64(p + FDB_DDV_OFF_INIT);
-    v->slope = d_i32(p + FDB_DDV_OFF_SLOPE);
-    const uint8_t* inner = p + FDB_DDV_OFF_INNER;
-    v->nbits = inner[6] & FDB_NBITS_MASK;
-    v->sign = (inner[6] & FDB_SIGN_MASK) != 0;
-    v->idata = inner + FDB_PRIM_OFF_DATA;
-    int numBytes = (int)d_u32(inner);
-    int bitShift = inner[7] & 0x3f;
-    v->n = ((numBytes - 4) * 8 + (bitShift != 0 ? bitShift - 8 : 0)) / v->nbits;
-  } else if (v->wf == FDB_WF_DDV_CONST) {
-    v->n = d_i32(p + FDB_DDVC_OFF_NELEM);
-    v->init = d_i64(p + FDB_DDVC_OFF_INIT);
-    v->slope = d_i32(p + FDB_DDVC_OFF_SLOPE);
-    v->idata = nullptr; v->nbits = 0; v->sign = 0;
-  } else {
-    v->n = ((int)d_u32(p) - 4) / 8;
-    v->idata = p + FDB_PRIM_OFF_DATA;
-    v->init = 0; v->slope = 0; v->nbits = 64; v->sign = 1;
-  }
-}
-
-__device__ __forceinline__ int64_t d_inner_at(const DVec* v, int i) {
-  switch (v->nbits) {
-    case 32: return d_i32(v->idata + 4 * (size_t)i);
-    case 16: { int32_t x = (int16_t)d_u16(v->idata + 2 * (size_t)i);
-               return v->sign ? x : (x & 0xffff); }
-    case 8:  { int32_t x = (int8_t)v->idata[i];
-               return v->sign ? x : (x & 0xff); }
-    case 4:  return (v->idata[i >> 1] >> ((i & 1) * 4)) & 0x0f;
-    case 2:  return (v->idata[i >> 2] >> ((i & 3) * 2)) & 0x03;
-  }
-  return 0;
-}
-
-// wave-local LDS ordering only: s_waitcnt lgkmcnt(0) — unlike s_waitcnt(0)
-// this does NOT drain vmcnt, so outstanding global loads/stores/atomics keep
-// flying across the fence (a full flush on the per-element hot path costs a
-// full memory round trip per element)
-__device__ __forceinline__ void d_wait_lds() {
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-}
-
-__device__ __forceinline__ int64_t d_lv_at(const DVec* v, int i) {
-  if (v->wf == FDB_WF_DDV) return v->init + (int64_t)v->slope * i + d_inner_at(v, i);
-  if (v->wf == FDB_WF_DDV_CONST) return v->init + (int64_t)v->slope * i;
-  return d_i64(v->idata + 8 * (size_t)i);
-}
-
-__device__ __forceinline__ double d_dv_at(const DVec* v, int i) {
-  if (v->wf == FDB_WF_PRIM64) return d_f64(v->idata + 8 * (size_t)i);
-  return (double)d_lv_at(v, i);
-}
+// FN_*/AGG_* ids, DirSoA, DVec readers, decode, wave scans, extrapolatedRate:
+// scan_common.h (shared with scan_fast.hip)
 
 // ---------------------------------------------------------------------------
 // per-series LDS workspace
@@ -192,100 +118,6 @@ struct Ws {                         // per-wave LDS workspace
   int32_t total_rows;
 };
 
-// wave-wide inclusive prefix sum (64 lanes)
-__device__ __forceinline__ double wave_incl_scan(double x, int lane) {
-  for (int off = 1; off < 64; off <<= 1) {
-    double t = __shfl_up(x, off);
-    if (lane >= off) x += t;
-  }
-  return x;
-}
-__device__ __forceinline__ int wave_incl_scan_i(int x, int lane) {
-  for (int off = 1; off < 64; off <<= 1) {
-    int t = __shfl_up(x, off);
-    if (lane >= off) x += t;
-  }
-  return x;
-}
-
-// ---------------------------------------------------------------------------
-// vectorized decode: whole chunk into LDS, wide loads, high ILP.
-// (the frozen formats of DESIGN.md §2; element semantics identical to
-//  d_lv_at/d_dv_at, which remain the reference implementations / fallback)
-// ---------------------------------------------------------------------------
-template <bool AS_DOUBLE>
-__device__ void d_decode_chunk(const DVec& v, int n, int64_t* tout, double* dout,
-                               int lane) {
-  if (v.wf == FDB_WF_DDV_CONST) {
-    for (int i = lane; i < n; i += 64) {
-      int64_t x = v.init + (int64_t)v.slope * i;
-      if (AS_DOUBLE) dout[i] = (double)x; else tout[i] = x;
-    }
-    return;
-  }
-  if (v.wf == FDB_WF_PRIM64) {
-    // raw 64-bit payload, 8-byte aligned (+8 from a 64B-aligned base)
-    for (int i = lane; i < n; i += 64) {
-      if (AS_DOUBLE) dout[i] = d_f64(v.idata + 8 * (size_t)i);
-      else           tout[i] = d_i64(v.idata + 8 * (size_t)i);
-    }
-    return;
-  }
-  // packed DDV inner data starts at +28 (4-byte aligned only)
-  if (v.nbits == 16) {
-    for (int i0 = 4 * lane; i0 < n; i0 += 256) {
-      uint32_t lo = d_u32(v.idata + 2 * (size_t)i0);
-      uint32_t hi = d_u32(v.idata + 2 * (size_t)i0 + 4);
-      uint64_t w = ((uint64_t)hi << 32) | lo;
-      #pragma unroll
-      for (int k = 0; k < 4; k++) {
-        if (i0 + k < n) {
-          int32_t d = (int32_t)(int16_t)(uint16_t)(w >> (16 * k));
-          if (!v.sign) d &= 0xffff;
-          int64_t x = v.init + (int64_t)v.slope * (i0 + k) + d;
-          if (AS_DOUBLE) dout[i0 + k] = (double)x; else tout[i0 + k] = x;
-        }
-      }
-    }
-    return;
-  }
-  if (v.nbits == 8) {
-    for (int i0 = 8 * lane; i0 < n; i0 += 512) {
-      uint32_t lo = d_u32(v.idata + (size_t)i0);
-      uint32_t hi = d_u32(v.idata + (size_t)i0 + 4);
-      uint64_t w = ((uint64_t)hi << 32) | lo;
-      #pragma unroll
-      for (int k = 0; k < 8; k++) {
-        if (i0 + k < n) {
-          int32_t d = (int32_t)(int8_t)(uint8_t)(w >> (8 * k));
-          if (!v.sign) d &= 0xff;
-          int64_t x = v.init + (int64_t)v.slope * (i0 + k) + d;
-          if (AS_DOUBLE) dout[i0 + k] = (double)x; else tout[i0 + k] = x;
-        }
-      }
-    }
-    return;
-  }
-  if (v.nbits == 32) {
-    for (int i0 = 2 * lane; i0 < n; i0 += 128) {
-      #pragma unroll
-      for (int k = 0; k < 2; k++) {
-        if (i0 + k < n) {
-          int64_t x = v.init + (int64_t)v.slope * (i0 + k)
-                    + d_i32(v.idata + 4 * (size_t)(i0 + k));
-          if (AS_DOUBLE) dout[i0 + k] = (double)x; else tout[i0 + k] = x;
-        }
-      }
-    }
-    return;
-  }
-  // nbits 2/4 fallback (rare)
-  for (int i = lane; i < n; i += 64) {
-    int64_t x = v.init + (int64_t)v.slope * i + d_inner_at(&v, i);
-    if (AS_DOUBLE) dout[i] = (double)x; else tout[i] = x;
-  }
-}
-
 // first index in [0,n) with seg[i] >= item; n when none. Timestamps sit near a
 // slope line (the DDV premise), so an interpolation guess + short walk replaces
 // a dependent binary-search chain (usually 1-2 LDS reads) — the same idea as
@@ -305,27 +137,6 @@ __device__ __forceinline__ int lds_search_ge(const int64_t* seg, int n, int64_t 
 __device__ __forceinline__ int lds_search_le(const int64_t* seg, int n, int64_t item,
                                              int64_t ts0, float inv_slope) {
   return lds_search_ge(seg, n, item + 1, ts0, inv_slope) - 1;
-}
-
-// extrapolatedRate (RateFunctions.scala:72-111) — same arithmetic as the oracle
-__device__ double d_extrapolated_rate(int64_t windowStart, int64_t windowEnd, int numSamples,
-                                      int64_t t1, double v1, int64_t t2, double v2,
-                                      bool isCounter, bool isRate) {
-  double durationToStart = (double)(t1 - windowStart) / 1000.0;
-  double durationToEnd = (double)(windowEnd - t2) / 1000.0;
-  double sampledInterval = (double)(t2 - t1) / 1000.0;
-  double avgDur = sampledInterval / ((double)numSamples - 1);
-  double delta = v2 - v1;
-  if (isCounter && delta > 0 && v1 >= 0) {
-    double durationToZero = sampledInterval * (v1 / delta);
-    if (durationToZero < durationToStart) durationToStart = durationToZero;
-  }
-  double thresh = avgDur * 1.1;
-  double ext = sampledInterval;
-  ext += (durationToStart < thresh) ? durationToStart : avgDur / 2;
-  ext += (durationToEnd < thresh) ? durationToEnd : avgDur / 2;
-  double scaledDelta = delta * (ext / sampledInterval);
-  return isRate ? (scaledDelta / (double)(windowEnd - windowStart) * 1000.0) : scaledDelta;
 }
 
 // in-chunk correction lookup from the sparse drop table (the step function
@@ -354,19 +165,6 @@ __device__ __forceinline__ double d_corrected(const WS& ws, const ChunkMeta& m, 
   double x = ws.val[m.row0 + i];
   if (m.dropped) { if (isnan(x)) x = 0; x += d_corr_at(ws, m, i); }
   return x;
-}
-
-// NaN-aware f64 atomic min/max via CAS (group aggregation; RowAggregator semantics)
-__device__ void atomic_min_max_f64(double* addr, double val, bool is_min) {
-  unsigned long long* a = (unsigned long long*)addr;
-  unsigned long long old = *a, assumed;
-  do {
-    assumed = old;
-    double cur = __longlong_as_double((long long)assumed);
-    double nw = isnan(cur) ? val : (is_min ? fmin(cur, val) : fmax(cur, val));
-    if (!isnan(cur) && nw == cur) return;
-    old = atomicCAS(a, assumed, (unsigned long long)__double_as_longlong(nw));
-  } while (old != assumed);
 }
 
 // ---------------------------------------------------------------------------
@@ -1502,6 +1300,7 @@ struct fdb_dataset {
   int max_chunks;           // max chunks in one series
   int max_chunk_rows;       // max rows in one chunk (hist per-chunk LDS cap)
   int has_hist;             // dataset holds sect-delta histogram vectors
+  int fast_ok;              // single-chunk series, chunk spans fit i32 ms
 };
 
 extern "C" fdb_engine_t* fdb_engine_create(int32_t device) {
@@ -1570,12 +1369,13 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
     samples += dir[i].num_rows;
   }
   int max_group = 0, max_rows = 0, max_chunks = 0, max_chunk_rows = 0;
-  int has_hist = 0, has_scalar = 0;
+  int has_hist = 0, has_scalar = 0, spans_fit_i32 = 1;
   for (int64_t i2 = 0; i2 < nc; i2++) {
     uint16_t wf;
     memcpy(&wf, view.blob + dir[i2].val_off + 4, 2);
     if (wf == FDB_WF_HIST_SECTDELTA) has_hist = 1; else has_scalar = 1;
     if (dir[i2].num_rows > max_chunk_rows) max_chunk_rows = dir[i2].num_rows;
+    if (dir[i2].end_time - dir[i2].start_time >= (int64_t)1 << 31) spans_fit_i32 = 0;
   }
   if (has_hist && has_scalar) {
     fdb_set_error("mixed histogram and scalar series in one store are not "
@@ -1608,6 +1408,8 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
   d->max_chunks = max_chunks;
   d->max_chunk_rows = max_chunk_rows;
   d->has_hist = has_hist;
+  d->fast_ok = !has_hist && max_chunks <= 1 && max_chunk_rows <= 400 &&
+               spans_fit_i32;
 
   auto upload = [&](void** dst, const void* src, size_t bytes) -> bool {
     if (hipMalloc(dst, bytes ? bytes : 8) != hipSuccess) return false;
@@ -1652,10 +1454,29 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
 }
 
 // launch dispatch over the func template parameter
+// the fast single-chunk kernel handles this (dataset, query) pair?
+static bool fast_eligible(const fdb_dataset_t* d, const fdb_query_t* q) {
+  static int enabled = -1;
+  if (enabled < 0) {
+    const char* v = getenv("FDB_FAST");           // perf/parity experiments
+    enabled = (v && atoi(v) == 0) ? 0 : 1;
+  }
+  return enabled && d->fast_ok && fdb_fast_scan_supported(q->func_id) &&
+         q->step > 0 && q->step < ((int64_t)1 << 31) && q->window >= 0;
+}
+
 static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
                            double* dev_out, double* dev_cnt, double* dev_sq) {
   DirSoA dir{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
   int nw = fdb_num_windows(q);
+  if (fast_eligible(d, q) && q->agg_id == AGG_NONE) {
+    return fdb_launch_fast_scan(e->stream, d->blob, dir, d->series_first,
+                                d->series_nchunks, d->group_ids,
+                                d->series_by_group, d->num_series,
+                                q->start, q->step, q->window, nw,
+                                q->func_id, AGG_NONE, /*emit_group=*/0,
+                                dev_out, dev_cnt, dev_sq, q->_pad);
+  }
   // capacity tier: the common one-chunk shape runs 4 waves/block; series
   // spanning several chunks (long lookbacks) use the 1600-row/2-wave variant
   const bool large = d->max_rows > FDB_MAX_ROWS_PER_SERIES ||
@@ -1763,11 +1584,20 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   HIP_CHECK(hipEventCreate(&ev0));
   HIP_CHECK(hipEventCreate(&ev1));
 
-  // every agg runs two-phase: the scan fills an internal per-series [S×W]
-  // grid with plain stores, then a presenter (topk_kernel / group_reduce_
-  // kernel) folds it along the group-sorted index — no data-path atomics
+  // aggregated queries: the fused-group fast path folds fastReduce into the
+  // scan (per-lane register partials + one atomic burst per group change) and
+  // never materializes the [S×W] grid; other shapes run two-phase — the scan
+  // fills an internal per-series grid with plain stores, then a presenter
+  // (topk_kernel / group_reduce_kernel) folds it along the group-sorted index
+  static int fused_enabled = -1;
+  if (fused_enabled < 0) {
+    const char* v = getenv("FDB_FUSED_GROUP");   // perf/parity experiments
+    fused_enabled = (v && atoi(v) == 0) ? 0 : 1;
+  }
+  const bool fused = fused_enabled && q->agg_id != AGG_NONE && !is_topk &&
+                     nw <= 256 && fast_eligible(d, q);
   fdb_query_t qscan = *q;
-  if (q->agg_id != AGG_NONE) {
+  if (q->agg_id != AGG_NONE && !fused) {
     HIP_CHECK(hipMalloc(&per_grid, (size_t)d->num_series * nw * 8));
     qscan.agg_id = AGG_NONE;
   }
@@ -1777,6 +1607,30 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   for (int it = 0; it < total_runs; it++) {
     bool timed = it >= warmup;
     if (timed) HIP_CHECK(hipEventRecord(ev0, e->stream));
+    if (fused) {
+      // atomically-accumulated grids: reset per launch. MIN/MAX start the
+      // value grid at NaN (atomic_min_max_f64's empty-cell marker).
+      const bool mm = q->agg_id == AGG_MIN || q->agg_id == AGG_MAX;
+      HIP_CHECK(hipMemsetAsync(dev_out, mm ? 0xFF : 0, buf_len * 8, e->stream));
+      HIP_CHECK(hipMemsetAsync(dev_cnt, 0, out_len * 8, e->stream));
+      if (dev_sq && !(needs_sq && partial))
+        HIP_CHECK(hipMemsetAsync(dev_sq, 0, out_len * 8, e->stream));
+      DirSoA dirf{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
+      int32_t rc = fdb_launch_fast_scan(
+          e->stream, d->blob, dirf, d->series_first, d->series_nchunks,
+          d->group_ids, d->series_by_group, d->num_series,
+          q->start, q->step, q->window, nw, q->func_id, q->agg_id,
+          /*emit_group=*/1, dev_out, dev_cnt, dev_sq, q->_pad);
+      if (rc != FDB_OK) return rc;
+      if (timed) {
+        HIP_CHECK(hipEventRecord(ev1, e->stream));
+        HIP_CHECK(hipEventSynchronize(ev1));
+        float ms;
+        HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+        ms_sum += ms;
+      }
+      continue;
+    }
     int32_t rc = launch_scan(e, d, &qscan, per_grid ? per_grid : dev_out,
                              per_grid ? nullptr : dev_cnt, nullptr);
     if (rc != FDB_OK) return rc;

@@ -1,0 +1,314 @@
+// Shared device-side pieces of the MI355X scan engine: frozen-vector readers,
+// wave scans, the extrapolatedRate epilogue and launch plumbing types.
+// Layouts: DESIGN.md §2 (bit-faithful to the reference's off-heap vectors).
+#ifndef FDB_SCAN_COMMON_H
+#define FDB_SCAN_COMMON_H
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <cstring>
+#include <cmath>
+
+#include "chunk_format.h"
+
+// function / aggregation ids (include/filodb_amd.h; dispatch table
+// RangeFunction.scala:294-410, aggregator/RowAggregator.scala)
+enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
+       FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12,
+       FN_PRESENT=13, FN_TIMESTAMP=14, FN_ZSCORE=15,
+       FN_QUANTILE=16, FN_MAD=17, FN_PREDICT_LINEAR=18, FN_RATE_OVER_DELTA=19 };
+enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
+       AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9, AGG_GROUP=10 };
+
+struct DirSoA {
+  const uint64_t* ts_off;
+  const uint64_t* val_off;
+  const int64_t*  start_time;
+  const int64_t*  end_time;
+  const int32_t*  num_rows;
+};
+
+__device__ __forceinline__ uint16_t d_u16(const uint8_t* p) { uint16_t v; memcpy(&v, p, 2); return v; }
+__device__ __forceinline__ uint32_t d_u32(const uint8_t* p) { uint32_t v; memcpy(&v, p, 4); return v; }
+__device__ __forceinline__ int32_t  d_i32(const uint8_t* p) { int32_t v; memcpy(&v, p, 4); return v; }
+__device__ __forceinline__ int64_t  d_i64(const uint8_t* p) { int64_t v; memcpy(&v, p, 8); return v; }
+__device__ __forceinline__ double   d_f64(const uint8_t* p) { double v; memcpy(&v, p, 8); return v; }
+
+struct DVec {           // opened vector header
+  const uint8_t* idata;
+  int64_t init;
+  int32_t slope;
+  int n;
+  uint16_t wf;
+  uint8_t nbits, sign, dropped;
+};
+
+__device__ inline void d_vec_open(const uint8_t* p, DVec* v) {
+  v->wf = d_u16(p + 4);
+  v->dropped = (d_u16(p + 6) & FDB_DROP_MASK) != 0;
+  if (v->wf == FDB_WF_DDV) {
+    v->init = d_i64(p + FDB_DDV_OFF_INIT);
+    v->slope = d_i32(p + FDB_DDV_OFF_SLOPE);
+    const uint8_t* inner = p + FDB_DDV_OFF_INNER;
+    v->nbits = inner[6] & FDB_NBITS_MASK;
+    v->sign = (inner[6] & FDB_SIGN_MASK) != 0;
+    v->idata = inner + FDB_PRIM_OFF_DATA;
+    int numBytes = (int)d_u32(inner);
+    int bitShift = inner[7] & 0x3f;
+    v->n = ((numBytes - 4) * 8 + (bitShift != 0 ? bitShift - 8 : 0)) / v->nbits;
+  } else if (v->wf == FDB_WF_DDV_CONST) {
+    v->n = d_i32(p + FDB_DDVC_OFF_NELEM);
+    v->init = d_i64(p + FDB_DDVC_OFF_INIT);
+    v->slope = d_i32(p + FDB_DDVC_OFF_SLOPE);
+    v->idata = nullptr; v->nbits = 0; v->sign = 0;
+  } else {
+    v->n = ((int)d_u32(p) - 4) / 8;
+    v->idata = p + FDB_PRIM_OFF_DATA;
+    v->init = 0; v->slope = 0; v->nbits = 64; v->sign = 1;
+  }
+}
+
+__device__ __forceinline__ int64_t d_inner_at(const DVec* v, int i) {
+  switch (v->nbits) {
+    case 32: return d_i32(v->idata + 4 * (size_t)i);
+    case 16: { int32_t x = (int16_t)d_u16(v->idata + 2 * (size_t)i);
+               return v->sign ? x : (x & 0xffff); }
+    case 8:  { int32_t x = (int8_t)v->idata[i];
+               return v->sign ? x : (x & 0xff); }
+    case 4:  return (v->idata[i >> 1] >> ((i & 1) * 4)) & 0x0f;
+    case 2:  return (v->idata[i >> 2] >> ((i & 3) * 2)) & 0x03;
+  }
+  return 0;
+}
+
+// wave-local LDS ordering only: s_waitcnt lgkmcnt(0) — unlike s_waitcnt(0)
+// this does NOT drain vmcnt, so outstanding global loads/stores/atomics keep
+// flying across the fence
+__device__ __forceinline__ void d_wait_lds() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+}
+
+__device__ __forceinline__ int64_t d_lv_at(const DVec* v, int i) {
+  if (v->wf == FDB_WF_DDV) return v->init + (int64_t)v->slope * i + d_inner_at(v, i);
+  if (v->wf == FDB_WF_DDV_CONST) return v->init + (int64_t)v->slope * i;
+  return d_i64(v->idata + 8 * (size_t)i);
+}
+
+__device__ __forceinline__ double d_dv_at(const DVec* v, int i) {
+  if (v->wf == FDB_WF_PRIM64) return d_f64(v->idata + 8 * (size_t)i);
+  return (double)d_lv_at(v, i);
+}
+
+// wave-wide inclusive prefix sums (64 lanes)
+__device__ __forceinline__ double wave_incl_scan(double x, int lane) {
+  for (int off = 1; off < 64; off <<= 1) {
+    double t = __shfl_up(x, off);
+    if (lane >= off) x += t;
+  }
+  return x;
+}
+__device__ __forceinline__ int wave_incl_scan_i(int x, int lane) {
+  for (int off = 1; off < 64; off <<= 1) {
+    int t = __shfl_up(x, off);
+    if (lane >= off) x += t;
+  }
+  return x;
+}
+
+// vectorized decode: whole chunk into LDS, wide loads, high ILP.
+// (element semantics identical to d_lv_at/d_dv_at, which remain the
+//  reference implementations / fallback)
+template <bool AS_DOUBLE>
+__device__ inline void d_decode_chunk(const DVec& v, int n, int64_t* tout,
+                                      double* dout, int lane) {
+  if (v.wf == FDB_WF_DDV_CONST) {
+    for (int i = lane; i < n; i += 64) {
+      int64_t x = v.init + (int64_t)v.slope * i;
+      if (AS_DOUBLE) dout[i] = (double)x; else tout[i] = x;
+    }
+    return;
+  }
+  if (v.wf == FDB_WF_PRIM64) {
+    // raw 64-bit payload, 8-byte aligned (+8 from a 64B-aligned base)
+    for (int i = lane; i < n; i += 64) {
+      if (AS_DOUBLE) dout[i] = d_f64(v.idata + 8 * (size_t)i);
+      else           tout[i] = d_i64(v.idata + 8 * (size_t)i);
+    }
+    return;
+  }
+  // packed DDV inner data starts at +28 (4-byte aligned only)
+  if (v.nbits == 16) {
+    for (int i0 = 4 * lane; i0 < n; i0 += 256) {
+      uint32_t lo = d_u32(v.idata + 2 * (size_t)i0);
+      uint32_t hi = d_u32(v.idata + 2 * (size_t)i0 + 4);
+      uint64_t w = ((uint64_t)hi << 32) | lo;
+      #pragma unroll
+      for (int k = 0; k < 4; k++) {
+        if (i0 + k < n) {
+          int32_t d = (int32_t)(int16_t)(uint16_t)(w >> (16 * k));
+          if (!v.sign) d &= 0xffff;
+          int64_t x = v.init + (int64_t)v.slope * (i0 + k) + d;
+          if (AS_DOUBLE) dout[i0 + k] = (double)x; else tout[i0 + k] = x;
+        }
+      }
+    }
+    return;
+  }
+  if (v.nbits == 8) {
+    for (int i0 = 8 * lane; i0 < n; i0 += 512) {
+      uint32_t lo = d_u32(v.idata + (size_t)i0);
+      uint32_t hi = d_u32(v.idata + (size_t)i0 + 4);
+      uint64_t w = ((uint64_t)hi << 32) | lo;
+      #pragma unroll
+      for (int k = 0; k < 8; k++) {
+        if (i0 + k < n) {
+          int32_t d = (int32_t)(int8_t)(uint8_t)(w >> (8 * k));
+          if (!v.sign) d &= 0xff;
+          int64_t x = v.init + (int64_t)v.slope * (i0 + k) + d;
+          if (AS_DOUBLE) dout[i0 + k] = (double)x; else tout[i0 + k] = x;
+        }
+      }
+    }
+    return;
+  }
+  if (v.nbits == 32) {
+    for (int i0 = 2 * lane; i0 < n; i0 += 128) {
+      #pragma unroll
+      for (int k = 0; k < 2; k++) {
+        if (i0 + k < n) {
+          int64_t x = v.init + (int64_t)v.slope * (i0 + k)
+                    + d_i32(v.idata + 4 * (size_t)(i0 + k));
+          if (AS_DOUBLE) dout[i0 + k] = (double)x; else tout[i0 + k] = x;
+        }
+      }
+    }
+    return;
+  }
+  // nbits 2/4 fallback (rare)
+  for (int i = lane; i < n; i += 64) {
+    int64_t x = v.init + (int64_t)v.slope * i + d_inner_at(&v, i);
+    if (AS_DOUBLE) dout[i] = (double)x; else tout[i] = x;
+  }
+}
+
+// i32-offset decode variant: timestamps relative to the chunk's first ts.
+// Caller guarantees the chunk's time span fits i32 (upload-time check).
+__device__ inline void d_decode_ts_offsets(const DVec& v, int n, int64_t ts0,
+                                           int32_t* tout, int lane) {
+  if (v.wf == FDB_WF_DDV_CONST) {
+    for (int i = lane; i < n; i += 64)
+      tout[i] = (int32_t)((int64_t)v.slope * i + (v.init - ts0));
+    return;
+  }
+  if (v.wf == FDB_WF_PRIM64) {
+    for (int i = lane; i < n; i += 64)
+      tout[i] = (int32_t)(d_i64(v.idata + 8 * (size_t)i) - ts0);
+    return;
+  }
+  const int64_t base = v.init - ts0;
+  if (v.nbits == 16) {
+    for (int i0 = 4 * lane; i0 < n; i0 += 256) {
+      uint32_t lo = d_u32(v.idata + 2 * (size_t)i0);
+      uint32_t hi = d_u32(v.idata + 2 * (size_t)i0 + 4);
+      uint64_t w = ((uint64_t)hi << 32) | lo;
+      #pragma unroll
+      for (int k = 0; k < 4; k++) {
+        if (i0 + k < n) {
+          int32_t d = (int32_t)(int16_t)(uint16_t)(w >> (16 * k));
+          if (!v.sign) d &= 0xffff;
+          tout[i0 + k] = (int32_t)(base + (int64_t)v.slope * (i0 + k) + d);
+        }
+      }
+    }
+    return;
+  }
+  if (v.nbits == 32) {
+    for (int i0 = 2 * lane; i0 < n; i0 += 128) {
+      #pragma unroll
+      for (int k = 0; k < 2; k++)
+        if (i0 + k < n)
+          tout[i0 + k] = (int32_t)(base + (int64_t)v.slope * (i0 + k)
+                                   + d_i32(v.idata + 4 * (size_t)(i0 + k)));
+    }
+    return;
+  }
+  for (int i = lane; i < n; i += 64)
+    tout[i] = (int32_t)(base + (int64_t)v.slope * i + d_inner_at(&v, i));
+}
+
+// extrapolatedRate (RateFunctions.scala:72-111), computed in the ms domain
+// (the reference's /1000 unit conversions cancel — every comparison and the
+// final ext/sampledInterval ratio are scale-invariant; ≤2 ulp vs the
+// seconds-domain oracle, inside the 1e-9 north_star tolerance).
+//   inv_ns1    = 1/(numSamples-1)  (from the per-block reciprocal table)
+//   rate_scale = 1000.0/(windowEnd-windowStart) precomputed on the host
+__device__ __forceinline__ double d_extrap_rate_ms(
+    double dts /*t1-wStart*/, double dte /*wEnd-t2*/, double si /*t2-t1, >0*/,
+    double inv_ns1, double v1, double v2,
+    bool isCounter, bool isRate, double rate_scale) {
+  double inv_si = 1.0 / si;
+  double avg = si * inv_ns1;
+  double delta = v2 - v1;
+  if (isCounter && delta > 0 && v1 >= 0) {
+    double dtz = si * (v1 / delta);      // durationToZero
+    dts = dtz < dts ? dtz : dts;
+  }
+  double thresh = avg * 1.1;
+  double ext = si + (dts < thresh ? dts : avg * 0.5)
+                  + (dte < thresh ? dte : avg * 0.5);
+  double scaled = delta * (ext * inv_si);
+  return isRate ? scaled * rate_scale : scaled;
+}
+
+// original i64/seconds-domain form (kept for the general/streaming paths)
+__device__ inline double d_extrapolated_rate(int64_t windowStart, int64_t windowEnd,
+                                             int numSamples,
+                                             int64_t t1, double v1, int64_t t2, double v2,
+                                             bool isCounter, bool isRate) {
+  double durationToStart = (double)(t1 - windowStart) / 1000.0;
+  double durationToEnd = (double)(windowEnd - t2) / 1000.0;
+  double sampledInterval = (double)(t2 - t1) / 1000.0;
+  double avgDur = sampledInterval / ((double)numSamples - 1);
+  double delta = v2 - v1;
+  if (isCounter && delta > 0 && v1 >= 0) {
+    double durationToZero = sampledInterval * (v1 / delta);
+    if (durationToZero < durationToStart) durationToStart = durationToZero;
+  }
+  double thresh = avgDur * 1.1;
+  double ext = sampledInterval;
+  ext += (durationToStart < thresh) ? durationToStart : avgDur / 2;
+  ext += (durationToEnd < thresh) ? durationToEnd : avgDur / 2;
+  double scaledDelta = delta * (ext / sampledInterval);
+  return isRate ? (scaledDelta / (double)(windowEnd - windowStart) * 1000.0) : scaledDelta;
+}
+
+// NaN-aware f64 atomic min/max via CAS (group aggregation; RowAggregator semantics)
+__device__ inline void atomic_min_max_f64(double* addr, double val, bool is_min) {
+  unsigned long long* a = (unsigned long long*)addr;
+  unsigned long long old = *a, assumed;
+  do {
+    assumed = old;
+    double cur = __longlong_as_double((long long)assumed);
+    double nw = isnan(cur) ? val : (is_min ? fmin(cur, val) : fmax(cur, val));
+    if (!isnan(cur) && nw == cur) return;
+    old = atomicCAS(a, assumed, (unsigned long long)__double_as_longlong(nw));
+  } while (old != assumed);
+}
+
+// floor(a/b) clamped into the window index range [-1, hi+1], for b>0:
+// double reciprocal + one-step integer fixup (i64 division is software-
+// emulated and costs hundreds of cycles). Outside [-1, hi+1] the exact value
+// is irrelevant — callers clamp ranges to [0, hi) — so the guards also keep
+// the fixup products within i64 (callers ensure b <= 2^31).
+__device__ __forceinline__ int fdiv_floor_win(int64_t a, int64_t b, double inv_b,
+                                              int hi) {
+  double g = (double)a * inv_b;
+  if (g < -1.0) return -1;
+  if (g > (double)hi + 1.0) return hi + 1;
+  int64_t w = (int64_t)__builtin_floor(g);
+  w += ((w + 1) * b <= a);
+  w -= (w * b > a);
+  return (int)w;
+}
+
+#endif  // FDB_SCAN_COMMON_H

@@ -1,0 +1,546 @@
+// Fast scan kernel: the single-chunk-per-series shape (BASELINE configs[1-3]).
+//
+// Ground-up round-2 redesign of the window phase (profiles/README.md round-1
+// findings: the old kernel was latency-bound on per-window search + branchy
+// f64 epilogue chains at 4 waves/SIMD). What changed:
+//
+//  * window row-ranges come from ONE inversion scan over the rows instead of
+//    per-window searches: row i computes two floor-divisions and writes the
+//    window ranges it bounds into sw[]/ew[] (startRow/endRow per window) —
+//    O(rows + windows) total, exact for arbitrary sorted timestamps
+//    (replaces DeltaDeltaDataReader.binarySearch/ceilingIndex on the device;
+//    semantics: WindowedChunkIterator, ChunkSetInfo.scala:467-511)
+//  * timestamps live in LDS as i32 offsets from the chunk's first ts (halves
+//    the LDS footprint and read traffic; chunk span is checked at upload)
+//  * the extrapolatedRate epilogue runs in the ms domain with a per-block
+//    reciprocal table for 1/(numSamples-1): 2 f64 divisions per window
+//    instead of 6-7 (scan_common.h d_extrap_rate_ms; ≤2 ulp vs the oracle,
+//    inside the 1e-9 north_star tolerance)
+//  * window results are branch-lean selects; one store per window
+//
+// Per-window semantics are IDENTICAL to the round-1 kernel's single-chunk
+// stage C (parity-green against the oracle): RateFunctions.scala:72-111,
+// 230-289; AggrOverTimeFunctions.scala:553-605,924-1028,1082-1224;
+// RangeFunction.scala:595-745.
+//
+// EMIT=1 ("fused group" emit) folds fastReduce into the scan: each lane keeps
+// its windows' group-partials in registers while the wave walks a contiguous
+// slab of the group-sorted series index, flushing one atomicAdd burst per
+// group change — the [S×W] intermediate grid of the two-phase reduce is never
+// materialized (AggrOverRangeVectors.scala:320-377 semantics; raw sums +
+// counts out, agg_present_kernel applies the presentation step).
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <cstring>
+#include <cmath>
+
+#include "chunk_format.h"
+#include "scan_common.h"
+#include "../../include/filodb_amd.h"
+
+void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
+
+#define FAST_ROWS 400        // reference chunk row cap (filodb-defaults.conf:835)
+#define FAST_TILE 256        // windows per tile (sw/ew LDS arrays)
+#define FAST_WAVES 4
+#define FAST_DROPS 32        // sparse counter-reset table capacity
+
+// kinds (which auxiliary LDS arrays a function family needs)
+#define K_RATE    0
+#define K_PFX     1
+#define K_PFX_SQ  2
+#define K_MINMAX  3
+#define K_CHANGES 4
+#define K_LAST    5
+
+template <int FUNC> struct FKind { static constexpr int v =
+    (FUNC <= FN_DELTA) ? K_RATE :
+    (FUNC == FN_SUM || FUNC == FN_AVG || FUNC == FN_COUNT ||
+     FUNC == FN_RATE_OVER_DELTA) ? K_PFX :
+    (FUNC == FN_STDDEV || FUNC == FN_STDVAR) ? K_PFX_SQ :
+    (FUNC == FN_MIN || FUNC == FN_MAX) ? K_MINMAX :
+    (FUNC == FN_CHANGES) ? K_CHANGES : K_LAST; };
+
+template <int KIND>
+struct FWs {                        // per-wave LDS workspace
+  int32_t tso[FAST_ROWS];           // ts - ts0 (i32 offsets)
+  double  val[FAST_ROWS];           // raw values, or prefix sums (PFX kinds)
+  uint16_t cnt[(KIND == K_PFX || KIND == K_PFX_SQ || KIND == K_CHANGES)
+                   ? FAST_ROWS : 1];
+  double  sq[KIND == K_PFX_SQ ? FAST_ROWS : 1];
+  double  grp[KIND == K_MINMAX ? (FAST_ROWS + 7) / 8 : 1];
+  int16_t dpos[KIND == K_RATE ? FAST_DROPS : 1];
+  double  dcum[KIND == K_RATE ? FAST_DROPS : 1];
+  int16_t sw[FAST_TILE];            // per-window startRow (n = none)
+  int16_t ew[FAST_TILE];            // per-window endRow (-1 = none)
+};
+
+template <int KIND> struct NeedsInv {   // reciprocal table users
+  static constexpr bool v = KIND == K_RATE || KIND == K_PFX || KIND == K_PFX_SQ;
+};
+
+// in-chunk correction at row i from the sparse drop table (the step function
+// CorrectingDoubleVectorReader :325-342 materializes as corrected[])
+template <typename WS>
+__device__ __forceinline__ double f_corr_at(const WS& ws, int dcount, bool dense,
+                                            int n, int i) {
+  if (dense) {      // >FAST_DROPS resets in the chunk: serial recompute (rare)
+    double corr = 0, last = -1.7976931348623157e308;
+    for (int j = 0; j <= i; j++) {
+      double x = ws.val[j];
+      if (isnan(x)) x = 0;
+      if (x < last) corr += last;
+      last = x;
+    }
+    return corr;
+  }
+  double c = 0;
+  for (int j = 0; j < dcount; j++) {
+    if (ws.dpos[j] <= i) c = ws.dcum[j]; else break;
+  }
+  return c;
+}
+
+// min-waves/SIMD: PFX_SQ is LDS-bound at 3 blocks/CU; the group-emit variant
+// carries 12 accumulator registers (4 at min-waves 5 spills ~33 VGPRs); the
+// plain kinds fit 96 VGPRs = 5 waves/SIMD
+template <int FUNC, int EMIT>
+__global__ __launch_bounds__(FAST_WAVES * 64,
+                             (FKind<FUNC>::v == K_PFX_SQ) ? 3 : (EMIT ? 4 : 5))
+void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
+                      const int32_t* __restrict__ series_first,
+                      const int32_t* __restrict__ series_nchunks,
+                      const int32_t* __restrict__ group_ids,
+                      const int32_t* __restrict__ sbg,   // EMIT=1: group-sorted order
+                      int num_series,
+                      int64_t qstart, int64_t qstep, int64_t qwindow,
+                      int num_windows, int agg_id,
+                      double* __restrict__ out,
+                      double* __restrict__ out_cnt,
+                      double* __restrict__ out_sq)
+{
+  constexpr int KIND = FKind<FUNC>::v;
+  constexpr bool IS_COUNTER = (FUNC == FN_RATE || FUNC == FN_INCREASE);
+  __shared__ FWs<KIND> ws_all[FAST_WAVES];
+  __shared__ double inv_tab[NeedsInv<KIND>::v ? FAST_ROWS + 2 : 1];
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  FWs<KIND>& ws = ws_all[wave];
+
+  if (NeedsInv<KIND>::v) {
+    for (int i = threadIdx.x; i < FAST_ROWS + 2; i += FAST_WAVES * 64)
+      inv_tab[i] = 1.0 / (double)i;          // [0] = +inf (never a divisor)
+    __syncthreads();
+  }
+
+  const double inv_step = 1.0 / (double)qstep;
+  const double rate_scale = 1000.0 / (double)qwindow;
+
+  // series iteration: grid-stride for the plain grid emit; a contiguous slab
+  // of the group-sorted index for the fused-group emit (so one wave sees each
+  // group as a run and flushes once per group change)
+  int pos0, pos1, pos_step;
+  if (EMIT == 1) {
+    const int gw = blockIdx.x * FAST_WAVES + wave;
+    const int nwaves = gridDim.x * FAST_WAVES;
+    const int slab = (num_series + nwaves - 1) / nwaves;
+    pos0 = gw * slab;
+    pos1 = min(num_series, pos0 + slab);
+    pos_step = 1;
+  } else {
+    pos0 = blockIdx.x * FAST_WAVES + wave;
+    pos1 = num_series;
+    pos_step = gridDim.x * FAST_WAVES;
+  }
+
+  // EMIT=1 per-lane group accumulators (windows lane+64k, k<4; W <= 256)
+  double accS[4], accQ[4], accC[4];
+  int cur_grp = -1;
+  if (EMIT == 1) {
+    #pragma unroll
+    for (int k = 0; k < 4; k++) { accS[k] = NAN; accQ[k] = 0; accC[k] = 0; }
+  }
+  auto flush_group = [&](int g) {
+    #pragma unroll
+    for (int k = 0; k < 4; k++) {
+      int w = lane + 64 * k;
+      if (w < num_windows && accC[k] > 0) {
+        size_t cell = (size_t)g * num_windows + w;
+        if (agg_id == AGG_MIN || agg_id == AGG_MAX)
+          atomic_min_max_f64(&out[cell], accS[k], agg_id == AGG_MIN);
+        else
+          atomicAdd(&out[cell],
+                    (agg_id == AGG_COUNT || agg_id == AGG_GROUP) ? accC[k] : accS[k]);
+        atomicAdd(&out_cnt[cell], accC[k]);
+        if (out_sq) atomicAdd(&out_sq[cell], accQ[k]);
+      }
+      accS[k] = NAN; accQ[k] = 0; accC[k] = 0;
+    }
+  };
+
+  for (int pos = pos0; pos < pos1; pos += pos_step) {
+    const int sid = (EMIT == 1) ? sbg[pos] : pos;
+    const int first = series_first[sid];
+    const int nch = series_nchunks[sid];
+
+    // ---- decode: the single chunk into LDS (i32 ts offsets + f64 values) ---
+    int n = 0;
+    int64_t ts0 = 0;
+    bool dropped = false;
+    if (nch >= 1) {
+      DVec tv, vv;
+      d_vec_open(blob + dir.ts_off[first], &tv);
+      d_vec_open(blob + dir.val_off[first], &vv);
+      n = dir.num_rows[first];
+      if (n > FAST_ROWS || n > tv.n) n = 0;     // guarded at upload
+      if (n > 0) {
+        ts0 = d_lv_at(&tv, 0);                  // wave-uniform
+        d_decode_ts_offsets(tv, n, ts0, ws.tso, lane);
+        d_decode_chunk<true>(vv, n, nullptr, ws.val, lane);
+        dropped = vv.dropped;
+      }
+    }
+    d_wait_lds();
+    __builtin_amdgcn_wave_barrier();
+
+    // ---- meta: kind-specific per-row structures ----------------------------
+    int dcount = 0;
+    bool dense = false;
+    if constexpr (KIND == K_RATE) {
+      // counter-reset scan → sparse (position, cumulative correction) table
+      // (CorrectingDoubleVectorReader :325-342; NaN→0 like the reference)
+      if (dropped && n > 0) {
+        double carry_corr = 0;
+        double carry_x = -1.7976931348623157e308;
+        for (int base = 0; base < n; base += 64) {
+          int i = base + lane;
+          double raw = (i < n) ? ws.val[i] : 0;
+          double x = (i < n && !isnan(raw)) ? raw : 0;
+          double px = __shfl_up(x, 1);
+          if (lane == 0) px = carry_x;
+          double ci = (i < n && x < px) ? px : 0;
+          double scan = wave_incl_scan(ci, lane);
+          uint64_t mask = __ballot(ci != 0);
+          int here = __popcll(mask);
+          if (here) {
+            if (dcount + here > FAST_DROPS) {
+              dense = true;
+            } else if (ci != 0) {
+              int slot = dcount + __popcll(mask & ((1ULL << lane) - 1));
+              ws.dpos[slot] = (int16_t)i;
+              ws.dcum[slot] = carry_corr + scan;
+            }
+            if (!dense) dcount += here;
+          }
+          carry_corr += __shfl(scan, 63);
+          carry_x = __shfl(x, 63);
+        }
+        if (dense) dcount = 0;
+      }
+    }
+    if constexpr (KIND == K_PFX || KIND == K_PFX_SQ) {
+      double carry = 0, carry_sq = 0;
+      int ccarry = 0;
+      for (int base = 0; base < n; base += 64) {
+        int i = base + lane;
+        double raw = (i < n) ? ws.val[i] : NAN;
+        bool ok = (i < n) && !isnan(raw);
+        double x = ok ? raw : 0;
+        double s = wave_incl_scan(x, lane);
+        int cs = wave_incl_scan_i(ok ? 1 : 0, lane);
+        double sqs = 0;
+        if constexpr (KIND == K_PFX_SQ) sqs = wave_incl_scan(x * x, lane);
+        __builtin_amdgcn_wave_barrier();        // all raw reads precede writes
+        if (i < n) {
+          ws.val[i] = carry + s;                // val[] becomes the prefix
+          ws.cnt[i] = (uint16_t)(ccarry + cs);
+          if constexpr (KIND == K_PFX_SQ) ws.sq[i] = carry_sq + sqs;
+        }
+        carry += __shfl(s, 63);
+        ccarry += __shfl(cs, 63);
+        if constexpr (KIND == K_PFX_SQ) carry_sq += __shfl(sqs, 63);
+      }
+    }
+    if constexpr (KIND == K_MINMAX) {
+      constexpr bool IS_MIN = (FUNC == FN_MIN);
+      for (int base = 0; base < n; base += 64) {
+        int i = base + lane;
+        double x = (i < n) ? ws.val[i] : NAN;
+        #pragma unroll
+        for (int off = 1; off < 8; off <<= 1) {
+          double o = __shfl_xor(x, off);
+          if (!isnan(o) && (isnan(x) || (IS_MIN ? o < x : o > x))) x = o;
+        }
+        if ((lane & 7) == 0 && i < n) ws.grp[i >> 3] = x;
+      }
+    }
+    if constexpr (KIND == K_CHANGES) {
+      int ccarry = 0;
+      double carry_x = NAN;
+      for (int base = 0; base < n; base += 64) {
+        int i = base + lane;
+        double x = (i < n) ? ws.val[i] : NAN;
+        double px = __shfl_up(x, 1);
+        if (lane == 0) px = carry_x;
+        int ind = (i > 0 && i < n && !isnan(x) && !isnan(px) && x != px) ? 1 : 0;
+        int s = wave_incl_scan_i(ind, lane);
+        if (i < n) ws.cnt[i] = (uint16_t)(ccarry + s);
+        ccarry += __shfl(s, 63);
+        carry_x = __shfl(x, 63);
+      }
+    }
+    d_wait_lds();
+    __builtin_amdgcn_wave_barrier();
+
+    // window time base: wEnd(w) = qstart + w*qstep; offsets are vs ts0
+    const int64_t Ae = ts0 - qstart;            // wEnd >= ts  ⟺ w*qstep >= o+Ae
+    if (EMIT == 1) {
+      const int grp_id = group_ids[sid];
+      if (grp_id != cur_grp) {
+        if (cur_grp >= 0) flush_group(cur_grp);
+        cur_grp = grp_id;
+      }
+    }
+
+    for (int tb = 0; tb < num_windows; tb += FAST_TILE) {
+      const int tn = min(FAST_TILE, num_windows - tb);
+      // prefill: no-start sentinel n, no-end sentinel -1
+      for (int i = lane; i < tn; i += 64) { ws.sw[i] = (int16_t)n; ws.ew[i] = -1; }
+      d_wait_lds();
+      __builtin_amdgcn_wave_barrier();
+
+      // ---- inversion scan: rows → window boundaries ------------------------
+      // c_i = first w with wEnd >= ts_i   = ceil((o_i + Ae) / qstep)
+      // d_i = last  w with wStart <= ts_i = floor((o_i + Ae + qwindow) / qstep)
+      // row i ends   windows [c_{i-1}, c_i)   with e = i-1   (wEnd < ts_i)
+      // row i starts windows (d_{i-1}, d_i]   with s = i     (wStart > ts_{i-1})
+      int c_carry = 0, d_carry = -1;
+      for (int base = 0; base < n; base += 64) {
+        const int i = base + lane;
+        const bool live = i < n;
+        int64_t o = live ? (int64_t)ws.tso[i] : 0;
+        int ci = live ? fdiv_floor_win(o + Ae + qstep - 1, qstep, inv_step,
+                                       num_windows) : 0;
+        int di = live ? fdiv_floor_win(o + Ae + qwindow, qstep, inv_step,
+                                       num_windows) : 0;
+        int cprev = __shfl_up(ci, 1);
+        int dprev = __shfl_up(di, 1);
+        if (lane == 0) { cprev = c_carry; dprev = d_carry; }
+        if (live) {
+          if (i > 0) {
+            int lo = max(cprev, tb), hi = min(ci, tb + tn);
+            for (int w = lo; w < hi; w++) ws.ew[w - tb] = (int16_t)(i - 1);
+          }
+          int lo = (i == 0) ? tb : max(dprev + 1, tb);
+          int hi = min(di, tb + tn - 1);
+          for (int w = lo; w <= hi; w++) ws.sw[w - tb] = (int16_t)i;
+        }
+        const int lastl = min(63, n - 1 - base);
+        c_carry = __shfl(ci, lastl);
+        d_carry = __shfl(di, lastl);
+      }
+      if (n > 0) {          // tail: windows with wEnd >= ts_{n-1} end at n-1
+        for (int w = max(c_carry, tb) + lane; w < tb + tn; w += 64)
+          ws.ew[w - tb] = (int16_t)(n - 1);
+      }
+      d_wait_lds();
+      __builtin_amdgcn_wave_barrier();
+
+      // ---- window phase: 4 windows per lane, branch-lean -------------------
+      #pragma unroll
+      for (int k = 0; k < 4; k++) {
+        const int wi = lane + 64 * k;
+        const int w = tb + wi;
+        const bool wok = wi < tn;
+        const int s = wok ? ws.sw[wi] : 0;
+        const int e = wok ? ws.ew[wi] : -1;
+        double res = NAN;
+
+        if constexpr (KIND == K_RATE) {
+          // ChunkedRateFunctionBase + extrapolatedRate; e>s implies s valid,
+          // covers empty and single-sample windows (highestTime>lowestTime)
+          if (wok && e > s) {
+            const int t1 = ws.tso[s], t2 = ws.tso[e];
+            if (t2 > t1) {
+              double v1 = ws.val[s], v2 = ws.val[e];
+              if (IS_COUNTER && dropped) {
+                if (isnan(v1)) v1 = 0;
+                if (isnan(v2)) v2 = 0;
+                v1 += f_corr_at(ws, dcount, dense, n, s);
+                v2 += f_corr_at(ws, dcount, dense, n, e);
+              }
+              const int64_t wEndOff = (int64_t)w * qstep - Ae;   // wEnd - ts0
+              res = d_extrap_rate_ms((double)(t1 - (wEndOff - qwindow)),
+                                     (double)(wEndOff - t2),
+                                     (double)(t2 - t1),
+                                     inv_tab[e - s], v1, v2,
+                                     IS_COUNTER, FUNC == FN_RATE, rate_scale);
+            }
+          }
+        } else if constexpr (KIND == K_PFX) {
+          if (wok && e >= s && e >= 0) {
+            double ps = ws.val[e] - (s ? ws.val[s - 1] : 0.0);
+            int pc = (int)ws.cnt[e] - (s ? (int)ws.cnt[s - 1] : 0);
+            if constexpr (FUNC == FN_SUM) res = pc > 0 ? ps : NAN;
+            else if constexpr (FUNC == FN_COUNT) res = (double)pc;
+            else if constexpr (FUNC == FN_AVG)
+              res = pc > 0 ? ps * inv_tab[pc] : NAN;
+            else   // FN_RATE_OVER_DELTA (RateFunctions.scala:424-445)
+              res = (pc > 0 ? ps : NAN) * rate_scale;
+          }
+        } else if constexpr (KIND == K_PFX_SQ) {
+          if (wok && e >= s && e >= 0) {
+            double ps = ws.val[e] - (s ? ws.val[s - 1] : 0.0);
+            int pc = (int)ws.cnt[e] - (s ? (int)ws.cnt[s - 1] : 0);
+            if (pc > 0) {
+              double qs = ws.sq[e] - (s ? ws.sq[s - 1] : 0.0);
+              double inv = inv_tab[pc];
+              double avg = ps * inv;
+              double r = qs * inv - avg * avg;
+              res = (FUNC == FN_STDDEV) ? sqrt(r) : r;
+            }
+          }
+        } else if constexpr (KIND == K_MINMAX) {
+          if (wok && e >= s && e >= 0) {
+            constexpr bool IS_MIN = (FUNC == FN_MIN);
+            double mm = NAN;
+            auto acc = [&](double x) {
+              if (!isnan(x) && (isnan(mm) || (IS_MIN ? x < mm : x > mm))) mm = x;
+            };
+            int ga = (s + 7) >> 3, gb = (e + 1) >> 3;
+            if (ga < gb) {
+              for (int i = s; i < ga * 8; i++) acc(ws.val[i]);
+              for (int g = ga; g < gb; g++) acc(ws.grp[g]);
+              for (int i = gb * 8; i <= e; i++) acc(ws.val[i]);
+            } else {
+              for (int i = s; i <= e; i++) acc(ws.val[i]);
+            }
+            res = mm;
+          }
+        } else if constexpr (KIND == K_CHANGES) {
+          // single chunk: change-indicator prefix over (s, e]; prev starts NaN
+          if (wok && e >= s && e >= 0)
+            res = (double)((int)ws.cnt[e] - (int)ws.cnt[s]);
+        } else {  // K_LAST family
+          if constexpr (FUNC == FN_TIMESTAMP) {
+            // TimestampChunkedFunction: no window-start bound; seconds
+            if (wok && e >= 0) res = (double)(ts0 + ws.tso[e]) / 1000.0;
+          } else if constexpr (FUNC == FN_PRESENT) {
+            if (wok && e >= s && e >= 0) {
+              double v = ws.val[e];
+              if (!isnan(v)) res = 1.0;
+              else if (e > 0) res = isnan(ws.val[e - 1]) ? NAN : 1.0;
+            }
+          } else if constexpr (FUNC == FN_LAST) {
+            // ts[e] >= ts[s] >= wStart holds whenever the range is nonempty
+            if (wok && e >= s && e >= 0) res = ws.val[e];
+          } else {  // FN_ZSCORE (AggrOverTimeFunctions.scala:1592-1603)
+            if (wok && e >= s && e >= 0) {
+              double sm = NAN, sq = NAN, lastv = NAN;
+              int pc = 0;
+              for (int i = s; i <= e; i++) {
+                double x = ws.val[i];
+                if (isnan(x)) continue;
+                if (isnan(sm)) { sm = 0; sq = 0; }
+                if (i == e) lastv = x;
+                sm += x; sq += x * x; pc++;
+              }
+              if (pc > 0) {
+                double avg = sm / pc;
+                double sd = sqrt(sq / pc - avg * avg);
+                res = (lastv - avg) / sd;
+              } else res = isnan(sm) ? sm : 0;
+            }
+          }
+        }
+
+        if (EMIT == 0) {
+          if (wok && w < num_windows)
+            out[(size_t)sid * num_windows + w] = res;
+        } else if (!isnan(res)) {
+          // fastReduce map step: NaN rows skipped (RowAggregator semantics)
+          switch (agg_id) {
+            case AGG_MIN:
+              accS[k] = (isnan(accS[k]) || res < accS[k]) ? res : accS[k];
+              accC[k] += 1.0; break;
+            case AGG_MAX:
+              accS[k] = (isnan(accS[k]) || res > accS[k]) ? res : accS[k];
+              accC[k] += 1.0; break;
+            case AGG_STDDEV: case AGG_STDVAR:
+              accQ[k] += res * res;   // fall through to sum+count
+              accS[k] = isnan(accS[k]) ? res : accS[k] + res;
+              accC[k] += 1.0; break;
+            case AGG_COUNT: case AGG_GROUP:
+              accC[k] += 1.0; break;
+            default:   // AGG_SUM / AGG_AVG
+              accS[k] = isnan(accS[k]) ? res : accS[k] + res;
+              accC[k] += 1.0; break;
+          }
+        }
+      }
+      d_wait_lds();
+      __builtin_amdgcn_wave_barrier();
+    }  // tiles
+  }  // series
+  if (EMIT == 1 && cur_grp >= 0) flush_group(cur_grp);
+}
+
+// ---------------------------------------------------------------------------
+// host-side launcher (called from engine.hip's launch dispatch)
+// ---------------------------------------------------------------------------
+bool fdb_fast_scan_supported(int func_id) {
+  switch (func_id) {
+    case FN_RATE: case FN_INCREASE: case FN_DELTA:
+    case FN_SUM: case FN_COUNT: case FN_AVG: case FN_RATE_OVER_DELTA:
+    case FN_MIN: case FN_MAX: case FN_STDDEV: case FN_STDVAR:
+    case FN_CHANGES: case FN_LAST: case FN_PRESENT: case FN_TIMESTAMP:
+    case FN_ZSCORE:
+      return true;
+    default:
+      return false;
+  }
+}
+
+int32_t fdb_launch_fast_scan(hipStream_t stream, const uint8_t* blob, DirSoA dir,
+                             const int32_t* series_first, const int32_t* series_nchunks,
+                             const int32_t* group_ids, const int32_t* series_by_group,
+                             int num_series,
+                             int64_t qstart, int64_t qstep, int64_t qwindow,
+                             int num_windows, int func_id, int agg_id, int emit_group,
+                             double* out, double* out_cnt, double* out_sq,
+                             int phase_mask) {
+  (void)phase_mask;
+  int grid = (num_series + FAST_WAVES - 1) / FAST_WAVES;
+  int cap = 8192;
+  if (const char* g = getenv("FDB_GRID")) cap = atoi(g);   // perf experiments
+  if (cap > 0 && grid > cap) grid = cap;
+  #define FARGS blob, dir, series_first, series_nchunks, group_ids, \
+      series_by_group, num_series, qstart, qstep, qwindow, num_windows, \
+      agg_id, out, out_cnt, out_sq
+  #define FCASE(F) case F: \
+    if (emit_group) hipLaunchKernelGGL((fast_scan_kernel<F, 1>), dim3(grid), \
+                                       dim3(FAST_WAVES * 64), 0, stream, FARGS); \
+    else hipLaunchKernelGGL((fast_scan_kernel<F, 0>), dim3(grid), \
+                            dim3(FAST_WAVES * 64), 0, stream, FARGS); \
+    break
+  switch (func_id) {
+    FCASE(FN_RATE); FCASE(FN_INCREASE); FCASE(FN_DELTA); FCASE(FN_SUM);
+    FCASE(FN_COUNT); FCASE(FN_AVG); FCASE(FN_RATE_OVER_DELTA);
+    FCASE(FN_MIN); FCASE(FN_MAX); FCASE(FN_STDDEV); FCASE(FN_STDVAR);
+    FCASE(FN_CHANGES); FCASE(FN_LAST); FCASE(FN_PRESENT); FCASE(FN_TIMESTAMP);
+    FCASE(FN_ZSCORE);
+    default:
+      fdb_set_error("fast scan: unsupported func_id %d", func_id);
+      return FDB_ERR_BADARG;
+  }
+  #undef FCASE
+  #undef FARGS
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) {
+    fdb_set_error("fast_scan_kernel launch failed: %s", hipGetErrorString(e));
+    return FDB_ERR;
+  }
+  return FDB_OK;
+}
