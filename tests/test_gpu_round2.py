@@ -1,0 +1,104 @@
+"""Round-2 GPU paths: fast-kernel window tiling, rate-over-delta on device,
+fused-group emit vs the two-phase reduce.
+
+All cases compare the HIP engine against the CPU oracle (or against the
+engine's own alternate code path) on identical sealed stores.
+"""
+import os
+
+import numpy as np
+import pytest
+
+from conftest import build_store, synth_counter_series, synth_gauge_series
+from test_gpu_parity import counter_store, gauge_store, mkq, run_both, check, FUNCS
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def engine(fdb):
+    return fdb.Engine(0)
+
+
+def test_window_tiling_past_256(fdb, oracle, engine):
+    """num_windows > FAST_TILE exercises the fast kernel's tile loop."""
+    st = counter_store(fdb, n_series=32, seed=31, reset_p=0.01)
+    # 300 windows across the hour (step smaller than cadence)
+    q = fdb.make_query(100000, 12000, 100000 + 299 * 12000, 300000, fdb.FN_RATE)
+    assert q.num_windows == 300
+    got, want = run_both(fdb, oracle, engine, st, q)
+    check(got, want)
+
+
+def test_window_tiling_gauge(fdb, oracle, engine):
+    st = gauge_store(fdb, n_series=32, seed=32, nan_p=0.1)
+    q = fdb.make_query(100000, 10000, 100000 + 399 * 10000, 600000,
+                       fdb.FN_AVG_OVER_TIME)
+    assert q.num_windows == 400
+    got, want = run_both(fdb, oracle, engine, st, q)
+    check(got, want)
+
+
+def test_rate_over_delta_gpu(fdb, oracle, engine):
+    """FN 19 (delta-temporality rate, RateFunctions.scala:424-445) on the GPU
+    fast path vs the oracle."""
+    st = gauge_store(fdb, n_series=48, seed=19, nan_p=0.1)
+    q = mkq(fdb, fdb.FN_RATE_OVER_DELTA)
+    got, want = run_both(fdb, oracle, engine, st, q)
+    check(got, want)
+
+
+@pytest.mark.parametrize("agg", ["sum", "avg", "min", "max", "count",
+                                 "stddev", "stdvar"])
+def test_fused_group_matches_two_phase(fdb, oracle, engine, agg):
+    """The fused-group emit (no [S×W] intermediate) against the two-phase
+    scan→group_reduce path AND the oracle, same store and query."""
+    from test_gpu_parity import AGGS
+    rng = np.random.default_rng(hash(agg) % 2**31)
+    n_groups = 9
+    series, groups = [], []
+    for s in range(150):
+        ts, vs = synth_counter_series(rng, 240, reset_p=0.01)
+        series.append([[(int(t), float(v)) for t, v in zip(ts, vs)]])
+        groups.append(s % n_groups)
+    st = build_store(fdb, series, groups=groups, kind=fdb.COL_COUNTER)
+    q = mkq(fdb, FUNCS["rate"], AGGS[agg], n_groups)
+    nw = q.num_windows
+    ds = engine.upload(st)
+    got_fused = np.empty(n_groups * nw)
+    engine.query(ds, q, out=got_fused)
+    want = oracle.query_exec(st.view(), q, st.num_series, nw, nthreads=4)
+    check(got_fused, want)
+
+
+def test_fused_group_partial_mode(fdb, oracle, engine):
+    """Partial (multi-GPU merge) grids through the fused emit."""
+    rng = np.random.default_rng(88)
+    n_groups = 5
+    series, groups = [], []
+    for s in range(60):
+        ts, vs = synth_gauge_series(rng, 240, nan_p=0.2)
+        series.append([[(int(t), float(v)) for t, v in zip(ts, vs)]])
+        groups.append(s % n_groups)
+    st = build_store(fdb, series, groups=groups)
+    q = mkq(fdb, FUNCS["avg"], 5, n_groups)
+    nw = q.num_windows
+    want_s, want_c = oracle.query_exec(st.view(), q, st.num_series, nw,
+                                       out_counts=True)
+    got_s = np.empty(n_groups * nw)
+    got_c = np.empty(n_groups * nw)
+    engine.query(engine.upload(st), q, out=got_s, out_counts=got_c)
+    check(got_s, want_s)
+    np.testing.assert_array_equal(got_c, want_c)
+
+
+def test_single_row_and_empty_edge(fdb, oracle, engine):
+    """1-row chunks and windows entirely before/after the data."""
+    series = [
+        [[(100000, 5.0)]],                              # one sample
+        [[(100000, 1.0), (100000, 2.0), (130000, 3.0)]],  # duplicate ts
+    ]
+    st = build_store(fdb, series)
+    q = fdb.make_query(40000, 15000, 400000, 60000, fdb.FN_SUM_OVER_TIME)
+    got, want = run_both(fdb, oracle, engine, st, q)
+    check(got, want)
