@@ -30,7 +30,7 @@ def oracle():
     return pyclient
 
 
-def build_store(fdb, series, kind=None, max_rows=200, groups=None):
+def build_store(fdb, series, kind=None, max_rows=400, groups=None):
     """series: list of list-of-(chunk tuples) — each series is a list of chunks,
     each chunk a list of (ts, value) pairs; chunk boundaries forced explicitly
     (mirrors RawDataWindowingSpec.timeValueRVPk + addChunkToRV,
