@@ -236,31 +236,11 @@ __device__ inline void d_decode_ts_offsets(const DVec& v, int n, int64_t ts0,
     tout[i] = (int32_t)(base + (int64_t)v.slope * i + d_inner_at(&v, i));
 }
 
-// extrapolatedRate (RateFunctions.scala:72-111), computed in the ms domain
-// (the reference's /1000 unit conversions cancel — every comparison and the
-// final ext/sampledInterval ratio are scale-invariant; ≤2 ulp vs the
-// seconds-domain oracle, inside the 1e-9 north_star tolerance).
-//   inv_ns1    = 1/(numSamples-1)  (from the per-block reciprocal table)
-//   rate_scale = 1000.0/(windowEnd-windowStart) precomputed on the host
-__device__ __forceinline__ double d_extrap_rate_ms(
-    double dts /*t1-wStart*/, double dte /*wEnd-t2*/, double si /*t2-t1, >0*/,
-    double inv_ns1, double v1, double v2,
-    bool isCounter, bool isRate, double rate_scale) {
-  double inv_si = 1.0 / si;
-  double avg = si * inv_ns1;
-  double delta = v2 - v1;
-  if (isCounter && delta > 0 && v1 >= 0) {
-    double dtz = si * (v1 / delta);      // durationToZero
-    dts = dtz < dts ? dtz : dts;
-  }
-  double thresh = avg * 1.1;
-  double ext = si + (dts < thresh ? dts : avg * 0.5)
-                  + (dte < thresh ? dte : avg * 0.5);
-  double scaled = delta * (ext * inv_si);
-  return isRate ? scaled * rate_scale : scaled;
-}
-
-// original i64/seconds-domain form (kept for the general/streaming paths)
+// extrapolatedRate (RateFunctions.scala:72-111) — the oracle's EXACT
+// operation sequence. Do not reassociate or replace the divisions: the
+// durationToZero/threshold comparisons are discontinuous and integer counter
+// data makes exact rational ties (v1/delta == 1.1/(numSamples-1)) common, so
+// the branch taken must follow the reference's own FP rounding.
 __device__ inline double d_extrapolated_rate(int64_t windowStart, int64_t windowEnd,
                                              int numSamples,
                                              int64_t t1, double v1, int64_t t2, double v2,

@@ -76,8 +76,10 @@ struct FWs {                        // per-wave LDS workspace
   int16_t ew[FAST_TILE];            // per-window endRow (-1 = none)
 };
 
-template <int KIND> struct NeedsInv {   // reciprocal table users
-  static constexpr bool v = KIND == K_RATE || KIND == K_PFX || KIND == K_PFX_SQ;
+template <int KIND> struct NeedsInv {   // reciprocal table users (AVG/STDDEV;
+  // the rate epilogue keeps the oracle's exact divisions — its threshold
+  // comparisons are discontinuous and must round identically, see below)
+  static constexpr bool v = KIND == K_PFX || KIND == K_PFX_SQ;
 };
 
 // in-chunk correction at row i from the sparse drop table (the step function
@@ -102,12 +104,13 @@ __device__ __forceinline__ double f_corr_at(const WS& ws, int dcount, bool dense
   return c;
 }
 
-// min-waves/SIMD: PFX_SQ is LDS-bound at 3 blocks/CU; the group-emit variant
-// carries 12 accumulator registers (4 at min-waves 5 spills ~33 VGPRs); the
-// plain kinds fit 96 VGPRs = 5 waves/SIMD
-template <int FUNC, int EMIT>
-__global__ __launch_bounds__(FAST_WAVES * 64,
-                             (FKind<FUNC>::v == K_PFX_SQ) ? 3 : (EMIT ? 4 : 5))
+// MINW = min waves/SIMD the register allocator must meet. PFX_SQ is
+// LDS-bound at 3 blocks/CU; the group-emit variant carries 12 accumulator
+// registers (4 waves); K_RATE (no reciprocal table, LDS fits 6 blocks/CU) is
+// built at BOTH 5 (96 VGPR, no spills) and 6 (80 VGPR, ~19 spilled) — the
+// launcher picks via FDB_RATE_WAVES, measured on hardware.
+template <int FUNC, int EMIT, int MINW>
+__global__ __launch_bounds__(FAST_WAVES * 64, MINW)
 void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                       const int32_t* __restrict__ series_first,
                       const int32_t* __restrict__ series_nchunks,
@@ -360,7 +363,14 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 
         if constexpr (KIND == K_RATE) {
           // ChunkedRateFunctionBase + extrapolatedRate; e>s implies s valid,
-          // covers empty and single-sample windows (highestTime>lowestTime)
+          // covers empty and single-sample windows (highestTime>lowestTime).
+          // The epilogue keeps the oracle's EXACT operation sequence
+          // (RateFunctions.scala:72-111): its durationToZero/threshold
+          // comparisons are discontinuous — counters make exact rational
+          // ties like v1/delta == 1.1/(numSamples-1) common, and the branch
+          // taken then depends on the reference's own FP rounding. All
+          // inputs are ts0-relative; extrapolatedRate uses differences only,
+          // so offset-domain i64s give bit-identical results.
           if (wok && e > s) {
             const int t1 = ws.tso[s], t2 = ws.tso[e];
             if (t2 > t1) {
@@ -372,11 +382,9 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                 v2 += f_corr_at(ws, dcount, dense, n, e);
               }
               const int64_t wEndOff = (int64_t)w * qstep - Ae;   // wEnd - ts0
-              res = d_extrap_rate_ms((double)(t1 - (wEndOff - qwindow)),
-                                     (double)(wEndOff - t2),
-                                     (double)(t2 - t1),
-                                     inv_tab[e - s], v1, v2,
-                                     IS_COUNTER, FUNC == FN_RATE, rate_scale);
+              res = d_extrapolated_rate(wEndOff - qwindow, wEndOff, e - s + 1,
+                                        t1, v1, t2, v2,
+                                        IS_COUNTER, FUNC == FN_RATE);
             }
           }
         } else if constexpr (KIND == K_PFX) {
@@ -516,26 +524,35 @@ int32_t fdb_launch_fast_scan(hipStream_t stream, const uint8_t* blob, DirSoA dir
   int cap = 8192;
   if (const char* g = getenv("FDB_GRID")) cap = atoi(g);   // perf experiments
   if (cap > 0 && grid > cap) grid = cap;
+  static int rate_w = 0;
+  if (!rate_w) {
+    const char* v = getenv("FDB_RATE_WAVES");   // occupancy experiment knob
+    rate_w = (v && atoi(v) == 5) ? 5 : 6;
+  }
   #define FARGS blob, dir, series_first, series_nchunks, group_ids, \
       series_by_group, num_series, qstart, qstep, qwindow, num_windows, \
       agg_id, out, out_cnt, out_sq
-  #define FCASE(F) case F: \
-    if (emit_group) hipLaunchKernelGGL((fast_scan_kernel<F, 1>), dim3(grid), \
-                                       dim3(FAST_WAVES * 64), 0, stream, FARGS); \
-    else hipLaunchKernelGGL((fast_scan_kernel<F, 0>), dim3(grid), \
-                            dim3(FAST_WAVES * 64), 0, stream, FARGS); \
-    break
+  #define LAUNCH(F, E, W) hipLaunchKernelGGL((fast_scan_kernel<F, E, W>), \
+      dim3(grid), dim3(FAST_WAVES * 64), 0, stream, FARGS)
+  #define FCASE(F, W) case F: \
+    if (emit_group) LAUNCH(F, 1, 4); else LAUNCH(F, 0, W); break
   switch (func_id) {
-    FCASE(FN_RATE); FCASE(FN_INCREASE); FCASE(FN_DELTA); FCASE(FN_SUM);
-    FCASE(FN_COUNT); FCASE(FN_AVG); FCASE(FN_RATE_OVER_DELTA);
-    FCASE(FN_MIN); FCASE(FN_MAX); FCASE(FN_STDDEV); FCASE(FN_STDVAR);
-    FCASE(FN_CHANGES); FCASE(FN_LAST); FCASE(FN_PRESENT); FCASE(FN_TIMESTAMP);
-    FCASE(FN_ZSCORE);
+    case FN_RATE:
+      if (emit_group) LAUNCH(FN_RATE, 1, 4);
+      else if (rate_w == 5) LAUNCH(FN_RATE, 0, 5);
+      else LAUNCH(FN_RATE, 0, 6);
+      break;
+    FCASE(FN_INCREASE, 6); FCASE(FN_DELTA, 6); FCASE(FN_SUM, 5);
+    FCASE(FN_COUNT, 5); FCASE(FN_AVG, 5); FCASE(FN_RATE_OVER_DELTA, 5);
+    FCASE(FN_MIN, 5); FCASE(FN_MAX, 5); FCASE(FN_STDDEV, 3); FCASE(FN_STDVAR, 3);
+    FCASE(FN_CHANGES, 5); FCASE(FN_LAST, 5); FCASE(FN_PRESENT, 5);
+    FCASE(FN_TIMESTAMP, 5); FCASE(FN_ZSCORE, 5);
     default:
       fdb_set_error("fast scan: unsupported func_id %d", func_id);
       return FDB_ERR_BADARG;
   }
   #undef FCASE
+  #undef LAUNCH
   #undef FARGS
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) {

