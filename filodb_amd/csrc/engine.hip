@@ -1456,12 +1456,9 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
 // launch dispatch over the func template parameter
 // the fast single-chunk kernel handles this (dataset, query) pair?
 static bool fast_eligible(const fdb_dataset_t* d, const fdb_query_t* q) {
-  static int enabled = -1;
-  if (enabled < 0) {
-    const char* v = getenv("FDB_FAST");           // perf/parity experiments
-    enabled = (v && atoi(v) == 0) ? 0 : 1;
-  }
-  return enabled && d->fast_ok && fdb_fast_scan_supported(q->func_id) &&
+  const char* v = getenv("FDB_FAST");             // perf/parity experiments
+  if (v && atoi(v) == 0) return false;
+  return d->fast_ok && fdb_fast_scan_supported(q->func_id) &&
          q->step > 0 && q->step < ((int64_t)1 << 31) && q->window >= 0;
 }
 
@@ -1589,12 +1586,8 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   // never materializes the [S×W] grid; other shapes run two-phase — the scan
   // fills an internal per-series grid with plain stores, then a presenter
   // (topk_kernel / group_reduce_kernel) folds it along the group-sorted index
-  static int fused_enabled = -1;
-  if (fused_enabled < 0) {
-    const char* v = getenv("FDB_FUSED_GROUP");   // perf/parity experiments
-    fused_enabled = (v && atoi(v) == 0) ? 0 : 1;
-  }
-  const bool fused = fused_enabled && q->agg_id != AGG_NONE && !is_topk &&
+  const char* fe = getenv("FDB_FUSED_GROUP");    // perf/parity experiments
+  const bool fused = !(fe && atoi(fe) == 0) && q->agg_id != AGG_NONE && !is_topk &&
                      nw <= 256 && fast_eligible(d, q);
   fdb_query_t qscan = *q;
   if (q->agg_id != AGG_NONE && !fused) {

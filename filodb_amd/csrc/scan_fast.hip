@@ -12,10 +12,9 @@
 //    semantics: WindowedChunkIterator, ChunkSetInfo.scala:467-511)
 //  * timestamps live in LDS as i32 offsets from the chunk's first ts (halves
 //    the LDS footprint and read traffic; chunk span is checked at upload)
-//  * the extrapolatedRate epilogue runs in the ms domain with a per-block
-//    reciprocal table for 1/(numSamples-1): 2 f64 divisions per window
-//    instead of 6-7 (scan_common.h d_extrap_rate_ms; ≤2 ulp vs the oracle,
-//    inside the 1e-9 north_star tolerance)
+//  * the extrapolatedRate epilogue keeps the oracle's exact operation
+//    sequence (its threshold comparisons are discontinuous — see the window
+//    phase); AVG/STDDEV use a per-block reciprocal table (continuous, ≤2 ulp)
 //  * window results are branch-lean selects; one store per window
 //
 // Per-window semantics are IDENTICAL to the round-1 kernel's single-chunk
@@ -524,11 +523,8 @@ int32_t fdb_launch_fast_scan(hipStream_t stream, const uint8_t* blob, DirSoA dir
   int cap = 8192;
   if (const char* g = getenv("FDB_GRID")) cap = atoi(g);   // perf experiments
   if (cap > 0 && grid > cap) grid = cap;
-  static int rate_w = 0;
-  if (!rate_w) {
-    const char* v = getenv("FDB_RATE_WAVES");   // occupancy experiment knob
-    rate_w = (v && atoi(v) == 5) ? 5 : 6;
-  }
+  const char* rw = getenv("FDB_RATE_WAVES");   // occupancy experiment knob
+  const int rate_w = (rw && atoi(rw) == 5) ? 5 : 6;
   #define FARGS blob, dir, series_first, series_nchunks, group_ids, \
       series_by_group, num_series, qstart, qstep, qwindow, num_windows, \
       agg_id, out, out_cnt, out_sq

@@ -1,0 +1,67 @@
+#!/usr/bin/env python3
+"""Perf sweep on one resident dataset: builds the headline (configs[1]) store
+once, then times kernel variants toggled by env knobs — avoids paying the
+host-side synth build per variant. Run on a GPU box via gpurun."""
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+
+import filodb_amd as fdb  # noqa: E402
+
+T0 = 100000
+
+
+def build(n_series=1_000_000, kind=None):
+    st = fdb.ChunkStore()
+    st.synth_generate(kind if kind is not None else fdb.COL_COUNTER,
+                      n_series, 240, start_ts=T0, step_ms=15000, jitter_ms=300,
+                      lam=10.0, reset_p=0.001, n_groups=1000, seed=42)
+    st.seal()
+    return st
+
+
+def bench(eng, ds, q, out, cnt=None, warmup=2, iters=8):
+    return eng.bench(ds, q, out, out_counts=cnt, on_device=True,
+                     warmup=warmup, iters=iters)
+
+
+def main():
+    import torch
+    n = int(os.environ.get("SWEEP_SERIES", "1000000"))
+    st = build(n)
+    eng = fdb.Engine(0)
+    ds = eng.upload(st)
+    span = 240 * 15000
+    q_rate = fdb.make_query(T0, 15000, T0 + span, 300000, fdb.FN_RATE)
+    q_sum = fdb.make_query(T0, 15000, T0 + span, 300000, fdb.FN_RATE,
+                           fdb.AGG_SUM, 1000)
+    nw = q_rate.num_windows
+    out = torch.empty(n * nw, dtype=torch.float64, device="cuda:0")
+    gout = torch.empty(1000 * nw, dtype=torch.float64, device="cuda:0")
+    gcnt = torch.empty_like(gout)
+
+    for waves in ("6", "5"):
+        os.environ["FDB_RATE_WAVES"] = waves
+        ms = bench(eng, ds, q_rate, out)
+        print(f"rate waves={waves}: {ms:.3f} ms", flush=True)
+    for fused in ("1", "0"):
+        os.environ["FDB_FUSED_GROUP"] = fused
+        ms = bench(eng, ds, q_sum, gout, gcnt)
+        print(f"sum-by-group fused={fused}: {ms:.3f} ms", flush=True)
+    # gauge workload on the same box (separate store: raw f64 values)
+    del ds
+    st2 = build(n, kind=fdb.COL_GAUGE)
+    ds2 = eng.upload(st2)
+    q_avg = fdb.make_query(T0, 15000, T0 + span, 600000, fdb.FN_AVG_OVER_TIME)
+    ms = bench(eng, ds2, q_avg, out)
+    print(f"avg_over_time[10m]: {ms:.3f} ms", flush=True)
+    q_min = fdb.make_query(T0, 15000, T0 + span, 600000, fdb.FN_MIN_OVER_TIME)
+    ms = bench(eng, ds2, q_min, out)
+    print(f"min_over_time[10m]: {ms:.3f} ms", flush=True)
+
+
+if __name__ == "__main__":
+    main()
