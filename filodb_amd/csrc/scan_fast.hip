@@ -120,7 +120,8 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                       int num_windows, int agg_id,
                       double* __restrict__ out,
                       double* __restrict__ out_cnt,
-                      double* __restrict__ out_sq)
+                      double* __restrict__ out_sq,
+                      int pm)    // bit 3: s_memtime phase split into out[]
 {
   constexpr int KIND = FKind<FUNC>::v;
   constexpr bool IS_COUNTER = (FUNC == FN_RATE || FUNC == FN_INCREASE);
@@ -139,6 +140,8 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 
   const double inv_step = 1.0 / (double)qstep;
   const double rate_scale = 1000.0 / (double)qwindow;
+  const bool timing = (pm & 8) != 0;    // perf ablation (clobbers out[])
+  uint64_t tD = 0, tM = 0, tI = 0, tW = 0, tt = 0;
 
   // series iteration: grid-stride for the plain grid emit; a contiguous slab
   // of the group-sorted index for the fused-group emit (so one wave sees each
@@ -183,6 +186,7 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   };
 
   for (int pos = pos0; pos < pos1; pos += pos_step) {
+    if (timing) tt = __builtin_amdgcn_s_memtime();
     const int sid = (EMIT == 1) ? sbg[pos] : pos;
     const int first = series_first[sid];
     const int nch = series_nchunks[sid];
@@ -206,6 +210,7 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     }
     d_wait_lds();
     __builtin_amdgcn_wave_barrier();
+    if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); tD += t - tt; tt = t; }
 
     // ---- meta: kind-specific per-row structures ----------------------------
     int dcount = 0;
@@ -295,6 +300,7 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     }
     d_wait_lds();
     __builtin_amdgcn_wave_barrier();
+    if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); tM += t - tt; tt = t; }
 
     // window time base: wEnd(w) = qstart + w*qstep; offsets are vs ts0
     const int64_t Ae = ts0 - qstart;            // wEnd >= ts  ⟺ w*qstep >= o+Ae
@@ -349,8 +355,124 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       }
       d_wait_lds();
       __builtin_amdgcn_wave_barrier();
+      if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); tI += t - tt; tt = t; }
 
-      // ---- window phase: 4 windows per lane, branch-lean -------------------
+      // ---- window phase: 4 windows per lane --------------------------------
+      // results land through emit_res: plain grid store or fastReduce
+      // register-accumulate (NaN rows skipped, RowAggregator semantics)
+      auto emit_res = [&](int k, int w, bool wok, double res) {
+        if (EMIT == 0) {
+          // timing mode suppresses result stores so the per-wave cycle
+          // records at out[0..nwaves*4) survive (perf ablation only)
+          if (wok && w < num_windows && !timing)
+            out[(size_t)sid * num_windows + w] = res;
+        } else if (!isnan(res)) {
+          switch (agg_id) {
+            case AGG_MIN:
+              accS[k] = (isnan(accS[k]) || res < accS[k]) ? res : accS[k];
+              accC[k] += 1.0; break;
+            case AGG_MAX:
+              accS[k] = (isnan(accS[k]) || res > accS[k]) ? res : accS[k];
+              accC[k] += 1.0; break;
+            case AGG_STDDEV: case AGG_STDVAR:
+              accQ[k] += res * res;
+              accS[k] = isnan(accS[k]) ? res : accS[k] + res;
+              accC[k] += 1.0; break;
+            case AGG_COUNT: case AGG_GROUP:
+              accC[k] += 1.0; break;
+            default:   // AGG_SUM / AGG_AVG
+              accS[k] = isnan(accS[k]) ? res : accS[k] + res;
+              accC[k] += 1.0; break;
+          }
+        }
+      };
+
+      if constexpr (KIND == K_RATE) {
+        // ChunkedRateFunctionBase + extrapolatedRate; e>s implies s valid,
+        // covers empty and single-sample windows (highestTime>lowestTime).
+        // The epilogue keeps the oracle's EXACT operation sequence
+        // (RateFunctions.scala:72-111): its durationToZero/threshold
+        // comparisons are discontinuous — counters make exact rational ties
+        // like v1/delta == 1.1/(numSamples-1) common, and the branch taken
+        // then depends on the reference's own FP rounding. All inputs are
+        // ts0-relative; extrapolatedRate uses differences only, so
+        // offset-domain i64s give bit-identical results.
+        // Staged: all 4 windows' boundary loads issue before any epilogue
+        // math so the LDS latency and the f64 division chains overlap.
+        int sA[4], eA[4], t1A[4], t2A[4];
+        double v1A[4], v2A[4];
+        #pragma unroll
+        for (int k = 0; k < 4; k++) {
+          const int wi = lane + 64 * k;
+          const bool wok = wi < tn;
+          sA[k] = wok ? ws.sw[wi] : 1;
+          eA[k] = wok ? ws.ew[wi] : -1;
+        }
+        #pragma unroll
+        for (int k = 0; k < 4; k++) {
+          const bool v = eA[k] > sA[k];
+          const int sc = v ? sA[k] : 0, ec = v ? eA[k] : 0;
+          t1A[k] = ws.tso[sc]; t2A[k] = ws.tso[ec];
+          v1A[k] = ws.val[sc]; v2A[k] = ws.val[ec];
+        }
+        #pragma unroll
+        for (int k = 0; k < 4; k++) {
+          const int wi = lane + 64 * k;
+          const int w = tb + wi;
+          const bool wok = wi < tn;
+          double res = NAN;
+          if (eA[k] > sA[k] && t2A[k] > t1A[k]) {
+            double v1 = v1A[k], v2 = v2A[k];
+            if (IS_COUNTER && dropped) {
+              if (isnan(v1)) v1 = 0;
+              if (isnan(v2)) v2 = 0;
+              v1 += f_corr_at(ws, dcount, dense, n, sA[k]);
+              v2 += f_corr_at(ws, dcount, dense, n, eA[k]);
+            }
+            const int64_t wEndOff = (int64_t)w * qstep - Ae;   // wEnd - ts0
+            res = d_extrapolated_rate(wEndOff - qwindow, wEndOff,
+                                      eA[k] - sA[k] + 1,
+                                      t1A[k], v1, t2A[k], v2,
+                                      IS_COUNTER, FUNC == FN_RATE);
+          }
+          emit_res(k, w, wok, res);
+        }
+      } else if constexpr (KIND == K_PFX) {
+        // staged prefix-diff loads, then the per-function epilogues
+        int sA[4], eA[4], pcA[4];
+        double psA[4];
+        #pragma unroll
+        for (int k = 0; k < 4; k++) {
+          const int wi = lane + 64 * k;
+          const bool wok = wi < tn;
+          sA[k] = wok ? ws.sw[wi] : 1;
+          eA[k] = wok ? ws.ew[wi] : -1;
+        }
+        #pragma unroll
+        for (int k = 0; k < 4; k++) {
+          const bool v = eA[k] >= sA[k] && eA[k] >= 0;
+          const int sc = v ? sA[k] : 1, ec = v ? eA[k] : 0;
+          psA[k] = ws.val[ec] - (sc ? ws.val[sc - 1] : 0.0);
+          pcA[k] = (int)ws.cnt[ec] - (sc ? (int)ws.cnt[sc - 1] : 0);
+        }
+        #pragma unroll
+        for (int k = 0; k < 4; k++) {
+          const int wi = lane + 64 * k;
+          const bool wok = wi < tn;
+          double res = NAN;
+          if (eA[k] >= sA[k] && eA[k] >= 0) {
+            const double ps = psA[k];
+            const int pc = pcA[k];
+            if constexpr (FUNC == FN_SUM) res = pc > 0 ? ps : NAN;
+            else if constexpr (FUNC == FN_COUNT) res = (double)pc;
+            else if constexpr (FUNC == FN_AVG)
+              res = pc > 0 ? ps * inv_tab[pc] : NAN;
+            else   // FN_RATE_OVER_DELTA (RateFunctions.scala:424-445)
+              res = (pc > 0 ? ps : NAN) * rate_scale;
+          }
+          emit_res(k, tb + wi, wok, res);
+        }
+      } else {
       #pragma unroll
       for (int k = 0; k < 4; k++) {
         const int wi = lane + 64 * k;
@@ -360,44 +482,7 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         const int e = wok ? ws.ew[wi] : -1;
         double res = NAN;
 
-        if constexpr (KIND == K_RATE) {
-          // ChunkedRateFunctionBase + extrapolatedRate; e>s implies s valid,
-          // covers empty and single-sample windows (highestTime>lowestTime).
-          // The epilogue keeps the oracle's EXACT operation sequence
-          // (RateFunctions.scala:72-111): its durationToZero/threshold
-          // comparisons are discontinuous — counters make exact rational
-          // ties like v1/delta == 1.1/(numSamples-1) common, and the branch
-          // taken then depends on the reference's own FP rounding. All
-          // inputs are ts0-relative; extrapolatedRate uses differences only,
-          // so offset-domain i64s give bit-identical results.
-          if (wok && e > s) {
-            const int t1 = ws.tso[s], t2 = ws.tso[e];
-            if (t2 > t1) {
-              double v1 = ws.val[s], v2 = ws.val[e];
-              if (IS_COUNTER && dropped) {
-                if (isnan(v1)) v1 = 0;
-                if (isnan(v2)) v2 = 0;
-                v1 += f_corr_at(ws, dcount, dense, n, s);
-                v2 += f_corr_at(ws, dcount, dense, n, e);
-              }
-              const int64_t wEndOff = (int64_t)w * qstep - Ae;   // wEnd - ts0
-              res = d_extrapolated_rate(wEndOff - qwindow, wEndOff, e - s + 1,
-                                        t1, v1, t2, v2,
-                                        IS_COUNTER, FUNC == FN_RATE);
-            }
-          }
-        } else if constexpr (KIND == K_PFX) {
-          if (wok && e >= s && e >= 0) {
-            double ps = ws.val[e] - (s ? ws.val[s - 1] : 0.0);
-            int pc = (int)ws.cnt[e] - (s ? (int)ws.cnt[s - 1] : 0);
-            if constexpr (FUNC == FN_SUM) res = pc > 0 ? ps : NAN;
-            else if constexpr (FUNC == FN_COUNT) res = (double)pc;
-            else if constexpr (FUNC == FN_AVG)
-              res = pc > 0 ? ps * inv_tab[pc] : NAN;
-            else   // FN_RATE_OVER_DELTA (RateFunctions.scala:424-445)
-              res = (pc > 0 ? ps : NAN) * rate_scale;
-          }
-        } else if constexpr (KIND == K_PFX_SQ) {
+        if constexpr (KIND == K_PFX_SQ) {
           if (wok && e >= s && e >= 0) {
             double ps = ws.val[e] - (s ? ws.val[s - 1] : 0.0);
             int pc = (int)ws.cnt[e] - (s ? (int)ws.cnt[s - 1] : 0);
@@ -463,35 +548,22 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
           }
         }
 
-        if (EMIT == 0) {
-          if (wok && w < num_windows)
-            out[(size_t)sid * num_windows + w] = res;
-        } else if (!isnan(res)) {
-          // fastReduce map step: NaN rows skipped (RowAggregator semantics)
-          switch (agg_id) {
-            case AGG_MIN:
-              accS[k] = (isnan(accS[k]) || res < accS[k]) ? res : accS[k];
-              accC[k] += 1.0; break;
-            case AGG_MAX:
-              accS[k] = (isnan(accS[k]) || res > accS[k]) ? res : accS[k];
-              accC[k] += 1.0; break;
-            case AGG_STDDEV: case AGG_STDVAR:
-              accQ[k] += res * res;   // fall through to sum+count
-              accS[k] = isnan(accS[k]) ? res : accS[k] + res;
-              accC[k] += 1.0; break;
-            case AGG_COUNT: case AGG_GROUP:
-              accC[k] += 1.0; break;
-            default:   // AGG_SUM / AGG_AVG
-              accS[k] = isnan(accS[k]) ? res : accS[k] + res;
-              accC[k] += 1.0; break;
-          }
-        }
+        emit_res(k, w, wok, res);
       }
+      }  // generic-kind loop
       d_wait_lds();
       __builtin_amdgcn_wave_barrier();
+      if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); tW += t - tt; tt = t; }
     }  // tiles
   }  // series
   if (EMIT == 1 && cur_grp >= 0) flush_group(cur_grp);
+  if (timing && EMIT == 0 && lane == 0) {
+    size_t gw = (size_t)blockIdx.x * FAST_WAVES + wave;
+    out[gw * 4 + 0] = (double)tD;
+    out[gw * 4 + 1] = (double)tM;
+    out[gw * 4 + 2] = (double)tI;
+    out[gw * 4 + 3] = (double)tW;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -527,7 +599,7 @@ int32_t fdb_launch_fast_scan(hipStream_t stream, const uint8_t* blob, DirSoA dir
   const int rate_w = (rw && atoi(rw) == 5) ? 5 : 6;
   #define FARGS blob, dir, series_first, series_nchunks, group_ids, \
       series_by_group, num_series, qstart, qstep, qwindow, num_windows, \
-      agg_id, out, out_cnt, out_sq
+      agg_id, out, out_cnt, out_sq, phase_mask
   #define LAUNCH(F, E, W) hipLaunchKernelGGL((fast_scan_kernel<F, E, W>), \
       dim3(grid), dim3(FAST_WAVES * 64), 0, stream, FARGS)
   #define FCASE(F, W) case F: \

@@ -51,6 +51,20 @@ def main():
         os.environ["FDB_FUSED_GROUP"] = fused
         ms = bench(eng, ds, q_sum, gout, gcnt)
         print(f"sum-by-group fused={fused}: {ms:.3f} ms", flush=True)
+    # phase split (s_memtime; pm bit 3 clobbers out[] with per-wave cycles)
+    os.environ["FDB_RATE_WAVES"] = "6"
+    q_rate._pad = 8
+    eng.query(ds, q_rate, out=out, on_device=True)
+    eng.synchronize()
+    nwaves = 8192 * 4
+    ph = out[:nwaves * 4].reshape(nwaves, 4).cpu().numpy()
+    tot = ph.sum(axis=0)
+    names = ["decode", "meta", "inversion", "windows"]
+    print("rate phase split:",
+          " ".join(f"{nm}={v / tot.sum() * 100:.0f}%" for nm, v in zip(names, tot)),
+          f"(sum {tot.sum() / 1e9:.2f} Gcyc over {nwaves} waves)", flush=True)
+    q_rate._pad = 0
+
     # gauge workload on the same box (separate store: raw f64 values)
     del ds
     st2 = build(n, kind=fdb.COL_GAUGE)
