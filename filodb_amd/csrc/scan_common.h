@@ -99,6 +99,55 @@ __device__ __forceinline__ double d_dv_at(const DVec* v, int i) {
   return (double)d_lv_at(v, i);
 }
 
+// one-load header open: vectors are 64-B aligned in the device blob (padded
+// tail), so a single 32-byte read covers every header field of all three
+// scalar layouts — replaces d_vec_open's ~6 dependent global loads with one
+// round trip. first_val (optional) returns element 0 (ts0 for a timestamp
+// vector) parsed from the same 32 bytes.
+__device__ __forceinline__ void d_vec_open_wide(const uint8_t* p, DVec* v,
+                                                int64_t* first_val) {
+  uint32_t h[8];
+  memcpy(h, p, 32);
+  const uint16_t wf = (uint16_t)(h[1] & 0xffff);
+  v->wf = wf;
+  v->dropped = ((h[1] >> 16) & FDB_DROP_MASK) != 0;
+  if (wf == FDB_WF_DDV) {
+    v->init = (int64_t)(((uint64_t)h[3] << 32) | h[2]);
+    v->slope = (int32_t)h[4];
+    const uint32_t w6 = h[6];                  // inner bytes 4..7
+    v->nbits = (uint8_t)((w6 >> 16) & FDB_NBITS_MASK);
+    v->sign = ((w6 >> 16) & FDB_SIGN_MASK) != 0;
+    const int bitShift = (int)(w6 >> 24) & 0x3f;
+    v->idata = p + FDB_DDV_OFF_INNER + FDB_PRIM_OFF_DATA;
+    const int numBytes = (int)h[5];            // inner length word
+    v->n = ((numBytes - 4) * 8 + (bitShift != 0 ? bitShift - 8 : 0)) / v->nbits;
+    if (first_val) {
+      int64_t in0;                             // inner element 0 is in h[7]
+      switch (v->nbits) {
+        case 32: in0 = (int32_t)h[7]; break;
+        case 16: { int32_t x = (int16_t)(h[7] & 0xffff);
+                   in0 = v->sign ? x : (x & 0xffff); } break;
+        case 8:  { int32_t x = (int8_t)(h[7] & 0xff);
+                   in0 = v->sign ? x : (x & 0xff); } break;
+        case 4:  in0 = h[7] & 0x0f; break;
+        default: in0 = h[7] & 0x03; break;
+      }
+      *first_val = v->init + in0;
+    }
+  } else if (wf == FDB_WF_DDV_CONST) {
+    v->n = (int32_t)h[2];
+    v->init = (int64_t)(((uint64_t)h[4] << 32) | h[3]);
+    v->slope = (int32_t)h[5];
+    v->idata = nullptr; v->nbits = 0; v->sign = 0;
+    if (first_val) *first_val = v->init;
+  } else {
+    v->n = ((int)h[0] - 4) / 8;
+    v->idata = p + FDB_PRIM_OFF_DATA;
+    v->init = 0; v->slope = 0; v->nbits = 64; v->sign = 1;
+    if (first_val) *first_val = (int64_t)(((uint64_t)h[3] << 32) | h[2]);
+  }
+}
+
 // wave-wide inclusive prefix sums (64 lanes)
 __device__ __forceinline__ double wave_incl_scan(double x, int lane) {
   for (int off = 1; off < 64; off <<= 1) {

@@ -185,11 +185,35 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     }
   };
 
+  // two-level software pipeline over the series loop: the dependent chain
+  // series_first[sid] → dir offsets → 32-B vector headers is 3 serial HBM
+  // round trips per series; prefetching each level a stage early hides them
+  // under the previous series' decode/window work (d_wait_lds is lgkm-only,
+  // so prefetch loads stay in flight across the phase fences)
+  int pf_sid = 0, pf_first = 0, pf_nch = 0, pf_nr = 0;
+  uint64_t pf_toff = 0, pf_voff = 0;
+  if (pos0 < pos1) {
+    pf_sid = (EMIT == 1) ? sbg[pos0] : pos0;
+    pf_first = series_first[pf_sid];
+    pf_nch = series_nchunks[pf_sid];
+    pf_toff = dir.ts_off[pf_first];
+    pf_voff = dir.val_off[pf_first];
+    pf_nr = dir.num_rows[pf_first];
+  }
   for (int pos = pos0; pos < pos1; pos += pos_step) {
     if (timing) tt = __builtin_amdgcn_s_memtime();
-    const int sid = (EMIT == 1) ? sbg[pos] : pos;
-    const int first = series_first[sid];
-    const int nch = series_nchunks[sid];
+    const int sid = pf_sid;
+    const int nch = pf_nch;
+    const uint64_t toff = pf_toff, voff = pf_voff;
+    const int nrows = pf_nr;
+    // prefetch stage 1: next series' directory row ids
+    const int npos = pos + pos_step;
+    int nsid = 0, nfirst = 0, nnch = 0;
+    if (npos < pos1) {
+      nsid = (EMIT == 1) ? sbg[npos] : npos;
+      nfirst = series_first[nsid];
+      nnch = series_nchunks[nsid];
+    }
 
     // ---- decode: the single chunk into LDS (i32 ts offsets + f64 values) ---
     int n = 0;
@@ -197,16 +221,23 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     bool dropped = false;
     if (nch >= 1) {
       DVec tv, vv;
-      d_vec_open(blob + dir.ts_off[first], &tv);
-      d_vec_open(blob + dir.val_off[first], &vv);
-      n = dir.num_rows[first];
-      if (n > FAST_ROWS || n > tv.n) n = 0;     // guarded at upload
+      d_vec_open_wide(blob + toff, &tv, &ts0);   // one 32-B load each
+      d_vec_open_wide(blob + voff, &vv, nullptr);
+      n = nrows;
+      if (n > FAST_ROWS || n > tv.n) { n = 0; ts0 = 0; }   // guarded at upload
       if (n > 0) {
-        ts0 = d_lv_at(&tv, 0);                  // wave-uniform
         d_decode_ts_offsets(tv, n, ts0, ws.tso, lane);
         d_decode_chunk<true>(vv, n, nullptr, ws.val, lane);
         dropped = vv.dropped;
       }
+    }
+    // prefetch stage 2: next series' chunk offsets (nfirst landed during the
+    // decode above; these land during this series' meta/window phases)
+    if (npos < pos1) {
+      pf_sid = nsid; pf_first = nfirst; pf_nch = nnch;
+      pf_toff = dir.ts_off[nfirst];
+      pf_voff = dir.val_off[nfirst];
+      pf_nr = dir.num_rows[nfirst];
     }
     d_wait_lds();
     __builtin_amdgcn_wave_barrier();
@@ -397,72 +428,45 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         // then depends on the reference's own FP rounding. All inputs are
         // ts0-relative; extrapolatedRate uses differences only, so
         // offset-domain i64s give bit-identical results.
-        // Staged: all 4 windows' boundary loads issue before any epilogue
-        // math so the LDS latency and the f64 division chains overlap.
-        int sA[4], eA[4], t1A[4], t2A[4];
-        double v1A[4], v2A[4];
-        #pragma unroll
-        for (int k = 0; k < 4; k++) {
-          const int wi = lane + 64 * k;
-          const bool wok = wi < tn;
-          sA[k] = wok ? ws.sw[wi] : 1;
-          eA[k] = wok ? ws.ew[wi] : -1;
-        }
-        #pragma unroll
-        for (int k = 0; k < 4; k++) {
-          const bool v = eA[k] > sA[k];
-          const int sc = v ? sA[k] : 0, ec = v ? eA[k] : 0;
-          t1A[k] = ws.tso[sc]; t2A[k] = ws.tso[ec];
-          v1A[k] = ws.val[sc]; v2A[k] = ws.val[ec];
-        }
+        // (an explicitly staged variant — all 4 windows' loads before any
+        // math — measured SLOWER: 2.84 -> 3.26 ms, register pressure)
         #pragma unroll
         for (int k = 0; k < 4; k++) {
           const int wi = lane + 64 * k;
           const int w = tb + wi;
           const bool wok = wi < tn;
+          const int s = wok ? ws.sw[wi] : 1;
+          const int e = wok ? ws.ew[wi] : -1;
           double res = NAN;
-          if (eA[k] > sA[k] && t2A[k] > t1A[k]) {
-            double v1 = v1A[k], v2 = v2A[k];
-            if (IS_COUNTER && dropped) {
-              if (isnan(v1)) v1 = 0;
-              if (isnan(v2)) v2 = 0;
-              v1 += f_corr_at(ws, dcount, dense, n, sA[k]);
-              v2 += f_corr_at(ws, dcount, dense, n, eA[k]);
+          if (e > s) {
+            const int t1 = ws.tso[s], t2 = ws.tso[e];
+            if (t2 > t1) {
+              double v1 = ws.val[s], v2 = ws.val[e];
+              if (IS_COUNTER && dropped) {
+                if (isnan(v1)) v1 = 0;
+                if (isnan(v2)) v2 = 0;
+                v1 += f_corr_at(ws, dcount, dense, n, s);
+                v2 += f_corr_at(ws, dcount, dense, n, e);
+              }
+              const int64_t wEndOff = (int64_t)w * qstep - Ae;   // wEnd - ts0
+              res = d_extrapolated_rate(wEndOff - qwindow, wEndOff, e - s + 1,
+                                        t1, v1, t2, v2,
+                                        IS_COUNTER, FUNC == FN_RATE);
             }
-            const int64_t wEndOff = (int64_t)w * qstep - Ae;   // wEnd - ts0
-            res = d_extrapolated_rate(wEndOff - qwindow, wEndOff,
-                                      eA[k] - sA[k] + 1,
-                                      t1A[k], v1, t2A[k], v2,
-                                      IS_COUNTER, FUNC == FN_RATE);
           }
           emit_res(k, w, wok, res);
         }
       } else if constexpr (KIND == K_PFX) {
-        // staged prefix-diff loads, then the per-function epilogues
-        int sA[4], eA[4], pcA[4];
-        double psA[4];
         #pragma unroll
         for (int k = 0; k < 4; k++) {
           const int wi = lane + 64 * k;
           const bool wok = wi < tn;
-          sA[k] = wok ? ws.sw[wi] : 1;
-          eA[k] = wok ? ws.ew[wi] : -1;
-        }
-        #pragma unroll
-        for (int k = 0; k < 4; k++) {
-          const bool v = eA[k] >= sA[k] && eA[k] >= 0;
-          const int sc = v ? sA[k] : 1, ec = v ? eA[k] : 0;
-          psA[k] = ws.val[ec] - (sc ? ws.val[sc - 1] : 0.0);
-          pcA[k] = (int)ws.cnt[ec] - (sc ? (int)ws.cnt[sc - 1] : 0);
-        }
-        #pragma unroll
-        for (int k = 0; k < 4; k++) {
-          const int wi = lane + 64 * k;
-          const bool wok = wi < tn;
+          const int s = wok ? ws.sw[wi] : 1;
+          const int e = wok ? ws.ew[wi] : -1;
           double res = NAN;
-          if (eA[k] >= sA[k] && eA[k] >= 0) {
-            const double ps = psA[k];
-            const int pc = pcA[k];
+          if (e >= s && e >= 0) {
+            double ps = ws.val[e] - (s ? ws.val[s - 1] : 0.0);
+            int pc = (int)ws.cnt[e] - (s ? (int)ws.cnt[s - 1] : 0);
             if constexpr (FUNC == FN_SUM) res = pc > 0 ? ps : NAN;
             else if constexpr (FUNC == FN_COUNT) res = (double)pc;
             else if constexpr (FUNC == FN_AVG)
