@@ -108,7 +108,7 @@ __device__ __forceinline__ double f_corr_at(const WS& ws, int dcount, bool dense
 // registers (4 waves); K_RATE (no reciprocal table, LDS fits 6 blocks/CU) is
 // built at BOTH 5 (96 VGPR, no spills) and 6 (80 VGPR, ~19 spilled) — the
 // launcher picks via FDB_RATE_WAVES, measured on hardware.
-template <int FUNC, int EMIT, int MINW>
+template <int FUNC, int EMIT, int MINW, bool TIMED = false>
 __global__ __launch_bounds__(FAST_WAVES * 64, MINW)
 void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                       const int32_t* __restrict__ series_first,
@@ -128,7 +128,11 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   __shared__ FWs<KIND> ws_all[FAST_WAVES];
   __shared__ double inv_tab[NeedsInv<KIND>::v ? FAST_ROWS + 2 : 1];
 
-  const int wave = threadIdx.x >> 6;
+  // readfirstlane pins the wave id (and everything derived from it — series
+  // ids, directory offsets, header pointers) as wave-uniform for the
+  // compiler: the whole per-series bookkeeping chain then lives in SGPRs
+  // with scalar loads instead of occupying VGPRs
+  const int wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
   FWs<KIND>& ws = ws_all[wave];
 
@@ -140,7 +144,8 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 
   const double inv_step = 1.0 / (double)qstep;
   const double rate_scale = 1000.0 / (double)qwindow;
-  const bool timing = (pm & 8) != 0;    // perf ablation (clobbers out[])
+  (void)pm;
+  constexpr bool timing = TIMED;        // perf ablation (clobbers out[])
   uint64_t tD = 0, tM = 0, tI = 0, tW = 0, tt = 0;
 
   // series iteration: grid-stride for the plain grid emit; a contiguous slab
@@ -608,14 +613,25 @@ int32_t fdb_launch_fast_scan(hipStream_t stream, const uint8_t* blob, DirSoA dir
       dim3(grid), dim3(FAST_WAVES * 64), 0, stream, FARGS)
   #define FCASE(F, W) case F: \
     if (emit_group) LAUNCH(F, 1, 4); else LAUNCH(F, 0, W); break
+  const bool timed = (phase_mask & 8) != 0;   // s_memtime phase ablation
   switch (func_id) {
     case FN_RATE:
       if (emit_group) LAUNCH(FN_RATE, 1, 4);
+      else if (timed) hipLaunchKernelGGL((fast_scan_kernel<FN_RATE, 0, 5, true>),
+                                         dim3(grid), dim3(FAST_WAVES * 64), 0,
+                                         stream, FARGS);
       else if (rate_w == 5) LAUNCH(FN_RATE, 0, 5);
       else LAUNCH(FN_RATE, 0, 6);
       break;
+    case FN_AVG:
+      if (emit_group) LAUNCH(FN_AVG, 1, 4);
+      else if (timed) hipLaunchKernelGGL((fast_scan_kernel<FN_AVG, 0, 5, true>),
+                                         dim3(grid), dim3(FAST_WAVES * 64), 0,
+                                         stream, FARGS);
+      else LAUNCH(FN_AVG, 0, 5);
+      break;
     FCASE(FN_INCREASE, 6); FCASE(FN_DELTA, 6); FCASE(FN_SUM, 5);
-    FCASE(FN_COUNT, 5); FCASE(FN_AVG, 5); FCASE(FN_RATE_OVER_DELTA, 5);
+    FCASE(FN_COUNT, 5); FCASE(FN_RATE_OVER_DELTA, 5);
     FCASE(FN_MIN, 5); FCASE(FN_MAX, 5); FCASE(FN_STDDEV, 3); FCASE(FN_STDVAR, 3);
     FCASE(FN_CHANGES, 5); FCASE(FN_LAST, 5); FCASE(FN_PRESENT, 5);
     FCASE(FN_TIMESTAMP, 5); FCASE(FN_ZSCORE, 5);
