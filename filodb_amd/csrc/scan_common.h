@@ -335,8 +335,11 @@ __device__ __forceinline__ int fdiv_floor_win(int64_t a, int64_t b, double inv_b
   if (g < -1.0) return -1;
   if (g > (double)hi + 1.0) return hi + 1;
   int64_t w = (int64_t)__builtin_floor(g);
-  w += ((w + 1) * b <= a);
-  w -= (w * b > a);
+  // residual a - w*b via fma: |residual| < 2b << 2^53, so the single-rounded
+  // fma result is exact — replaces two software i64 multiplies
+  double r = __builtin_fma(-(double)w, (double)b, (double)a);
+  w += (r >= (double)b);
+  w -= (r < 0.0);
   return (int)w;
 }
 

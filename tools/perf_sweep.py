@@ -47,6 +47,12 @@ def main():
         os.environ["FDB_RATE_WAVES"] = waves
         ms = bench(eng, ds, q_rate, out)
         print(f"rate waves={waves}: {ms:.3f} ms", flush=True)
+    os.environ["FDB_RATE_WAVES"] = "6"
+    for gridcap in ("1536", "3072", "16384"):
+        os.environ["FDB_GRID"] = gridcap
+        ms = bench(eng, ds, q_rate, out)
+        print(f"rate grid={gridcap}: {ms:.3f} ms", flush=True)
+    os.environ.pop("FDB_GRID")
     for fused in ("1", "0"):
         os.environ["FDB_FUSED_GROUP"] = fused
         ms = bench(eng, ds, q_sum, gout, gcnt)
