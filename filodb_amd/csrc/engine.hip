@@ -42,752 +42,32 @@ int32_t fdb_launch_fast_scan(hipStream_t stream, const uint8_t* blob, DirSoA dir
                              int phase_mask);
 bool fdb_fast_scan_supported(int func_id);
 
+// scan_stream.hip: unbounded multi-chunk general path (chunk summaries +
+// per-window walk)
+int32_t fdb_launch_summaries(hipStream_t stream, const uint8_t* blob, DirSoA dir,
+                             int64_t num_chunks, void* sums);
+int64_t fdb_chunksum_bytes(int64_t num_chunks);
+bool fdb_stream_walk_supported(int func_id);
+int32_t fdb_launch_stream_walk(hipStream_t stream, const uint8_t* blob,
+                               DirSoA dir, const void* sums,
+                               const int32_t* series_first,
+                               const int32_t* series_nchunks, int num_series,
+                               int64_t qstart, int64_t qstep, int64_t qend,
+                               int64_t qwindow, int num_windows, int func_id,
+                               double* out);
+
 #define HIP_CHECK(expr) do { hipError_t _e = (expr); if (_e != hipSuccess) { \
   fdb_set_error("%s failed: %s", #expr, hipGetErrorString(_e)); return FDB_ERR; } } while (0)
 #define HIP_CHECK_NULL(expr) do { hipError_t _e = (expr); if (_e != hipSuccess) { \
   fdb_set_error("%s failed: %s", #expr, hipGetErrorString(_e)); return nullptr; } } while (0)
 
-// ---------------------------------------------------------------------------
-// capacity limits (round 1)
-// ---------------------------------------------------------------------------
-#define FDB_MAX_ROWS_PER_SERIES 400   // LDS-resident rows per series (= chunk row cap)
-#define FDB_MAX_CHUNKS_PER_SERIES 8   /* small tier; longer series use the large tier */
+// The round-1 LDS-resident scan kernel (two capacity tiers, 1600 rows / 16
+// chunks) is gone: single-chunk series run scan_fast.hip, everything else
+// runs scan_stream.hip with no capacity caps.
+#define FDB_MAX_ROWS_PER_SERIES 400   // per-chunk row cap (reference default)
 #define WAVES_PER_BLOCK 4
-#define BLOCK_THREADS (WAVES_PER_BLOCK * 64)
 
-// FN_*/AGG_* ids, DirSoA, DVec readers, decode, wave scans, extrapolatedRate:
-// scan_common.h (shared with scan_fast.hip)
-
-// ---------------------------------------------------------------------------
-// per-series LDS workspace
-// ---------------------------------------------------------------------------
-struct ChunkMeta {
-  int32_t row0, nrows;
-  int64_t start_time, end_time;    // directory times (pre-encoding, like ChunkSetInfo)
-  int64_t ts0;                     // decoded first timestamp (search guess base)
-  float   inv_slope;               // (nrows-1)/(ts_last-ts0); 0 when degenerate
-  int32_t dropped;
-  double last_for_update;          // updateCorrection lastValue (DoubleVector.scala:375-391,190-195)
-  double chunk_correction;         // CorrectingDoubleVectorReader._correction total
-  int16_t dstart, dcount;          // slice of the wave drop table
-  int16_t dense_corr;              // >MAX_DROPS drops: recompute corrections serially
-  int16_t v0_nan;                  // isNaN(apply(0)) for the single-row-NaN rule
-};
-
-#define FDB_MAX_DROPS 32            // wave-wide counter-reset table capacity
-
-// which auxiliary LDS arrays a function family needs
-#define K_RATE    0   // ts, val, drop table
-#define K_PFX     1   // ts, prefix-sum (in place of val) + count prefix
-#define K_PFX_SQ  2   // + squared prefix (stddev/stdvar)
-#define K_MINMAX  3   // ts, val, 8-element group min/max
-#define K_CHANGES 4   // ts, val, change-indicator prefix
-#define K_LAST    5   // ts, val only (LastSampleChunkedFunctionD)
-
-template <int FUNC> struct KKind { static constexpr int v =
-    (FUNC <= FN_DELTA) ? K_RATE :
-    (FUNC == FN_SUM || FUNC == FN_AVG || FUNC == FN_COUNT) ? K_PFX :
-    (FUNC == FN_STDDEV || FUNC == FN_STDVAR) ? K_PFX_SQ :
-    (FUNC == FN_MIN || FUNC == FN_MAX) ? K_MINMAX :
-    (FUNC == FN_LAST || FUNC == FN_PRESENT || FUNC == FN_TIMESTAMP)
-        ? K_LAST : K_CHANGES; };   // FN_ZSCORE uses K_CHANGES raw values
-
-#define FDB_NGROUPS ((FDB_MAX_ROWS_PER_SERIES + 7) / 8)
-
-// two capacity tiers: the common one-chunk-per-span shape (4 waves/block) and
-// a long-lookback tier for series spanning several 400-row chunks
-// (2 waves/block; SURVEY §5: long windows crossing many chunks)
-#define FDB_ROWS_LARGE 1600
-#define FDB_CHUNKS_LARGE 16
-
-template <int KIND, int CAP, int MAXC>
-struct Ws {                         // per-wave LDS workspace
-  int64_t ts[CAP];
-  double  val[CAP];                 // raw values, or inclusive NaN-zeroed
-                                    // prefix sums for the PFX kinds
-  // counter resets are rare: (position, cumulative in-chunk correction) pairs
-  // instead of a full corrected[] copy (CorrectingDoubleVectorReader :325-342)
-  int16_t dpos[KIND == K_RATE ? FDB_MAX_DROPS : 1];
-  double  dcum[KIND == K_RATE ? FDB_MAX_DROPS : 1];
-  // non-NaN count prefix (PFX kinds) / change-indicator prefix (CHANGES)
-  uint16_t cnt[(KIND == K_PFX || KIND == K_PFX_SQ || KIND == K_CHANGES) ? CAP : 1];
-  double  sq[KIND == K_PFX_SQ ? CAP : 1];   // squared prefix
-  double  grp[KIND == K_MINMAX ? ((CAP + 7) / 8 + MAXC) : 1]; // 8-elem group min/max (chunk-relative, padded per chunk)
-  ChunkMeta cm[MAXC];
-  int32_t nchunks;
-  int32_t total_rows;
-};
-
-// first index in [0,n) with seg[i] >= item; n when none. Timestamps sit near a
-// slope line (the DDV premise), so an interpolation guess + short walk replaces
-// a dependent binary-search chain (usually 1-2 LDS reads) — the same idea as
-// DeltaDeltaDataReader.binarySearch (DeltaDeltaVector.scala:159-188).
-__device__ __forceinline__ int lds_search_ge(const int64_t* seg, int n, int64_t item,
-                                             int64_t ts0, float inv_slope) {
-  if (n <= 0) return 0;
-  int g = (int)((float)(item - ts0) * inv_slope);
-  if (g < 0) g = 0;
-  if (g > n - 1) g = n - 1;
-  while (g > 0 && seg[g - 1] >= item) g--;
-  while (g < n && seg[g] < item) g++;
-  return g;
-}
-// last index with seg[i] <= item; -1 when none (ceilingIndex semantics).
-// Integer timestamps: last <= item == (first >= item+1) - 1.
-__device__ __forceinline__ int lds_search_le(const int64_t* seg, int n, int64_t item,
-                                             int64_t ts0, float inv_slope) {
-  return lds_search_ge(seg, n, item + 1, ts0, inv_slope) - 1;
-}
-
-// in-chunk correction lookup from the sparse drop table (the step function
-// CorrectingDoubleVectorReader :325-342 materializes as corrected[])
-template <typename WS>
-__device__ __forceinline__ double d_corr_at(const WS& ws, const ChunkMeta& m, int i) {
-  if (m.dense_corr) {
-    // serial recompute — only when one chunk held >FDB_MAX_DROPS resets
-    double corr = 0, last = -1.7976931348623157e308;
-    for (int j = 0; j <= i; j++) {
-      double x = ws.val[m.row0 + j];
-      if (isnan(x)) x = 0;
-      if (x < last) corr += last;
-      last = x;
-    }
-    return corr;
-  }
-  double c = 0;
-  for (int j = m.dstart; j < m.dstart + m.dcount; j++) {
-    if (ws.dpos[j] <= i) c = ws.dcum[j]; else break;
-  }
-  return c;
-}
-template <typename WS>
-__device__ __forceinline__ double d_corrected(const WS& ws, const ChunkMeta& m, int i) {
-  double x = ws.val[m.row0 + i];
-  if (m.dropped) { if (isnan(x)) x = 0; x += d_corr_at(ws, m, i); }
-  return x;
-}
-
-// ---------------------------------------------------------------------------
-// the scan kernel: one wavefront per series
-//   decode → per-chunk meta (search slopes, correction table, prefixes) →
-//   per-window evaluation with O(1)-ish row location and O(1) prefix lookups
-// ---------------------------------------------------------------------------
-// rate family carries more live state (correction meta): forcing 5 waves/SIMD
-// makes it spill; the gauge kinds fit 5 waves cleanly
-template <int FUNC, int CAP = FDB_MAX_ROWS_PER_SERIES,
-          int MAXC = FDB_MAX_CHUNKS_PER_SERIES, int WAVES = WAVES_PER_BLOCK>
-__global__ __launch_bounds__(WAVES * 64, (FUNC <= FN_DELTA) ? 4 : 5)
-void scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
-                 const int32_t* __restrict__ series_first,
-                 const int32_t* __restrict__ series_nchunks,
-                 const int32_t* __restrict__ group_ids,
-                 int num_series,
-                 int64_t qstart, int64_t qstep, int64_t qend, int64_t qwindow,
-                 int num_windows,
-                 int agg_id,
-                 double* __restrict__ out,        // [S×W] when AGG_NONE else [G×W] sums
-                 double* __restrict__ out_cnt,    // [G×W] contribution counts (agg) or null
-                 double* __restrict__ out_sq,     // [G×W] sum-of-squares (stddev aggs) or null
-                 int phase_mask)                  // debug ablation: 1=decode 2=windows
-{
-  constexpr int KIND = KKind<FUNC>::v;
-  constexpr bool RATE_FAMILY = (KIND == K_RATE);
-  __shared__ Ws<KIND, CAP, MAXC> ws_all[WAVES];
-
-  const int wave = threadIdx.x >> 6;
-  const int lane = threadIdx.x & 63;
-  Ws<KIND, CAP, MAXC>& ws = ws_all[wave];
-
-  // phase timing (ablation: phase_mask & 8) — wall cycles per wave per phase
-  const bool timing = (phase_mask & 8) != 0;
-  uint64_t t_decode = 0, t_meta = 0, t_win = 0, tt0 = 0;
-  if (timing) tt0 = __builtin_amdgcn_s_memtime();
-  // residency census (ablation: phase_mask & 16): concurrent-block high-water
-  // in out_cnt[0..1] (u64 counters)
-  if ((phase_mask & 16) && threadIdx.x == 0) {
-    unsigned long long cur =
-        atomicAdd((unsigned long long*)&out_cnt[0], 1ULL) + 1;
-    atomicMax((unsigned long long*)&out_cnt[1], cur);
-  }
-
-  // grid-stride over series: a few resident blocks loop over the shard instead
-  // of one block per 4 series — 250k tiny-workgroup dispatches cost multiple ms
-  // of pure launch churn (measured ~20-26ns each), so the loop, not the grid,
-  // walks the series
-  for (int sid = blockIdx.x * WAVES + wave; sid < num_series;
-       sid += gridDim.x * WAVES) {
-  // ---- decode phase: all chunks of this series into LDS --------------------
-  const int first = series_first[sid];
-  int nchunks = series_nchunks[sid];
-  if (nchunks > MAXC) nchunks = MAXC;
-  if (lane == 0) { ws.nchunks = nchunks; ws.total_rows = 0; }
-
-  int row0 = 0;
-  for (int c = 0; c < nchunks; c++) {
-    int n = dir.num_rows[first + c];
-    if (row0 + n > CAP) n = 0;   // guarded at upload
-    DVec vv;
-    if (phase_mask & 1) {
-      DVec tv;
-      d_vec_open(blob + dir.ts_off[first + c], &tv);
-      d_vec_open(blob + dir.val_off[first + c], &vv);
-      d_decode_chunk<false>(tv, n, ws.ts + row0, nullptr, lane);
-      d_decode_chunk<true>(vv, n, nullptr, ws.val + row0, lane);
-    } else {                       // window-only ablation: synthesize rows
-      vv.dropped = 0;
-      for (int i = lane; i < n; i += 64) {
-        ws.ts[row0 + i] = dir.start_time[first + c] + (int64_t)i * 15000;
-        ws.val[row0 + i] = (double)(i * 10);
-      }
-    }
-    if (lane == 0) {
-      ChunkMeta& m = ws.cm[c];
-      m.row0 = row0; m.nrows = n;
-      m.start_time = dir.start_time[first + c];
-      m.end_time = dir.end_time[first + c];
-      m.dropped = vv.dropped;
-      m.chunk_correction = 0; m.last_for_update = 0;
-      m.dcount = 0; m.dense_corr = 0; m.v0_nan = 0;
-    }
-    row0 += n;
-  }
-  if (lane == 0) ws.total_rows = row0;
-  d_wait_lds();                    // wave-local LDS visibility
-  __builtin_amdgcn_wave_barrier();
-  if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_decode += t - tt0; tt0 = t; }
-
-  // ---- meta phase ----------------------------------------------------------
-  // every kind: search-guess slope per chunk
-  // K_RATE: counter-correction scan (CorrectingDoubleVectorReader :325-342)
-  //         recorded as a sparse drop table + updateCorrection scalars
-  // K_PFX(_SQ): NaN-zeroed value (and square) prefix + non-NaN count prefix —
-  //         the ~97% window overlap then costs O(1) per window, not O(rows)
-  // K_MINMAX: 8-element group min/max
-  // K_CHANGES: change-indicator prefix (DoubleVectorDataReader64.changes :283-302)
-  int table_used = 0;
-  for (int c = 0; c < nchunks; c++) {
-    ChunkMeta& m = ws.cm[c];
-    const int r0 = m.row0, n = m.nrows;
-    if (n == 0) continue;
-    if (lane == 0) {
-      m.ts0 = ws.ts[r0];
-      int64_t tl = ws.ts[r0 + n - 1];
-      m.inv_slope = (tl > m.ts0) ? (float)(n - 1) / (float)(tl - m.ts0) : 0.0f;
-    }
-
-    if constexpr (RATE_FAMILY) {
-      if (lane == 0) m.v0_nan = isnan(ws.val[r0]);
-      if (m.dropped) {
-        double carry_corr = 0;
-        double carry_x = -1.7976931348623157e308;   // 'last' starts Double.MinValue
-        int last_idx = -1;
-        double last_val = 0;
-        int dstart = table_used, dcnt = 0;
-        bool overflow = false;
-        for (int base = 0; base < n; base += 64) {
-          int i = base + lane;
-          double raw = (i < n) ? ws.val[r0 + i] : 0;
-          double x = (i < n && !isnan(raw)) ? raw : 0;
-          double px = __shfl_up(x, 1);
-          if (lane == 0) px = carry_x;
-          double ci = (i < n && x < px) ? px : 0;
-          double scan = wave_incl_scan(ci, lane);
-          uint64_t mask = __ballot(ci != 0);
-          int here = __popcll(mask);
-          if (here) {
-            if (dstart + dcnt + here > FDB_MAX_DROPS) {
-              overflow = true;
-            } else if (ci != 0) {
-              int slot = dstart + dcnt + __popcll(mask & ((1ULL << lane) - 1));
-              ws.dpos[slot] = (int16_t)i;
-              ws.dcum[slot] = carry_corr + scan;
-            }
-            dcnt += here;
-          }
-          carry_corr += __shfl(scan, 63);
-          carry_x = __shfl(x, 63);
-          if (i < n && !isnan(raw)) { last_idx = i; last_val = raw; }
-        }
-        for (int off = 32; off > 0; off >>= 1) {
-          int oi = __shfl_down(last_idx, off);
-          double ov = __shfl_down(last_val, off);
-          if (oi > last_idx) { last_idx = oi; last_val = ov; }
-        }
-        if (!overflow) table_used = dstart + dcnt;
-        if (lane == 0) {
-          m.chunk_correction = carry_corr;
-          m.last_for_update = (last_idx >= 0) ? last_val : 0;
-          m.dstart = (int16_t)dstart;
-          m.dcount = overflow ? 0 : (int16_t)dcnt;
-          m.dense_corr = overflow ? 1 : 0;
-        }
-      } else if (lane == 0) {
-        m.last_for_update = ws.val[r0 + n - 1];     // default updateCorrection
-      }
-    }
-
-    if constexpr (KIND == K_PFX || KIND == K_PFX_SQ) {
-      double carry = 0, carry_sq = 0;
-      int ccarry = 0;
-      for (int base = 0; base < n; base += 64) {
-        int i = base + lane;
-        double raw = (i < n) ? ws.val[r0 + i] : NAN;
-        bool ok = (i < n) && !isnan(raw);
-        double x = ok ? raw : 0;
-        double s = wave_incl_scan(x, lane);
-        int cs = wave_incl_scan_i(ok ? 1 : 0, lane);
-        double sqs = 0;
-        if constexpr (KIND == K_PFX_SQ) sqs = wave_incl_scan(x * x, lane);
-        __builtin_amdgcn_wave_barrier();            // all raw reads precede writes
-        if (i < n) {
-          ws.val[r0 + i] = carry + s;               // val[] becomes the prefix
-          ws.cnt[r0 + i] = (uint16_t)(ccarry + cs);
-          if constexpr (KIND == K_PFX_SQ) ws.sq[r0 + i] = carry_sq + sqs;
-        }
-        carry += __shfl(s, 63);
-        ccarry += __shfl(cs, 63);
-        if constexpr (KIND == K_PFX_SQ) carry_sq += __shfl(sqs, 63);
-      }
-    }
-
-    if constexpr (KIND == K_MINMAX) {
-      constexpr bool IS_MIN = (FUNC == FN_MIN);
-      for (int base = 0; base < n; base += 64) {
-        int i = base + lane;
-        double x = (i < n) ? ws.val[r0 + i] : NAN;
-        // 8-lane group reduce (elements 8g..8g+7 are lanes 8k..8k+7)
-        #pragma unroll
-        for (int off = 1; off < 8; off <<= 1) {
-          double o = __shfl_xor(x, off);
-          if (!isnan(o) && (isnan(x) || (IS_MIN ? o < x : o > x))) x = o;
-        }
-        if ((lane & 7) == 0 && i < n)
-          ws.grp[(r0 >> 3) + c + (i >> 3)] = x;   // chunk-relative groups;
-                                                  // +c pads across chunk seams
-      }
-    }
-
-    if constexpr (KIND == K_CHANGES) {
-      // ind[i] = in-chunk change at i (i>0); prefix stored in cnt[]
-      int ccarry = 0;
-      double carry_x = NAN;
-      for (int base = 0; base < n; base += 64) {
-        int i = base + lane;
-        double x = (i < n) ? ws.val[r0 + i] : NAN;
-        double px = __shfl_up(x, 1);
-        if (lane == 0) px = carry_x;
-        int ind = (i > 0 && i < n && !isnan(x) && !isnan(px) && x != px) ? 1 : 0;
-        int s = wave_incl_scan_i(ind, lane);
-        if (i < n) ws.cnt[r0 + i] = (uint16_t)(ccarry + s);
-        ccarry += __shfl(s, 63);
-        carry_x = __shfl(x, 63);
-      }
-    }
-  }
-  d_wait_lds();
-  __builtin_amdgcn_wave_barrier();
-  if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_meta += t - tt0; tt0 = t; }
-
-  // ---- window phase: lanes split the windows -------------------------------
-  if (!(phase_mask & 2)) {            // decode-only ablation: publish a checksum
-    if (lane == 0 && ws.total_rows > 0)
-      out[(size_t)sid * num_windows] = ws.val[0] + (double)ws.ts[0];
-    continue;
-  }
-  const int grp_id = group_ids[sid];
-  // single-chunk fast path (the common shape: one 400-row-capped chunk per
-  // query span): chunk meta lives in registers, no per-window chunk loop
-  if (ws.nchunks == 1 && ws.cm[0].nrows > 0) {
-    const ChunkMeta mr = ws.cm[0];       // one LDS read burst, then registers
-    const int64_t* seg = ws.ts;
-    const int n1 = mr.nrows;
-    // staged window batches: the search walks and the f64 epilogue chains of a
-    // lane's 4 windows are independent — computing them in unrolled stages lets
-    // the scheduler overlap their latency chains (phase timing showed the
-    // serial per-window loop was 58% of wave time)
-    for (int wbase = 0; wbase < num_windows; wbase += 64 * 4) {
-      int sRk[4], eRk[4];
-      // stage A: interpolation guess + 3-step branchless walk, all windows
-      #pragma unroll
-      for (int k = 0; k < 4; k++) {
-        int w = wbase + lane + 64 * k;
-        sRk[k] = 0x7fffffff;
-        if (w < num_windows) {
-          int64_t wEnd = qstart + (int64_t)w * qstep;
-          int64_t wStart = wEnd - qwindow;
-          if (mr.end_time >= wStart) {
-            int g = (int)((float)(wStart - mr.ts0) * mr.inv_slope);
-            g = g < 0 ? 0 : (g > n1 - 1 ? n1 - 1 : g);
-            int h = (int)((float)(wEnd + 1 - mr.ts0) * mr.inv_slope);
-            h = h < 0 ? 0 : (h > n1 - 1 ? n1 - 1 : h);
-            #pragma unroll
-            for (int it = 0; it < 3; it++) {
-              int64_t ga = seg[g > 0 ? g - 1 : 0], gb = seg[g < n1 ? g : n1 - 1];
-              g -= (g > 0 && ga >= wStart) ? 1 : 0;
-              g += (g < n1 && gb < wStart) ? 1 : 0;
-              int64_t ha = seg[h > 0 ? h - 1 : 0], hb = seg[h < n1 ? h : n1 - 1];
-              h -= (h > 0 && ha >= wEnd + 1) ? 1 : 0;
-              h += (h < n1 && hb < wEnd + 1) ? 1 : 0;
-            }
-            sRk[k] = g; eRk[k] = h;
-          }
-        }
-      }
-      // stage B: verify; rare irregular-cadence lanes take the full walk
-      #pragma unroll
-      for (int k = 0; k < 4; k++) {
-        if (sRk[k] == 0x7fffffff) continue;
-        int w = wbase + lane + 64 * k;
-        int64_t wEnd = qstart + (int64_t)w * qstep;
-        int64_t wStart = wEnd - qwindow;
-        int g = sRk[k], h = eRk[k];
-        if (!((g == 0 || seg[g - 1] < wStart) && (g == n1 || seg[g] >= wStart)))
-          g = lds_search_ge(seg, n1, wStart, mr.ts0, mr.inv_slope);
-        if (!((h == 0 || seg[h - 1] <= wEnd) && (h == n1 || seg[h] > wEnd)))
-          h = lds_search_ge(seg, n1, wEnd + 1, mr.ts0, mr.inv_slope);
-        sRk[k] = g; eRk[k] = h - 1;
-      }
-      // stage C: evaluate + emit
-      #pragma unroll
-      for (int k = 0; k < 4; k++) {
-      int w = wbase + lane + 64 * k;
-      if (w >= num_windows) continue;
-      const int64_t wEnd = qstart + (int64_t)w * qstep;
-      const int64_t wStart = wEnd - qwindow;
-      double result = NAN;
-      if constexpr (FUNC == FN_TIMESTAMP) {
-        // TimestampChunkedFunction (RangeFunction.scala:705-723): last ts <=
-        // wEnd in the window's chunk list, no window-start bound; seconds
-        int endRow = sRk[k] == 0x7fffffff ? -1 : eRk[k];
-        if (endRow >= 0 && endRow < mr.nrows)
-          result = (double)seg[endRow] / 1000.0;
-      } else {
-        int startRow = sRk[k] == 0x7fffffff ? 1 : sRk[k];
-        int endRow = sRk[k] == 0x7fffffff ? 0 : eRk[k];
-        if (startRow <= endRow && endRow < mr.nrows) {
-          if constexpr (RATE_FAMILY) {
-            constexpr bool isCounter = (FUNC != FN_DELTA);
-            if (!(isCounter && startRow == 0 && endRow == 0 && mr.v0_nan)) {
-              int64_t st = seg[startRow], en = seg[endRow];
-              if (en > st) {
-                int numSamples = endRow - startRow + 1;
-                double lo = isCounter ? d_corrected(ws, mr, startRow)
-                                      : ws.val[startRow];
-                double hi = isCounter ? d_corrected(ws, mr, endRow)
-                                      : ws.val[endRow];
-                result = d_extrapolated_rate(wStart, wEnd, numSamples, st, lo,
-                                             en, hi, isCounter, FUNC == FN_RATE);
-              }
-            }
-          } else if constexpr (KIND == K_PFX || KIND == K_PFX_SQ) {
-            double ps = ws.val[endRow] - (startRow ? ws.val[startRow - 1] : 0.0);
-            int pc = (int)ws.cnt[endRow] - (startRow ? (int)ws.cnt[startRow - 1] : 0);
-            if (FUNC == FN_SUM) result = pc > 0 ? ps : NAN;
-            else if (FUNC == FN_COUNT) result = (double)pc;
-            else if (FUNC == FN_AVG) result = pc > 0 ? ps / pc : NAN;
-            else {  // stddev / stdvar over one chunk range
-              if (pc > 0) {
-                double qs = ws.sq[endRow] - (startRow ? ws.sq[startRow - 1] : 0.0);
-                double avg = ps / pc;
-                double r = qs / pc - avg * avg;
-                result = (FUNC == FN_STDDEV) ? sqrt(r) : r;
-              } else result = NAN;
-            }
-          } else if constexpr (KIND == K_MINMAX) {
-            constexpr bool IS_MIN = (FUNC == FN_MIN);
-            double mm = NAN;
-            auto acc = [&](double x) {
-              if (!isnan(x) && (isnan(mm) || (IS_MIN ? x < mm : x > mm))) mm = x;
-            };
-            int ga = (startRow + 7) >> 3, gb = (endRow + 1) >> 3;
-            if (ga < gb) {
-              for (int i = startRow; i < ga * 8; i++) acc(ws.val[i]);
-              for (int g = ga; g < gb; g++) acc(ws.grp[g]);
-              for (int i = gb * 8; i <= endRow; i++) acc(ws.val[i]);
-            } else {
-              for (int i = startRow; i <= endRow; i++) acc(ws.val[i]);
-            }
-            result = mm;
-          } else if constexpr (FUNC == FN_PRESENT) {
-            // PresentOverTimeChunkedFunctionD (RangeFunction.scala:725-745):
-            // non-NaN -> 1; a NaN stale marker steps back one row
-            double v = ws.val[endRow];
-            if (!isnan(v)) result = 1.0;
-            else if (endRow > 0)
-              result = isnan(ws.val[endRow - 1]) ? NAN : 1.0;
-          } else if constexpr (KIND == K_LAST) {
-            // last sample <= wEnd within window; raw value (NaN propagates)
-            result = seg[endRow] >= wStart ? ws.val[endRow] : NAN;
-          } else if constexpr (FUNC == FN_ZSCORE) {
-            // ZScoreChunkedFunctionD (AggrOverTimeFunctions.scala:1592-1603)
-            double s = NAN, sq = NAN, lastv = NAN;
-            int pc = 0;
-            for (int i = startRow; i <= endRow; i++) {
-              double x = ws.val[i];
-              if (isnan(x)) continue;
-              if (isnan(s)) { s = 0; sq = 0; }
-              if (i == endRow) lastv = x;
-              s += x; sq += x * x; pc++;
-            }
-            if (pc > 0) {
-              double avg = s / pc;
-              double sd = sqrt(sq / pc - avg * avg);
-              result = (lastv - avg) / sd;
-            } else result = isnan(s) ? s : 0;
-          } else {  // K_CHANGES, single chunk: prefix diff, prev starts NaN
-            result = (double)((int)ws.cnt[endRow] - (int)ws.cnt[startRow]);
-          }
-        }
-      }
-      if (agg_id == AGG_NONE) {
-        out[(size_t)sid * num_windows + w] = result;
-      } else if (!isnan(result)) {
-        size_t cell = (size_t)grp_id * num_windows + w;
-        switch (agg_id) {
-          case AGG_SUM: case AGG_AVG:
-            atomicAdd(&out[cell], result); atomicAdd(&out_cnt[cell], 1.0); break;
-          case AGG_STDDEV: case AGG_STDVAR:
-            atomicAdd(&out[cell], result); atomicAdd(&out_sq[cell], result * result);
-            atomicAdd(&out_cnt[cell], 1.0); break;
-          case AGG_COUNT:
-            atomicAdd(&out[cell], 1.0); atomicAdd(&out_cnt[cell], 1.0); break;
-          case AGG_MIN: atomic_min_max_f64(&out[cell], result, true);
-                        atomicAdd(&out_cnt[cell], 1.0); break;
-          case AGG_MAX: atomic_min_max_f64(&out[cell], result, false);
-                        atomicAdd(&out_cnt[cell], 1.0); break;
-        }
-      }
-      }  // stage C per-window
-    }  // window batches
-    d_wait_lds();
-    __builtin_amdgcn_wave_barrier();
-    if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_win += t - tt0; tt0 = t; }
-    continue;
-  }
-
-  for (int w = lane; w < num_windows; w += 64) {
-    const int64_t wEnd = qstart + (int64_t)w * qstep;
-    const int64_t wStart = wEnd - qwindow;
-    double result = NAN;
-
-    if constexpr (RATE_FAMILY) {
-      bool meta_has = false;
-      double meta_last = 0, meta_corr = 0;
-      int numSamples = 0;
-      int64_t lowestTime = INT64_MAX, highestTime = 0;
-      double lowestValue = NAN, highestValue = NAN;
-      constexpr bool isCounter = (FUNC != FN_DELTA);
-      for (int c = 0; c < ws.nchunks; c++) {
-        const ChunkMeta& m = ws.cm[c];
-        if (m.end_time < wStart) continue;          // WindowedChunkIterator drop rule
-        const int64_t* seg = ws.ts + m.row0;
-        int startRow = lds_search_ge(seg, m.nrows, wStart, m.ts0, m.inv_slope);
-        int endRow = lds_search_le(seg, m.nrows, wEnd, m.ts0, m.inv_slope);
-        if (isCounter && meta_has) {                // detectDropAndCorrection
-          double firstv = ws.val[m.row0];
-          if (isnan(firstv) || firstv < meta_last) meta_corr += meta_last;
-        }
-        if (startRow <= endRow && endRow < m.nrows) {
-          bool skip = isCounter && startRow == 0 && endRow == 0 && m.v0_nan;
-          if (!skip) {
-            int64_t st = seg[startRow], en = seg[endRow];
-            if (st < lowestTime || en > highestTime) {
-              numSamples += endRow - startRow + 1;
-              if (st < lowestTime) {
-                lowestTime = st;
-                lowestValue = isCounter ? d_corrected(ws, m, startRow) + meta_corr
-                                        : ws.val[m.row0 + startRow];
-              }
-              if (en > highestTime) {
-                highestTime = en;
-                highestValue = isCounter ? d_corrected(ws, m, endRow) + meta_corr
-                                         : ws.val[m.row0 + endRow];
-              }
-            }
-          }
-        }
-        if (isCounter) {                            // updateCorrection
-          if (m.dropped) meta_corr += m.chunk_correction;
-          meta_last = m.last_for_update;
-          meta_has = true;
-        }
-        if (m.end_time >= wEnd) break;
-      }
-      if (highestTime > lowestTime)
-        result = d_extrapolated_rate(wStart, wEnd, numSamples,
-                                     lowestTime, lowestValue, highestTime, highestValue,
-                                     isCounter, FUNC == FN_RATE);
-    } else {
-      double sum = NAN, sqsum = NAN, mm = NAN;
-      double changes = NAN, prev = NAN;
-      double last_val = NAN;
-      int64_t last_ts = -1;
-      int icount = 0;
-      bool started = false;                        // CountOverTime: saw a nonempty range
-      for (int c = 0; c < ws.nchunks; c++) {
-        const ChunkMeta& m = ws.cm[c];
-        if (m.end_time < wStart) continue;
-        const int64_t* seg = ws.ts + m.row0;
-        int startRow = lds_search_ge(seg, m.nrows, wStart, m.ts0, m.inv_slope);
-        int endRow = lds_search_le(seg, m.nrows, wEnd, m.ts0, m.inv_slope);
-        if constexpr (KIND == K_LAST) {
-          // LastSampleChunkedFunction.addChunks (RangeFunction.scala:599-614):
-          // no startRow search; last ts <= wEnd wins if within the window.
-          // FN_PRESENT maps to 1 with a one-row step-back on NaN markers
-          // (:725-745); FN_TIMESTAMP takes the bare ts without the
-          // window-start bound (:705-723).
-          if (endRow >= 0 && endRow < m.nrows) {
-            int64_t t = seg[endRow];
-            if constexpr (FUNC == FN_TIMESTAMP) {
-              if (t > last_ts) { last_ts = t; last_val = (double)t / 1000.0; }
-            } else if constexpr (FUNC == FN_PRESENT) {
-              if (t >= wStart && t > last_ts) {
-                double v = ws.val[m.row0 + endRow];
-                if (!isnan(v)) { last_ts = t; last_val = 1.0; }
-                else if (endRow > 0) {
-                  last_ts = t;
-                  last_val = isnan(ws.val[m.row0 + endRow - 1]) ? NAN : 1.0;
-                }
-              }
-            } else {
-              if (t >= wStart && t > last_ts) { last_ts = t; last_val = ws.val[m.row0 + endRow]; }
-            }
-          }
-          if (m.end_time >= wEnd) break;
-          continue;
-        }
-        if (startRow <= endRow && endRow < m.nrows) {
-          const int a = m.row0 + startRow, b = m.row0 + endRow;
-          if constexpr (KIND == K_PFX || KIND == K_PFX_SQ) {
-            // chunk-range aggregates from the prefixes (O(1) per window)
-            double ps = ws.val[b] - (startRow ? ws.val[a - 1] : 0.0);
-            int pc = (int)ws.cnt[b] - (startRow ? (int)ws.cnt[a - 1] : 0);
-            double cs = pc > 0 ? ps : NAN;          // all-NaN range sums to NaN
-            if (!isnan(cs) && isnan(sum)) sum = 0;
-            sum += cs;                              // NaN-poison quirk preserved
-            icount += pc;
-            started = true;
-            if constexpr (KIND == K_PFX_SQ) {
-              double qs = ws.sq[b] - (startRow ? ws.sq[a - 1] : 0.0);
-              double csq = pc > 0 ? qs : NAN;
-              if (!isnan(csq) && isnan(sqsum)) sqsum = 0;
-              sqsum += csq;
-            }
-          } else if constexpr (KIND == K_MINMAX) {
-            constexpr bool IS_MIN = (FUNC == FN_MIN);
-            auto acc = [&](double x) {
-              if (!isnan(x) && (isnan(mm) || (IS_MIN ? x < mm : x > mm))) mm = x;
-            };
-            // chunk-relative full groups in [ga, gb), slot base sb
-            int ga = (startRow + 7) >> 3, gb = (endRow + 1) >> 3;
-            int sb = (m.row0 >> 3) + c;
-            if (ga < gb) {
-              for (int i = startRow; i < ga * 8; i++) acc(ws.val[m.row0 + i]);
-              for (int g = ga; g < gb; g++) acc(ws.grp[sb + g]);
-              for (int i = gb * 8; i <= endRow; i++) acc(ws.val[m.row0 + i]);
-            } else {
-              for (int i = a; i <= b; i++) acc(ws.val[i]);
-            }
-          } else if constexpr (FUNC == FN_ZSCORE) {
-            // VarOverTimeChunkedFunctionD accumulation over raw values
-            // (AggrOverTimeFunctions.scala:1082-1115): lastSample set only
-            // when the range's endRow value is non-NaN (:1103)
-            double cs = NAN, csq = NAN;
-            int cc = 0;
-            for (int i = a; i <= b; i++) {
-              double x = ws.val[i];
-              if (isnan(x)) continue;
-              if (isnan(cs)) { cs = 0; csq = 0; }
-              if (i == b) last_val = x;       // lastSample carry across chunks
-              cs += x; csq += x * x; cc++;
-            }
-            if (!isnan(cs) && isnan(sum)) sum = 0;
-            sum += cs;
-            if (!isnan(csq) && isnan(sqsum)) sqsum = 0;
-            sqsum += csq;
-            icount += cc;
-          } else {  // K_CHANGES
-            if (isnan(changes)) changes = 0;
-            // in-chunk changes from the indicator prefix over (startRow, endRow]
-            double ch = (double)((int)ws.cnt[b] - (int)ws.cnt[a]);
-            // boundary: first sample vs carry from the previous chunk range
-            double vs = ws.val[a];
-            if (!isnan(vs) && !isnan(prev) && vs != prev) ch += 1;
-            changes += ch;
-            prev = ws.val[b];                       // carried raw (NaN possible)
-          }
-        }
-        if (m.end_time >= wEnd) break;
-      }
-      switch (FUNC) {
-        case FN_SUM:   result = sum; break;
-        case FN_COUNT: result = started ? (double)icount : NAN; break;
-        case FN_AVG:   result = icount > 0 ? sum / icount : (isnan(sum) ? sum : 0); break;
-        case FN_MIN: case FN_MAX: result = mm; break;
-        case FN_STDDEV: case FN_STDVAR: {
-          if (icount > 0) {
-            double avg = sum / icount;
-            double r = sqsum / icount - avg * avg;
-            result = (FUNC == FN_STDDEV) ? sqrt(r) : r;
-          } else result = isnan(sum) ? sum : 0;
-        } break;
-        case FN_CHANGES: result = changes; break;
-        case FN_LAST: case FN_PRESENT: case FN_TIMESTAMP:
-          result = last_val; break;
-        case FN_ZSCORE: {
-          // ZScoreChunkedFunctionD (AggrOverTimeFunctions.scala:1592-1603)
-          if (icount > 0) {
-            double avg = sum / icount;
-            double sd = sqrt(sqsum / icount - avg * avg);
-            result = (last_val - avg) / sd;
-          } else result = isnan(sum) ? sum : 0;
-        } break;
-      }
-    }
-
-    if (agg_id == AGG_NONE) {
-      out[(size_t)sid * num_windows + w] = result;
-    } else if (!isnan(result)) {
-      // fastReduce fused: RowAggregator merge per group cell
-      size_t cell = (size_t)grp_id * num_windows + w;
-      switch (agg_id) {
-        case AGG_SUM: case AGG_AVG:
-          atomicAdd(&out[cell], result);
-          atomicAdd(&out_cnt[cell], 1.0);
-          break;
-        case AGG_STDDEV: case AGG_STDVAR:
-          atomicAdd(&out[cell], result);
-          atomicAdd(&out_sq[cell], result * result);
-          atomicAdd(&out_cnt[cell], 1.0);
-          break;
-        case AGG_COUNT:
-          atomicAdd(&out[cell], 1.0);
-          atomicAdd(&out_cnt[cell], 1.0);
-          break;
-        case AGG_MIN: atomic_min_max_f64(&out[cell], result, true);
-                      atomicAdd(&out_cnt[cell], 1.0); break;
-        case AGG_MAX: atomic_min_max_f64(&out[cell], result, false);
-                      atomicAdd(&out_cnt[cell], 1.0); break;
-      }
-    }
-  }
-  // next series reuses this wave's LDS slot: wave-local ordering is enough
-  d_wait_lds();
-  __builtin_amdgcn_wave_barrier();
-  if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); t_win += t - tt0; tt0 = t; }
-  }  // series grid-stride loop
-  if ((phase_mask & 16) && threadIdx.x == 0)
-    atomicAdd((unsigned long long*)&out_cnt[0], (unsigned long long)-1ll);
-  if (timing && lane == 0) {
-    size_t gw = (size_t)blockIdx.x * WAVES_PER_BLOCK + wave;
-    out[gw * 4 + 0] = (double)t_decode;
-    out[gw * 4 + 1] = (double)t_meta;
-    out[gw * 4 + 2] = (double)t_win;
-    out[gw * 4 + 3] = 0;
-  }
-}
-
-
+// floor division helper used by the hist kernel below
 // ---------------------------------------------------------------------------
 // histogram scan kernel (BASELINE config #4)
 //   histogram_quantile(q, sum(rate(hist[w])) by group)
@@ -1301,6 +581,7 @@ struct fdb_dataset {
   int max_chunk_rows;       // max rows in one chunk (hist per-chunk LDS cap)
   int has_hist;             // dataset holds sect-delta histogram vectors
   int fast_ok;              // single-chunk series, chunk spans fit i32 ms
+  void* sums;               // ChunkSum[num_chunks] for the streaming walk
 };
 
 extern "C" fdb_engine_t* fdb_engine_create(int32_t device) {
@@ -1340,6 +621,7 @@ extern "C" void fdb_dataset_destroy(fdb_dataset_t* d) {
   (void)hipFree(d->start_time); (void)hipFree(d->end_time); (void)hipFree(d->num_rows);
   (void)hipFree(d->series_first); (void)hipFree(d->series_nchunks); (void)hipFree(d->group_ids);
   (void)hipFree(d->series_by_group); (void)hipFree(d->group_offsets);
+  (void)hipFree(d->sums);
   delete d;
 }
 
@@ -1389,12 +671,11 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
       total += dir[view.series_first[sid] + c].num_rows;
     if (total > max_rows) max_rows = total;
     if (view.series_nchunks[sid] > max_chunks) max_chunks = view.series_nchunks[sid];
-    if (total > 1600 || view.series_nchunks[sid] > 16) {
-      fdb_set_error("series %d exceeds capacity (%d rows / %d chunks; caps 1600/16 "
-                    "— longer lookbacks are round-2 streaming work)",
-                    sid, total, view.series_nchunks[sid]);
-      return nullptr;
-    }
+  }
+  if (max_chunk_rows > 400) {
+    fdb_set_error("chunk has %d rows; the reference's chunk cap is 400 "
+                  "(filodb-defaults.conf:835)", max_chunk_rows);
+    return nullptr;
   }
 
   auto* d = new fdb_dataset();
@@ -1450,6 +731,20 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
     fdb_dataset_destroy(d);
     return nullptr;
   }
+  // chunk summaries for the streaming walk (query-independent; skipped when
+  // every series takes the single-chunk fast path)
+  if (has_scalar && !d->fast_ok) {
+    if (hipMalloc(&d->sums, (size_t)fdb_chunksum_bytes(nc)) != hipSuccess) {
+      fdb_set_error("summary allocation failed (out of HBM?)");
+      fdb_dataset_destroy(d);
+      return nullptr;
+    }
+    DirSoA dirx{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
+    if (fdb_launch_summaries(e->stream, d->blob, dirx, nc, d->sums) != FDB_OK) {
+      fdb_dataset_destroy(d);
+      return nullptr;
+    }
+  }
   return d;
 }
 
@@ -1474,37 +769,21 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
                                 q->func_id, AGG_NONE, /*emit_group=*/0,
                                 dev_out, dev_cnt, dev_sq, q->_pad);
   }
-  // capacity tier: the common one-chunk shape runs 4 waves/block; series
-  // spanning several chunks (long lookbacks) use the 1600-row/2-wave variant
-  const bool large = d->max_rows > FDB_MAX_ROWS_PER_SERIES ||
-                     d->max_chunks > FDB_MAX_CHUNKS_PER_SERIES;
-  const int waves = large ? 2 : WAVES_PER_BLOCK;
-  int grid = (d->num_series + waves - 1) / waves;
-  // enough blocks to fill every CU at worst-case occupancy, few enough that
-  // workgroup dispatch is off the critical path
-  int cap = 8192;
-  if (const char* g = getenv("FDB_GRID")) cap = atoi(g);   // perf experiments
-  if (cap > 0 && grid > cap) grid = cap;
-  #define ARGS d->blob, dir, d->series_first, d->series_nchunks, d->group_ids, \
-      d->num_series, q->start, q->step, q->end, q->window, nw, q->agg_id, \
-      dev_out, dev_cnt, dev_sq, q->_pad == 0 ? 3 : q->_pad
-  #define CASE(F) case F: \
-    if (large) hipLaunchKernelGGL((scan_kernel<F, FDB_ROWS_LARGE, FDB_CHUNKS_LARGE, 2>), \
-                                  dim3(grid), dim3(128), 0, e->stream, ARGS); \
-    else hipLaunchKernelGGL((scan_kernel<F>), dim3(grid), dim3(BLOCK_THREADS), 0, \
-                            e->stream, ARGS); \
-    break
-  switch (q->func_id) {
-    CASE(FN_RATE); CASE(FN_INCREASE); CASE(FN_DELTA); CASE(FN_SUM); CASE(FN_COUNT);
-    CASE(FN_AVG); CASE(FN_MIN); CASE(FN_MAX); CASE(FN_STDDEV); CASE(FN_STDVAR);
-    CASE(FN_CHANGES); CASE(FN_LAST);
-    CASE(FN_PRESENT); CASE(FN_TIMESTAMP); CASE(FN_ZSCORE);
-    default: fdb_set_error("bad func_id %d", q->func_id); return FDB_ERR_BADARG;
+  // everything else: the unbounded summary+walk general path. It writes the
+  // per-series [S×W] grid only (aggregation is the two-phase reduce outside).
+  if (!fdb_stream_walk_supported(q->func_id)) {
+    fdb_set_error("bad func_id %d", q->func_id);
+    return FDB_ERR_BADARG;
   }
-  #undef CASE
-  #undef ARGS
-  HIP_CHECK(hipGetLastError());
-  return FDB_OK;
+  if (!d->sums) {
+    fdb_set_error("dataset has no chunk summaries (internal)");
+    return FDB_ERR;
+  }
+  (void)dev_cnt; (void)dev_sq;
+  return fdb_launch_stream_walk(e->stream, d->blob, dir, d->sums,
+                                d->series_first, d->series_nchunks,
+                                d->num_series, q->start, q->step, q->end,
+                                q->window, nw, q->func_id, dev_out);
 }
 
 static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
