@@ -92,6 +92,52 @@ def test_fused_group_partial_mode(fdb, oracle, engine):
     np.testing.assert_array_equal(got_c, want_c)
 
 
+def _long_store(fdb, n_series, n, kind, seed, reset_p=0.005, nan_p=0.0):
+    rng = np.random.default_rng(seed)
+    series = []
+    for _ in range(n_series):
+        if kind == "counter":
+            ts, vs = synth_counter_series(rng, n, reset_p=reset_p)
+        else:
+            ts, vs = synth_gauge_series(rng, n, nan_p=nan_p)
+        series.append([[(int(t), float(v)) for t, v in zip(ts, vs)]])
+    # one long append per series; the builder cuts 400-row chunks itself
+    return build_store(fdb, series,
+                       kind=fdb.COL_COUNTER if kind == "counter" else None)
+
+
+def test_24h_lookback_sum(fdb, oracle, engine):
+    """Round-1 caps (1600 rows / 16 chunks) lifted: 24h@15s = 5760 rows in
+    15 auto-cut chunks per series, sum_over_time[5m] step=15s, 5761 windows."""
+    st = _long_store(fdb, 12, 5760, "gauge", seed=241, nan_p=0.05)
+    assert st.num_chunks(0) == 15
+    q = fdb.make_query(100000, 15000, 100000 + 5760 * 15000, 300000,
+                       fdb.FN_SUM_OVER_TIME)
+    got, want = run_both(fdb, oracle, engine, st, q)
+    check(got, want)
+
+
+def test_24h_lookback_rate_1h_window(fdb, oracle, engine):
+    """rate[1h] step=15s across a 24h span (window/step ratio 240 — beyond
+    every round-1 tier), with counter resets crossing chunk boundaries."""
+    st = _long_store(fdb, 8, 5760, "counter", seed=242, reset_p=0.002)
+    q = fdb.make_query(100000, 15000, 100000 + 5760 * 15000, 3600_000,
+                       fdb.FN_RATE)
+    got, want = run_both(fdb, oracle, engine, st, q)
+    check(got, want)
+
+
+@pytest.mark.parametrize("func", ["avg", "min", "max", "stddev", "changes",
+                                  "count", "last", "zscore"])
+def test_long_series_func_matrix(fdb, oracle, engine, func):
+    """Gauge function matrix on 3h series (720 rows, 2 chunks) through the
+    streaming walk (summaries + boundary decodes)."""
+    st = _long_store(fdb, 16, 720, "gauge", seed=hash(func) % 2**31, nan_p=0.1)
+    q = mkq(fdb, FUNCS[func], end=Q["start"] + 600 * 15000)
+    got, want = run_both(fdb, oracle, engine, st, q)
+    check(got, want)
+
+
 def test_single_row_and_empty_edge(fdb, oracle, engine):
     """1-row chunks and windows entirely before/after the data."""
     series = [
