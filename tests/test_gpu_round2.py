@@ -10,7 +10,8 @@ import numpy as np
 import pytest
 
 from conftest import build_store, synth_counter_series, synth_gauge_series
-from test_gpu_parity import counter_store, gauge_store, mkq, run_both, check, FUNCS
+from test_gpu_parity import (counter_store, gauge_store, mkq, run_both,
+                             check, FUNCS, AGGS, Q)
 
 pytestmark = pytest.mark.gpu
 
@@ -53,7 +54,6 @@ def test_rate_over_delta_gpu(fdb, oracle, engine):
 def test_fused_group_matches_two_phase(fdb, oracle, engine, agg):
     """The fused-group emit (no [S×W] intermediate) against the two-phase
     scan→group_reduce path AND the oracle, same store and query."""
-    from test_gpu_parity import AGGS
     rng = np.random.default_rng(hash(agg) % 2**31)
     n_groups = 9
     series, groups = [], []
@@ -170,3 +170,20 @@ def test_duplicate_ts_pinned_semantics(fdb, engine):
     assert np.isnan(got[0:4]).all()
     assert got[4] == 3.0
     assert got[8] == 7.0
+
+
+def test_multichunk_group_aggregation(fdb, oracle, engine):
+    """sum by(group)(rate) where series span chunks: walk kernel feeding the
+    two-phase group reduce."""
+    rng = np.random.default_rng(55)
+    n_groups = 6
+    series, groups = [], []
+    for s in range(90):
+        ts, vs = synth_counter_series(rng, 720, reset_p=0.01)
+        series.append([[(int(t), float(v)) for t, v in zip(ts, vs)]])
+        groups.append(s % n_groups)
+    st = build_store(fdb, series, groups=groups, kind=fdb.COL_COUNTER)
+    q = mkq(fdb, FUNCS["rate"], AGGS["sum"], n_groups,
+            end=Q["start"] + 500 * 15000)
+    got, want = run_both(fdb, oracle, engine, st, q)
+    check(got, want)
