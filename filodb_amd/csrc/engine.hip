@@ -48,6 +48,15 @@ int32_t fdb_launch_summaries(hipStream_t stream, const uint8_t* blob, DirSoA dir
                              int64_t num_chunks, void* sums);
 int64_t fdb_chunksum_bytes(int64_t num_chunks);
 bool fdb_stream_walk_supported(int func_id);
+
+// hist2.hip: two-cursor histogram walk (unbounded chunks / window ratio)
+int32_t fdb_launch_hist2(hipStream_t stream, const uint8_t* blob, DirSoA dir,
+                         const int32_t* series_first,
+                         const int32_t* series_nchunks,
+                         const int32_t* group_ids, int num_series,
+                         int64_t qstart, int64_t qstep, int64_t qwindow,
+                         int num_windows, int nb,
+                         double* out_sums, double* out_cnt);
 int32_t fdb_launch_stream_walk(hipStream_t stream, const uint8_t* blob,
                                DirSoA dir, const void* sums,
                                const int32_t* series_first,
@@ -101,69 +110,6 @@ __device__ __forceinline__ int64_t d_fdiv_fast(int64_t a, int64_t b, double inv_
   int64_t w = (int64_t)floor((double)a * inv_b);
   while ((w + 1) * b <= a) w++;
   while (w * b > a) w--;
-  return w;
-}
-
-// wave-wide inclusive prefix sum over i64 (bucket-cumulative reconstruction)
-__device__ __forceinline__ int64_t wave_incl_scan_i64(int64_t x, int lane) {
-  for (int off = 1; off < 64; off <<= 1) {
-    int64_t t = __shfl_up(x, off);
-    if (lane >= off) x += t;
-  }
-  return x;
-}
-
-// --- wave-staged element stream -------------------------------------------
-// One coalesced dword-per-lane load stages 256 B of an element's NibblePack
-// stream into the wave's registers; all subsequent byte/u16/u64 reads are
-// register shuffles (ds_bpermute) instead of dependent global loads. The
-// stage for element e+1 is issued before element e is parsed, so the load
-// latency overlaps a full element's parse work. Elements longer than the
-// staged range (elen+shift > 252, i.e. huge sections) fall back to direct
-// global reads; the blob is tail-padded 256 B at upload so staging never
-// faults.
-__device__ __forceinline__ uint32_t estream_stage(const uint8_t* p, int lane) {
-  const uint8_t* base = (const uint8_t*)((uintptr_t)p & ~(uintptr_t)3);
-  uint32_t d;
-  memcpy(&d, base + lane * 4, 4);
-  return d;
-}
-// `staged` must be WAVE-UNIFORM and every call site must have the full wave
-// active: __shfl is ds_bpermute, and a source lane that is inactive (or a
-// divergent caller) yields undefined data.
-__device__ __forceinline__ uint32_t estream_byte(bool staged, uint32_t buf,
-                                                 int shift, const uint8_t* g,
-                                                 int k) {
-  if (staged) {
-    int kk = k + shift;
-    uint32_t d = __shfl(buf, kk >> 2);
-    return (d >> ((kk & 3) * 8)) & 0xff;
-  }
-  return g[k];
-}
-__device__ __forceinline__ uint32_t estream_u16(bool staged, uint32_t buf,
-                                                int shift, const uint8_t* g,
-                                                int k) {
-  return estream_byte(staged, buf, shift, g, k) |
-         (estream_byte(staged, buf, shift, g, k + 1) << 8);
-}
-// little-endian 8-byte window at offset k (per-lane k; bpermute shuffles)
-__device__ __forceinline__ uint64_t estream_w64(bool staged, uint32_t buf,
-                                                int shift, const uint8_t* g,
-                                                int k) {
-  if (staged) {
-    int kk = k + shift;
-    int dw = kk >> 2;
-    uint64_t a = __shfl(buf, dw);
-    uint64_t b = __shfl(buf, dw + 1);
-    uint64_t c = __shfl(buf, dw + 2);
-    int sh = (kk & 3) * 8;
-    uint64_t w = (a | (b << 32)) >> sh;
-    if (sh) w |= c << (64 - sh);
-    return w;
-  }
-  uint64_t w;
-  memcpy(&w, g + k, 8);
   return w;
 }
 
@@ -969,19 +915,16 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
     fdb_set_error("not a histogram dataset");
     return FDB_ERR_BADARG;
   }
-  if (d->max_chunks > FDB_HIST_MAX_CHUNKS) {
-    fdb_set_error("histogram series span %d chunks; the streaming walk handles "
-                  "up to %d", d->max_chunks, FDB_HIST_MAX_CHUNKS);
-    return FDB_ERR_BADARG;
-  }
   if (d->max_chunk_rows > FDB_MAX_ROWS_PER_SERIES) {
     fdb_set_error("histogram chunk has %d rows; per-chunk cap is %d",
                   d->max_chunk_rows, FDB_MAX_ROWS_PER_SERIES);
     return FDB_ERR_BADARG;
   }
-  if (q->window / q->step + 2 > FDB_HIST_RING) {
-    fdb_set_error("window/step ratio %lld exceeds round-1 ring capacity %d",
-                  (long long)(q->window / q->step), FDB_HIST_RING);
+  const char* hv1 = getenv("FDB_HIST_V1");   // round-1 kernel for A/B runs
+  const bool use_v1 = hv1 && atoi(hv1) == 1;
+  if (use_v1 && (d->max_chunks > FDB_HIST_MAX_CHUNKS ||
+                 q->window / q->step + 2 > FDB_HIST_RING)) {
+    fdb_set_error("FDB_HIST_V1 set but the query exceeds the v1 ring caps");
     return FDB_ERR_BADARG;
   }
   if (d->max_group >= q->num_groups) { fdb_set_error("num_groups too small"); return FDB_ERR_BADARG; }
@@ -1006,7 +949,13 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
   if (const char* g = getenv("FDB_HIST_GRID")) hcap = atoi(g);  // perf experiments
   if (hcap > 0 && grid > hcap) grid = hcap;
   int dbg = getenv("FDB_HIST_TIME") ? 1 : 0;
-  if (d->max_chunks > 1)
+  if (!use_v1) {
+    int32_t rc = fdb_launch_hist2(e->stream, d->blob, dir, d->series_first,
+                                  d->series_nchunks, d->group_ids,
+                                  d->num_series, q->start, q->step, q->window,
+                                  nw, nb, dev_sums, dev_cnt);
+    if (rc != FDB_OK) return rc;
+  } else if (d->max_chunks > 1)
     hipLaunchKernelGGL(hist_scan_kernel<FDB_HIST_MAX_CHUNKS>, dim3(grid),
                        dim3(HIST_WAVES * 64), 0, e->stream,
                        d->blob, dir, d->series_first, d->series_nchunks, d->group_ids,

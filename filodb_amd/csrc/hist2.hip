@@ -1,0 +1,280 @@
+// Histogram scan v2: two-cursor monotonic walk over the sect-delta streams.
+//
+// Round 1 streamed every element once, holding per-bucket window-start values
+// in a ring of active windows — which capped window/step at 22 and chunks at
+// 4, and serialized on the per-element header parse (43.7 ms for the
+// BASELINE config #4 workload; profiles/README.md). v2 removes the ring:
+//
+//   rate(hist[w]) per window needs only the FIRST element >= wStart and the
+//   LAST element <= wEnd (HistogramRateFunctionBase, RateFunctions.scala:
+//   330-400). Windows are processed in order by ONE wave per series with two
+//   monotone cursors — S tracks the window-start element, E the window-end
+//   element. A cursor hops elements by their u16 length prefix and fully
+//   decodes an element only when it stops on it (or at section bases, since
+//   sect-delta diffs are against the section's first element —
+//   HistogramVector.scala:491-545) — so the dependent header-parse chain runs
+//   ~twice per element worst case, with the two cursors' chains independent
+//   and overlapping, at far higher occupancy (no 12 KB ring in LDS).
+//
+// Unbounded: chunks per series, window/step ratio, num_windows. Per-chunk
+// rows <= 400 (the per-cursor LDS timestamp buffer; the reference's own
+// chunk cap). Corrections: TypeDrop sections add apply(e-1) per bucket
+// (Section.scala:17-24 type byte); chunk-boundary drops use
+// Histogram.compare's top-bucket-down order (Histogram.scala:204-214) —
+// same rules and per-window correction base as v1 (parity-green).
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <cstring>
+#include <cmath>
+
+#include "chunk_format.h"
+#include "scan_common.h"
+#include "../../include/filodb_amd.h"
+
+void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
+
+#define H2_ROWS 400
+#define H2_WAVES 4
+
+// parse one element's per-lane bucket delta from its NibblePack stream.
+// ep = element start (at the u16 length); wave-uniform. All lanes must be
+// active (estream shuffles). Returns the lane's (<<trailing-zeroes) delta.
+__device__ __forceinline__ int64_t h2_parse(const uint8_t* ep, int elen,
+                                            int nb, int b, bool live,
+                                            int lane) {
+  const uint32_t ebuf = estream_stage(ep, lane);
+  const int eshift = (int)((uintptr_t)ep & 3);
+  const bool est = elen + 14 + eshift <= 256;
+  // 8-value group headers, walked serially (wave-uniform)
+  const int my_group = b >> 3;
+  int off = 0, gOff = 0, gBits = 0, gTrail = 0;
+  uint32_t gMask = 0;
+  for (int g = 0; g * 8 < nb; g++) {
+    uint32_t mask = estream_byte(est, ebuf, eshift, ep, 2 + off);
+    int numBits = 0, trail = 0, glen;
+    if (mask == 0) {
+      glen = 1;
+    } else {
+      int widths = (int)estream_byte(est, ebuf, eshift, ep, 2 + off + 1);
+      numBits = ((widths >> 4) + 1) * 4;
+      trail = (widths & 0x0f) * 4;
+      glen = 2 + (numBits * __popc(mask) + 7) / 8;
+    }
+    if (g == my_group) { gOff = off; gBits = numBits; gTrail = trail; gMask = mask; }
+    off += glen;
+  }
+  const int bit = b & 7;
+  const uint32_t in_mask = gMask & (1u << bit);
+  int slot = __popc(gMask & ((1u << bit) - 1));
+  int bitpos = slot * gBits;
+  int koff = 2 + gOff + 2 + (bitpos >> 3);
+  uint64_t w64 = estream_w64(est, ebuf, eshift, ep, koff);
+  uint32_t b8 = estream_byte(est, ebuf, eshift, ep, koff + 8);
+  int64_t delta = 0;
+  if (live && in_mask) {
+    int sh = bitpos & 7;
+    uint64_t v = w64 >> sh;
+    if (gBits > 64 - sh) v |= (uint64_t)b8 << (64 - sh);
+    uint64_t m = gBits >= 64 ? ~0ULL : ((1ULL << gBits) - 1);
+    delta = (int64_t)((v & m) << gTrail);
+  }
+  return delta;
+}
+
+struct H2Cursor {
+  const uint8_t* ep;     // current element start (u16 len prefix)
+  const uint8_t* sp;     // NEXT section header
+  int elen;              // current element's payload length
+  int sect_left;         // elements left in section AFTER the current one
+  int c;                 // chunk index within the series
+  int e_local, e_global; // element indices (current)
+  int nrows;             // rows in current chunk
+  bool decoded;          // val_b holds the current element
+  bool sect_first;       // current element is its section's base
+  double base_b;         // per-lane section base value
+  double val_b;          // per-lane raw value of current element (if decoded)
+  double C_b;            // per-lane correction total at current position
+  double Centry_b;       // C at current chunk entry (after boundary drop)
+  double prevlast_b;     // raw value of the previous chunk's last element
+};
+
+template <int UNUSED = 0>
+__global__ __launch_bounds__(H2_WAVES * 64, 5)
+void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
+                  const int32_t* __restrict__ series_first,
+                  const int32_t* __restrict__ series_nchunks,
+                  const int32_t* __restrict__ group_ids,
+                  int num_series,
+                  int64_t qstart, int64_t qstep, int64_t qwindow,
+                  int num_windows, int nb,
+                  double* __restrict__ out_sums,   // [G × W × nb]
+                  double* __restrict__ out_cnt) {  // [G × W]
+  __shared__ int64_t tsS_all[H2_WAVES][H2_ROWS];
+  __shared__ int64_t tsE_all[H2_WAVES][H2_ROWS];
+  const int wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  const int b = lane;
+  const bool live = b < nb;
+
+  for (int sid = blockIdx.x * H2_WAVES + wave; sid < num_series;
+       sid += gridDim.x * H2_WAVES) {
+    const int first = series_first[sid];
+    const int nchunks = series_nchunks[sid];
+    if (nchunks < 1) continue;
+    const int grp = group_ids[sid];
+
+    // open a chunk for a cursor: decode its timestamps into the cursor's LDS
+    // buffer and position at element 0 (section base, decoded)
+    auto open_chunk = [&](H2Cursor& cu, int c, int64_t* tsbuf) {
+      const uint8_t* hv = blob + dir.val_off[first + c];
+      DVec tv;
+      d_vec_open_wide(blob + dir.ts_off[first + c], &tv, nullptr);
+      cu.nrows = dir.num_rows[first + c];
+      if (cu.nrows > H2_ROWS || cu.nrows > tv.n) cu.nrows = 0;
+      d_decode_chunk<false>(tv, cu.nrows, tsbuf, nullptr, lane);
+      d_wait_lds();
+      __builtin_amdgcn_wave_barrier();
+      const int defsz = d_u16(hv + FDB_HIST_OFF_DEFSIZE);
+      cu.sp = hv + FDB_HIST_OFF_DEF + defsz;
+      cu.c = c;
+      cu.e_local = -1;
+      cu.sect_left = 0;
+      cu.decoded = false;
+    };
+
+    // decode the current element (value = section base + scanned deltas;
+    // a section base element's value is the scan of its own deltas)
+    auto decode_cur = [&](H2Cursor& cu) {
+      if (cu.decoded) return;
+      int64_t delta = h2_parse(cu.ep, cu.elen, nb, b, live, lane);
+      int64_t scan = wave_incl_scan_i64(live ? delta : 0, lane);
+      if (cu.sect_first) { cu.val_b = (double)scan; cu.base_b = cu.val_b; }
+      else cu.val_b = cu.base_b + (double)scan;
+      cu.decoded = true;
+    };
+
+    // step to the next element; returns false when the series is exhausted.
+    // tsbuf is the cursor's chunk-timestamp LDS buffer.
+    auto step = [&](H2Cursor& cu, int64_t* tsbuf) -> bool {
+      if (cu.e_local + 1 >= cu.nrows) {
+        if (cu.c + 1 >= nchunks) return false;
+        // leaving a chunk: its last element's raw value feeds the boundary
+        // drop detection (and TypeDrop at the next chunk's head)
+        decode_cur(cu);
+        cu.prevlast_b = cu.val_b;
+        open_chunk(cu, cu.c + 1, tsbuf);
+      }
+      bool new_sect = false;
+      if (cu.sect_left == 0) {
+        // entering a section; TypeDrop adds apply(e-1) (the element we just
+        // ensured is decoded below / at the chunk seam above)
+        int stype = cu.sp[3];
+        cu.sect_left = cu.sp[2];
+        cu.ep = cu.sp + 4;
+        cu.sp += 4 + d_u16(cu.sp);
+        new_sect = true;
+        if (stype == 1 && cu.e_local >= 0) cu.C_b += cu.val_b;
+      } else {
+        cu.ep += 2 + cu.elen;
+      }
+      cu.elen = (int)d_u16(cu.ep);
+      cu.sect_left--;
+      cu.e_local++;
+      cu.e_global++;
+      cu.sect_first = new_sect;
+      cu.decoded = false;
+      if (new_sect) decode_cur(cu);           // section base always decoded
+      else if (cu.sect_left == 0) decode_cur(cu);  // section last: TypeDrop feed
+      if (cu.e_local == 0) {
+        // first element of a chunk: boundary drop detection
+        // (Histogram.compare top-bucket-down, Histogram.scala:204-214)
+        if (cu.c > 0) {
+          unsigned long long diff = __ballot(live && cu.val_b != cu.prevlast_b);
+          if (diff) {
+            int L = 63 - __clzll(diff);
+            int lt = __shfl((int)(cu.val_b < cu.prevlast_b), L);
+            if (lt) cu.C_b += cu.prevlast_b;
+          }
+        }
+        cu.Centry_b = cu.C_b;
+      }
+      return true;
+    };
+
+    // initialize both cursors at element 0
+    H2Cursor S, E;
+    memset(&S, 0, sizeof(S)); memset(&E, 0, sizeof(E));
+    S.e_global = -1; E.e_global = -1;
+    S.C_b = 0; E.C_b = 0;
+    open_chunk(S, 0, tsS_all[wave]);
+    if (S.nrows == 0) continue;
+    if (!step(S, tsS_all[wave])) continue;
+    open_chunk(E, 0, tsE_all[wave]);
+    step(E, tsE_all[wave]);
+    bool s_more = true, e_more = true;
+
+    for (int w = 0; w < num_windows; w++) {
+      const int64_t wEnd = qstart + (int64_t)w * qstep;
+      const int64_t wStart = wEnd - qwindow;
+      // E → last element with ts <= wEnd
+      for (;;) {
+        int64_t nxt;
+        if (E.e_local + 1 < E.nrows) nxt = tsE_all[wave][E.e_local + 1];
+        else if (E.c + 1 < nchunks) nxt = dir.start_time[first + E.c + 1];
+        else break;
+        if (nxt > wEnd) break;
+        if (!step(E, tsE_all[wave])) { e_more = false; break; }
+      }
+      // S → first element with ts >= wStart
+      while (s_more && tsS_all[wave][S.e_local] < wStart) {
+        if (!step(S, tsS_all[wave])) { s_more = false; break; }
+      }
+      (void)e_more;
+      if (!s_more) break;                     // no element >= wStart: done
+      const int64_t t1 = tsS_all[wave][S.e_local];
+      const int64_t t2 = tsE_all[wave][E.e_local];
+      if (t1 > wEnd || t2 < wStart) continue; // empty window
+      if (!(t2 > t1)) continue;               // highestTime > lowestTime rule
+      decode_cur(S);
+      decode_cur(E);
+      const int numSamples = E.e_global - S.e_global + 1;
+      if (live) {
+        // corrections relative to the window's first chunk: the reference's
+        // per-window CorrectionMeta starts NoCorrection there
+        double v1 = S.val_b + (S.C_b - S.Centry_b);
+        double v2 = E.val_b + (E.C_b - S.Centry_b);
+        double r = d_extrapolated_rate(wStart, wEnd, numSamples,
+                                       t1, v1, t2, v2, true, true);
+        atomicAdd(&out_sums[((size_t)grp * num_windows + w) * nb + b], r);
+      }
+      if (lane == 0)
+        atomicAdd(&out_cnt[(size_t)grp * num_windows + w], 1.0);
+    }
+    d_wait_lds();
+    __builtin_amdgcn_wave_barrier();
+  }
+}
+
+int32_t fdb_launch_hist2(hipStream_t stream, const uint8_t* blob, DirSoA dir,
+                         const int32_t* series_first,
+                         const int32_t* series_nchunks,
+                         const int32_t* group_ids, int num_series,
+                         int64_t qstart, int64_t qstep, int64_t qwindow,
+                         int num_windows, int nb,
+                         double* out_sums, double* out_cnt) {
+  int grid = (num_series + H2_WAVES - 1) / H2_WAVES;
+  int cap = 8192;
+  if (const char* g = getenv("FDB_HIST_GRID")) cap = atoi(g);
+  if (cap > 0 && grid > cap) grid = cap;
+  hipLaunchKernelGGL((hist2_kernel<0>), dim3(grid), dim3(H2_WAVES * 64), 0,
+                     stream, blob, dir, series_first, series_nchunks,
+                     group_ids, num_series, qstart, qstep, qwindow,
+                     num_windows, nb, out_sums, out_cnt);
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) {
+    fdb_set_error("hist2_kernel launch failed: %s", hipGetErrorString(e));
+    return FDB_ERR;
+  }
+  return FDB_OK;
+}

@@ -343,4 +343,68 @@ __device__ __forceinline__ int fdiv_floor_win(int64_t a, int64_t b, double inv_b
   return (int)w;
 }
 
+// wave-wide inclusive prefix sum over i64 (bucket-cumulative reconstruction)
+__device__ __forceinline__ int64_t wave_incl_scan_i64(int64_t x, int lane) {
+  for (int off = 1; off < 64; off <<= 1) {
+    int64_t t = __shfl_up(x, off);
+    if (lane >= off) x += t;
+  }
+  return x;
+}
+
+// --- wave-staged element stream -------------------------------------------
+// One coalesced dword-per-lane load stages 256 B of an element's NibblePack
+// stream into the wave's registers; all subsequent byte/u16/u64 reads are
+// register shuffles (ds_bpermute) instead of dependent global loads. The
+// stage for element e+1 is issued before element e is parsed, so the load
+// latency overlaps a full element's parse work. Elements longer than the
+// staged range (elen+shift > 252, i.e. huge sections) fall back to direct
+// global reads; the blob is tail-padded 256 B at upload so staging never
+// faults.
+__device__ __forceinline__ uint32_t estream_stage(const uint8_t* p, int lane) {
+  const uint8_t* base = (const uint8_t*)((uintptr_t)p & ~(uintptr_t)3);
+  uint32_t d;
+  memcpy(&d, base + lane * 4, 4);
+  return d;
+}
+// `staged` must be WAVE-UNIFORM and every call site must have the full wave
+// active: __shfl is ds_bpermute, and a source lane that is inactive (or a
+// divergent caller) yields undefined data.
+__device__ __forceinline__ uint32_t estream_byte(bool staged, uint32_t buf,
+                                                 int shift, const uint8_t* g,
+                                                 int k) {
+  if (staged) {
+    int kk = k + shift;
+    uint32_t d = __shfl(buf, kk >> 2);
+    return (d >> ((kk & 3) * 8)) & 0xff;
+  }
+  return g[k];
+}
+__device__ __forceinline__ uint32_t estream_u16(bool staged, uint32_t buf,
+                                                int shift, const uint8_t* g,
+                                                int k) {
+  return estream_byte(staged, buf, shift, g, k) |
+         (estream_byte(staged, buf, shift, g, k + 1) << 8);
+}
+// little-endian 8-byte window at offset k (per-lane k; bpermute shuffles)
+__device__ __forceinline__ uint64_t estream_w64(bool staged, uint32_t buf,
+                                                int shift, const uint8_t* g,
+                                                int k) {
+  if (staged) {
+    int kk = k + shift;
+    int dw = kk >> 2;
+    uint64_t a = __shfl(buf, dw);
+    uint64_t b = __shfl(buf, dw + 1);
+    uint64_t c = __shfl(buf, dw + 2);
+    int sh = (kk & 3) * 8;
+    uint64_t w = (a | (b << 32)) >> sh;
+    if (sh) w |= c << (64 - sh);
+    return w;
+  }
+  uint64_t w;
+  memcpy(&w, g + k, 8);
+  return w;
+}
+
+
 #endif  // FDB_SCAN_COMMON_H
