@@ -99,8 +99,8 @@ struct H2Cursor {
   double prevlast_b;     // raw value of the previous chunk's last element
 };
 
-template <int UNUSED = 0>
-__global__ __launch_bounds__(H2_WAVES * 64, 5)
+template <int MINW>
+__global__ __launch_bounds__(H2_WAVES * 64, MINW)
 void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                   const int32_t* __restrict__ series_first,
                   const int32_t* __restrict__ series_nchunks,
@@ -267,10 +267,17 @@ int32_t fdb_launch_hist2(hipStream_t stream, const uint8_t* blob, DirSoA dir,
   int cap = 8192;
   if (const char* g = getenv("FDB_HIST_GRID")) cap = atoi(g);
   if (cap > 0 && grid > cap) grid = cap;
-  hipLaunchKernelGGL((hist2_kernel<0>), dim3(grid), dim3(H2_WAVES * 64), 0,
-                     stream, blob, dir, series_first, series_nchunks,
-                     group_ids, num_series, qstart, qstep, qwindow,
-                     num_windows, nb, out_sums, out_cnt);
+  const char* hw = getenv("FDB_HIST_WAVES");   // occupancy experiment knob
+  if (hw && atoi(hw) == 5)
+    hipLaunchKernelGGL((hist2_kernel<5>), dim3(grid), dim3(H2_WAVES * 64), 0,
+                       stream, blob, dir, series_first, series_nchunks,
+                       group_ids, num_series, qstart, qstep, qwindow,
+                       num_windows, nb, out_sums, out_cnt);
+  else
+    hipLaunchKernelGGL((hist2_kernel<4>), dim3(grid), dim3(H2_WAVES * 64), 0,
+                       stream, blob, dir, series_first, series_nchunks,
+                       group_ids, num_series, qstart, qstep, qwindow,
+                       num_windows, nb, out_sums, out_cnt);
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) {
     fdb_set_error("hist2_kernel launch failed: %s", hipGetErrorString(e));
