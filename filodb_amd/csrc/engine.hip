@@ -49,6 +49,15 @@ int32_t fdb_launch_summaries(hipStream_t stream, const uint8_t* blob, DirSoA dir
 int64_t fdb_chunksum_bytes(int64_t num_chunks);
 bool fdb_stream_walk_supported(int func_id);
 
+// window_sample.hip: per-(series,window) sample-buffer functions (FN 16-18)
+bool fdb_window_sample_supported(int func_id);
+int32_t fdb_launch_window_sample(hipStream_t stream, const uint8_t* blob,
+                                 DirSoA dir, const int32_t* series_first,
+                                 const int32_t* series_nchunks, int num_series,
+                                 int64_t qstart, int64_t qstep, int64_t qwindow,
+                                 int num_windows, int func_id, double param,
+                                 double* out, int32_t* overflow);
+
 // hist2.hip: two-cursor histogram walk (unbounded chunks / window ratio)
 int32_t fdb_launch_hist2(hipStream_t stream, const uint8_t* blob, DirSoA dir,
                          const int32_t* series_first,
@@ -508,6 +517,7 @@ __global__ void group_reduce_kernel(const double* __restrict__ grid,
 struct fdb_engine {
   int device;
   hipStream_t stream;
+  int32_t* dev_flag;        // sample-kernel overflow flag (device)
 };
 
 struct fdb_dataset {
@@ -547,11 +557,19 @@ extern "C" fdb_engine_t* fdb_engine_create(int32_t device) {
     delete eng;
     return nullptr;
   }
+  if (hipMalloc(&eng->dev_flag, 4) != hipSuccess ||
+      hipMemset(eng->dev_flag, 0, 4) != hipSuccess) {
+    fdb_set_error("engine flag allocation failed");
+    (void)hipStreamDestroy(eng->stream);
+    delete eng;
+    return nullptr;
+  }
   return eng;
 }
 
 extern "C" void fdb_engine_destroy(fdb_engine_t* e) {
   if (!e) return;
+  (void)hipFree(e->dev_flag);
   (void)hipStreamDestroy(e->stream);
   delete e;
 }
@@ -714,6 +732,14 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
                                 q->start, q->step, q->window, nw,
                                 q->func_id, AGG_NONE, /*emit_group=*/0,
                                 dev_out, dev_cnt, dev_sq, q->_pad);
+  }
+  // sample-buffer functions (quantile/MAD/predict_linear) have their own
+  // per-(series,window) kernel, on any dataset shape
+  if (fdb_window_sample_supported(q->func_id)) {
+    return fdb_launch_window_sample(e->stream, d->blob, dir, d->series_first,
+                                    d->series_nchunks, d->num_series,
+                                    q->start, q->step, q->window, nw,
+                                    q->func_id, q->param, dev_out, e->dev_flag);
   }
   // everything else: the unbounded summary+walk general path. It writes the
   // per-series [S×W] grid only (aggregation is the two-phase reduce outside).
@@ -892,6 +918,16 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
     HIP_CHECK(hipGetLastError());
   }
   HIP_CHECK(hipStreamSynchronize(e->stream));
+  if (fdb_window_sample_supported(q->func_id)) {
+    int32_t flag = 0;
+    HIP_CHECK(hipMemcpy(&flag, e->dev_flag, 4, hipMemcpyDeviceToHost));
+    if (flag) {
+      (void)hipMemset(e->dev_flag, 0, 4);
+      fdb_set_error("a window exceeded the %d-sample buffer of the "
+                    "quantile/MAD kernel", 1024);
+      return FDB_ERR;
+    }
+  }
 
   if (!out_on_device) {
     HIP_CHECK(hipMemcpy(out, dev_out, buf_len * 8, hipMemcpyDeviceToHost));

@@ -187,3 +187,31 @@ def test_multichunk_group_aggregation(fdb, oracle, engine):
             end=Q["start"] + 500 * 15000)
     got, want = run_both(fdb, oracle, engine, st, q)
     check(got, want)
+
+
+@pytest.mark.parametrize("func,param", [("quantile", 0.9), ("quantile", 0.0),
+                                        ("quantile", 1.0), ("mad", 0.0),
+                                        ("predict", 600.0)])
+def test_sample_funcs_gpu(fdb, oracle, engine, func, param):
+    """FN 16/17/18 (quantile_over_time / MAD / predict_linear) through the
+    per-(series,window) sample kernel vs the oracle, single- and multi-chunk."""
+    fid = {"quantile": fdb.FN_QUANTILE_OVER_TIME, "mad": fdb.FN_MAD_OVER_TIME,
+           "predict": fdb.FN_PREDICT_LINEAR}[func]
+    for chunking in [(240,), (100, 100, 40)]:
+        st = gauge_store(fdb, n_series=24, seed=hash((func, chunking)) % 2**31,
+                         nan_p=0.15, chunking=chunking)
+        q = fdb.make_query(Q["start"], 15000, Q["start"] + 120 * 15000,
+                           600000, fid)
+        q.param = param
+        got, want = run_both(fdb, oracle, engine, st, q)
+        check(got, want)
+
+
+def test_quantile_out_of_range_params(fdb, oracle, engine):
+    """q<0 → -inf, q>1 → +inf on touched windows (oracle rule)."""
+    st = gauge_store(fdb, n_series=4, seed=7, nan_p=0.0)
+    for p in (-0.5, 1.5):
+        q = mkq(fdb, fdb.FN_QUANTILE_OVER_TIME)
+        q.param = p
+        got, want = run_both(fdb, oracle, engine, st, q)
+        check(got, want)
