@@ -21,13 +21,14 @@ FN_STDDEV_OVER_TIME, FN_STDVAR_OVER_TIME, FN_CHANGES = 8, 9, 10
 FN_HIST_RATE = 11
 FN_LAST = 12
 FN_PRESENT, FN_TIMESTAMP, FN_ZSCORE = 13, 14, 15
-FN_QUANTILE_OVER_TIME, FN_MAD_OVER_TIME = 16, 17   # oracle-only (kernel: round 2)
-FN_PREDICT_LINEAR = 18                             # oracle-only (kernel: round 2)
-FN_RATE_OVER_DELTA = 19                            # oracle-only (kernel: round 2)
+FN_QUANTILE_OVER_TIME, FN_MAD_OVER_TIME = 16, 17
+FN_PREDICT_LINEAR = 18
+FN_RATE_OVER_DELTA = 19
 # aggregation ids (RowAggregator implementations)
 AGG_NONE, AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX, AGG_AVG = 0, 1, 2, 3, 4, 5
 AGG_TOPK, AGG_BOTTOMK = 6, 7
 AGG_STDDEV, AGG_STDVAR, AGG_GROUP = 8, 9, 10
+AGG_QUANTILE, AGG_COUNT_VALUES = 11, 12
 # column kinds
 COL_GAUGE, COL_COUNTER, COL_HIST = 0, 1, 2
 
@@ -134,6 +135,10 @@ def lib():
         L.fdb_query_bench.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
                                       _c_double_p, _c_double_p, ctypes.c_int32,
                                       ctypes.c_int32, ctypes.c_int32, _c_double_p]
+        L.fdb_query_exec_count_values.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
+            ctypes.c_int32, _c_double_p, _c_double_p,
+            ctypes.POINTER(ctypes.c_int32)]
         _lib = L
     return _lib
 
@@ -285,6 +290,22 @@ class Engine:
             _as_f64_ptr(out_bucket_sums), _as_f64_ptr(out_counts),
             _as_f64_ptr(out_quantile), 1 if on_device else 0)
         _check(rc, "query_exec_hist")
+
+    def count_values(self, dataset, q: Query, k_cap=64):
+        """CountValuesRowAggregator: per (group, window) distinct values with
+        frequencies, sorted ascending. Returns (values, counts, n) with
+        values/counts shaped [G*W, k_cap] and n [G*W]."""
+        nw = q.num_windows
+        cells = q.num_groups * nw
+        vals = np.zeros(cells * k_cap, dtype=np.float64)
+        cnts = np.zeros(cells * k_cap, dtype=np.float64)
+        n = np.zeros(cells, dtype=np.int32)
+        rc = lib().fdb_query_exec_count_values(
+            self._h, dataset._h, ctypes.byref(q), k_cap,
+            _as_f64_ptr(vals), _as_f64_ptr(cnts),
+            n.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)))
+        _check(rc, "count_values")
+        return vals.reshape(cells, k_cap), cnts.reshape(cells, k_cap), n
 
     def synchronize(self):
         _check(lib().fdb_engine_synchronize(self._h), "synchronize")

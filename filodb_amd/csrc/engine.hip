@@ -51,6 +51,16 @@ bool fdb_stream_walk_supported(int func_id);
 
 // window_sample.hip: per-(series,window) sample-buffer functions (FN 16-18)
 bool fdb_window_sample_supported(int func_id);
+// agg_extra.hip: t-digest quantile + count_values cell presenters
+int32_t fdb_launch_quantile_cells(hipStream_t stream, const double* grid,
+                                  const int32_t* sbg, const int32_t* goff,
+                                  int ng, int nw, double q, double* out);
+int32_t fdb_launch_count_values(hipStream_t stream, const double* grid,
+                                const int32_t* sbg, const int32_t* goff,
+                                int ng, int nw, int k_cap,
+                                double* out_vals, double* out_cnts,
+                                int32_t* out_n, int32_t* overflow);
+
 int32_t fdb_launch_window_sample(hipStream_t stream, const uint8_t* blob,
                                  DirSoA dir, const int32_t* series_first,
                                  const int32_t* series_nchunks, int num_series,
@@ -837,8 +847,15 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   // never materializes the [S×W] grid; other shapes run two-phase — the scan
   // fills an internal per-series grid with plain stores, then a presenter
   // (topk_kernel / group_reduce_kernel) folds it along the group-sorted index
+  const bool is_quant = q->agg_id == AGG_QUANTILE;
+  if (is_quant && out_counts) {
+    fdb_set_error("quantile aggregation has no partial (multi-shard) mode — "
+                  "digest shipping is not implemented");
+    return FDB_ERR_BADARG;
+  }
   const char* fe = getenv("FDB_FUSED_GROUP");    // perf/parity experiments
-  const bool fused = !(fe && atoi(fe) == 0) && q->agg_id != AGG_NONE && !is_topk &&
+  const bool fused = !(fe && atoi(fe) == 0) && !is_quant &&
+                     q->agg_id != AGG_NONE && !is_topk &&
                      nw <= 256 && fast_eligible(d, q);
   fdb_query_t qscan = *q;
   if (q->agg_id != AGG_NONE && !fused) {
@@ -885,6 +902,12 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
           q->num_groups, nw, kk, q->agg_id == AGG_TOPK ? 1 : 0,
           dev_out, dev_cnt);
       HIP_CHECK(hipGetLastError());
+    } else if (is_quant) {
+      int32_t rcq = fdb_launch_quantile_cells(e->stream, per_grid,
+                                              d->series_by_group,
+                                              d->group_offsets, q->num_groups,
+                                              nw, q->param, dev_out);
+      if (rcq != FDB_OK) return rcq;
     } else if (q->agg_id != AGG_NONE) {
       size_t cells = (size_t)q->num_groups * nw;
       static int variant = -1;
@@ -912,7 +935,7 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
       ms_sum += ms;
     }
   }
-  if (q->agg_id != AGG_NONE && !is_topk) {
+  if (q->agg_id != AGG_NONE && !is_topk && !is_quant) {
     agg_present_kernel<<<(unsigned)((out_len + 255) / 256), 256, 0, e->stream>>>(
         dev_out, dev_cnt, dev_sq, out_len, q->agg_id, partial);
     HIP_CHECK(hipGetLastError());
@@ -1034,6 +1057,66 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
 extern "C" int32_t fdb_query_exec(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
                                   double* out, double* out_counts, int32_t out_on_device) {
   return run_query(e, d, q, out, out_counts, out_on_device, 0, 1, nullptr);
+}
+
+// count_values cross-series aggregation (CountValuesRowAggregator.scala):
+// per (group, window) the distinct non-NaN values with frequencies, sorted
+// ascending; more than k_cap distinct values in any cell is an error (the
+// reference throws at its 1000-value limit). Host-memory outputs:
+//   out_vals/out_cnts: [num_groups × num_windows × k_cap]
+//   out_n:             [num_groups × num_windows]
+extern "C" int32_t fdb_query_exec_count_values(
+    fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
+    int32_t k_cap, double* out_vals, double* out_cnts, int32_t* out_n) {
+  HIP_CHECK(hipSetDevice(e->device));
+  if (d->has_hist) { fdb_set_error("histogram dataset"); return FDB_ERR_BADARG; }
+  int nw = fdb_num_windows(q);
+  if (nw <= 0 || q->num_groups <= 0 || d->max_group >= q->num_groups) {
+    fdb_set_error("bad count_values query params");
+    return FDB_ERR_BADARG;
+  }
+  size_t cells = (size_t)q->num_groups * nw;
+  double *per_grid = nullptr, *dv = nullptr, *dc = nullptr;
+  int32_t* dn = nullptr;
+  int32_t rc = FDB_ERR;
+  fdb_query_t qscan = *q;
+  qscan.agg_id = AGG_NONE;
+  if (hipMalloc(&per_grid, (size_t)d->num_series * nw * 8) != hipSuccess ||
+      hipMalloc(&dv, cells * (size_t)k_cap * 8) != hipSuccess ||
+      hipMalloc(&dc, cells * (size_t)k_cap * 8) != hipSuccess ||
+      hipMalloc(&dn, cells * 4) != hipSuccess) {
+    fdb_set_error("count_values allocation failed");
+    goto done;
+  }
+  rc = launch_scan(e, d, &qscan, per_grid, nullptr, nullptr);
+  if (rc != FDB_OK) goto done;
+  rc = fdb_launch_count_values(e->stream, per_grid, d->series_by_group,
+                               d->group_offsets, q->num_groups, nw, k_cap,
+                               dv, dc, dn, e->dev_flag);
+  if (rc != FDB_OK) goto done;
+  if (hipStreamSynchronize(e->stream) != hipSuccess) { rc = FDB_ERR; goto done; }
+  {
+    int32_t flag = 0;
+    (void)hipMemcpy(&flag, e->dev_flag, 4, hipMemcpyDeviceToHost);
+    if (flag) {
+      (void)hipMemset(e->dev_flag, 0, 4);
+      fdb_set_error("count_values: a cell exceeded %d distinct values "
+                    "(the reference throws at its 1000-value limit)", k_cap);
+      rc = FDB_ERR;
+      goto done;
+    }
+  }
+  if (hipMemcpy(out_vals, dv, cells * (size_t)k_cap * 8,
+                hipMemcpyDeviceToHost) != hipSuccess ||
+      hipMemcpy(out_cnts, dc, cells * (size_t)k_cap * 8,
+                hipMemcpyDeviceToHost) != hipSuccess ||
+      hipMemcpy(out_n, dn, cells * 4, hipMemcpyDeviceToHost) != hipSuccess) {
+    fdb_set_error("count_values copy-back failed");
+    rc = FDB_ERR;
+  }
+done:
+  (void)hipFree(per_grid); (void)hipFree(dv); (void)hipFree(dc); (void)hipFree(dn);
+  return rc;
 }
 
 extern "C" int32_t fdb_query_bench(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,

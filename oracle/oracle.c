@@ -26,6 +26,7 @@
 #include <stdlib.h>
 
 #include "../filodb_amd/csrc/chunk_format.h"
+#include "../filodb_amd/csrc/tdigest_impl.h"
 
 #ifdef _OPENMP
 #include <omp.h>
@@ -72,7 +73,8 @@ static double interp_quantile(double q, const double* vals, int n) {
   return vals[lower] * (1 - weight) + vals[upper] * weight;
 }
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
-       AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9, AGG_GROUP=10 };
+       AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9, AGG_GROUP=10,
+       AGG_QUANTILE=11, AGG_COUNT_VALUES=12 };
 
 static inline uint16_t rd_u16(const uint8_t* p) { uint16_t v; memcpy(&v, p, 2); return v; }
 static inline uint32_t rd_u32(const uint8_t* p) { uint32_t v; memcpy(&v, p, 4); return v; }
@@ -774,6 +776,29 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
     return 0;
   }
 
+  if (q->agg_id == AGG_QUANTILE) {
+    /* QuantileRowAggregator (QuantileRowAggregator.scala:21-76): one t-digest
+     * per (group, window), samples added in ascending series order, present
+     * step emits digest.quantile(q). Partial-mode digest shipping is not
+     * implemented (out_counts ignored). */
+    if (gridlen > 2000000) return -1;
+    tdigest_t* tds = (tdigest_t*)malloc(gridlen * sizeof(tdigest_t));
+    if (!tds) return -1;
+    for (size_t i = 0; i < gridlen; i++) td_init(&tds[i]);
+    double* row = (double*)malloc((size_t)nw * sizeof(double));
+    eval_ctx_t ctx; ctx.scratch = (double*)malloc(64 * 512 * sizeof(double));
+    for (int s = 0; s < ns; s++) {
+      eval_series(view, s, q, &ctx, row);
+      int grp = view->group_ids[s];
+      for (int w = 0; w < nw; w++)
+        td_add(&tds[(size_t)grp * nw + w], row[w]);
+    }
+    for (size_t i = 0; i < gridlen; i++)
+      out[i] = td_quantile(&tds[i], q->param);
+    free(tds); free(row); free(ctx.scratch);
+    return 0;
+  }
+
   for (size_t i = 0; i < gridlen; i++) out[i] = NAN;
   double* counts = out_counts;
   double* owned_counts = 0;
@@ -926,6 +951,56 @@ EXPORT int32_t oracle_query_exec(const fdb_view_t* view, const fdb_query_t* q,
 }
 
 /* convenience: evaluate one series only (tests) */
+/* CountValuesRowAggregator (CountValuesRowAggregator.scala:26-100): per
+ * (group, window) a value→frequency map over non-NaN series results; more
+ * than `limit` distinct values is an error (the reference throws at 1000).
+ * Output: per cell, n distinct pairs sorted by value ascending. */
+EXPORT int32_t oracle_count_values(const fdb_view_t* view, const fdb_query_t* q,
+                                   int32_t k_cap, double* out_vals,
+                                   double* out_cnts, int32_t* out_n) {
+  int nw = (int)((q->end - q->start) / q->step) + 1;
+  int ns = view->num_series;
+  int ng = q->num_groups;
+  size_t gridlen = (size_t)ng * nw;
+  if (k_cap < 1 || k_cap > 1000) return -1;
+  double* vals = (double*)calloc(gridlen * (size_t)k_cap, sizeof(double));
+  double* cnts = (double*)calloc(gridlen * (size_t)k_cap, sizeof(double));
+  if (!vals || !cnts) { free(vals); free(cnts); return -1; }
+  memset(out_n, 0, gridlen * sizeof(int32_t));
+  double* row = (double*)malloc((size_t)nw * sizeof(double));
+  eval_ctx_t ctx; ctx.scratch = (double*)malloc(64 * 512 * sizeof(double));
+  int rc = 0;
+  for (int s = 0; s < ns && rc == 0; s++) {
+    eval_series(view, s, q, &ctx, row);
+    int grp = view->group_ids[s];
+    for (int w = 0; w < nw; w++) {
+      double x = row[w];
+      if (isnan(x)) continue;
+      size_t cell = (size_t)grp * nw + w;
+      double* cv = vals + cell * k_cap;
+      double* cc = cnts + cell * k_cap;
+      int n = out_n[cell];
+      /* binary search for x in the sorted distinct list */
+      int lo = 0, hi = n;
+      while (lo < hi) { int mid = (lo + hi) / 2;
+        if (cv[mid] < x) lo = mid + 1; else hi = mid; }
+      if (lo < n && cv[lo] == x) { cc[lo] += 1; }
+      else {
+        if (n >= k_cap) { rc = -2; break; }   /* reference throws at limit */
+        for (int j = n; j > lo; j--) { cv[j] = cv[j-1]; cc[j] = cc[j-1]; }
+        cv[lo] = x; cc[lo] = 1;
+        out_n[cell] = n + 1;
+      }
+    }
+  }
+  if (rc == 0) {
+    memcpy(out_vals, vals, gridlen * (size_t)k_cap * sizeof(double));
+    memcpy(out_cnts, cnts, gridlen * (size_t)k_cap * sizeof(double));
+  }
+  free(vals); free(cnts); free(row); free(ctx.scratch);
+  return rc;
+}
+
 EXPORT int32_t oracle_eval_series(const fdb_view_t* view, int32_t sid, const fdb_query_t* q,
                                   double* out) {
   eval_ctx_t ctx; ctx.scratch = (double*)malloc(64 * 512 * sizeof(double));

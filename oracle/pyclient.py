@@ -106,6 +106,26 @@ def hist_quantile(q, values, first, mult):
     return L.oracle_hist_quantile(q, v.ctypes.data_as(_c_double_p), len(v), first, mult)
 
 
+def count_values(view, q, k_cap=64):
+    import ctypes as ct
+    nw = q.num_windows
+    cells = q.num_groups * nw
+    vals = np.zeros(cells * k_cap, dtype=np.float64)
+    cnts = np.zeros(cells * k_cap, dtype=np.float64)
+    n = np.zeros(cells, dtype=np.int32)
+    L = lib()
+    L.oracle_count_values.argtypes = [ct.c_void_p, ct.c_void_p, ct.c_int32,
+                                      _c_double_p, _c_double_p,
+                                      ct.POINTER(ct.c_int32)]
+    rc = L.oracle_count_values(ct.byref(view), ct.byref(q), k_cap,
+                               vals.ctypes.data_as(_c_double_p),
+                               cnts.ctypes.data_as(_c_double_p),
+                               n.ctypes.data_as(ct.POINTER(ct.c_int32)))
+    if rc != 0:
+        raise RuntimeError(f"oracle_count_values failed rc={rc}")
+    return vals.reshape(cells, k_cap), cnts.reshape(cells, k_cap), n
+
+
 def query_exec(view, q, num_series, num_windows, out_counts=False, nthreads=1):
     """Runs the oracle over a store view (filodb_amd.View) with query q
     (filodb_amd.Query). Returns the result grid as numpy."""
