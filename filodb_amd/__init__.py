@@ -135,6 +135,14 @@ def lib():
         L.fdb_query_bench.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
                                       _c_double_p, _c_double_p, ctypes.c_int32,
                                       ctypes.c_int32, ctypes.c_int32, _c_double_p]
+        L.fdb_series_append_hist_mm.argtypes = [
+            ctypes.c_void_p, ctypes.c_int32, ctypes.POINTER(ctypes.c_int64),
+            ctypes.POINTER(ctypes.c_uint64), _c_double_p, _c_double_p,
+            ctypes.c_int32, ctypes.c_int32, ctypes.c_double, ctypes.c_double]
+        L.fdb_query_exec_hist_mm.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
+            ctypes.c_int32, _c_double_p, _c_double_p, _c_double_p,
+            _c_double_p, _c_double_p, ctypes.c_int32]
         L.fdb_query_exec_count_values.argtypes = [
             ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
             ctypes.c_int32, _c_double_p, _c_double_p,
@@ -208,6 +216,19 @@ class ChunkStore:
             self._h, sid, ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
             bv.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)), n, nb,
             first, mult), "append_hist")
+
+    def append_hist_mm(self, sid, ts, bucket_values, maxs, mins,
+                       first=2.0, mult=2.0):
+        ts = np.ascontiguousarray(ts, dtype=np.int64)
+        bv = np.ascontiguousarray(bucket_values, dtype=np.uint64)
+        mx = np.ascontiguousarray(maxs, dtype=np.float64)
+        mn = np.ascontiguousarray(mins, dtype=np.float64)
+        n, nb = bv.shape
+        _check(lib().fdb_series_append_hist_mm(
+            self._h, sid, ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+            bv.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            _as_f64_ptr(mx), _as_f64_ptr(mn), n, nb, first, mult),
+            "append_hist_mm")
 
     def cut_chunk(self, sid):
         _check(lib().fdb_series_cut_chunk(self._h, sid), "cut_chunk")
@@ -290,6 +311,16 @@ class Engine:
             _as_f64_ptr(out_bucket_sums), _as_f64_ptr(out_counts),
             _as_f64_ptr(out_quantile), 1 if on_device else 0)
         _check(rc, "query_exec_hist")
+
+    def query_hist_mm(self, dataset, q: Query, num_buckets,
+                      out_bucket_sums=None, out_counts=None, out_max=None,
+                      out_min=None, out_quantile=None, on_device=False):
+        rc = lib().fdb_query_exec_hist_mm(
+            self._h, dataset._h, ctypes.byref(q), num_buckets,
+            _as_f64_ptr(out_bucket_sums), _as_f64_ptr(out_counts),
+            _as_f64_ptr(out_max), _as_f64_ptr(out_min),
+            _as_f64_ptr(out_quantile), 1 if on_device else 0)
+        _check(rc, "query_exec_hist_mm")
 
     def count_values(self, dataset, q: Query, k_cap=64):
         """CountValuesRowAggregator: per (group, window) distinct values with

@@ -345,9 +345,11 @@ static void encode_timestamps(const int64_t* ts, int n, bytes& out) {
 // ---------------------------------------------------------------------------
 struct Chunk {
   uint64_t ts_off, val_off;      // into store blob (valid after seal)
+  uint64_t max_off = 0, min_off = 0;   // hist companion columns (0 = absent)
   int32_t  num_rows;
   int64_t  start_time, end_time;
   bytes    ts_bytes, val_bytes;  // cleared after seal (moved into blob)
+  bytes    max_bytes, min_bytes;
 };
 
 struct Series {
@@ -359,6 +361,8 @@ struct Series {
   double   buf_last = -1.7976931348623157e308;  // Double.MinValue
   // histogram column state (FDB_COL_HIST)
   std::vector<uint64_t> buf_hist;     // row-major [rows × num_buckets]
+  std::vector<double> buf_max, buf_min;  // companion columns (otel max/min)
+  bool     has_mm = false;
   int32_t  num_buckets = 0;
   double   bucket_first = 0, bucket_mult = 0;
   std::vector<Chunk> chunks;
@@ -496,6 +500,11 @@ static int32_t cut_chunk(fdb_store_t* s, Series& se) {
   if (se.col_kind == FDB_COL_HIST) {
     encode_hist_chunk(se, n, c.val_bytes);
     se.buf_hist.clear();
+    if (se.has_mm) {
+      encode_doubles(se.buf_max.data(), n, false, c.max_bytes);
+      encode_doubles(se.buf_min.data(), n, false, c.min_bytes);
+      se.buf_max.clear(); se.buf_min.clear();
+    }
   } else {
     encode_doubles(se.buf_vals.data(), n, se.col_kind == FDB_COL_COUNTER && se.buf_drop,
                    c.val_bytes);
@@ -535,6 +544,34 @@ extern "C" int32_t fdb_series_append_hist(fdb_store_t* s, int32_t sid,
   return FDB_OK;
 }
 
+// histogram rows with otel max/min companion double columns
+// (SumAndMaxOverTimeFuncHD / CumulativeHistRateAndMinMaxFunction inputs,
+//  AggrOverTimeFunctions.scala:612-813)
+extern "C" int32_t fdb_series_append_hist_mm(fdb_store_t* s, int32_t sid,
+                                             const int64_t* ts,
+                                             const uint64_t* bucket_values,
+                                             const double* maxs, const double* mins,
+                                             int32_t n, int32_t num_buckets,
+                                             double bucket_first, double bucket_mult) {
+  if (sid < 0 || sid >= (int32_t)s->series.size()) { fdb_set_error("bad series id"); return FDB_ERR_BADARG; }
+  Series& se = s->series[(size_t)sid];
+  if (!se.buf_ts.empty() && !se.has_mm && se.col_kind == FDB_COL_HIST && !se.buf_hist.empty()) {
+    fdb_set_error("series %d mixes hist rows with and without max/min", sid);
+    return FDB_ERR_BADARG;
+  }
+  se.has_mm = true;
+  // append row-by-row so the auto-cut keeps columns aligned
+  for (int32_t i = 0; i < n; i++) {
+    se.buf_max.push_back(maxs[i]);
+    se.buf_min.push_back(mins[i]);
+    int32_t rc = fdb_series_append_hist(s, sid, ts + i,
+                                        bucket_values + (size_t)i * num_buckets,
+                                        1, num_buckets, bucket_first, bucket_mult);
+    if (rc != FDB_OK) return rc;
+  }
+  return FDB_OK;
+}
+
 extern "C" int32_t fdb_series_append(fdb_store_t* s, int32_t sid,
                                      const int64_t* ts, const double* vals, int32_t n) {
   if (s->sealed) { fdb_set_error("store is sealed"); return FDB_ERR_BADARG; }
@@ -570,6 +607,7 @@ extern "C" int32_t fdb_store_seal(fdb_store_t* s) {
     cut_chunk(s, se);
     for (auto& c : se.chunks) {
       total += (c.ts_bytes.size() + 63 & ~size_t(63)) + (c.val_bytes.size() + 63 & ~size_t(63));
+      total += (c.max_bytes.size() + 63 & ~size_t(63)) + (c.min_bytes.size() + 63 & ~size_t(63));
       nchunks++;
     }
   }
@@ -590,8 +628,17 @@ extern "C" int32_t fdb_store_seal(fdb_store_t* s) {
       c.val_off = off;
       memcpy(s->blob.data() + off, c.val_bytes.data(), c.val_bytes.size());
       off = (off + c.val_bytes.size() + 63) & ~size_t(63);
+      if (!c.max_bytes.empty()) {
+        c.max_off = off;
+        memcpy(s->blob.data() + off, c.max_bytes.data(), c.max_bytes.size());
+        off = (off + c.max_bytes.size() + 63) & ~size_t(63);
+        c.min_off = off;
+        memcpy(s->blob.data() + off, c.min_bytes.data(), c.min_bytes.size());
+        off = (off + c.min_bytes.size() + 63) & ~size_t(63);
+      }
       fdb_dir_entry_t e;
       e.ts_off = c.ts_off; e.val_off = c.val_off;
+      e.max_off = c.max_off; e.min_off = c.min_off;
       e.start_time = c.start_time; e.end_time = c.end_time;
       e.num_rows = c.num_rows; e._pad = 0;
       s->dir.push_back(e);

@@ -84,6 +84,12 @@ typedef struct {
   int64_t  end_time;
   int32_t  num_rows;
   int32_t  _pad;
+  /* histogram companion double columns (otel max/min schema,
+   * AggrOverTimeFunctions.scala:612-813); 0 = column absent. The first
+   * vector in the blob is always a timestamp vector, so offset 0 is never a
+   * companion. */
+  uint64_t max_off;
+  uint64_t min_off;
 } fdb_dir_entry_t;
 
 #endif /* FDB_CHUNK_FORMAT_H */

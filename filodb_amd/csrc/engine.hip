@@ -70,12 +70,14 @@ int32_t fdb_launch_window_sample(hipStream_t stream, const uint8_t* blob,
 
 // hist2.hip: two-cursor histogram walk (unbounded chunks / window ratio)
 int32_t fdb_launch_hist2(hipStream_t stream, const uint8_t* blob, DirSoA dir,
+                         const uint64_t* max_off, const uint64_t* min_off,
                          const int32_t* series_first,
                          const int32_t* series_nchunks,
                          const int32_t* group_ids, int num_series,
                          int64_t qstart, int64_t qstep, int64_t qwindow,
-                         int num_windows, int nb,
-                         double* out_sums, double* out_cnt);
+                         int num_windows, int nb, int hfunc,
+                         double* out_sums, double* out_cnt,
+                         double* out_max, double* out_min);
 int32_t fdb_launch_stream_walk(hipStream_t stream, const uint8_t* blob,
                                DirSoA dir, const void* sums,
                                const int32_t* series_first,
@@ -548,6 +550,8 @@ struct fdb_dataset {
   int has_hist;             // dataset holds sect-delta histogram vectors
   int fast_ok;              // single-chunk series, chunk spans fit i32 ms
   void* sums;               // ChunkSum[num_chunks] for the streaming walk
+  uint64_t *max_off, *min_off;  // hist companion column offsets (0 = absent)
+  int has_mm;               // every hist chunk carries max/min columns
 };
 
 extern "C" fdb_engine_t* fdb_engine_create(int32_t device) {
@@ -596,6 +600,7 @@ extern "C" void fdb_dataset_destroy(fdb_dataset_t* d) {
   (void)hipFree(d->series_first); (void)hipFree(d->series_nchunks); (void)hipFree(d->group_ids);
   (void)hipFree(d->series_by_group); (void)hipFree(d->group_offsets);
   (void)hipFree(d->sums);
+  (void)hipFree(d->max_off); (void)hipFree(d->min_off);
   delete d;
 }
 
@@ -611,17 +616,26 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
   int64_t nc = view.num_chunks;
 
   // SoA host staging
-  std::vector<uint64_t> ts_off(nc), val_off(nc);
+  std::vector<uint64_t> ts_off(nc), val_off(nc), mx_off(nc), mn_off(nc);
   std::vector<int64_t> st(nc), en(nc);
   std::vector<int32_t> nr(nc);
   int64_t payload = 0, samples = 0;
+  int all_mm = 1, any_mm = 0;
   for (int64_t i = 0; i < nc; i++) {
     ts_off[i] = dir[i].ts_off; val_off[i] = dir[i].val_off;
+    mx_off[i] = dir[i].max_off; mn_off[i] = dir[i].min_off;
+    if (dir[i].max_off) any_mm = 1; else all_mm = 0;
     st[i] = dir[i].start_time; en[i] = dir[i].end_time; nr[i] = dir[i].num_rows;
     uint32_t tl, vl;
     memcpy(&tl, view.blob + dir[i].ts_off, 4);
     memcpy(&vl, view.blob + dir[i].val_off, 4);
     payload += (int64_t)tl + 4 + (int64_t)vl + 4;
+    if (dir[i].max_off) {
+      uint32_t xl, nl;
+      memcpy(&xl, view.blob + dir[i].max_off, 4);
+      memcpy(&nl, view.blob + dir[i].min_off, 4);
+      payload += (int64_t)xl + 4 + (int64_t)nl + 4;
+    }
     samples += dir[i].num_rows;
   }
   int max_group = 0, max_rows = 0, max_chunks = 0, max_chunk_rows = 0;
@@ -665,6 +679,7 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
   d->has_hist = has_hist;
   d->fast_ok = !has_hist && max_chunks <= 1 && max_chunk_rows <= 400 &&
                spans_fit_i32;
+  d->has_mm = has_hist && any_mm && all_mm;
 
   auto upload = [&](void** dst, const void* src, size_t bytes) -> bool {
     if (hipMalloc(dst, bytes ? bytes : 8) != hipSuccess) return false;
@@ -699,7 +714,9 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
     && upload((void**)&d->series_nchunks, view.series_nchunks, (size_t)view.num_series * 4)
     && upload((void**)&d->group_ids, view.group_ids, (size_t)view.num_series * 4)
     && upload((void**)&d->series_by_group, sbg.data(), sbg.size() * 4)
-    && upload((void**)&d->group_offsets, goff.data(), goff.size() * 4);
+    && upload((void**)&d->group_offsets, goff.data(), goff.size() * 4)
+    && upload((void**)&d->max_off, mx_off.data(), nc * 8)
+    && upload((void**)&d->min_off, mn_off.data(), nc * 8);
   if (!ok) {
     fdb_set_error("device upload failed (out of HBM?)");
     fdb_dataset_destroy(d);
@@ -962,16 +979,28 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
 }
 
 
-extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
-                                       const fdb_query_t* q, int32_t nb,
-                                       double* out_bucket_sums, double* out_counts,
-                                       double* out_quantile, int32_t out_on_device) {
+static int32_t run_hist(fdb_engine_t* e, const fdb_dataset_t* d,
+                        const fdb_query_t* q, int32_t nb,
+                        double* out_bucket_sums, double* out_counts,
+                        double* out_max, double* out_min,
+                        double* out_quantile, int32_t out_on_device) {
   HIP_CHECK(hipSetDevice(e->device));
   int nw = fdb_num_windows(q);
   if (nw <= 0 || q->num_groups <= 0) { fdb_set_error("bad hist query params"); return FDB_ERR_BADARG; }
   if (nb < 1 || nb > 64) { fdb_set_error("num_buckets must be 1..64"); return FDB_ERR_BADARG; }
   if (!d->has_hist) {
     fdb_set_error("not a histogram dataset");
+    return FDB_ERR_BADARG;
+  }
+  const int hfunc = (q->func_id == FDB_FN_SUM_OVER_TIME) ? 1 : 0;
+  if (q->func_id != FDB_FN_HIST_RATE && q->func_id != FDB_FN_SUM_OVER_TIME) {
+    fdb_set_error("histogram queries support FDB_FN_HIST_RATE and "
+                  "FDB_FN_SUM_OVER_TIME (got %d)", q->func_id);
+    return FDB_ERR_BADARG;
+  }
+  if ((out_max || out_min) && !d->has_mm) {
+    fdb_set_error("dataset has no max/min companion columns "
+                  "(fdb_series_append_hist_mm)");
     return FDB_ERR_BADARG;
   }
   if (d->max_chunk_rows > FDB_MAX_ROWS_PER_SERIES) {
@@ -982,25 +1011,38 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
   const char* hv1 = getenv("FDB_HIST_V1");   // round-1 kernel for A/B runs
   const bool use_v1 = hv1 && atoi(hv1) == 1;
   if (use_v1 && (d->max_chunks > FDB_HIST_MAX_CHUNKS ||
-                 q->window / q->step + 2 > FDB_HIST_RING)) {
-    fdb_set_error("FDB_HIST_V1 set but the query exceeds the v1 ring caps");
+                 q->window / q->step + 2 > FDB_HIST_RING ||
+                 hfunc == 1 || out_max || out_min)) {
+    fdb_set_error("FDB_HIST_V1 supports only rate without companions within "
+                  "the v1 ring caps");
     return FDB_ERR_BADARG;
   }
   if (d->max_group >= q->num_groups) { fdb_set_error("num_groups too small"); return FDB_ERR_BADARG; }
   size_t cells = (size_t)q->num_groups * nw;
 
   double *dev_sums = nullptr, *dev_cnt = nullptr, *dev_quant = nullptr;
-  bool own_sums = true, own_cnt = true, own_quant = false;
+  double *dev_max = nullptr, *dev_min = nullptr;
+  bool own_sums = true, own_cnt = true, own_quant = false, own_mm = false;
   if (out_on_device) {
     dev_sums = out_bucket_sums; own_sums = dev_sums == nullptr;
     dev_cnt = out_counts; own_cnt = dev_cnt == nullptr;
     dev_quant = out_quantile;
+    dev_max = out_max; dev_min = out_min;
   }
   if (!dev_sums) HIP_CHECK(hipMalloc(&dev_sums, cells * nb * 8));
   if (!dev_cnt) HIP_CHECK(hipMalloc(&dev_cnt, cells * 8));
   if (out_quantile && !dev_quant) { HIP_CHECK(hipMalloc(&dev_quant, cells * 8)); own_quant = true; }
+  if (out_max && !dev_max) {
+    HIP_CHECK(hipMalloc(&dev_max, cells * 8));
+    HIP_CHECK(hipMalloc(&dev_min, cells * 8));
+    own_mm = true;
+  }
   HIP_CHECK(hipMemsetAsync(dev_sums, 0, cells * nb * 8, e->stream));
   HIP_CHECK(hipMemsetAsync(dev_cnt, 0, cells * 8, e->stream));
+  if (dev_max) {    // NaN start for the maxIgnoreNaN/minIgnoreNaN merges
+    HIP_CHECK(hipMemsetAsync(dev_max, 0xFF, cells * 8, e->stream));
+    HIP_CHECK(hipMemsetAsync(dev_min, 0xFF, cells * 8, e->stream));
+  }
 
   DirSoA dir{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
   int grid = (d->num_series + HIST_WAVES - 1) / HIST_WAVES;
@@ -1009,10 +1051,12 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
   if (hcap > 0 && grid > hcap) grid = hcap;
   int dbg = getenv("FDB_HIST_TIME") ? 1 : 0;
   if (!use_v1) {
-    int32_t rc = fdb_launch_hist2(e->stream, d->blob, dir, d->series_first,
+    int32_t rc = fdb_launch_hist2(e->stream, d->blob, dir, d->max_off,
+                                  d->min_off, d->series_first,
                                   d->series_nchunks, d->group_ids,
                                   d->num_series, q->start, q->step, q->window,
-                                  nw, nb, dev_sums, dev_cnt);
+                                  nw, nb, hfunc, dev_sums, dev_cnt,
+                                  dev_max, dev_min);
     if (rc != FDB_OK) return rc;
   } else if (d->max_chunks > 1)
     hipLaunchKernelGGL(hist_scan_kernel<FDB_HIST_MAX_CHUNKS>, dim3(grid),
@@ -1047,11 +1091,36 @@ extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
       HIP_CHECK(hipMemcpy(out_counts, dev_cnt, cells * 8, hipMemcpyDeviceToHost));
     if (out_quantile)
       HIP_CHECK(hipMemcpy(out_quantile, dev_quant, cells * 8, hipMemcpyDeviceToHost));
+    if (out_max && dev_max)
+      HIP_CHECK(hipMemcpy(out_max, dev_max, cells * 8, hipMemcpyDeviceToHost));
+    if (out_min && dev_min)
+      HIP_CHECK(hipMemcpy(out_min, dev_min, cells * 8, hipMemcpyDeviceToHost));
   }
   if (own_sums) (void)hipFree(dev_sums);
   if (own_cnt) (void)hipFree(dev_cnt);
   if (own_quant) (void)hipFree(dev_quant);
+  if (own_mm) { (void)hipFree(dev_max); (void)hipFree(dev_min); }
   return FDB_OK;
+}
+
+extern "C" int32_t fdb_query_exec_hist(fdb_engine_t* e, const fdb_dataset_t* d,
+                                       const fdb_query_t* q, int32_t nb,
+                                       double* out_bucket_sums, double* out_counts,
+                                       double* out_quantile, int32_t out_on_device) {
+  return run_hist(e, d, q, nb, out_bucket_sums, out_counts, nullptr, nullptr,
+                  out_quantile, out_on_device);
+}
+
+// histogram query with otel max/min companion outputs [G × W]
+// (SumAndMaxOverTimeFuncHD / CumulativeHistRateAndMinMaxFunction,
+//  AggrOverTimeFunctions.scala:612-813; HistMaxMinSumAggregator merges)
+extern "C" int32_t fdb_query_exec_hist_mm(fdb_engine_t* e, const fdb_dataset_t* d,
+                                          const fdb_query_t* q, int32_t nb,
+                                          double* out_bucket_sums, double* out_counts,
+                                          double* out_max, double* out_min,
+                                          double* out_quantile, int32_t out_on_device) {
+  return run_hist(e, d, q, nb, out_bucket_sums, out_counts, out_max, out_min,
+                  out_quantile, out_on_device);
 }
 
 extern "C" int32_t fdb_query_exec(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,

@@ -97,11 +97,20 @@ struct H2Cursor {
   double C_b;            // per-lane correction total at current position
   double Centry_b;       // C at current chunk entry (after boundary drop)
   double prevlast_b;     // raw value of the previous chunk's last element
+  double psum_b;         // per-lane running value prefix (HFUNC==1)
 };
 
-template <int MINW>
+// HFUNC 0 = counter-corrected rate (HistRateFunction); 1 = SumOverTime of
+// the histograms (raw bucket sums — SumOverTimeChunkedFunctionH, no
+// corrections; both cursors then decode every element they pass and carry a
+// per-bucket running prefix). out_max/out_min (nullable) add the otel
+// companion-column max/min per window, merged across series with
+// maxIgnoreNaN/minIgnoreNaN (HistMaxMinSumAggregator).
+template <int MINW, int HFUNC>
 __global__ __launch_bounds__(H2_WAVES * 64, MINW)
 void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
+                  const uint64_t* __restrict__ max_off,
+                  const uint64_t* __restrict__ min_off,
                   const int32_t* __restrict__ series_first,
                   const int32_t* __restrict__ series_nchunks,
                   const int32_t* __restrict__ group_ids,
@@ -109,7 +118,9 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                   int64_t qstart, int64_t qstep, int64_t qwindow,
                   int num_windows, int nb,
                   double* __restrict__ out_sums,   // [G × W × nb]
-                  double* __restrict__ out_cnt) {  // [G × W]
+                  double* __restrict__ out_cnt,    // [G × W]
+                  double* __restrict__ out_max,    // [G × W] or null
+                  double* __restrict__ out_min) {  // [G × W] or null
   __shared__ int64_t tsS_all[H2_WAVES][H2_ROWS];
   __shared__ int64_t tsE_all[H2_WAVES][H2_ROWS];
   const int wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
@@ -152,6 +163,7 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       if (cu.sect_first) { cu.val_b = (double)scan; cu.base_b = cu.val_b; }
       else cu.val_b = cu.base_b + (double)scan;
       cu.decoded = true;
+      if (HFUNC == 1) cu.psum_b += cu.val_b;   // value prefix through current
     };
 
     // step to the next element; returns false when the series is exhausted.
@@ -184,7 +196,8 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       cu.e_global++;
       cu.sect_first = new_sect;
       cu.decoded = false;
-      if (new_sect) decode_cur(cu);           // section base always decoded
+      if (HFUNC == 1) decode_cur(cu);         // sum mode: full prefix
+      else if (new_sect) decode_cur(cu);      // section base always decoded
       else if (cu.sect_left == 0) decode_cur(cu);  // section last: TypeDrop feed
       if (cu.e_local == 0) {
         // first element of a chunk: boundary drop detection
@@ -207,6 +220,7 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     memset(&S, 0, sizeof(S)); memset(&E, 0, sizeof(E));
     S.e_global = -1; E.e_global = -1;
     S.C_b = 0; E.C_b = 0;
+    S.psum_b = 0; E.psum_b = 0;
     open_chunk(S, 0, tsS_all[wave]);
     if (S.nrows == 0) continue;
     if (!step(S, tsS_all[wave])) continue;
@@ -235,21 +249,60 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       const int64_t t1 = tsS_all[wave][S.e_local];
       const int64_t t2 = tsE_all[wave][E.e_local];
       if (t1 > wEnd || t2 < wStart) continue; // empty window
-      if (!(t2 > t1)) continue;               // highestTime > lowestTime rule
-      decode_cur(S);
-      decode_cur(E);
-      const int numSamples = E.e_global - S.e_global + 1;
-      if (live) {
-        // corrections relative to the window's first chunk: the reference's
-        // per-window CorrectionMeta starts NoCorrection there
-        double v1 = S.val_b + (S.C_b - S.Centry_b);
-        double v2 = E.val_b + (E.C_b - S.Centry_b);
-        double r = d_extrapolated_rate(wStart, wEnd, numSamples,
-                                       t1, v1, t2, v2, true, true);
-        atomicAdd(&out_sums[((size_t)grp * num_windows + w) * nb + b], r);
+      if (t2 < t1) continue;
+      const size_t cell = (size_t)grp * num_windows + w;
+      // otel companion columns: NaN-ignoring max/min over the window's rows
+      // [S..E], read straight from the raw double vectors chunk by chunk
+      if (out_max || out_min) {
+        double wmax = NAN, wmin = NAN;
+        for (int c2 = S.c; c2 <= E.c; c2++) {
+          if (max_off[first + c2] == 0) continue;
+          DVec xv, nv;
+          d_vec_open_wide(blob + max_off[first + c2], &xv, nullptr);
+          d_vec_open_wide(blob + min_off[first + c2], &nv, nullptr);
+          const int lo = (c2 == S.c) ? S.e_local : 0;
+          const int hi = (c2 == E.c) ? E.e_local : dir.num_rows[first + c2] - 1;
+          for (int i = lo + lane; i <= hi; i += 64) {
+            double mx = d_dv_at(&xv, i);
+            double mn = d_dv_at(&nv, i);
+            if (!isnan(mx) && (isnan(wmax) || mx > wmax)) wmax = mx;
+            if (!isnan(mn) && (isnan(wmin) || mn < wmin)) wmin = mn;
+          }
+        }
+        for (int off = 32; off > 0; off >>= 1) {
+          double o = __shfl_down(wmax, off);
+          if (!isnan(o) && (isnan(wmax) || o > wmax)) wmax = o;
+          o = __shfl_down(wmin, off);
+          if (!isnan(o) && (isnan(wmin) || o < wmin)) wmin = o;
+        }
+        if (lane == 0) {
+          if (out_max && !isnan(wmax)) atomic_min_max_f64(&out_max[cell], wmax, false);
+          if (out_min && !isnan(wmin)) atomic_min_max_f64(&out_min[cell], wmin, true);
+        }
       }
-      if (lane == 0)
-        atomicAdd(&out_cnt[(size_t)grp * num_windows + w], 1.0);
+      if (HFUNC == 0) {
+        if (!(t2 > t1)) continue;             // highestTime > lowestTime rule
+        decode_cur(S);
+        decode_cur(E);
+        const int numSamples = E.e_global - S.e_global + 1;
+        if (live) {
+          // corrections relative to the window's first chunk: the reference's
+          // per-window CorrectionMeta starts NoCorrection there
+          double v1 = S.val_b + (S.C_b - S.Centry_b);
+          double v2 = E.val_b + (E.C_b - S.Centry_b);
+          double r = d_extrapolated_rate(wStart, wEnd, numSamples,
+                                         t1, v1, t2, v2, true, true);
+          atomicAdd(&out_sums[cell * nb + b], r);
+        }
+        if (lane == 0) atomicAdd(&out_cnt[cell], 1.0);
+      } else {
+        // SumOverTime: prefix difference over [S..E] inclusive
+        if (live) {
+          double sum_b = E.psum_b - S.psum_b + S.val_b;
+          atomicAdd(&out_sums[cell * nb + b], sum_b);
+        }
+        if (lane == 0) atomicAdd(&out_cnt[cell], 1.0);
+      }
     }
     d_wait_lds();
     __builtin_amdgcn_wave_barrier();
@@ -257,27 +310,30 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 }
 
 int32_t fdb_launch_hist2(hipStream_t stream, const uint8_t* blob, DirSoA dir,
+                         const uint64_t* max_off, const uint64_t* min_off,
                          const int32_t* series_first,
                          const int32_t* series_nchunks,
                          const int32_t* group_ids, int num_series,
                          int64_t qstart, int64_t qstep, int64_t qwindow,
-                         int num_windows, int nb,
-                         double* out_sums, double* out_cnt) {
+                         int num_windows, int nb, int hfunc,
+                         double* out_sums, double* out_cnt,
+                         double* out_max, double* out_min) {
   int grid = (num_series + H2_WAVES - 1) / H2_WAVES;
   int cap = 8192;
   if (const char* g = getenv("FDB_HIST_GRID")) cap = atoi(g);
   if (cap > 0 && grid > cap) grid = cap;
   const char* hw = getenv("FDB_HIST_WAVES");   // occupancy experiment knob
-  if (hw && atoi(hw) == 5)
-    hipLaunchKernelGGL((hist2_kernel<5>), dim3(grid), dim3(H2_WAVES * 64), 0,
-                       stream, blob, dir, series_first, series_nchunks,
-                       group_ids, num_series, qstart, qstep, qwindow,
-                       num_windows, nb, out_sums, out_cnt);
+  #define H2ARGS blob, dir, max_off, min_off, series_first, series_nchunks,       group_ids, num_series, qstart, qstep, qwindow, num_windows, nb,       out_sums, out_cnt, out_max, out_min
+  if (hfunc == 1)
+    hipLaunchKernelGGL((hist2_kernel<4, 1>), dim3(grid), dim3(H2_WAVES * 64), 0,
+                       stream, H2ARGS);
+  else if (hw && atoi(hw) == 5)
+    hipLaunchKernelGGL((hist2_kernel<5, 0>), dim3(grid), dim3(H2_WAVES * 64), 0,
+                       stream, H2ARGS);
   else
-    hipLaunchKernelGGL((hist2_kernel<4>), dim3(grid), dim3(H2_WAVES * 64), 0,
-                       stream, blob, dir, series_first, series_nchunks,
-                       group_ids, num_series, qstart, qstep, qwindow,
-                       num_windows, nb, out_sums, out_cnt);
+    hipLaunchKernelGGL((hist2_kernel<4, 0>), dim3(grid), dim3(H2_WAVES * 64), 0,
+                       stream, H2ARGS);
+  #undef H2ARGS
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) {
     fdb_set_error("hist2_kernel launch failed: %s", hipGetErrorString(e));

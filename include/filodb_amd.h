@@ -272,6 +272,27 @@ int32_t fdb_query_exec_count_values(fdb_engine_t* e, const fdb_dataset_t* d,
                                     double* out_vals, double* out_cnts,
                                     int32_t* out_n);
 
+/* Histogram rows with otel max/min companion double columns
+ * (SumAndMaxOverTimeFuncHD / CumulativeHistRateAndMinMaxFunction inputs,
+ *  AggrOverTimeFunctions.scala:612-813). */
+int32_t fdb_series_append_hist_mm(fdb_store_t* s, int32_t sid,
+                                  const int64_t* ts,
+                                  const uint64_t* bucket_values,
+                                  const double* maxs, const double* mins,
+                                  int32_t n, int32_t num_buckets,
+                                  double bucket_first, double bucket_mult);
+
+/* Histogram query with companion max/min outputs [num_groups × windows]
+ * (merged across series with maxIgnoreNaN/minIgnoreNaN per
+ *  HistMaxMinSumAggregator). func_id FDB_FN_HIST_RATE = counter-corrected
+ * rate + min/max; FDB_FN_SUM_OVER_TIME = SumOverTime of the histograms
+ * (SumAndMaxOverTimeFuncHD shape). out_max/out_min may be NULL. */
+int32_t fdb_query_exec_hist_mm(fdb_engine_t* e, const fdb_dataset_t* d,
+                               const fdb_query_t* q, int32_t nb,
+                               double* out_bucket_sums, double* out_counts,
+                               double* out_max, double* out_min,
+                               double* out_quantile, int32_t out_on_device);
+
 /* Histogram pipeline for BASELINE config #4:
  * histogram_quantile(param, sum(rate(hist[window])) by group)
  * — HistRateFunction per series (per-bucket extrapolated rate with counter

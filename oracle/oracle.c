@@ -1163,6 +1163,17 @@ EXPORT int32_t oracle_hist_corrections(const uint8_t* vec, int64_t* out, int32_t
 /* histogram_quantile(param, sum by(group)(rate(hist[window]))) — the full
  * config-#4 pipeline. out_bucket_sums [G×W×nb] / out_counts [G×W] /
  * out_quantile [G×W]; any may be NULL. */
+/* companion-column variant: adds per-(group,window) max/min over the otel
+ * max/min double columns (SumAndMaxOverTimeFuncHD and
+ * CumulativeHistRateAndMinMaxFunction, AggrOverTimeFunctions.scala:612-813;
+ * cross-series merge maxIgnoreNaN/minIgnoreNaN per HistMaxMinSumAggregator).
+ * func_id FN_HIST_RATE -> counter-corrected rate; FN_SUM -> SumOverTime of
+ * the histograms (raw bucket sums, no corrections). out_max/out_min NULLable. */
+EXPORT int32_t oracle_query_exec_hist_mm(const fdb_view_t* view, const fdb_query_t* q,
+                                         int32_t nb, double* sums, double* cnts,
+                                         double* out_max, double* out_min,
+                                         double* out_quantile);
+
 EXPORT int32_t oracle_query_exec_hist(const fdb_view_t* view, const fdb_query_t* q,
                                       int32_t nb,
                                       double* out_bucket_sums, double* out_counts,
@@ -1276,5 +1287,153 @@ EXPORT int32_t oracle_query_exec_hist(const fdb_view_t* view, const fdb_query_t*
   }
   if (!out_bucket_sums) free(sums);
   if (!out_counts) free(cnts);
+  return 0;
+}
+
+
+EXPORT int32_t oracle_query_exec_hist_mm(const fdb_view_t* view, const fdb_query_t* q,
+                                         int32_t nb, double* sums, double* cnts,
+                                         double* out_max, double* out_min,
+                                         double* out_quantile) {
+  int nw = (int)((q->end - q->start) / q->step) + 1;
+  int ng = q->num_groups;
+  size_t cells = (size_t)ng * nw;
+  int is_rate = q->func_id == 11; /* FN_HIST_RATE */
+  for (size_t i = 0; i < cells * nb; i++) sums[i] = 0;
+  for (size_t i = 0; i < cells; i++) cnts[i] = 0;
+  if (out_max) for (size_t i = 0; i < cells; i++) out_max[i] = NAN;
+  if (out_min) for (size_t i = 0; i < cells; i++) out_min[i] = NAN;
+
+  int64_t* lastv = (int64_t*)malloc((size_t)nb * 8);
+  int64_t* corr  = (int64_t*)malloc((size_t)nb * 8);
+  double* lo = (double*)malloc((size_t)nb * 8);
+  double* hi = (double*)malloc((size_t)nb * 8);
+  double* wsum = (double*)malloc((size_t)nb * 8);
+  for (int sid = 0; sid < view->num_series; sid++) {
+    int first = view->series_first[sid];
+    int nchunks = view->series_nchunks[sid];
+    if (nchunks > 64) nchunks = 64;
+    const fdb_dir_entry_t* dir = view->dir + first;
+    vec_t tsv[64], mxv[64], mnv[64];
+    int have_mm[64];
+    hist_chunk_t hv[64];
+    for (int c = 0; c < nchunks; c++) {
+      vec_open(view->blob + dir[c].ts_off, &tsv[c]);
+      have_mm[c] = dir[c].max_off != 0;
+      if (have_mm[c]) {
+        vec_open(view->blob + dir[c].max_off, &mxv[c]);
+        vec_open(view->blob + dir[c].min_off, &mnv[c]);
+      }
+      if (hist_open(view->blob + dir[c].val_off, &hv[c]) != 0) {
+        for (int c2 = 0; c2 < c; c2++) hist_close(&hv[c2]);
+        free(lastv); free(corr); free(lo); free(hi); free(wsum);
+        return -1;
+      }
+    }
+    int grp = view->group_ids[sid];
+    for (int w = 0; w < nw; w++) {
+      int64_t wEnd = q->start + (int64_t)w * q->step;
+      int64_t wStart = wEnd - q->window;
+      int meta_has = 0;
+      memset(corr, 0, (size_t)nb * 8);
+      int numSamples = 0;
+      int64_t lowestTime = INT64_MAX, highestTime = 0;
+      int have_lo = 0, have_hi = 0;
+      int have_sum = 0;
+      double wmax = NAN, wmin = NAN;
+      for (int b = 0; b < nb; b++) wsum[b] = 0;
+      for (int c = 0; c < nchunks; c++) {
+        if (dir[c].end_time < wStart) continue;
+        const vec_t* tv = &tsv[c];
+        const hist_chunk_t* h = &hv[c];
+        int startRow = lv_binary_search(tv, wStart) & 0x7fffffff;
+        int endRow = lv_ceiling(tv, wEnd);
+        if (endRow > dir[c].num_rows - 1) endRow = dir[c].num_rows - 1;
+        if (is_rate && meta_has) {
+          if (hist_less(h->cum, lastv, nb))
+            for (int b = 0; b < nb; b++) corr[b] += lastv[b];
+        }
+        if (startRow <= endRow) {
+          if (is_rate) {
+            int64_t st = lv_at(tv, startRow), en = lv_at(tv, endRow);
+            if (st < lowestTime || en > highestTime) {
+              numSamples += endRow - startRow + 1;
+              if (st < lowestTime) {
+                lowestTime = st;
+                for (int b = 0; b < nb; b++)
+                  lo[b] = (double)(h->cum[(size_t)startRow * nb + b]
+                                   + h->corr[(size_t)startRow * nb + b] + corr[b]);
+                have_lo = 1;
+              }
+              if (en > highestTime) {
+                highestTime = en;
+                for (int b = 0; b < nb; b++)
+                  hi[b] = (double)(h->cum[(size_t)endRow * nb + b]
+                                   + h->corr[(size_t)endRow * nb + b] + corr[b]);
+                have_hi = 1;
+              }
+            }
+          } else {
+            /* SumOverTimeChunkedFunctionH: raw histogram add per row */
+            for (int i = startRow; i <= endRow; i++)
+              for (int b = 0; b < nb; b++)
+                wsum[b] += (double)h->cum[(size_t)i * nb + b];
+            have_sum = 1;
+          }
+          if (have_mm[c]) {
+            for (int i = startRow; i <= endRow; i++) {
+              double mx = dv_at(&mxv[c], i);
+              double mn = dv_at(&mnv[c], i);
+              if (!isnan(mx) && (isnan(wmax) || mx > wmax)) wmax = mx;
+              if (!isnan(mn) && (isnan(wmin) || mn < wmin)) wmin = mn;
+            }
+          }
+        }
+        if (is_rate) {
+          for (int b = 0; b < nb; b++) {
+            corr[b] += h->chunk_corr[b];
+            lastv[b] = h->cum[(size_t)(h->n - 1) * nb + b];
+          }
+          meta_has = 1;
+        }
+        if (dir[c].end_time >= wEnd) break;
+      }
+      size_t cell = (size_t)grp * nw + w;
+      if (is_rate) {
+        if (highestTime > lowestTime && have_lo && have_hi) {
+          for (int b = 0; b < nb; b++)
+            sums[cell * nb + b] += extrapolated_rate(wStart, wEnd, numSamples,
+                                                     lowestTime, lo[b],
+                                                     highestTime, hi[b], 1, 1);
+          cnts[cell] += 1;
+        }
+      } else if (have_sum) {
+        for (int b = 0; b < nb; b++) sums[cell * nb + b] += wsum[b];
+        cnts[cell] += 1;
+      }
+      /* HistMaxMinSumAggregator: maxIgnoreNaN / minIgnoreNaN merges */
+      if (out_max && !isnan(wmax) &&
+          (isnan(out_max[cell]) || wmax > out_max[cell])) out_max[cell] = wmax;
+      if (out_min && !isnan(wmin) &&
+          (isnan(out_min[cell]) || wmin < out_min[cell])) out_min[cell] = wmin;
+    }
+    for (int c = 0; c < nchunks; c++) hist_close(&hv[c]);
+  }
+  free(lastv); free(corr); free(lo); free(hi); free(wsum);
+
+  if (out_quantile) {
+    double first = 0, mult = 0;
+    for (int sid = 0; sid < view->num_series && first == 0; sid++) {
+      const fdb_dir_entry_t* dir = view->dir + view->series_first[sid];
+      hist_chunk_t hc;
+      if (hist_open(view->blob + dir[0].val_off, &hc) == 0) {
+        first = hc.first; mult = hc.mult;
+        hist_close(&hc);
+      }
+    }
+    for (size_t i = 0; i < cells; i++)
+      out_quantile[i] = cnts[i] > 0
+        ? oracle_hist_quantile(q->param, sums + i * nb, nb, first, mult) : NAN;
+  }
   return 0;
 }

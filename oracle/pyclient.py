@@ -106,6 +106,30 @@ def hist_quantile(q, values, first, mult):
     return L.oracle_hist_quantile(q, v.ctypes.data_as(_c_double_p), len(v), first, mult)
 
 
+def query_exec_hist_mm(view, q, num_buckets, with_quantile=True):
+    import ctypes as ct
+    nw = q.num_windows
+    ng = q.num_groups
+    sums = np.zeros(ng * nw * num_buckets, dtype=np.float64)
+    cnts = np.zeros(ng * nw, dtype=np.float64)
+    mx = np.zeros(ng * nw, dtype=np.float64)
+    mn = np.zeros(ng * nw, dtype=np.float64)
+    quant = np.zeros(ng * nw, dtype=np.float64) if with_quantile else None
+    L = lib()
+    L.oracle_query_exec_hist_mm.argtypes = [ct.c_void_p, ct.c_void_p,
+                                            ct.c_int32, _c_double_p,
+                                            _c_double_p, _c_double_p,
+                                            _c_double_p, _c_double_p]
+    rc = L.oracle_query_exec_hist_mm(
+        ct.byref(view), ct.byref(q), num_buckets,
+        sums.ctypes.data_as(_c_double_p), cnts.ctypes.data_as(_c_double_p),
+        mx.ctypes.data_as(_c_double_p), mn.ctypes.data_as(_c_double_p),
+        quant.ctypes.data_as(_c_double_p) if quant is not None else None)
+    if rc != 0:
+        raise RuntimeError(f"oracle_query_exec_hist_mm failed rc={rc}")
+    return sums, cnts, mx, mn, quant
+
+
 def count_values(view, q, k_cap=64):
     import ctypes as ct
     nw = q.num_windows
