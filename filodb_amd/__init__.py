@@ -143,6 +143,14 @@ def lib():
             ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
             ctypes.c_int32, _c_double_p, _c_double_p, _c_double_p,
             _c_double_p, _c_double_p, ctypes.c_int32]
+        L.fdb_gpu_unpack_doubles_xor.argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64,
+            ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32),
+            ctypes.POINTER(ctypes.c_int64), ctypes.c_int32, _c_double_p,
+            ctypes.c_int64]
+        L.fdb_nibblepack_unpack_doubles.argtypes = [
+            ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32, _c_double_p,
+            ctypes.c_int32]
         L.fdb_query_exec_count_values.argtypes = [
             ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
             ctypes.c_int32, _c_double_p, _c_double_p,
@@ -385,6 +393,42 @@ def make_query(start, step, end, window, func_id, agg_id=AGG_NONE, num_groups=0,
     q.func_id, q.agg_id, q.num_groups = func_id, agg_id, num_groups
     q.param = param
     return q
+
+
+def nibblepack_unpack_doubles(data, n):
+    """Host-side unpackDoubleXOR (NibblePack.scala:360-394)."""
+    buf = (ctypes.c_uint8 * len(data)).from_buffer_copy(data)
+    out = np.empty(n, dtype=np.float64)
+    _check(lib().fdb_nibblepack_unpack_doubles(buf, len(data), _as_f64_ptr(out), n),
+           "unpack_doubles")
+    return out
+
+
+def gpu_unpack_doubles_xor(engine, streams):
+    """Decodes a list of packed XOR streams [(bytes, n), ...] on the GPU
+    (wavefront prefix-XOR kernel). Returns a list of numpy arrays."""
+    blob = b"".join(s for s, _ in streams)
+    offs, counts, out_offs = [], [], []
+    pos = 0
+    opos = 0
+    for s, n in streams:
+        offs.append(pos)
+        counts.append(n)
+        out_offs.append(opos)
+        pos += len(s)
+        opos += n
+    b = (ctypes.c_uint8 * len(blob)).from_buffer_copy(blob)
+    offs_a = np.array(offs, dtype=np.int64)
+    cnts_a = np.array(counts, dtype=np.int32)
+    oo_a = np.array(out_offs, dtype=np.int64)
+    out = np.empty(opos, dtype=np.float64)
+    _check(lib().fdb_gpu_unpack_doubles_xor(
+        engine._h, b, len(blob),
+        offs_a.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        cnts_a.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+        oo_a.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        len(streams), _as_f64_ptr(out), opos), "gpu_unpack_doubles_xor")
+    return [out[o:o + n] for o, n in zip(out_offs, counts)]
 
 
 def nibblepack_pack8(vals8):

@@ -159,6 +159,30 @@ extern "C" int32_t fdb_nibblepack_pack_doubles(const double* in, int32_t n, uint
   return (int32_t)len;
 }
 
+// unpackDoubleXOR (NibblePack.scala:360-394): product-side host decoder for
+// Gorilla-XOR packed double streams (first double raw, rest XOR-chained
+// through 8-value groups).
+extern "C" int32_t fdb_nibblepack_unpack8(const uint8_t* in, int32_t inlen, int64_t out[8],
+                                          int32_t* consumed);
+extern "C" int32_t fdb_nibblepack_unpack_doubles(const uint8_t* in, int32_t inlen,
+                                                 double* out, int32_t n) {
+  if (n < 1 || inlen < 8) { fdb_set_error("stream too short"); return FDB_ERR_BADARG; }
+  uint64_t last; memcpy(&last, in, 8);
+  memcpy(&out[0], &last, 8);
+  int32_t pos = 8, i = 1;
+  while (i < n) {
+    int64_t grp[8]; int32_t consumed;
+    if (fdb_nibblepack_unpack8(in + pos, inlen - pos, grp, &consumed) != FDB_OK)
+      return FDB_ERR_BADARG;
+    pos += consumed;
+    for (int k = 0; k < 8 && i < n; k++, i++) {
+      last ^= (uint64_t)grp[k];
+      memcpy(&out[i], &last, 8);
+    }
+  }
+  return FDB_OK;
+}
+
 // unpack8 (NibblePack.scala:395-447). Returns FDB_OK and *consumed, or error.
 extern "C" int32_t fdb_nibblepack_unpack8(const uint8_t* in, int32_t inlen, int64_t out[8],
                                           int32_t* consumed) {

@@ -588,6 +588,8 @@ extern "C" void fdb_engine_destroy(fdb_engine_t* e) {
   delete e;
 }
 
+hipStream_t fdb_engine_stream(fdb_engine_t* e) { return e->stream; }
+
 extern "C" int32_t fdb_engine_synchronize(fdb_engine_t* e) {
   HIP_CHECK(hipStreamSynchronize(e->stream));
   return FDB_OK;

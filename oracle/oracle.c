@@ -1055,6 +1055,27 @@ typedef struct {
 } hist_chunk_t;
 
 /* NibblePack delta-decode nb longs from a stream (DeltaSink semantics) */
+/* unpackDoubleXOR (NibblePack.scala:360-394): first double raw, the rest
+ * XOR-chained through 8-value nibble-packed groups */
+EXPORT int32_t oracle_nibblepack_unpack_doubles(const uint8_t* in, int32_t inlen,
+                                                double* out, int32_t n) {
+  if (n < 1 || inlen < 8) return -1;
+  uint64_t last; memcpy(&last, in, 8);
+  memcpy(&out[0], &last, 8);
+  int pos = 8, i = 1;
+  while (i < n) {
+    int64_t grp[8]; int consumed;
+    if (oracle_nibblepack_unpack8(in + pos, inlen - pos, grp, &consumed) != 0)
+      return -1;
+    pos += consumed;
+    for (int k = 0; k < 8 && i < n; k++, i++) {
+      last ^= (uint64_t)grp[k];
+      memcpy(&out[i], &last, 8);
+    }
+  }
+  return 0;
+}
+
 static int np_unpack_delta(const uint8_t* in, int inlen, int64_t* out, int nb) {
   int pos = 0, i = 0;
   int64_t current = 0;

@@ -6,6 +6,7 @@ core/src/test/scala/filodb.memory/format/NibblePackTest.scala:13-78 (the
 (fdb_nibblepack_*) and the test oracle (oracle_nibblepack_unpack8) are checked.
 """
 import numpy as np
+import pytest
 
 
 # NibblePackTest.scala:16-30
@@ -84,3 +85,56 @@ def test_pack_delta_roundtrip(fdb, oracle):
             vals.append(current)
         pos += consumed
     assert vals[:len(inputs)] == inputs
+
+
+def test_unpack_doubles_xor_roundtrip(fdb, oracle):
+    """§8a9 close-out: packDoubles → unpackDoubleXOR bit-exact roundtrip,
+    host/product decoder vs the oracle restatement (NibblePack.scala:360-394)."""
+    import ctypes as ct
+    rng = np.random.default_rng(11)
+    for n in (1, 2, 7, 8, 9, 63, 64, 65, 500):
+        vals = rng.normal(0, 100, n)
+        vals[rng.random(n) < 0.1] = 0.0
+        packed = _pack_doubles(fdb, vals)
+        got = fdb.nibblepack_unpack_doubles(packed, n)
+        np.testing.assert_array_equal(got, vals)
+        # oracle restatement decodes identically
+        L = oracle.lib()
+        L.oracle_nibblepack_unpack_doubles.argtypes = [
+            ct.POINTER(ct.c_uint8), ct.c_int32,
+            ct.POINTER(ct.c_double), ct.c_int32]
+        buf = (ct.c_uint8 * len(packed)).from_buffer_copy(packed)
+        out = np.empty(n, dtype=np.float64)
+        rc = L.oracle_nibblepack_unpack_doubles(
+            buf, len(packed), out.ctypes.data_as(ct.POINTER(ct.c_double)), n)
+        assert rc == 0
+        np.testing.assert_array_equal(out, vals)
+
+
+def _pack_doubles(fdb, vals):
+    import ctypes as ct
+    a = np.ascontiguousarray(vals, dtype=np.float64)
+    cap = len(a) * 10 + 64
+    out = (ct.c_uint8 * cap)()
+    n = fdb.lib().fdb_nibblepack_pack_doubles(
+        a.ctypes.data_as(ct.POINTER(ct.c_double)), len(a), out, cap)
+    assert n > 0
+    return bytes(out[:n])
+
+
+@pytest.mark.gpu
+def test_gpu_xor_decode_matches_host(fdb, oracle):
+    """The wavefront prefix-XOR kernel decodes packed streams bit-exactly
+    (vs the host decoder) across sizes and batches."""
+    rng = np.random.default_rng(12)
+    eng = fdb.Engine(0)
+    streams = []
+    wants = []
+    for n in (1, 2, 9, 64, 65, 129, 400, 1000):
+        vals = np.round(rng.normal(50, 20, n), 3)
+        vals[rng.random(n) < 0.05] = 0.0
+        streams.append((_pack_doubles(fdb, vals), n))
+        wants.append(vals)
+    outs = fdb.gpu_unpack_doubles_xor(eng, streams)
+    for got, want in zip(outs, wants):
+        np.testing.assert_array_equal(got, want)
