@@ -197,6 +197,9 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   // so prefetch loads stay in flight across the phase fences)
   int pf_sid = 0, pf_first = 0, pf_nch = 0, pf_nr = 0;
   uint64_t pf_toff = 0, pf_voff = 0;
+  DVec pf_tv, pf_vv;
+  int64_t pf_ts0 = 0;
+  bool pf_hdr = false;      // pf_tv/pf_vv/pf_ts0 hold the next series' headers
   if (pos0 < pos1) {
     pf_sid = (EMIT == 1) ? sbg[pos0] : pos0;
     pf_first = series_first[pf_sid];
@@ -226,8 +229,12 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     bool dropped = false;
     if (nch >= 1) {
       DVec tv, vv;
-      d_vec_open_wide(blob + toff, &tv, &ts0);   // one 32-B load each
-      d_vec_open_wide(blob + voff, &vv, nullptr);
+      if (pf_hdr) {                    // stage-3 prefetch landed last series
+        tv = pf_tv; vv = pf_vv; ts0 = pf_ts0;
+      } else {
+        d_vec_open_wide(blob + toff, &tv, &ts0);   // one 32-B load each
+        d_vec_open_wide(blob + voff, &vv, nullptr);
+      }
       n = nrows;
       if (n > FAST_ROWS || n > tv.n) { n = 0; ts0 = 0; }   // guarded at upload
       if (n > 0) {
@@ -564,6 +571,14 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       __builtin_amdgcn_wave_barrier();
       if (timing) { uint64_t t = __builtin_amdgcn_s_memtime(); tW += t - tt; tt = t; }
     }  // tiles
+    // prefetch stage 3: next series' 32-B vector headers (the stage-2 offset
+    // loads landed during the window phase; these fly across the loop edge so
+    // the next decode starts with only the payload round-trip outstanding)
+    pf_hdr = npos < pos1 && pf_nch >= 1;
+    if (pf_hdr) {
+      d_vec_open_wide(blob + pf_toff, &pf_tv, &pf_ts0);
+      d_vec_open_wide(blob + pf_voff, &pf_vv, nullptr);
+    }
   }  // series
   if (EMIT == 1 && cur_grp >= 0) flush_group(cur_grp);
   if (timing && EMIT == 0 && lane == 0) {
