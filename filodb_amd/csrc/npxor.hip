@@ -57,8 +57,8 @@ void xor_unpack_kernel(const uint8_t* __restrict__ blob,
       const uint8_t* gp = p + pos;
       uint32_t stg = estream_stage(gp, lane);
       const int shift = (int)((uintptr_t)gp & 3);
-      int goff[8], gbits[8], gtrail[8];
-      uint32_t gmask[8];
+      int goff[8] = {0}, gbits[8] = {0}, gtrail[8] = {0};
+      uint32_t gmask[8] = {0};
       int off = 0;
       int ngroups = 0;
       for (int g = 0; g < 8 && i + ngroups * 8 < n + 7; g++) {
@@ -107,15 +107,18 @@ void xor_unpack_kernel(const uint8_t* __restrict__ blob,
         pos += glen;
         continue;
       }
-      // lane = (group, slot): extract this lane's delta
+      // lane = (group, slot): extract this lane's delta. The estream reads
+      // are SHUFFLES, so every lane must execute them (inactive source lanes
+      // yield undefined data) — only the final delta is predicated.
       const int g = lane >> 3, k = lane & 7;
+      const bool lvalid = g < ngroups && (gmask[g] & (1u << k));
+      int slot = lvalid ? __popc(gmask[g] & ((1u << k) - 1)) : 0;
+      int bitpos = slot * gbits[g];
+      int koff = goff[g] + 2 + (bitpos >> 3);
+      uint64_t w = estream_w64(true, stg, shift, gp, koff);
+      uint32_t b8 = estream_byte(true, stg, shift, gp, koff + 8);
       uint64_t delta = 0;
-      if (g < ngroups && (gmask[g] & (1u << k))) {
-        int slot = __popc(gmask[g] & ((1u << k) - 1));
-        int bitpos = slot * gbits[g];
-        int koff = goff[g] + 2 + (bitpos >> 3);
-        uint64_t w = estream_w64(true, stg, shift, gp, koff);
-        uint32_t b8 = estream_byte(true, stg, shift, gp, koff + 8);
+      if (lvalid) {
         int sh = bitpos & 7;
         uint64_t v = w >> sh;
         if (gbits[g] > 64 - sh) v |= (uint64_t)b8 << (64 - sh);
