@@ -367,39 +367,36 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       // d_i = last  w with wStart <= ts_i = floor((o_i + Ae + qwindow) / qstep)
       // row i ends   windows [c_{i-1}, c_i)   with e = i-1   (wEnd < ts_i)
       // row i starts windows (d_{i-1}, d_i]   with s = i     (wStart > ts_{i-1})
-      // neighbor boundaries are RECOMPUTED from ts[i-1] instead of shuffled:
-      // two extra pure-ALU floor-divisions per row are cheaper than the
-      // cross-lane + batch-carry latency chains they replace
+      // (a shuffle-free variant recomputing the neighbor boundaries from
+      // ts[i-1] — 4 fdivs/row, no carries — measured SLOWER: the inversion
+      // phase grew 8.2 → 11.1 Gcyc; the shfl form stays)
+      int c_carry = 0, d_carry = -1;
       for (int base = 0; base < n; base += 64) {
         const int i = base + lane;
         const bool live = i < n;
         int64_t o = live ? (int64_t)ws.tso[i] : 0;
-        int64_t op = (live && i > 0) ? (int64_t)ws.tso[i - 1] : 0;
         int ci = live ? fdiv_floor_win(o + Ae + qstep - 1, qstep, inv_step,
                                        num_windows) : 0;
         int di = live ? fdiv_floor_win(o + Ae + qwindow, qstep, inv_step,
                                        num_windows) : 0;
+        int cprev = __shfl_up(ci, 1);
+        int dprev = __shfl_up(di, 1);
+        if (lane == 0) { cprev = c_carry; dprev = d_carry; }
         if (live) {
           if (i > 0) {
-            int cprev = fdiv_floor_win(op + Ae + qstep - 1, qstep, inv_step,
-                                       num_windows);
-            int dprev = fdiv_floor_win(op + Ae + qwindow, qstep, inv_step,
-                                       num_windows);
             int lo = max(cprev, tb), hi = min(ci, tb + tn);
             for (int w = lo; w < hi; w++) ws.ew[w - tb] = (int16_t)(i - 1);
-            lo = max(dprev + 1, tb);
-            hi = min(di, tb + tn - 1);
-            for (int w = lo; w <= hi; w++) ws.sw[w - tb] = (int16_t)i;
-          } else {
-            int hi = min(di, tb + tn - 1);
-            for (int w = tb; w <= hi; w++) ws.sw[w - tb] = (int16_t)i;
           }
+          int lo = (i == 0) ? tb : max(dprev + 1, tb);
+          int hi = min(di, tb + tn - 1);
+          for (int w = lo; w <= hi; w++) ws.sw[w - tb] = (int16_t)i;
         }
+        const int lastl = min(63, n - 1 - base);
+        c_carry = __shfl(ci, lastl);
+        d_carry = __shfl(di, lastl);
       }
       if (n > 0) {          // tail: windows with wEnd >= ts_{n-1} end at n-1
-        const int clast = fdiv_floor_win((int64_t)ws.tso[n - 1] + Ae + qstep - 1,
-                                         qstep, inv_step, num_windows);
-        for (int w = max(clast, tb) + lane; w < tb + tn; w += 64)
+        for (int w = max(c_carry, tb) + lane; w < tb + tn; w += 64)
           ws.ew[w - tb] = (int16_t)(n - 1);
       }
       d_wait_lds();
