@@ -66,13 +66,21 @@ def test_fused_group_matches_two_phase(fdb, oracle, engine, agg):
     nw = q.num_windows
     ds = engine.upload(st)
     got_fused = np.empty(n_groups * nw)
-    engine.query(ds, q, out=got_fused)
+    os.environ["FDB_FUSED_GROUP"] = "1"
+    try:
+        engine.query(ds, q, out=got_fused)
+    finally:
+        os.environ.pop("FDB_FUSED_GROUP")
+    got_two = np.empty(n_groups * nw)
+    engine.query(ds, q, out=got_two)
     want = oracle.query_exec(st.view(), q, st.num_series, nw, nthreads=4)
     check(got_fused, want)
+    check(got_two, want)
 
 
 def test_fused_group_partial_mode(fdb, oracle, engine):
     """Partial (multi-GPU merge) grids through the fused emit."""
+    os.environ["FDB_FUSED_GROUP"] = "1"
     rng = np.random.default_rng(88)
     n_groups = 5
     series, groups = [], []
@@ -87,7 +95,10 @@ def test_fused_group_partial_mode(fdb, oracle, engine):
                                        out_counts=True)
     got_s = np.empty(n_groups * nw)
     got_c = np.empty(n_groups * nw)
-    engine.query(engine.upload(st), q, out=got_s, out_counts=got_c)
+    try:
+        engine.query(engine.upload(st), q, out=got_s, out_counts=got_c)
+    finally:
+        os.environ.pop("FDB_FUSED_GROUP", None)
     check(got_s, want_s)
     np.testing.assert_array_equal(got_c, want_c)
 

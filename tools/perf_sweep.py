@@ -57,6 +57,7 @@ def main():
         os.environ["FDB_FUSED_GROUP"] = fused
         ms = bench(eng, ds, q_sum, gout, gcnt)
         print(f"sum-by-group fused={fused}: {ms:.3f} ms", flush=True)
+    os.environ.pop("FDB_FUSED_GROUP")
     # phase split (s_memtime; pm bit 3 clobbers out[] with per-wave cycles)
     os.environ["FDB_RATE_WAVES"] = "6"
     q_rate._pad = 8
@@ -81,6 +82,29 @@ def main():
     q_min = fdb.make_query(T0, 15000, T0 + span, 600000, fdb.FN_MIN_OVER_TIME)
     ms = bench(eng, ds2, q_min, out)
     print(f"min_over_time[10m]: {ms:.3f} ms", flush=True)
+
+    # hist workload (config #4 shape): v2 occupancy A/B
+    del ds2
+    st3 = build(100_000, kind=fdb.COL_HIST)
+    ds3 = eng.upload(st3)
+    qh = fdb.make_query(T0, 15000, T0 + span, 300000, fdb.FN_HIST_RATE,
+                        fdb.AGG_SUM, 1000, param=0.99)
+    nwh = qh.num_windows
+    hs = torch.zeros(1000 * nwh * 64, dtype=torch.float64, device="cuda:0")
+    hc = torch.zeros(1000 * nwh, dtype=torch.float64, device="cuda:0")
+    hq = torch.zeros(1000 * nwh, dtype=torch.float64, device="cuda:0")
+    for hw in ("4", "5"):
+        os.environ["FDB_HIST_WAVES"] = hw
+        eng.query_hist(ds3, qh, 64, out_bucket_sums=hs, out_counts=hc,
+                       out_quantile=hq, on_device=True)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(3):
+            eng.query_hist(ds3, qh, 64, out_bucket_sums=hs, out_counts=hc,
+                           out_quantile=hq, on_device=True)
+        torch.cuda.synchronize()
+        print(f"hist waves={hw}: {(time.perf_counter() - t0) / 3 * 1000:.3f} ms",
+              flush=True)
 
 
 if __name__ == "__main__":
