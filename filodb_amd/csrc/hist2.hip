@@ -323,15 +323,16 @@ int32_t fdb_launch_hist2(hipStream_t stream, const uint8_t* blob, DirSoA dir,
   if (const char* g = getenv("FDB_HIST_GRID")) cap = atoi(g);
   if (cap > 0 && grid > cap) grid = cap;
   const char* hw = getenv("FDB_HIST_WAVES");   // occupancy experiment knob
+  const bool w4 = hw && atoi(hw) == 4;         // 5 waves/SIMD measured best
   #define H2ARGS blob, dir, max_off, min_off, series_first, series_nchunks,       group_ids, num_series, qstart, qstep, qwindow, num_windows, nb,       out_sums, out_cnt, out_max, out_min
   if (hfunc == 1)
     hipLaunchKernelGGL((hist2_kernel<4, 1>), dim3(grid), dim3(H2_WAVES * 64), 0,
                        stream, H2ARGS);
-  else if (hw && atoi(hw) == 5)
-    hipLaunchKernelGGL((hist2_kernel<5, 0>), dim3(grid), dim3(H2_WAVES * 64), 0,
+  else if (w4)
+    hipLaunchKernelGGL((hist2_kernel<4, 0>), dim3(grid), dim3(H2_WAVES * 64), 0,
                        stream, H2ARGS);
   else
-    hipLaunchKernelGGL((hist2_kernel<4, 0>), dim3(grid), dim3(H2_WAVES * 64), 0,
+    hipLaunchKernelGGL((hist2_kernel<5, 0>), dim3(grid), dim3(H2_WAVES * 64), 0,
                        stream, H2ARGS);
   #undef H2ARGS
   hipError_t e = hipGetLastError();
