@@ -120,7 +120,8 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                   double* __restrict__ out_sums,   // [G × W × nb]
                   double* __restrict__ out_cnt,    // [G × W]
                   double* __restrict__ out_max,    // [G × W] or null
-                  double* __restrict__ out_min) {  // [G × W] or null
+                  double* __restrict__ out_min,    // [G × W] or null
+                  int abl) {   // perf ablation: 1=skip emits, 2=skip decodes
   __shared__ int64_t tsS_all[H2_WAVES][H2_ROWS];
   __shared__ int64_t tsE_all[H2_WAVES][H2_ROWS];
   const int wave = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
@@ -158,6 +159,7 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     // a section base element's value is the scan of its own deltas)
     auto decode_cur = [&](H2Cursor& cu) {
       if (cu.decoded) return;
+      if (abl & 2) { cu.decoded = true; cu.val_b = 0; cu.base_b = 0; return; }
       int64_t delta = h2_parse(cu.ep, cu.elen, nb, b, live, lane);
       int64_t scan = wave_incl_scan_i64(live ? delta : 0, lane);
       if (cu.sect_first) { cu.val_b = (double)scan; cu.base_b = cu.val_b; }
@@ -285,6 +287,7 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         decode_cur(S);
         decode_cur(E);
         const int numSamples = E.e_global - S.e_global + 1;
+        if (abl & 1) continue;
         if (live) {
           // corrections relative to the window's first chunk: the reference's
           // per-window CorrectionMeta starts NoCorrection there
@@ -324,7 +327,9 @@ int32_t fdb_launch_hist2(hipStream_t stream, const uint8_t* blob, DirSoA dir,
   if (cap > 0 && grid > cap) grid = cap;
   const char* hw = getenv("FDB_HIST_WAVES");   // occupancy experiment knob
   const bool w4 = hw && atoi(hw) == 4;         // 5 waves/SIMD measured best
-  #define H2ARGS blob, dir, max_off, min_off, series_first, series_nchunks,       group_ids, num_series, qstart, qstep, qwindow, num_windows, nb,       out_sums, out_cnt, out_max, out_min
+  const char* ab = getenv("FDB_HIST_ABLATE");  // 1=skip emits, 2=skip decodes
+  const int abl = ab ? atoi(ab) : 0;
+  #define H2ARGS blob, dir, max_off, min_off, series_first, series_nchunks,       group_ids, num_series, qstart, qstep, qwindow, num_windows, nb,       out_sums, out_cnt, out_max, out_min, abl
   if (hfunc == 1)
     hipLaunchKernelGGL((hist2_kernel<4, 1>), dim3(grid), dim3(H2_WAVES * 64), 0,
                        stream, H2ARGS);

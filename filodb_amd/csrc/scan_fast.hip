@@ -108,7 +108,7 @@ __device__ __forceinline__ double f_corr_at(const WS& ws, int dcount, bool dense
 // registers (4 waves); K_RATE (no reciprocal table, LDS fits 6 blocks/CU) is
 // built at BOTH 5 (96 VGPR, no spills) and 6 (80 VGPR, ~19 spilled) — the
 // launcher picks via FDB_RATE_WAVES, measured on hardware.
-template <int FUNC, int EMIT, int MINW, bool TIMED = false>
+template <int FUNC, int EMIT, int MINW, bool TIMED = false, bool PF = true>
 __global__ __launch_bounds__(FAST_WAVES * 64, MINW)
 void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                       const int32_t* __restrict__ series_first,
@@ -200,7 +200,7 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   DVec pf_tv, pf_vv;
   int64_t pf_ts0 = 0;
   bool pf_hdr = false;      // pf_tv/pf_vv/pf_ts0 hold the next series' headers
-  if (pos0 < pos1) {
+  if (PF && pos0 < pos1) {
     pf_sid = (EMIT == 1) ? sbg[pos0] : pos0;
     pf_first = series_first[pf_sid];
     pf_nch = series_nchunks[pf_sid];
@@ -210,14 +210,23 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   }
   for (int pos = pos0; pos < pos1; pos += pos_step) {
     if (timing) tt = __builtin_amdgcn_s_memtime();
-    const int sid = pf_sid;
-    const int nch = pf_nch;
-    const uint64_t toff = pf_toff, voff = pf_voff;
-    const int nrows = pf_nr;
+    int sid, nch, nrows;
+    uint64_t toff, voff;
+    if (PF) {
+      sid = pf_sid; nch = pf_nch; nrows = pf_nr;
+      toff = pf_toff; voff = pf_voff;
+    } else {
+      sid = (EMIT == 1) ? sbg[pos] : pos;
+      const int first0 = series_first[sid];
+      nch = series_nchunks[sid];
+      toff = dir.ts_off[first0];
+      voff = dir.val_off[first0];
+      nrows = dir.num_rows[first0];
+    }
     // prefetch stage 1: next series' directory row ids
     const int npos = pos + pos_step;
     int nsid = 0, nfirst = 0, nnch = 0;
-    if (npos < pos1) {
+    if (PF && npos < pos1) {
       nsid = (EMIT == 1) ? sbg[npos] : npos;
       nfirst = series_first[nsid];
       nnch = series_nchunks[nsid];
@@ -245,7 +254,7 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     }
     // prefetch stage 2: next series' chunk offsets (nfirst landed during the
     // decode above; these land during this series' meta/window phases)
-    if (npos < pos1) {
+    if (PF && npos < pos1) {
       pf_sid = nsid; pf_first = nfirst; pf_nch = nnch;
       pf_toff = dir.ts_off[nfirst];
       pf_voff = dir.val_off[nfirst];
@@ -577,7 +586,7 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     // prefetch stage 3: next series' 32-B vector headers (the stage-2 offset
     // loads landed during the window phase; these fly across the loop edge so
     // the next decode starts with only the payload round-trip outstanding)
-    pf_hdr = npos < pos1 && pf_nch >= 1;
+    pf_hdr = PF && npos < pos1 && pf_nch >= 1;
     if (pf_hdr) {
       d_vec_open_wide(blob + pf_toff, &pf_tv, &pf_ts0);
       d_vec_open_wide(blob + pf_voff, &pf_vv, nullptr);
@@ -638,8 +647,10 @@ int32_t fdb_launch_fast_scan(hipStream_t stream, const uint8_t* blob, DirSoA dir
       else if (timed) hipLaunchKernelGGL((fast_scan_kernel<FN_RATE, 0, 5, true>),
                                          dim3(grid), dim3(FAST_WAVES * 64), 0,
                                          stream, FARGS);
-      else if (rate_w == 5) LAUNCH(FN_RATE, 0, 5);
-      else LAUNCH(FN_RATE, 0, 6);
+      else if (rate_w == 6)   // 6 waves/SIMD: prefetch stripped (register diet)
+        hipLaunchKernelGGL((fast_scan_kernel<FN_RATE, 0, 6, false, false>),
+                           dim3(grid), dim3(FAST_WAVES * 64), 0, stream, FARGS);
+      else LAUNCH(FN_RATE, 0, 5);
       break;
     case FN_AVG:
       if (emit_group) LAUNCH(FN_AVG, 1, 4);
