@@ -93,8 +93,7 @@ def main():
     hs = torch.zeros(1000 * nwh * 64, dtype=torch.float64, device="cuda:0")
     hc = torch.zeros(1000 * nwh, dtype=torch.float64, device="cuda:0")
     hq = torch.zeros(1000 * nwh, dtype=torch.float64, device="cuda:0")
-    for hw in ("4", "5"):
-        os.environ["FDB_HIST_WAVES"] = hw
+    def hist_ms(tag):
         eng.query_hist(ds3, qh, 64, out_bucket_sums=hs, out_counts=hc,
                        out_quantile=hq, on_device=True)
         torch.cuda.synchronize()
@@ -103,8 +102,16 @@ def main():
             eng.query_hist(ds3, qh, 64, out_bucket_sums=hs, out_counts=hc,
                            out_quantile=hq, on_device=True)
         torch.cuda.synchronize()
-        print(f"hist waves={hw}: {(time.perf_counter() - t0) / 3 * 1000:.3f} ms",
+        print(f"hist {tag}: {(time.perf_counter() - t0) / 3 * 1000:.3f} ms",
               flush=True)
+    for hw in ("4", "5"):
+        os.environ["FDB_HIST_WAVES"] = hw
+        hist_ms(f"waves={hw}")
+    os.environ.pop("FDB_HIST_WAVES")
+    for abl in ("1", "3"):   # 1 = no emits; 3 = no emits + no decodes
+        os.environ["FDB_HIST_ABLATE"] = abl
+        hist_ms(f"ablate={abl}")
+    os.environ.pop("FDB_HIST_ABLATE")
 
 
 if __name__ == "__main__":
