@@ -52,12 +52,12 @@ __device__ __forceinline__ int64_t h2_parse_buf(const uint8_t* ep, int elen,
   int off = 0, gOff = 0, gBits = 0, gTrail = 0;
   uint32_t gMask = 0;
   for (int g = 0; g * 8 < nb; g++) {
-    uint32_t mask = estream_byte(est, ebuf, eshift, ep, 2 + off);
+    uint32_t mask = estream_byte_uni(est, ebuf, eshift, ep, 2 + off);
     int numBits = 0, trail = 0, glen;
     if (mask == 0) {
       glen = 1;
     } else {
-      int widths = (int)estream_byte(est, ebuf, eshift, ep, 2 + off + 1);
+      int widths = (int)estream_byte_uni(est, ebuf, eshift, ep, 2 + off + 1);
       numBits = ((widths >> 4) + 1) * 4;
       trail = (widths & 0x0f) * 4;
       glen = 2 + (numBits * __popc(mask) + 7) / 8;

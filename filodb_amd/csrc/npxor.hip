@@ -62,12 +62,12 @@ void xor_unpack_kernel(const uint8_t* __restrict__ blob,
       int off = 0;
       int ngroups = 0;
       for (int g = 0; g < 8 && i + ngroups * 8 < n + 7; g++) {
-        uint32_t mask = estream_byte(true, stg, shift, gp, off);
+        uint32_t mask = estream_byte_uni(true, stg, shift, gp, off);
         int numBits = 0, trail = 0, glen;
         if (mask == 0) {
           glen = 1;
         } else {
-          int widths = (int)estream_byte(true, stg, shift, gp, off + 1);
+          int widths = (int)estream_byte_uni(true, stg, shift, gp, off + 1);
           numBits = ((widths >> 4) + 1) * 4;
           trail = (widths & 0x0f) * 4;
           glen = 2 + (numBits * __popc(mask) + 7) / 8;

@@ -387,6 +387,20 @@ __device__ __forceinline__ uint32_t estream_u16(bool staged, uint32_t buf,
   return estream_byte(staged, buf, shift, g, k) |
          (estream_byte(staged, buf, shift, g, k + 1) << 8);
 }
+// wave-UNIFORM byte read from the staged window: kk>>2 is the same for every
+// lane, so v_readlane (scalar result, no LDS pipe) replaces ds_bpermute —
+// the header walks of the NibblePack parsers are built from these
+__device__ __forceinline__ uint32_t estream_byte_uni(bool staged, uint32_t buf,
+                                                     int shift, const uint8_t* g,
+                                                     int k) {
+  if (staged) {
+    int kk = k + shift;
+    uint32_t d = __builtin_amdgcn_readlane(buf, kk >> 2);
+    return (d >> ((kk & 3) * 8)) & 0xff;
+  }
+  return g[k];
+}
+
 // little-endian 8-byte window at offset k (per-lane k; bpermute shuffles)
 __device__ __forceinline__ uint64_t estream_w64(bool staged, uint32_t buf,
                                                 int shift, const uint8_t* g,
