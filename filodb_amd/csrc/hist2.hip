@@ -93,6 +93,8 @@ __device__ __forceinline__ int64_t h2_parse(const uint8_t* ep, int elen,
 struct H2Cursor {
   const uint8_t* ep;     // current element start (u16 len prefix)
   const uint8_t* sp;     // NEXT section header
+  const uint8_t* sb;     // 4-aligned base of the 256-B register stage (null = none)
+  uint32_t sbuf;         // this lane's dword of the stage
   int elen;              // current element's payload length
   int sect_left;         // elements left in section AFTER the current one
   int c;                 // chunk index within the series
@@ -144,6 +146,30 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     if (nchunks < 1) continue;
     const int grp = group_ids[sid];
 
+    // per-cursor persistent 256-B register stage: one coalesced global load
+    // serves the element hops, section headers and parses of the next ~4-8
+    // elements (the raw dependent d_u16 hop loads were the serial chain —
+    // ~16k cycles per element of wall, sweep ablate data). Returns the byte
+    // offset of ptr within the stage, restaging when [ptr, ptr+need) leaves it.
+    auto ensure_stage = [&](H2Cursor& cu, const uint8_t* ptr, int need) -> int {
+      int off = (int)(ptr - cu.sb);
+      if (cu.sb == nullptr || off < 0 || off + need > 256) {
+        cu.sb = (const uint8_t*)((uintptr_t)ptr & ~(uintptr_t)3);
+        cu.sbuf = estream_stage(ptr, lane);
+        off = (int)(ptr - cu.sb);
+      }
+      return off;
+    };
+    auto staged_byte = [&](H2Cursor& cu, const uint8_t* ptr) -> uint32_t {
+      int off = ensure_stage(cu, ptr, 1);
+      return estream_byte_uni(true, cu.sbuf, 0, ptr, off);
+    };
+    auto staged_u16 = [&](H2Cursor& cu, const uint8_t* ptr) -> uint32_t {
+      int off = ensure_stage(cu, ptr, 2);
+      return estream_byte_uni(true, cu.sbuf, 0, ptr, off) |
+             (estream_byte_uni(true, cu.sbuf, 0, ptr, off + 1) << 8);
+    };
+
     // open a chunk for a cursor: decode its timestamps into the cursor's LDS
     // buffer and position at element 0 (section base, decoded)
     auto open_chunk = [&](H2Cursor& cu, int c, int64_t* tsbuf) {
@@ -161,6 +187,7 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       cu.e_local = -1;
       cu.sect_left = 0;
       cu.decoded = false;
+      cu.sb = nullptr;
     };
 
     // decode the current element (value = section base + scanned deltas;
@@ -168,7 +195,9 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     auto decode_cur = [&](H2Cursor& cu) {
       if (cu.decoded) return;
       if (abl & 2) { cu.decoded = true; cu.val_b = 0; cu.base_b = 0; return; }
-      int64_t delta = h2_parse(cu.ep, cu.elen, nb, b, live, lane);
+      const int off = ensure_stage(cu, cu.ep, min(cu.elen + 14, 256));
+      int64_t delta = h2_parse_buf(cu.ep, cu.elen, nb, b, live, lane,
+                                   cu.sbuf, off);
       int64_t scan = wave_incl_scan_i64(live ? delta : 0, lane);
       if (cu.sect_first) { cu.val_b = (double)scan; cu.base_b = cu.val_b; }
       else cu.val_b = cu.base_b + (double)scan;
@@ -191,16 +220,17 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       if (cu.sect_left == 0) {
         // entering a section; TypeDrop adds apply(e-1) (the element we just
         // ensured is decoded below / at the chunk seam above)
-        int stype = cu.sp[3];
-        cu.sect_left = cu.sp[2];
+        int stype = (int)staged_byte(cu, cu.sp + 3);
+        cu.sect_left = (int)staged_byte(cu, cu.sp + 2);
+        int slen = (int)staged_u16(cu, cu.sp);
         cu.ep = cu.sp + 4;
-        cu.sp += 4 + d_u16(cu.sp);
+        cu.sp += 4 + slen;
         new_sect = true;
         if (stype == 1 && cu.e_local >= 0) cu.C_b += cu.val_b;
       } else {
         cu.ep += 2 + cu.elen;
       }
-      cu.elen = (int)d_u16(cu.ep);
+      cu.elen = (int)staged_u16(cu, cu.ep);
       cu.sect_left--;
       cu.e_local++;
       cu.e_global++;
@@ -296,12 +326,12 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         // two (independent) header-parse chains interleave — the per-element
         // parse latency was 19 of the 36 ms on config #4 (sweep ablate=3)
         if (!S.decoded && !E.decoded && !(abl & 2)) {
-          uint32_t bufS = estream_stage(S.ep, lane);
-          uint32_t bufE = estream_stage(E.ep, lane);
-          int64_t dS = h2_parse_buf(S.ep, S.elen, nb, b, live, lane, bufS,
-                                    (int)((uintptr_t)S.ep & 3));
-          int64_t dE = h2_parse_buf(E.ep, E.elen, nb, b, live, lane, bufE,
-                                    (int)((uintptr_t)E.ep & 3));
+          const int offS = ensure_stage(S, S.ep, min(S.elen + 14, 256));
+          const int offE = ensure_stage(E, E.ep, min(E.elen + 14, 256));
+          int64_t dS = h2_parse_buf(S.ep, S.elen, nb, b, live, lane, S.sbuf,
+                                    offS);
+          int64_t dE = h2_parse_buf(E.ep, E.elen, nb, b, live, lane, E.sbuf,
+                                    offE);
           int64_t sS = wave_incl_scan_i64(live ? dS : 0, lane);
           int64_t sE = wave_incl_scan_i64(live ? dE : 0, lane);
           if (S.sect_first) { S.val_b = (double)sS; S.base_b = S.val_b; }
