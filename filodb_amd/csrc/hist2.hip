@@ -138,6 +138,8 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
   const int lane = threadIdx.x & 63;
   const int b = lane;
   const bool live = b < nb;
+  const bool tmg = (abl & 8) != 0;   // s_memtime phase split into out_cnt[0..3]
+  uint64_t tAdv = 0, tDec = 0, tEmit = 0, tOther = 0, tc = 0;
 
   for (int sid = blockIdx.x * H2_WAVES + wave; sid < num_series;
        sid += gridDim.x * H2_WAVES) {
@@ -269,6 +271,7 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     bool s_more = true, e_more = true;
 
     for (int w = 0; w < num_windows; w++) {
+      if (tmg) tc = __builtin_amdgcn_s_memtime();
       const int64_t wEnd = qstart + (int64_t)w * qstep;
       const int64_t wStart = wEnd - qwindow;
       // E → last element with ts <= wEnd
@@ -284,6 +287,7 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       while (s_more && tsS_all[wave][S.e_local] < wStart) {
         if (!step(S, tsS_all[wave])) { s_more = false; break; }
       }
+      if (tmg) { uint64_t t2m = __builtin_amdgcn_s_memtime(); tAdv += t2m - tc; tc = t2m; }
       (void)e_more;
       if (!s_more) break;                     // no element >= wStart: done
       const int64_t t1 = tsS_all[wave][S.e_local];
@@ -342,6 +346,7 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
         }
         decode_cur(S);
         decode_cur(E);
+        if (tmg) { uint64_t t2m = __builtin_amdgcn_s_memtime(); tDec += t2m - tc; tc = t2m; }
         const int numSamples = E.e_global - S.e_global + 1;
         if (abl & 1) continue;
         if (live) {
@@ -354,6 +359,7 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
           atomicAdd(&out_sums[cell * nb + b], r);
         }
         if (lane == 0) atomicAdd(&out_cnt[cell], 1.0);
+        if (tmg) { uint64_t t2m = __builtin_amdgcn_s_memtime(); tEmit += t2m - tc; tc = t2m; }
       } else {
         // SumOverTime: prefix difference over [S..E] inclusive
         if (live) {
@@ -365,6 +371,12 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     }
     d_wait_lds();
     __builtin_amdgcn_wave_barrier();
+  }
+  if (tmg && lane == 0) {
+    atomicAdd(&out_cnt[0], (double)tAdv);
+    atomicAdd(&out_cnt[1], (double)tDec);
+    atomicAdd(&out_cnt[2], (double)tEmit);
+    atomicAdd(&out_cnt[3], (double)tOther);
   }
 }
 

@@ -111,6 +111,17 @@ def main():
     for abl in ("1", "3"):   # 1 = no emits; 3 = no emits + no decodes
         os.environ["FDB_HIST_ABLATE"] = abl
         hist_ms(f"ablate={abl}")
+    # in-kernel phase split (abl bit 3): advance / decode / emit cycles
+    os.environ["FDB_HIST_ABLATE"] = "8"
+    hc.zero_()
+    eng.query_hist(ds3, qh, 64, out_bucket_sums=hs, out_counts=hc,
+                   out_quantile=None, on_device=True)
+    eng.synchronize()
+    ph = hc[:4].cpu().numpy()
+    tot = ph.sum()
+    print("hist phase split: advance=%.0f%% decode=%.0f%% emit=%.0f%% "
+          "(sum %.2f Gcyc)" % (ph[0]/tot*100, ph[1]/tot*100, ph[2]/tot*100,
+                               tot/1e9), flush=True)
     os.environ.pop("FDB_HIST_ABLATE")
 
 
