@@ -112,7 +112,7 @@ def main():
         os.environ["FDB_HIST_ABLATE"] = abl
         hist_ms(f"ablate={abl}")
     # in-kernel phase split (abl bit 3): advance / decode / emit cycles
-    os.environ["FDB_HIST_ABLATE"] = "8"
+    os.environ["FDB_HIST_ABLATE"] = "9"   # timing + no emits: out_cnt holds only cycles
     hc.zero_()
     eng.query_hist(ds3, qh, 64, out_bucket_sums=hs, out_counts=hc,
                    out_quantile=None, on_device=True)
