@@ -1,6 +1,7 @@
 #!/usr/bin/env python3
-"""Phase ablation of the scan kernel on the 1M-series rate workload.
-Phases: 3=full, 1=decode-only, 2=window-only (synthesized LDS rows)."""
+"""ROUND-1 tool (stale): phase ablation of the deleted round-1 scan kernel
+via q._pad bits 1/2. The round-2 fast kernel honors only bit 3 (s_memtime
+split — see tools/perf_sweep.py, which supersedes this)."""
 import os
 import sys
 

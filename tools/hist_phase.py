@@ -1,5 +1,6 @@
-"""In-kernel s_memtime phase split of hist_scan_kernel on the bench workload.
-Run with FDB_HIST_TIME=1 on a GPU box; prints cycles/element per phase."""
+"""ROUND-1 tool (stale): FDB_HIST_TIME phase split of the round-1 hist ring
+kernel (kept behind FDB_HIST_V1=1). hist2 uses FDB_HIST_ABLATE — see
+tools/perf_sweep.py."""
 import os, sys
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
