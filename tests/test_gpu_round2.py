@@ -226,3 +226,43 @@ def test_quantile_out_of_range_params(fdb, oracle, engine):
         q.param = p
         got, want = run_both(fdb, oracle, engine, st, q)
         check(got, want)
+
+
+@pytest.mark.parametrize("seed", [101, 202, 303])
+def test_randomized_matrix_vs_oracle(fdb, oracle, engine, seed):
+    """Randomized sweep: random chunkings (1..6 chunks), NaN/reset densities,
+    query geometry and function — fast path, stream walk and sample kernel
+    all exercised against the oracle in one go."""
+    rng = np.random.default_rng(seed)
+    funcs = [0, 1, 2, 3, 4, 5, 6, 7, 8, 9, 10, 12, 13, 14, 15, 16, 17, 18, 19]
+    for trial in range(6):
+        fid = int(rng.choice(funcs))
+        n = int(rng.integers(5, 500))
+        nchunks = int(rng.integers(1, 7))
+        cuts = sorted(rng.choice(np.arange(1, max(2, n)), size=nchunks - 1,
+                                 replace=False).tolist()) if nchunks > 1 else []
+        series = []
+        for _ in range(int(rng.integers(2, 10))):
+            if fid <= 2 or fid == 19:
+                ts, vs = synth_counter_series(rng, n, reset_p=0.02)
+                if rng.random() < 0.3:
+                    vs[rng.random(n) < 0.05] = np.nan
+            else:
+                ts, vs = synth_gauge_series(rng, n, nan_p=0.15)
+            bounds = [0] + cuts + [n]
+            chunks = []
+            for a, b in zip(bounds[:-1], bounds[1:]):
+                if a < b:
+                    chunks.append([(int(t), float(v))
+                                   for t, v in zip(ts[a:b], vs[a:b])])
+            series.append(chunks)
+        st = build_store(fdb, series,
+                         kind=fdb.COL_COUNTER if fid <= 2 else None)
+        start = 100000 + int(rng.integers(0, 50)) * 15000
+        step = int(rng.choice([5000, 15000, 60000]))
+        window = int(rng.choice([60000, 300000, 1800000]))
+        nw = int(rng.integers(2, 120))
+        q = fdb.make_query(start, step, start + (nw - 1) * step, window, fid)
+        q.param = 0.8 if fid in (16, 17) else 600.0
+        got, want = run_both(fdb, oracle, engine, st, q)
+        check(got, want)
