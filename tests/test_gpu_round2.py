@@ -238,8 +238,8 @@ def test_randomized_matrix_vs_oracle(fdb, oracle, engine, seed):
     for trial in range(6):
         fid = int(rng.choice(funcs))
         n = int(rng.integers(5, 500))
-        nchunks = int(rng.integers(1, 7))
-        cuts = sorted(rng.choice(np.arange(1, max(2, n)), size=nchunks - 1,
+        nchunks = min(int(rng.integers(1, 7)), n - 1)
+        cuts = sorted(rng.choice(np.arange(1, n), size=max(0, nchunks - 1),
                                  replace=False).tolist()) if nchunks > 1 else []
         series = []
         for _ in range(int(rng.integers(2, 10))):
