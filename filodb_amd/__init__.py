@@ -143,6 +143,36 @@ def lib():
             ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
             ctypes.c_int32, _c_double_p, _c_double_p, _c_double_p,
             _c_double_p, _c_double_p, ctypes.c_int32]
+        L.fdb_brv2_builder_create.restype = ctypes.c_void_p
+        L.fdb_brv2_builder_create.argtypes = [ctypes.c_int64]
+        L.fdb_brv2_builder_destroy.argtypes = [ctypes.c_void_p]
+        L.fdb_brv2_add_record.argtypes = [
+            ctypes.c_void_p, ctypes.c_int64, ctypes.c_double, ctypes.c_char_p,
+            ctypes.POINTER(ctypes.c_char_p), ctypes.c_int32, ctypes.c_int32]
+        L.fdb_brv2_finish.argtypes = [ctypes.c_void_p,
+                                      ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32]
+        L.fdb_brv2_read.argtypes = [
+            ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32, ctypes.c_int32,
+            ctypes.POINTER(ctypes.c_int64), _c_double_p,
+            ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32,
+            ctypes.POINTER(ctypes.c_int32), ctypes.POINTER(ctypes.c_int32),
+            ctypes.POINTER(ctypes.c_int32)]
+        L.fdb_brv2_index_create.restype = ctypes.c_void_p
+        L.fdb_brv2_index_destroy.argtypes = [ctypes.c_void_p]
+        L.fdb_store_ingest_brv2.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint8),
+            ctypes.c_int32, ctypes.c_int32, ctypes.POINTER(ctypes.c_int32)]
+        L.fdb_chunkid.restype = ctypes.c_int64
+        L.fdb_chunkid.argtypes = [ctypes.c_int64, ctypes.c_int64]
+        L.fdb_chunkid_start_time.restype = ctypes.c_int64
+        L.fdb_chunkid_start_time.argtypes = [ctypes.c_int64]
+        L.fdb_store_persist.argtypes = [
+            ctypes.c_void_p, ctypes.c_int32, ctypes.POINTER(ctypes.c_uint8),
+            ctypes.c_int32, ctypes.c_int64, ctypes.POINTER(ctypes.c_uint8),
+            ctypes.c_int32, ctypes.POINTER(ctypes.c_int32)]
+        L.fdb_store_restore.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint8),
+            ctypes.c_int32, ctypes.c_int32, ctypes.POINTER(ctypes.c_int32)]
         L.fdb_gpu_unpack_doubles_xor.argtypes = [
             ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64,
             ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32),
@@ -447,3 +477,111 @@ def nibblepack_unpack8(data):
     _check(lib().fdb_nibblepack_unpack8(buf, len(data), out, ctypes.byref(consumed)),
            "unpack8")
     return list(out), consumed.value
+
+
+class BRv2Builder:
+    """BinaryRecord v2 ingestion-container builder (layout per
+    RecordContainer.scala / RecordSchema.scala / RecordBuilder.scala)."""
+
+    def __init__(self, ts_header=0):
+        self._h = lib().fdb_brv2_builder_create(ts_header)
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            try:
+                lib().fdb_brv2_builder_destroy(self._h)
+            except (TypeError, AttributeError):
+                pass
+            self._h = None
+
+    def add(self, ts, value, metric, tags=None, schema_id=1):
+        tags = tags or {}
+        arr = (ctypes.c_char_p * (2 * len(tags)))()
+        for i, (k, v) in enumerate(tags.items()):
+            arr[2 * i] = k.encode()
+            arr[2 * i + 1] = v.encode()
+        _check(lib().fdb_brv2_add_record(self._h, ts, value, metric.encode(),
+                                         arr, len(tags), schema_id), "brv2_add")
+
+    def finish(self):
+        cap = 1 << 22
+        out = (ctypes.c_uint8 * cap)()
+        n = _check(lib().fdb_brv2_finish(self._h, out, cap), "brv2_finish")
+        return bytes(out[:n])
+
+
+def brv2_read(container, idx):
+    """Returns (ts, value, partkey_bytes, schema_id, part_hash) of record idx."""
+    buf = (ctypes.c_uint8 * len(container)).from_buffer_copy(container)
+    ts = ctypes.c_int64()
+    val = ctypes.c_double()
+    pk = (ctypes.c_uint8 * 4096)()
+    pk_len = ctypes.c_int32()
+    sid = ctypes.c_int32()
+    ph = ctypes.c_int32()
+    _check(lib().fdb_brv2_read(buf, len(container), idx, ctypes.byref(ts),
+                               ctypes.byref(val), pk, 4096,
+                               ctypes.byref(pk_len), ctypes.byref(sid),
+                               ctypes.byref(ph)), "brv2_read")
+    return ts.value, val.value, bytes(pk[:pk_len.value]), sid.value, ph.value
+
+
+class BRv2Index:
+    """Partition-key → series index for container ingestion."""
+
+    def __init__(self):
+        self._h = lib().fdb_brv2_index_create()
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            try:
+                lib().fdb_brv2_index_destroy(self._h)
+            except (TypeError, AttributeError):
+                pass
+            self._h = None
+
+
+def ingest_brv2(store, index, container, kind=COL_GAUGE):
+    """Consumes a BinaryRecord v2 container into the chunk store. Returns
+    (num_records, num_new_series)."""
+    buf = (ctypes.c_uint8 * len(container)).from_buffer_copy(container)
+    new_series = ctypes.c_int32()
+    n = _check(lib().fdb_store_ingest_brv2(store._h, index._h, buf,
+                                           len(container), kind,
+                                           ctypes.byref(new_series)),
+               "ingest_brv2")
+    return n, new_series.value
+
+
+def chunkid(start_time, ingestion_time):
+    """Chunk-table row key packing (core/.../store/package.scala:112-123)."""
+    return lib().fdb_chunkid(start_time, ingestion_time)
+
+
+def chunkid_start_time(cid):
+    return lib().fdb_chunkid_start_time(cid)
+
+
+def persist_series(store, sid, partkey, ingestion_time=0):
+    """Serializes every chunk of a series as Cassandra chunk-table rows
+    (TimeSeriesChunksTable.scala row shape in a flat framing). Returns the
+    bytes."""
+    pk = (ctypes.c_uint8 * max(1, len(partkey))).from_buffer_copy(
+        partkey or b"\x00")
+    cap = 1 << 24
+    out = (ctypes.c_uint8 * cap)()
+    out_len = ctypes.c_int32()
+    _check(lib().fdb_store_persist(store._h, sid, pk, len(partkey),
+                                   ingestion_time, out, cap,
+                                   ctypes.byref(out_len)), "persist")
+    return bytes(out[:out_len.value])
+
+
+def restore_rows(store, index, rows, kind=COL_GAUGE):
+    """Restores persisted chunk-table rows into an unsealed store (frozen
+    vector bytes reused unchanged). Returns the number of rows consumed."""
+    buf = (ctypes.c_uint8 * len(rows)).from_buffer_copy(rows)
+    nrows = ctypes.c_int32()
+    _check(lib().fdb_store_restore(store._h, index._h, buf, len(rows), kind,
+                                   ctypes.byref(nrows)), "restore")
+    return nrows.value

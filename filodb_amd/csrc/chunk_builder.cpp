@@ -702,6 +702,33 @@ extern "C" int32_t fdb_chunk_get(const fdb_store_t* s, int32_t sid, int32_t ci, 
   return FDB_OK;
 }
 
+// Paging-side restore (TimeSeriesChunksTable read path): appends a chunk whose
+// frozen vector bytes come from persisted storage UNCHANGED — no re-encode, the
+// same bytes the ODP reader hands to the query engine.
+extern "C" int32_t fdb_store_add_encoded_chunk(fdb_store_t* s, int32_t sid,
+                                               const uint8_t* ts_bytes, int32_t ts_len,
+                                               const uint8_t* val_bytes, int32_t val_len,
+                                               int32_t num_rows,
+                                               int64_t start_time, int64_t end_time) {
+  if (s->sealed) { fdb_set_error("store is sealed"); return FDB_ERR_BADARG; }
+  if (sid < 0 || sid >= (int32_t)s->series.size()) { fdb_set_error("bad series id %d", sid); return FDB_ERR_BADARG; }
+  if (ts_len < 8 || val_len < 8 || num_rows < 1) { fdb_set_error("bad encoded chunk"); return FDB_ERR_BADARG; }
+  Series& se = s->series[(size_t)sid];
+  if (!se.buf_ts.empty()) { fdb_set_error("series %d has unsealed buffered rows", sid); return FDB_ERR_BADARG; }
+  if (!se.chunks.empty() && start_time < se.chunks.back().end_time) {
+    fdb_set_error("restored chunks must arrive time-ordered (series %d)", sid);
+    return FDB_ERR_BADARG;
+  }
+  Chunk c;
+  c.num_rows = num_rows;
+  c.start_time = start_time;
+  c.end_time = end_time;
+  c.ts_bytes.assign(ts_bytes, ts_bytes + ts_len);
+  c.val_bytes.assign(val_bytes, val_bytes + val_len);
+  se.chunks.push_back(std::move(c));
+  return FDB_OK;
+}
+
 // ---------------------------------------------------------------------------
 // synthetic workload generator (TestTimeseriesProducer shapes; BASELINE configs)
 // ---------------------------------------------------------------------------

@@ -287,6 +287,60 @@ int32_t fdb_series_append_hist_mm(fdb_store_t* s, int32_t sid,
  *  HistMaxMinSumAggregator). func_id FDB_FN_HIST_RATE = counter-corrected
  * rate + min/max; FDB_FN_SUM_OVER_TIME = SumOverTime of the histograms
  * (SumAndMaxOverTimeFuncHD shape). out_max/out_min may be NULL. */
+/* ------------------------------------------------------------------ */
+/* BinaryRecord v2 ingestion containers (layouts restated
+ * from core/.../binaryrecord2/RecordContainer.scala:15-27,
+ * RecordSchema.scala:60-75, RecordBuilder.scala:109-125,372-404,461-478).
+ * The builder produces on-wire containers of the gauge/counter ingestion
+ * schema {timestamp, value, metric, tags}; fdb_store_ingest_brv2 consumes
+ * them into a chunk store, keying series by the binary partition-key
+ * region exactly as the reference compares part keys. */
+typedef struct fdb_brv2_builder fdb_brv2_builder_t;
+typedef struct fdb_brv2_index fdb_brv2_index_t;
+
+fdb_brv2_builder_t* fdb_brv2_builder_create(int64_t ts_header);
+void fdb_brv2_builder_destroy(fdb_brv2_builder_t* b);
+/* tag_kv = [k0, v0, k1, v1, ...] (ntags pairs) */
+int32_t fdb_brv2_add_record(fdb_brv2_builder_t* b, int64_t ts, double val,
+                            const char* metric, const char* const* tag_kv,
+                            int32_t ntags, int32_t schema_id);
+int32_t fdb_brv2_finish(fdb_brv2_builder_t* b, uint8_t* out, int32_t cap);
+int32_t fdb_brv2_read(const uint8_t* bytes, int32_t len, int32_t idx,
+                      int64_t* ts, double* val, uint8_t* pk_out,
+                      int32_t pk_cap, int32_t* pk_len, int32_t* schema_id,
+                      int32_t* part_hash);
+fdb_brv2_index_t* fdb_brv2_index_create(void);
+void fdb_brv2_index_destroy(fdb_brv2_index_t* ix);
+int32_t fdb_store_ingest_brv2(fdb_store_t* s, fdb_brv2_index_t* ix,
+                              const uint8_t* bytes, int32_t len,
+                              int32_t col_kind, int32_t* out_new_series);
+
+/* Cassandra chunk-table persistence (paging side).
+ * Row shape restated from cassandra/.../columnstore/TimeSeriesChunksTable.scala:35-103:
+ * (partition blob, chunkid bigint, info = first 28 B of the ChunkSetInfo record
+ * {chunkID i64, numRows i32, ingestionTime i64, endTime i64}
+ * (core/.../store/ChunkSetInfo.scala:133-154, toBytes :250-254), chunks =
+ * frozen per-column vector blobs in schema order). chunkid packing from
+ * core/.../store/package.scala:112-123 (startTimeShift = 22, :16). Rows are
+ * framed into a flat byte stream (no Cassandra here):
+ * u32 pk_len + pk | i64 chunkid | u32 info_len + info | u16 nchunks |
+ * (u32 len + bytes)*. fdb_store_restore feeds the frozen bytes back
+ * UNCHANGED via fdb_store_add_encoded_chunk (a bit-faithful round trip). */
+int64_t fdb_chunkid(int64_t start_time, int64_t ingestion_time);
+int64_t fdb_chunkid_start_time(int64_t chunkid);
+int32_t fdb_store_persist(const fdb_store_t* s, int32_t series_id,
+                          const uint8_t* partkey, int32_t pk_len,
+                          int64_t ingestion_time,
+                          uint8_t* out, int32_t cap, int32_t* out_len);
+int32_t fdb_store_restore(fdb_store_t* s, fdb_brv2_index_t* ix,
+                          const uint8_t* bytes, int32_t len,
+                          int32_t col_kind, int32_t* out_rows);
+int32_t fdb_store_add_encoded_chunk(fdb_store_t* s, int32_t series_id,
+                                    const uint8_t* ts_bytes, int32_t ts_len,
+                                    const uint8_t* val_bytes, int32_t val_len,
+                                    int32_t num_rows,
+                                    int64_t start_time, int64_t end_time);
+
 int32_t fdb_query_exec_hist_mm(fdb_engine_t* e, const fdb_dataset_t* d,
                                const fdb_query_t* q, int32_t nb,
                                double* out_bucket_sums, double* out_counts,

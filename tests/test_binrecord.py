@@ -1,0 +1,261 @@
+"""BinaryRecord v2 ingestion containers.
+
+Byte-level layout assertions follow the reference sources directly:
+  RecordContainer.scala:15-27 (16-B header: len, version<<24, timestamp)
+  RecordSchema.scala:60-75    (fixedStart 6, field offsets, hash at
+                               offsets.last, variableAreaStart +4)
+  RecordBuilder.scala:109-125 (schemaID u16 at +4), :182-190 (blob =
+                               u16 len + bytes), :372-404 (map pairs =
+                               1-B key len + key, u16 val len + val),
+                               :461-478 (4-byte record alignment)
+"""
+import struct
+
+import numpy as np
+import pytest
+
+
+def test_container_header_layout(fdb):
+    b = fdb.BRv2Builder(ts_header=987654321)
+    b.add(100000, 1.5, "m", {})
+    c = b.finish()
+    nbytes, verword = struct.unpack_from("<II", c, 0)
+    assert nbytes == len(c) - 4            # length word counts bytes after it
+    assert (verword >> 24) & 0xFF == 1     # RecordBuilder.Version
+    assert struct.unpack_from("<q", c, 8)[0] == 987654321
+    # first record at +16
+    rec_len = struct.unpack_from("<i", c, 16)[0]
+    assert 16 + 4 + rec_len <= len(c)
+
+
+def test_record_field_layout(fdb):
+    b = fdb.BRv2Builder()
+    b.add(123456789, 42.25, "heap_usage", {"job": "api"}, schema_id=7)
+    c = b.finish()
+    rec = 16
+    # +4 u16 schemaID; +6 i64 ts; +14 f64 value
+    assert struct.unpack_from("<H", c, rec + 4)[0] == 7
+    assert struct.unpack_from("<q", c, rec + 6)[0] == 123456789
+    assert struct.unpack_from("<d", c, rec + 14)[0] == 42.25
+    # +22/+26: u32 offsets (from record start) to the metric/tags blobs
+    moff = struct.unpack_from("<I", c, rec + 22)[0]
+    toff = struct.unpack_from("<I", c, rec + 26)[0]
+    assert moff == 34                      # variableAreaStart
+    mlen = struct.unpack_from("<H", c, rec + moff)[0]
+    assert c[rec + moff + 2:rec + moff + 2 + mlen] == b"heap_usage"
+    assert toff == moff + 2 + mlen
+    # map: u16 total len; pair = 1-B key len + key + u16 val len + val
+    map_len = struct.unpack_from("<H", c, rec + toff)[0]
+    p = rec + toff + 2
+    klen = c[p]
+    assert klen < 0xC0                     # not a predefined-key code
+    assert c[p + 1:p + 1 + klen] == b"job"
+    vlen = struct.unpack_from("<H", c, p + 1 + klen)[0]
+    assert c[p + 3 + klen:p + 3 + klen + vlen] == b"api"
+    assert map_len == 1 + klen + 2 + vlen
+    # record length word counts bytes after it; container is 4-byte aligned
+    rec_len = struct.unpack_from("<i", c, rec)[0]
+    assert rec + 4 + rec_len == rec + toff + 2 + map_len
+    assert len(c) % 4 == 0
+
+
+def test_records_align_and_iterate(fdb):
+    b = fdb.BRv2Builder()
+    for i in range(5):
+        b.add(100000 + i * 15000, float(i), "m" + "x" * i, {"k": str(i)})
+    c = b.finish()
+    seen = [fdb.brv2_read(c, i) for i in range(5)]
+    assert [s[0] for s in seen] == [100000 + i * 15000 for i in range(5)]
+    assert [s[1] for s in seen] == [float(i) for i in range(5)]
+    # identical part keys hash identically; different ones differ
+    b2 = fdb.BRv2Builder()
+    b2.add(1, 0.0, "m", {"k": "0"})
+    b2.add(2, 1.0, "m", {"k": "0"})
+    b2.add(3, 1.0, "m", {"k": "1"})
+    c2 = b2.finish()
+    h = [fdb.brv2_read(c2, i)[4] for i in range(3)]
+    assert h[0] == h[1] != h[2]
+
+
+def test_ingest_roundtrip_matches_direct_store(fdb, oracle):
+    """Containers → ingest → seal → oracle query equals a directly-built
+    store with the same samples (the ingest-side step feeding the path)."""
+    rng = np.random.default_rng(17)
+    n = 100
+    series_samples = {}
+    b = fdb.BRv2Builder()
+    for s in range(6):
+        ts = (100000 + np.arange(n) * 15000
+              + rng.integers(-250, 251, n)).astype(np.int64)
+        ts = np.maximum.accumulate(ts)
+        vs = np.cumsum(rng.normal(0, 1, n)) + 0.5
+        series_samples[s] = (ts, vs)
+    # interleave records across series (ingestion order is arrival order)
+    for i in range(n):
+        for s in range(6):
+            ts, vs = series_samples[s]
+            b.add(int(ts[i]), float(vs[i]), "metric", {"job": f"j{s}"})
+    container = b.finish()
+
+    st = fdb.ChunkStore()
+    ix = fdb.BRv2Index()
+    nrec, nnew = fdb.ingest_brv2(st, ix, container, kind=fdb.COL_GAUGE)
+    assert nrec == 6 * n and nnew == 6
+    st.seal()
+
+    st2 = fdb.ChunkStore()
+    for s in range(6):
+        sid = st2.add_series(0, fdb.COL_GAUGE)
+        ts, vs = series_samples[s]
+        st2.append(sid, ts, vs)
+    st2.seal()
+
+    q = fdb.make_query(100000 + 20 * 15000, 15000, 100000 + 90 * 15000,
+                       300000, fdb.FN_SUM_OVER_TIME)
+    nw = q.num_windows
+    got = oracle.query_exec(st.view(), q, st.num_series, nw)
+    want = oracle.query_exec(st2.view(), q, st2.num_series, nw)
+    np.testing.assert_array_equal(got, want)
+
+
+def test_ingest_across_containers_same_index(fdb):
+    """Series identity persists across containers through the part-key index."""
+    st = fdb.ChunkStore()
+    ix = fdb.BRv2Index()
+    for batch in range(3):
+        b = fdb.BRv2Builder()
+        for i in range(10):
+            b.add(100000 + (batch * 10 + i) * 15000, float(i), "m",
+                  {"job": "a"})
+            b.add(100000 + (batch * 10 + i) * 15000, float(i) * 2, "m",
+                  {"job": "b"})
+        n, new = fdb.ingest_brv2(st, ix, b.finish())
+        assert n == 20
+        assert new == (2 if batch == 0 else 0)
+    st.seal()
+    assert st.num_series == 2
+
+
+def test_bad_containers_rejected(fdb):
+    st = fdb.ChunkStore()
+    ix = fdb.BRv2Index()
+    with pytest.raises(RuntimeError):
+        fdb.ingest_brv2(st, ix, b"\x00" * 8)          # too short
+    bad = bytearray(fdb.BRv2Builder().finish())
+    bad[7] = 0x02                                      # wrong version byte
+    with pytest.raises(RuntimeError):
+        fdb.ingest_brv2(st, ix, bytes(bad))
+
+# ---------------------------------------------------------------------------
+# Cassandra chunk-table persistence (paging side).
+# chunkid packing: core/.../store/package.scala:112-123 (startTimeShift=22 :16)
+# info blob: ChunkSetInfo.scala:133-154 (chunkID@0, numRows@8,
+#            ingestionTime@12, endTime@20), toBytes :250-254
+# row shape: cassandra/.../columnstore/TimeSeriesChunksTable.scala:35-103
+# ---------------------------------------------------------------------------
+
+def _floor_mod(a, m):
+    return a - (a // m) * m
+
+
+def test_chunkid_packing(fdb):
+    mod = 48 * 24 * 3600
+    for st, it in [(0, 0), (1000000, 12345), (1694700000000, 999999999),
+                   (5, -7), (1 << 40, mod - 1)]:
+        cid = fdb.chunkid(st, it)
+        want = ((1 << 63) ^ (st << 22) | _floor_mod(it, mod))
+        want = want - (1 << 64) if want >= (1 << 63) else want   # as i64
+        assert cid == want, (st, it)
+        assert fdb.chunkid_start_time(cid) == st
+
+
+def test_persist_row_layout(fdb):
+    st = fdb.ChunkStore()
+    sid = st.add_series(0, fdb.COL_GAUGE)
+    ts = np.arange(100000, 100000 + 10 * 15000, 15000, dtype=np.int64)
+    st.append(sid, ts, np.arange(10, dtype=np.float64))
+    st.seal()
+    pk = b"\x0a\x00heap_usage"
+    rows = fdb.persist_series(st, sid, pk, ingestion_time=777)
+    # frame: u32 pk_len + pk | i64 chunkid | u32 info_len + info(28) |
+    #        u16 nchunks | (u32 len + bytes)*
+    pl = struct.unpack_from("<I", rows, 0)[0]
+    assert pl == len(pk) and rows[4:4 + pl] == pk
+    p = 4 + pl
+    cid = struct.unpack_from("<q", rows, p)[0]
+    assert cid == fdb.chunkid(100000, 777)
+    p += 8
+    il = struct.unpack_from("<I", rows, p)[0]
+    assert il == 28
+    p += 4
+    info_cid, nrows = struct.unpack_from("<qi", rows, p)
+    ing, endt = struct.unpack_from("<qq", rows, p + 12)
+    assert (info_cid, nrows, ing, endt) == (cid, 10, 777, int(ts[-1]))
+    p += 28
+    nc = struct.unpack_from("<H", rows, p)[0]
+    assert nc == 2
+    p += 2
+    tl = struct.unpack_from("<I", rows, p)[0]
+    # frozen vector bytes: leading u32 length word counts bytes after it
+    assert struct.unpack_from("<I", rows, p + 4)[0] == tl - 4
+    p += 4 + tl
+    vl = struct.unpack_from("<I", rows, p)[0]
+    assert struct.unpack_from("<I", rows, p + 4)[0] == vl - 4
+    assert p + 4 + vl == len(rows)
+
+
+def test_persist_restore_roundtrip_query(fdb, oracle):
+    """store → chunk-table rows → fresh store: queries agree bit-exactly
+    (frozen vector bytes travel UNCHANGED through the row blobs)."""
+    rng = np.random.default_rng(23)
+    st = fdb.ChunkStore()
+    st.set_max_rows(120)                    # force multiple chunks per series
+    n = 400
+    for s in range(5):
+        sid = st.add_series(0, fdb.COL_COUNTER)
+        ts = (100000 + np.arange(n) * 15000
+              + rng.integers(-250, 251, n)).astype(np.int64)
+        ts = np.maximum.accumulate(ts)
+        vs = np.cumsum(rng.exponential(2.0, n))
+        st.append(sid, ts, vs)
+    st.seal()
+
+    st2 = fdb.ChunkStore()
+    st2.set_max_rows(120)
+    ix = fdb.BRv2Index()
+    total = 0
+    for s in range(5):
+        rows = fdb.persist_series(st, s, b"pk-%d" % s, ingestion_time=1234)
+        total += fdb.restore_rows(st2, ix, rows, kind=fdb.COL_COUNTER)
+    assert total == sum(4 for _ in range(5))  # 400 rows / 120 → 4 chunks each
+    st2.seal()
+
+    q = fdb.make_query(100000 + 30 * 15000, 15000, 100000 + 390 * 15000,
+                       300000, fdb.FN_RATE)
+    got = oracle.query_exec(st2.view(), q, st2.num_series, q.num_windows)
+    want = oracle.query_exec(st.view(), q, st.num_series, q.num_windows)
+    np.testing.assert_array_equal(got, want)
+
+
+def test_restore_series_identity_and_order(fdb):
+    """Same partkey across restore calls lands in the same series; stream
+    interleaving is fine; out-of-order chunk times are rejected."""
+    st = fdb.ChunkStore()
+    a = st.add_series(0, fdb.COL_GAUGE)
+    b = st.add_series(0, fdb.COL_GAUGE)
+    ts = np.arange(100000, 100000 + 50 * 15000, 15000, dtype=np.int64)
+    st.append(a, ts, np.ones(50))
+    st.append(b, ts, np.full(50, 2.0))
+    st.seal()
+    ra = fdb.persist_series(st, a, b"A")
+    rb = fdb.persist_series(st, b, b"B")
+    st2 = fdb.ChunkStore()
+    ix = fdb.BRv2Index()
+    fdb.restore_rows(st2, ix, ra)
+    fdb.restore_rows(st2, ix, rb)
+    fdb.restore_rows(st2, ix, b"")          # empty stream is a no-op
+    assert st2.num_series == 2
+    with pytest.raises(RuntimeError):       # same partkey, earlier chunk again
+        fdb.restore_rows(st2, ix, ra)
+    with pytest.raises(RuntimeError):       # truncated stream
+        fdb.restore_rows(st2, ix, ra[:10])
