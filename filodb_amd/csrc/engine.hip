@@ -797,6 +797,10 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   }
   int nw = fdb_num_windows(q);
   if (nw <= 0) { fdb_set_error("bad window params"); return FDB_ERR_BADARG; }
+  if (q->window > ((int64_t)1 << 33)) {   // d_div1000 exactness domain
+    fdb_set_error("window length > 2^33 ms (~99 days) unsupported");
+    return FDB_ERR_BADARG;
+  }
   const bool is_topk = q->agg_id == AGG_TOPK || q->agg_id == AGG_BOTTOMK;
   const int kk = is_topk ? (int)q->param : 0;
   if (is_topk && (kk < 1 || kk > 16)) {
@@ -993,6 +997,10 @@ static int32_t run_hist(fdb_engine_t* e, const fdb_dataset_t* d,
   HIP_CHECK(hipSetDevice(e->device));
   int nw = fdb_num_windows(q);
   if (nw <= 0 || q->num_groups <= 0) { fdb_set_error("bad hist query params"); return FDB_ERR_BADARG; }
+  if (q->window > ((int64_t)1 << 33)) {   // d_div1000 exactness domain
+    fdb_set_error("window length > 2^33 ms (~99 days) unsupported");
+    return FDB_ERR_BADARG;
+  }
   if (nb < 1 || nb > 64) { fdb_set_error("num_buckets must be 1..64"); return FDB_ERR_BADARG; }
   if (!d->has_hist) {
     fdb_set_error("not a histogram dataset");

@@ -632,7 +632,7 @@ int32_t fdb_launch_fast_scan(hipStream_t stream, const uint8_t* blob, DirSoA dir
   if (const char* g = getenv("FDB_GRID")) cap = atoi(g);   // perf experiments
   if (cap > 0 && grid > cap) grid = cap;
   const char* rw = getenv("FDB_RATE_WAVES");   // occupancy experiment knob
-  const int rate_w = (rw && atoi(rw) == 5) ? 5 : 6;   // 6-wave prefetch-free measured best (2.48 vs 2.62 ms)
+  const int rate_w = rw ? atoi(rw) : 6;   // 6-wave prefetch-free measured best (2.48 vs 2.62 ms)
   #define FARGS blob, dir, series_first, series_nchunks, group_ids, \
       series_by_group, num_series, qstart, qstep, qwindow, num_windows, \
       agg_id, out, out_cnt, out_sq, phase_mask
