@@ -358,6 +358,29 @@ __device__ __forceinline__ int fdiv_floor_win(int64_t a, int64_t b, double inv_b
   return (int)w;
 }
 
+// fdiv_floor_win widened: clamp low bound is a parameter and the exact
+// "residual nonzero" bit comes back, so ONE division yields both inversion
+// boundaries when b | window: f = floor(a/b), ci = f + (a mod b != 0) [ceil],
+// di = f + window/b. Outside [lo_clamp, hi+1] the value is only ever compared
+// against in-range window indexes (callers take min/max), so the clamped
+// return and a pessimistic nz are safe; the 1.0 gap between adjacent
+// quotients dwarfs the ~1e-7 absolute error of g, so the clamp tests cannot
+// fire for a true quotient inside the range.
+__device__ __forceinline__ int fdiv_floor_rem(int64_t a, int64_t b, double inv_b,
+                                              int hi, int lo_clamp, bool* nz) {
+  double g = (double)a * inv_b;
+  if (g < (double)lo_clamp) { *nz = true; return lo_clamp; }
+  if (g > (double)hi + 1.0) { *nz = true; return hi + 1; }
+  int64_t w = (int64_t)__builtin_floor(g);
+  double r = __builtin_fma(-(double)w, (double)b, (double)a);
+  const bool up = (r >= (double)b), dn = (r < 0.0);
+  w += up;
+  w -= dn;
+  double rr = up ? r - (double)b : (dn ? r + (double)b : r);
+  *nz = (rr != 0.0);
+  return (int)w;
+}
+
 // wave-wide inclusive prefix sum over i64 (bucket-cumulative reconstruction)
 __device__ __forceinline__ int64_t wave_incl_scan_i64(int64_t x, int lane) {
   for (int off = 1; off < 64; off <<= 1) {

@@ -379,15 +379,31 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       // (a shuffle-free variant recomputing the neighbor boundaries from
       // ts[i-1] — 4 fdivs/row, no carries — measured SLOWER: the inversion
       // phase grew 8.2 → 11.1 Gcyc; the shfl form stays)
+      // when qstep | qwindow (the usual Prometheus shape) one division gives
+      // both boundaries: f = floor((o+Ae)/qstep) with its exact mod-zero bit
+      // ⇒ ci = f + (mod != 0) [= ceil], di = f + qwindow/qstep. The wider low
+      // clamp keeps di's derivation exact down to di = -1 (see fdiv_floor_rem).
+      const bool kdiv = (qwindow % qstep) == 0;
+      const int Kwin = kdiv ? (int)(qwindow / qstep) : 0;
       int c_carry = 0, d_carry = -1;
       for (int base = 0; base < n; base += 64) {
         const int i = base + lane;
         const bool live = i < n;
         int64_t o = live ? (int64_t)ws.tso[i] : 0;
-        int ci = live ? fdiv_floor_win(o + Ae + qstep - 1, qstep, inv_step,
-                                       num_windows) : 0;
-        int di = live ? fdiv_floor_win(o + Ae + qwindow, qstep, inv_step,
-                                       num_windows) : 0;
+        int ci, di;
+        if (kdiv) {             // wave-uniform branch
+          bool nz = false;
+          int f = live ? fdiv_floor_rem(o + Ae, qstep, inv_step, num_windows,
+                                        -1 - Kwin, &nz) : 0;
+          ci = f + (nz ? 1 : 0);
+          di = f + Kwin;
+          if (!live) { ci = 0; di = 0; }
+        } else {
+          ci = live ? fdiv_floor_win(o + Ae + qstep - 1, qstep, inv_step,
+                                     num_windows) : 0;
+          di = live ? fdiv_floor_win(o + Ae + qwindow, qstep, inv_step,
+                                     num_windows) : 0;
+        }
         int cprev = __shfl_up(ci, 1);
         int dprev = __shfl_up(di, 1);
         if (lane == 0) { cprev = c_carry; dprev = d_carry; }

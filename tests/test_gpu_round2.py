@@ -266,3 +266,23 @@ def test_randomized_matrix_vs_oracle(fdb, oracle, engine, seed):
         q.param = 0.8 if fid in (16, 17) else 600.0
         got, want = run_both(fdb, oracle, engine, st, q)
         check(got, want)
+
+
+@pytest.mark.parametrize("step,window", [(15000, 70000), (10000, 45000),
+                                         (7000, 300000), (15000, 300000),
+                                         (15000, 15000), (20000, 10000)])
+def test_window_step_ratio_matrix(fdb, oracle, engine, step, window):
+    """Inversion-scan boundary parity across divisible and non-divisible
+    window/step pairs (the divisible shape derives both boundaries from one
+    floor-division; the others keep the two-division path)."""
+    st = counter_store(fdb, n_series=32, n=240, seed=step + window,
+                       reset_p=0.02, nan_p=0.05)
+    q = fdb.make_query(Q["start"] - 2 * step, step,
+                       Q["start"] + 230 * 15000, window, fdb.FN_RATE)
+    got, want = run_both(fdb, oracle, engine, st, q)
+    check(got, want)
+    q2 = fdb.make_query(Q["start"], step, Q["start"] + 230 * 15000, window,
+                        fdb.FN_SUM_OVER_TIME)
+    st2 = gauge_store(fdb, n_series=16, seed=window, nan_p=0.2)
+    got, want = run_both(fdb, oracle, engine, st2, q2)
+    check(got, want)
