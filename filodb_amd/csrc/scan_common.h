@@ -245,9 +245,13 @@ __device__ inline void d_decode_chunk(const DVec& v, int n, int64_t* tout,
 // Caller guarantees the chunk's time span fits i32 (upload-time check).
 __device__ inline void d_decode_ts_offsets(const DVec& v, int n, int64_t ts0,
                                            int32_t* tout, int lane) {
+  // all arithmetic in i32: the upload guard pins the chunk's span (and so
+  // base + slope*i + inner, = ts_i - ts0) inside i32, and i32 wrap-around
+  // equals the i64-then-truncate result whenever the true value fits
   if (v.wf == FDB_WF_DDV_CONST) {
+    const int32_t cbase = (int32_t)(v.init - ts0);
     for (int i = lane; i < n; i += 64)
-      tout[i] = (int32_t)((int64_t)v.slope * i + (v.init - ts0));
+      tout[i] = (int32_t)((uint32_t)v.slope * (uint32_t)i + (uint32_t)cbase);
     return;
   }
   if (v.wf == FDB_WF_PRIM64) {
@@ -255,7 +259,7 @@ __device__ inline void d_decode_ts_offsets(const DVec& v, int n, int64_t ts0,
       tout[i] = (int32_t)(d_i64(v.idata + 8 * (size_t)i) - ts0);
     return;
   }
-  const int64_t base = v.init - ts0;
+  const int32_t base = (int32_t)(v.init - ts0);
   if (v.nbits == 16) {
     for (int i0 = 4 * lane; i0 < n; i0 += 256) {
       uint32_t lo = d_u32(v.idata + 2 * (size_t)i0);
@@ -266,7 +270,8 @@ __device__ inline void d_decode_ts_offsets(const DVec& v, int n, int64_t ts0,
         if (i0 + k < n) {
           int32_t d = (int32_t)(int16_t)(uint16_t)(w >> (16 * k));
           if (!v.sign) d &= 0xffff;
-          tout[i0 + k] = (int32_t)(base + (int64_t)v.slope * (i0 + k) + d);
+          tout[i0 + k] = (int32_t)((uint32_t)base
+              + (uint32_t)v.slope * (uint32_t)(i0 + k) + (uint32_t)d);
         }
       }
     }
@@ -277,13 +282,15 @@ __device__ inline void d_decode_ts_offsets(const DVec& v, int n, int64_t ts0,
       #pragma unroll
       for (int k = 0; k < 2; k++)
         if (i0 + k < n)
-          tout[i0 + k] = (int32_t)(base + (int64_t)v.slope * (i0 + k)
-                                   + d_i32(v.idata + 4 * (size_t)(i0 + k)));
+          tout[i0 + k] = (int32_t)((uint32_t)base
+              + (uint32_t)v.slope * (uint32_t)(i0 + k)
+              + (uint32_t)d_i32(v.idata + 4 * (size_t)(i0 + k)));
     }
     return;
   }
   for (int i = lane; i < n; i += 64)
-    tout[i] = (int32_t)(base + (int64_t)v.slope * i + d_inner_at(&v, i));
+    tout[i] = (int32_t)((uint32_t)base + (uint32_t)v.slope * (uint32_t)i
+                        + (uint32_t)d_inner_at(&v, i));
 }
 
 // Correctly-rounded x/1000 in 3 ops (mul + 2 fma, Markstein fixup) instead of
@@ -324,6 +331,33 @@ __device__ inline double d_extrapolated_rate(int64_t windowStart, int64_t window
   ext += (durationToEnd < thresh) ? durationToEnd : avgDur / 2;
   double scaledDelta = delta * (ext / sampledInterval);
   return isRate ? (scaledDelta / (double)(windowEnd - windowStart) * 1000.0) : scaledDelta;
+}
+
+// i32-difference variant of d_extrapolated_rate for the fast path: all three
+// millisecond gaps fit i32 when the window does (caller checks), so each
+// i64 subtract + 4-instruction i64→f64 convert becomes a truncate + single
+// v_cvt_f64_i32; the window duration is wave-uniform and passed pre-converted.
+// Every operation after the converts matches d_extrapolated_rate exactly
+// (same values, same rounding) — results are bit-identical.
+__device__ inline double d_extrapolated_rate_i32(int32_t toStart_ms, int32_t toEnd_ms,
+                                                 int numSamples, int32_t sampled_ms,
+                                                 double dur_ms, double v1, double v2,
+                                                 bool isCounter, bool isRate) {
+  double durationToStart = d_div1000((double)toStart_ms);
+  double durationToEnd = d_div1000((double)toEnd_ms);
+  double sampledInterval = d_div1000((double)sampled_ms);
+  double avgDur = sampledInterval / ((double)numSamples - 1);
+  double delta = v2 - v1;
+  if (isCounter && delta > 0 && v1 >= 0) {
+    double durationToZero = sampledInterval * (v1 / delta);
+    if (durationToZero < durationToStart) durationToStart = durationToZero;
+  }
+  double thresh = avgDur * 1.1;
+  double ext = sampledInterval;
+  ext += (durationToStart < thresh) ? durationToStart : avgDur / 2;
+  ext += (durationToEnd < thresh) ? durationToEnd : avgDur / 2;
+  double scaledDelta = delta * (ext / sampledInterval);
+  return isRate ? (scaledDelta / dur_ms * 1000.0) : scaledDelta;
 }
 
 // NaN-aware f64 atomic min/max via CAS (group aggregation; RowAggregator semantics)

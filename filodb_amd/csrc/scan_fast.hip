@@ -144,6 +144,8 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 
   const double inv_step = 1.0 / (double)qstep;
   const double rate_scale = 1000.0 / (double)qwindow;
+  const bool qw32 = qwindow < ((int64_t)1 << 31);   // i32 epilogue eligibility
+  const double dur_ms = (double)qwindow;
   (void)pm;
   constexpr bool timing = TIMED;        // perf ablation (clobbers out[])
   uint64_t tD = 0, tM = 0, tI = 0, tW = 0, tt = 0;
@@ -489,9 +491,15 @@ void fast_scan_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                 v2 += f_corr_at(ws, dcount, dense, n, e);
               }
               const int64_t wEndOff = (int64_t)w * qstep - Ae;   // wEnd - ts0
-              res = d_extrapolated_rate(wEndOff - qwindow, wEndOff, e - s + 1,
-                                        t1, v1, t2, v2,
-                                        IS_COUNTER, FUNC == FN_RATE);
+              if (qw32)      // uniform: window fits i32 ⇒ so do all three gaps
+                res = d_extrapolated_rate_i32(
+                    (int32_t)(t1 - (wEndOff - qwindow)), (int32_t)(wEndOff - t2),
+                    e - s + 1, t2 - t1, dur_ms, v1, v2,
+                    IS_COUNTER, FUNC == FN_RATE);
+              else
+                res = d_extrapolated_rate(wEndOff - qwindow, wEndOff, e - s + 1,
+                                          t1, v1, t2, v2,
+                                          IS_COUNTER, FUNC == FN_RATE);
             }
           }
           emit_res(k, w, wok, res);
