@@ -42,22 +42,26 @@ void fdb_set_error(const char* fmt, ...);   // chunk_builder.cpp
 // active (estream shuffles). Returns the lane's (<<trailing-zeroes) delta.
 // The _buf form takes a pre-staged 256-B register window so two elements'
 // stage loads can be issued together and their parse chains interleaved.
-__device__ __forceinline__ int64_t h2_parse_buf(const uint8_t* ep, int elen,
+template <bool EST>
+__device__ __forceinline__ int64_t h2_parse_est(const uint8_t* ep, int elen,
                                                 int nb, int b, bool live,
                                                 int lane, uint32_t ebuf,
                                                 int eshift) {
-  const bool est = elen + 14 + eshift <= 256;
-  // 8-value group headers, walked serially (wave-uniform)
+  // 8-value group headers, walked serially (wave-uniform). EST is a
+  // compile-time constant: with it true the walk is pure readlane + scalar
+  // arithmetic (SGPR chain, no exec-mask churn); the runtime est ? : form
+  // made every access an if-converted dual path whose global-load arm
+  // poisoned the uniformity analysis and kept the whole walk on the VALU.
   const int my_group = b >> 3;
   int off = 0, gOff = 0, gBits = 0, gTrail = 0;
   uint32_t gMask = 0;
   for (int g = 0; g * 8 < nb; g++) {
-    uint32_t mask = estream_byte_uni(est, ebuf, eshift, ep, 2 + off);
+    uint32_t mask = estream_byte_uni(EST, ebuf, eshift, ep, 2 + off);
     int numBits = 0, trail = 0, glen;
     if (mask == 0) {
       glen = 1;
     } else {
-      int widths = (int)estream_byte_uni(est, ebuf, eshift, ep, 2 + off + 1);
+      int widths = (int)estream_byte_uni(EST, ebuf, eshift, ep, 2 + off + 1);
       numBits = ((widths >> 4) + 1) * 4;
       trail = (widths & 0x0f) * 4;
       glen = 2 + (numBits * __popc(mask) + 7) / 8;
@@ -70,8 +74,8 @@ __device__ __forceinline__ int64_t h2_parse_buf(const uint8_t* ep, int elen,
   int slot = __popc(gMask & ((1u << bit) - 1));
   int bitpos = slot * gBits;
   int koff = 2 + gOff + 2 + (bitpos >> 3);
-  uint64_t w64 = estream_w64(est, ebuf, eshift, ep, koff);
-  uint32_t b8 = estream_byte(est, ebuf, eshift, ep, koff + 8);
+  uint64_t w64 = estream_w64(EST, ebuf, eshift, ep, koff);
+  uint32_t b8 = estream_byte(EST, ebuf, eshift, ep, koff + 8);
   int64_t delta = 0;
   if (live && in_mask) {
     int sh = bitpos & 7;
@@ -81,6 +85,17 @@ __device__ __forceinline__ int64_t h2_parse_buf(const uint8_t* ep, int elen,
     delta = (int64_t)((v & m) << gTrail);
   }
   return delta;
+}
+
+// uniform-branch dispatcher: big elements (stage window overrun) take the
+// unstaged global-load path; everything else stays on the scalarized walk
+__device__ __forceinline__ int64_t h2_parse_buf(const uint8_t* ep, int elen,
+                                                int nb, int b, bool live,
+                                                int lane, uint32_t ebuf,
+                                                int eshift) {
+  if (elen + 14 + eshift <= 256)
+    return h2_parse_est<true>(ep, elen, nb, b, live, lane, ebuf, eshift);
+  return h2_parse_est<false>(ep, elen, nb, b, live, lane, ebuf, eshift);
 }
 
 __device__ __forceinline__ int64_t h2_parse(const uint8_t* ep, int elen,
@@ -240,7 +255,12 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
       cu.decoded = false;
       if (HFUNC == 1) decode_cur(cu);         // sum mode: full prefix
       else if (new_sect) decode_cur(cu);      // section base always decoded
-      else if (cu.sect_left == 0) decode_cur(cu);  // section last: TypeDrop feed
+      else if (cu.sect_left == 0 && cu.e_local + 1 < cu.nrows &&
+               staged_byte(cu, cu.sp + 3) == 1)
+        decode_cur(cu);   // section last feeds the NEXT section's TypeDrop
+                          // correction — peek its type byte and decode only
+                          // then (drops are rare; the chunk seam decodes its
+                          // own last element unconditionally in step())
       if (cu.e_local == 0) {
         // first element of a chunk: boundary drop detection
         // (Histogram.compare top-bucket-down, Histogram.scala:204-214)
