@@ -259,3 +259,66 @@ def test_restore_series_identity_and_order(fdb):
         fdb.restore_rows(st2, ix, ra)
     with pytest.raises(RuntimeError):       # truncated stream
         fdb.restore_rows(st2, ix, ra[:10])
+
+
+@pytest.mark.gpu
+def test_ingest_to_gpu_query(fdb, oracle):
+    """Containers → ingest → seal → upload → GPU engine query equals the
+    oracle on the same store (the full ingest-to-scan path end to end)."""
+    rng = np.random.default_rng(31)
+    n = 240
+    b = fdb.BRv2Builder()
+    series_vals = []
+    for s in range(24):
+        ts = (100000 + np.arange(n) * 15000
+              + rng.integers(-250, 251, n)).astype(np.int64)
+        ts = np.maximum.accumulate(ts)
+        vs = np.cumsum(rng.exponential(2.0, n))
+        series_vals.append((ts, vs))
+        for t, v in zip(ts, vs):
+            b.add(int(t), float(v), "req_total", {"job": f"j{s}"})
+    st = fdb.ChunkStore()
+    ix = fdb.BRv2Index()
+    fdb.ingest_brv2(st, ix, b.finish(), kind=fdb.COL_COUNTER)
+    st.seal()
+    eng = fdb.Engine(0)
+    q = fdb.make_query(100000 + 20 * 15000, 15000, 100000 + 239 * 15000,
+                       300000, fdb.FN_RATE)
+    nw = q.num_windows
+    want = oracle.query_exec(st.view(), q, st.num_series, nw)
+    got = np.empty(st.num_series * nw, dtype=np.float64)
+    eng.query(eng.upload(st), q, out=got)
+    np.testing.assert_allclose(got, want, rtol=1e-9, atol=1e-12,
+                               equal_nan=True)
+
+
+@pytest.mark.gpu
+def test_restore_to_gpu_query(fdb, oracle):
+    """Persist → restore → upload → GPU query equals the original store's
+    GPU results exactly (frozen bytes unchanged through the row blobs)."""
+    rng = np.random.default_rng(37)
+    st = fdb.ChunkStore()
+    st.set_max_rows(120)
+    for s in range(16):
+        sid = st.add_series(0, fdb.COL_COUNTER)
+        ts = (100000 + np.arange(360) * 15000
+              + rng.integers(-250, 251, 360)).astype(np.int64)
+        ts = np.maximum.accumulate(ts)
+        st.append(sid, ts, np.cumsum(rng.exponential(1.5, 360)))
+    st.seal()
+    st2 = fdb.ChunkStore()
+    st2.set_max_rows(120)
+    ix = fdb.BRv2Index()
+    for s in range(16):
+        fdb.restore_rows(st2, ix, fdb.persist_series(st, s, b"pk%d" % s, 42),
+                         kind=fdb.COL_COUNTER)
+    st2.seal()
+    eng = fdb.Engine(0)
+    q = fdb.make_query(100000 + 30 * 15000, 15000, 100000 + 350 * 15000,
+                       300000, fdb.FN_RATE)
+    nw = q.num_windows
+    a = np.empty(16 * nw, dtype=np.float64)
+    bb = np.empty(16 * nw, dtype=np.float64)
+    eng.query(eng.upload(st), q, out=a)
+    eng.query(eng.upload(st2), q, out=bb)
+    np.testing.assert_array_equal(a, bb)
