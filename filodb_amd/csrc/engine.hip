@@ -699,11 +699,12 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
       sbg[(size_t)cur[(size_t)view.group_ids[s2]]++] = s2;
   }
 
-  // +256 B tail pad: the hist walk's wave-staged element reads may extend up
-  // to 256 B past the last element's start (estream staging)
+  // +512 B tail pad: the hist walk's wave-staged element reads may extend up
+  // to 256 B past the last element's start, and its linear stage PREFETCH one
+  // further 256-B window beyond that
   auto upload_pad = [&](void** dst, const void* src, size_t bytes) -> bool {
-    if (hipMalloc(dst, bytes + 256) != hipSuccess) return false;
-    if (hipMemset((char*)*dst + bytes, 0, 256) != hipSuccess) return false;
+    if (hipMalloc(dst, bytes + 512) != hipSuccess) return false;
+    if (hipMemset((char*)*dst + bytes, 0, 512) != hipSuccess) return false;
     return hipMemcpy(*dst, src, bytes, hipMemcpyHostToDevice) == hipSuccess;
   };
   bool ok = upload_pad((void**)&d->blob, view.blob, (size_t)view.blob_len)

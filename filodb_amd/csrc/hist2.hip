@@ -110,6 +110,7 @@ struct H2Cursor {
   const uint8_t* sp;     // NEXT section header
   const uint8_t* sb;     // 4-aligned base of the 256-B register stage (null = none)
   uint32_t sbuf;         // this lane's dword of the stage
+  uint32_t sbuf2;        // prefetched next 256-B window (load in flight)
   int elen;              // current element's payload length
   int sect_left;         // elements left in section AFTER the current one
   int c;                 // chunk index within the series
@@ -168,11 +169,25 @@ void hist2_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     // elements (the raw dependent d_u16 hop loads were the serial chain —
     // ~16k cycles per element of wall, sweep ablate data). Returns the byte
     // offset of ptr within the stage, restaging when [ptr, ptr+need) leaves it.
+    // dual-buffer linear streaming: sbuf2 always holds an ISSUED load of the
+    // NEXT 256-B window (the stream is consumed forward), so the common
+    // restage is a register swap whose vmcnt wait overlapped the previous
+    // ~2-3 elements of parse work instead of a cold ~600-cycle stall. A cold
+    // (re-anchoring) restage only happens at chunk opens and when an element
+    // straddles the window boundary. The +512 blob tail pad covers the
+    // prefetch past the last vector.
     auto ensure_stage = [&](H2Cursor& cu, const uint8_t* ptr, int need) -> int {
       int off = (int)(ptr - cu.sb);
       if (cu.sb == nullptr || off < 0 || off + need > 256) {
-        cu.sb = (const uint8_t*)((uintptr_t)ptr & ~(uintptr_t)3);
-        cu.sbuf = estream_stage(ptr, lane);
+        if (cu.sb != nullptr && off >= 256 && off + need <= 512) {
+          cu.sb += 256;                        // swap in the prefetched window
+          cu.sbuf = cu.sbuf2;
+        } else {
+          // cold anchor at ptr (chunk open, straddling element, or jump)
+          cu.sb = (const uint8_t*)((uintptr_t)ptr & ~(uintptr_t)3);
+          cu.sbuf = estream_stage(cu.sb, lane);
+        }
+        cu.sbuf2 = estream_stage(cu.sb + 256, lane);   // issue next prefetch
         off = (int)(ptr - cu.sb);
       }
       return off;
