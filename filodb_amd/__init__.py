@@ -173,6 +173,16 @@ def lib():
         L.fdb_store_restore.argtypes = [
             ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint8),
             ctypes.c_int32, ctypes.c_int32, ctypes.POINTER(ctypes.c_int32)]
+        L.fdb_store_add_encoded_chunk.argtypes = [
+            ctypes.c_void_p, ctypes.c_int32, ctypes.POINTER(ctypes.c_uint8),
+            ctypes.c_int32, ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32,
+            ctypes.c_int32, ctypes.c_int64, ctypes.c_int64]
+        L.fdb_gpu_encode_chunks.argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_int64), _c_double_p,
+            ctypes.POINTER(ctypes.c_int64), ctypes.c_int32, ctypes.c_int32,
+            ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64,
+            ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int64),
+            ctypes.POINTER(ctypes.c_int32), ctypes.POINTER(ctypes.c_int32)]
         L.fdb_gpu_unpack_doubles_xor.argtypes = [
             ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint8), ctypes.c_int64,
             ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32),
@@ -585,3 +595,51 @@ def restore_rows(store, index, rows, kind=COL_GAUGE):
     _check(lib().fdb_store_restore(store._h, index._h, buf, len(rows), kind,
                                    ctypes.byref(nrows)), "restore")
     return nrows.value
+
+
+def add_encoded_chunk(store, sid, ts_bytes, val_bytes, num_rows,
+                      start_time, end_time):
+    """Appends a pre-encoded (frozen-bytes) chunk to an unsealed store."""
+    tb = (ctypes.c_uint8 * len(ts_bytes)).from_buffer_copy(ts_bytes)
+    vb = (ctypes.c_uint8 * len(val_bytes)).from_buffer_copy(val_bytes)
+    _check(lib().fdb_store_add_encoded_chunk(store._h, sid, tb, len(ts_bytes),
+                                             vb, len(val_bytes), num_rows,
+                                             start_time, end_time),
+           "add_encoded_chunk")
+
+
+def gpu_encode_chunks(engine, chunks, kind=COL_GAUGE):
+    """GPU ingest encoder: chunks = list of (ts_i64_array, vals_f64_array).
+    Returns a list of (ts_bytes, val_bytes) — byte-identical to the host
+    encoder's frozen vectors."""
+    import numpy as _np
+    nch = len(chunks)
+    row_offs = _np.zeros(nch + 1, dtype=_np.int64)
+    for i, (t, v) in enumerate(chunks):
+        assert len(t) == len(v)
+        row_offs[i + 1] = row_offs[i] + len(t)
+    ts = _np.concatenate([_np.asarray(t, dtype=_np.int64) for t, _ in chunks])
+    vals = _np.concatenate([_np.asarray(v, dtype=_np.float64)
+                            for _, v in chunks])
+    cap = int(sum(2 * ((len(t) * 8 + 64 + 63) & ~63) for t, _ in chunks))
+    out = _np.zeros(cap, dtype=_np.uint8)
+    toff = _np.zeros(nch, dtype=_np.int64)
+    voff = _np.zeros(nch, dtype=_np.int64)
+    tlen = _np.zeros(nch, dtype=_np.int32)
+    vlen = _np.zeros(nch, dtype=_np.int32)
+    _check(lib().fdb_gpu_encode_chunks(
+        engine._h,
+        ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        vals.ctypes.data_as(_c_double_p),
+        row_offs.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        nch, kind,
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)), cap,
+        toff.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        voff.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        tlen.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+        vlen.ctypes.data_as(ctypes.POINTER(ctypes.c_int32))), "gpu_encode")
+    res = []
+    for c in range(nch):
+        res.append((out[toff[c]:toff[c] + tlen[c]].tobytes(),
+                    out[voff[c]:voff[c] + vlen[c]].tobytes()))
+    return res

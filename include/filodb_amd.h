@@ -341,6 +341,21 @@ int32_t fdb_store_add_encoded_chunk(fdb_store_t* s, int32_t series_id,
                                     int32_t num_rows,
                                     int64_t start_time, int64_t end_time);
 
+/* GPU ingest-side chunk encoder: encodes num_chunks scalar
+ * chunks (ts rows + double rows, delimited by row_offs[c]..row_offs[c+1])
+ * into frozen vectors BYTE-IDENTICAL to the host encoder / the reference's
+ * formats (DeltaDeltaVector.scala:63-135, IntBinaryVector.scala:52-177,
+ * DoubleVector.scala:86-96,457-476). One wavefront per chunk. Outputs land in
+ * `out` at the returned per-chunk offsets/lengths; histogram columns stay
+ * host-side. */
+int32_t fdb_gpu_encode_chunks(fdb_engine_t* e,
+                              const int64_t* ts, const double* vals,
+                              const int64_t* row_offs,
+                              int32_t num_chunks, int32_t col_kind,
+                              uint8_t* out, int64_t out_cap,
+                              int64_t* out_ts_off, int64_t* out_val_off,
+                              int32_t* out_ts_len, int32_t* out_val_len);
+
 int32_t fdb_query_exec_hist_mm(fdb_engine_t* e, const fdb_dataset_t* d,
                                const fdb_query_t* q, int32_t nb,
                                double* out_bucket_sums, double* out_counts,
