@@ -322,3 +322,36 @@ def test_restore_to_gpu_query(fdb, oracle):
     eng.query(eng.upload(st), q, out=a)
     eng.query(eng.upload(st2), q, out=bb)
     np.testing.assert_array_equal(a, bb)
+
+
+def test_persist_truncation_fuzz(fdb):
+    """Every strict prefix of a persisted stream is rejected (no partial-row
+    state leaks into the store)."""
+    st = fdb.ChunkStore()
+    sid = st.add_series(0, fdb.COL_GAUGE)
+    ts = np.arange(100000, 100000 + 30 * 15000, 15000, dtype=np.int64)
+    st.append(sid, ts, np.arange(30, dtype=np.float64))
+    st.seal()
+    rows = fdb.persist_series(st, sid, b"pk")
+    for cut in range(1, len(rows), 7):
+        st2 = fdb.ChunkStore()
+        ix = fdb.BRv2Index()
+        with pytest.raises(RuntimeError):
+            fdb.restore_rows(st2, ix, rows[:cut])
+        assert st2.num_series in (0, 1)      # no partial series beyond the key
+
+
+def test_brv2_reader_truncation_fuzz(fdb):
+    """Truncated containers never read out of bounds (error or clean count)."""
+    b = fdb.BRv2Builder()
+    for i in range(8):
+        b.add(100000 + i * 15000, float(i), "metric", {"job": "a", "dc": "x"})
+    c = b.finish()
+    for cut in range(0, len(c), 5):
+        st = fdb.ChunkStore()
+        ix = fdb.BRv2Index()
+        try:
+            n, _ = fdb.ingest_brv2(st, ix, c[:cut])
+            assert 0 <= n <= 8
+        except RuntimeError:
+            pass

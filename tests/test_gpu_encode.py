@@ -141,3 +141,33 @@ def test_encode_feeds_queryable_store(fdb, oracle, engine):
     engine.query(engine.upload(st), q, out=a)
     engine.query(engine.upload(st2), q, out=b)
     np.testing.assert_array_equal(a, b)
+
+
+def test_encode_fuzz_vs_host(fdb, engine):
+    """Randomized shapes: mixed magnitudes, NaN densities, monotone and
+    near-const streams — every chunk byte-equal to the host encoder."""
+    rng = np.random.default_rng(59)
+    chunks = []
+    for _ in range(60):
+        n = int(rng.integers(1, 400))
+        style = rng.integers(0, 5)
+        t0 = int(rng.integers(10**5, 10**14))
+        step = int(rng.choice([1000, 15000, 60000]))
+        jit = int(rng.choice([0, 100, 251, 5000, 10**6]))
+        ts = t0 + np.arange(n) * step
+        if jit:
+            ts = np.maximum.accumulate(ts + rng.integers(-jit, jit + 1, n))
+        ts = ts.astype(np.int64)
+        if style == 0:
+            vs = rng.normal(0, 10, n)
+        elif style == 1:
+            vs = np.cumsum(rng.integers(0, 100, n)).astype(np.float64)
+        elif style == 2:
+            vs = rng.integers(-5, 5, n).astype(np.float64)
+        elif style == 3:
+            vs = np.where(rng.random(n) < 0.3, np.nan, rng.normal(0, 1, n))
+        else:
+            vs = np.full(n, float(rng.integers(0, 10)))
+        chunks.append((ts, vs))
+    assert_match(fdb, engine, chunks, fdb.COL_GAUGE)
+    assert_match(fdb, engine, chunks, fdb.COL_COUNTER)
