@@ -156,3 +156,74 @@ def test_two_shard_stddev_merge_equals_global(fdb, oracle):
     expected = oracle.query_exec(st.view(), q, st.num_series, q.num_windows)
     np.testing.assert_allclose(merged, expected, rtol=1e-9, atol=1e-12,
                                equal_nan=True)
+
+
+def _worker_hist(rank, result_queue):
+    import sys
+    sys.path.insert(0, REPO)
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import filodb_amd as fdb
+    import pyclient as oracle
+    from test_hist import make_hist_store, synth_hist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29521"
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+
+    n_groups, nb = 3, 8
+    rng = np.random.default_rng(2000 + rank)
+    series = [synth_hist(rng, 60, nb=nb) for _ in range(12)]
+    groups = [(rank * 12 + s) % n_groups for s in range(12)]
+    st = make_hist_store(fdb, series, nb=nb, groups=groups)
+    start = 100000 + 25 * 15000
+    q = fdb.make_query(start, 30000, start + 300000, 300000,
+                       fdb.FN_HIST_RATE, fdb.AGG_SUM, n_groups)
+    q.param = 0.9
+    nw = q.num_windows
+    # per-shard [G×W×nb] bucket-rate sums + counts; ONE all-reduce each, then
+    # the quantile presentation runs on the MERGED sums (the reference's
+    # cross-shard ReduceAggregateExec → present split)
+    sums, cnts, _ = oracle.query_exec_hist(st.view(), q, nb)
+    t_s = torch.from_numpy(sums)
+    t_c = torch.from_numpy(cnts)
+    dist.all_reduce(t_s)
+    dist.all_reduce(t_c)
+    if rank == 0:
+        merged = t_s.numpy()
+        quant = np.full(n_groups * nw, np.nan)
+        for cell in range(n_groups * nw):
+            if t_c.numpy()[cell] > 0:
+                quant[cell] = oracle.hist_quantile(
+                    0.9, merged[cell * nb:(cell + 1) * nb], 2.0, 2.0)
+        result_queue.put(quant)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_two_shard_hist_quantile_merge_equals_global(fdb, oracle):
+    """Cross-shard histogram pipeline: bucket-rate sums all-reduce BEFORE the
+    quantile presentation; merged result equals the single-store global."""
+    from test_hist import make_hist_store, synth_hist
+    ctx = mp.get_context("spawn")
+    queue = ctx.Queue()
+    procs = [ctx.Process(target=_worker_hist, args=(r, queue)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    merged = queue.get(timeout=180)
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+
+    n_groups, nb = 3, 8
+    series, groups = [], []
+    for rank in range(WORLD):
+        rng = np.random.default_rng(2000 + rank)
+        series += [synth_hist(rng, 60, nb=nb) for _ in range(12)]
+        groups += [(rank * 12 + s) % n_groups for s in range(12)]
+    st = make_hist_store(fdb, series, nb=nb, groups=groups)
+    start = 100000 + 25 * 15000
+    q = fdb.make_query(start, 30000, start + 300000, 300000,
+                       fdb.FN_HIST_RATE, fdb.AGG_SUM, n_groups)
+    q.param = 0.9
+    _, _, want = oracle.query_exec_hist(st.view(), q, nb)
+    np.testing.assert_allclose(merged, want, rtol=1e-9, equal_nan=True)
