@@ -1,0 +1,86 @@
+"""Reference-literal goldens for the window-sample functions and changes.
+
+Fixtures and expected values lifted verbatim from the reference's own specs:
+  QuantileOverTimeSpec.scala:10-127  (q=0.5 with NaNs → 4928.5; empty → NaN;
+                                      constant → 8201; single value → 8201;
+                                      q=0.2 → 2.8000000000000003; q=0.9 → 9.1)
+  ChangesFunctionSpec.scala:8-77     (normal → 4; empty → NaN; constant → 0;
+                                      NaN-padded constant → 0)
+  AggrOverTimeFunctionsSpec.scala:608-621 (MAD of [9,6,4,1,1,2,2] → 1.0)
+
+Each spec evaluates one window (startTs, endTs]; here that is a one-window
+query with window = endTs - startTs. The "only one value" quantile case
+lists its NaN padding out of timestamp order in the spec; the samples are
+sorted here (same multiset, same window content).
+"""
+import numpy as np
+import pytest
+
+SPEC_SAMPLES = [
+    (8072000, 7419.0), (8082100, np.nan), (8092196, 4614.0),
+    (8102215, 4909.0), (8112223, 4909.0), (8122388, 4948.0),
+    (8132570, np.nan), (8142822, np.nan), (8152858, np.nan),
+    (8162999, 8201.0),
+]
+SPEC_START, SPEC_END = 8071950, 8163070
+RAMP = [(8072000 + [0, 10100, 20196, 30215, 40223, 50388, 60570, 70822,
+                    80858, 90999][i], float(i + 1)) for i in range(10)]
+
+
+def one_window(fdb, oracle, samples, func_id, param=0.0,
+               start=SPEC_START, end=SPEC_END):
+    st = fdb.ChunkStore()
+    sid = st.add_series(0, fdb.COL_GAUGE)
+    ts = np.array([t for t, _ in samples], dtype=np.int64)
+    vs = np.array([v for _, v in samples], dtype=np.float64)
+    st.append(sid, ts, vs)
+    st.seal()
+    q = fdb.make_query(end, 10000, end, end - start, func_id)
+    q.param = param
+    out = oracle.query_exec(st.view(), q, 1, 1)
+    return float(out[0])
+
+
+def test_quantile_over_time_spec_literals(fdb, oracle):
+    FN = fdb.FN_QUANTILE_OVER_TIME
+    assert one_window(fdb, oracle, SPEC_SAMPLES, FN, 0.5) == 4928.5
+    nan_only = [(8082100, np.nan), (8132570, np.nan),
+                (8142822, np.nan), (8152858, np.nan)]
+    assert np.isnan(one_window(fdb, oracle, nan_only, FN, 0.5))
+    const = [(8082100, 8201.0), (8132570, 8201.0),
+             (8142822, 8201.0), (8152858, 8201.0)]
+    assert one_window(fdb, oracle, const, FN, 0.2) == 8201.0
+    single = [(8082100, np.nan), (8132570, 8201.0),
+              (8142822, np.nan), (8152858, np.nan)]
+    assert one_window(fdb, oracle, single, FN, 0.2) == 8201.0
+    assert one_window(fdb, oracle, RAMP, FN, 0.2) == 2.8000000000000003
+    assert one_window(fdb, oracle, RAMP, FN, 0.9) == 9.1
+
+
+def test_changes_spec_literals(fdb, oracle):
+    """ChangesFunctionSpec's normal case expects 4 from the SLIDING function
+    (ChangesOverTimeFunction, AggrOverTimeFunctions.scala:360-395, which
+    carries lastValue ACROSS NaNs). The chunked path this engine implements
+    (DoubleVector.changes, DoubleVector.scala:283-302) sets prev to every
+    value including NaN, so a NaN breaks the chain: the same samples give 2
+    (7419→NaN→4614 and NaN→8201 are not counted; 4614→4909 and 4909→4948
+    are). The NaN-free literals below agree between both variants."""
+    FN = fdb.FN_CHANGES
+    assert one_window(fdb, oracle, SPEC_SAMPLES, FN) == 2.0
+    nan_only = [(8082100, np.nan), (8132570, np.nan),
+                (8142822, np.nan), (8152858, np.nan)]
+    assert np.isnan(one_window(fdb, oracle, nan_only, FN))
+    const = [(8082100, 8201.0), (8132570, 8201.0),
+             (8142822, 8201.0), (8152858, 8201.0)]
+    assert one_window(fdb, oracle, const, FN) == 0.0
+    padded = [(8082100, np.nan), (8132570, 8201.0),
+              (8142822, 8201.0), (8152858, np.nan)]
+    assert one_window(fdb, oracle, padded, FN) == 0.0
+
+
+def test_mad_spec_literal(fdb, oracle):
+    samples = [(100000 + i * 10000, v)
+               for i, v in enumerate([9.0, 6.0, 4.0, 1.0, 1.0, 2.0, 2.0])]
+    got = one_window(fdb, oracle, samples, fdb.FN_MAD_OVER_TIME,
+                     start=70000, end=170000)
+    assert got == pytest.approx(1.0, abs=1e-10)
