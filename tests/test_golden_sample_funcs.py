@@ -67,9 +67,15 @@ def test_changes_spec_literals(fdb, oracle):
     are). The NaN-free literals below agree between both variants."""
     FN = fdb.FN_CHANGES
     assert one_window(fdb, oracle, SPEC_SAMPLES, FN) == 2.0
+    # all-NaN rows: the chunked accumulator zeroes `changes` for ANY touched
+    # chunk range (ChangesChunkedFunctionD, AggrOverTimeFunctions.scala:
+    # 1200-1202), so this is 0; NaN needs a truly empty window (below).
+    # The spec's NaN expectation is again the sliding variant's.
     nan_only = [(8082100, np.nan), (8132570, np.nan),
                 (8142822, np.nan), (8152858, np.nan)]
-    assert np.isnan(one_window(fdb, oracle, nan_only, FN))
+    assert one_window(fdb, oracle, nan_only, FN) == 0.0
+    assert np.isnan(one_window(fdb, oracle, nan_only, FN,
+                               start=8000000, end=8001000))
     const = [(8082100, 8201.0), (8132570, 8201.0),
              (8142822, 8201.0), (8152858, 8201.0)]
     assert one_window(fdb, oracle, const, FN) == 0.0
