@@ -181,3 +181,40 @@ def test_huge_timestamp_jump_falls_back_to_raw(fdb, oracle):
     q2 = fdb.make_query(130000, 15000, 130000, 30000, fdb.FN_SUM_OVER_TIME)
     out2 = oracle.eval_series(st.view(), 0, q2, 1)
     assert out2[0] == 1.5 + 2.5 + 3.5
+
+
+def test_ddv_size_goldens_from_reference_tests(fdb):
+    """Frozen-vector size/type literals from the reference's own vector tests
+    (LongVectorTest.scala:124-136 const DDV = 24 B; :100-109 packed DDV of
+    [0,2,1,4,3] = 28 overhead + 3 data bytes at nbits 4; :172-185 decreasing
+    const DDV = 24 B)."""
+    import struct
+
+    def enc(ts, vs, kind):
+        st = fdb.ChunkStore()
+        sid = st.add_series(0, kind)
+        st.append(sid, np.asarray(ts, dtype=np.int64),
+                  np.asarray(vs, dtype=np.float64))
+        st.seal()
+        return st.chunk(0, 0)[:2]
+
+    # increasing 10s-spaced timestamps → DeltaDeltaConstVector, 24 bytes
+    base = 1_694_700_000_000
+    ts = [base + i * 10000 for i in range(51)]
+    tsb, _ = enc(ts, [0.0] * 51, fdb.COL_GAUGE)
+    assert len(tsb) == 24
+    assert struct.unpack_from("<I", tsb, 4)[0] & 0xFFFF == 0x0608  # const wf
+
+    # integral values [0,2,1,4,3]: slope 0, deltas in [0,4] → nbits 4,
+    # ceil(5*4/8)=3 data bytes after the 28-byte DDV overhead
+    _, vb = enc(ts[:5], [0.0, 2.0, 1.0, 4.0, 3.0], fdb.COL_GAUGE)
+    assert len(vb) == 28 + 3
+    assert struct.unpack_from("<I", vb, 4)[0] & 0xFFFF == 0x0808   # packed wf
+    inner_nbits = vb[26] & 0x3F
+    assert inner_nbits == 4
+
+    # exactly-decreasing integral values → const DDV (all line deltas 0)
+    _, vb2 = enc(ts, [float(10_000_000 - i * 100) for i in range(51)],
+                 fdb.COL_GAUGE)
+    assert len(vb2) == 24
+    assert struct.unpack_from("<i", vb2, 20)[0] == -100            # slope
