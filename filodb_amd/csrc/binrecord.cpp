@@ -293,10 +293,18 @@ extern "C" int64_t fdb_chunkid_start_time(int64_t chunkid) {
   return (int64_t)(((uint64_t)((1LL << 63) ^ chunkid)) >> kStartTimeShift);
 }
 
+extern "C" int32_t fdb_store_is_sealed(const fdb_store_t* s);
+
 extern "C" int32_t fdb_store_persist(const fdb_store_t* s, int32_t sid,
                                      const uint8_t* partkey, int32_t pk_len,
                                      int64_t ingestion_time,
                                      uint8_t* out, int32_t cap, int32_t* out_len) {
+  if (!fdb_store_is_sealed(s)) {
+    // unsealed series may hold buffered rows not yet cut into chunks —
+    // persisting would silently drop them
+    fdb_set_error("persist requires a sealed store");
+    return FDB_ERR_BADARG;
+  }
   int32_t nch = fdb_series_num_chunks(s, sid);
   if (nch < 0) return nch;
   std::vector<uint8_t> buf;

@@ -673,6 +673,7 @@ extern "C" int32_t fdb_store_seal(fdb_store_t* s) {
 }
 
 extern "C" int32_t fdb_store_num_series(const fdb_store_t* s) { return (int32_t)s->series.size(); }
+extern "C" int32_t fdb_store_is_sealed(const fdb_store_t* s) { return s->sealed ? 1 : 0; }
 extern "C" int32_t fdb_series_num_chunks(const fdb_store_t* s, int32_t sid) {
   if (sid < 0 || sid >= (int32_t)s->series.size()) return FDB_ERR_BADARG;
   return (int32_t)s->series[(size_t)sid].chunks.size();

@@ -287,6 +287,12 @@ extern "C" int32_t fdb_gpu_encode_chunks(fdb_engine_t* e,
     fdb_set_error("gpu_encode: histogram columns encode host-side");
     return FDB_ERR_BADARG;
   }
+  for (int c = 0; c < num_chunks; c++) {
+    if (row_offs[c + 1] < row_offs[c] || row_offs[c] < 0) {
+      fdb_set_error("gpu_encode: row_offs must be nonnegative and nondecreasing");
+      return FDB_ERR_BADARG;
+    }
+  }
   const int64_t nrows = row_offs[num_chunks];
   // padded upper bounds: ts ≤ max(DDV 28+4n, raw 8+8n)+align; val ≤ 12+8n
   int64_t need = 0;

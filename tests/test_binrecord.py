@@ -355,3 +355,12 @@ def test_brv2_reader_truncation_fuzz(fdb):
             assert 0 <= n <= 8
         except RuntimeError:
             pass
+
+
+def test_persist_requires_sealed_store(fdb):
+    st = fdb.ChunkStore()
+    sid = st.add_series(0, fdb.COL_GAUGE)
+    st.append(sid, np.array([100000, 115000], dtype=np.int64),
+              np.array([1.0, 2.0]))
+    with pytest.raises(RuntimeError):      # buffered rows would be dropped
+        fdb.persist_series(st, sid, b"pk")
