@@ -177,6 +177,12 @@ def lib():
             ctypes.c_void_p, ctypes.c_int32, ctypes.POINTER(ctypes.c_uint8),
             ctypes.c_int32, ctypes.POINTER(ctypes.c_uint8), ctypes.c_int32,
             ctypes.c_int32, ctypes.c_int64, ctypes.c_int64]
+        L.fdb_series_append_sc.argtypes = [
+            ctypes.c_void_p, ctypes.c_int32, ctypes.POINTER(ctypes.c_int64),
+            _c_double_p, _c_double_p, ctypes.c_int32]
+        L.fdb_query_exec_avg_sc.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(Query),
+            _c_double_p, ctypes.c_int32]
         L.fdb_gpu_encode_chunks.argtypes = [
             ctypes.c_void_p, ctypes.POINTER(ctypes.c_int64), _c_double_p,
             ctypes.POINTER(ctypes.c_int64), ctypes.c_int32, ctypes.c_int32,
@@ -247,6 +253,18 @@ class ChunkStore:
         _check(lib().fdb_series_append(
             self._h, sid, ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
             vals.ctypes.data_as(_c_double_p), len(ts)), "append")
+
+    def append_sc(self, sid, ts, sums, counts):
+        """Downsampled rows: per-row pre-aggregated sum + sample count (the
+        count column rides the chunk's companion slot)."""
+        ts = np.ascontiguousarray(ts, dtype=np.int64)
+        sums = np.ascontiguousarray(sums, dtype=np.float64)
+        counts = np.ascontiguousarray(counts, dtype=np.float64)
+        assert len(ts) == len(sums) == len(counts)
+        _check(lib().fdb_series_append_sc(
+            self._h, sid, ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+            sums.ctypes.data_as(_c_double_p),
+            counts.ctypes.data_as(_c_double_p), len(ts)), "append_sc")
 
     def synth_generate(self, kind, n_series, n_samples, start_ts=100000,
                        step_ms=15000, jitter_ms=250, lam=10.0, reset_p=0.001,
@@ -327,6 +345,18 @@ class Engine:
         if not d:
             raise RuntimeError(f"fdb_dataset_upload failed: {_err()}")
         return Dataset(d)
+
+    def query_avg_sc(self, dataset, q: Query, out=None, on_device=False):
+        """avg over downsampled sum+count columns (AvgWithSumAndCountOverTime):
+        SumOverTime(sum)/SumOverTime(count) per window."""
+        nw = q.num_windows
+        if out is None:
+            out = np.empty(dataset.num_series * nw, dtype=np.float64)
+        ptr = (out.ctypes.data_as(_c_double_p) if isinstance(out, np.ndarray)
+               else ctypes.cast(out.data_ptr(), _c_double_p))
+        _check(lib().fdb_query_exec_avg_sc(self._h, dataset._h, q, ptr,
+                                           1 if on_device else 0), "avg_sc")
+        return out
 
     def query(self, dataset, q: Query, out=None, out_counts=None, on_device=False):
         """Executes one (shard, query). Returns `out` (allocated as numpy when None

@@ -385,8 +385,10 @@ struct Series {
   double   buf_last = -1.7976931348623157e308;  // Double.MinValue
   // histogram column state (FDB_COL_HIST)
   std::vector<uint64_t> buf_hist;     // row-major [rows × num_buckets]
-  std::vector<double> buf_max, buf_min;  // companion columns (otel max/min)
+  std::vector<double> buf_max, buf_min;  // companion columns (otel max/min,
+                                          // or the downsample count column)
   bool     has_mm = false;
+  bool     has_sc = false;            // scalar sum+count (buf_max = counts)
   int32_t  num_buckets = 0;
   double   bucket_first = 0, bucket_mult = 0;
   std::vector<Chunk> chunks;
@@ -532,6 +534,14 @@ static int32_t cut_chunk(fdb_store_t* s, Series& se) {
   } else {
     encode_doubles(se.buf_vals.data(), n, se.col_kind == FDB_COL_COUNTER && se.buf_drop,
                    c.val_bytes);
+    if (se.has_sc) {                  // downsample count column rides max_off
+      if ((int)se.buf_max.size() != n) {
+        fdb_set_error("sum/count columns out of step (append_sc only on this series)");
+        return FDB_ERR_BADARG;
+      }
+      encode_doubles(se.buf_max.data(), n, false, c.max_bytes);
+      se.buf_max.clear();
+    }
   }
   se.chunks.push_back(std::move(c));
   se.buf_ts.clear(); se.buf_vals.clear();
@@ -563,7 +573,10 @@ extern "C" int32_t fdb_series_append_hist(fdb_store_t* s, int32_t sid,
     se.buf_ts.push_back(ts[i]);
     se.buf_hist.insert(se.buf_hist.end(), bucket_values + (size_t)i * num_buckets,
                        bucket_values + (size_t)(i + 1) * num_buckets);
-    if ((int32_t)se.buf_ts.size() >= s->max_rows) cut_chunk(s, se);
+    if ((int32_t)se.buf_ts.size() >= s->max_rows) {
+      int32_t crc = cut_chunk(s, se);
+      if (crc != FDB_OK) return crc;
+    }
   }
   return FDB_OK;
 }
@@ -596,6 +609,29 @@ extern "C" int32_t fdb_series_append_hist_mm(fdb_store_t* s, int32_t sid,
   return FDB_OK;
 }
 
+// Downsampled scalar series: per row a pre-aggregated sum and its sample
+// count (the downsample schema's avg path — AvgWithSumAndCountOverTimeFuncD,
+// AggrOverTimeFunctions.scala:820-860: avg(window) = SumOverTime(sum col) /
+// SumOverTime(count col)). The count column rides the chunk's max_off slot.
+extern "C" int32_t fdb_series_append_sc(fdb_store_t* s, int32_t sid,
+                                        const int64_t* ts, const double* sums,
+                                        const double* counts, int32_t n) {
+  if (sid < 0 || sid >= (int32_t)s->series.size()) { fdb_set_error("bad series id"); return FDB_ERR_BADARG; }
+  Series& se = s->series[(size_t)sid];
+  if (se.col_kind == FDB_COL_HIST) { fdb_set_error("append_sc needs a scalar series"); return FDB_ERR_BADARG; }
+  if (!se.buf_ts.empty() && !se.has_sc) {
+    fdb_set_error("series %d mixes rows with and without counts", sid);
+    return FDB_ERR_BADARG;
+  }
+  se.has_sc = true;
+  for (int32_t i = 0; i < n; i++) {      // row-wise so auto-cut stays aligned
+    se.buf_max.push_back(counts[i]);
+    int32_t rc = fdb_series_append(s, sid, ts + i, sums + i, 1);
+    if (rc != FDB_OK) return rc;
+  }
+  return FDB_OK;
+}
+
 extern "C" int32_t fdb_series_append(fdb_store_t* s, int32_t sid,
                                      const int64_t* ts, const double* vals, int32_t n) {
   if (s->sealed) { fdb_set_error("store is sealed"); return FDB_ERR_BADARG; }
@@ -614,7 +650,10 @@ extern "C" int32_t fdb_series_append(fdb_store_t* s, int32_t sid,
     }
     se.buf_ts.push_back(ts[i]);
     se.buf_vals.push_back(vals[i]);
-    if ((int32_t)se.buf_ts.size() >= s->max_rows) cut_chunk(s, se);
+    if ((int32_t)se.buf_ts.size() >= s->max_rows) {
+      int32_t crc = cut_chunk(s, se);
+      if (crc != FDB_OK) return crc;
+    }
   }
   return FDB_OK;
 }
@@ -628,7 +667,8 @@ extern "C" int32_t fdb_store_seal(fdb_store_t* s) {
   if (s->sealed) return FDB_OK;
   size_t total = 0, nchunks = 0;
   for (auto& se : s->series) {
-    cut_chunk(s, se);
+    int32_t crc = cut_chunk(s, se);
+    if (crc != FDB_OK) return crc;
     for (auto& c : se.chunks) {
       total += (c.ts_bytes.size() + 63 & ~size_t(63)) + (c.val_bytes.size() + 63 & ~size_t(63));
       total += (c.max_bytes.size() + 63 & ~size_t(63)) + (c.min_bytes.size() + 63 & ~size_t(63));
@@ -656,6 +696,8 @@ extern "C" int32_t fdb_store_seal(fdb_store_t* s) {
         c.max_off = off;
         memcpy(s->blob.data() + off, c.max_bytes.data(), c.max_bytes.size());
         off = (off + c.max_bytes.size() + 63) & ~size_t(63);
+      }
+      if (!c.min_bytes.empty()) {
         c.min_off = off;
         memcpy(s->blob.data() + off, c.min_bytes.data(), c.min_bytes.size());
         off = (off + c.min_bytes.size() + 63) & ~size_t(63);
@@ -799,7 +841,8 @@ extern "C" int32_t fdb_synth_generate(fdb_store_t* s, int32_t kind, int32_t n_se
           row += cum_int[(size_t)b];
           se.buf_hist[at + (size_t)b] = row;
         }
-        if ((int32_t)se.buf_ts.size() >= s->max_rows) cut_chunk(s, se);
+        if ((int32_t)se.buf_ts.size() >= s->max_rows)
+          (void)cut_chunk(s, se);   // synth columns are always aligned
         continue;
       }
       double v;
@@ -817,7 +860,8 @@ extern "C" int32_t fdb_synth_generate(fdb_store_t* s, int32_t kind, int32_t n_se
       }
       se.buf_ts.push_back(ts);
       se.buf_vals.push_back(v);
-      if ((int32_t)se.buf_ts.size() >= s->max_rows) cut_chunk(s, se);
+      if ((int32_t)se.buf_ts.size() >= s->max_rows)
+        (void)cut_chunk(s, se);     // synth columns are always aligned
     }
     cut_chunk(s, se);
   }

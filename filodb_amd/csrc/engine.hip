@@ -552,6 +552,7 @@ struct fdb_dataset {
   void* sums;               // ChunkSum[num_chunks] for the streaming walk
   uint64_t *max_off, *min_off;  // hist companion column offsets (0 = absent)
   int has_mm;               // every hist chunk carries max/min columns
+  int has_sc;               // every scalar chunk carries a count column
 };
 
 extern "C" fdb_engine_t* fdb_engine_create(int32_t device) {
@@ -682,6 +683,7 @@ extern "C" fdb_dataset_t* fdb_dataset_upload(fdb_engine_t* e, const fdb_store_t*
   d->fast_ok = !has_hist && max_chunks <= 1 && max_chunk_rows <= 400 &&
                spans_fit_i32;
   d->has_mm = has_hist && any_mm && all_mm;
+  d->has_sc = has_scalar && any_mm && all_mm;   // count column rides max_off
 
   auto upload = [&](void** dst, const void* src, size_t bytes) -> bool {
     if (hipMalloc(dst, bytes ? bytes : 8) != hipSuccess) return false;
@@ -752,8 +754,10 @@ static bool fast_eligible(const fdb_dataset_t* d, const fdb_query_t* q) {
 }
 
 static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
-                           double* dev_out, double* dev_cnt, double* dev_sq) {
-  DirSoA dir{d->ts_off, d->val_off, d->start_time, d->end_time, d->num_rows};
+                           double* dev_out, double* dev_cnt, double* dev_sq,
+                           const uint64_t* val_override = nullptr) {
+  DirSoA dir{d->ts_off, val_override ? val_override : d->val_off,
+             d->start_time, d->end_time, d->num_rows};
   int nw = fdb_num_windows(q);
   if (fast_eligible(d, q) && q->agg_id == AGG_NONE) {
     return fdb_launch_fast_scan(e->stream, d->blob, dir, d->series_first,
@@ -1141,6 +1145,81 @@ extern "C" int32_t fdb_query_exec_hist_mm(fdb_engine_t* e, const fdb_dataset_t* 
 extern "C" int32_t fdb_query_exec(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_query_t* q,
                                   double* out, double* out_counts, int32_t out_on_device) {
   return run_query(e, d, q, out, out_counts, out_on_device, 0, 1, nullptr);
+}
+
+__global__ void avg_div_kernel(double* __restrict__ out,
+                               const double* __restrict__ a,
+                               const double* __restrict__ b, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = a[i] / b[i];
+}
+
+// Downsample avg: AvgWithSumAndCountOverTimeFuncD (AggrOverTimeFunctions.
+// scala:820-860) — per window, SumOverTime of the sum column divided by
+// SumOverTime of the count column (plain IEEE division, each column with the
+// NaN-poison sum semantics). The count column rides the dataset's max_off
+// slots, so the second pass is the SAME fast scan with the directory's value
+// offsets swapped — no new scan code.
+extern "C" int32_t fdb_query_exec_avg_sc(fdb_engine_t* e, const fdb_dataset_t* d,
+                                         const fdb_query_t* q, double* out,
+                                         int32_t out_on_device) {
+  HIP_CHECK(hipSetDevice(e->device));
+  if (!d->has_sc) {
+    fdb_set_error("dataset has no count columns (fdb_series_append_sc)");
+    return FDB_ERR_BADARG;
+  }
+  int nw = fdb_num_windows(q);
+  if (nw <= 0) { fdb_set_error("bad window params"); return FDB_ERR_BADARG; }
+  if (q->window > ((int64_t)1 << 33)) {
+    fdb_set_error("window length > 2^33 ms unsupported");
+    return FDB_ERR_BADARG;
+  }
+  if (q->agg_id != AGG_NONE) {
+    fdb_set_error("avg_sc emits the [series x windows] grid only");
+    return FDB_ERR_BADARG;
+  }
+  fdb_query_t q2 = *q;
+  q2.func_id = FDB_FN_SUM_OVER_TIME;
+  if (!fast_eligible(d, &q2)) {
+    fdb_set_error("avg_sc needs a fast-eligible dataset (single-chunk "
+                  "series, spans inside i32 ms)");
+    return FDB_ERR_BADARG;
+  }
+  const size_t cells = (size_t)d->num_series * (size_t)nw;
+  double *dsum = nullptr, *dcnt = nullptr, *dout = nullptr;
+  int32_t rc = FDB_ERR;
+  if (hipMalloc(&dsum, cells * 8) != hipSuccess) goto done;
+  if (hipMalloc(&dcnt, cells * 8) != hipSuccess) goto done;
+  if (out_on_device) dout = out;
+  else if (hipMalloc(&dout, cells * 8) != hipSuccess) goto done;
+  rc = launch_scan(e, d, &q2, dsum, nullptr, nullptr);
+  if (rc != FDB_OK) goto done;
+  rc = launch_scan(e, d, &q2, dcnt, nullptr, nullptr, d->max_off);
+  if (rc != FDB_OK) goto done;
+  {
+    const int threads = 256;
+    const size_t grid = (cells + threads - 1) / threads;
+    hipLaunchKernelGGL(avg_div_kernel, dim3((uint32_t)grid), dim3(threads), 0,
+                       e->stream, dout, dsum, dcnt, cells);
+    if (hipGetLastError() != hipSuccess) {
+      fdb_set_error("avg_div_kernel launch failed");
+      rc = FDB_ERR;
+      goto done;
+    }
+  }
+  if (hipStreamSynchronize(e->stream) != hipSuccess) { rc = FDB_ERR; goto done; }
+  if (!out_on_device &&
+      hipMemcpy(out, dout, cells * 8, hipMemcpyDeviceToHost) != hipSuccess) {
+    fdb_set_error("avg_sc copy-out failed");
+    rc = FDB_ERR;
+    goto done;
+  }
+  rc = FDB_OK;
+done:
+  (void)hipFree(dsum);
+  (void)hipFree(dcnt);
+  if (!out_on_device) (void)hipFree(dout);
+  return rc;
 }
 
 // count_values cross-series aggregation (CountValuesRowAggregator.scala):

@@ -348,6 +348,19 @@ int32_t fdb_store_add_encoded_chunk(fdb_store_t* s, int32_t series_id,
  * DoubleVector.scala:86-96,457-476). One wavefront per chunk. Outputs land in
  * `out` at the returned per-chunk offsets/lengths; histogram columns stay
  * host-side. */
+/* Downsampled scalar series: pre-aggregated sums with per-row sample counts
+ * (the downsample schema's avg path). fdb_series_append_sc stores the count
+ * column alongside the sum column; fdb_query_exec_avg_sc computes
+ * avg(window) = SumOverTime(sum col) / SumOverTime(count col) per
+ * AvgWithSumAndCountOverTimeFuncD (AggrOverTimeFunctions.scala:820-860),
+ * emitting the [series x windows] grid. */
+int32_t fdb_series_append_sc(fdb_store_t* s, int32_t series_id,
+                             const int64_t* ts, const double* sums,
+                             const double* counts, int32_t n);
+int32_t fdb_query_exec_avg_sc(fdb_engine_t* e, const fdb_dataset_t* d,
+                              const fdb_query_t* q, double* out,
+                              int32_t out_on_device);
+
 int32_t fdb_gpu_encode_chunks(fdb_engine_t* e,
                               const int64_t* ts, const double* vals,
                               const int64_t* row_offs,

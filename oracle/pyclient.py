@@ -242,3 +242,36 @@ def corrected_doubles(vec_bytes, cap=100000):
     if n < 0:
         raise RuntimeError("corrected_doubles failed")
     return out[:n].copy()
+
+
+def query_exec_avg_sc(view, q, num_series, num_windows):
+    """AvgWithSumAndCountOverTimeFuncD composition (AggrOverTimeFunctions.
+    scala:820-860): SumOverTime over the sum column / SumOverTime over the
+    count column, per window — the count column rides each dir entry's
+    max_off slot, so the second pass runs the SAME oracle with the value
+    offsets swapped."""
+    import ctypes as ct
+
+    class DirEntry(ct.Structure):       # fdb_dir_entry_t (chunk_format.h)
+        _fields_ = [("ts_off", ct.c_uint64), ("val_off", ct.c_uint64),
+                    ("start_time", ct.c_int64), ("end_time", ct.c_int64),
+                    ("num_rows", ct.c_int32), ("_pad", ct.c_int32),
+                    ("max_off", ct.c_uint64), ("min_off", ct.c_uint64)]
+
+    nq = type(q)()
+    ct.memmove(ct.byref(nq), ct.byref(q), ct.sizeof(q))
+    nq.func_id = 3                       # FDB_FN_SUM_OVER_TIME
+    sums = query_exec(view, nq, num_series, num_windows)
+    n = int(view.num_chunks)
+    src = ct.cast(view.dir, ct.POINTER(DirEntry))
+    swapped = (DirEntry * n)()
+    for i in range(n):
+        swapped[i] = src[i]
+        if swapped[i].max_off == 0:
+            raise RuntimeError("chunk %d has no count column" % i)
+        swapped[i].val_off = swapped[i].max_off
+    view2 = type(view)()
+    ct.memmove(ct.byref(view2), ct.byref(view), ct.sizeof(view))
+    view2.dir = ct.cast(swapped, ct.c_void_p)
+    counts = query_exec(view2, nq, num_series, num_windows)
+    return sums / counts
