@@ -344,7 +344,9 @@ class Engine:
         d = lib().fdb_dataset_upload(self._h, store._h)
         if not d:
             raise RuntimeError(f"fdb_dataset_upload failed: {_err()}")
-        return Dataset(d)
+        ds = Dataset(d)
+        ds.num_series = store.num_series
+        return ds
 
     def query_avg_sc(self, dataset, q: Query, out=None, on_device=False):
         """avg over downsampled sum+count columns (AvgWithSumAndCountOverTime):
