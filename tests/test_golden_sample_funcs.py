@@ -90,3 +90,28 @@ def test_mad_spec_literal(fdb, oracle):
     got = one_window(fdb, oracle, samples, fdb.FN_MAD_OVER_TIME,
                      start=70000, end=170000)
     assert got == pytest.approx(1.0, abs=1e-10)
+
+
+DELTA_TS = [8072000, 8082100, 8092196, 8102215, 8112223, 8122388, 8132570,
+            8142822, 8152858, 8162999]
+DELTA_VALS = [111.0, 92.0, 103.0, 110.0, 185.0, 39.0, 52.0, 95.0, 7.0, 99.0]
+
+
+def test_rate_over_delta_spec_literals(fdb, oracle):
+    """PeriodicRateFunctionsSpec.scala:26-80 — RateOverDeltaChunkedFunctionD:
+    rate = sum of in-window delta samples / window_ms * 1000."""
+    FN = fdb.FN_RATE_OVER_DELTA
+    samples = list(zip(DELTA_TS, DELTA_VALS))
+    got = one_window(fdb, oracle, samples, FN)
+    want = sum(DELTA_VALS) / (SPEC_END - SPEC_START) * 1000
+    assert got == pytest.approx(want, abs=1e-7)
+    # one-sample window is NOT NaN (:60-68)
+    got1 = one_window(fdb, oracle, samples, FN, start=8101215, end=8103215)
+    assert not np.isnan(got1)
+    assert got1 == pytest.approx(110.0 / 2000 * 1000, abs=1e-7)
+    # flat (non-increasing) delta samples still rate > 0 (:69-79)
+    flat = [(t, 111.0) for t in DELTA_TS]
+    gotf = one_window(fdb, oracle, flat, FN)
+    assert gotf != 0.0
+    assert gotf == pytest.approx(1110.0 / (SPEC_END - SPEC_START) * 1000,
+                                 abs=1e-7)
