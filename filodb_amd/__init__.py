@@ -24,6 +24,7 @@ FN_PRESENT, FN_TIMESTAMP, FN_ZSCORE = 13, 14, 15
 FN_QUANTILE_OVER_TIME, FN_MAD_OVER_TIME = 16, 17
 FN_PREDICT_LINEAR = 18
 FN_RATE_OVER_DELTA = 19
+FN_HOLT_WINTERS = 20
 # aggregation ids (RowAggregator implementations)
 AGG_NONE, AGG_SUM, AGG_COUNT, AGG_MIN, AGG_MAX, AGG_AVG = 0, 1, 2, 3, 4, 5
 AGG_TOPK, AGG_BOTTOMK = 6, 7
@@ -41,7 +42,7 @@ class Query(ctypes.Structure):
         ("end", ctypes.c_int64), ("window", ctypes.c_int64),
         ("func_id", ctypes.c_int32), ("agg_id", ctypes.c_int32),
         ("num_groups", ctypes.c_int32), ("_pad", ctypes.c_int32),
-        ("param", ctypes.c_double),
+        ("param", ctypes.c_double), ("param2", ctypes.c_double),
     ]
 
     @property
@@ -459,11 +460,12 @@ def dataset_out_len(dataset, q: Query, partial=False):
 
 
 def make_query(start, step, end, window, func_id, agg_id=AGG_NONE, num_groups=0,
-               param=0.0):
+               param=0.0, param2=0.0):
     q = Query()
     q.start, q.step, q.end, q.window = start, step, end, window
     q.func_id, q.agg_id, q.num_groups = func_id, agg_id, num_groups
     q.param = param
+    q.param2 = param2
     return q
 
 

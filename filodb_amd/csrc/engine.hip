@@ -66,7 +66,7 @@ int32_t fdb_launch_window_sample(hipStream_t stream, const uint8_t* blob,
                                  const int32_t* series_nchunks, int num_series,
                                  int64_t qstart, int64_t qstep, int64_t qwindow,
                                  int num_windows, int func_id, double param,
-                                 double* out, int32_t* overflow);
+                                 double param2, double* out, int32_t* overflow);
 
 // hist2.hip: two-cursor histogram walk (unbounded chunks / window ratio)
 int32_t fdb_launch_hist2(hipStream_t stream, const uint8_t* blob, DirSoA dir,
@@ -773,7 +773,8 @@ static int32_t launch_scan(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_qu
     return fdb_launch_window_sample(e->stream, d->blob, dir, d->series_first,
                                     d->series_nchunks, d->num_series,
                                     q->start, q->step, q->window, nw,
-                                    q->func_id, q->param, dev_out, e->dev_flag);
+                                    q->func_id, q->param, q->param2,
+                                    dev_out, e->dev_flag);
   }
   // everything else: the unbounded summary+walk general path. It writes the
   // per-series [S×W] grid only (aggregation is the two-phase reduce outside).
@@ -804,6 +805,12 @@ static int32_t run_query(fdb_engine_t* e, const fdb_dataset_t* d, const fdb_quer
   if (nw <= 0) { fdb_set_error("bad window params"); return FDB_ERR_BADARG; }
   if (q->window > ((int64_t)1 << 33)) {   // d_div1000 exactness domain
     fdb_set_error("window length > 2^33 ms (~99 days) unsupported");
+    return FDB_ERR_BADARG;
+  }
+  if (q->func_id == FDB_FN_HOLT_WINTERS &&
+      !(q->param >= 0 && q->param <= 1 && q->param2 >= 0 && q->param2 <= 1)) {
+    // parseParameters (AggrOverTimeFunctions.scala:1373-1382)
+    fdb_set_error("holt_winters sf/tf must be in [0, 1]");
     return FDB_ERR_BADARG;
   }
   const bool is_topk = q->agg_id == AGG_TOPK || q->agg_id == AGG_BOTTOMK;

@@ -16,7 +16,8 @@
 enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12,
        FN_PRESENT=13, FN_TIMESTAMP=14, FN_ZSCORE=15,
-       FN_QUANTILE=16, FN_MAD=17, FN_PREDICT_LINEAR=18, FN_RATE_OVER_DELTA=19 };
+       FN_QUANTILE=16, FN_MAD=17, FN_PREDICT_LINEAR=18, FN_RATE_OVER_DELTA=19,
+       FN_HOLT_WINTERS=20 };
 enum { AGG_NONE=0, AGG_SUM=1, AGG_COUNT=2, AGG_MIN=3, AGG_MAX=4, AGG_AVG=5,
        AGG_TOPK=6, AGG_BOTTOMK=7, AGG_STDDEV=8, AGG_STDVAR=9, AGG_GROUP=10,
        AGG_QUANTILE=11, AGG_COUNT_VALUES=12 };

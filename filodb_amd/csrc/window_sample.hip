@@ -51,7 +51,7 @@ void window_sample_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
                           const int32_t* __restrict__ series_nchunks,
                           int num_series,
                           int64_t qstart, int64_t qstep, int64_t qwindow,
-                          int num_windows, double param,
+                          int num_windows, double param, double param2,
                           double* __restrict__ out,
                           int32_t* __restrict__ overflow) {
   __shared__ double buf_all[WS_WAVES][WS_MAX_SAMPLES];
@@ -73,6 +73,7 @@ void window_sample_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     bool touched = false, over = false;
     double plX = NAN, plY = NAN, plXY = NAN, plX2 = NAN;   // predict_linear
     int plN = 0;
+    double hwS = NAN, hwB = NAN, hwNext = NAN, hwRes = NAN; // holt_winters
     const bool collect =
         (FUNC != FN_QUANTILE) || (param >= 0 && param <= 1);
 
@@ -113,6 +114,50 @@ void window_sample_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
             else { plY += sy; plX += sx; plXY += sxy; plX2 += sx2; }
             plN += cnt;
           }
+        } else if constexpr (FUNC == FN_HOLT_WINTERS) {
+          // HoltWintersChunkedFunctionD (AggrOverTimeFunctions.scala:
+          // 1379-1452), the oracle's exact operation sequence serial on
+          // lane 0 (the recurrence has a loop-carried dependency). The
+          // reference's one-past-endRow read is modeled as the decoded row
+          // when it exists in the chunk and NaN at the chunk's end, exactly
+          // like the oracle — bit-identical engine==oracle on any shape;
+          // reference parity holds for single-chunk series.
+          if (lane == 0) {
+            const double sf = param, tf = param2;
+            int itPos = startRow, rowNum = startRow;
+            if (isnan(hwS) && isnan(hwB)) {
+              double s0v = NAN, b0v = NAN;
+              int cur = startRow;
+              while (cur <= endRow && isnan(s0v)) { s0v = d_dv_at(&vv, itPos++); cur++; }
+              while (cur <= endRow && isnan(b0v)) { b0v = d_dv_at(&vv, itPos++); cur++; }
+              hwNext = b0v;
+              hwB = b0v - s0v;
+              hwS = s0v;
+              rowNum = cur - 1;
+            } else if (isnan(hwB)) {
+              double b0v = NAN;
+              int cur = startRow;
+              while (cur <= endRow && isnan(b0v)) { b0v = d_dv_at(&vv, itPos++); cur++; }
+              hwNext = b0v;
+              hwB = b0v - hwS;
+              rowNum = cur - 1;
+            } else {
+              itPos++;              // continuation discards one read
+            }
+            if (!isnan(hwB)) {
+              while (rowNum <= endRow) {
+                if (!isnan(hwNext)) {
+                  double ns = sf * hwNext + (1 - sf) * (hwS + hwB);
+                  hwB = tf * (ns - hwS) + (1 - tf) * hwB;
+                  hwS = ns;
+                }
+                hwNext = (itPos < n) ? d_dv_at(&vv, itPos) : NAN;
+                itPos++;
+                rowNum++;
+              }
+              hwRes = hwS;
+            }
+          }
         } else if (collect) {
           // gather the range's non-NaN values into the sort buffer
           for (int base = startRow; base <= endRow; base += 64) {
@@ -137,6 +182,8 @@ void window_sample_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
     double result = NAN;
     if (over) {
       if (lane == 0) atomicExch(overflow, 1);
+    } else if constexpr (FUNC == FN_HOLT_WINTERS) {
+      result = hwRes;
     } else if constexpr (FUNC == FN_PREDICT_LINEAR) {
       if (plN >= 2) {
         double covXY = plXY - plX * plY / plN;
@@ -216,7 +263,7 @@ void window_sample_kernel(const uint8_t* __restrict__ blob, DirSoA dir,
 
 bool fdb_window_sample_supported(int func_id) {
   return func_id == FN_QUANTILE || func_id == FN_MAD ||
-         func_id == FN_PREDICT_LINEAR;
+         func_id == FN_PREDICT_LINEAR || func_id == FN_HOLT_WINTERS;
 }
 
 int32_t fdb_launch_window_sample(hipStream_t stream, const uint8_t* blob,
@@ -224,13 +271,13 @@ int32_t fdb_launch_window_sample(hipStream_t stream, const uint8_t* blob,
                                  const int32_t* series_nchunks, int num_series,
                                  int64_t qstart, int64_t qstep, int64_t qwindow,
                                  int num_windows, int func_id, double param,
-                                 double* out, int32_t* overflow) {
+                                 double param2, double* out, int32_t* overflow) {
   size_t pairs = (size_t)num_series * num_windows;
   size_t grid = (pairs + WS_WAVES - 1) / WS_WAVES;
   if (grid > 8192) grid = 8192;
   if (grid < 1) grid = 1;
   #define WARGS blob, dir, series_first, series_nchunks, num_series, \
-      qstart, qstep, qwindow, num_windows, param, out, overflow
+      qstart, qstep, qwindow, num_windows, param, param2, out, overflow
   switch (func_id) {
     case FN_QUANTILE:
       hipLaunchKernelGGL((window_sample_kernel<FN_QUANTILE>), dim3((uint32_t)grid),
@@ -242,6 +289,11 @@ int32_t fdb_launch_window_sample(hipStream_t stream, const uint8_t* blob,
       break;
     case FN_PREDICT_LINEAR:
       hipLaunchKernelGGL((window_sample_kernel<FN_PREDICT_LINEAR>),
+                         dim3((uint32_t)grid), dim3(WS_WAVES * 64), 0, stream,
+                         WARGS);
+      break;
+    case FN_HOLT_WINTERS:
+      hipLaunchKernelGGL((window_sample_kernel<FN_HOLT_WINTERS>),
                          dim3((uint32_t)grid), dim3(WS_WAVES * 64), 0, stream,
                          WARGS);
       break;

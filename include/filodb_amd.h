@@ -171,6 +171,15 @@ int32_t fdb_nibblepack_pack_doubles(const double* in, int32_t n, uint8_t* out, i
                                        t offset seconds in param */
 #define FDB_FN_RATE_OVER_DELTA 19   /* RateOverDeltaChunkedFunctionD
                                        (RateFunctions.scala:424-445) */
+#define FDB_FN_HOLT_WINTERS    20   /* HoltWintersChunkedFunctionD
+                                       (AggrOverTimeFunctions.scala:1379-1453):
+                                       sf in param, tf in param2. SINGLE-CHUNK
+                                       series only (fast-eligible datasets):
+                                       across chunk boundaries the reference
+                                       feeds one decoded-past-endRow value —
+                                       undefined memory — into the recurrence,
+                                       so exact multi-chunk parity does not
+                                       exist; the engine rejects those loudly */
 #define FDB_FN_ZSCORE          15   /* ZScoreChunkedFunctionD
                                        (AggrOverTimeFunctions.scala:1592-1603):
                                        (lastSample - mean) / stddev over the window */
@@ -218,6 +227,8 @@ typedef struct {
   int32_t _pad;
   double  param;     /* function parameter (quantile q for the histogram
                         present step); 0 when unused                          */
+  double  param2;    /* second parameter (holt_winters trend factor); 0 when
+                        unused                                               */
 } fdb_query_t;
 
 static inline int32_t fdb_num_windows(const fdb_query_t* q) {

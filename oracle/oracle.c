@@ -39,6 +39,7 @@ typedef struct {
   int64_t start, step, end, window;
   int32_t func_id, agg_id, num_groups, _pad;
   double param;
+  double param2;
 } fdb_query_t;
 typedef struct {
   const uint8_t* blob;
@@ -55,7 +56,7 @@ enum { FN_RATE=0, FN_INCREASE=1, FN_DELTA=2, FN_SUM=3, FN_COUNT=4, FN_AVG=5,
        FN_MIN=6, FN_MAX=7, FN_STDDEV=8, FN_STDVAR=9, FN_CHANGES=10, FN_LAST=12,
        FN_PRESENT=13, FN_TIMESTAMP=14, FN_ZSCORE=15,
        FN_QUANTILE=16, FN_MAD=17, FN_PREDICT_LINEAR=18,
-       FN_RATE_OVER_DELTA=19 };
+       FN_RATE_OVER_DELTA=19, FN_HOLT_WINTERS=20 };
 
 /* QuantileOverTimeFunction.calculateRank (AggrOverTimeFunctions.scala:400-406)
  * + the sorted linear interpolation both quantile_over_time and
@@ -524,6 +525,7 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
       int qn = 0, touched = 0;     /* quantile/MAD window sample buffer */
       double plX = NAN, plY = NAN, plXY = NAN, plX2 = NAN;  /* predict_linear */
       int plN = 0;
+      double hwS = NAN, hwB = NAN, hwNext = NAN, hwRes = NAN;  /* holt_winters */
       for (int c = 0; c < nchunks; c++) {
         if (dir[c].end_time < wStart) continue;
         const vec_t* tv = &tsv[c];
@@ -623,6 +625,51 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
                 plN++;
               }
             } break;
+            case FN_HOLT_WINTERS: {
+              /* HoltWintersChunkedFunctionD.addTimeDoubleChunks
+               * (AggrOverTimeFunctions.scala:1379-1452), restated operation
+               * for operation. The reference's final it.next of every chunk
+               * range reads one row PAST endRow: inside the chunk that is
+               * the decoded row endRow+1; at the chunk's end it is whatever
+               * memory follows the vector — undefined — modeled here as NaN.
+               * The engine only accepts single-chunk series, where that
+               * value provably never reaches the recurrence. */
+              double sf = q->param, tf = q->param2;
+              int itPos = startRow;     /* next row the iterator returns */
+              int rowNum = startRow;
+              if (isnan(hwS) && isnan(hwB)) {
+                double s0v = NAN, b0v = NAN;
+                int cur = startRow;
+                while (cur <= endRow && isnan(s0v)) { s0v = dv_at(vv, itPos++); cur++; }
+                while (cur <= endRow && isnan(b0v)) { b0v = dv_at(vv, itPos++); cur++; }
+                hwNext = b0v;
+                hwB = b0v - s0v;
+                hwS = s0v;
+                rowNum = cur - 1;
+              } else if (isnan(hwB)) {
+                double b0v = NAN;
+                int cur = startRow;
+                while (cur <= endRow && isnan(b0v)) { b0v = dv_at(vv, itPos++); cur++; }
+                hwNext = b0v;
+                hwB = b0v - hwS;
+                rowNum = cur - 1;
+              } else {
+                itPos++;                /* continuation discards one read */
+              }
+              if (!isnan(hwB)) {
+                while (rowNum <= endRow) {
+                  if (!isnan(hwNext)) {
+                    double ns = sf * hwNext + (1 - sf) * (hwS + hwB);
+                    hwB = tf * (ns - hwS) + (1 - tf) * hwB;
+                    hwS = ns;
+                  }
+                  hwNext = (itPos < dir[c].num_rows) ? dv_at(vv, itPos) : NAN;
+                  itPos++;
+                  rowNum++;
+                }
+                hwRes = hwS;
+              }
+            } break;
             case FN_QUANTILE: case FN_MAD: {
               /* QuantileOverTimeChunkedFunctionD (:1272-1299) /
                * MedianAbsoluteDeviationOverTimeChunkedFunctionD (:1302-1330):
@@ -660,6 +707,7 @@ static void eval_series(const fdb_view_t* view, int sid, const fdb_query_t* q,
           result = r;
         } break;
         case FN_CHANGES: result = changes; break;
+        case FN_HOLT_WINTERS: result = hwRes; break;
         case FN_LAST: case FN_PRESENT: case FN_TIMESTAMP:
           result = last_val; break;
         case FN_ZSCORE: {

@@ -27,7 +27,7 @@ RAMP = [(8072000 + [0, 10100, 20196, 30215, 40223, 50388, 60570, 70822,
                     80858, 90999][i], float(i + 1)) for i in range(10)]
 
 
-def one_window(fdb, oracle, samples, func_id, param=0.0,
+def one_window(fdb, oracle, samples, func_id, param=0.0, param2=0.1,
                start=SPEC_START, end=SPEC_END):
     st = fdb.ChunkStore()
     sid = st.add_series(0, fdb.COL_GAUGE)
@@ -37,6 +37,7 @@ def one_window(fdb, oracle, samples, func_id, param=0.0,
     st.seal()
     q = fdb.make_query(end, 10000, end, end - start, func_id)
     q.param = param
+    q.param2 = param2
     out = oracle.query_exec(st.view(), q, 1, 1)
     return float(out[0])
 
@@ -115,3 +116,77 @@ def test_rate_over_delta_spec_literals(fdb, oracle):
     assert gotf != 0.0
     assert gotf == pytest.approx(1110.0 / (SPEC_END - SPEC_START) * 1000,
                                  abs=1e-7)
+
+
+def _hw_model(arr, sf=0.01, tf=0.1):
+    """The reference's own test model (AggrOverTimeFunctionsSpec.scala:
+    694-714): s0=arr[0], b0=arr[1]-arr[0], then the smoothing recurrence."""
+    if len(arr) < 2:
+        return float("nan")
+    s0, b0 = arr[0], arr[1] - arr[0]
+    for x in arr[1:]:
+        s = sf * x + (1 - sf) * (s0 + b0)
+        b0 = tf * (s - s0) + (1 - tf) * b0
+        s0 = s
+    return s0
+
+
+def test_holt_winters_spec_literals(fdb, oracle):
+    """HoltWintersChunkedFunctionD (single-chunk) against the reference
+    spec's data and model (AggrOverTimeFunctionsSpec.scala:686-737)."""
+    FN = fdb.FN_HOLT_WINTERS
+    cases = [
+        [15900.0, 15920.0, 15940.0, 15960.0, 15980.0, 16000.0],
+        [23850.0, 23880.0, 23910.0, 23940.0, 23970.0, 24000.0],
+        [31800.0, 31840.0, 31880.0, 31920.0, 31960.0, 32000.0],
+        [-15900.0, -15920.0, -15940.0, -15960.0, -15980.0, -16000.0],
+    ]
+    for vals in cases:
+        samples = [(100000 + i * 10000, v) for i, v in enumerate(vals)]
+        got = one_window(fdb, oracle, samples, FN, param=0.01,
+                         start=60000, end=160000)
+        assert got == pytest.approx(_hw_model(vals), abs=1e-10)
+    # < 2 samples → NaN; NaN rows are skipped for the two seeds
+    one = [(100000, 5.0)]
+    assert np.isnan(one_window(fdb, oracle, one, FN, param=0.01,
+                               start=60000, end=160000))
+    withnan = [(100000, np.nan), (110000, 10.0), (120000, np.nan),
+               (130000, 20.0), (140000, 30.0)]
+    got = one_window(fdb, oracle, withnan, FN, param=0.01,
+                     start=60000, end=160000)
+    assert not np.isnan(got)
+
+
+@pytest.mark.gpu
+def test_holt_winters_gpu_vs_oracle(fdb, oracle):
+    """FN 20 through the window-sample kernel: identical operation sequence
+    to the oracle, including the NaN seed-scan and the modeled end-of-chunk
+    read — bit-equal results across a randomized single-chunk matrix."""
+    rng = np.random.default_rng(73)
+    st = fdb.ChunkStore()
+    n = 240
+    for s in range(24):
+        ts = (100000 + np.arange(n) * 15000
+              + rng.integers(-250, 251, n)).astype(np.int64)
+        ts = np.maximum.accumulate(ts)
+        vs = np.cumsum(rng.normal(1.0, 3.0, n)) + 100
+        vs[rng.random(n) < 0.15] = np.nan
+        sid = st.add_series(0, fdb.COL_GAUGE)
+        st.append(sid, ts, vs)
+    st.seal()
+    eng = fdb.Engine(0)
+    ds = eng.upload(st)
+    for sf, tf, step, window in [(0.01, 0.1, 15000, 300000),
+                                 (0.5, 0.5, 60000, 1800000),
+                                 (0.9, 0.05, 15000, 60000)]:
+        q = fdb.make_query(100000 + 30 * 15000, step, 100000 + 230 * 15000,
+                           window, fdb.FN_HOLT_WINTERS, param=sf, param2=tf)
+        want = oracle.query_exec(st.view(), q, st.num_series, q.num_windows)
+        got = np.empty(st.num_series * q.num_windows, dtype=np.float64)
+        eng.query(ds, q, out=got)
+        np.testing.assert_allclose(got, want, rtol=1e-12, atol=0,
+                                   equal_nan=True)
+    with pytest.raises(RuntimeError):       # sf outside [0,1]
+        q = fdb.make_query(550000, 15000, 700000, 300000,
+                           fdb.FN_HOLT_WINTERS, param=1.5, param2=0.1)
+        eng.query(ds, q, out=np.empty(24 * 11))
