@@ -190,3 +190,60 @@ def test_holt_winters_gpu_vs_oracle(fdb, oracle):
         q = fdb.make_query(550000, 15000, 700000, 300000,
                            fdb.FN_HOLT_WINTERS, param=1.5, param2=0.1)
         eng.query(ds, q, out=np.empty(24 * 11))
+
+
+def _hw_chunked(ts, vs, ws, we, sf, tf):
+    """Independent Python restatement of HoltWintersChunkedFunctionD for ONE
+    single-chunk window (rows [s..e] = first ts >= ws .. last ts <= we):
+    NaN seed scans, the double-counted second seed, the one-past-endRow read
+    (in-chunk row or the modeled NaN at the chunk's end)."""
+    n = len(ts)
+    s = next((i for i in range(n) if ts[i] >= ws), n)
+    e = n - 1
+    while e >= 0 and ts[e] > we:
+        e -= 1
+    if s > e:
+        return float("nan")
+    itp = s
+    s0v = float("nan")
+    cur = s
+    while cur <= e and np.isnan(s0v):
+        s0v = vs[itp]; itp += 1; cur += 1
+    b0v = float("nan")
+    while cur <= e and np.isnan(b0v):
+        b0v = vs[itp]; itp += 1; cur += 1
+    nxt, b0, s0 = b0v, b0v - s0v, s0v
+    row = cur - 1
+    res = float("nan")
+    if not np.isnan(b0):
+        while row <= e:
+            if not np.isnan(nxt):
+                ns = sf * nxt + (1 - sf) * (s0 + b0)
+                b0 = tf * (ns - s0) + (1 - tf) * b0
+                s0 = ns
+            nxt = vs[itp] if itp < n else float("nan")
+            itp += 1
+            row += 1
+        res = s0
+    return res
+
+
+def test_holt_winters_nan_fuzz(fdb, oracle):
+    """Randomized NaN patterns and window boundaries: the oracle equals an
+    independent Python restatement of the chunked function bit for bit."""
+    rng = np.random.default_rng(79)
+    for trial in range(25):
+        n = int(rng.integers(2, 120))
+        ts = (100000 + np.arange(n) * 10000).astype(np.int64)
+        vs = np.round(rng.normal(100, 20, n), 6)
+        vs[rng.random(n) < rng.choice([0.0, 0.2, 0.6])] = np.nan
+        sf = float(rng.uniform(0.01, 0.99))
+        tf = float(rng.uniform(0.01, 0.99))
+        end = int(ts[0] + rng.integers(1, n + 3) * 10000)
+        win = int(rng.integers(1, n + 3) * 10000)
+        samples = list(zip(ts.tolist(), vs.tolist()))
+        got = one_window(fdb, oracle, samples, fdb.FN_HOLT_WINTERS,
+                         param=sf, param2=tf, start=end - win, end=end)
+        want = _hw_chunked(ts, vs, end - win, end, sf, tf)
+        assert (np.isnan(got) and np.isnan(want)) or got == want, \
+            (trial, got, want)
