@@ -356,11 +356,13 @@ extern "C" int32_t fdb_store_restore(fdb_store_t* s, fdb_brv2_index_t* ix,
                                      const uint8_t* bytes, int32_t len,
                                      int32_t col_kind, int32_t* out_rows) {
   int32_t pos = 0, nrows = 0;
-  auto need = [&](int32_t n) { return pos + n <= len; };
+  // 64-bit bound check: field lengths come from the (untrusted) stream, so
+  // a huge u32 must not wrap the 32-bit sum into an in-bounds value
+  auto need = [&](int64_t n) { return n >= 0 && (int64_t)pos + n <= (int64_t)len; };
   while (pos < len) {
     if (!need(4)) { fdb_set_error("truncated row header"); return FDB_ERR_BADARG; }
     uint32_t pl = rd_u32b(bytes + pos); pos += 4;
-    if (!need((int32_t)pl + 8 + 4)) { fdb_set_error("truncated partkey"); return FDB_ERR_BADARG; }
+    if (!need((int64_t)pl + 8 + 4)) { fdb_set_error("truncated partkey"); return FDB_ERR_BADARG; }
     std::string key((const char*)bytes + pos, pl); pos += pl;
     int64_t cid = rd_i64b(bytes + pos); pos += 8;
     uint32_t il = rd_u32b(bytes + pos); pos += 4;
@@ -373,10 +375,10 @@ extern "C" int32_t fdb_store_restore(fdb_store_t* s, fdb_brv2_index_t* ix,
     if (nc != 2) { fdb_set_error("expected 2 column blobs, got %d", nc); return FDB_ERR_BADARG; }
     if (!need(4)) return FDB_ERR_BADARG;
     uint32_t tl = rd_u32b(bytes + pos); pos += 4;
-    if (!need((int32_t)tl + 4)) { fdb_set_error("truncated ts blob"); return FDB_ERR_BADARG; }
+    if (!need((int64_t)tl + 4)) { fdb_set_error("truncated ts blob"); return FDB_ERR_BADARG; }
     const uint8_t* tsb = bytes + pos; pos += tl;
     uint32_t vl = rd_u32b(bytes + pos); pos += 4;
-    if (!need((int32_t)vl)) { fdb_set_error("truncated value blob"); return FDB_ERR_BADARG; }
+    if (!need((int64_t)vl)) { fdb_set_error("truncated value blob"); return FDB_ERR_BADARG; }
     const uint8_t* vab = bytes + pos; pos += vl;
     auto it = ix->by_pk.find(key);
     int32_t sid;

@@ -1195,6 +1195,7 @@ extern "C" int32_t fdb_query_exec_avg_sc(fdb_engine_t* e, const fdb_dataset_t* d
   const size_t cells = (size_t)d->num_series * (size_t)nw;
   double *dsum = nullptr, *dcnt = nullptr, *dout = nullptr;
   int32_t rc = FDB_ERR;
+  fdb_set_error("avg_sc: device allocation failed");
   if (hipMalloc(&dsum, cells * 8) != hipSuccess) goto done;
   if (hipMalloc(&dcnt, cells * 8) != hipSuccess) goto done;
   if (out_on_device) dout = out;
