@@ -387,3 +387,25 @@ def test_persist_roundtrip_idempotent(fdb):
     st2.seal()
     rows2 = fdb.persist_series(st2, 0, b"pk", ingestion_time=555)
     assert rows2 == rows1
+
+
+def test_restore_hostile_lengths(fdb):
+    """Huge u32 field lengths in the stream must not wrap the bound checks
+    (they used to overflow int32 into an in-bounds value)."""
+    st = fdb.ChunkStore()
+    ix = fdb.BRv2Index()
+    hostile = struct.pack("<I", 0xFFFFFFF0) + b"\x00" * 64
+    with pytest.raises(RuntimeError):
+        fdb.restore_rows(st, ix, hostile)
+    st2 = fdb.ChunkStore()
+    sid = st2.add_series(0, fdb.COL_GAUGE)
+    ts = np.arange(100000, 100000 + 20 * 15000, 15000, dtype=np.int64)
+    st2.append(sid, ts, np.arange(20, dtype=np.float64))
+    st2.seal()
+    rows = bytearray(fdb.persist_series(st2, sid, b"pk"))
+    # corrupt the ts-blob length to a huge value
+    pl = struct.unpack_from("<I", rows, 0)[0]
+    off = 4 + pl + 8 + 4 + 28 + 2
+    struct.pack_into("<I", rows, off, 0xFFFFFF00)
+    with pytest.raises(RuntimeError):
+        fdb.restore_rows(fdb.ChunkStore(), fdb.BRv2Index(), bytes(rows))
