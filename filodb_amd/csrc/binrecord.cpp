@@ -171,7 +171,8 @@ static int32_t walk_container(const uint8_t* bytes, int32_t len, F&& cb) {
   int32_t count = 0;
   while (cur < end) {
     int32_t rec_len = (int32_t)rd_u32b(bytes + cur);
-    if (rec_len < kVarAreaStart - 4 || cur + 4 + rec_len > end) {
+    if (rec_len < kVarAreaStart - 4 ||
+        (int64_t)cur + 4 + (int64_t)rec_len > (int64_t)end) {
       fdb_set_error("bad record at +%d (len %d)", cur, rec_len);
       return FDB_ERR_BADARG;
     }
@@ -197,6 +198,11 @@ extern "C" int32_t fdb_brv2_read(const uint8_t* bytes, int32_t len, int32_t idx,
     if (part_hash) *part_hash = (int32_t)rd_u32b(rec + kHashOffset);
     uint32_t pk_start = rd_u32b(rec + kOffMetric);      // first part field
     uint32_t pk_end = (uint32_t)rlen;                   // var area runs to end
+    if (pk_start < (uint32_t)kVarAreaStart || pk_start > pk_end) {
+      // untrusted offset: must stay inside the record's var area
+      fdb_set_error("bad metric offset %u in record %d", pk_start, i);
+      return FDB_ERR_BADARG;
+    }
     if (pk_len) *pk_len = (int32_t)(pk_end - pk_start);
     if (pk_out) {
       if ((int32_t)(pk_end - pk_start) > pk_cap) { fdb_set_error("pk buffer too small"); return FDB_ERR_BADARG; }

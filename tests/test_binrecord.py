@@ -409,3 +409,14 @@ def test_restore_hostile_lengths(fdb):
     struct.pack_into("<I", rows, off, 0xFFFFFF00)
     with pytest.raises(RuntimeError):
         fdb.restore_rows(fdb.ChunkStore(), fdb.BRv2Index(), bytes(rows))
+
+
+def test_brv2_read_hostile_offset(fdb):
+    """A corrupted metric-offset field must not drive the part-key copy out
+    of bounds."""
+    b = fdb.BRv2Builder()
+    b.add(100000, 1.0, "m", {"k": "v"})
+    c = bytearray(b.finish())
+    struct.pack_into("<I", c, 16 + 22, 0xFFFFFFF0)   # record's metric offset
+    with pytest.raises(RuntimeError):
+        fdb.brv2_read(bytes(c), 0)
