@@ -745,6 +745,40 @@ extern "C" int32_t fdb_chunk_get(const fdb_store_t* s, int32_t sid, int32_t ci, 
   return FDB_OK;
 }
 
+// structural check of one frozen scalar vector: the length word matches the
+// buffer, the wireformat is known, and the decoded element count covers
+// num_rows — so restored bytes can never drive the GPU decoders out of
+// bounds. (Histogram vectors carry their own counts and are checked by the
+// upload guard.)
+static bool frozen_vec_ok(const uint8_t* b, int32_t len, int32_t num_rows,
+                          int allow_hist) {
+  if (len < 8) return false;
+  uint32_t lw; memcpy(&lw, b, 4);
+  if ((int64_t)lw + 4 != (int64_t)len) return false;
+  uint16_t wf; memcpy(&wf, b + 4, 2);
+  if (wf == FDB_WF_HIST_SECTDELTA) return allow_hist != 0;
+  if (wf == FDB_WF_DDV_CONST) {
+    if (len < FDB_DDVC_BYTES) return false;
+    int32_t n; memcpy(&n, b + FDB_DDVC_OFF_NELEM, 4);
+    return n >= num_rows;
+  }
+  if (wf == FDB_WF_DDV) {
+    if (len < FDB_DDV_OFF_INNER + FDB_PRIM_OFF_DATA) return false;
+    const uint8_t* inner = b + FDB_DDV_OFF_INNER;
+    uint32_t ilw; memcpy(&ilw, inner, 4);
+    if ((int64_t)FDB_DDV_OFF_INNER + 4 + ilw > (int64_t)len) return false;
+    int nbits = inner[6] & FDB_NBITS_MASK;
+    if (nbits != 2 && nbits != 4 && nbits != 8 && nbits != 16 && nbits != 32)
+      return false;
+    int bitShift = inner[7] & 0x3f;
+    int64_t n = (((int64_t)ilw - 4) * 8 + (bitShift ? bitShift - 8 : 0)) / nbits;
+    return n >= num_rows;
+  }
+  if (wf == FDB_WF_PRIM64)
+    return ((int64_t)lw - 4) / 8 >= num_rows;
+  return false;
+}
+
 // Paging-side restore (TimeSeriesChunksTable read path): appends a chunk whose
 // frozen vector bytes come from persisted storage UNCHANGED — no re-encode, the
 // same bytes the ODP reader hands to the query engine.
@@ -756,6 +790,11 @@ extern "C" int32_t fdb_store_add_encoded_chunk(fdb_store_t* s, int32_t sid,
   if (s->sealed) { fdb_set_error("store is sealed"); return FDB_ERR_BADARG; }
   if (sid < 0 || sid >= (int32_t)s->series.size()) { fdb_set_error("bad series id %d", sid); return FDB_ERR_BADARG; }
   if (ts_len < 8 || val_len < 8 || num_rows < 1) { fdb_set_error("bad encoded chunk"); return FDB_ERR_BADARG; }
+  if (!frozen_vec_ok(ts_bytes, ts_len, num_rows, 0) ||
+      !frozen_vec_ok(val_bytes, val_len, num_rows, 1)) {
+    fdb_set_error("malformed frozen vector bytes (series %d)", sid);
+    return FDB_ERR_BADARG;
+  }
   Series& se = s->series[(size_t)sid];
   if (!se.buf_ts.empty()) { fdb_set_error("series %d has unsealed buffered rows", sid); return FDB_ERR_BADARG; }
   if (!se.chunks.empty() && start_time < se.chunks.back().end_time) {

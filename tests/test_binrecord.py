@@ -420,3 +420,28 @@ def test_brv2_read_hostile_offset(fdb):
     struct.pack_into("<I", c, 16 + 22, 0xFFFFFFF0)   # record's metric offset
     with pytest.raises(RuntimeError):
         fdb.brv2_read(bytes(c), 0)
+
+
+def test_add_encoded_chunk_rejects_malformed_vectors(fdb):
+    """Restored vector bytes are structurally validated before any GPU
+    decoder sees them: bad length word, unknown wireformat, element count
+    short of num_rows."""
+    st = fdb.ChunkStore()
+    sid = st.add_series(0, fdb.COL_GAUGE)
+    ts = np.arange(100000, 100000 + 20 * 15000, 15000, dtype=np.int64)
+    st2 = fdb.ChunkStore()
+    s2 = st2.add_series(0, fdb.COL_GAUGE)
+    st2.append(s2, ts, np.arange(20, dtype=np.float64))
+    st2.seal()
+    tb, vb, n, t0, t1 = st2.chunk(0, 0)
+    fdb.add_encoded_chunk(st, sid, tb, vb, n, t0, t1)      # well-formed: OK
+    bad_len = bytearray(tb)
+    struct.pack_into("<I", bad_len, 0, len(tb) + 100)      # length word lies
+    with pytest.raises(RuntimeError):
+        fdb.add_encoded_chunk(st, sid, bytes(bad_len), vb, n, t1 + 1, t1 + 2)
+    bad_wf = bytearray(tb)
+    struct.pack_into("<H", bad_wf, 4, 0x1234)              # unknown wireformat
+    with pytest.raises(RuntimeError):
+        fdb.add_encoded_chunk(st, sid, bytes(bad_wf), vb, n, t1 + 1, t1 + 2)
+    with pytest.raises(RuntimeError):                      # count < num_rows
+        fdb.add_encoded_chunk(st, sid, tb, vb, n + 50, t1 + 1, t1 + 2)
