@@ -101,3 +101,31 @@ def test_gpu_avg_sc_rejects_plain_dataset(fdb):
     q = fdb.make_query(400000, 15000, 700000, 300000, fdb.FN_AVG_OVER_TIME)
     with pytest.raises(RuntimeError):
         eng.query_avg_sc(eng.upload(st), q)
+
+
+def test_oracle_avg_sc_multichunk(fdb, oracle):
+    """The oracle composition handles chunked series (both sum passes walk
+    chunks); the engine currently restricts avg_sc to single-chunk datasets,
+    so this pins the semantics a stream-walk extension must match."""
+    rng = np.random.default_rng(83)
+    st = fdb.ChunkStore()
+    st.set_max_rows(90)                  # 240 rows → 3 chunks per series
+    data = []
+    for _ in range(8):
+        ts = (100000 + np.arange(240) * 15000
+              + rng.integers(-250, 251, 240)).astype(np.int64)
+        ts = np.maximum.accumulate(ts)
+        sums = rng.normal(50, 20, 240) * rng.integers(1, 30, 240)
+        counts = rng.integers(1, 30, 240).astype(np.float64)
+        sums[rng.random(240) < 0.1] = np.nan
+        sid = st.add_series(0, fdb.COL_GAUGE)
+        st.append_sc(sid, ts, sums, counts)
+        data.append((ts, sums, counts))
+    st.seal()
+    assert st.num_chunks(0) == 3
+    q = fdb.make_query(100000 + 25 * 15000, 15000, 100000 + 235 * 15000,
+                       300000, fdb.FN_AVG_OVER_TIME)
+    got = oracle.query_exec_avg_sc(st.view(), q, st.num_series, q.num_windows)
+    want = naive_avg(data, q)
+    np.testing.assert_allclose(got, want, rtol=1e-9, atol=1e-12,
+                               equal_nan=True)
