@@ -247,3 +247,63 @@ def test_holt_winters_nan_fuzz(fdb, oracle):
         want = _hw_chunked(ts, vs, end - win, end, sf, tf)
         assert (np.isnan(got) and np.isnan(want)) or got == want, \
             (trial, got, want)
+
+
+def test_nan_sequence_matrix_spec(fdb, oracle):
+    """AggrOverTimeFunctionsSpec.scala:914-965 — sum/avg/stdvar/stddev/
+    zscore/present over the spec's NaN-pattern sequences, against its own
+    NaN-skipping models (window (60000, 160000] over 10s-spaced samples)."""
+    nan = float("nan")
+    test_data = [
+        [15900.0, 15920.0, 15940.0, 15960.0, 15980.0, 16000.0, 16020.0],
+        [-15900.0, -15920.0, -15940.0, -15960.0, -15980.0, -16000.0],
+        [15900.0, 15920.0, 15940.0, 15960.0, 15980.0, 16000.0, nan],
+        [23850.0, 23880.0, 23910.0, 23940.0, 23970.0, 24000.0],
+        [31800.0, 31840.0, 31880.0, 31920.0, 31960.0, 32000.0],
+        [31800.0, 31840.0, 31880.0, nan, 31920.0, 31960.0, 32000.0],
+        [nan, 31800.0, 31840.0, 31880.0, 31920.0, 31960.0, 32000.0],
+        [nan] * 7,
+        [],
+    ]
+    for vals in test_data:
+        samples = [(100000 + i * 10000, v) for i, v in enumerate(vals)]
+        if not samples:
+            samples = []
+        clean = [v for v in vals if not np.isnan(v)]
+        n = len(clean)
+        exp_sum = sum(clean) if n else nan
+        exp_avg = (exp_sum / n) if n else nan
+        exp_var = (sum((x - exp_avg) ** 2 for x in clean) / n) if n else nan
+        # zscore uses the reference's lastSample rule: set only when the
+        # range's LAST row is non-NaN (VarOverTimeChunkedFunctionD:1103) —
+        # a NaN-tailed window emits NaN. (The reference additionally never
+        # resets lastSample between windows, so a long-lived iterator can
+        # leak a previous window's sample into a NaN-tailed window; the
+        # spec's own assertions accept NaN there, and this stateless engine
+        # always emits NaN — DESIGN.md §9.)
+        exp_z = (((vals[-1] - exp_avg) / np.sqrt(exp_var))
+                 if n and not np.isnan(vals[-1]) else nan)
+
+        def run(fn):
+            if not samples:       # truly empty series
+                st = fdb.ChunkStore()
+                st.add_series(0, fdb.COL_GAUGE)
+                st.seal()
+                q = fdb.make_query(160000, 10000, 160000, 100000, fn)
+                return float(oracle.query_exec(st.view(), q, 1, 1)[0])
+            return one_window(fdb, oracle, samples, fn,
+                              start=60000, end=160000)
+
+        for fn, want in [(fdb.FN_SUM_OVER_TIME, exp_sum),
+                         (fdb.FN_AVG_OVER_TIME, exp_avg),
+                         (fdb.FN_STDVAR_OVER_TIME, exp_var),
+                         (fdb.FN_STDDEV_OVER_TIME,
+                          np.sqrt(exp_var) if n else nan),
+                         (fdb.FN_ZSCORE, exp_z)]:
+            got = run(fn)
+            if np.isnan(want):
+                assert np.isnan(got), (vals, fn, got)
+            else:
+                assert got == pytest.approx(want, rel=1e-9), (vals, fn, got)
+        if n:
+            assert run(fdb.FN_PRESENT) == 1.0
