@@ -227,3 +227,63 @@ def test_two_shard_hist_quantile_merge_equals_global(fdb, oracle):
     q.param = 0.9
     _, _, want = oracle.query_exec_hist(st.view(), q, nb)
     np.testing.assert_allclose(merged, want, rtol=1e-9, equal_nan=True)
+
+
+def _worker_w4(rank, result_queue):
+    import sys
+    sys.path.insert(0, REPO)
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import filodb_amd as fdb
+    import pyclient as oracle
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29523"
+    dist.init_process_group("gloo", rank=rank, world_size=4)
+
+    n_groups = 4
+    st, _ = _make_shard(fdb, rank, n_groups)
+    start = 100000 + 20 * 10000
+    q = fdb.make_query(start, 30000, start + 200000, 100000,
+                       fdb.FN_SUM_OVER_TIME, fdb.AGG_SUM, n_groups)
+    nw = q.num_windows
+    sums, counts = oracle.query_exec(st.view(), q, st.num_series, nw,
+                                     out_counts=True)
+    t_sum = torch.from_numpy(sums)
+    t_cnt = torch.from_numpy(counts)
+    dist.all_reduce(t_sum)
+    dist.all_reduce(t_cnt)
+    merged = torch.where(t_cnt > 0, t_sum,
+                         torch.full_like(t_sum, float("nan")))
+    if rank == 0:
+        result_queue.put(merged.numpy())
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_four_shard_sum_merge_equals_global(fdb, oracle):
+    """world_size=4: the same one-collective merge at a deeper reduce tree."""
+    ctx = mp.get_context("spawn")
+    queue = ctx.Queue()
+    procs = [ctx.Process(target=_worker_w4, args=(r, queue)) for r in range(4)]
+    for p in procs:
+        p.start()
+    merged = queue.get(timeout=240)
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+
+    n_groups = 4
+    series, groups = [], []
+    for rank in range(4):
+        rng = np.random.default_rng(1000 + rank)
+        for s in range(20):
+            ts, vs = synth_gauge_series(rng, 50, step=10000, jitter=400,
+                                        nan_p=0.1)
+            series.append([[(int(t), float(v)) for t, v in zip(ts, vs)]])
+            groups.append((rank * 20 + s) % n_groups)
+    st = build_store(fdb, series, groups=groups)
+    start = 100000 + 20 * 10000
+    q = fdb.make_query(start, 30000, start + 200000, 100000,
+                       fdb.FN_SUM_OVER_TIME, fdb.AGG_SUM, n_groups)
+    expected = oracle.query_exec(st.view(), q, st.num_series, q.num_windows)
+    np.testing.assert_allclose(merged, expected, rtol=1e-9, equal_nan=True)
