@@ -11,7 +11,8 @@
  *                   tools/verify_div1000.c -lm && ./verify_div1000
  * Last run (this container, 2026-09-15): bad=0 of 8589934593; the naive
  * q0-only reciprocal differs on ~13% of them (sanity that the harness can
- * detect mismatches).
+ * detect mismatches). The negative domain -2^33..0 was verified explicitly
+ * as well (bad=0) — not just by the sign-symmetry argument.
  */
 #include <stdio.h>
 #include <math.h>
